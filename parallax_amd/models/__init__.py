@@ -1,0 +1,8 @@
+"""Model zoo: one module per architecture family, auto-registered for the shard
+loader (parity with the reference's EntryClass autodiscovery, shard_loader.py:79)."""
+
+from .registry import MODEL_REGISTRY, get_model_class, register_model
+
+from . import llama  # noqa: F401  (registers LlamaForCausalLM / Qwen2ForCausalLM)
+from . import qwen3  # noqa: F401
+from . import qwen3_moe  # noqa: F401

@@ -1,0 +1,285 @@
+"""Llama-family decoder (covers LlamaForCausalLM and Qwen2ForCausalLM — the
+flagship DeepSeek-R1-Distill-Llama-8B is this architecture).
+
+Fresh MI355X-first design (reference analogue: src/parallax/models/llama.py /
+qwen2.py): RMSNorm / fused-add-RMSNorm, RoPE, KV scatter, paged attention and
+SwiGLU all route through parallax_amd.ops (HIP kernels on GPU); QKV and gate/up
+are fused single GEMMs (hipBLASLt); TP via column/row-parallel layers with RCCL
+all-reduce over xGMI.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..parallel.comm import get_comm
+from ..parallel.layers import (
+    ColumnParallelLinear,
+    MergedColumnParallelLinear,
+    RowParallelLinear,
+    VocabEmbedding,
+)
+from .config import ModelConfig
+from .forward_meta import ForwardMeta
+from .registry import register_model
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden_size: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(hidden_size), requires_grad=False)
+        self.eps = eps
+
+    def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor] = None):
+        if residual is None:
+            return ops.rmsnorm(x, self.weight, self.eps)
+        return ops.fused_add_rmsnorm(x, residual, self.weight, self.eps)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int, local_layer_idx: int):
+        super().__init__()
+        comm = get_comm()
+        self.layer_idx = layer_idx
+        self.local_layer_idx = local_layer_idx
+        self.head_dim = cfg.head_dim
+        self.num_heads = cfg.num_heads // comm.tp_size
+        self.num_kv_heads = max(1, cfg.num_kv_heads // comm.tp_size)
+        self.scale = self.head_dim ** -0.5
+        self.sliding_window = (
+            cfg.sliding_window if cfg.layer_type(layer_idx) == "sliding_attention" else -1
+        )
+        self.qk_norm = cfg.qk_norm
+        h = cfg.hidden_size
+        self.qkv_proj = MergedColumnParallelLinear(
+            h,
+            [cfg.num_heads * cfg.head_dim, cfg.num_kv_heads * cfg.head_dim,
+             cfg.num_kv_heads * cfg.head_dim],
+            bias=cfg.attention_bias,
+        )
+        self.o_proj = RowParallelLinear(
+            cfg.num_heads * cfg.head_dim, h, bias=cfg.o_proj_bias
+        )
+        if self.qk_norm:
+            self.q_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
+            self.k_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
+
+    def forward(self, x: torch.Tensor, meta: ForwardMeta, rope_cache: torch.Tensor):
+        T = x.shape[0]
+        q, k, v = self.qkv_proj.split_output(self.qkv_proj(x))
+        q = q.view(T, self.num_heads, self.head_dim)
+        k = k.view(T, self.num_kv_heads, self.head_dim)
+        v = v.view(T, self.num_kv_heads, self.head_dim)
+        if self.qk_norm:
+            q = self.q_norm(q)
+            k = self.k_norm(k)
+        q = q.contiguous()
+        k = k.contiguous()
+        ops.rope_inplace(q, k, meta.positions, rope_cache)
+
+        k_cache, v_cache = meta.kv_cache.layer(self.local_layer_idx)
+        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+
+        if meta.is_prefill:
+            attn = ops.prefill_attention(
+                q, k_cache, v_cache, meta.block_tables, meta.seq_lens,
+                meta.query_lens, self.scale, self.sliding_window,
+            )
+        else:
+            attn = ops.paged_attention_decode(
+                q, k_cache, v_cache, meta.block_tables, meta.seq_lens,
+                self.scale, self.sliding_window,
+            )
+        return self.o_proj(attn.reshape(T, self.num_heads * self.head_dim))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, intermediate_size: Optional[int] = None):
+        super().__init__()
+        inter = intermediate_size or cfg.intermediate_size
+        self.gate_up_proj = MergedColumnParallelLinear(
+            cfg.hidden_size, [inter, inter], bias=cfg.mlp_bias
+        )
+        self.down_proj = RowParallelLinear(inter, cfg.hidden_size, bias=cfg.mlp_bias)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int, local_layer_idx: int):
+        super().__init__()
+        self.self_attn = LlamaAttention(cfg, layer_idx, local_layer_idx)
+        self.mlp = LlamaMLP(cfg)
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+
+    def forward(self, x, residual, meta: ForwardMeta, rope_cache):
+        if residual is None:
+            residual = x
+            x = self.input_layernorm(x)
+        else:
+            x, residual = self.input_layernorm(x, residual)
+        x = self.self_attn(x, meta, rope_cache)
+        x, residual = self.post_attention_layernorm(x, residual)
+        x = self.mlp(x)
+        return x, residual
+
+
+@register_model("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM")
+class LlamaShardModel(nn.Module):
+    """A contiguous layer range [start_layer, end_layer) of a Llama-family model.
+    First shard owns the embedding; last shard owns final norm + lm_head
+    (reference base_executor.py:124-125 stage-role convention)."""
+
+    decoder_layer_cls = LlamaDecoderLayer
+
+    def __init__(self, cfg: ModelConfig, start_layer: int = 0, end_layer: Optional[int] = None):
+        super().__init__()
+        self.cfg = cfg
+        self.start_layer = start_layer
+        self.end_layer = end_layer if end_layer is not None else cfg.num_layers
+        self.is_first = start_layer == 0
+        self.is_last = self.end_layer == cfg.num_layers
+
+        if self.is_first:
+            self.embed_tokens = VocabEmbedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(
+            [
+                self.decoder_layer_cls(cfg, g, i)
+                for i, g in enumerate(range(start_layer, self.end_layer))
+            ]
+        )
+        if self.is_last:
+            self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+            self.lm_head = ColumnParallelLinear(cfg.hidden_size, cfg.vocab_size,
+                                                gather_output=True)
+        self.register_buffer(
+            "rope_cache",
+            ops.build_rope_cache(
+                cfg.max_position_embeddings,
+                cfg.head_dim,
+                cfg.rope_theta,
+                scaling_factor=_rope_scaling_factor(cfg),
+            ),
+            persistent=False,
+        )
+
+    @property
+    def num_local_layers(self) -> int:
+        return len(self.layers)
+
+    def embed(self, input_ids: torch.Tensor) -> torch.Tensor:
+        assert self.is_first, "only the first shard embeds tokens"
+        return self.embed_tokens(input_ids)
+
+    def forward(self, hidden: torch.Tensor, meta: ForwardMeta) -> torch.Tensor:
+        """hidden: [T, hidden] — embeddings on the first shard, the previous
+        stage's output elsewhere. Returns [T, hidden] post final-norm on the
+        last shard, pre-norm activations otherwise."""
+        residual = None
+        for layer in self.layers:
+            hidden, residual = layer(hidden, residual, meta, self.rope_cache)
+        if self.is_last:
+            hidden, _ = self.norm(hidden, residual)
+        else:
+            hidden = hidden + residual
+        return hidden
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        assert self.is_last, "only the last shard computes logits"
+        return self.lm_head(hidden.to(self.lm_head.weight.dtype)).float()
+
+    # -- weight loading ---------------------------------------------------------
+
+    _STACKED = {
+        "q_proj": ("qkv_proj", 0),
+        "k_proj": ("qkv_proj", 1),
+        "v_proj": ("qkv_proj", 2),
+        "gate_proj": ("gate_up_proj", 0),
+        "up_proj": ("gate_up_proj", 1),
+    }
+
+    def map_global_layer(self, name: str) -> Optional[str]:
+        """'model.layers.<g>.rest' -> 'layers.<local>.rest' (None if out of range);
+        parity with the reference shard loader's key remap (shard_loader.py:229)."""
+        if name.startswith("model.layers."):
+            parts = name.split(".")
+            g = int(parts[2])
+            if not (self.start_layer <= g < self.end_layer):
+                return None
+            return ".".join(["layers", str(g - self.start_layer)] + parts[3:])
+        if name.startswith("model.embed_tokens."):
+            if self.is_first:
+                return name.replace("model.", "", 1)
+            if self.is_last and self.cfg.tie_word_embeddings:
+                return name.replace("model.embed_tokens", "lm_head")
+            return None
+        if name.startswith("model.norm."):
+            return name.replace("model.", "", 1) if self.is_last else None
+        if name.startswith("lm_head."):
+            return name if self.is_last and not self.cfg.tie_word_embeddings else None
+        return None
+
+    def load_hf_weight(self, name: str, tensor: torch.Tensor) -> bool:
+        """Route one HF-named tensor into this shard. Returns True if consumed."""
+        local = self.map_global_layer(name)
+        if local is None:
+            return False
+        parts = local.split(".")
+        leaf = parts[-1]  # 'weight' or 'bias'
+        stem = parts[-2]
+        tensor = tensor.to(torch.bfloat16)
+        if stem in self._STACKED:
+            target_name, idx = self._STACKED[stem]
+            module = self._resolve(parts[:-2] + [target_name])
+            module.load_full_weight_part(idx, tensor if leaf == "weight" else None,
+                                         tensor if leaf == "bias" else None)
+            return True
+        module = self._resolve(parts[:-1])
+        if isinstance(module, (ColumnParallelLinear, RowParallelLinear)):
+            module.load_full_weight(tensor if leaf == "weight" else None,
+                                    tensor if leaf == "bias" else None)
+        else:
+            getattr(module, leaf).data.copy_(tensor)
+        return True
+
+    def _resolve(self, path_parts):
+        mod = self
+        for p in path_parts:
+            mod = mod[int(p)] if p.isdigit() else getattr(mod, p)
+        return mod
+
+    @torch.no_grad()
+    def init_random(self, seed: int = 1234) -> None:
+        """Random-init all parameters (synthetic-weight benchmarking; no network
+        for checkpoints). Norm weights -> 1, linears -> N(0, 0.02/sqrt(2L))."""
+        g = torch.Generator().manual_seed(seed)
+        std = 0.02 / math.sqrt(2 * max(1, self.cfg.num_layers))
+        for name, p in self.named_parameters():
+            if "norm" in name.lower():
+                p.data.fill_(1.0)
+            elif name.endswith("bias"):
+                p.data.zero_()
+            elif "embed" in name or "lm_head" in name:
+                p.data.copy_(
+                    torch.randn(p.shape, generator=g, dtype=torch.float32).mul_(0.02)
+                    .to(p.dtype)
+                )
+            else:
+                p.data.copy_(
+                    torch.randn(p.shape, generator=g, dtype=torch.float32).mul_(std)
+                    .to(p.dtype)
+                )
+
+
+def _rope_scaling_factor(cfg: ModelConfig) -> float:
+    rs = cfg.rope_scaling or {}
+    if rs.get("rope_type") == "linear" or rs.get("type") == "linear":
+        return float(rs.get("factor", 1.0))
+    return 1.0

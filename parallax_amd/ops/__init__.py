@@ -1,0 +1,259 @@
+"""Op dispatch: hand-written HIP/CDNA4 kernels on GPU, eager torch on CPU.
+
+Policy (per the MI355X-first design): on a CUDA (= ROCm/HIP) device the compiled
+extension `parallax_amd._C` is REQUIRED for the core hot ops — if it is missing
+the op raises instead of silently falling back to eager torch, so a GPU run can
+never "pass" on a Python fallback. On CPU the fp32 reference implementations run
+(plumbing tests, no GPU in CI).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Tuple
+
+import torch
+
+from . import reference as ref
+from ..utils.logging_config import get_logger
+
+logger = get_logger("ops")
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _try_load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from . import _C  # built in-tree by setup.py build_ext --inplace
+
+        _EXT = _C
+        logger.info("loaded HIP extension parallax_amd.ops._C")
+    except ImportError as e:  # pragma: no cover - exercised only without build
+        _EXT_ERR = str(e)
+    return _EXT
+
+
+def has_extension() -> bool:
+    return _try_load_extension() is not None
+
+
+def _require_ext(op_name: str):
+    ext = _try_load_extension()
+    if ext is None:
+        raise RuntimeError(
+            f"parallax_amd HIP extension is required for {op_name} on GPU but is not "
+            f"built (import error: {_EXT_ERR}). Build it with "
+            f"`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950)."
+        )
+    return ext
+
+
+# -- normalization ------------------------------------------------------------
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("rmsnorm")
+        out = torch.empty_like(x)
+        ext.rmsnorm(out, x.contiguous(), weight, eps)
+        return out
+    return ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Returns (normed, new_residual). On GPU this mutates x (normed) and
+    residual (sum) in place to avoid two extra HBM round trips."""
+    if x.is_cuda:
+        ext = _require_ext("fused_add_rmsnorm")
+        ext.fused_add_rmsnorm(x, residual, weight, eps)
+        return x, residual
+    return ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+# -- rotary -------------------------------------------------------------------
+
+build_rope_cache = ref.build_rope_cache
+
+
+def rope_inplace(
+    q: torch.Tensor,
+    k: Optional[torch.Tensor],
+    positions: torch.Tensor,
+    cos_sin: torch.Tensor,
+    is_neox: bool = True,
+) -> None:
+    if q.is_cuda:
+        ext = _require_ext("rope")
+        ext.rope_inplace(
+            q,
+            k if k is not None else q[:0],
+            positions.to(torch.int32),
+            cos_sin,
+            is_neox,
+        )
+        return
+    ref.rope_inplace(q, k, positions, cos_sin, is_neox)
+
+
+# -- cache scatter ---------------------------------------------------------------
+
+
+def reshape_and_cache(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if k.is_cuda:
+        ext = _require_ext("reshape_and_cache")
+        ext.reshape_and_cache(
+            k.contiguous(), v.contiguous(), k_cache, v_cache, slot_mapping.to(torch.int64)
+        )
+        return
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def mla_reshape_and_cache(
+    kv_latent: torch.Tensor,
+    k_rope: torch.Tensor,
+    cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if kv_latent.is_cuda:
+        ext = _require_ext("mla_reshape_and_cache")
+        ext.mla_reshape_and_cache(
+            kv_latent.contiguous(), k_rope.contiguous(), cache, slot_mapping.to(torch.int64)
+        )
+        return
+    ref.mla_reshape_and_cache(kv_latent, k_rope, cache, slot_mapping)
+
+
+# -- attention ---------------------------------------------------------------------
+
+
+def paged_attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+    sliding_window: int = -1,
+    softcap: float = 0.0,
+    sinks: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    if q.is_cuda:
+        ext = _require_ext("paged_attention_decode")
+        out = torch.empty_like(q)
+        ext.paged_attention_decode(
+            out,
+            q.contiguous(),
+            k_cache,
+            v_cache,
+            block_tables.to(torch.int32),
+            seq_lens.to(torch.int32),
+            scale,
+            sliding_window,
+            softcap,
+            sinks if sinks is not None else q.new_empty(0),
+        )
+        return out
+    return ref.paged_attention_decode(
+        q, k_cache, v_cache, block_tables, seq_lens, scale, sliding_window, softcap, sinks
+    )
+
+
+def prefill_attention(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    query_lens: torch.Tensor,
+    scale: float,
+    sliding_window: int = -1,
+    softcap: float = 0.0,
+    sinks: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    if q.is_cuda:
+        ext = _require_ext("prefill_attention")
+        out = torch.empty_like(q)
+        # varlen metadata: cumulative query offsets
+        cu_q = torch.zeros(len(query_lens) + 1, dtype=torch.int32, device=q.device)
+        torch.cumsum(query_lens.to(torch.int32), dim=0, out=cu_q[1:])
+        ext.prefill_attention(
+            out,
+            q.contiguous(),
+            k_cache,
+            v_cache,
+            block_tables.to(torch.int32),
+            seq_lens.to(torch.int32),
+            cu_q,
+            scale,
+            sliding_window,
+            softcap,
+            sinks if sinks is not None else q.new_empty(0),
+        )
+        return out
+    return ref.prefill_attention(
+        q, k_cache, v_cache, block_tables, seq_lens, query_lens, scale,
+        sliding_window, softcap, sinks,
+    )
+
+
+def mla_paged_attention_decode(
+    q_latent: torch.Tensor,
+    q_rope: torch.Tensor,
+    cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if q_latent.is_cuda:
+        ext = _require_ext("mla_paged_attention_decode")
+        out = torch.empty_like(q_latent)
+        ext.mla_paged_attention_decode(
+            out, q_latent.contiguous(), q_rope.contiguous(), cache,
+            block_tables.to(torch.int32), seq_lens.to(torch.int32), scale,
+        )
+        return out
+    return ref.mla_paged_attention_decode(
+        q_latent, q_rope, cache, block_tables, seq_lens, scale
+    )
+
+
+# -- activations ----------------------------------------------------------------------
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("silu_and_mul")
+        half = x.shape[-1] // 2
+        out = x.new_empty(*x.shape[:-1], half)
+        ext.silu_and_mul(out, x.contiguous())
+        return out
+    return ref.silu_and_mul(x)
+
+
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("gelu_and_mul")
+        half = x.shape[-1] // 2
+        out = x.new_empty(*x.shape[:-1], half)
+        ext.gelu_and_mul(out, x.contiguous())
+        return out
+    return ref.gelu_and_mul(x)
+
+
+# -- sampling (torch ops; GPU path uses torch's ROCm kernels — not a hot spot
+#    relative to the model forward, custom kernel is a later optimization) -------
+
+sample_tokens = ref.sample_tokens
+apply_penalties = ref.apply_penalties

@@ -1,0 +1,416 @@
+"""Single-host pipeline-parallel serving engine (the xGMI fast path).
+
+Design: SPMD replicated scheduling. Every pipeline stage runs an identical
+continuous-batching scheduler on identical request state; only hidden states
+move point-to-point (RCCL send/recv over xGMI) and sampled tokens are broadcast
+once per step. This removes the reference's per-hop serialize+RPC cost
+(SURVEY.md §3.3 — decode pays a Lattica RPC of (1, hidden) per token per hop)
+for the in-host path; the packet-driven peer executor (p2p/) remains the
+multi-host path.
+
+Replicated determinism: admission, batch formation, cache allocation and LRU
+eviction are all single-threaded and driven by the same event sequence on every
+rank, so per-rank block tables stay identical without any wire traffic.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..models import get_model_class
+from ..models.config import ModelConfig
+from ..models.forward_meta import ForwardMeta
+from ..parallel.comm import CommContext, get_comm
+from ..utils.logging_config import get_logger
+from .cache_manager import CacheManager
+from .kv_cache import (
+    KVCacheSpec,
+    PagedKVCache,
+    build_block_table_tensor,
+    slot_mapping_for_positions,
+)
+from .request import InitialRequest, RequestStatus, new_request_id
+from .sampler import Sampler
+from .sampling_params import SamplingParams
+from .scheduler import PrefillChunk, ScheduledBatch, Scheduler
+
+logger = get_logger("server.engine")
+
+
+@dataclass
+class StepOutput:
+    rid: str
+    token_id: int
+    finished: bool
+    finish_reason: Optional[str] = None
+
+
+@dataclass
+class EngineArgs:
+    block_size: int = 32
+    max_batch_size: int = 128
+    max_num_tokens_per_batch: int = 16384
+    prefill_chunk_size: int = 8192
+    cache_memory_fraction: float = 0.80
+    num_kv_blocks: Optional[int] = None  # override (tests / CPU)
+    micro_batches: int = 1
+    enable_prefix_cache: bool = True
+    dtype: torch.dtype = torch.bfloat16
+    seed: Optional[int] = 0
+    request_timeout_s: float = 600.0
+    start_layer: Optional[int] = None   # explicit layer range (decentralized mode)
+    end_layer: Optional[int] = None
+
+
+def partition_layers(num_layers: int, pp_size: int, pp_rank: int) -> Tuple[int, int]:
+    """Contiguous near-even split; earlier stages get the remainder (they also
+    host the embedding, but on 288 GB the difference is noise)."""
+    base, rem = divmod(num_layers, pp_size)
+    start = pp_rank * base + min(pp_rank, rem)
+    return start, start + base + (1 if pp_rank < rem else 0)
+
+
+class Engine:
+    def __init__(
+        self,
+        cfg: ModelConfig,
+        args: EngineArgs = EngineArgs(),
+        comm: Optional[CommContext] = None,
+        model_path: Optional[str] = None,
+        random_weights: bool = False,
+    ):
+        self.cfg = cfg
+        self.args = args
+        self.comm = comm or get_comm()
+        self.device = self.comm.device
+
+        if args.start_layer is not None:
+            start, end = args.start_layer, args.end_layer
+        else:
+            start, end = partition_layers(cfg.num_layers, self.comm.pp_size, self.comm.pp_rank)
+        self.start_layer, self.end_layer = start, end
+
+        model_cls = get_model_class(cfg.architecture)
+        with torch.device("meta"):
+            pass  # (meta-init then materialize is a later memory optimization)
+        self.model = model_cls(cfg, start, end)
+        if random_weights:
+            self.model.init_random()
+        elif model_path is not None:
+            from .shard_loader import load_shard_weights
+
+            load_shard_weights(self.model, model_path)
+        self.model = self.model.to(device=self.device, dtype=args.dtype)
+        # rope cache stays fp32
+        self.model.rope_cache = self.model.rope_cache.float()
+        self.model.eval()
+
+        spec = KVCacheSpec(
+            num_layers=end - start,
+            num_kv_heads=max(1, cfg.num_kv_heads // self.comm.tp_size),
+            head_dim=cfg.head_dim,
+            block_size=args.block_size,
+            dtype=args.dtype,
+        )
+        num_blocks = args.num_kv_blocks or CacheManager.num_blocks_from_memory(
+            spec, self.device, args.cache_memory_fraction
+        )
+        self.kv_cache = PagedKVCache(spec, num_blocks, self.device)
+        self.cache_manager = CacheManager(
+            args.block_size, num_blocks, enable_prefix_cache=args.enable_prefix_cache
+        )
+        self.scheduler = Scheduler(
+            self.cache_manager,
+            max_batch_size=args.max_batch_size,
+            max_num_tokens_per_batch=args.max_num_tokens_per_batch,
+            prefill_chunk_size=args.prefill_chunk_size,
+            request_timeout_s=args.request_timeout_s,
+            eos_token_ids=cfg.eos_token_ids,
+        )
+        self.sampler = Sampler(self.device, args.seed)
+        self._pending_adds: List[InitialRequest] = []
+        self._pending_aborts: List[str] = []
+        self.step_count = 0
+        logger.info(
+            "engine up: layers [%d,%d) of %d, %d KV blocks x %d tokens (%.2f GiB), device %s",
+            start, end, cfg.num_layers, num_blocks, args.block_size,
+            num_blocks * spec.bytes_per_block() / (1 << 30), self.device,
+        )
+
+    # -- public API (head rank) ----------------------------------------------------
+
+    def submit(
+        self,
+        prompt_token_ids: List[int],
+        sampling_params: Optional[SamplingParams] = None,
+        rid: Optional[str] = None,
+    ) -> str:
+        req = InitialRequest(
+            rid=rid or new_request_id(),
+            prompt_token_ids=list(prompt_token_ids),
+            sampling_params=sampling_params or SamplingParams(),
+        )
+        self._pending_adds.append(req)
+        return req.rid
+
+    def abort(self, rid: str) -> None:
+        self._pending_aborts.append(rid)
+
+    @property
+    def has_work(self) -> bool:
+        return self.scheduler.has_work or bool(self._pending_adds)
+
+    # -- the step ---------------------------------------------------------------------
+
+    def step(self) -> List[StepOutput]:
+        """One engine iteration on every rank. Returns newly sampled tokens
+        (meaningful on the head rank; identical on all ranks)."""
+        self._sync_ingress()
+        self.scheduler.sweep_aborted()
+        if self.step_count % 64 == 0:
+            self.scheduler.sweep_timeouts()
+        self.scheduler.admit_requests()
+        batch = self.scheduler.form_batch()
+        self.step_count += 1
+        if batch.is_empty:
+            return []
+
+        outputs: List[StepOutput] = []
+        sample_reqs: List[InitialRequest] = []
+        logits_parts: List[torch.Tensor] = []
+
+        if batch.prefill_chunks:
+            logits, samp = self._run_prefill(batch.prefill_chunks)
+            if samp:
+                sample_reqs.extend(samp)
+                if logits is not None:
+                    logits_parts.append(logits)
+        if batch.decode_reqs:
+            logits = self._run_decode(batch.decode_reqs)
+            sample_reqs.extend(batch.decode_reqs)
+            if logits is not None:
+                logits_parts.append(logits)
+
+        # bookkeeping that must precede token commit
+        for chunk in batch.prefill_chunks:
+            self.scheduler.complete_prefill_chunk(chunk)
+
+        if sample_reqs:
+            token_ids = self._sample_and_broadcast(logits_parts, sample_reqs)
+            for req, tok in zip(sample_reqs, token_ids):
+                finished = self.scheduler.commit_token(req.rid, tok)
+                outputs.append(
+                    StepOutput(
+                        rid=req.rid,
+                        token_id=tok,
+                        finished=finished is not None,
+                        finish_reason=req.status.finish_reason,
+                    )
+                )
+        return outputs
+
+    # -- ingress replication ----------------------------------------------------------
+
+    def _sync_ingress(self) -> None:
+        if self.comm.world_size > 1:
+            if self.comm.rank == 0:
+                payload = [
+                    [
+                        (r.rid, r.prompt_token_ids, r.sampling_params.to_dict())
+                        for r in self._pending_adds
+                    ],
+                    list(self._pending_aborts),
+                ]
+            else:
+                payload = [None, None]
+            dist.broadcast_object_list(payload, src=0)
+            if self.comm.rank != 0:
+                adds, aborts = payload
+                self._pending_adds = [
+                    InitialRequest(
+                        rid=rid,
+                        prompt_token_ids=toks,
+                        sampling_params=SamplingParams.from_dict(sp),
+                    )
+                    for rid, toks, sp in adds
+                ]
+                self._pending_aborts = list(aborts)
+        for req in self._pending_adds:
+            self.scheduler.add_request(req)
+        for rid in self._pending_aborts:
+            self.scheduler.abort_request(rid)
+        self._pending_adds = []
+        self._pending_aborts = []
+
+    # -- forward passes ------------------------------------------------------------------
+
+    def _build_prefill_meta(self, chunks: List[PrefillChunk]) -> Tuple[ForwardMeta, torch.Tensor]:
+        positions, input_ids, slot_mapping = [], [], []
+        block_tables, seq_lens, query_lens, logits_idx = [], [], [], []
+        t = 0
+        for c in chunks:
+            state = self.cache_manager.get(c.req.rid)
+            positions.extend(range(c.start, c.start + c.num_tokens))
+            input_ids.extend(c.req.prompt_token_ids[c.start : c.start + c.num_tokens])
+            slot_mapping.extend(
+                slot_mapping_for_positions(
+                    state.block_table, c.start, c.num_tokens, self.args.block_size
+                )
+            )
+            block_tables.append(state.block_table)
+            seq_lens.append(c.start + c.num_tokens)
+            query_lens.append(c.num_tokens)
+            t += c.num_tokens
+            logits_idx.append(t - 1)
+        dev = self.device
+        meta = ForwardMeta(
+            is_prefill=True,
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
+            block_tables=build_block_table_tensor(block_tables, dev),
+            seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
+            query_lens=torch.tensor(query_lens, dtype=torch.int32, device=dev),
+            kv_cache=self.kv_cache,
+            logits_indices=torch.tensor(logits_idx, dtype=torch.int64, device=dev),
+        )
+        ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
+        return meta, ids
+
+    def _build_decode_meta(self, reqs: List[InitialRequest]) -> Tuple[ForwardMeta, torch.Tensor]:
+        positions, input_ids, slot_mapping, block_tables, seq_lens = [], [], [], [], []
+        for r in reqs:
+            state = self.cache_manager.get(r.rid)
+            pos = r.total_len - 1  # position of the token being fed in
+            positions.append(pos)
+            input_ids.append(r.output_token_ids[-1])
+            slot_mapping.extend(
+                slot_mapping_for_positions(state.block_table, pos, 1, self.args.block_size)
+            )
+            block_tables.append(state.block_table)
+            seq_lens.append(r.total_len)
+        dev = self.device
+        meta = ForwardMeta(
+            is_prefill=False,
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
+            block_tables=build_block_table_tensor(block_tables, dev),
+            seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
+            kv_cache=self.kv_cache,
+            logits_indices=torch.arange(len(reqs), dtype=torch.int64, device=dev),
+        )
+        ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
+        return meta, ids
+
+    def _pipeline_forward(
+        self, meta: ForwardMeta, input_ids: torch.Tensor, need_logits_at: torch.Tensor
+    ) -> Optional[torch.Tensor]:
+        """Run the shard; move hidden states along the pipeline. Returns logits
+        on the last stage, None elsewhere."""
+        comm = self.comm
+        T = meta.num_tokens
+        h = self.cfg.hidden_size
+        if comm.is_first_stage:
+            hidden = self.model.embed(input_ids).to(self.args.dtype)
+        else:
+            hidden = comm.pp_recv((T, h), self.args.dtype, comm.pp_rank - 1)
+        hidden = self.model(hidden, meta)
+        if not comm.is_last_stage:
+            comm.pp_send(hidden, comm.pp_rank + 1)
+            return None
+        if need_logits_at.numel() == 0:
+            return None
+        return self.model.compute_logits(hidden[need_logits_at])
+
+    def _run_prefill(self, chunks: List[PrefillChunk]):
+        meta, input_ids = self._build_prefill_meta(chunks)
+        samp_reqs = [c.req for c in chunks if c.is_last_chunk]
+        # logits only at last tokens of final chunks
+        keep = torch.tensor(
+            [i for i, c in enumerate(chunks) if c.is_last_chunk], dtype=torch.int64
+        )
+        need = meta.logits_indices.cpu()[keep].to(self.device) if keep.numel() else \
+            torch.empty(0, dtype=torch.int64, device=self.device)
+        with torch.inference_mode():
+            logits = self._pipeline_forward(meta, input_ids, need)
+        return logits, samp_reqs
+
+    def _run_decode(self, reqs: List[InitialRequest]) -> Optional[torch.Tensor]:
+        mb = max(1, self.args.micro_batches) if self.comm.pp_size > 1 else 1
+        if mb == 1 or len(reqs) < mb:
+            meta, ids = self._build_decode_meta(reqs)
+            with torch.inference_mode():
+                return self._pipeline_forward(meta, ids, meta.logits_indices)
+        # micro-batch pipelining: split the decode batch, pipeline the chunks
+        groups: List[List[InitialRequest]] = [list(x) for x in _split(reqs, mb)]
+        logits_parts = []
+        with torch.inference_mode():
+            for g in groups:
+                meta, ids = self._build_decode_meta(g)
+                logits_parts.append(self._pipeline_forward(meta, ids, meta.logits_indices))
+        if self.comm.is_last_stage:
+            return torch.cat([l for l in logits_parts if l is not None], dim=0)
+        return None
+
+    # -- sampling + commit ------------------------------------------------------------------
+
+    def _sample_and_broadcast(
+        self, logits_parts: List[torch.Tensor], sample_reqs: List[InitialRequest]
+    ) -> List[int]:
+        comm = self.comm
+        if comm.pp_size == 1:
+            logits = torch.cat(logits_parts, dim=0)
+            return self.sampler.sample(logits, sample_reqs)
+        token_ids: List[Optional[int]]
+        if comm.is_last_stage and comm.tp_rank == 0:
+            logits = torch.cat(logits_parts, dim=0)
+            token_ids = self.sampler.sample(logits, sample_reqs)
+        else:
+            token_ids = [None] * len(sample_reqs)
+        if comm.world_size > 1:
+            payload = [token_ids]
+            src = comm.stage_rank(comm.pp_size - 1) - comm.tp_rank  # tp_rank 0 of last stage
+            dist.broadcast_object_list(payload, src=src)
+            token_ids = payload[0]
+        return token_ids  # type: ignore[return-value]
+
+    # -- convenience: synchronous generation (tests, chat CLI) --------------------------------
+
+    def generate(
+        self,
+        prompts: List[List[int]],
+        sampling_params: Optional[List[SamplingParams]] = None,
+        max_steps: int = 100000,
+    ) -> Dict[str, List[int]]:
+        rids = []
+        for i, p in enumerate(prompts):
+            sp = sampling_params[i] if sampling_params else SamplingParams()
+            rids.append(self.submit(p, sp))
+        done: Dict[str, List[int]] = {}
+        outputs: Dict[str, List[int]] = {rid: [] for rid in rids}
+        for _ in range(max_steps):
+            if not self.has_work:
+                break
+            for out in self.step():
+                if out.rid in outputs:
+                    outputs[out.rid].append(out.token_id)
+                    if out.finished:
+                        done[out.rid] = outputs[out.rid]
+            if len(done) == len(rids):
+                break
+        return {rid: outputs[rid] for rid in rids}
+
+
+def _split(xs, n):
+    k, m = divmod(len(xs), n)
+    out, i = [], 0
+    for j in range(n):
+        size = k + (1 if j < m else 0)
+        if size:
+            out.append(xs[i : i + size])
+        i += size
+    return out

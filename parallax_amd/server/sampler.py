@@ -1,0 +1,53 @@
+"""Batched sampler: temperature / top-k / top-p / min-p / penalties.
+
+Parity with the reference's Sampler + SamplingBatchInfo (sampling/sampler.py);
+fresh torch design. Penalty bookkeeping uses the head-stage request state."""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+from .request import InitialRequest
+
+
+class Sampler:
+    def __init__(self, device: torch.device, seed: Optional[int] = None):
+        self.device = device
+        self.generator = None
+        if seed is not None:
+            self.generator = torch.Generator(
+                device=device if device.type == "cuda" else "cpu"
+            ).manual_seed(seed)
+
+    def sample(self, logits: torch.Tensor, reqs: List[InitialRequest]) -> List[int]:
+        """logits: [B, vocab] fp32, row i belongs to reqs[i]. Returns token ids."""
+        B = logits.shape[0]
+        assert B == len(reqs)
+        sp = [r.sampling_params for r in reqs]
+        need_penalties = any(
+            s.repetition_penalty != 1.0 or s.presence_penalty != 0.0
+            or s.frequency_penalty != 0.0
+            for s in sp
+        )
+        if need_penalties:
+            logits = ops.apply_penalties(
+                logits,
+                [r.output_token_ids for r in reqs],
+                [r.prompt_token_ids for r in reqs],
+                torch.tensor([s.repetition_penalty for s in sp], device=logits.device),
+                torch.tensor([s.presence_penalty for s in sp], device=logits.device),
+                torch.tensor([s.frequency_penalty for s in sp], device=logits.device),
+            )
+        dev = logits.device
+        tokens = ops.sample_tokens(
+            logits,
+            torch.tensor([s.temperature for s in sp], device=dev),
+            torch.tensor([s.top_p for s in sp], device=dev),
+            torch.tensor([s.top_k for s in sp], device=dev, dtype=torch.long),
+            torch.tensor([s.min_p for s in sp], device=dev),
+            generator=self.generator,
+        )
+        return tokens.tolist()

@@ -1,0 +1,89 @@
+"""Allocator / radix prefix cache / cache-manager unit tests (CPU).
+
+Test strategy mirrors the reference's cache tests: pure
+bookkeeping, no tensors needed.
+"""
+
+import pytest
+
+from parallax_amd.server.allocator import BlockAllocator, OutOfBlocksError
+from parallax_amd.server.cache_manager import CacheManager
+from parallax_amd.server.radix_cache import BlockRadixCache
+
+
+def test_block_allocator_basic():
+    a = BlockAllocator(8, 16)
+    blocks = a.allocate(3)
+    assert len(blocks) == 3 and a.num_free_blocks == 5
+    a.free(blocks)
+    assert a.num_free_blocks == 8
+    with pytest.raises(OutOfBlocksError):
+        a.allocate(9)
+
+
+def test_block_allocator_refcount():
+    a = BlockAllocator(4, 16)
+    (b,) = a.allocate(1)
+    a.incref(b)
+    assert a.decref(b) == 1
+    assert a.num_free_blocks == 3
+    assert a.decref(b) == 0
+    assert a.num_free_blocks == 4
+
+
+def test_radix_match_and_insert():
+    a = BlockAllocator(32, 4)
+    r = BlockRadixCache(a, 4)
+    toks = list(range(10))  # 2 full blocks + 2 spare
+    blocks = a.allocate(3)
+    r.insert(toks, blocks)
+    hit_blocks, n = r.match_prefix(toks)
+    assert n == 8 and hit_blocks == blocks[:2]
+    # diverging suffix only matches the shared prefix
+    hit_blocks, n = r.match_prefix(list(range(4)) + [99, 98, 97, 96])
+    assert n == 4 and hit_blocks == blocks[:1]
+
+
+def test_radix_eviction_returns_blocks():
+    a = BlockAllocator(4, 4)
+    r = BlockRadixCache(a, 4)
+    blocks = a.allocate(2)
+    r.insert(list(range(8)), blocks)
+    a.free(blocks)  # request done; only the tree holds them
+    assert a.num_free_blocks == 2
+    assert r.evict(2) == 2
+    assert a.num_free_blocks == 4
+
+
+def test_cache_manager_prefix_reuse():
+    cm = CacheManager(block_size=4, num_blocks=16)
+    s1 = cm.allocate_request("r1", list(range(10)))
+    assert s1.num_cached_tokens == 0
+    cm.publish_prefill("r1", list(range(10)))
+    s2 = cm.allocate_request("r2", list(range(10)) + [50, 51])
+    assert s2.num_cached_tokens == 8  # two full blocks reused
+    assert s2.block_table[:2] == s1.block_table[:2]
+    cm.free_request("r1")
+    cm.free_request("r2")
+    # all blocks recoverable after eviction
+    cm.radix.evict(100)
+    assert cm.num_free_blocks == 16
+
+
+def test_cache_manager_whole_prompt_cached_recomputes_last_block():
+    cm = CacheManager(block_size=4, num_blocks=16)
+    cm.allocate_request("r1", list(range(8)))
+    cm.publish_prefill("r1", list(range(8)))
+    s2 = cm.allocate_request("r2", list(range(8)))
+    # must leave at least the last block to recompute (hidden states needed)
+    assert s2.num_cached_tokens == 4
+
+
+def test_cache_manager_decode_growth():
+    cm = CacheManager(block_size=4, num_blocks=8)
+    s = cm.allocate_request("r1", [1, 2, 3])
+    assert len(s.block_table) == 1
+    cm.append_tokens("r1", 5)
+    assert len(cm.get("r1").block_table) == 2
+    cm.append_tokens("r1", 5)  # idempotent
+    assert len(cm.get("r1").block_table) == 2
