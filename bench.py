@@ -1,0 +1,198 @@
+"""Flagship benchmark: continuous-batching decode throughput (output tokens/s)
+of DeepSeek-R1-Distill-Llama-8B (Llama-3.1-8B architecture), bf16, synthetic
+data, random-init weights — the BASELINE.json PP=1 headline config.
+
+Single GPU by default. With --gpus N (launched via torch.distributed.run, one
+rank per GPU) the model runs pipeline-parallel over N stages with the global
+batch scaled by N (weak scaling) and N micro-batches in flight.
+
+A "step" = one engine decode iteration: every running request advances one
+token (forward through all local layers + paged attention + sampling + commit).
+Prefill happens before the timed region (it is the TTFT story, not the decode
+throughput story); warmup decode steps precede timing; the timed region is
+bracketed by barrier + torch.cuda.synchronize on both sides; the reported time
+is the MAX across ranks.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def deepseek_r1_distill_llama_8b() -> dict:
+    # architecture of deepseek-ai/DeepSeek-R1-Distill-Llama-8B (= Llama-3.1-8B)
+    return {
+        "architectures": ["LlamaForCausalLM"],
+        "model_type": "llama",
+        "vocab_size": 128256,
+        "hidden_size": 4096,
+        "num_hidden_layers": 32,
+        "num_attention_heads": 32,
+        "num_key_value_heads": 8,
+        "head_dim": 128,
+        "intermediate_size": 14336,
+        "rms_norm_eps": 1e-5,
+        "rope_theta": 500000.0,
+        "max_position_embeddings": 131072,
+        "tie_word_embeddings": False,
+        "torch_dtype": "bfloat16",
+        "eos_token_id": 128001,
+    }
+
+
+def qwen2_05b() -> dict:
+    return {
+        "architectures": ["Qwen2ForCausalLM"],
+        "model_type": "qwen2",
+        "vocab_size": 151936,
+        "hidden_size": 896,
+        "num_hidden_layers": 24,
+        "num_attention_heads": 14,
+        "num_key_value_heads": 2,
+        "head_dim": 64,
+        "intermediate_size": 4864,
+        "rms_norm_eps": 1e-6,
+        "rope_theta": 1000000.0,
+        "max_position_embeddings": 32768,
+        "qkv_bias": True,
+        "tie_word_embeddings": True,
+        "torch_dtype": "bfloat16",
+    }
+
+
+MODELS = {
+    "deepseek-r1-distill-llama-8b": (deepseek_r1_distill_llama_8b, "DeepSeek-R1-Distill-Llama-8B"),
+    "qwen2-0.5b": (qwen2_05b, "Qwen2-0.5B"),
+}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=32)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--model", default="deepseek-r1-distill-llama-8b", choices=MODELS)
+    ap.add_argument("--batch-per-gpu", type=int, default=64,
+                    help="decode batch per GPU (global batch = N * this)")
+    ap.add_argument("--prompt-len", type=int, default=512)
+    ap.add_argument("--block-size", type=int, default=32)
+    ap.add_argument("--micro-batches", type=int, default=0,
+                    help="0 = one per pipeline stage")
+    ap.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
+    args = ap.parse_args()
+
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.parallel.comm import init_distributed
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = args.gpus if world == 1 else world
+    comm = init_distributed(pp_size=world, tp_size=1)
+
+    cfg_fn, model_name = MODELS[args.model]
+    cfg = ModelConfig.from_hf_config(cfg_fn())
+    if args.cpu:
+        cfg.num_layers = 2
+        cfg.vocab_size = 1024
+        cfg.hidden_size = 256
+        cfg.num_heads, cfg.num_kv_heads, cfg.head_dim = 4, 2, 64
+        cfg.intermediate_size = 512
+
+    use_gpu = torch.cuda.is_available() and not args.cpu
+    global_batch = args.batch_per_gpu * world
+    total_new_tokens = args.warmup + args.steps + 8
+    eargs = EngineArgs(
+        block_size=args.block_size,
+        max_batch_size=max(128, global_batch),
+        max_num_tokens_per_batch=max(16384, global_batch * 2),
+        micro_batches=args.micro_batches or world,
+        dtype=torch.bfloat16 if use_gpu else torch.float32,
+        num_kv_blocks=None if use_gpu else 4096,
+        enable_prefix_cache=False,  # unique synthetic prompts; skip radix overhead
+        seed=0,
+    )
+    engine = Engine(cfg, eargs, comm=comm, random_weights=True)
+
+    # synthetic prompts, unique tokens so nothing prefix-shares
+    g = torch.Generator().manual_seed(1234 + 7)
+    sp = SamplingParams(
+        temperature=1.0, top_p=1.0, top_k=-1,
+        max_new_tokens=total_new_tokens, ignore_eos=True,
+    )
+    if comm.rank == 0:
+        for i in range(global_batch):
+            prompt = torch.randint(0, cfg.vocab_size, (args.prompt_len,), generator=g).tolist()
+            engine.submit(prompt, sp)
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        comm.barrier()
+
+    # prefill everything (untimed; ends when all requests are decoding)
+    t_prefill0 = time.perf_counter()
+    while True:
+        engine.step()
+        running = engine.scheduler.running
+        if running and all(r.prefill_done and r.num_output_tokens >= 1 for r in running.values()):
+            break
+        if not engine.has_work:
+            raise RuntimeError("all requests finished during prefill phase?")
+    sync()
+    prefill_s = time.perf_counter() - t_prefill0
+
+    for _ in range(args.warmup):
+        engine.step()
+    sync()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        outs = engine.step()
+        assert len(outs) == global_batch, f"batch shrank: {len(outs)}"
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if comm.world_size > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if comm.device.type == "cuda":
+            t = t.to(comm.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    tokens = global_batch * args.steps
+    value = tokens / elapsed
+    if comm.rank == 0:
+        result = {
+            "metric": "output_tokens_per_sec",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": world if use_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": global_batch,
+                "seq_len": args.prompt_len,
+                "parallelism": f"pp{world}",
+                "micro_batches": eargs.micro_batches,
+                "prefill_s": round(prefill_s, 3),
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

@@ -31,6 +31,9 @@ class ForwardMeta:
     # indices of the last token of each request within the packed token dim
     # (where logits are needed; for decode this is arange(B))
     logits_indices: Optional[torch.Tensor] = None
+    # host-side max of seq_lens (avoids a device sync when picking the
+    # flash-decoding partition count)
+    max_seq_len: int = 0
 
     @property
     def num_tokens(self) -> int:

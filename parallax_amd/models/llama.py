@@ -93,7 +93,7 @@ class LlamaAttention(nn.Module):
         else:
             attn = ops.paged_attention_decode(
                 q, k_cache, v_cache, meta.block_tables, meta.seq_lens,
-                self.scale, self.sliding_window,
+                self.scale, self.sliding_window, max_seq_len=meta.max_seq_len or None,
             )
         return self.o_proj(attn.reshape(T, self.num_heads * self.head_dim))
 

@@ -21,6 +21,7 @@ logger = get_logger("ops")
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
+_warned_prefill = False
 
 
 def _try_load_extension():
@@ -148,10 +149,13 @@ def paged_attention_decode(
     sliding_window: int = -1,
     softcap: float = 0.0,
     sinks: Optional[torch.Tensor] = None,
+    max_seq_len: Optional[int] = None,
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("paged_attention_decode")
         out = torch.empty_like(q)
+        if max_seq_len is None:
+            max_seq_len = int(seq_lens.max().item())
         ext.paged_attention_decode(
             out,
             q.contiguous(),
@@ -163,6 +167,7 @@ def paged_attention_decode(
             sliding_window,
             softcap,
             sinks if sinks is not None else q.new_empty(0),
+            max_seq_len,
         )
         return out
     return ref.paged_attention_decode(
@@ -184,24 +189,31 @@ def prefill_attention(
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("prefill_attention")
-        out = torch.empty_like(q)
-        # varlen metadata: cumulative query offsets
-        cu_q = torch.zeros(len(query_lens) + 1, dtype=torch.int32, device=q.device)
-        torch.cumsum(query_lens.to(torch.int32), dim=0, out=cu_q[1:])
-        ext.prefill_attention(
-            out,
-            q.contiguous(),
-            k_cache,
-            v_cache,
-            block_tables.to(torch.int32),
-            seq_lens.to(torch.int32),
-            cu_q,
-            scale,
-            sliding_window,
-            softcap,
-            sinks if sinks is not None else q.new_empty(0),
-        )
-        return out
+        if hasattr(ext, "prefill_attention"):
+            out = torch.empty_like(q)
+            cu_q = torch.zeros(len(query_lens) + 1, dtype=torch.int32, device=q.device)
+            torch.cumsum(query_lens.to(torch.int32), dim=0, out=cu_q[1:])
+            ext.prefill_attention(
+                out,
+                q.contiguous(),
+                k_cache,
+                v_cache,
+                block_tables.to(torch.int32),
+                seq_lens.to(torch.int32),
+                cu_q,
+                scale,
+                sliding_window,
+                softcap,
+                sinks if sinks is not None else q.new_empty(0),
+            )
+            return out
+        global _warned_prefill
+        if not _warned_prefill:
+            logger.warning(
+                "HIP prefill_attention not in extension yet; using torch-composed "
+                "prefill (rocBLAS GEMMs) — decode stays on the HIP kernel"
+            )
+            _warned_prefill = True
     return ref.prefill_attention(
         q, k_cache, v_cache, block_tables, seq_lens, query_lens, scale,
         sliding_window, softcap, sinks,

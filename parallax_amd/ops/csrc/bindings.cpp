@@ -1,0 +1,192 @@
+// Python bindings for the parallax_amd gfx950 kernels (torch extension).
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+#define CHECK_BF16(x) \
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, #x " must be bf16")
+
+extern "C" {
+void launch_rmsnorm(void*, void*, const void*, float, int, int, hipStream_t);
+void launch_fused_add_rmsnorm(void*, void*, const void*, float, int, int,
+                              hipStream_t);
+void launch_rope(void*, void*, const int*, const float*, int, int, int, int,
+                 int, bool, hipStream_t);
+void launch_reshape_and_cache(const void*, const void*, void*, void*,
+                              const int64_t*, int, int, int, int, hipStream_t);
+void launch_mla_reshape_and_cache(const void*, const void*, void*,
+                                  const int64_t*, int, int, int, int,
+                                  hipStream_t);
+void launch_act_and_mul(void*, const void*, int64_t, int, bool, hipStream_t);
+void launch_paged_attention_decode(void*, const void*, const void*, const void*,
+                                   const int*, const int*, int, int, int, int,
+                                   int, int, float, int, float, const float*,
+                                   int, int, float*, float*, hipStream_t,
+                                   bool*);
+}
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_BF16(x);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  const int rows = x.numel() / H;
+  launch_rmsnorm(out.data_ptr(), x.data_ptr(), w.data_ptr(), (float)eps, rows,
+                 H, cur_stream());
+}
+
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual, torch::Tensor w,
+                       double eps) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_CONTIG(residual);
+  CHECK_BF16(x);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  const int rows = x.numel() / H;
+  launch_fused_add_rmsnorm(x.data_ptr(), residual.data_ptr(), w.data_ptr(),
+                           (float)eps, rows, H, cur_stream());
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                  torch::Tensor cos_sin, bool is_neox) {
+  CHECK_GPU(q);
+  CHECK_CONTIG(q);
+  CHECK_BF16(q);
+  TORCH_CHECK(positions.scalar_type() == at::kInt);
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
+  const int T = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hk = k.numel() > 0 ? k.size(1) : 0;
+  const int rot = cos_sin.size(-1);
+  launch_rope(q.data_ptr(), Hk ? k.data_ptr() : nullptr,
+              positions.data_ptr<int>(), cos_sin.data_ptr<float>(), T, Hq, Hk,
+              D, rot, is_neox, cur_stream());
+}
+
+void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                       torch::Tensor v_cache, torch::Tensor slot_mapping) {
+  CHECK_GPU(k);
+  CHECK_CONTIG(k);
+  CHECK_BF16(k_cache);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  const int T = k.size(0);
+  const int Hk = k.size(1);
+  const int D = k.size(2);
+  const int BS = k_cache.size(2);
+  TORCH_CHECK((Hk * D) % 8 == 0);
+  launch_reshape_and_cache(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+                           v_cache.data_ptr(), slot_mapping.data_ptr<int64_t>(),
+                           T, Hk, D, BS, cur_stream());
+}
+
+void mla_reshape_and_cache(torch::Tensor latent, torch::Tensor k_rope,
+                           torch::Tensor cache, torch::Tensor slot_mapping) {
+  CHECK_GPU(latent);
+  CHECK_CONTIG(latent);
+  const int T = latent.size(0);
+  const int R = latent.size(-1);
+  const int DR = k_rope.size(-1);
+  const int BS = cache.size(1);
+  TORCH_CHECK(R % 8 == 0 && DR % 8 == 0);
+  launch_mla_reshape_and_cache(latent.data_ptr(), k_rope.data_ptr(),
+                               cache.data_ptr(),
+                               slot_mapping.data_ptr<int64_t>(), T, R, DR, BS,
+                               cur_stream());
+}
+
+void silu_and_mul(torch::Tensor out, torch::Tensor x) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_BF16(x);
+  const int I = out.size(-1);
+  TORCH_CHECK(I % 8 == 0);
+  launch_act_and_mul(out.data_ptr(), x.data_ptr(), x.numel() / (2 * I), I,
+                     false, cur_stream());
+}
+
+void gelu_and_mul(torch::Tensor out, torch::Tensor x) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_BF16(x);
+  const int I = out.size(-1);
+  TORCH_CHECK(I % 8 == 0);
+  launch_act_and_mul(out.data_ptr(), x.data_ptr(), x.numel() / (2 * I), I, true,
+                     cur_stream());
+}
+
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor seq_lens,
+                            double scale, int64_t sliding_window,
+                            double softcap, torch::Tensor sinks,
+                            int64_t max_seq_len) {
+  CHECK_GPU(q);
+  CHECK_CONTIG(q);
+  CHECK_BF16(q);
+  CHECK_BF16(k_cache);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt);
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hk = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  const int max_blocks = block_tables.size(1);
+  const int G = Hq / Hk;
+  TORCH_CHECK(Hq % Hk == 0 && G <= 16, "GQA group must divide and be <= 16");
+  TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
+  TORCH_CHECK(BS == 16 || BS == 32 || BS == 64, "block size must be 16/32/64");
+
+  const float* sinks_ptr = nullptr;
+  torch::Tensor sinks_f;
+  if (sinks.numel() > 0) {
+    sinks_f = sinks.to(at::kFloat).contiguous();
+    sinks_ptr = sinks_f.data_ptr<float>();
+  }
+
+  // flash-decoding split: partition long contexts to fill the 256-CU chip
+  const int part_tokens = 1024;
+  int num_parts = 1;
+  if (max_seq_len > part_tokens) {
+    num_parts = std::min<int64_t>((max_seq_len + part_tokens - 1) / part_tokens, 128);
+  }
+  torch::Tensor tmp_acc, tmp_ml;
+  float *acc_ptr = nullptr, *ml_ptr = nullptr;
+  if (num_parts > 1) {
+    auto opts = q.options().dtype(at::kFloat);
+    tmp_acc = torch::empty({B, Hq, num_parts, D}, opts);
+    tmp_ml = torch::empty({B, Hq, num_parts, 2}, opts);
+    acc_ptr = tmp_acc.data_ptr<float>();
+    ml_ptr = tmp_ml.data_ptr<float>();
+  }
+  bool launched = false;
+  launch_paged_attention_decode(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hq, Hk, D, BS,
+      max_blocks, (float)scale, (int)sliding_window, (float)softcap, sinks_ptr,
+      num_parts, part_tokens, acc_ptr, ml_ptr, cur_stream(), &launched);
+  TORCH_CHECK(launched, "no kernel instantiation for D=", D, " BS=", BS,
+              " G=", G);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm);
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.def("rope_inplace", &rope_inplace);
+  m.def("reshape_and_cache", &reshape_and_cache);
+  m.def("mla_reshape_and_cache", &mla_reshape_and_cache);
+  m.def("silu_and_mul", &silu_and_mul);
+  m.def("gelu_and_mul", &gelu_and_mul);
+  m.def("paged_attention_decode", &paged_attention_decode);
+}

@@ -1,0 +1,231 @@
+// Memory-bound elementwise / normalization kernels for gfx950.
+// All bf16 traffic is vectorized 16 B/lane (bf16x8) per the CDNA guide (G13).
+//
+// Reference analogues: RMSNorm/RoPE/activation live inside MLX / SGLang in the
+// reference; here they are first-class HIP kernels fused where it pays
+// (fused_add_rmsnorm saves two HBM round-trips per layer).
+
+#include "common.h"
+
+// ---- rmsnorm -------------------------------------------------------------------
+// x: [rows, H] bf16, w: [H] bf16, out: [rows, H]. One block per row.
+// fp32 accumulation; H must be a multiple of 8.
+
+template <bool FUSED_ADD>
+__global__ void rmsnorm_kernel(
+    uint16_t* __restrict__ out,          // [rows, H] (bf16 bits)
+    uint16_t* __restrict__ x,            // [rows, H] — input; mutated if FUSED_ADD
+    uint16_t* __restrict__ residual,     // [rows, H] or nullptr
+    const uint16_t* __restrict__ w,      // [H]
+    const float eps,
+    const int H) {
+  __shared__ float red[16];
+  const int row = blockIdx.x;
+  uint16_t* xr = x + (size_t)row * H;
+  uint16_t* rr = FUSED_ADD ? residual + (size_t)row * H : nullptr;
+  uint16_t* orow = out + (size_t)row * H;
+
+  const int vecs = H / 8;
+  float ss = 0.f;
+  // pass 1: (optional add) + sum of squares; FUSED_ADD writes the new residual
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 xv = load_bf16x8(xr + i * 8);
+    float vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[j] = bf16x8_get(xv, j);
+    if (FUSED_ADD) {
+      bf16x8 rv = load_bf16x8(rr + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] += bf16x8_get(rv, j);
+      store_bf16x8(rr + i * 8, vals);  // residual' = x + residual
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ss = fmaf(vals[j], vals[j], ss);
+  }
+  ss = block_reduce_sum(ss, red);
+  const float rstd = rsqrtf(ss / H + eps);
+  __syncthreads();
+
+  // pass 2: normalize * weight  (rows are L2-resident between passes)
+  const uint16_t* src = FUSED_ADD ? rr : xr;
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 xv = load_bf16x8(src + i * 8);
+    bf16x8 wv = load_bf16x8(w + i * 8);
+    float vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      vals[j] = bf16x8_get(xv, j) * rstd * bf16x8_get(wv, j);
+    store_bf16x8(orow + i * 8, vals);
+  }
+}
+
+extern "C" void launch_rmsnorm(
+    void* out, void* x, const void* w, float eps, int rows, int H,
+    hipStream_t stream) {
+  int threads = min(1024, max(64, ((H / 8) + 63) / 64 * 64));
+  rmsnorm_kernel<false><<<rows, threads, 0, stream>>>(
+      (uint16_t*)out, (uint16_t*)x, nullptr, (const uint16_t*)w, eps, H);
+}
+
+extern "C" void launch_fused_add_rmsnorm(
+    void* x, void* residual, const void* w, float eps, int rows, int H,
+    hipStream_t stream) {
+  int threads = min(1024, max(64, ((H / 8) + 63) / 64 * 64));
+  // out == x (in-place): x <- rmsnorm(x + residual), residual <- x + residual
+  rmsnorm_kernel<true><<<rows, threads, 0, stream>>>(
+      (uint16_t*)x, (uint16_t*)x, (uint16_t*)residual, (const uint16_t*)w, eps, H);
+}
+
+// ---- RoPE (neox + interleaved) ---------------------------------------------------
+// q: [T, Hq, D], k: [T, Hk, D] (bf16, mutated in place), positions: [T] i32,
+// cos_sin: [P, rot] fp32 (cos | sin halves). Host-precomputed table (guide §B).
+// One block per token; threads sweep (head, pair).
+
+template <bool NEOX>
+__global__ void rope_kernel(
+    uint16_t* __restrict__ q, uint16_t* __restrict__ k,
+    const int* __restrict__ positions, const float* __restrict__ cos_sin,
+    const int Hq, const int Hk, const int D, const int rot) {
+  const int t = blockIdx.x;
+  const int half = rot / 2;
+  const float* cs = cos_sin + (size_t)positions[t] * rot;
+  const int total = (Hq + Hk) * half;
+  for (int i = threadIdx.x; i < total; i += blockDim.x) {
+    const int h = i / half;
+    const int j = i % half;
+    uint16_t* base = (h < Hq) ? q + ((size_t)t * Hq + h) * D
+                              : k + ((size_t)t * Hk + (h - Hq)) * D;
+    const float c = cs[j], s = cs[half + j];
+    int i1, i2;
+    if (NEOX) {
+      i1 = j; i2 = j + half;
+    } else {
+      i1 = 2 * j; i2 = 2 * j + 1;
+    }
+    const float x1 = bf16_bits_to_f32(base[i1]);
+    const float x2 = bf16_bits_to_f32(base[i2]);
+    base[i1] = f32_to_bf16_bits(fmaf(x1, c, -x2 * s));
+    base[i2] = f32_to_bf16_bits(fmaf(x2, c, x1 * s));
+  }
+}
+
+extern "C" void launch_rope(
+    void* q, void* k, const int* positions, const float* cos_sin,
+    int T, int Hq, int Hk, int D, int rot, bool neox, hipStream_t stream) {
+  int threads = min(512, max(64, ceil_div((Hq + Hk) * rot / 2, 64) * 64));
+  if (neox)
+    rope_kernel<true><<<T, threads, 0, stream>>>(
+        (uint16_t*)q, (uint16_t*)k, positions, cos_sin, Hq, Hk, D, rot);
+  else
+    rope_kernel<false><<<T, threads, 0, stream>>>(
+        (uint16_t*)q, (uint16_t*)k, positions, cos_sin, Hq, Hk, D, rot);
+}
+
+// ---- reshape_and_cache ------------------------------------------------------------
+// k/v: [T, Hk, D] bf16 -> k_cache/v_cache [NB, Hk, BS, D] via slot_mapping [T]
+// (slot = block * BS + off; -1 skips). One block per token, vectorized rows.
+
+__global__ void reshape_and_cache_kernel(
+    const uint16_t* __restrict__ k, const uint16_t* __restrict__ v,
+    uint16_t* __restrict__ k_cache, uint16_t* __restrict__ v_cache,
+    const int64_t* __restrict__ slot_mapping,
+    const int Hk, const int D, const int BS) {
+  const int t = blockIdx.x;
+  const int64_t slot = slot_mapping[t];
+  if (slot < 0) return;
+  const int64_t blk = slot / BS, off = slot % BS;
+  const int vecs = Hk * D / 8;
+  const uint16_t* ksrc = k + (size_t)t * Hk * D;
+  const uint16_t* vsrc = v + (size_t)t * Hk * D;
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    const int h = (i * 8) / D;
+    const int d = (i * 8) % D;
+    // dst row: cache[blk][h][off][d]
+    const size_t dst = (((size_t)blk * Hk + h) * BS + off) * D + d;
+    *reinterpret_cast<int4*>(k_cache + dst) =
+        *reinterpret_cast<const int4*>(ksrc + i * 8);
+    *reinterpret_cast<int4*>(v_cache + dst) =
+        *reinterpret_cast<const int4*>(vsrc + i * 8);
+  }
+}
+
+extern "C" void launch_reshape_and_cache(
+    const void* k, const void* v, void* k_cache, void* v_cache,
+    const int64_t* slot_mapping, int T, int Hk, int D, int BS,
+    hipStream_t stream) {
+  int threads = min(512, max(64, ceil_div(Hk * D / 8, 64) * 64));
+  reshape_and_cache_kernel<<<T, threads, 0, stream>>>(
+      (const uint16_t*)k, (const uint16_t*)v, (uint16_t*)k_cache,
+      (uint16_t*)v_cache, slot_mapping, Hk, D, BS);
+}
+
+// MLA variant: latent [T, R] + rope [T, dr] -> cache [NB, BS, R+dr]
+__global__ void mla_reshape_and_cache_kernel(
+    const uint16_t* __restrict__ latent, const uint16_t* __restrict__ k_rope,
+    uint16_t* __restrict__ cache, const int64_t* __restrict__ slot_mapping,
+    const int R, const int DR, const int BS) {
+  const int t = blockIdx.x;
+  const int64_t slot = slot_mapping[t];
+  if (slot < 0) return;
+  const int64_t blk = slot / BS, off = slot % BS;
+  uint16_t* dst = cache + ((size_t)blk * BS + off) * (R + DR);
+  for (int i = threadIdx.x * 8; i < R; i += blockDim.x * 8)
+    *reinterpret_cast<int4*>(dst + i) =
+        *reinterpret_cast<const int4*>(latent + (size_t)t * R + i);
+  for (int i = threadIdx.x * 8; i < DR; i += blockDim.x * 8)
+    *reinterpret_cast<int4*>(dst + R + i) =
+        *reinterpret_cast<const int4*>(k_rope + (size_t)t * DR + i);
+}
+
+extern "C" void launch_mla_reshape_and_cache(
+    const void* latent, const void* k_rope, void* cache,
+    const int64_t* slot_mapping, int T, int R, int DR, int BS,
+    hipStream_t stream) {
+  mla_reshape_and_cache_kernel<<<T, 128, 0, stream>>>(
+      (const uint16_t*)latent, (const uint16_t*)k_rope, (uint16_t*)cache,
+      slot_mapping, R, DR, BS);
+}
+
+// ---- activations ----------------------------------------------------------------------
+// x: [T, 2I] (gate | up) -> out: [T, I];  grid-stride, vectorized.
+
+template <bool GELU>
+__global__ void act_and_mul_kernel(
+    uint16_t* __restrict__ out, const uint16_t* __restrict__ x,
+    const int64_t T, const int I) {
+  const int64_t total = T * (I / 8);
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t t = idx / (I / 8);
+    const int i = (idx % (I / 8)) * 8;
+    bf16x8 g = load_bf16x8(x + t * 2 * I + i);
+    bf16x8 u = load_bf16x8(x + t * 2 * I + I + i);
+    float vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gv = bf16x8_get(g, j);
+      const float uv = bf16x8_get(u, j);
+      float a;
+      if (GELU) {  // tanh approximation
+        const float c = 0.7978845608028654f;
+        a = 0.5f * gv * (1.f + tanhf(c * (gv + 0.044715f * gv * gv * gv)));
+      } else {  // silu
+        a = gv / (1.f + __expf(-gv));
+      }
+      vals[j] = a * uv;
+    }
+    store_bf16x8(out + t * I + i, vals);
+  }
+}
+
+extern "C" void launch_act_and_mul(
+    void* out, const void* x, int64_t T, int I, bool gelu, hipStream_t stream) {
+  long long total = (long long)T * (I / 8);
+  int blocks = clamp_int((total + 255) / 256, 1, 2048);
+  if (gelu)
+    act_and_mul_kernel<true><<<blocks, 256, 0, stream>>>(
+        (uint16_t*)out, (const uint16_t*)x, T, I);
+  else
+    act_and_mul_kernel<false><<<blocks, 256, 0, stream>>>(
+        (uint16_t*)out, (const uint16_t*)x, T, I);
+}

@@ -277,6 +277,7 @@ class Engine:
             query_lens=torch.tensor(query_lens, dtype=torch.int32, device=dev),
             kv_cache=self.kv_cache,
             logits_indices=torch.tensor(logits_idx, dtype=torch.int64, device=dev),
+            max_seq_len=max(seq_lens),
         )
         ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
         return meta, ids
@@ -302,6 +303,7 @@ class Engine:
             seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
             kv_cache=self.kv_cache,
             logits_indices=torch.arange(len(reqs), dtype=torch.int64, device=dev),
+            max_seq_len=max(seq_lens),
         )
         ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
         return meta, ids
