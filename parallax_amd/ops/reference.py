@@ -72,7 +72,9 @@ def rope_inplace(
     sin = cs[:, half:].unsqueeze(1)
 
     def _apply(t: torch.Tensor) -> None:
-        rot = t[..., :rot_dim].float()
+        # clone: .float() on an fp32 tensor is a VIEW — without the clone the
+        # first store below would clobber x1/x2 before o2 is evaluated
+        rot = t[..., :rot_dim].clone().float()
         if is_neox:
             x1, x2 = rot[..., :half], rot[..., half:]
             o1 = x1 * cos - x2 * sin
@@ -81,8 +83,10 @@ def rope_inplace(
             t[..., half:rot_dim] = o2.to(t.dtype)
         else:  # interleaved (GPT-J style)
             x1, x2 = rot[..., 0::2], rot[..., 1::2]
-            t[..., 0:rot_dim:2] = (x1 * cos - x2 * sin).to(t.dtype)
-            t[..., 1:rot_dim:2] = (x2 * cos + x1 * sin).to(t.dtype)
+            o1 = x1 * cos - x2 * sin
+            o2 = x2 * cos + x1 * sin
+            t[..., 0:rot_dim:2] = o1.to(t.dtype)
+            t[..., 1:rot_dim:2] = o2.to(t.dtype)
 
     _apply(q)
     if k is not None:
