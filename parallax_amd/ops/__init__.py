@@ -189,31 +189,32 @@ def prefill_attention(
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("prefill_attention")
-        if hasattr(ext, "prefill_attention"):
-            out = torch.empty_like(q)
-            cu_q = torch.zeros(len(query_lens) + 1, dtype=torch.int32, device=q.device)
-            torch.cumsum(query_lens.to(torch.int32), dim=0, out=cu_q[1:])
-            ext.prefill_attention(
-                out,
-                q.contiguous(),
-                k_cache,
-                v_cache,
-                block_tables.to(torch.int32),
-                seq_lens.to(torch.int32),
-                cu_q,
-                scale,
-                sliding_window,
-                softcap,
-                sinks if sinks is not None else q.new_empty(0),
-            )
-            return out
-        global _warned_prefill
-        if not _warned_prefill:
-            logger.warning(
-                "HIP prefill_attention not in extension yet; using torch-composed "
-                "prefill (rocBLAS GEMMs) — decode stays on the HIP kernel"
-            )
-            _warned_prefill = True
+        out = torch.empty_like(q)
+        ql = query_lens.cpu().tolist()
+        cu = [0]
+        tile_req, tile_row0 = [], []
+        for i, l in enumerate(ql):
+            cu.append(cu[-1] + int(l))
+            for r0 in range(0, int(l), 32):
+                tile_req.append(i)
+                tile_row0.append(r0)
+        dev = q.device
+        ext.prefill_attention(
+            out,
+            q.contiguous(),
+            k_cache,
+            v_cache,
+            block_tables.to(torch.int32),
+            seq_lens.to(torch.int32),
+            torch.tensor(cu, dtype=torch.int32, device=dev),
+            torch.tensor(tile_req, dtype=torch.int32, device=dev),
+            torch.tensor(tile_row0, dtype=torch.int32, device=dev),
+            scale,
+            sliding_window,
+            softcap,
+            sinks if sinks is not None else q.new_empty(0),
+        )
+        return out
     return ref.prefill_attention(
         q, k_cache, v_cache, block_tables, seq_lens, query_lens, scale,
         sliding_window, softcap, sinks,

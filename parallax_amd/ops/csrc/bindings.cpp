@@ -27,6 +27,10 @@ void launch_paged_attention_decode(void*, const void*, const void*, const void*,
                                    int, int, float, int, float, const float*,
                                    int, int, float*, float*, hipStream_t,
                                    bool*);
+void launch_prefill_attention(void*, const void*, const void*, const void*,
+                              const int*, const int*, const int*, const int*,
+                              const int*, int, int, int, int, int, int, float,
+                              int, float, const float*, hipStream_t, bool*);
 }
 
 static hipStream_t cur_stream() {
@@ -155,11 +159,21 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     sinks_ptr = sinks_f.data_ptr<float>();
   }
 
-  // flash-decoding split: partition long contexts to fill the 256-CU chip
-  const int part_tokens = 1024;
+  // flash-decoding split: pick the partition size so B*Hk*parts fills the
+  // 256-CU chip (≈2 workgroups per CU), bounded below by 256-token partitions
   int num_parts = 1;
-  if (max_seq_len > part_tokens) {
-    num_parts = std::min<int64_t>((max_seq_len + part_tokens - 1) / part_tokens, 128);
+  int part_tokens = 0;
+  if (max_seq_len > 256) {
+    const long long wgs = (long long)B * Hk;
+    // target ~4 co-resident workgroups per CU for latency hiding
+    const int target_parts = (int)std::max<long long>(1, (1024 + wgs - 1) / wgs);
+    const int max_parts = (int)((max_seq_len + 255) / 256);
+    num_parts = std::min(std::min(target_parts, max_parts), 128);
+    if (num_parts > 1) {
+      part_tokens = (int)((max_seq_len + num_parts - 1) / num_parts);
+      part_tokens = (part_tokens + 127) / 128 * 128;  // chunk-aligned
+      num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
+    }
   }
   torch::Tensor tmp_acc, tmp_ml;
   float *acc_ptr = nullptr, *ml_ptr = nullptr;
@@ -180,7 +194,46 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
               " G=", G);
 }
 
+void prefill_attention(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor block_tables, torch::Tensor seq_lens,
+                       torch::Tensor cu_q, torch::Tensor tile_req,
+                       torch::Tensor tile_row0, double scale,
+                       int64_t sliding_window, double softcap,
+                       torch::Tensor sinks) {
+  CHECK_GPU(q);
+  CHECK_CONTIG(q);
+  CHECK_BF16(q);
+  CHECK_BF16(k_cache);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt);
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
+  TORCH_CHECK(cu_q.scalar_type() == at::kInt);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hk = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  const int max_blocks = block_tables.size(1);
+  const int n_tiles = tile_req.size(0);
+  TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
+
+  const float* sinks_ptr = nullptr;
+  torch::Tensor sinks_f;
+  if (sinks.numel() > 0) {
+    sinks_f = sinks.to(at::kFloat).contiguous();
+    sinks_ptr = sinks_f.data_ptr<float>();
+  }
+  bool launched = false;
+  launch_prefill_attention(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+      cu_q.data_ptr<int>(), tile_req.data_ptr<int>(), tile_row0.data_ptr<int>(),
+      n_tiles, Hq, Hk, D, BS, max_blocks, (float)scale, (int)sliding_window,
+      (float)softcap, sinks_ptr, cur_stream(), &launched);
+  TORCH_CHECK(launched, "no prefill kernel instantiation for D=", D);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("prefill_attention", &prefill_attention);
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_inplace", &rope_inplace);
