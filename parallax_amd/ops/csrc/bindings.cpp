@@ -159,21 +159,15 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     sinks_ptr = sinks_f.data_ptr<float>();
   }
 
-  // flash-decoding split: pick the partition size so B*Hk*parts fills the
-  // 256-CU chip (≈2 workgroups per CU), bounded below by 256-token partitions
-  int num_parts = 1;
-  int part_tokens = 0;
-  if (max_seq_len > 256) {
-    const long long wgs = (long long)B * Hk;
-    // target ~4 co-resident workgroups per CU for latency hiding
-    const int target_parts = (int)std::max<long long>(1, (1024 + wgs - 1) / wgs);
-    const int max_parts = (int)((max_seq_len + 255) / 256);
-    num_parts = std::min(std::min(target_parts, max_parts), 128);
-    if (num_parts > 1) {
-      part_tokens = (int)((max_seq_len + num_parts - 1) / num_parts);
-      part_tokens = (part_tokens + 127) / 128 * 128;  // chunk-aligned
-      num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
-    }
+  // flash-decoding split: fixed 256-token partitions (graph-capture friendly —
+  // num_parts depends only on max_seq_len, and partitions past a sequence's
+  // actual length exit immediately / are skipped by the reduce kernel)
+  int part_tokens = 256;
+  int num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
+  if (num_parts > 128) {
+    part_tokens = (int)((max_seq_len + 127) / 128);
+    part_tokens = (part_tokens + 127) / 128 * 128;  // chunk-aligned
+    num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
   }
   torch::Tensor tmp_acc, tmp_ml;
   float *acc_ptr = nullptr, *ml_ptr = nullptr;

@@ -58,15 +58,29 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
     tok_end = min(tok_end, (p + 1) * part_tokens);
   }
 
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  if (tok_begin >= tok_end) {
+    // empty partition: record neutral (l=0 -> reduce kernel skips the acc read)
+    if (PARTITIONED) {
+      const int p = blockIdx.z;
+      for (int g = tid; g < G; g += ATTN_THREADS) {
+        const size_t hq = (size_t)hk * G + g;
+        float* ml = tmp_ml + (((size_t)seq * Hk * G + hq) * gridDim.z + p) * 2;
+        ml[0] = -1e30f;
+        ml[1] = 0.f;
+      }
+    }
+    return;
+  }
+
   __shared__ float q_lds[GMAX][HEAD_DIM];
   __shared__ float p_lds[GMAX][CHUNK_TOKENS];
   __shared__ float red_lds[ATTN_THREADS * 8];  // [NT_PAR][HEAD_DIM] final reduce
   __shared__ float m_lds[GMAX], l_lds[GMAX], rescale_lds[GMAX];
   __shared__ float wmax_lds[4][GMAX], wsum_lds[4][GMAX];
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wid = tid >> 6;
 
   // ---- load q (scaled) into LDS -------------------------------------------------
   for (int i = tid; i < G * HEAD_DIM; i += ATTN_THREADS) {
@@ -79,20 +93,6 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
     l_lds[tid] = 0.f;
   }
   __syncthreads();
-
-  if (tok_begin >= tok_end) {
-    // empty partition: record neutral (m=-inf, l=0)
-    if (PARTITIONED) {
-      const int p = blockIdx.z;
-      for (int g = tid; g < G; g += ATTN_THREADS) {
-        const size_t hq = (size_t)hk * G + g;
-        float* ml = tmp_ml + (((size_t)seq * Hk * G + hq) * gridDim.z + p) * 2;
-        ml[0] = -1e30f;
-        ml[1] = 0.f;
-      }
-    }
-    return;
-  }
 
   // phase-B accumulators: this thread owns dim-chunk dc for token-group tp
   const int dc = tid % DC;
@@ -280,6 +280,7 @@ __global__ void paged_attention_reduce_kernel(
   for (int d0 = threadIdx.x * 8; d0 < HEAD_DIM; d0 += blockDim.x * 8) {
     float vals[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     for (int p = 0; p < num_parts; ++p) {
+      if (tmp_ml[(base * num_parts + p) * 2 + 1] <= 0.f) continue;  // empty part
       const float w = __expf(tmp_ml[(base * num_parts + p) * 2] - M);
       const float* src = tmp_acc + (base * num_parts + p) * HEAD_DIM + d0;
 #pragma unroll

@@ -65,6 +65,8 @@ class EngineArgs:
     request_timeout_s: float = 600.0
     start_layer: Optional[int] = None   # explicit layer range (decentralized mode)
     end_layer: Optional[int] = None
+    max_model_len: int = 8192           # context ceiling (sizes graph buffers)
+    enable_graphs: bool = True          # hipGraph-captured decode forward
 
 
 def partition_layers(num_layers: int, pp_size: int, pp_rank: int) -> Tuple[int, int]:
@@ -133,6 +135,17 @@ class Engine:
             eos_token_ids=cfg.eos_token_ids,
         )
         self.sampler = Sampler(self.device, args.seed)
+        self.graph_runner = None
+        if args.enable_graphs and self.device.type == "cuda":
+            from .graph_runner import DecodeGraphRunner
+
+            self.graph_runner = DecodeGraphRunner(
+                self.model, self.kv_cache, self.device, args.dtype,
+                max_batch=args.max_batch_size, max_model_len=args.max_model_len,
+                block_size=args.block_size, hidden_size=cfg.hidden_size,
+                is_first_stage=self.comm.is_first_stage,
+                is_last_stage=self.comm.is_last_stage,
+            )
         self._pending_adds: List[InitialRequest] = []
         self._pending_aborts: List[str] = []
         self.step_count = 0
@@ -341,22 +354,51 @@ class Engine:
             logits = self._pipeline_forward(meta, input_ids, need)
         return logits, samp_reqs
 
+    def _decode_one(self, reqs: List[InitialRequest]) -> Optional[torch.Tensor]:
+        """One decode forward for one (micro-)batch, graph-captured on GPU."""
+        if self.graph_runner is None:
+            meta, ids = self._build_decode_meta(reqs)
+            return self._pipeline_forward(meta, ids, meta.logits_indices)
+        comm = self.comm
+        input_ids, positions, slots, btabs, seq_lens = [], [], [], [], []
+        for r in reqs:
+            state = self.cache_manager.get(r.rid)
+            pos = r.total_len - 1
+            positions.append(pos)
+            input_ids.append(r.output_token_ids[-1])
+            slots.append(
+                state.block_table[pos // self.args.block_size] * self.args.block_size
+                + pos % self.args.block_size
+            )
+            btabs.append(state.block_table)
+            seq_lens.append(r.total_len)
+        hidden_in = None
+        if not comm.is_first_stage:
+            hidden_in = comm.pp_recv(
+                (len(reqs), self.cfg.hidden_size), self.args.dtype, comm.pp_rank - 1
+            )
+        out = self.graph_runner.run(input_ids, positions, slots, btabs, seq_lens,
+                                    hidden_in=hidden_in)
+        if not comm.is_last_stage:
+            comm.pp_send(out, comm.pp_rank + 1)
+            return None
+        return out
+
     def _run_decode(self, reqs: List[InitialRequest]) -> Optional[torch.Tensor]:
         mb = max(1, self.args.micro_batches) if self.comm.pp_size > 1 else 1
-        if mb == 1 or len(reqs) < mb:
-            meta, ids = self._build_decode_meta(reqs)
-            with torch.inference_mode():
-                return self._pipeline_forward(meta, ids, meta.logits_indices)
-        # micro-batch pipelining: split the decode batch, pipeline the chunks
-        groups: List[List[InitialRequest]] = [list(x) for x in _split(reqs, mb)]
-        logits_parts = []
         with torch.inference_mode():
+            if mb == 1 or len(reqs) < mb:
+                return self._decode_one(reqs)
+            # micro-batch pipelining: split the decode batch, pipeline the chunks
+            groups: List[List[InitialRequest]] = [list(x) for x in _split(reqs, mb)]
+            logits_parts = []
             for g in groups:
-                meta, ids = self._build_decode_meta(g)
-                logits_parts.append(self._pipeline_forward(meta, ids, meta.logits_indices))
-        if self.comm.is_last_stage:
-            return torch.cat([l for l in logits_parts if l is not None], dim=0)
-        return None
+                out = self._decode_one(g)
+                if out is not None:
+                    logits_parts.append(out.clone())  # graph output buffer is reused
+            if self.comm.is_last_stage:
+                return torch.cat(logits_parts, dim=0)
+            return None
 
     # -- sampling + commit ------------------------------------------------------------------
 

@@ -1,0 +1,169 @@
+"""hipGraph-captured decode forward.
+
+The decode step of a 32-layer shard launches ~350 kernels; at ~10-20 us of
+launch+gap each that dominates the ~3 ms of real work. This runner captures the
+whole per-stage decode forward (embedding -> layers -> final norm -> lm_head)
+into a hipGraph per batch-size bucket and replays it with inputs written into
+persistent device buffers. Sampling and PP send/recv stay eager (outside the
+graph), so the same runner serves every pipeline stage.
+
+Capture-shape invariants:
+- batch padded up to the bucket; pad rows get seq_len=1, slot -1 (no KV write),
+  block-table row 0 (reads one garbage token — confined to the pad row).
+- block_tables buffer is [max_batch, max_blocks(max_model_len)]; real tables
+  are copied into the leading columns.
+- the attention partition count is a function of max_model_len only (see
+  bindings.cpp), so kernel grids are capture-stable.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..models.forward_meta import ForwardMeta
+from ..utils.logging_config import get_logger
+
+logger = get_logger("server.graph_runner")
+
+DEFAULT_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256]
+
+
+class DecodeGraphRunner:
+    def __init__(
+        self,
+        model,
+        kv_cache,
+        device: torch.device,
+        dtype: torch.dtype,
+        max_batch: int,
+        max_model_len: int,
+        block_size: int,
+        hidden_size: int,
+        is_first_stage: bool,
+        is_last_stage: bool,
+        buckets: Optional[List[int]] = None,
+    ):
+        self.model = model
+        self.kv_cache = kv_cache
+        self.device = device
+        self.dtype = dtype
+        self.max_model_len = max_model_len
+        self.block_size = block_size
+        self.is_first = is_first_stage
+        self.is_last = is_last_stage
+        self.max_blocks = (max_model_len + block_size - 1) // block_size
+        self.buckets = sorted(b for b in (buckets or DEFAULT_BUCKETS) if b <= max_batch)
+        if not self.buckets or self.buckets[-1] < max_batch:
+            self.buckets.append(max_batch)
+        B = self.buckets[-1]
+
+        dev = device
+        self.input_ids = torch.zeros(B, dtype=torch.long, device=dev)
+        self.hidden_in = torch.zeros(B, hidden_size, dtype=dtype, device=dev)
+        self.positions = torch.zeros(B, dtype=torch.int32, device=dev)
+        self.slot_mapping = torch.full((B,), -1, dtype=torch.int64, device=dev)
+        self.block_tables = torch.zeros(B, self.max_blocks, dtype=torch.int32, device=dev)
+        self.seq_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        # pinned host staging (one copy per step)
+        self.h_input_ids = torch.zeros(B, dtype=torch.long, pin_memory=True)
+        self.h_positions = torch.zeros(B, dtype=torch.int32, pin_memory=True)
+        self.h_slot_mapping = torch.full((B,), -1, dtype=torch.int64, pin_memory=True)
+        self.h_block_tables = torch.zeros(B, self.max_blocks, dtype=torch.int32,
+                                          pin_memory=True)
+        self.h_seq_lens = torch.ones(B, dtype=torch.int32, pin_memory=True)
+
+        self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self._outputs: Dict[int, torch.Tensor] = {}
+        self._pool = None
+
+    def _meta(self, bucket: int) -> ForwardMeta:
+        return ForwardMeta(
+            is_prefill=False,
+            positions=self.positions[:bucket],
+            slot_mapping=self.slot_mapping[:bucket],
+            block_tables=self.block_tables[:bucket],
+            seq_lens=self.seq_lens[:bucket],
+            kv_cache=self.kv_cache,
+            logits_indices=None,
+            max_seq_len=self.max_model_len,
+        )
+
+    def _forward(self, bucket: int) -> torch.Tensor:
+        meta = self._meta(bucket)
+        if self.is_first:
+            hidden = self.model.embed(self.input_ids[:bucket]).to(self.dtype)
+        else:
+            hidden = self.hidden_in[:bucket]
+        hidden = self.model(hidden, meta)
+        if self.is_last:
+            return self.model.compute_logits(hidden)
+        return hidden
+
+    def _capture(self, bucket: int) -> None:
+        logger.info("capturing decode graph for batch bucket %d", bucket)
+        # neutralize buffers: warmup/capture must not write into the live KV
+        # cache (slot -1 = skip) or read past block-table row 0
+        self.input_ids.zero_()
+        self.positions.zero_()
+        self.slot_mapping.fill_(-1)
+        self.block_tables.zero_()
+        self.seq_lens.fill_(1)
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warmup (caching allocator, rocBLAS heuristics)
+                self._forward(bucket)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph, pool=self._pool):
+            out = self._forward(bucket)
+        if self._pool is None:
+            self._pool = graph.pool()
+        self._graphs[bucket] = graph
+        self._outputs[bucket] = out
+
+    def bucket_for(self, batch: int) -> int:
+        for b in self.buckets:
+            if b >= batch:
+                return b
+        return self.buckets[-1]
+
+    def run(
+        self,
+        input_ids: List[int],
+        positions: List[int],
+        slot_mapping: List[int],
+        block_tables: List[List[int]],
+        seq_lens: List[int],
+        hidden_in: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """Returns logits[:B] (last stage) or hidden[:B] (other stages)."""
+        B = len(seq_lens)
+        bucket = self.bucket_for(B)
+        if bucket not in self._graphs:
+            self._capture(bucket)
+
+        self.h_input_ids[:B] = torch.tensor(input_ids, dtype=torch.long)
+        self.h_positions[:B] = torch.tensor(positions, dtype=torch.int32)
+        self.h_slot_mapping[:B] = torch.tensor(slot_mapping, dtype=torch.int64)
+        self.h_slot_mapping[B:bucket] = -1
+        self.h_seq_lens[:B] = torch.tensor(seq_lens, dtype=torch.int32)
+        self.h_seq_lens[B:bucket] = 1
+        self.h_block_tables[:B].zero_()
+        for i, bt in enumerate(block_tables):
+            self.h_block_tables[i, : len(bt)] = torch.tensor(bt, dtype=torch.int32)
+
+        self.input_ids[:bucket].copy_(self.h_input_ids[:bucket], non_blocking=True)
+        self.positions[:bucket].copy_(self.h_positions[:bucket], non_blocking=True)
+        self.slot_mapping[:bucket].copy_(self.h_slot_mapping[:bucket], non_blocking=True)
+        self.seq_lens[:bucket].copy_(self.h_seq_lens[:bucket], non_blocking=True)
+        self.block_tables[:bucket].copy_(self.h_block_tables[:bucket], non_blocking=True)
+        if hidden_in is not None:
+            self.hidden_in[:B].copy_(hidden_in)
+
+        self._graphs[bucket].replay()
+        return self._outputs[bucket][:B]
