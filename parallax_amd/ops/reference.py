@@ -281,50 +281,60 @@ def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
 
 
 def sample_tokens(
-    logits: torch.Tensor,        # [B, vocab]
-    temperatures: torch.Tensor,  # [B]
-    top_ps: torch.Tensor,        # [B]
-    top_ks: torch.Tensor,        # [B] int (-1 = off)
-    min_ps: torch.Tensor,        # [B]
+    logits: torch.Tensor,        # [B, vocab] (any device)
+    temperatures,                # [B] floats (list or CPU tensor)
+    top_ps,
+    top_ks,                      # ints, -1 = off
+    min_ps,
     generator: Optional[torch.Generator] = None,
 ) -> torch.Tensor:
-    """Fused temperature / top-k / top-p / min-p filtering + sampling.
-    Greedy rows (temperature == 0) take argmax."""
+    """Batched temperature / top-k / top-p / min-p filtering + sampling.
+
+    Fully vectorized — sampling params are HOST values (never .item()'d off the
+    device), so the hot path issues no device syncs. Greedy rows
+    (temperature == 0) take argmax."""
     B, V = logits.shape
+    temps = [float(t) for t in temperatures]
+    tps = [float(p) for p in top_ps]
+    tks = [int(k) for k in top_ks]
+    mps = [float(m) for m in min_ps]
+    dev = logits.device
     logits = logits.float()
-    greedy = temperatures <= 0.0
-    out = torch.empty(B, dtype=torch.long, device=logits.device)
-    if greedy.any():
-        out[greedy] = logits[greedy].argmax(dim=-1)
-    rows = (~greedy).nonzero(as_tuple=True)[0]
-    if rows.numel() == 0:
+
+    greedy = [t <= 0.0 for t in temps]
+    need_filter = any(
+        (not g) and (0 < k < V or p < 1.0 or m > 0.0)
+        for g, k, p, m in zip(greedy, tks, tps, mps)
+    )
+    out = logits.argmax(dim=-1)  # greedy default for every row
+    if all(greedy):
         return out
-    lr = logits[rows] / temperatures[rows].unsqueeze(-1).clamp_min(1e-6)
-    probs = torch.softmax(lr, dim=-1)
-    # top-k
-    for j, r in enumerate(rows.tolist()):
-        k = int(top_ks[r])
-        p = probs[j]
-        if 0 < k < V:
-            thresh = torch.topk(p, k).values[-1]
-            p = torch.where(p < thresh, torch.zeros_like(p), p)
-        tp = float(top_ps[r])
-        if tp < 1.0:
-            sorted_p, idx = torch.sort(p, descending=True)
-            cum = torch.cumsum(sorted_p, dim=-1)
-            keep = cum - sorted_p < tp * p.sum()
-            keep[0] = True
-            mask = torch.zeros_like(p, dtype=torch.bool)
-            mask[idx[keep]] = True
-            p = torch.where(mask, p, torch.zeros_like(p))
-        mp = float(min_ps[r])
-        if mp > 0.0:
-            p = torch.where(p < mp * p.max(), torch.zeros_like(p), p)
-        probs[j] = p
-    probs = probs / probs.sum(dim=-1, keepdim=True)
-    sampled = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
-    out[rows] = sampled
-    return out
+
+    t_gpu = torch.tensor(
+        [max(t, 1e-6) for t in temps], dtype=torch.float32, device=dev
+    ).unsqueeze(-1)
+    probs = torch.softmax(logits / t_gpu, dim=-1)
+    if need_filter:
+        sorted_p, idx = torch.sort(probs, dim=-1, descending=True)
+        cum = torch.cumsum(sorted_p, dim=-1)
+        tp_g = torch.tensor(tps, dtype=torch.float32, device=dev).unsqueeze(-1)
+        keep = (cum - sorted_p) < tp_g                       # top-p (first always kept)
+        tk = torch.tensor(
+            [k if 0 < k < V else V for k in tks], dtype=torch.long, device=dev
+        ).unsqueeze(-1)
+        keep &= torch.arange(V, device=dev).unsqueeze(0) < tk  # top-k
+        mp_g = torch.tensor(mps, dtype=torch.float32, device=dev).unsqueeze(-1)
+        keep &= sorted_p >= mp_g * sorted_p[:, :1]             # min-p
+        keep[:, 0] = True
+        sorted_p = torch.where(keep, sorted_p, torch.zeros_like(sorted_p))
+        sorted_p = sorted_p / sorted_p.sum(dim=-1, keepdim=True)
+        picked = torch.multinomial(sorted_p, 1, generator=generator)
+        sampled = idx.gather(-1, picked).squeeze(-1)
+    else:
+        sampled = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+
+    greedy_mask = torch.tensor(greedy, dtype=torch.bool, device=dev)
+    return torch.where(greedy_mask, out, sampled)
 
 
 def apply_penalties(

@@ -378,7 +378,8 @@ class Engine:
                 (len(reqs), self.cfg.hidden_size), self.args.dtype, comm.pp_rank - 1
             )
         out = self.graph_runner.run(input_ids, positions, slots, btabs, seq_lens,
-                                    hidden_in=hidden_in)
+                                    hidden_in=hidden_in,
+                                    rids=[r.rid for r in reqs])
         if not comm.is_last_stage:
             comm.pp_send(out, comm.pp_rank + 1)
             return None

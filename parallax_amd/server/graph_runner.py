@@ -77,6 +77,10 @@ class DecodeGraphRunner:
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._outputs: Dict[int, torch.Tensor] = {}
         self._pool = None
+        # block-table row cache: rows only change when a request crosses a
+        # block boundary or the batch composition changes
+        self._cached_rids: List[str] = []
+        self._cached_btlens: List[int] = []
 
     def _meta(self, bucket: int) -> ForwardMeta:
         return ForwardMeta(
@@ -140,12 +144,14 @@ class DecodeGraphRunner:
         block_tables: List[List[int]],
         seq_lens: List[int],
         hidden_in: Optional[torch.Tensor] = None,
+        rids: Optional[List[str]] = None,
     ) -> torch.Tensor:
         """Returns logits[:B] (last stage) or hidden[:B] (other stages)."""
         B = len(seq_lens)
         bucket = self.bucket_for(B)
         if bucket not in self._graphs:
             self._capture(bucket)
+            self._cached_rids = []
 
         self.h_input_ids[:B] = torch.tensor(input_ids, dtype=torch.long)
         self.h_positions[:B] = torch.tensor(positions, dtype=torch.int32)
@@ -153,15 +159,33 @@ class DecodeGraphRunner:
         self.h_slot_mapping[B:bucket] = -1
         self.h_seq_lens[:B] = torch.tensor(seq_lens, dtype=torch.int32)
         self.h_seq_lens[B:bucket] = 1
-        self.h_block_tables[:B].zero_()
-        for i, bt in enumerate(block_tables):
-            self.h_block_tables[i, : len(bt)] = torch.tensor(bt, dtype=torch.int32)
+
+        btlens = [len(bt) for bt in block_tables]
+        same_batch = rids is not None and rids == self._cached_rids
+        if same_batch:
+            # steady-state decode: copy only rows whose table grew
+            for i, bt in enumerate(block_tables):
+                if btlens[i] != self._cached_btlens[i]:
+                    self.h_block_tables[i, : btlens[i]] = torch.tensor(
+                        bt, dtype=torch.int32
+                    )
+                    self.block_tables[i, : btlens[i]].copy_(
+                        self.h_block_tables[i, : btlens[i]], non_blocking=True
+                    )
+        else:
+            self.h_block_tables[:B].zero_()
+            for i, bt in enumerate(block_tables):
+                self.h_block_tables[i, : btlens[i]] = torch.tensor(bt, dtype=torch.int32)
+            self.block_tables[:bucket].copy_(
+                self.h_block_tables[:bucket], non_blocking=True
+            )
+        self._cached_rids = list(rids) if rids is not None else []
+        self._cached_btlens = btlens
 
         self.input_ids[:bucket].copy_(self.h_input_ids[:bucket], non_blocking=True)
         self.positions[:bucket].copy_(self.h_positions[:bucket], non_blocking=True)
         self.slot_mapping[:bucket].copy_(self.h_slot_mapping[:bucket], non_blocking=True)
         self.seq_lens[:bucket].copy_(self.h_seq_lens[:bucket], non_blocking=True)
-        self.block_tables[:bucket].copy_(self.h_block_tables[:bucket], non_blocking=True)
         if hidden_in is not None:
             self.hidden_in[:B].copy_(hidden_in)
 

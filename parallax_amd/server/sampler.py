@@ -41,13 +41,12 @@ class Sampler:
                 torch.tensor([s.presence_penalty for s in sp], device=logits.device),
                 torch.tensor([s.frequency_penalty for s in sp], device=logits.device),
             )
-        dev = logits.device
         tokens = ops.sample_tokens(
             logits,
-            torch.tensor([s.temperature for s in sp], device=dev),
-            torch.tensor([s.top_p for s in sp], device=dev),
-            torch.tensor([s.top_k for s in sp], device=dev, dtype=torch.long),
-            torch.tensor([s.min_p for s in sp], device=dev),
+            [s.temperature for s in sp],
+            [s.top_p for s in sp],
+            [s.top_k for s in sp],
+            [s.min_p for s in sp],
             generator=self.generator,
         )
         return tokens.tolist()
