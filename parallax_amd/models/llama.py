@@ -71,19 +71,22 @@ class LlamaAttention(nn.Module):
 
     def forward(self, x: torch.Tensor, meta: ForwardMeta, rope_cache: torch.Tensor):
         T = x.shape[0]
-        q, k, v = self.qkv_proj.split_output(self.qkv_proj(x))
+        qkv = self.qkv_proj(x)
+        q, k, v = self.qkv_proj.split_output(qkv)
+        # strided views into the fused QKV buffer — no contiguous copies; the
+        # HIP kernels take a row stride
         q = q.view(T, self.num_heads, self.head_dim)
         k = k.view(T, self.num_kv_heads, self.head_dim)
         v = v.view(T, self.num_kv_heads, self.head_dim)
         if self.qk_norm:
-            q = self.q_norm(q)
-            k = self.k_norm(k)
-        q = q.contiguous()
-        k = k.contiguous()
-        ops.rope_inplace(q, k, meta.positions, rope_cache)
-
+            q = self.q_norm(q.contiguous())
+            k = self.k_norm(k.contiguous())
+            v = v.contiguous()
         k_cache, v_cache = meta.kv_cache.layer(self.local_layer_idx)
-        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+        ops.rope_and_cache(
+            q, k, v, k_cache, v_cache, meta.positions, rope_cache,
+            meta.slot_mapping,
+        )
 
         if meta.is_prefill:
             attn = ops.prefill_attention(

@@ -105,6 +105,30 @@ def rope_inplace(
 # -- cache scatter ---------------------------------------------------------------
 
 
+def rope_and_cache(
+    q: torch.Tensor,             # [T, Hq, D] (rows may be strided, e.g. QKV views)
+    k: torch.Tensor,             # [T, Hk, D]
+    v: torch.Tensor,             # [T, Hk, D]
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    positions: torch.Tensor,
+    cos_sin: torch.Tensor,
+    slot_mapping: torch.Tensor,
+    is_neox: bool = True,
+) -> None:
+    """Fused: rope(q) in place; rope(k) + v scattered straight into the paged
+    cache. One kernel per layer instead of three."""
+    if q.is_cuda:
+        ext = _require_ext("rope_and_cache")
+        ext.rope_and_cache(
+            q, k, v, k_cache, v_cache, positions.to(torch.int32), cos_sin,
+            slot_mapping.to(torch.int64), is_neox,
+        )
+        return
+    ref.rope_inplace(q, k, positions, cos_sin, is_neox)
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
 def reshape_and_cache(
     k: torch.Tensor,
     v: torch.Tensor,
@@ -153,12 +177,12 @@ def paged_attention_decode(
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("paged_attention_decode")
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         if max_seq_len is None:
             max_seq_len = int(seq_lens.max().item())
         ext.paged_attention_decode(
             out,
-            q.contiguous(),
+            q,
             k_cache,
             v_cache,
             block_tables.to(torch.int32),
@@ -189,7 +213,7 @@ def prefill_attention(
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("prefill_attention")
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         ql = query_lens.cpu().tolist()
         cu = [0]
         tile_req, tile_row0 = [], []
@@ -201,7 +225,7 @@ def prefill_attention(
         dev = q.device
         ext.prefill_attention(
             out,
-            q.contiguous(),
+            q,
             k_cache,
             v_cache,
             block_tables.to(torch.int32),

@@ -15,7 +15,11 @@ void launch_rmsnorm(void*, void*, const void*, float, int, int, hipStream_t);
 void launch_fused_add_rmsnorm(void*, void*, const void*, float, int, int,
                               hipStream_t);
 void launch_rope(void*, void*, const int*, const float*, int, int, int, int,
-                 int, bool, hipStream_t);
+                 int, bool, int64_t, int64_t, hipStream_t);
+void launch_rope_and_cache(void*, const void*, const void*, void*, void*,
+                           const int*, const float*, const int64_t*, int, int,
+                           int, int, int, int, bool, int64_t, int64_t, int64_t,
+                           hipStream_t);
 void launch_reshape_and_cache(const void*, const void*, void*, void*,
                               const int64_t*, int, int, int, int, hipStream_t);
 void launch_mla_reshape_and_cache(const void*, const void*, void*,
@@ -24,17 +28,26 @@ void launch_mla_reshape_and_cache(const void*, const void*, void*,
 void launch_act_and_mul(void*, const void*, int64_t, int, bool, hipStream_t);
 void launch_paged_attention_decode(void*, const void*, const void*, const void*,
                                    const int*, const int*, int, int, int, int,
-                                   int, int, float, int, float, const float*,
-                                   int, int, float*, float*, hipStream_t,
-                                   bool*);
+                                   int, int, int64_t, float, int, float,
+                                   const float*, int, int, float*, float*,
+                                   hipStream_t, bool*);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const int*, const int*, const int*, const int*,
-                              const int*, int, int, int, int, int, int, float,
-                              int, float, const float*, hipStream_t, bool*);
+                              const int*, int, int, int, int, int, int, int64_t,
+                              float, int, float, const float*, hipStream_t,
+                              bool*);
 }
 
 static hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
+}
+
+// [T, H, D] view whose rows may live inside a wider fused tensor (e.g. the QKV
+// GEMM output): inner two dims must be dense, the token stride may be larger.
+static int64_t row_stride(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 && t.stride(1) == t.size(2),
+              name, " must be [T, H, D] with dense inner dims");
+  return t.stride(0);
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
@@ -64,7 +77,6 @@ void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual, torch::Tensor w,
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   torch::Tensor cos_sin, bool is_neox) {
   CHECK_GPU(q);
-  CHECK_CONTIG(q);
   CHECK_BF16(q);
   TORCH_CHECK(positions.scalar_type() == at::kInt);
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
@@ -73,9 +85,36 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
   const int D = q.size(2);
   const int Hk = k.numel() > 0 ? k.size(1) : 0;
   const int rot = cos_sin.size(-1);
+  const int64_t qs = row_stride(q, "q");
+  const int64_t ks = Hk ? row_stride(k, "k") : 0;
   launch_rope(q.data_ptr(), Hk ? k.data_ptr() : nullptr,
               positions.data_ptr<int>(), cos_sin.data_ptr<float>(), T, Hq, Hk,
-              D, rot, is_neox, cur_stream());
+              D, rot, is_neox, qs, ks, cur_stream());
+}
+
+void rope_and_cache(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor k_cache, torch::Tensor v_cache,
+                    torch::Tensor positions, torch::Tensor cos_sin,
+                    torch::Tensor slot_mapping, bool is_neox) {
+  CHECK_GPU(q);
+  CHECK_BF16(q);
+  CHECK_BF16(k_cache);
+  TORCH_CHECK(positions.scalar_type() == at::kInt);
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  const int T = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hk = k.size(1);
+  const int BS = k_cache.size(2);
+  const int rot = cos_sin.size(-1);
+  TORCH_CHECK(rot == D || (rot < D && (D - rot) % 8 == 0));
+  launch_rope_and_cache(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+      v_cache.data_ptr(), positions.data_ptr<int>(), cos_sin.data_ptr<float>(),
+      slot_mapping.data_ptr<int64_t>(), T, Hq, Hk, D, rot, BS, is_neox,
+      row_stride(q, "q"), row_stride(k, "k"), row_stride(v, "v"),
+      cur_stream());
 }
 
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
@@ -136,7 +175,6 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             double softcap, torch::Tensor sinks,
                             int64_t max_seq_len) {
   CHECK_GPU(q);
-  CHECK_CONTIG(q);
   CHECK_BF16(q);
   CHECK_BF16(k_cache);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
@@ -182,8 +220,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   launch_paged_attention_decode(
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hq, Hk, D, BS,
-      max_blocks, (float)scale, (int)sliding_window, (float)softcap, sinks_ptr,
-      num_parts, part_tokens, acc_ptr, ml_ptr, cur_stream(), &launched);
+      max_blocks, row_stride(q, "q"), (float)scale, (int)sliding_window,
+      (float)softcap, sinks_ptr, num_parts, part_tokens, acc_ptr, ml_ptr,
+      cur_stream(), &launched);
   TORCH_CHECK(launched, "no kernel instantiation for D=", D, " BS=", BS,
               " G=", G);
 }
@@ -196,7 +235,6 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                        int64_t sliding_window, double softcap,
                        torch::Tensor sinks) {
   CHECK_GPU(q);
-  CHECK_CONTIG(q);
   CHECK_BF16(q);
   CHECK_BF16(k_cache);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
@@ -221,8 +259,8 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
       cu_q.data_ptr<int>(), tile_req.data_ptr<int>(), tile_row0.data_ptr<int>(),
-      n_tiles, Hq, Hk, D, BS, max_blocks, (float)scale, (int)sliding_window,
-      (float)softcap, sinks_ptr, cur_stream(), &launched);
+      n_tiles, Hq, Hk, D, BS, max_blocks, row_stride(q, "q"), (float)scale,
+      (int)sliding_window, (float)softcap, sinks_ptr, cur_stream(), &launched);
   TORCH_CHECK(launched, "no prefill kernel instantiation for D=", D);
 }
 
@@ -231,6 +269,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_inplace", &rope_inplace);
+  m.def("rope_and_cache", &rope_and_cache);
   m.def("reshape_and_cache", &reshape_and_cache);
   m.def("mla_reshape_and_cache", &mla_reshape_and_cache);
   m.def("silu_and_mul", &silu_and_mul);

@@ -86,7 +86,8 @@ template <bool NEOX>
 __global__ void rope_kernel(
     uint16_t* __restrict__ q, uint16_t* __restrict__ k,
     const int* __restrict__ positions, const float* __restrict__ cos_sin,
-    const int Hq, const int Hk, const int D, const int rot) {
+    const int Hq, const int Hk, const int D, const int rot,
+    const int64_t q_stride, const int64_t k_stride) {
   const int t = blockIdx.x;
   const int half = rot / 2;
   const float* cs = cos_sin + (size_t)positions[t] * rot;
@@ -94,8 +95,8 @@ __global__ void rope_kernel(
   for (int i = threadIdx.x; i < total; i += blockDim.x) {
     const int h = i / half;
     const int j = i % half;
-    uint16_t* base = (h < Hq) ? q + ((size_t)t * Hq + h) * D
-                              : k + ((size_t)t * Hk + (h - Hq)) * D;
+    uint16_t* base = (h < Hq) ? q + (size_t)t * q_stride + h * D
+                              : k + (size_t)t * k_stride + (h - Hq) * D;
     const float c = cs[j], s = cs[half + j];
     int i1, i2;
     if (NEOX) {
@@ -112,14 +113,102 @@ __global__ void rope_kernel(
 
 extern "C" void launch_rope(
     void* q, void* k, const int* positions, const float* cos_sin,
-    int T, int Hq, int Hk, int D, int rot, bool neox, hipStream_t stream) {
+    int T, int Hq, int Hk, int D, int rot, bool neox, int64_t q_stride,
+    int64_t k_stride, hipStream_t stream) {
   int threads = min(512, max(64, ceil_div((Hq + Hk) * rot / 2, 64) * 64));
   if (neox)
    hipLaunchKernelGGL(( rope_kernel<true>), dim3(T), dim3(threads), 0, stream, 
-        (uint16_t*)q, (uint16_t*)k, positions, cos_sin, Hq, Hk, D, rot);
+        (uint16_t*)q, (uint16_t*)k, positions, cos_sin, Hq, Hk, D, rot,
+        q_stride, k_stride);
   else
    hipLaunchKernelGGL(( rope_kernel<false>), dim3(T), dim3(threads), 0, stream, 
-        (uint16_t*)q, (uint16_t*)k, positions, cos_sin, Hq, Hk, D, rot);
+        (uint16_t*)q, (uint16_t*)k, positions, cos_sin, Hq, Hk, D, rot,
+        q_stride, k_stride);
+}
+
+// ---- fused rope + cache scatter ----------------------------------------------------
+// One launch per decode/prefill step replaces {rope(q,k) ; reshape_and_cache}:
+// q roped in place (strided rows OK), k roped straight into the paged cache,
+// v copied vectorized. Saves one k round-trip and two kernel launches per layer.
+template <bool NEOX>
+__global__ void rope_and_cache_kernel(
+    uint16_t* __restrict__ q,        // [T, Hq, D] rows at q_stride
+    const uint16_t* __restrict__ k,  // [T, Hk, D] rows at k_stride
+    const uint16_t* __restrict__ v,
+    uint16_t* __restrict__ k_cache,  // [NB, Hk, BS, D]
+    uint16_t* __restrict__ v_cache,
+    const int* __restrict__ positions, const float* __restrict__ cos_sin,
+    const int64_t* __restrict__ slot_mapping,
+    const int Hq, const int Hk, const int D, const int rot, const int BS,
+    const int64_t q_stride, const int64_t k_stride, const int64_t v_stride) {
+  const int t = blockIdx.x;
+  const int64_t slot = slot_mapping[t];
+  const int half = rot / 2;
+  const float* cs = cos_sin + (size_t)positions[t] * rot;
+  const int64_t blk = slot >= 0 ? slot / BS : 0;
+  const int64_t off = slot >= 0 ? slot % BS : 0;
+  uint16_t* kdst = k_cache + (((size_t)blk * Hk) * BS + off) * D;
+  uint16_t* vdst = v_cache + (((size_t)blk * Hk) * BS + off) * D;
+
+  // q: rope in place
+  for (int i = threadIdx.x; i < Hq * half; i += blockDim.x) {
+    const int h = i / half, j = i % half;
+    uint16_t* base = q + (size_t)t * q_stride + h * D;
+    const float c = cs[j], ss = cs[half + j];
+    const int i1 = NEOX ? j : 2 * j;
+    const int i2 = NEOX ? j + half : 2 * j + 1;
+    const float x1 = bf16_bits_to_f32(base[i1]);
+    const float x2 = bf16_bits_to_f32(base[i2]);
+    base[i1] = f32_to_bf16_bits(fmaf(x1, c, -x2 * ss));
+    base[i2] = f32_to_bf16_bits(fmaf(x2, c, x1 * ss));
+  }
+  if (slot < 0) return;
+  // k: rope -> cache (rot dims) + copy (pass-through dims if rot < D)
+  for (int i = threadIdx.x; i < Hk * half; i += blockDim.x) {
+    const int h = i / half, j = i % half;
+    const uint16_t* src = k + (size_t)t * k_stride + h * D;
+    uint16_t* dst = kdst + (size_t)h * BS * D;
+    const float c = cs[j], ss = cs[half + j];
+    const int i1 = NEOX ? j : 2 * j;
+    const int i2 = NEOX ? j + half : 2 * j + 1;
+    const float x1 = bf16_bits_to_f32(src[i1]);
+    const float x2 = bf16_bits_to_f32(src[i2]);
+    dst[i1] = f32_to_bf16_bits(fmaf(x1, c, -x2 * ss));
+    dst[i2] = f32_to_bf16_bits(fmaf(x2, c, x1 * ss));
+  }
+  if (rot < D) {
+    for (int i = threadIdx.x; i < Hk * (D - rot) / 8; i += blockDim.x) {
+      const int h = (i * 8) / (D - rot);
+      const int d = rot + (i * 8) % (D - rot);
+      *reinterpret_cast<int4*>(kdst + (size_t)h * BS * D + d) =
+          *reinterpret_cast<const int4*>(k + (size_t)t * k_stride + h * D + d);
+    }
+  }
+  // v: straight vectorized copy into the cache
+  for (int i = threadIdx.x; i < Hk * D / 8; i += blockDim.x) {
+    const int h = (i * 8) / D;
+    const int d = (i * 8) % D;
+    *reinterpret_cast<int4*>(vdst + (size_t)h * BS * D + d) =
+        *reinterpret_cast<const int4*>(v + (size_t)t * v_stride + h * D + d);
+  }
+}
+
+extern "C" void launch_rope_and_cache(
+    void* q, const void* k, const void* v, void* k_cache, void* v_cache,
+    const int* positions, const float* cos_sin, const int64_t* slot_mapping,
+    int T, int Hq, int Hk, int D, int rot, int BS, bool neox,
+    int64_t q_stride, int64_t k_stride, int64_t v_stride, hipStream_t stream) {
+  int threads = min(512, max(128, ceil_div(Hk * D / 8, 64) * 64));
+  if (neox)
+   hipLaunchKernelGGL(( rope_and_cache_kernel<true>), dim3(T), dim3(threads), 0, stream, 
+        (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
+        (uint16_t*)k_cache, (uint16_t*)v_cache, positions, cos_sin,
+        slot_mapping, Hq, Hk, D, rot, BS, q_stride, k_stride, v_stride);
+  else
+   hipLaunchKernelGGL(( rope_and_cache_kernel<false>), dim3(T), dim3(threads), 0, stream, 
+        (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
+        (uint16_t*)k_cache, (uint16_t*)v_cache, positions, cos_sin,
+        slot_mapping, Hq, Hk, D, rot, BS, q_stride, k_stride, v_stride);
 }
 
 // ---- reshape_and_cache ------------------------------------------------------------

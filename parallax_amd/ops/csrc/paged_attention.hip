@@ -36,6 +36,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
     const int* __restrict__ seq_lens,      // [B]
     const int max_blocks,
     const int Hk, const int G,
+    const int64_t q_stride,
     const float scale,
     const int sliding_window,           // <=0: full
     const float softcap,                // <=0: off
@@ -86,7 +87,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
   for (int i = tid; i < G * HEAD_DIM; i += ATTN_THREADS) {
     const int g = i / HEAD_DIM, d = i % HEAD_DIM;
     q_lds[g][d] =
-        bf16_bits_to_f32(q[((size_t)seq * Hk * G + hk * G + g) * HEAD_DIM + d]) * scale;
+        bf16_bits_to_f32(q[(size_t)seq * q_stride + (hk * G + g) * HEAD_DIM + d]) * scale;
   }
   if (tid < GMAX) {
     m_lds[tid] = -1e30f;
@@ -336,9 +337,9 @@ __global__ void paged_attention_reduce_kernel(
 extern "C" void launch_paged_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* seq_lens, int B, int Hq, int Hk, int D,
-    int BS, int max_blocks, float scale, int sliding_window, float softcap,
-    const float* sinks, int num_parts, int part_tokens, float* tmp_acc,
-    float* tmp_ml, hipStream_t stream, bool* launched) {
+    int BS, int max_blocks, int64_t q_stride, float scale, int sliding_window,
+    float softcap, const float* sinks, int num_parts, int part_tokens,
+    float* tmp_acc, float* tmp_ml, hipStream_t stream, bool* launched) {
   const int G = Hq / Hk;
   *launched = false;
   if (num_parts <= 1) {
@@ -348,8 +349,8 @@ extern "C" void launch_paged_attention_decode(
           <<<grid, ATTN_THREADS, 0, stream>>>(
               (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q,
               (const uint16_t*)k_cache, (const uint16_t*)v_cache, block_tables,
-              seq_lens, max_blocks, Hk, G, scale, sliding_window, softcap,
-              sinks, 0);
+              seq_lens, max_blocks, Hk, G, q_stride, scale, sliding_window,
+              softcap, sinks, 0);
       *launched = true;
     })));
   } else {
@@ -359,8 +360,8 @@ extern "C" void launch_paged_attention_decode(
           <<<grid, ATTN_THREADS, 0, stream>>>(
               nullptr, tmp_acc, tmp_ml, (const uint16_t*)q,
               (const uint16_t*)k_cache, (const uint16_t*)v_cache, block_tables,
-              seq_lens, max_blocks, Hk, G, scale, sliding_window, softcap,
-              sinks, part_tokens);
+              seq_lens, max_blocks, Hk, G, q_stride, scale, sliding_window,
+              softcap, sinks, part_tokens);
       dim3 rgrid(Hq, B, 1);
       paged_attention_reduce_kernel<HEAD_DIM><<<rgrid, 64, 0, stream>>>(
           (uint16_t*)out, tmp_acc, tmp_ml, num_parts, sinks, Hq);

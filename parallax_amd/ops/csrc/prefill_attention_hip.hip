@@ -40,8 +40,8 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
     const int* __restrict__ tile_req,      // [n_tiles] request of each q tile
     const int* __restrict__ tile_row0,     // [n_tiles] first q row of the tile
     const int max_blocks, const int Hq, const int Hk, const int BS,
-    const float scale, const int sliding_window, const float softcap,
-    const float* __restrict__ sinks) {
+    const int64_t q_stride, const float scale, const int sliding_window,
+    const float softcap, const float* __restrict__ sinks) {
   const int h = blockIdx.x;
   const int tile = blockIdx.y;
   const int req = tile_req[tile];
@@ -77,7 +77,7 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
       int4 val = make_int4(0, 0, 0, 0);
       if (ok)
         val = *reinterpret_cast<const int4*>(
-            q + ((size_t)(q0 + r0 + qrow) * Hq + h) * HEAD_DIM + d + c * 8);
+            q + (size_t)(q0 + r0 + qrow) * q_stride + h * HEAD_DIM + d + c * 8);
       const int byte = swz(qrow * HEAD_DIM * 2 + (d + c * 8) * 2, qrow);
       *reinterpret_cast<int4*>(reinterpret_cast<char*>(Ql) + byte) = val;
     }
@@ -270,23 +270,24 @@ extern "C" void launch_prefill_attention(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* seq_lens, const int* cu_q,
     const int* tile_req, const int* tile_row0, int n_tiles, int Hq, int Hk,
-    int D, int BS, int max_blocks, float scale, int sliding_window,
-    float softcap, const float* sinks, hipStream_t stream, bool* launched) {
+    int D, int BS, int max_blocks, int64_t q_stride, float scale,
+    int sliding_window, float softcap, const float* sinks, hipStream_t stream,
+    bool* launched) {
   *launched = false;
   dim3 grid(Hq, n_tiles, 1);
   if (D == 128) {
    hipLaunchKernelGGL(( prefill_attention_kernel<128>), dim3(grid), dim3(PF_THREADS), 0, stream, 
         (uint16_t*)out, (const uint16_t*)q, (const uint16_t*)k_cache,
         (const uint16_t*)v_cache, block_tables, seq_lens, cu_q, tile_req,
-        tile_row0, max_blocks, Hq, Hk, BS, scale, sliding_window, softcap,
-        sinks);
+        tile_row0, max_blocks, Hq, Hk, BS, q_stride, scale, sliding_window,
+        softcap, sinks);
     *launched = true;
   } else if (D == 64) {
    hipLaunchKernelGGL(( prefill_attention_kernel<64>), dim3(grid), dim3(PF_THREADS), 0, stream, 
         (uint16_t*)out, (const uint16_t*)q, (const uint16_t*)k_cache,
         (const uint16_t*)v_cache, block_tables, seq_lens, cu_q, tile_req,
-        tile_row0, max_blocks, Hq, Hk, BS, scale, sliding_window, softcap,
-        sinks);
+        tile_row0, max_blocks, Hq, Hk, BS, q_stride, scale, sliding_window,
+        softcap, sinks);
     *launched = true;
   }
 }

@@ -4,7 +4,11 @@ bookkeeping. Run on a GPU box:
 """
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
@@ -12,8 +16,6 @@ from parallax_amd.models.config import ModelConfig
 from parallax_amd.server.engine import Engine, EngineArgs
 from parallax_amd.server.sampling_params import SamplingParams
 
-import sys, os
-sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from bench import deepseek_r1_distill_llama_8b  # noqa: E402
 
 

@@ -201,3 +201,50 @@ def test_engine_gpu_decode_deterministic():
     o1, o2 = run(), run()
     assert [len(v) for v in o1.values()] == [6, 6]
     assert list(o1.values()) == list(o2.values())
+
+
+def test_rope_and_cache_fused():
+    """Fused rope+cache (strided qkv views) vs separate reference ops."""
+    torch.manual_seed(9)
+    T, Hq, Hk, D, BS, NB = 33, 8, 2, 128, 32, 16
+    qkv = torch.randn(T, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device="cuda")
+    q = qkv[:, : Hq * D].view(T, Hq, D)
+    k = qkv[:, Hq * D : (Hq + Hk) * D].view(T, Hk, D)
+    v = qkv[:, (Hq + Hk) * D :].view(T, Hk, D)
+    kc = torch.zeros(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros_like(kc)
+    pos = torch.randint(0, 500, (T,), dtype=torch.int32, device="cuda")
+    cs = ops.build_rope_cache(512, D, 10000.0).cuda()
+    slots = torch.randperm(NB * BS, device="cuda")[:T].to(torch.int64)
+    slots[3] = -1
+
+    q_ref = q.float().cpu().clone()
+    k_ref = k.float().cpu().clone()
+    kc_ref = torch.zeros(NB, Hk, BS, D).float()
+    vc_ref = torch.zeros_like(kc_ref)
+    ref.rope_inplace(q_ref, k_ref, pos.cpu(), cs.cpu(), True)
+    ref.reshape_and_cache(k_ref, v.float().cpu(), kc_ref, vc_ref, slots.cpu())
+
+    ops.rope_and_cache(q, k, v, kc, vc, pos, cs, slots)
+    _assert_close(q, q_ref)
+    _assert_close(kc, kc_ref)
+    _assert_close(vc, vc_ref)
+
+
+def test_paged_attention_strided_q():
+    torch.manual_seed(10)
+    Hk, G, D, BS = 2, 4, 128, 32
+    Hq = G * Hk
+    B = 3
+    ctxs = [100, 40, 7]
+    kc, vc, bt = _make_paged_kv(B, Hk, D, BS, max(ctxs))
+    qkv = torch.randn(B, (Hq + 4) * D, dtype=torch.bfloat16, device="cuda")
+    q = qkv[:, : Hq * D].view(B, Hq, D)  # strided rows
+    seq_lens = torch.tensor(ctxs, dtype=torch.int32, device="cuda")
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, kc, vc, bt, seq_lens, scale)
+    expect = ref.paged_attention_decode(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(),
+        seq_lens.cpu(), scale,
+    )
+    _assert_close(out, expect, atol=3e-2, rtol=3e-2)
