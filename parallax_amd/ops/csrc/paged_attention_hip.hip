@@ -80,7 +80,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
 
   __shared__ float q_lds[GMAX][HEAD_DIM];
   __shared__ float p_lds[GMAX][CHUNK_TOKENS];
-  __shared__ float red_lds[ATTN_THREADS * 8];  // [NT_PAR][HEAD_DIM] final reduce
+  __shared__ float redw[GMAX][4][HEAD_DIM];  // per-wave output partials
   __shared__ float m_lds[GMAX], l_lds[GMAX], rescale_lds[GMAX];
   __shared__ float wmax_lds[4][GMAX], wsum_lds[4][GMAX];
 
@@ -183,18 +183,29 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) acc[g][j] *= r;
     }
-    for (int it = 0; it < CHUNK_TOKENS / NT_PAR; ++it) {
+    // preload every V vector for this chunk first so the HBM latencies
+    // overlap (a guarded load inside the FMA loop serializes ~900 cy each)
+    constexpr int VIT = CHUNK_TOKENS / NT_PAR;
+    bf16x8 vv[VIT];
+#pragma unroll
+    for (int it = 0; it < VIT; ++it) {
+      const int tok2 = base_tok + tp + it * NT_PAR;
+      if (tok2 >= tok_begin && tok2 < tok_end) {
+        const int blk = btab[tok2 / BLOCK_SIZE];
+        const int off = tok2 % BLOCK_SIZE;
+        vv[it] = load_bf16x8(
+            v_cache + (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM +
+            dc * 8);
+      } else {
+        vv[it].raw = make_int4(0, 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int it = 0; it < VIT; ++it) {
       const int tt = tp + it * NT_PAR;
-      const int tok2 = base_tok + tt;
-      if (tok2 < tok_begin || tok2 >= tok_end) continue;
-      const int blk = btab[tok2 / BLOCK_SIZE];
-      const int off = tok2 % BLOCK_SIZE;
-      const bf16x8 vv = load_bf16x8(
-          v_cache + (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM +
-          dc * 8);
       float vf[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vf[j] = bf16x8_get(vv, j);
+      for (int j = 0; j < 8; ++j) vf[j] = bf16x8_get(vv[it], j);
 #pragma unroll
       for (int g = 0; g < GMAX; ++g) {
         if (g >= G) break;
@@ -206,45 +217,52 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
     __syncthreads();  // p_lds reused next chunk
   }
 
-  // ---- final: reduce the NT_PAR partial accumulators, normalize, write ----------
-  float* red = red_lds;  // [NT_PAR][HEAD_DIM] fp32
-  static_assert(NT_PAR * HEAD_DIM == ATTN_THREADS * 8, "reduce buffer sized");
-  for (int g = 0; g < G; ++g) {
+  // ---- final reduce: shfl across the tp groups inside each wave (no LDS),
+  // then one barrier to combine the 4 wave partials ------------------------------
 #pragma unroll
-    for (int j = 0; j < 8; ++j) red[tp * HEAD_DIM + dc * 8 + j] = acc[g][j];
-    __syncthreads();
-    for (int s = NT_PAR / 2; s > 0; s >>= 1) {
-      if (tp < s) {
+  for (int g = 0; g < GMAX; ++g) {
+    if (g >= G) break;
+    for (int off = DC; off < WAVE_SIZE; off <<= 1) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          red[tp * HEAD_DIM + dc * 8 + j] += red[(tp + s) * HEAD_DIM + dc * 8 + j];
-      }
-      __syncthreads();
+      for (int j = 0; j < 8; ++j)
+        acc[g][j] += __shfl_xor(acc[g][j], off, WAVE_SIZE);
     }
-    const size_t hq = (size_t)hk * G + g;
+    if (lane < DC) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) redw[g][wid][dc * 8 + j] = acc[g][j];
+    }
+  }
+  __syncthreads();
+
+  // one thread per (g, dc) combines the 4 wave partials and writes out
+  const int g2 = tid / DC;
+  const int dc2 = tid % DC;
+  if (g2 < G) {
+    const size_t hq = (size_t)hk * G + g2;
+    float vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      vals[j] = redw[g2][0][dc2 * 8 + j] + redw[g2][1][dc2 * 8 + j] +
+                redw[g2][2][dc2 * 8 + j] + redw[g2][3][dc2 * 8 + j];
     if (PARTITIONED) {
       const int p = blockIdx.z;
-      if (tp == 0) {
-        float* dst =
-            tmp_acc + (((size_t)seq * Hk * G + hq) * gridDim.z + p) * HEAD_DIM;
+      float* dst =
+          tmp_acc + (((size_t)seq * Hk * G + hq) * gridDim.z + p) * HEAD_DIM;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) dst[dc * 8 + j] = red[dc * 8 + j];
-      }
-      if (tid == 0) {
+      for (int j = 0; j < 8; ++j) dst[dc2 * 8 + j] = vals[j];
+      if (dc2 == 0) {
         float* ml = tmp_ml + (((size_t)seq * Hk * G + hq) * gridDim.z + p) * 2;
-        ml[0] = m_lds[g];
-        ml[1] = l_lds[g];
+        ml[0] = m_lds[g2];
+        ml[1] = l_lds[g2];
       }
-    } else if (tp == 0) {
-      float l = l_lds[g];
-      if (sinks != nullptr) l += __expf(sinks[hq] - m_lds[g]);
+    } else {
+      float l = l_lds[g2];
+      if (sinks != nullptr) l += __expf(sinks[hq] - m_lds[g2]);
       const float inv = 1.f / l;
-      float vals[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vals[j] = red[dc * 8 + j] * inv;
-      store_bf16x8(out + ((size_t)seq * Hk * G + hq) * HEAD_DIM + dc * 8, vals);
+      for (int j = 0; j < 8; ++j) vals[j] *= inv;
+      store_bf16x8(out + ((size_t)seq * Hk * G + hq) * HEAD_DIM + dc2 * 8, vals);
     }
-    __syncthreads();
   }
 }
 
