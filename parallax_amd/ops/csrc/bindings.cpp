@@ -5,6 +5,8 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 #define CHECK_BF16(x) \
@@ -31,6 +33,12 @@ void launch_paged_attention_decode(void*, const void*, const void*, const void*,
                                    int, int, int64_t, float, int, float,
                                    const float*, int, int, float*, float*,
                                    hipStream_t, bool*);
+void launch_paged_decode_mfma(void*, const void*, const void*, const void*,
+                              const int*, const int*, int, int, int, int, int,
+                              int, int64_t, float, int, float, const float*,
+                              int, int, float*, float*, hipStream_t, bool*);
+void launch_paged_attention_reduce(void*, const float*, const float*, int, int,
+                                   int, int, const float*, hipStream_t);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const int*, const int*, const int*, const int*,
                               const int*, int, int, int, int, int, int, int64_t,
@@ -217,14 +225,28 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     ml_ptr = tmp_ml.data_ptr<float>();
   }
   bool launched = false;
-  launch_paged_attention_decode(
-      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hq, Hk, D, BS,
-      max_blocks, row_stride(q, "q"), (float)scale, (int)sliding_window,
-      (float)softcap, sinks_ptr, num_parts, part_tokens, acc_ptr, ml_ptr,
-      cur_stream(), &launched);
+  static const bool use_valu = std::getenv("PARALLAX_ATTN_VALU") != nullptr;
+  if (!use_valu) {
+    launch_paged_decode_mfma(
+        out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hq, Hk, D,
+        BS, max_blocks, row_stride(q, "q"), (float)scale, (int)sliding_window,
+        (float)softcap, sinks_ptr, num_parts, part_tokens, acc_ptr, ml_ptr,
+        cur_stream(), &launched);
+  }
+  if (!launched) {
+    launch_paged_attention_decode(
+        out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hq, Hk, D,
+        BS, max_blocks, row_stride(q, "q"), (float)scale, (int)sliding_window,
+        (float)softcap, sinks_ptr, num_parts, part_tokens, acc_ptr, ml_ptr,
+        cur_stream(), &launched);
+  }
   TORCH_CHECK(launched, "no kernel instantiation for D=", D, " BS=", BS,
               " G=", G);
+  if (num_parts > 1 && launched)
+    launch_paged_attention_reduce(out.data_ptr(), acc_ptr, ml_ptr, B, Hq, D,
+                                  num_parts, sinks_ptr, cur_stream());
 }
 
 void prefill_attention(torch::Tensor out, torch::Tensor q,

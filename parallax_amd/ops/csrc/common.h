@@ -105,6 +105,13 @@ DEVINL float block_reduce_sum(float v, float* lds_scratch) {
   return lds_scratch[0];
 }
 
+typedef short bf16x8v __attribute__((ext_vector_type(8)));
+typedef float f32x4v __attribute__((ext_vector_type(4)));
+
+// XOR-swizzle a byte offset within a row-major LDS tile: spreads the 16-lane
+// row-stride access pattern across banks (guide §6 G4)
+DEVINL int swz(int byte_in, int row) { return byte_in ^ ((row & 7) << 4); }
+
 __host__ __device__ __forceinline__ int ceil_div(int a, int b) {
   return (a + b - 1) / b;
 }

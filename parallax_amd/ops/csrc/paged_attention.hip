@@ -352,6 +352,18 @@ __global__ void paged_attention_reduce_kernel(
     __VA_ARGS__;                       \
   }
 
+extern "C" void launch_paged_attention_reduce(
+    void* out, const float* tmp_acc, const float* tmp_ml, int B, int Hq, int D,
+    int num_parts, const float* sinks, hipStream_t stream) {
+  dim3 rgrid(Hq, B, 1);
+  if (D == 128)
+    paged_attention_reduce_kernel<128><<<rgrid, 64, 0, stream>>>(
+        (uint16_t*)out, tmp_acc, tmp_ml, num_parts, sinks, Hq);
+  else if (D == 64)
+    paged_attention_reduce_kernel<64><<<rgrid, 64, 0, stream>>>(
+        (uint16_t*)out, tmp_acc, tmp_ml, num_parts, sinks, Hq);
+}
+
 extern "C" void launch_paged_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* seq_lens, int B, int Hq, int Hk, int D,
@@ -380,9 +392,6 @@ extern "C" void launch_paged_attention_decode(
               (const uint16_t*)k_cache, (const uint16_t*)v_cache, block_tables,
               seq_lens, max_blocks, Hk, G, q_stride, scale, sliding_window,
               softcap, sinks, part_tokens);
-      dim3 rgrid(Hq, B, 1);
-      paged_attention_reduce_kernel<HEAD_DIM><<<rgrid, 64, 0, stream>>>(
-          (uint16_t*)out, tmp_acc, tmp_ml, num_parts, sinks, Hq);
       *launched = true;
     })));
   }

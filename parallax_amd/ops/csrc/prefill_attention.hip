@@ -22,11 +22,6 @@
 #define QTILE 32
 #define KTILE 64
 
-typedef short bf16x8v __attribute__((ext_vector_type(8)));
-typedef float f32x4v __attribute__((ext_vector_type(4)));
-
-DEVINL int swz(int byte_in, int row) { return byte_in ^ ((row & 7) << 4); }
-
 template <int HEAD_DIM>
 __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
     uint16_t* __restrict__ out,            // [T, Hq, D]

@@ -18,7 +18,7 @@ from parallax_amd import ops
 def run_case(B, Hk, G, D, ctx, max_seq_len, BS=32, iters=50):
     torch.manual_seed(0)
     Hq = Hk * G
-    max_blocks = (max_seq_len + BS - 1) // BS
+    max_blocks = (max(max_seq_len, ctx) + BS - 1) // BS
     nb = B * ((ctx + BS - 1) // BS) + 1
     kc = torch.randn(nb, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
     vc = torch.randn(nb, Hk, BS, D, dtype=torch.bfloat16, device="cuda")

@@ -20,6 +20,7 @@ sources = [
     os.path.join(CSRC, "bindings.cpp"),
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "paged_attention.hip"),
+    os.path.join(CSRC, "paged_attention_mfma.hip"),
     os.path.join(CSRC, "prefill_attention.hip"),
     os.path.join(CSRC, "moe.hip"),
 ]
