@@ -438,14 +438,14 @@ class Engine:
         done: Dict[str, List[int]] = {}
         outputs: Dict[str, List[int]] = {rid: [] for rid in rids}
         for _ in range(max_steps):
-            if not self.has_work:
-                break
             for out in self.step():
                 if out.rid in outputs:
                     outputs[out.rid].append(out.token_id)
                     if out.finished:
                         done[out.rid] = outputs[out.rid]
-            if len(done) == len(rids):
+            # check AFTER the step: step() syncs ingress across ranks, so the
+            # loop exits on the same iteration on every pipeline stage
+            if not self.has_work or len(done) == len(rids):
                 break
         return {rid: outputs[rid] for rid in rids}
 
