@@ -1,0 +1,115 @@
+"""Threaded engine host: runs the engine step loop and demultiplexes token
+output to per-request consumers (the seam between the async HTTP frontend and
+the synchronous engine).
+
+Reference analogue: the vllm-rs frontend <-> executor engine-core ZMQ boundary
+(engine_core_protocol.py). MI355X-first design: the single-host frontend shares
+the process (no serialize hop); the multi-host path uses p2p/message codecs.
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from ..utils.logging_config import get_logger
+from .engine import Engine, StepOutput
+from .request import new_request_id
+from .sampling_params import SamplingParams
+
+logger = get_logger("server.engine_server")
+
+
+@dataclass
+class RequestStream:
+    rid: str
+    out_queue: "queue.Queue[Optional[StepOutput]]" = field(
+        default_factory=lambda: queue.Queue()
+    )
+    created: float = field(default_factory=time.monotonic)
+
+
+class EngineServer:
+    """Owns the engine step thread. submit() is thread-safe; consumers read
+    StepOutputs from their stream queue (None terminates)."""
+
+    def __init__(self, engine: Engine, idle_sleep_s: float = 0.002):
+        self.engine = engine
+        self.idle_sleep_s = idle_sleep_s
+        self._streams: Dict[str, RequestStream] = {}
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        # serving metrics (parity with reference request_metrics)
+        self.total_requests = 0
+        self.total_output_tokens = 0
+
+    # -- lifecycle ------------------------------------------------------------
+
+    def start(self) -> None:
+        assert self._thread is None
+        self._thread = threading.Thread(target=self._run_loop, daemon=True,
+                                        name="engine-step-loop")
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=10)
+            self._thread = None
+
+    # -- ingress ----------------------------------------------------------------
+
+    def submit(
+        self,
+        prompt_token_ids: List[int],
+        sampling_params: SamplingParams,
+        rid: Optional[str] = None,
+    ) -> RequestStream:
+        rid = rid or new_request_id()
+        stream = RequestStream(rid=rid)
+        with self._lock:
+            self._streams[rid] = stream
+            self.engine.submit(prompt_token_ids, sampling_params, rid=rid)
+            self.total_requests += 1
+        return stream
+
+    def abort(self, rid: str) -> None:
+        with self._lock:
+            self.engine.abort(rid)
+
+    # -- the loop -----------------------------------------------------------------
+
+    def _run_loop(self) -> None:
+        logger.info("engine step loop running")
+        while not self._stop.is_set():
+            with self._lock:
+                has_work = self.engine.has_work
+            if not has_work:
+                time.sleep(self.idle_sleep_s)
+                continue
+            with self._lock:
+                outputs = self.engine.step()
+            for out in outputs:
+                self.total_output_tokens += 1
+                stream = self._streams.get(out.rid)
+                if stream is not None:
+                    stream.out_queue.put(out)
+                    if out.finished:
+                        stream.out_queue.put(None)
+                        with self._lock:
+                            self._streams.pop(out.rid, None)
+        logger.info("engine step loop stopped")
+
+    def stats(self) -> dict:
+        return {
+            "total_requests": self.total_requests,
+            "total_output_tokens": self.total_output_tokens,
+            "running": self.engine.scheduler.num_running,
+            "waiting": len(self.engine.scheduler.wait_queue),
+            "free_kv_blocks": self.engine.cache_manager.num_free_blocks,
+            "prefix_cache_hit_rate": self.engine.cache_manager.radix.hit_rate,
+        }

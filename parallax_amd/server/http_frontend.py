@@ -1,0 +1,165 @@
+"""OpenAI-compatible HTTP frontend (FastAPI).
+
+Endpoints (parity with the reference's vllm-rs frontend + backend API surface):
+POST /v1/chat/completions (stream + non-stream), POST /v1/completions,
+GET /v1/models, GET /health, GET /stats.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import List, Optional
+
+from fastapi import FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from ..utils.logging_config import get_logger
+from .engine_server import EngineServer
+from .sampling_params import SamplingParams
+from .tokenizer_util import TokenizerWrapper
+
+logger = get_logger("server.http_frontend")
+
+
+def create_app(
+    server: EngineServer,
+    tokenizer: TokenizerWrapper,
+    model_name: str = "parallax-amd-model",
+) -> FastAPI:
+    app = FastAPI(title="parallax_amd", version="0.1.0")
+
+    def _params(body: dict) -> SamplingParams:
+        sp = SamplingParams.from_openai(body)
+        if tokenizer.eos_token_id is not None and not sp.stop_token_ids:
+            sp.stop_token_ids = []
+        return sp
+
+    async def _collect(stream, sp: SamplingParams):
+        """Drain a request stream fully (non-streaming path)."""
+        loop = asyncio.get_event_loop()
+        token_ids: List[int] = []
+        finish_reason = "stop"
+        while True:
+            out = await loop.run_in_executor(None, stream.out_queue.get)
+            if out is None:
+                break
+            token_ids.append(out.token_id)
+            if out.finished:
+                finish_reason = out.finish_reason or "stop"
+        return token_ids, finish_reason
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/v1/models")
+    async def models():
+        return {
+            "object": "list",
+            "data": [{"id": model_name, "object": "model", "owned_by": "parallax_amd"}],
+        }
+
+    @app.get("/stats")
+    async def stats():
+        return server.stats()
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        body = await request.json()
+        messages = body.get("messages")
+        if not messages:
+            raise HTTPException(400, "messages required")
+        prompt_ids = tokenizer.chat_prompt_ids(messages)
+        sp = _params(body)
+        rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        stream = server.submit(prompt_ids, sp, rid=rid)
+        created = int(time.time())
+
+        if body.get("stream"):
+            async def sse():
+                loop = asyncio.get_event_loop()
+                token_ids: List[int] = []
+                sent_len = 0
+                t_start = time.monotonic()
+                first_token_t = None
+                while True:
+                    out = await loop.run_in_executor(None, stream.out_queue.get)
+                    if out is None:
+                        break
+                    if first_token_t is None:
+                        first_token_t = time.monotonic()
+                    token_ids.append(out.token_id)
+                    text = tokenizer.decode(token_ids)
+                    delta, sent_len = text[sent_len:], len(text)
+                    chunk = {
+                        "id": rid, "object": "chat.completion.chunk",
+                        "created": created, "model": model_name,
+                        "choices": [{
+                            "index": 0,
+                            "delta": {"content": delta},
+                            "finish_reason": out.finish_reason if out.finished else None,
+                        }],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                # final usage chunk (reference logs TPS/TTFT from this)
+                elapsed = time.monotonic() - t_start
+                usage = {
+                    "prompt_tokens": len(prompt_ids),
+                    "completion_tokens": len(token_ids),
+                    "total_tokens": len(prompt_ids) + len(token_ids),
+                    "ttft_ms": round(((first_token_t or time.monotonic()) - t_start) * 1e3, 2),
+                    "tps": round(len(token_ids) / max(elapsed, 1e-6), 2),
+                }
+                yield f"data: {json.dumps({'id': rid, 'object': 'chat.completion.chunk', 'created': created, 'model': model_name, 'choices': [], 'usage': usage})}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        token_ids, finish_reason = await _collect(stream, sp)
+        return JSONResponse({
+            "id": rid, "object": "chat.completion", "created": created,
+            "model": model_name,
+            "choices": [{
+                "index": 0,
+                "message": {"role": "assistant", "content": tokenizer.decode(token_ids)},
+                "finish_reason": finish_reason,
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": len(token_ids),
+                "total_tokens": len(prompt_ids) + len(token_ids),
+            },
+        })
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        prompt = body.get("prompt")
+        if prompt is None:
+            raise HTTPException(400, "prompt required")
+        if isinstance(prompt, list) and prompt and isinstance(prompt[0], int):
+            prompt_ids = prompt
+        else:
+            prompt_ids = tokenizer.encode(prompt)
+        sp = _params(body)
+        rid = f"cmpl-{uuid.uuid4().hex[:24]}"
+        stream = server.submit(prompt_ids, sp, rid=rid)
+        token_ids, finish_reason = await _collect(stream, sp)
+        return JSONResponse({
+            "id": rid, "object": "text_completion", "created": int(time.time()),
+            "model": model_name,
+            "choices": [{
+                "index": 0, "text": tokenizer.decode(token_ids),
+                "finish_reason": finish_reason,
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": len(token_ids),
+                "total_tokens": len(prompt_ids) + len(token_ids),
+            },
+        })
+
+    return app

@@ -1,0 +1,100 @@
+"""OpenAI API frontend over a tiny CPU engine (TestClient, no network)."""
+
+import json
+
+import pytest
+import torch
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient
+
+from parallax_amd.models.config import ModelConfig
+from parallax_amd.server.engine import Engine, EngineArgs
+from parallax_amd.server.engine_server import EngineServer
+from parallax_amd.server.http_frontend import create_app
+from parallax_amd.server.tokenizer_util import TokenizerWrapper
+
+
+@pytest.fixture(scope="module")
+def client():
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=64,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+        intermediate_size=128, max_position_embeddings=512,
+        eos_token_ids=[2],
+    )
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=256,
+                                 dtype=torch.float32), random_weights=True)
+    server = EngineServer(eng)
+    server.start()
+    tok = TokenizerWrapper(vocab_size=cfg.vocab_size)
+    app = create_app(server, tok, model_name="tiny-test-model")
+    with TestClient(app) as c:
+        yield c
+    server.stop()
+
+
+def test_health_and_models(client):
+    assert client.get("/health").json()["status"] == "ok"
+    models = client.get("/v1/models").json()
+    assert models["data"][0]["id"] == "tiny-test-model"
+
+
+def test_chat_completion(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-test-model",
+        "messages": [{"role": "user", "content": "hello"}],
+        "max_tokens": 5, "temperature": 0.0,
+    })
+    assert r.status_code == 200
+    body = r.json()
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    assert body["usage"]["completion_tokens"] >= 1
+    assert body["choices"][0]["finish_reason"] in ("stop", "length")
+
+
+def test_chat_completion_streaming(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+        "model": "tiny-test-model",
+        "messages": [{"role": "user", "content": "stream test"}],
+        "max_tokens": 4, "temperature": 0.0, "stream": True,
+    }) as r:
+        assert r.status_code == 200
+        chunks = []
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                chunks.append(line[6:])
+    assert chunks[-1] == "[DONE]"
+    usage = json.loads(chunks[-2])["usage"]
+    assert usage["completion_tokens"] == 4
+    assert "ttft_ms" in usage and "tps" in usage
+    deltas = [json.loads(c) for c in chunks[:-2]]
+    assert all(d["object"] == "chat.completion.chunk" for d in deltas)
+
+
+def test_completions(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-test-model", "prompt": "once upon a time",
+        "max_tokens": 3, "temperature": 0.0,
+    })
+    assert r.status_code == 200
+    assert r.json()["usage"]["completion_tokens"] == 3
+
+
+def test_concurrent_requests(client):
+    import concurrent.futures as cf
+
+    def one(i):
+        return client.post("/v1/completions", json={
+            "prompt": f"request number {i}", "max_tokens": 4,
+            "temperature": 0.0,
+        }).json()["usage"]["completion_tokens"]
+
+    with cf.ThreadPoolExecutor(8) as ex:
+        results = list(ex.map(one, range(8)))
+    assert results == [4] * 8
+
+
+def test_stats(client):
+    s = client.get("/stats").json()
+    assert s["total_requests"] >= 1 and s["total_output_tokens"] >= 1
