@@ -6,3 +6,4 @@ from .registry import MODEL_REGISTRY, get_model_class, register_model
 from . import llama  # noqa: F401  (registers LlamaForCausalLM / Qwen2ForCausalLM)
 from . import qwen3  # noqa: F401
 from . import qwen3_moe  # noqa: F401
+from . import deepseek_v3  # noqa: F401

@@ -139,7 +139,9 @@ class ModelConfig:
                 cfg.get("moe_layer_freq"), list) else 1,
             norm_topk_prob=cfg.get("norm_topk_prob", True),
             routed_scaling_factor=cfg.get("routed_scaling_factor", 1.0),
-            scoring_func=cfg.get("scoring_func", "softmax"),
+            scoring_func=cfg.get("scoring_func")
+            or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "kimi_k2")
+                else "softmax"),
             topk_group=cfg.get("topk_group", 0),
             n_group=cfg.get("n_group", 0),
             # MLA

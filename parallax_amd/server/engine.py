@@ -30,6 +30,7 @@ from ..utils.logging_config import get_logger
 from .cache_manager import CacheManager
 from .kv_cache import (
     KVCacheSpec,
+    MLAKVCache,
     PagedKVCache,
     build_block_table_tensor,
     slot_mapping_for_positions,
@@ -107,22 +108,45 @@ class Engine:
             from .shard_loader import load_shard_weights
 
             load_shard_weights(self.model, model_path)
+        if hasattr(self.model, "finalize_weights"):
+            self.model.finalize_weights()
         self.model = self.model.to(device=self.device, dtype=args.dtype)
         # rope cache stays fp32
         self.model.rope_cache = self.model.rope_cache.float()
         self.model.eval()
 
-        spec = KVCacheSpec(
-            num_layers=end - start,
-            num_kv_heads=max(1, cfg.num_kv_heads // self.comm.tp_size),
-            head_dim=cfg.head_dim,
-            block_size=args.block_size,
-            dtype=args.dtype,
-        )
-        num_blocks = args.num_kv_blocks or CacheManager.num_blocks_from_memory(
-            spec, self.device, args.cache_memory_fraction
-        )
-        self.kv_cache = PagedKVCache(spec, num_blocks, self.device)
+        self.is_mla = cfg.is_mla
+        if self.is_mla:
+            mla_block_bytes = MLAKVCache.bytes_per_block(
+                end - start, cfg.kv_lora_rank, cfg.qk_rope_head_dim,
+                args.block_size, args.dtype,
+            )
+            if args.num_kv_blocks:
+                num_blocks = args.num_kv_blocks
+            else:
+                if self.device.type == "cuda":
+                    free_b, _ = torch.cuda.mem_get_info(self.device)
+                    budget = int(free_b * args.cache_memory_fraction) - (2 << 30)
+                else:
+                    budget = 1 << 30
+                num_blocks = max(16, budget // mla_block_bytes)
+            self.kv_cache = MLAKVCache(
+                end - start, cfg.kv_lora_rank, cfg.qk_rope_head_dim,
+                args.block_size, num_blocks, self.device,
+                args.dtype if args.dtype != torch.float32 else torch.float32,
+            )
+        else:
+            spec = KVCacheSpec(
+                num_layers=end - start,
+                num_kv_heads=max(1, cfg.num_kv_heads // self.comm.tp_size),
+                head_dim=cfg.head_dim,
+                block_size=args.block_size,
+                dtype=args.dtype,
+            )
+            num_blocks = args.num_kv_blocks or CacheManager.num_blocks_from_memory(
+                spec, self.device, args.cache_memory_fraction
+            )
+            self.kv_cache = PagedKVCache(spec, num_blocks, self.device)
         self.cache_manager = CacheManager(
             args.block_size, num_blocks, enable_prefix_cache=args.enable_prefix_cache
         )
@@ -145,14 +169,15 @@ class Engine:
                 block_size=args.block_size, hidden_size=cfg.hidden_size,
                 is_first_stage=self.comm.is_first_stage,
                 is_last_stage=self.comm.is_last_stage,
+                is_mla=self.is_mla,
             )
         self._pending_adds: List[InitialRequest] = []
         self._pending_aborts: List[str] = []
         self.step_count = 0
         logger.info(
-            "engine up: layers [%d,%d) of %d, %d KV blocks x %d tokens (%.2f GiB), device %s",
+            "engine up: layers [%d,%d) of %d, %d KV blocks x %d tokens, mla=%s, device %s",
             start, end, cfg.num_layers, num_blocks, args.block_size,
-            num_blocks * spec.bytes_per_block() / (1 << 30), self.device,
+            self.is_mla, self.device,
         )
 
     # -- public API (head rank) ----------------------------------------------------
@@ -288,7 +313,8 @@ class Engine:
             block_tables=build_block_table_tensor(block_tables, dev),
             seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
             query_lens=torch.tensor(query_lens, dtype=torch.int32, device=dev),
-            kv_cache=self.kv_cache,
+            kv_cache=None if self.is_mla else self.kv_cache,
+            mla_cache=self.kv_cache if self.is_mla else None,
             logits_indices=torch.tensor(logits_idx, dtype=torch.int64, device=dev),
             max_seq_len=max(seq_lens),
         )
@@ -314,7 +340,8 @@ class Engine:
             slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
             block_tables=build_block_table_tensor(block_tables, dev),
             seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
-            kv_cache=self.kv_cache,
+            kv_cache=None if self.is_mla else self.kv_cache,
+            mla_cache=self.kv_cache if self.is_mla else None,
             logits_indices=torch.arange(len(reqs), dtype=torch.int64, device=dev),
             max_seq_len=max(seq_lens),
         )

@@ -44,7 +44,9 @@ class DecodeGraphRunner:
         is_first_stage: bool,
         is_last_stage: bool,
         buckets: Optional[List[int]] = None,
+        is_mla: bool = False,
     ):
+        self.is_mla = is_mla
         self.model = model
         self.kv_cache = kv_cache
         self.device = device
@@ -89,7 +91,8 @@ class DecodeGraphRunner:
             slot_mapping=self.slot_mapping[:bucket],
             block_tables=self.block_tables[:bucket],
             seq_lens=self.seq_lens[:bucket],
-            kv_cache=self.kv_cache,
+            kv_cache=None if self.is_mla else self.kv_cache,
+            mla_cache=self.kv_cache if self.is_mla else None,
             logits_indices=None,
             max_seq_len=self.max_model_len,
         )
