@@ -33,6 +33,7 @@ class ModelConfig:
     o_proj_bias: bool = False
     mlp_bias: bool = False
     qk_norm: bool = False            # qwen3-style per-head q/k RMSNorm
+    partial_rotary_factor: float = 1.0
     sliding_window: int = -1         # -1 = full attention
     # which layers use the sliding window ("full" layers interleave, gpt-oss)
     layer_types: Optional[List[str]] = None
@@ -68,6 +69,10 @@ class ModelConfig:
     raw: Dict[str, Any] = field(default_factory=dict, repr=False)
 
     @property
+    def rot_dim(self) -> int:
+        return int(self.head_dim * self.partial_rotary_factor)
+
+    @property
     def is_moe(self) -> bool:
         return self.num_experts > 0
 
@@ -98,6 +103,7 @@ class ModelConfig:
         num_heads = cfg.get("num_attention_heads", 32)
         hidden = cfg.get("hidden_size", 4096)
         head_dim = cfg.get("head_dim") or hidden // num_heads
+        rope_params = cfg.get("rope_parameters") or {}
         eos = cfg.get("eos_token_id", 2)
         eos_ids = eos if isinstance(eos, list) else [eos]
         sliding = cfg.get("sliding_window") or -1
@@ -115,12 +121,18 @@ class ModelConfig:
             head_dim=head_dim,
             intermediate_size=cfg.get("intermediate_size", 4 * hidden),
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-5),
-            rope_theta=cfg.get("rope_theta", 10000.0),
-            rope_scaling=cfg.get("rope_scaling"),
+            rope_theta=cfg.get("rope_theta") or rope_params.get("rope_theta", 10000.0),
+            rope_scaling=cfg.get("rope_scaling") or (rope_params or None),
+            partial_rotary_factor=cfg.get("partial_rotary_factor")
+            or rope_params.get("partial_rotary_factor", 1.0),
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", cfg.get("qkv_bias", False)),
-            qk_norm=cfg.get("model_type", "") in ("qwen3", "qwen3_moe"),
+            o_proj_bias=cfg.get("attention_bias", False)
+            and cfg.get("model_type") == "gpt_oss",
+            attention_sinks=cfg.get("model_type") == "gpt_oss",
+            qk_norm=cfg.get("use_qk_norm", False)
+            or cfg.get("model_type", "") in ("qwen3", "qwen3_moe"),
             sliding_window=sliding,
             layer_types=layer_types,
             eos_token_ids=[e for e in eos_ids if e is not None],
@@ -140,7 +152,8 @@ class ModelConfig:
             norm_topk_prob=cfg.get("norm_topk_prob", True),
             routed_scaling_factor=cfg.get("routed_scaling_factor", 1.0),
             scoring_func=cfg.get("scoring_func")
-            or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "kimi_k2")
+            or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "kimi_k2",
+                                                       "glm4_moe", "glm4v_moe")
                 else "softmax"),
             topk_group=cfg.get("topk_group", 0),
             n_group=cfg.get("n_group", 0),

@@ -171,9 +171,9 @@ class DeepseekV3ShardModel(LlamaShardModel):
     def __init__(self, cfg: ModelConfig, start_layer: int = 0, end_layer=None):
         super().__init__(cfg, start_layer, end_layer)
         # MLA ropes only the decoupled qk_rope dims
-        self.rope_cache = ops.build_rope_cache(
-            cfg.max_position_embeddings, cfg.qk_rope_head_dim, cfg.rope_theta
-        )
+        from .rope import build_rope_cache_for
+
+        self.rope_cache = build_rope_cache_for(cfg, rot_dim=cfg.qk_rope_head_dim)
 
     def finalize_weights(self) -> None:
         for layer in self.layers:

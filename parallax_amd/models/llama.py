@@ -27,6 +27,7 @@ from ..parallel.layers import (
 from .config import ModelConfig
 from .forward_meta import ForwardMeta
 from .registry import register_model
+from .rope import build_rope_cache_for
 
 
 class RMSNorm(nn.Module):
@@ -68,6 +69,11 @@ class LlamaAttention(nn.Module):
         if self.qk_norm:
             self.q_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
             self.k_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
+        self.sinks: Optional[nn.Parameter] = None
+        if cfg.attention_sinks:
+            self.sinks = nn.Parameter(
+                torch.empty(self.num_heads, dtype=torch.float32), requires_grad=False
+            )
 
     def forward(self, x: torch.Tensor, meta: ForwardMeta, rope_cache: torch.Tensor):
         T = x.shape[0]
@@ -88,15 +94,17 @@ class LlamaAttention(nn.Module):
             meta.slot_mapping,
         )
 
+        sinks = self.sinks.float() if self.sinks is not None else None
         if meta.is_prefill:
             attn = ops.prefill_attention(
                 q, k_cache, v_cache, meta.block_tables, meta.seq_lens,
-                meta.query_lens, self.scale, self.sliding_window,
+                meta.query_lens, self.scale, self.sliding_window, sinks=sinks,
             )
         else:
             attn = ops.paged_attention_decode(
                 q, k_cache, v_cache, meta.block_tables, meta.seq_lens,
-                self.scale, self.sliding_window, max_seq_len=meta.max_seq_len or None,
+                self.scale, self.sliding_window, sinks=sinks,
+                max_seq_len=meta.max_seq_len or None,
             )
         return self.o_proj(attn.reshape(T, self.num_heads * self.head_dim))
 
@@ -164,12 +172,7 @@ class LlamaShardModel(nn.Module):
                                                 gather_output=True)
         self.register_buffer(
             "rope_cache",
-            ops.build_rope_cache(
-                cfg.max_position_embeddings,
-                cfg.head_dim,
-                cfg.rope_theta,
-                scaling_factor=_rope_scaling_factor(cfg),
-            ),
+            build_rope_cache_for(cfg, rot_dim=cfg.rot_dim),
             persistent=False,
         )
 
@@ -281,8 +284,3 @@ class LlamaShardModel(nn.Module):
                 )
 
 
-def _rope_scaling_factor(cfg: ModelConfig) -> float:
-    rs = cfg.rope_scaling or {}
-    if rs.get("rope_type") == "linear" or rs.get("type") == "linear":
-        return float(rs.get("factor", 1.0))
-    return 1.0

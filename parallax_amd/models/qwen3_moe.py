@@ -51,6 +51,12 @@ class Qwen3MoEShardModel(LlamaShardModel):
             layer.mlp.experts.router.weight.data.copy_(t)
             return True
         if parts[3] == "experts":
+            if parts[4] == "gate_up_proj":  # fused [E, 2I, H] (transformers >= 5)
+                layer.mlp.experts.w_gate_up.data.copy_(t)
+                return True
+            if parts[4] == "down_proj":
+                layer.mlp.experts.w_down.data.copy_(t)
+                return True
             e = int(parts[4])
             proj = parts[5]
             if proj == "gate_proj":

@@ -77,14 +77,15 @@ class MergedColumnParallelLinear(nn.Module):
         self.bias = nn.Parameter(torch.empty(total), requires_grad=False) if bias else None
 
     def load_full_weight_part(
-        self, idx: int, w: torch.Tensor, b: Optional[torch.Tensor] = None
+        self, idx: int, w: Optional[torch.Tensor], b: Optional[torch.Tensor] = None
     ) -> None:
         """Load one sub-projection (e.g. k_proj into the fused QKV) from its
         full (unsharded) HF tensor."""
         comm = get_comm()
         off = sum(self.shard_sizes[:idx])
         start, per = _shard(self.out_sizes[idx], comm.tp_size, comm.tp_rank)
-        self.weight.data[off : off + per].copy_(w[start : start + per])
+        if w is not None:
+            self.weight.data[off : off + per].copy_(w[start : start + per])
         if self.bias is not None and b is not None:
             self.bias.data[off : off + per].copy_(b[start : start + per])
 
