@@ -101,6 +101,7 @@ class MLAAttention(nn.Module):
             out_latent = ops.mla_paged_attention_decode(
                 q_latent.contiguous(), q_pe.contiguous(), cache,
                 meta.block_tables, meta.seq_lens, self.scale,
+                max_seq_len=meta.max_seq_len or None,
             )
             attn = torch.einsum(
                 "bhr,hvr->bhv", out_latent.float(), self.w_uv.float()

@@ -252,13 +252,17 @@ def mla_paged_attention_decode(
     block_tables: torch.Tensor,
     seq_lens: torch.Tensor,
     scale: float,
+    max_seq_len: Optional[int] = None,
 ) -> torch.Tensor:
     if q_latent.is_cuda:
         ext = _require_ext("mla_paged_attention_decode")
         out = torch.empty_like(q_latent)
+        if max_seq_len is None:
+            max_seq_len = int(seq_lens.max().item())
         ext.mla_paged_attention_decode(
             out, q_latent.contiguous(), q_rope.contiguous(), cache,
             block_tables.to(torch.int32), seq_lens.to(torch.int32), scale,
+            max_seq_len,
         )
         return out
     return ref.mla_paged_attention_decode(

@@ -39,6 +39,11 @@ void launch_paged_decode_mfma(void*, const void*, const void*, const void*,
                               int, int, float*, float*, hipStream_t, bool*);
 void launch_paged_attention_reduce(void*, const float*, const float*, int, int,
                                    int, int, const float*, hipStream_t);
+void launch_mla_paged_attention_decode(void*, const void*, const void*,
+                                       const void*, const int*, const int*,
+                                       int, int, int, int, int, int, float,
+                                       int, int, float*, float*, hipStream_t,
+                                       bool*);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const int*, const int*, const int*, const int*,
                               const int*, int, int, int, int, int, int, int64_t,
@@ -286,7 +291,56 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(launched, "no prefill kernel instantiation for D=", D);
 }
 
+void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
+                                torch::Tensor q_pe, torch::Tensor cache,
+                                torch::Tensor block_tables,
+                                torch::Tensor seq_lens, double scale,
+                                int64_t max_seq_len) {
+  CHECK_GPU(q_latent);
+  CHECK_CONTIG(q_latent);
+  CHECK_CONTIG(q_pe);
+  CHECK_BF16(q_latent);
+  CHECK_BF16(cache);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt);
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
+  const int B = q_latent.size(0);
+  const int H = q_latent.size(1);
+  const int R = q_latent.size(2);
+  const int DR = q_pe.size(2);
+  const int BS = cache.size(1);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(cache.size(2) == R + DR, "cache entry dim mismatch");
+
+  int part_tokens = 256;
+  int num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
+  if (num_parts > 128) {
+    part_tokens = (int)((max_seq_len + 127) / 128);
+    part_tokens = (part_tokens + 127) / 128 * 128;
+    num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
+  }
+  torch::Tensor tmp_acc, tmp_ml;
+  float *acc_ptr = nullptr, *ml_ptr = nullptr;
+  if (num_parts > 1) {
+    auto opts = q_latent.options().dtype(at::kFloat);
+    tmp_acc = torch::empty({B, H, num_parts, R}, opts);
+    tmp_ml = torch::empty({B, H, num_parts, 2}, opts);
+    acc_ptr = tmp_acc.data_ptr<float>();
+    ml_ptr = tmp_ml.data_ptr<float>();
+  }
+  bool launched = false;
+  launch_mla_paged_attention_decode(
+      out.data_ptr(), q_latent.data_ptr(), q_pe.data_ptr(), cache.data_ptr(),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, H, R, DR, BS,
+      max_blocks, (float)scale, num_parts, part_tokens, acc_ptr, ml_ptr,
+      cur_stream(), &launched);
+  TORCH_CHECK(launched, "no MLA kernel for R=", R, " DR=", DR, " BS=", BS);
+  if (num_parts > 1)
+    launch_paged_attention_reduce(out.data_ptr(), acc_ptr, ml_ptr, B, H, R,
+                                  num_parts, nullptr, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mla_paged_attention_decode", &mla_paged_attention_decode);
   m.def("prefill_attention", &prefill_attention);
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);

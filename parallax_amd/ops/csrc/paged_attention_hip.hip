@@ -363,6 +363,9 @@ extern "C" void launch_paged_attention_reduce(
   else if (D == 64)
    hipLaunchKernelGGL(( paged_attention_reduce_kernel<64>), dim3(rgrid), dim3(64), 0, stream, 
         (uint16_t*)out, tmp_acc, tmp_ml, num_parts, sinks, Hq);
+  else if (D == 512)  // MLA latent dim
+   hipLaunchKernelGGL(( paged_attention_reduce_kernel<512>), dim3(rgrid), dim3(64), 0, stream, 
+        (uint16_t*)out, tmp_acc, tmp_ml, num_parts, sinks, Hq);
 }
 
 extern "C" void launch_paged_attention_decode(

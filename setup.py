@@ -21,6 +21,7 @@ sources = [
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "paged_attention.hip"),
     os.path.join(CSRC, "paged_attention_mfma.hip"),
+    os.path.join(CSRC, "mla_attention.hip"),
     os.path.join(CSRC, "prefill_attention.hip"),
     os.path.join(CSRC, "moe.hip"),
 ]
