@@ -89,6 +89,10 @@ class FusedMoE(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
         topk_ids, topk_w = self.router(x)           # [T,k]
+        if x.is_cuda:
+            return ops.fused_moe_forward(
+                x, self.w_gate_up, self.w_down, topk_ids, topk_w
+            ).to(x.dtype)
         out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
         flat_ids = topk_ids.reshape(-1)              # [T*k]
         flat_w = topk_w.reshape(-1)

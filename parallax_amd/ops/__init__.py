@@ -293,6 +293,37 @@ def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return ref.gelu_and_mul(x)
 
 
+# -- MoE ----------------------------------------------------------------------
+
+
+def fused_moe_forward(
+    x: torch.Tensor,          # [T, H] bf16
+    w_gate_up: torch.Tensor,  # [E, 2I, H] bf16
+    w_down: torch.Tensor,     # [E, H, I] bf16
+    topk_ids: torch.Tensor,   # [T, k] long
+    topk_weights: torch.Tensor,  # [T, k] float
+    activation: str = "silu",
+    limit: float = 0.0,
+) -> torch.Tensor:
+    """Grouped-GEMM MoE (device-side routing; graph-capture safe). Returns
+    fp32 [T, H]."""
+    assert x.is_cuda
+    ext = _require_ext("moe_forward")
+    T, H = x.shape
+    E = w_gate_up.shape[0]
+    k = topk_ids.shape[1]
+    flat = topk_ids.reshape(-1)
+    sorted_ids, perm = torch.sort(flat)
+    seg = torch.searchsorted(
+        sorted_ids, torch.arange(E + 1, device=x.device, dtype=sorted_ids.dtype)
+    ).to(torch.int32)
+    route_w = topk_weights.reshape(-1).float()[perm].contiguous()
+    out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+    ext.moe_forward(out, x.contiguous(), w_gate_up, w_down, perm.contiguous(),
+                    seg.contiguous(), route_w, k, activation == "gelu", limit)
+    return out
+
+
 # -- sampling (torch ops; GPU path uses torch's ROCm kernels — not a hot spot
 #    relative to the model forward, custom kernel is a later optimization) -------
 

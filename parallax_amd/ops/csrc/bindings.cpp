@@ -44,6 +44,13 @@ void launch_mla_paged_attention_decode(void*, const void*, const void*,
                                        int, int, int, int, int, int, float,
                                        int, int, float*, float*, hipStream_t,
                                        bool*);
+void launch_build_moe_tiles(int*, int*, const int*, int, int, hipStream_t);
+void launch_moe_gate_up(void*, const void*, const void*, const int*,
+                        const int*, const int64_t*, const int*, int, int, int,
+                        int, int, bool, float, hipStream_t);
+void launch_moe_down(void*, const void*, const void*, const float*, const int*,
+                     const int*, const int64_t*, const int*, int, int, int,
+                     int, int, hipStream_t);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const int*, const int*, const int*, const int*,
                               const int*, int, int, int, int, int, int, int64_t,
@@ -339,7 +346,47 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
                                   num_parts, nullptr, cur_stream());
 }
 
+void moe_forward(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
+                 torch::Tensor w_down, torch::Tensor perm,
+                 torch::Tensor seg_offsets, torch::Tensor route_w,
+                 int64_t topk, bool gelu, double limit) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_BF16(x);
+  CHECK_BF16(w_gu);
+  CHECK_BF16(w_down);
+  TORCH_CHECK(out.scalar_type() == at::kFloat && out.is_contiguous());
+  TORCH_CHECK(perm.scalar_type() == at::kLong);
+  TORCH_CHECK(seg_offsets.scalar_type() == at::kInt);
+  TORCH_CHECK(route_w.scalar_type() == at::kFloat);
+  const int E = w_gu.size(0);
+  const int I = w_gu.size(1) / 2;
+  const int H = w_gu.size(2);
+  const int n_assign = perm.size(0);
+  TORCH_CHECK(H % 32 == 0 && I % 64 == 0,
+              "MoE dims must be multiples of 32/64 (H=", H, " I=", I, ")");
+  const int max_tiles = (n_assign + 15) / 16 + E;
+  auto iopts = x.options().dtype(at::kInt);
+  torch::Tensor tile_expert = torch::empty({max_tiles}, iopts);
+  torch::Tensor tile_row0 = torch::empty({max_tiles}, iopts);
+  torch::Tensor h_buf =
+      torch::empty({n_assign, I}, x.options().dtype(at::kBFloat16));
+  auto stream = cur_stream();
+  launch_build_moe_tiles(tile_expert.data_ptr<int>(), tile_row0.data_ptr<int>(),
+                         seg_offsets.data_ptr<int>(), E, max_tiles, stream);
+  launch_moe_gate_up(h_buf.data_ptr(), x.data_ptr(), w_gu.data_ptr(),
+                     tile_expert.data_ptr<int>(), tile_row0.data_ptr<int>(),
+                     perm.data_ptr<int64_t>(), seg_offsets.data_ptr<int>(), E,
+                     (int)topk, H, I, max_tiles, gelu, (float)limit, stream);
+  launch_moe_down(out.data_ptr(), h_buf.data_ptr(), w_down.data_ptr(),
+                  route_w.data_ptr<float>(), tile_expert.data_ptr<int>(),
+                  tile_row0.data_ptr<int>(), perm.data_ptr<int64_t>(),
+                  seg_offsets.data_ptr<int>(), E, (int)topk, H, I, max_tiles,
+                  stream);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("moe_forward", &moe_forward);
   m.def("mla_paged_attention_decode", &mla_paged_attention_decode);
   m.def("prefill_attention", &prefill_attention);
   m.def("rmsnorm", &rmsnorm);
