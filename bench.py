@@ -63,9 +63,65 @@ def qwen2_05b() -> dict:
     }
 
 
+def qwen2_72b() -> dict:
+    return {
+        "architectures": ["Qwen2ForCausalLM"],
+        "model_type": "qwen2",
+        "vocab_size": 152064,
+        "hidden_size": 8192,
+        "num_hidden_layers": 80,
+        "num_attention_heads": 64,
+        "num_key_value_heads": 8,
+        "head_dim": 128,
+        "intermediate_size": 29568,
+        "rms_norm_eps": 1e-6,
+        "rope_theta": 1000000.0,
+        "max_position_embeddings": 32768,
+        "qkv_bias": True,
+        "tie_word_embeddings": False,
+        "torch_dtype": "bfloat16",
+    }
+
+
+def deepseek_v3() -> dict:
+    # DeepSeek-V3/R1 671B (also the Kimi-K2 architecture class). bf16 weights
+    # need ~168 GB/GPU at PP=8 — run with --gpus 8.
+    return {
+        "architectures": ["DeepseekV3ForCausalLM"],
+        "model_type": "deepseek_v3",
+        "vocab_size": 129280,
+        "hidden_size": 7168,
+        "num_hidden_layers": 61,
+        "num_attention_heads": 128,
+        "num_key_value_heads": 128,
+        "intermediate_size": 18432,
+        "moe_intermediate_size": 2048,
+        "n_routed_experts": 256,
+        "num_experts_per_tok": 8,
+        "n_shared_experts": 1,
+        "n_group": 8,
+        "topk_group": 4,
+        "routed_scaling_factor": 2.5,
+        "norm_topk_prob": True,
+        "first_k_dense_replace": 3,
+        "q_lora_rank": 1536,
+        "kv_lora_rank": 512,
+        "qk_nope_head_dim": 128,
+        "qk_rope_head_dim": 64,
+        "v_head_dim": 128,
+        "rms_norm_eps": 1e-6,
+        "rope_theta": 10000.0,
+        "max_position_embeddings": 163840,
+        "tie_word_embeddings": False,
+        "torch_dtype": "bfloat16",
+    }
+
+
 MODELS = {
     "deepseek-r1-distill-llama-8b": (deepseek_r1_distill_llama_8b, "DeepSeek-R1-Distill-Llama-8B"),
     "qwen2-0.5b": (qwen2_05b, "Qwen2-0.5B"),
+    "qwen2-72b": (qwen2_72b, "Qwen2-72B"),
+    "deepseek-v3": (deepseek_v3, "DeepSeek-V3-671B"),
 }
 
 

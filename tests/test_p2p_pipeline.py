@@ -1,0 +1,100 @@
+"""Decentralized pipeline over the loopback transport: 2/3-stage peer
+executors produce the same greedy tokens as the single-process engine
+(the reference's in-process pipeline emulation test strategy,
+tests/test_executor.py analogue). Also: codec round-trip."""
+
+import pytest
+import torch
+
+from parallax_amd.p2p import codec
+from parallax_amd.p2p.peer_executor import PeerExecutor
+from parallax_amd.p2p.transport import LoopbackTransport
+from parallax_amd.server.request import IntermediateRequest
+from parallax_amd.server.sampling_params import SamplingParams
+
+from tests.test_pipeline_parallel import PROMPTS, full_state_dict, run_single_process, tiny_cfg
+
+
+def test_codec_roundtrip():
+    h = torch.randn(5, 64, dtype=torch.bfloat16)
+    req = IntermediateRequest(
+        rid="r1", routing_table=["a", "b"], current_position=7,
+        num_new_tokens=5, is_prefill=True, hidden_states=h,
+        sampling_params=SamplingParams(temperature=0.5, top_k=5),
+        input_ids=[1, 2, 3, 4, 5],
+    )
+    msg = codec.decode(codec.encode_forward([req]))
+    assert msg["kind"] == "forward"
+    r = msg["reqs"][0]
+    assert r.rid == "r1" and r.routing_table == ["a", "b"]
+    assert r.current_position == 7 and r.is_prefill
+    assert torch.equal(r.hidden_states, h)
+    assert r.sampling_params.temperature == 0.5 and r.sampling_params.top_k == 5
+
+    ctl = codec.decode(codec.encode_control("release", ["r1", "r2"]))
+    assert ctl == {"kind": "release", "rids": ["r1", "r2"]}
+    tok = codec.decode(codec.encode_tokens([("r1", 42)]))
+    assert tok["kind"] == "token" and tok["tokens"][0] == ["r1", 42] or \
+        tok["tokens"][0] == ("r1", 42)
+
+
+@pytest.mark.parametrize("n_stages", [2, 3])
+def test_loopback_pipeline_matches_single(n_stages):
+    expected = run_single_process()
+    cfg = tiny_cfg()
+    registry = {}
+    peer_ids = [f"peer{i}" for i in range(n_stages)]
+    # contiguous layer split
+    base, rem = divmod(cfg.num_layers, n_stages)
+    spans, pos = [], 0
+    for i in range(n_stages):
+        n = base + (1 if i < rem else 0)
+        spans.append((pos, pos + n))
+        pos += n
+    peers = []
+    sd = full_state_dict(cfg)
+    for pid, (s, e) in zip(peer_ids, spans):
+        t = LoopbackTransport(pid, registry)
+        px = PeerExecutor(cfg, s, e, pid, t, dtype=torch.float32,
+                          num_kv_blocks=128, block_size=8)
+        for name, w in sd.items():
+            px.model.load_hf_weight(name, w)
+        peers.append(px)
+
+    head = peers[0]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6, ignore_eos=True)
+    rids = [head.submit(p, sp, peer_ids) for p in PROMPTS]
+    outputs = {rid: [] for rid in rids}
+    done = set()
+    for _ in range(2000):
+        for px in peers:
+            px.step(recv_timeout=0.001)
+        for out in head.drain_outputs():
+            outputs[out.rid].append(out.token_id)
+            if out.finished:
+                done.add(out.rid)
+        if len(done) == len(rids):
+            break
+    assert [outputs[r] for r in rids] == expected
+    # release control packets freed downstream cache state
+    for _ in range(20):
+        for px in peers:
+            px.step(recv_timeout=0.001)
+    for px in peers[1:]:
+        assert not px._peer_positions
+
+
+def test_tcp_transport_roundtrip():
+    from parallax_amd.p2p.transport import TcpTransport
+
+    a = TcpTransport("a", "127.0.0.1", 0)
+    b = TcpTransport("b", "127.0.0.1", 0)
+    a.set_peer_addr("b", "127.0.0.1", b.port)
+    b.set_peer_addr("a", "127.0.0.1", a.port)
+    a.send("b", b"hello" * 1000)
+    got = b.recv(timeout=5)
+    assert got == b"hello" * 1000
+    b.send("a", b"pong")
+    assert a.recv(timeout=5) == b"pong"
+    a.close()
+    b.close()
