@@ -1,0 +1,79 @@
+"""OpenAI frontend served by the HEAD peer of a decentralized pipeline.
+
+The gateway (backend/service.py) routes a request to this app with the chosen
+routing_table in the body; tokens stream back from the last stage through the
+node agent's output demux. Reference analogue: the vllm-rs frontend launched on
+the head peer (start_layer == 0, launch.py)."""
+
+from __future__ import annotations
+
+import asyncio
+import time
+import uuid
+from typing import List
+
+from fastapi import FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse
+
+from ..server.sampling_params import SamplingParams
+from ..server.tokenizer_util import TokenizerWrapper
+from ..utils.logging_config import get_logger
+from .node_agent import NodeAgent
+
+logger = get_logger("p2p.head_frontend")
+
+
+def create_head_app(
+    agent: NodeAgent, tokenizer: TokenizerWrapper, model_name: str = "model"
+) -> FastAPI:
+    app = FastAPI(title="parallax_amd-head", version="0.1.0")
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok", "node_id": agent.node_id}
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        body = await request.json()
+        messages = body.get("messages")
+        if not messages:
+            raise HTTPException(400, "messages required")
+        routing_table: List[str] = body.get("routing_table") or [agent.node_id]
+        prompt_ids = tokenizer.chat_prompt_ids(messages)
+        sp = SamplingParams.from_openai(body)
+        rid, q = agent.submit(prompt_ids, sp, routing_table)
+        loop = asyncio.get_event_loop()
+        token_ids, finish_reason = [], "stop"
+        t0 = time.monotonic()
+        first_t = None
+        while True:
+            out = await loop.run_in_executor(None, q.get)
+            if out is None:
+                break
+            if first_t is None:
+                first_t = time.monotonic()
+            token_ids.append(out.token_id)
+            if out.finished:
+                finish_reason = out.finish_reason or "stop"
+        elapsed = time.monotonic() - t0
+        return JSONResponse({
+            "id": f"chatcmpl-{uuid.uuid4().hex[:16]}",
+            "object": "chat.completion",
+            "created": int(time.time()),
+            "model": model_name,
+            "choices": [{
+                "index": 0,
+                "message": {"role": "assistant",
+                            "content": tokenizer.decode(token_ids)},
+                "finish_reason": finish_reason,
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": len(token_ids),
+                "total_tokens": len(prompt_ids) + len(token_ids),
+                "ttft_ms": round(((first_t or time.monotonic()) - t0) * 1e3, 2),
+                "tps": round(len(token_ids) / max(elapsed, 1e-6), 2),
+            },
+        })
+
+    return app
