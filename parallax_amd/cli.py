@@ -1,0 +1,182 @@
+"""parallax_amd CLI: run / join / serve / chat (reference analogue: cli.py).
+
+  run    — start the scheduler service (control plane + gateway)
+  join   — start a worker node agent that joins a scheduler
+  serve  — single-host serving: engine + OpenAI frontend in one process
+           (PP over local GPUs via torch.distributed.run when --gpus > 1)
+  chat   — minimal terminal chat client against any OpenAI endpoint
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+
+from .utils.logging_config import get_logger
+
+logger = get_logger("cli")
+
+
+def cmd_run(args) -> None:
+    import uvicorn
+
+    from .backend.service import SchedulerService, create_backend_app
+
+    svc = SchedulerService()
+    if args.model_path:
+        from .models.config import ModelConfig
+
+        with open(os.path.join(args.model_path, "config.json")) as f:
+            hf_cfg = json.load(f)
+        svc.init_model(args.model_name or args.model_path, hf_cfg,
+                       min_nodes=args.min_nodes)
+    app = create_backend_app(svc)
+    uvicorn.run(app, host=args.host, port=args.port)
+
+
+def cmd_join(args) -> None:
+    import threading
+
+    import torch
+    import uvicorn
+
+    from .p2p.head_frontend import create_head_app
+    from .p2p.node_agent import NodeAgent
+    from .server.tokenizer_util import TokenizerWrapper
+
+    device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    agent = NodeAgent(
+        args.scheduler_url, node_id=args.node_id, host=args.host,
+        http_port=args.port, model_path=args.model_path,
+        random_weights=args.model_path is None, device=device, dtype=dtype,
+        num_kv_blocks=args.num_kv_blocks, block_size=args.block_size,
+    )
+    agent.join(timeout_s=args.join_timeout)
+    agent.start()
+    logger.info("node %s serving layers [%d,%d)", agent.node_id,
+                agent.assignment["start_layer"], agent.assignment["end_layer"])
+    if agent.assignment["start_layer"] == 0:
+        tok = TokenizerWrapper(args.model_path,
+                               vocab_size=agent.cfg.vocab_size)
+        app = create_head_app(agent, tok, args.model_name or "model")
+        uvicorn.run(app, host=args.host, port=args.port)
+    else:
+        threading.Event().wait()  # worker runs in background threads
+
+
+def cmd_serve(args) -> None:
+    import torch
+    import uvicorn
+
+    from .models.config import ModelConfig
+    from .parallel.comm import init_distributed
+    from .server.engine import Engine, EngineArgs
+    from .server.engine_server import EngineServer
+    from .server.http_frontend import create_app
+    from .server.tokenizer_util import TokenizerWrapper
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    comm = init_distributed(pp_size=world, tp_size=1)
+    if args.model_path:
+        cfg = ModelConfig.from_pretrained(args.model_path)
+    else:
+        from bench import MODELS  # synthetic configs
+
+        cfg = ModelConfig.from_hf_config(MODELS[args.model][0]())
+    eargs = EngineArgs(
+        block_size=args.block_size,
+        max_batch_size=args.max_batch_size,
+        max_num_tokens_per_batch=args.max_num_tokens_per_batch,
+        micro_batches=world,
+        dtype=torch.bfloat16 if torch.cuda.is_available() else torch.float32,
+        num_kv_blocks=None if torch.cuda.is_available() else 4096,
+        max_model_len=args.max_model_len,
+    )
+    engine = Engine(cfg, eargs, comm=comm, model_path=args.model_path,
+                    random_weights=args.model_path is None)
+    if comm.rank == 0:
+        server = EngineServer(engine)
+        server.start()
+        tok = TokenizerWrapper(args.model_path, vocab_size=cfg.vocab_size)
+        app = create_app(server, tok, args.model_name or args.model or "model")
+        uvicorn.run(app, host=args.host, port=args.port)
+    else:
+        # non-head ranks just run the SPMD step loop
+        import time
+
+        while True:
+            if engine.has_work:
+                engine.step()
+            else:
+                engine.step() if engine.scheduler.has_work else time.sleep(0.002)
+
+
+def cmd_chat(args) -> None:
+    import httpx
+
+    url = args.url.rstrip("/") + "/v1/chat/completions"
+    history = []
+    print(f"chatting with {url} (ctrl-d to exit)")
+    while True:
+        try:
+            user = input("you> ")
+        except EOFError:
+            break
+        history.append({"role": "user", "content": user})
+        r = httpx.post(url, json={"model": "default", "messages": history,
+                                  "max_tokens": args.max_tokens}, timeout=600)
+        msg = r.json()["choices"][0]["message"]
+        history.append(msg)
+        print(f"assistant> {msg['content']}")
+
+
+def main(argv=None) -> None:
+    p = argparse.ArgumentParser(prog="parallax_amd")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    pr = sub.add_parser("run", help="start the scheduler service")
+    pr.add_argument("--host", default="0.0.0.0")
+    pr.add_argument("--port", type=int, default=3001)
+    pr.add_argument("--model-path", default=None)
+    pr.add_argument("--model-name", default=None)
+    pr.add_argument("--min-nodes", type=int, default=1)
+    pr.set_defaults(fn=cmd_run)
+
+    pj = sub.add_parser("join", help="join a scheduler as a worker node")
+    pj.add_argument("--scheduler-url", required=True)
+    pj.add_argument("--node-id", default=None)
+    pj.add_argument("--host", default="127.0.0.1")
+    pj.add_argument("--port", type=int, default=3010)
+    pj.add_argument("--model-path", default=None)
+    pj.add_argument("--model-name", default=None)
+    pj.add_argument("--num-kv-blocks", type=int, default=4096)
+    pj.add_argument("--block-size", type=int, default=32)
+    pj.add_argument("--join-timeout", type=float, default=600.0)
+    pj.set_defaults(fn=cmd_join)
+
+    ps = sub.add_parser("serve", help="single-host OpenAI server")
+    ps.add_argument("--host", default="0.0.0.0")
+    ps.add_argument("--port", type=int, default=3000)
+    ps.add_argument("--model-path", default=None)
+    ps.add_argument("--model", default="deepseek-r1-distill-llama-8b")
+    ps.add_argument("--model-name", default=None)
+    ps.add_argument("--block-size", type=int, default=32)
+    ps.add_argument("--max-batch-size", type=int, default=128)
+    ps.add_argument("--max-num-tokens-per-batch", type=int, default=16384)
+    ps.add_argument("--max-model-len", type=int, default=8192)
+    ps.set_defaults(fn=cmd_serve)
+
+    pc = sub.add_parser("chat", help="terminal chat client")
+    pc.add_argument("--url", default="http://127.0.0.1:3000")
+    pc.add_argument("--max-tokens", type=int, default=256)
+    pc.set_defaults(fn=cmd_chat)
+
+    args = p.parse_args(argv)
+    args.fn(args)
+
+
+if __name__ == "__main__":
+    main()

@@ -86,6 +86,8 @@ class SamplingParams:
                 or body.get("max_completion_tokens")
                 or default_max_tokens
             ),
+            min_new_tokens=int(body.get("min_tokens", 0)),
             stop=stop,
+            ignore_eos=bool(body.get("ignore_eos", False)),
             seed=body.get("seed"),
         )

@@ -1,0 +1,79 @@
+"""Router LB unit tests + serving-benchmark harness smoke against a live
+tiny engine server."""
+
+import threading
+import time
+
+import pytest
+import torch
+
+pytest.importorskip("uvicorn")
+import asyncio
+
+import httpx
+
+from parallax_amd.router.lb import ClusterEndpoint, LoadBalancer, create_router_app
+
+
+def test_lb_round_robin_and_health():
+    lb = LoadBalancer(["http://a", "http://b"], strategy="round_robin")
+    picks = {lb.pick().url for _ in range(4)}
+    assert picks == {"http://a", "http://b"}
+    lb.endpoints["http://a"].healthy = False
+    assert all(lb.pick().url == "http://b" for _ in range(3))
+    lb.endpoints["http://b"].healthy = False
+    assert lb.pick() is None
+
+
+def test_lb_performance_strategy():
+    lb = LoadBalancer(["http://fast", "http://slow"], strategy="performance")
+    lb.endpoints["http://fast"].update_metrics(ttft_ms=50, tps=100)
+    lb.endpoints["http://slow"].update_metrics(ttft_ms=2000, tps=5)
+    assert lb.pick().url == "http://fast"
+
+
+def test_lb_runtime_reconfig():
+    lb = LoadBalancer(["http://a"])
+    lb.add_endpoint("http://b")
+    lb.remove_endpoint("http://a")
+    assert list(lb.endpoints) == ["http://b"]
+    lb.set_strategy("performance")
+    assert lb.strategy == "performance"
+
+
+@pytest.mark.timeout(180)
+def test_benchmark_harness_end_to_end():
+    """benchmark_serving against a live tiny engine (streaming SSE)."""
+    import uvicorn
+
+    from parallax_amd.benchmark.benchmark_serving import run_benchmark
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.engine_server import EngineServer
+    from parallax_amd.server.http_frontend import create_app
+    from parallax_amd.server.tokenizer_util import TokenizerWrapper
+    from tests.test_cluster_integration import free_port, serve_in_thread
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=64,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+        intermediate_size=128, max_position_embeddings=1024, eos_token_ids=[2],
+    )
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=1024,
+                                 dtype=torch.float32), random_weights=True)
+    server = EngineServer(eng)
+    server.start()
+    app = create_app(server, TokenizerWrapper(vocab_size=512), "bench-tiny")
+    port = free_port()
+    srv, _ = serve_in_thread(app, port)
+
+    result = asyncio.run(run_benchmark(
+        f"http://127.0.0.1:{port}", num_prompts=8, request_rate=50.0,
+        input_len=16, output_len=6,
+    ))
+    assert result["completed"] == 8, result
+    assert result["output_token_throughput_tps"] > 0
+    assert result["ttft_ms"]["mean"] > 0
+    assert result["itl_ms"] is not None
+    srv.should_exit = True
+    server.stop()
