@@ -132,6 +132,7 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
                 layer_latency_ms=body.get("layer_latency_ms"),
                 current_requests=body.get("current_requests"),
                 rtt_ms=body.get("rtt_ms"),
+                last_refit_time=body.get("last_refit_time"),
             )
         return {
             "assignment": assignment.__dict__ if assignment else None,

@@ -112,6 +112,7 @@ class ClusterScheduler:
         layer_latency_ms: Optional[float] = None,
         current_requests: Optional[int] = None,
         rtt_ms: Optional[Dict[str, float]] = None,
+        last_refit_time: Optional[float] = None,
     ) -> Optional[LayerAssignment]:
         """Heartbeat: returns the node's current assignment so the worker can
         detect re-allocation (reference p2p/server.py:786-815 mismatch check)."""
@@ -125,6 +126,8 @@ class ClusterScheduler:
             node.current_requests = current_requests
         if rtt_ms:
             node.rtt_ms.update(rtt_ms)
+        if last_refit_time is not None:
+            node.last_refit_time = last_refit_time
         return self._assignment(node) if node.has_assignment else None
 
     def sweep_heartbeats(self) -> List[str]:

@@ -450,6 +450,42 @@ class Engine:
             token_ids = payload[0]
         return token_ids  # type: ignore[return-value]
 
+    # -- weight refit (runtime weight update, reference §3.5) ---------------------------------
+
+    def update_weights_from_disk(self, model_path: str) -> int:
+        """Reload this shard's weights from a (new) checkpoint directory and
+        invalidate the prefix cache (cached KV was computed with old weights).
+        Reference analogue: model_runner.update_weights_from_disk
+        (sglang/model_runner.py:406-422)."""
+        from .shard_loader import load_shard_weights
+
+        was_gpu = self.device.type == "cuda"
+        model_cpu = self.model.to("cpu") if was_gpu else self.model
+        n = load_shard_weights(model_cpu, model_path)
+        if hasattr(model_cpu, "finalize_weights"):
+            model_cpu.finalize_weights()
+        self.model = model_cpu.to(device=self.device, dtype=self.args.dtype)
+        self.model.rope_cache = self.model.rope_cache.float()
+        if self.graph_runner is not None:
+            self.graph_runner.model = self.model
+            # weights moved: captured graphs reference stale parameter storage
+            self.graph_runner._graphs.clear()
+            self.graph_runner._outputs.clear()
+        self.cache_manager.reset_prefix_cache()
+        logger.info("weight refit: reloaded %d tensors from %s", n, model_path)
+        return n
+
+    def update_weights_from_tensors(self, named_tensors) -> int:
+        """In-place update from (name, tensor) pairs (RL weight push)."""
+        n = 0
+        for name, t in named_tensors:
+            if self.model.load_hf_weight(name, t.cpu()):
+                n += 1
+        if hasattr(self.model, "finalize_weights"):
+            self.model.finalize_weights()
+        self.cache_manager.reset_prefix_cache()
+        return n
+
     # -- convenience: synchronous generation (tests, chat CLI) --------------------------------
 
     def generate(

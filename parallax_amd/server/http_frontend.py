@@ -66,6 +66,16 @@ def create_app(
     async def stats():
         return server.stats()
 
+    @app.post("/update_weights")
+    async def update_weights(request: Request):
+        body = await request.json()
+        path = body.get("model_path")
+        if not path:
+            raise HTTPException(400, "model_path required")
+        with server._lock:
+            n = server.engine.update_weights_from_disk(path)
+        return {"updated_tensors": n}
+
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
         body = await request.json()
