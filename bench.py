@@ -137,6 +137,7 @@ def main():
     ap.add_argument("--block-size", type=int, default=32)
     ap.add_argument("--micro-batches", type=int, default=0,
                     help="0 = one per pipeline stage")
+    ap.add_argument("--kv-dtype", default="auto", choices=["auto", "fp8"])
     ap.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
     args = ap.parse_args()
 
@@ -169,6 +170,7 @@ def main():
         dtype=torch.bfloat16 if use_gpu else torch.float32,
         num_kv_blocks=None if use_gpu else 4096,
         enable_prefix_cache=False,  # unique synthetic prompts; skip radix overhead
+        kv_cache_dtype=args.kv_dtype,
         seed=0,
     )
     engine = Engine(cfg, eargs, comm=comm, random_weights=True)
@@ -243,6 +245,7 @@ def main():
                 "global_batch": global_batch,
                 "seq_len": args.prompt_len,
                 "parallelism": f"pp{world}",
+                "kv_dtype": args.kv_dtype,
                 "micro_batches": eargs.micro_batches,
                 "prefill_s": round(prefill_s, 3),
             },

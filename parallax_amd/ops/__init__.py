@@ -115,14 +115,17 @@ def rope_and_cache(
     cos_sin: torch.Tensor,
     slot_mapping: torch.Tensor,
     is_neox: bool = True,
+    k_scale: float = 1.0,
+    v_scale: float = 1.0,
 ) -> None:
     """Fused: rope(q) in place; rope(k) + v scattered straight into the paged
-    cache. One kernel per layer instead of three."""
+    cache (quantizing when the cache is fp8_e4m3fn). One kernel per layer
+    instead of three."""
     if q.is_cuda:
         ext = _require_ext("rope_and_cache")
         ext.rope_and_cache(
             q, k, v, k_cache, v_cache, positions.to(torch.int32), cos_sin,
-            slot_mapping.to(torch.int64), is_neox,
+            slot_mapping.to(torch.int64), is_neox, k_scale, v_scale,
         )
         return
     ref.rope_inplace(q, k, positions, cos_sin, is_neox)
@@ -174,6 +177,8 @@ def paged_attention_decode(
     softcap: float = 0.0,
     sinks: Optional[torch.Tensor] = None,
     max_seq_len: Optional[int] = None,
+    k_scale: float = 1.0,
+    v_scale: float = 1.0,
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("paged_attention_decode")
@@ -192,6 +197,8 @@ def paged_attention_decode(
             softcap,
             sinks if sinks is not None else q.new_empty(0),
             max_seq_len,
+            k_scale,
+            v_scale,
         )
         return out
     return ref.paged_attention_decode(
@@ -210,6 +217,8 @@ def prefill_attention(
     sliding_window: int = -1,
     softcap: float = 0.0,
     sinks: Optional[torch.Tensor] = None,
+    k_scale: float = 1.0,
+    v_scale: float = 1.0,
 ) -> torch.Tensor:
     if q.is_cuda:
         ext = _require_ext("prefill_attention")
@@ -237,6 +246,8 @@ def prefill_attention(
             sliding_window,
             softcap,
             sinks if sinks is not None else q.new_empty(0),
+            k_scale,
+            v_scale,
         )
         return out
     return ref.prefill_attention(

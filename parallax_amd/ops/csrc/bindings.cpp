@@ -20,8 +20,8 @@ void launch_rope(void*, void*, const int*, const float*, int, int, int, int,
                  int, bool, int64_t, int64_t, hipStream_t);
 void launch_rope_and_cache(void*, const void*, const void*, void*, void*,
                            const int*, const float*, const int64_t*, int, int,
-                           int, int, int, int, bool, int64_t, int64_t, int64_t,
-                           hipStream_t);
+                           int, int, int, int, bool, bool, float, float,
+                           int64_t, int64_t, int64_t, hipStream_t);
 void launch_reshape_and_cache(const void*, const void*, void*, void*,
                               const int64_t*, int, int, int, int, hipStream_t);
 void launch_mla_reshape_and_cache(const void*, const void*, void*,
@@ -36,7 +36,8 @@ void launch_paged_attention_decode(void*, const void*, const void*, const void*,
 void launch_paged_decode_mfma(void*, const void*, const void*, const void*,
                               const int*, const int*, int, int, int, int, int,
                               int, int64_t, float, int, float, const float*,
-                              int, int, float*, float*, hipStream_t, bool*);
+                              int, int, float*, float*, bool, float, float,
+                              hipStream_t, bool*);
 void launch_paged_attention_reduce(void*, const float*, const float*, int, int,
                                    int, int, const float*, hipStream_t);
 void launch_mla_paged_attention_decode(void*, const void*, const void*,
@@ -54,8 +55,8 @@ void launch_moe_down(void*, const void*, const void*, const float*, const int*,
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const int*, const int*, const int*, const int*,
                               const int*, int, int, int, int, int, int, int64_t,
-                              float, int, float, const float*, hipStream_t,
-                              bool*);
+                              float, int, float, const float*, bool, float,
+                              float, hipStream_t, bool*);
 }
 
 static hipStream_t cur_stream() {
@@ -64,6 +65,10 @@ static hipStream_t cur_stream() {
 
 // [T, H, D] view whose rows may live inside a wider fused tensor (e.g. the QKV
 // GEMM output): inner two dims must be dense, the token stride may be larger.
+static bool is_fp8(const torch::Tensor& t) {
+  return t.scalar_type() == at::kFloat8_e4m3fn;
+}
+
 static int64_t row_stride(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 && t.stride(1) == t.size(2),
               name, " must be [T, H, D] with dense inner dims");
@@ -115,10 +120,13 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
 void rope_and_cache(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor k_cache, torch::Tensor v_cache,
                     torch::Tensor positions, torch::Tensor cos_sin,
-                    torch::Tensor slot_mapping, bool is_neox) {
+                    torch::Tensor slot_mapping, bool is_neox, double k_scale,
+                    double v_scale) {
   CHECK_GPU(q);
   CHECK_BF16(q);
-  CHECK_BF16(k_cache);
+  const bool fp8 = is_fp8(k_cache);
+  TORCH_CHECK(fp8 || k_cache.scalar_type() == at::kBFloat16,
+              "k_cache must be bf16 or fp8_e4m3fn");
   TORCH_CHECK(positions.scalar_type() == at::kInt);
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
   TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
@@ -132,9 +140,9 @@ void rope_and_cache(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   launch_rope_and_cache(
       q.data_ptr(), k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
       v_cache.data_ptr(), positions.data_ptr<int>(), cos_sin.data_ptr<float>(),
-      slot_mapping.data_ptr<int64_t>(), T, Hq, Hk, D, rot, BS, is_neox,
-      row_stride(q, "q"), row_stride(k, "k"), row_stride(v, "v"),
-      cur_stream());
+      slot_mapping.data_ptr<int64_t>(), T, Hq, Hk, D, rot, BS, is_neox, fp8,
+      (float)k_scale, (float)v_scale, row_stride(q, "q"), row_stride(k, "k"),
+      row_stride(v, "v"), cur_stream());
 }
 
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
@@ -193,10 +201,13 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
                             double scale, int64_t sliding_window,
                             double softcap, torch::Tensor sinks,
-                            int64_t max_seq_len) {
+                            int64_t max_seq_len, double k_scale,
+                            double v_scale) {
   CHECK_GPU(q);
   CHECK_BF16(q);
-  CHECK_BF16(k_cache);
+  const bool kv_fp8 = is_fp8(k_cache);
+  TORCH_CHECK(kv_fp8 || k_cache.scalar_type() == at::kBFloat16,
+              "k_cache must be bf16 or fp8_e4m3fn");
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
   const int B = q.size(0);
@@ -238,13 +249,13 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   }
   bool launched = false;
   static const bool use_valu = std::getenv("PARALLAX_ATTN_VALU") != nullptr;
-  if (!use_valu) {
+  if (!use_valu || kv_fp8) {
     launch_paged_decode_mfma(
         out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
         block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hq, Hk, D,
         BS, max_blocks, row_stride(q, "q"), (float)scale, (int)sliding_window,
         (float)softcap, sinks_ptr, num_parts, part_tokens, acc_ptr, ml_ptr,
-        cur_stream(), &launched);
+        kv_fp8, (float)k_scale, (float)v_scale, cur_stream(), &launched);
   }
   if (!launched) {
     launch_paged_attention_decode(
@@ -267,10 +278,12 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                        torch::Tensor cu_q, torch::Tensor tile_req,
                        torch::Tensor tile_row0, double scale,
                        int64_t sliding_window, double softcap,
-                       torch::Tensor sinks) {
+                       torch::Tensor sinks, double k_scale, double v_scale) {
   CHECK_GPU(q);
   CHECK_BF16(q);
-  CHECK_BF16(k_cache);
+  const bool kv_fp8 = is_fp8(k_cache);
+  TORCH_CHECK(kv_fp8 || k_cache.scalar_type() == at::kBFloat16,
+              "k_cache must be bf16 or fp8_e4m3fn");
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
   TORCH_CHECK(cu_q.scalar_type() == at::kInt);
@@ -294,7 +307,8 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
       cu_q.data_ptr<int>(), tile_req.data_ptr<int>(), tile_row0.data_ptr<int>(),
       n_tiles, Hq, Hk, D, BS, max_blocks, row_stride(q, "q"), (float)scale,
-      (int)sliding_window, (float)softcap, sinks_ptr, cur_stream(), &launched);
+      (int)sliding_window, (float)softcap, sinks_ptr, kv_fp8, (float)k_scale,
+      (float)v_scale, cur_stream(), &launched);
   TORCH_CHECK(launched, "no prefill kernel instantiation for D=", D);
 }
 

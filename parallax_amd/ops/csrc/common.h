@@ -5,6 +5,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
+#include <hip/hip_fp8.h>
 #include <stdint.h>
 
 #define WAVE_SIZE 64
@@ -103,6 +104,19 @@ DEVINL float block_reduce_sum(float v, float* lds_scratch) {
   }
   __syncthreads();
   return lds_scratch[0];
+}
+
+// ---- FP8 (OCP e4m3fn — gfx950 native; NOT the MI300X fnuz variant) -----------
+
+DEVINL float fp8_e4m3_to_f32(uint8_t b) {
+  __hip_fp8_e4m3 v;
+  v.__x = b;
+  return static_cast<float>(v);
+}
+
+DEVINL uint8_t f32_to_fp8_e4m3(float f) {
+  __hip_fp8_e4m3 v(f);
+  return v.__x;
 }
 
 typedef short bf16x8v __attribute__((ext_vector_type(8)));

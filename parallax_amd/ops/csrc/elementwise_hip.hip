@@ -130,25 +130,29 @@ extern "C" void launch_rope(
 // One launch per decode/prefill step replaces {rope(q,k) ; reshape_and_cache}:
 // q roped in place (strided rows OK), k roped straight into the paged cache,
 // v copied vectorized. Saves one k round-trip and two kernel launches per layer.
-template <bool NEOX>
+template <bool NEOX, bool FP8>
 __global__ void rope_and_cache_kernel(
     uint16_t* __restrict__ q,        // [T, Hq, D] rows at q_stride
     const uint16_t* __restrict__ k,  // [T, Hk, D] rows at k_stride
     const uint16_t* __restrict__ v,
-    uint16_t* __restrict__ k_cache,  // [NB, Hk, BS, D]
-    uint16_t* __restrict__ v_cache,
+    void* __restrict__ k_cache_v,    // [NB, Hk, BS, D] bf16 | fp8
+    void* __restrict__ v_cache_v,
     const int* __restrict__ positions, const float* __restrict__ cos_sin,
     const int64_t* __restrict__ slot_mapping,
     const int Hq, const int Hk, const int D, const int rot, const int BS,
-    const int64_t q_stride, const int64_t k_stride, const int64_t v_stride) {
+    const int64_t q_stride, const int64_t k_stride, const int64_t v_stride,
+    const float inv_k_scale, const float inv_v_scale) {
   const int t = blockIdx.x;
   const int64_t slot = slot_mapping[t];
   const int half = rot / 2;
   const float* cs = cos_sin + (size_t)positions[t] * rot;
   const int64_t blk = slot >= 0 ? slot / BS : 0;
   const int64_t off = slot >= 0 ? slot % BS : 0;
-  uint16_t* kdst = k_cache + (((size_t)blk * Hk) * BS + off) * D;
-  uint16_t* vdst = v_cache + (((size_t)blk * Hk) * BS + off) * D;
+  const size_t base_off = (((size_t)blk * Hk) * BS + off) * D;
+  uint16_t* kdst16 = FP8 ? nullptr : (uint16_t*)k_cache_v + base_off;
+  uint16_t* vdst16 = FP8 ? nullptr : (uint16_t*)v_cache_v + base_off;
+  uint8_t* kdst8 = FP8 ? (uint8_t*)k_cache_v + base_off : nullptr;
+  uint8_t* vdst8 = FP8 ? (uint8_t*)v_cache_v + base_off : nullptr;
 
   // q: rope in place
   for (int i = threadIdx.x; i < Hq * half; i += blockDim.x) {
@@ -167,48 +171,62 @@ __global__ void rope_and_cache_kernel(
   for (int i = threadIdx.x; i < Hk * half; i += blockDim.x) {
     const int h = i / half, j = i % half;
     const uint16_t* src = k + (size_t)t * k_stride + h * D;
-    uint16_t* dst = kdst + (size_t)h * BS * D;
     const float c = cs[j], ss = cs[half + j];
     const int i1 = NEOX ? j : 2 * j;
     const int i2 = NEOX ? j + half : 2 * j + 1;
     const float x1 = bf16_bits_to_f32(src[i1]);
     const float x2 = bf16_bits_to_f32(src[i2]);
-    dst[i1] = f32_to_bf16_bits(fmaf(x1, c, -x2 * ss));
-    dst[i2] = f32_to_bf16_bits(fmaf(x2, c, x1 * ss));
-  }
-  if (rot < D) {
-    for (int i = threadIdx.x; i < Hk * (D - rot) / 8; i += blockDim.x) {
-      const int h = (i * 8) / (D - rot);
-      const int d = rot + (i * 8) % (D - rot);
-      *reinterpret_cast<int4*>(kdst + (size_t)h * BS * D + d) =
-          *reinterpret_cast<const int4*>(k + (size_t)t * k_stride + h * D + d);
+    const float o1 = fmaf(x1, c, -x2 * ss);
+    const float o2 = fmaf(x2, c, x1 * ss);
+    if (FP8) {
+      kdst8[(size_t)h * BS * D + i1] = f32_to_fp8_e4m3(o1 * inv_k_scale);
+      kdst8[(size_t)h * BS * D + i2] = f32_to_fp8_e4m3(o2 * inv_k_scale);
+    } else {
+      kdst16[(size_t)h * BS * D + i1] = f32_to_bf16_bits(o1);
+      kdst16[(size_t)h * BS * D + i2] = f32_to_bf16_bits(o2);
     }
   }
-  // v: straight vectorized copy into the cache
-  for (int i = threadIdx.x; i < Hk * D / 8; i += blockDim.x) {
-    const int h = (i * 8) / D;
-    const int d = (i * 8) % D;
-    *reinterpret_cast<int4*>(vdst + (size_t)h * BS * D + d) =
-        *reinterpret_cast<const int4*>(v + (size_t)t * v_stride + h * D + d);
+  if (rot < D) {
+    for (int i = threadIdx.x; i < Hk * (D - rot); i += blockDim.x) {
+      const int h = i / (D - rot);
+      const int d = rot + i % (D - rot);
+      const float x = bf16_bits_to_f32(k[(size_t)t * k_stride + h * D + d]);
+      if (FP8)
+        kdst8[(size_t)h * BS * D + d] = f32_to_fp8_e4m3(x * inv_k_scale);
+      else
+        kdst16[(size_t)h * BS * D + d] = f32_to_bf16_bits(x);
+    }
+  }
+  // v: copy (quantized when FP8)
+  for (int i = threadIdx.x; i < Hk * D; i += blockDim.x) {
+    const int h = i / D;
+    const int d = i % D;
+    const float x = bf16_bits_to_f32(v[(size_t)t * v_stride + h * D + d]);
+    if (FP8)
+      vdst8[(size_t)h * BS * D + d] = f32_to_fp8_e4m3(x * inv_v_scale);
+    else
+      vdst16[(size_t)h * BS * D + d] = f32_to_bf16_bits(x);
   }
 }
 
 extern "C" void launch_rope_and_cache(
     void* q, const void* k, const void* v, void* k_cache, void* v_cache,
     const int* positions, const float* cos_sin, const int64_t* slot_mapping,
-    int T, int Hq, int Hk, int D, int rot, int BS, bool neox,
-    int64_t q_stride, int64_t k_stride, int64_t v_stride, hipStream_t stream) {
+    int T, int Hq, int Hk, int D, int rot, int BS, bool neox, bool fp8,
+    float k_scale, float v_scale, int64_t q_stride, int64_t k_stride,
+    int64_t v_stride, hipStream_t stream) {
   int threads = min(512, max(128, ceil_div(Hk * D / 8, 64) * 64));
-  if (neox)
-   hipLaunchKernelGGL(( rope_and_cache_kernel<true>), dim3(T), dim3(threads), 0, stream, 
-        (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
-        (uint16_t*)k_cache, (uint16_t*)v_cache, positions, cos_sin,
-        slot_mapping, Hq, Hk, D, rot, BS, q_stride, k_stride, v_stride);
-  else
-   hipLaunchKernelGGL(( rope_and_cache_kernel<false>), dim3(T), dim3(threads), 0, stream, 
-        (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
-        (uint16_t*)k_cache, (uint16_t*)v_cache, positions, cos_sin,
-        slot_mapping, Hq, Hk, D, rot, BS, q_stride, k_stride, v_stride);
+  const float iks = 1.f / k_scale, ivs = 1.f / v_scale;
+#define RC_LAUNCH(NEOX, FP8)                                                 \
+ hipLaunchKernelGGL(( rope_and_cache_kernel<NEOX, FP8>), dim3(T), dim3(threads), 0, stream,               \
+      (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v, k_cache,        \
+      v_cache, positions, cos_sin, slot_mapping, Hq, Hk, D, rot, BS,        \
+      q_stride, k_stride, v_stride, iks, ivs)
+  if (neox && fp8) RC_LAUNCH(true, true);
+  else if (neox) RC_LAUNCH(true, false);
+  else if (fp8) RC_LAUNCH(false, true);
+  else RC_LAUNCH(false, false);
+#undef RC_LAUNCH
 }
 
 // ---- reshape_and_cache ------------------------------------------------------------

@@ -15,19 +15,20 @@
 #define DM_THREADS 256
 #define DM_KTILE 64
 
-template <int HEAD_DIM, int BLOCK_SIZE, bool PARTITIONED>
+template <int HEAD_DIM, int BLOCK_SIZE, bool PARTITIONED, bool KV_FP8>
 __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     uint16_t* __restrict__ out,        // [B, Hq, D] (final mode)
     float* __restrict__ tmp_acc,       // [B, Hq, P, D] (partitioned)
     float* __restrict__ tmp_ml,        // [B, Hq, P, 2]
     const uint16_t* __restrict__ q,    // [B, Hq, D] rows at q_stride
-    const uint16_t* __restrict__ k_cache,  // [NB, Hk, BS, D]
-    const uint16_t* __restrict__ v_cache,
+    const void* __restrict__ k_cache_v,  // [NB, Hk, BS, D] bf16 | fp8
+    const void* __restrict__ v_cache_v,
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     const int max_blocks, const int Hk, const int G, const int64_t q_stride,
     const float scale, const int sliding_window, const float softcap,
-    const float* __restrict__ sinks, const int part_tokens) {
+    const float* __restrict__ sinks, const int part_tokens,
+    const float k_scale, const float v_scale) {
   const int seq = blockIdx.y;
   const int hk = blockIdx.x;
   const int L = seq_lens[seq];
@@ -114,24 +115,43 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
       const int d0 = dv * (HEAD_DIM / 4);
       const int gtok = kbase + tok;
       const bool ok = gtok >= tok_begin && gtok < tok_end;
-      const uint16_t* krow = nullptr;
-      const uint16_t* vrow = nullptr;
+      size_t row_off = 0;
       if (ok) {
         const int blk = btab[gtok / BLOCK_SIZE];
         const int off = gtok % BLOCK_SIZE;
-        krow = k_cache + (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM;
-        vrow = v_cache + (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM;
+        row_off = (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM;
       }
 #pragma unroll
       for (int c = 0; c < HEAD_DIM / 32; ++c) {
         const int d = d0 + c * 8;
         int4 kval = make_int4(0, 0, 0, 0);
-        if (ok) kval = *reinterpret_cast<const int4*>(krow + d);
+        int4 vval = make_int4(0, 0, 0, 0);
+        if (ok) {
+          if (KV_FP8) {
+            // 8 fp8 bytes -> 8 bf16 (scale folded in)
+            const uint8_t* k8 = (const uint8_t*)k_cache_v + row_off + d;
+            const uint8_t* v8 = (const uint8_t*)v_cache_v + row_off + d;
+            uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
+            uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
+            const uint64_t kraw = *reinterpret_cast<const uint64_t*>(k8);
+            const uint64_t vraw = *reinterpret_cast<const uint64_t*>(v8);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              ks[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+              vsp[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
+            }
+          } else {
+            kval = *reinterpret_cast<const int4*>(
+                (const uint16_t*)k_cache_v + row_off + d);
+            vval = *reinterpret_cast<const int4*>(
+                (const uint16_t*)v_cache_v + row_off + d);
+          }
+        }
         const int kb = swz(tok * HEAD_DIM * 2 + d * 2, tok);
         *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kval;
 
-        int4 vval = make_int4(0, 0, 0, 0);
-        if (ok) vval = *reinterpret_cast<const int4*>(vrow + d);
         const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vval);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -274,30 +294,32 @@ extern "C" void launch_paged_decode_mfma(
     const int* block_tables, const int* seq_lens, int B, int Hq, int Hk, int D,
     int BS, int max_blocks, int64_t q_stride, float scale, int sliding_window,
     float softcap, const float* sinks, int num_parts, int part_tokens,
-    float* tmp_acc, float* tmp_ml, hipStream_t stream, bool* launched) {
+    float* tmp_acc, float* tmp_ml, bool kv_fp8, float k_scale, float v_scale,
+    hipStream_t stream, bool* launched) {
   const int G = Hq / Hk;
   *launched = false;
   if (G > 16) return;
 
-#define DM_LAUNCH(HD, BSZ)                                                     \
+#define DM_LAUNCH2(HD, BSZ, FP8)                                               \
   if (num_parts <= 1) {                                                        \
     dim3 grid(Hk, B, 1);                                                       \
-    paged_decode_mfma_kernel<HD, BSZ, false>                                   \
+    paged_decode_mfma_kernel<HD, BSZ, false, FP8>                              \
         <<<grid, DM_THREADS, 0, stream>>>(                                     \
-            (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q,              \
-            (const uint16_t*)k_cache, (const uint16_t*)v_cache, block_tables,  \
-            seq_lens, max_blocks, Hk, G, q_stride, scale, sliding_window,      \
-            softcap, sinks, 0);                                                \
+            (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q, k_cache,     \
+            v_cache, block_tables, seq_lens, max_blocks, Hk, G, q_stride,      \
+            scale, sliding_window, softcap, sinks, 0, k_scale, v_scale);       \
   } else {                                                                     \
     dim3 grid(Hk, B, num_parts);                                               \
-    paged_decode_mfma_kernel<HD, BSZ, true>                                    \
+    paged_decode_mfma_kernel<HD, BSZ, true, FP8>                               \
         <<<grid, DM_THREADS, 0, stream>>>(                                     \
-            nullptr, tmp_acc, tmp_ml, (const uint16_t*)q,                      \
-            (const uint16_t*)k_cache, (const uint16_t*)v_cache, block_tables,  \
-            seq_lens, max_blocks, Hk, G, q_stride, scale, sliding_window,      \
-            softcap, sinks, part_tokens);                                      \
+            nullptr, tmp_acc, tmp_ml, (const uint16_t*)q, k_cache, v_cache,    \
+            block_tables, seq_lens, max_blocks, Hk, G, q_stride, scale,        \
+            sliding_window, softcap, sinks, part_tokens, k_scale, v_scale);    \
   }                                                                            \
   *launched = true;
+
+#define DM_LAUNCH(HD, BSZ)                                                     \
+  if (kv_fp8) { DM_LAUNCH2(HD, BSZ, true) } else { DM_LAUNCH2(HD, BSZ, false) }
 
   if (D == 128 && BS == 32) { DM_LAUNCH(128, 32) }
   else if (D == 128 && BS == 16) { DM_LAUNCH(128, 16) }
@@ -306,4 +328,5 @@ extern "C" void launch_paged_decode_mfma(
   else if (D == 64 && BS == 16) { DM_LAUNCH(64, 16) }
   else if (D == 64 && BS == 64) { DM_LAUNCH(64, 64) }
 #undef DM_LAUNCH
+#undef DM_LAUNCH2
 }

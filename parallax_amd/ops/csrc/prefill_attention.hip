@@ -22,12 +22,12 @@
 #define QTILE 32
 #define KTILE 64
 
-template <int HEAD_DIM>
+template <int HEAD_DIM, bool KV_FP8>
 __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
     uint16_t* __restrict__ out,            // [T, Hq, D]
     const uint16_t* __restrict__ q,        // [T, Hq, D]
-    const uint16_t* __restrict__ k_cache,  // [NB, Hk, BS, D]
-    const uint16_t* __restrict__ v_cache,
+    const void* __restrict__ k_cache_v,    // [NB, Hk, BS, D] bf16 | fp8
+    const void* __restrict__ v_cache_v,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ seq_lens,      // [B] total ctx (prefix + new)
     const int* __restrict__ cu_q,          // [B+1] query offsets
@@ -35,7 +35,8 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
     const int* __restrict__ tile_row0,     // [n_tiles] first q row of the tile
     const int max_blocks, const int Hq, const int Hk, const int BS,
     const int64_t q_stride, const float scale, const int sliding_window,
-    const float softcap, const float* __restrict__ sinks) {
+    const float softcap, const float* __restrict__ sinks,
+    const float k_scale, const float v_scale) {
   const int h = blockIdx.x;
   const int tile = blockIdx.y;
   const int req = tile_req[tile];
@@ -102,23 +103,41 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
       const int d = dv * (HEAD_DIM / 4);
       const int gtok = kbase + tok;
       const bool ok = gtok < L;
-      const uint16_t* krow = nullptr;
-      const uint16_t* vrow = nullptr;
+      size_t row_off = 0;
       if (ok) {
         const int blk = btab[gtok / BS];
         const int off = gtok % BS;
-        krow = k_cache + (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
-        vrow = v_cache + (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
+        row_off = (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
       }
 #pragma unroll
       for (int c = 0; c < HEAD_DIM / 32; ++c) {  // 8 elems per step
         int4 kval = make_int4(0, 0, 0, 0);
-        if (ok) kval = *reinterpret_cast<const int4*>(krow + d + c * 8);
+        int4 vval = make_int4(0, 0, 0, 0);
+        if (ok) {
+          if (KV_FP8) {
+            const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
+                (const uint8_t*)k_cache_v + row_off + d + c * 8);
+            const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
+                (const uint8_t*)v_cache_v + row_off + d + c * 8);
+            uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
+            uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              ks[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+              vsp[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
+            }
+          } else {
+            kval = *reinterpret_cast<const int4*>(
+                (const uint16_t*)k_cache_v + row_off + d + c * 8);
+            vval = *reinterpret_cast<const int4*>(
+                (const uint16_t*)v_cache_v + row_off + d + c * 8);
+          }
+        }
         const int kb = swz(tok * HEAD_DIM * 2 + (d + c * 8) * 2, tok);
         *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kval;
 
-        int4 vval = make_int4(0, 0, 0, 0);
-        if (ok) vval = *reinterpret_cast<const int4*>(vrow + d + c * 8);
         const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vval);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -265,23 +284,19 @@ extern "C" void launch_prefill_attention(
     const int* block_tables, const int* seq_lens, const int* cu_q,
     const int* tile_req, const int* tile_row0, int n_tiles, int Hq, int Hk,
     int D, int BS, int max_blocks, int64_t q_stride, float scale,
-    int sliding_window, float softcap, const float* sinks, hipStream_t stream,
-    bool* launched) {
+    int sliding_window, float softcap, const float* sinks, bool kv_fp8,
+    float k_scale, float v_scale, hipStream_t stream, bool* launched) {
   *launched = false;
   dim3 grid(Hq, n_tiles, 1);
-  if (D == 128) {
-    prefill_attention_kernel<128><<<grid, PF_THREADS, 0, stream>>>(
-        (uint16_t*)out, (const uint16_t*)q, (const uint16_t*)k_cache,
-        (const uint16_t*)v_cache, block_tables, seq_lens, cu_q, tile_req,
-        tile_row0, max_blocks, Hq, Hk, BS, q_stride, scale, sliding_window,
-        softcap, sinks);
-    *launched = true;
-  } else if (D == 64) {
-    prefill_attention_kernel<64><<<grid, PF_THREADS, 0, stream>>>(
-        (uint16_t*)out, (const uint16_t*)q, (const uint16_t*)k_cache,
-        (const uint16_t*)v_cache, block_tables, seq_lens, cu_q, tile_req,
-        tile_row0, max_blocks, Hq, Hk, BS, q_stride, scale, sliding_window,
-        softcap, sinks);
-    *launched = true;
-  }
+#define PF_LAUNCH(HD, FP8)                                                   \
+  prefill_attention_kernel<HD, FP8><<<grid, PF_THREADS, 0, stream>>>(        \
+      (uint16_t*)out, (const uint16_t*)q, k_cache, v_cache, block_tables,    \
+      seq_lens, cu_q, tile_req, tile_row0, max_blocks, Hq, Hk, BS, q_stride, \
+      scale, sliding_window, softcap, sinks, k_scale, v_scale);              \
+  *launched = true;
+  if (D == 128 && kv_fp8) { PF_LAUNCH(128, true) }
+  else if (D == 128) { PF_LAUNCH(128, false) }
+  else if (D == 64 && kv_fp8) { PF_LAUNCH(64, true) }
+  else if (D == 64) { PF_LAUNCH(64, false) }
+#undef PF_LAUNCH
 }

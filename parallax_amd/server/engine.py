@@ -68,6 +68,7 @@ class EngineArgs:
     end_layer: Optional[int] = None
     max_model_len: int = 8192           # context ceiling (sizes graph buffers)
     enable_graphs: bool = True          # hipGraph-captured decode forward
+    kv_cache_dtype: str = "auto"        # "auto" (= engine dtype) | "fp8"(e4m3)
 
 
 def partition_layers(num_layers: int, pp_size: int, pp_rank: int) -> Tuple[int, int]:
@@ -136,12 +137,16 @@ class Engine:
                 args.dtype if args.dtype != torch.float32 else torch.float32,
             )
         else:
+            kv_dtype = args.dtype
+            if args.kv_cache_dtype == "fp8":
+                assert self.device.type == "cuda", "fp8 KV needs the HIP kernels"
+                kv_dtype = torch.float8_e4m3fn
             spec = KVCacheSpec(
                 num_layers=end - start,
                 num_kv_heads=max(1, cfg.num_kv_heads // self.comm.tp_size),
                 head_dim=cfg.head_dim,
                 block_size=args.block_size,
-                dtype=args.dtype,
+                dtype=kv_dtype,
             )
             num_blocks = args.num_kv_blocks or CacheManager.num_blocks_from_memory(
                 spec, self.device, args.cache_memory_fraction

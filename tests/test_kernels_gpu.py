@@ -248,3 +248,70 @@ def test_paged_attention_strided_q():
         seq_lens.cpu(), scale,
     )
     _assert_close(out, expect, atol=3e-2, rtol=3e-2)
+
+
+def _fp8_roundtrip_cpu(t: torch.Tensor) -> torch.Tensor:
+    return t.to(torch.float8_e4m3fn).float()
+
+
+def test_rope_and_cache_fp8():
+    torch.manual_seed(11)
+    T, Hq, Hk, D, BS, NB = 20, 8, 2, 128, 32, 8
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(NB, Hk, BS, D, dtype=torch.float8_e4m3fn, device="cuda")
+    vc = torch.zeros_like(kc)
+    pos = torch.randint(0, 400, (T,), dtype=torch.int32, device="cuda")
+    cs = ops.build_rope_cache(512, D, 10000.0).cuda()
+    slots = torch.randperm(NB * BS, device="cuda")[:T].to(torch.int64)
+
+    k_ref = k.float().cpu().clone()
+    q_ref = q.float().cpu().clone()
+    ref.rope_inplace(q_ref, k_ref, pos.cpu(), cs.cpu(), True)
+    ops.rope_and_cache(q, k, v, kc, vc, pos, cs, slots)
+    # cache holds fp8-quantized roped K / raw V
+    blk, off = slots.cpu() // BS, slots.cpu() % BS
+    got_k = kc.cpu().float()[blk, :, off]
+    got_v = vc.cpu().float()[blk, :, off]
+    torch.testing.assert_close(got_k, _fp8_roundtrip_cpu(k_ref), atol=8e-2, rtol=8e-2)
+    torch.testing.assert_close(got_v, _fp8_roundtrip_cpu(v.float().cpu()),
+                               atol=8e-2, rtol=8e-2)
+
+
+def test_paged_attention_decode_fp8_kv():
+    torch.manual_seed(12)
+    Hk, G, D, BS = 2, 4, 128, 32
+    ctxs = [300, 64, 1500]
+    B, Hq = len(ctxs), G * 2
+    kc_b, vc_b, bt = _make_paged_kv(B, Hk, D, BS, max(ctxs), seed=3)
+    kc = kc_b.to(torch.float8_e4m3fn)
+    vc = vc_b.to(torch.float8_e4m3fn)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(ctxs, dtype=torch.int32, device="cuda")
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, kc, vc, bt, seq_lens, scale)
+    expect = ref.paged_attention_decode(
+        q.float().cpu(), kc.cpu().float(), vc.cpu().float(), bt.cpu(),
+        seq_lens.cpu(), scale,
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=5e-2, rtol=5e-2)
+
+
+def test_engine_fp8_kv_end_to_end():
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=256,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=128,
+        intermediate_size=512, max_position_embeddings=2048, eos_token_ids=[],
+    )
+    eng = Engine(cfg, EngineArgs(num_kv_blocks=256, kv_cache_dtype="fp8"),
+                 random_weights=True)
+    out = eng.generate(
+        [[1, 2, 3, 4, 5] * 8, [9] * 11],
+        [SamplingParams(temperature=0.0, max_new_tokens=5, ignore_eos=True)] * 2,
+    )
+    assert all(len(v) == 5 for v in out.values())
