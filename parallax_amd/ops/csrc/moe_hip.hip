@@ -18,7 +18,8 @@
 #include "common.h"
 
 #define MOE_THREADS 256
-#define MOE_TM 16  // assignment rows per tile
+#define MOE_TM 64  // assignment rows per tile (4 MFMA M-tiles; big tiles
+                   // amortize the expert-weight stream across more tokens)
 
 __global__ void build_moe_tiles_kernel(
     int* __restrict__ tile_expert,   // [max_tiles]
@@ -70,49 +71,55 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_kernel(
   const uint16_t* wg_row = w_gu + ((size_t)e * 2 * I + (n0 + l15)) * H;
   const uint16_t* wu_row = w_gu + ((size_t)e * 2 * I + (I + n0 + l15)) * H;
 
-  // this lane's A row (token) — fixed for the whole K loop
-  const int arow = row0 + l15;
-  const bool arow_ok = arow < rows_end;
-  const uint16_t* xrow =
-      arow_ok ? x + (size_t)(perm[arow] / topk) * H : x;
-
-  f32x4v acc_g = {}, acc_u = {};
-  for (int k = 0; k < H; k += 32) {
-    bf16x8v afrag = {};
-    if (arow_ok)
-      afrag = *reinterpret_cast<const bf16x8v*>(xrow + k + l4 * 8);
-    const bf16x8v bg = *reinterpret_cast<const bf16x8v*>(wg_row + k + l4 * 8);
-    const bf16x8v bu = *reinterpret_cast<const bf16x8v*>(wu_row + k + l4 * 8);
-    acc_g = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bg, acc_g, 0, 0, 0);
-    acc_u = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bu, acc_u, 0, 0, 0);
+  // 4 token M-tiles share each streamed W fragment
+  const uint16_t* xrow[4];
+  bool arow_ok[4];
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+    const int arow = row0 + mt * 16 + l15;
+    arow_ok[mt] = arow < rows_end;
+    xrow[mt] = arow_ok[mt] ? x + (size_t)(perm[arow] / topk) * H : x;
   }
 
-  // acc layout: col = n (l15... wait: A rows = tokens = M; B cols = W rows = N
-  // D: col(lane&15) = N index (output feature), row = M (token)  — but our A
-  // fragment used lane&15 as the TOKEN row and B used lane&15 as the W row;
-  // with both operands "row-major k-consecutive", D[m][n]: m = A row, n = B row:
-  // col = lane&15 -> n? No: mfma semantics fix A rows to M. We loaded
-  // A[m = l15] and B[n = l15]; D element (lane, reg): row m = (lane>>4)*4+reg,
-  // col n = lane&15.
+  f32x4v acc_g[4] = {}, acc_u[4] = {};
+  for (int k = 0; k < H; k += 32) {
+    const bf16x8v bg = *reinterpret_cast<const bf16x8v*>(wg_row + k + l4 * 8);
+    const bf16x8v bu = *reinterpret_cast<const bf16x8v*>(wu_row + k + l4 * 8);
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int m = l4 * 4 + r;          // token row within tile
-    const int grow = row0 + m;
-    if (grow >= rows_end) continue;
-    const int n = n0 + l15;            // column within I
-    float g = acc_g[r], u = acc_u[r];
-    float a;
-    if (limit > 0.f) {                 // gpt-oss clamped interleaved-style act
-      g = fminf(g, limit);
-      u = fminf(fmaxf(u, -limit), limit);
-      a = (u + 1.f) * (g / (1.f + __expf(-g * 1.702f)));
-    } else if (GELU) {
-      const float c = 0.7978845608028654f;
-      a = 0.5f * g * (1.f + tanhf(c * (g + 0.044715f * g * g * g))) * u;
-    } else {
-      a = g / (1.f + __expf(-g)) * u;
+    for (int mt = 0; mt < 4; ++mt) {
+      if (row0 + mt * 16 >= rows_end) break;
+      bf16x8v afrag = {};
+      if (arow_ok[mt])
+        afrag = *reinterpret_cast<const bf16x8v*>(xrow[mt] + k + l4 * 8);
+      acc_g[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bg, acc_g[mt], 0, 0, 0);
+      acc_u[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bu, acc_u[mt], 0, 0, 0);
     }
-    h_buf[(size_t)grow * I + n] = f32_to_bf16_bits(a);
+  }
+
+  // D layout (verified by the numerics tests): element (lane, reg) is
+  // row m = (lane>>4)*4 + reg (token), col n = lane&15 (output feature).
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mt * 16 + l4 * 4 + r;  // token row within tile
+      const int grow = row0 + m;
+      if (grow >= rows_end) continue;
+      const int n = n0 + l15;              // column within I
+      float g = acc_g[mt][r], u = acc_u[mt][r];
+      float a;
+      if (limit > 0.f) {                   // gpt-oss clamped act
+        g = fminf(g, limit);
+        u = fminf(fmaxf(u, -limit), limit);
+        a = (u + 1.f) * (g / (1.f + __expf(-g * 1.702f)));
+      } else if (GELU) {
+        const float c = 0.7978845608028654f;
+        a = 0.5f * g * (1.f + tanhf(c * (g + 0.044715f * g * g * g))) * u;
+      } else {
+        a = g / (1.f + __expf(-g)) * u;
+      }
+      h_buf[(size_t)grow * I + n] = f32_to_bf16_bits(a);
+    }
   }
 }
 
@@ -142,36 +149,52 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_kernel(
   const int l4 = lane >> 4;
 
   const int n0 = blockIdx.y * 256 + wid * 64;  // 4 waves x 4 n-tiles = 256 cols
-  const int arow = row0 + l15;
-  const bool arow_ok = arow < rows_end;
-  const uint16_t* hrow = arow_ok ? h_buf + (size_t)arow * I : h_buf;
+  const uint16_t* hrow[4];
+  bool arow_ok[4];
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+    const int arow = row0 + mt * 16 + l15;
+    arow_ok[mt] = arow < rows_end;
+    hrow[mt] = arow_ok[mt] ? h_buf + (size_t)arow * I : h_buf;
+  }
 
-  f32x4v acc[4] = {};
+  f32x4v acc[4][4] = {};
   for (int k = 0; k < I; k += 32) {
-    bf16x8v afrag = {};
-    if (arow_ok)
-      afrag = *reinterpret_cast<const bf16x8v*>(hrow + k + l4 * 8);
+    bf16x8v bfr[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int n = n0 + nt * 16 + l15;
+      bfr[nt] = {};
+      if (n < H)
+        bfr[nt] = *reinterpret_cast<const bf16x8v*>(
+            w_down + ((size_t)e * H + n) * I + k + l4 * 8);
+    }
+#pragma unroll
+    for (int mt = 0; mt < 4; ++mt) {
+      if (row0 + mt * 16 >= rows_end) break;
+      bf16x8v afrag = {};
+      if (arow_ok[mt])
+        afrag = *reinterpret_cast<const bf16x8v*>(hrow[mt] + k + l4 * 8);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfr[nt], acc[mt][nt], 0, 0, 0);
+    }
+  }
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       const int n = n0 + nt * 16 + l15;
       if (n >= H) break;
-      const bf16x8v b = *reinterpret_cast<const bf16x8v*>(
-          w_down + ((size_t)e * H + n) * I + k + l4 * 8);
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, b, acc[nt], 0, 0, 0);
-    }
-  }
 #pragma unroll
-  for (int nt = 0; nt < 4; ++nt) {
-    const int n = n0 + nt * 16 + l15;
-    if (n >= H) break;
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = l4 * 4 + r;
-      const int grow = row0 + m;
-      if (grow >= rows_end) continue;
-      const float w = route_w[grow];
-      const int64_t token = perm[grow] / topk;
-      atomicAdd(out + (size_t)token * H + n, acc[nt][r] * w);
+      for (int r = 0; r < 4; ++r) {
+        const int grow = row0 + mt * 16 + l4 * 4 + r;
+        if (grow >= rows_end) continue;
+        const float w = route_w[grow];
+        const int64_t token = perm[grow] / topk;
+        atomicAdd(out + (size_t)token * H + n, acc[mt][nt][r] * w);
+      }
     }
   }
 }
