@@ -304,6 +304,35 @@ def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return ref.gelu_and_mul(x)
 
 
+# -- dense linear (skinny-M fast path) ------------------------------------------
+
+SKINNY_GEMM_MAX_M = int(os.environ.get("PARALLAX_SKINNY_GEMM_MAX_M", "256"))
+
+
+def linear(
+    x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None
+) -> torch.Tensor:
+    """F.linear with a custom streaming kernel for decode-sized M (hipBLASLt
+    tile picks are 2-5x off the weight-streaming roofline at M <= 256)."""
+    if (
+        x.is_cuda
+        and x.dim() == 2
+        and x.dtype == torch.bfloat16
+        and 0 < x.shape[0] <= SKINNY_GEMM_MAX_M
+        and weight.shape[0] % 64 == 0
+        and weight.shape[1] % 32 == 0
+        and x.stride(1) == 1
+    ):
+        ext = _require_ext("skinny_gemm")
+        out = torch.empty(
+            x.shape[0], weight.shape[0], dtype=x.dtype, device=x.device
+        )
+        if ext.skinny_gemm(out, x, weight,
+                           bias if bias is not None else x.new_empty(0)):
+            return out
+    return torch.nn.functional.linear(x, weight, bias)
+
+
 # -- MoE ----------------------------------------------------------------------
 
 

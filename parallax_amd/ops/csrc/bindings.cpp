@@ -52,6 +52,8 @@ void launch_moe_gate_up(void*, const void*, const void*, const int*,
 void launch_moe_down(void*, const void*, const void*, const float*, const int*,
                      const int*, const int64_t*, const int*, int, int, int,
                      int, int, hipStream_t);
+void launch_skinny_gemm(void*, const void*, const void*, const void*, int,
+                        int, int, int64_t, int64_t, hipStream_t, bool*);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const int*, const int*, const int*, const int*,
                               const int*, int, int, int, int, int, int, int64_t,
@@ -399,7 +401,26 @@ void moe_forward(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
                   stream);
 }
 
+bool skinny_gemm(torch::Tensor c, torch::Tensor x, torch::Tensor w,
+                 torch::Tensor bias) {
+  CHECK_GPU(x);
+  CHECK_BF16(x);
+  CHECK_BF16(w);
+  TORCH_CHECK(w.is_contiguous());
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1);
+  TORCH_CHECK(c.dim() == 2 && c.stride(1) == 1);
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int N = w.size(0);
+  bool launched = false;
+  launch_skinny_gemm(c.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     bias.numel() ? bias.data_ptr() : nullptr, M, N, K,
+                     x.stride(0), c.stride(0), cur_stream(), &launched);
+  return launched;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("skinny_gemm", &skinny_gemm);
   m.def("moe_forward", &moe_forward);
   m.def("mla_paged_attention_decode", &mla_paged_attention_decode);
   m.def("prefill_attention", &prefill_attention);

@@ -1,0 +1,73 @@
+#include "hip/hip_runtime.h"
+// Skinny-M dense GEMM: C[M,N] = X[M,K] . W[N,K]^T (+bias), M <= 256.
+//
+// Decode-step GEMMs are M = batch (64-256) against multi-hundred-MB weight
+// matrices — pure weight streaming. hipBLASLt's tile picks run 2-5x off the
+// HBM roofline at these shapes (measured, profiles/README.md); this kernel
+// streams each W row exactly once: one workgroup per 64-column slab, all M
+// rows accumulated in registers (up to 16 MFMA M-tiles), X fragments served
+// from L1/L2 (X is tiny and shared by every slab).
+
+#include "common.h"
+
+#define SG_THREADS 256
+
+__global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
+    uint16_t* __restrict__ c,        // [M, N] bf16
+    const uint16_t* __restrict__ x,  // [M, K] rows at x_stride
+    const uint16_t* __restrict__ w,  // [N, K]
+    const uint16_t* __restrict__ bias,  // [N] or nullptr
+    const int M, const int N, const int K, const int64_t x_stride,
+    const int64_t c_stride) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+
+  const int n = blockIdx.x * 64 + wid * 16 + l15;  // this lane's W row / C col
+  const uint16_t* wrow = w + (size_t)n * K;
+  const int mtiles = (M + 15) / 16;
+
+  f32x4v acc[16] = {};
+  for (int k = 0; k < K; k += 32) {
+    const bf16x8v bfrag =
+        *reinterpret_cast<const bf16x8v*>(wrow + k + l4 * 8);
+#pragma unroll
+    for (int mt = 0; mt < 16; ++mt) {
+      if (mt >= mtiles) break;
+      const int row = mt * 16 + l15;
+      bf16x8v afrag = {};
+      if (row < M)
+        afrag = *reinterpret_cast<const bf16x8v*>(
+            x + (size_t)row * x_stride + k + l4 * 8);
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[mt],
+                                                        0, 0, 0);
+    }
+  }
+
+  const float b = bias != nullptr ? bf16_bits_to_f32(bias[n]) : 0.f;
+#pragma unroll
+  for (int mt = 0; mt < 16; ++mt) {
+    if (mt >= mtiles) break;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = mt * 16 + l4 * 4 + r;
+      if (row < M)
+        c[(size_t)row * c_stride + n] = f32_to_bf16_bits(acc[mt][r] + b);
+    }
+  }
+}
+
+extern "C" void launch_skinny_gemm(
+    void* c, const void* x, const void* w, const void* bias, int M, int N,
+    int K, int64_t x_stride, int64_t c_stride, hipStream_t stream,
+    bool* launched) {
+  *launched = false;
+  if (M > 256 || N % 64 != 0 || K % 32 != 0) return;
+  dim3 grid(N / 64, 1, 1);
+ hipLaunchKernelGGL(( skinny_gemm_kernel), dim3(grid), dim3(SG_THREADS), 0, stream, 
+      (uint16_t*)c, (const uint16_t*)x, (const uint16_t*)w,
+      (const uint16_t*)bias, M, N, K, x_stride, c_stride);
+  *launched = true;
+}

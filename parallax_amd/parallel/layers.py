@@ -17,6 +17,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .comm import get_comm
+from .. import ops
 
 
 def _shard(dim_size: int, tp_size: int, tp_rank: int) -> tuple:
@@ -55,7 +56,7 @@ class ColumnParallelLinear(nn.Module):
             self.bias.data.copy_(b[start : start + per])
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight, self.bias)
+        y = ops.linear(x, self.weight, self.bias)
         if self.gather_output and self.tp_size > 1:
             y = get_comm().tp_all_gather(y, dim=-1)
         return y
@@ -102,7 +103,7 @@ class MergedColumnParallelLinear(nn.Module):
             off += per
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight, self.bias)
+        return ops.linear(x, self.weight, self.bias)
 
     def split_output(self, y: torch.Tensor) -> List[torch.Tensor]:
         return list(torch.split(y, self.shard_sizes, dim=-1))
@@ -137,7 +138,7 @@ class RowParallelLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         comm = get_comm()
-        y = F.linear(x, self.weight)
+        y = ops.linear(x, self.weight)
         y = comm.tp_all_reduce(y)
         # after the all-reduce every rank holds the full sum; bias is added once
         if self.bias is not None:

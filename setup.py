@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "mla_attention.hip"),
     os.path.join(CSRC, "prefill_attention.hip"),
     os.path.join(CSRC, "moe.hip"),
+    os.path.join(CSRC, "skinny_gemm.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 

@@ -315,3 +315,31 @@ def test_engine_fp8_kv_end_to_end():
         [SamplingParams(temperature=0.0, max_new_tokens=5, ignore_eos=True)] * 2,
     )
     assert all(len(v) == 5 for v in out.values())
+
+
+@pytest.mark.parametrize("M,N,K,bias", [
+    (64, 4096, 4096, False),
+    (256, 28672, 4096, False),
+    (1, 4096, 14336, False),
+    (77, 6144, 4096, True),
+    (128, 128256, 4096, False),  # lm_head shape
+])
+def test_skinny_gemm(M, N, K, bias):
+    torch.manual_seed(13)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.2
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.02
+    b = torch.randn(N, dtype=torch.bfloat16, device="cuda") if bias else None
+    out = ops.linear(x, w, b)
+    expect = torch.nn.functional.linear(x.float(), w.float(),
+                                        b.float() if bias else None)
+    torch.testing.assert_close(out.float(), expect.cuda(), atol=8e-2, rtol=8e-2)
+
+
+def test_skinny_gemm_strided_rows():
+    torch.manual_seed(14)
+    buf = torch.randn(32, 512, dtype=torch.bfloat16, device="cuda") * 0.2
+    x = buf[:, :256]  # row stride 512
+    w = torch.randn(128, 256, dtype=torch.bfloat16, device="cuda") * 0.05
+    out = ops.linear(x, w)
+    expect = torch.nn.functional.linear(x.float(), w.float())
+    torch.testing.assert_close(out.float(), expect, atol=8e-2, rtol=8e-2)
