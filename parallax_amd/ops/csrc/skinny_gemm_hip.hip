@@ -1,19 +1,22 @@
 #include "hip/hip_runtime.h"
 // Skinny-M dense GEMM: C[M,N] = X[M,K] . W[N,K]^T (+bias), M <= 256.
 //
-// Decode-step GEMMs are M = batch (64-256) against multi-hundred-MB weight
+// Decode-step GEMMs are M = batch (1-256) against multi-hundred-MB weight
 // matrices — pure weight streaming. hipBLASLt's tile picks run 2-5x off the
-// HBM roofline at these shapes (measured, profiles/README.md); this kernel
-// streams each W row exactly once: one workgroup per 64-column slab, all M
-// rows accumulated in registers (up to 16 MFMA M-tiles), X fragments served
-// from L1/L2 (X is tiny and shared by every slab).
+// HBM roofline at these shapes (measured, profiles/README.md). Structure:
+// one workgroup per 64-column W slab; the X K-slab (M x 128) is staged in LDS
+// once per iteration (cooperative coalesced loads, padded rows against bank
+// conflicts) and every wave's MFMA A-fragments come from LDS; each W row is
+// streamed exactly once per slab by the lane that owns that output column.
 
 #include "common.h"
 
 #define SG_THREADS 256
+#define SG_BK 128           // K elements staged per iteration
+#define SG_PROW (SG_BK + 8) // padded LDS row (elements)
 
 __global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
-    uint16_t* __restrict__ c,        // [M, N] bf16
+    uint16_t* __restrict__ c,        // [M, N] bf16 rows at c_stride
     const uint16_t* __restrict__ x,  // [M, K] rows at x_stride
     const uint16_t* __restrict__ w,  // [N, K]
     const uint16_t* __restrict__ bias,  // [N] or nullptr
@@ -29,20 +32,36 @@ __global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
   const uint16_t* wrow = w + (size_t)n * K;
   const int mtiles = (M + 15) / 16;
 
+  __shared__ uint16_t Xl[256 * SG_PROW];
+
   f32x4v acc[16] = {};
-  for (int k = 0; k < K; k += 32) {
-    const bf16x8v bfrag =
-        *reinterpret_cast<const bf16x8v*>(wrow + k + l4 * 8);
+  for (int k0 = 0; k0 < K; k0 += SG_BK) {
+    __syncthreads();
+    // stage X[:, k0:k0+128]: (row, 16B chunk) pairs over all threads so small
+    // M still engages the whole workgroup, chunks coalesce within a row
+    for (int idx = tid; idx < M * (SG_BK / 8); idx += SG_THREADS) {
+      const int row = idx / (SG_BK / 8);
+      const int cc = idx % (SG_BK / 8);
+      *reinterpret_cast<int4*>(Xl + row * SG_PROW + cc * 8) =
+          *reinterpret_cast<const int4*>(x + (size_t)row * x_stride + k0 + cc * 8);
+    }
+    __syncthreads();
+
 #pragma unroll
-    for (int mt = 0; mt < 16; ++mt) {
-      if (mt >= mtiles) break;
-      const int row = mt * 16 + l15;
-      bf16x8v afrag = {};
-      if (row < M)
-        afrag = *reinterpret_cast<const bf16x8v*>(
-            x + (size_t)row * x_stride + k + l4 * 8);
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[mt],
-                                                        0, 0, 0);
+    for (int ks = 0; ks < SG_BK / 32; ++ks) {
+      const bf16x8v bfrag = *reinterpret_cast<const bf16x8v*>(
+          wrow + k0 + ks * 32 + l4 * 8);
+#pragma unroll
+      for (int mt = 0; mt < 16; ++mt) {
+        if (mt >= mtiles) break;
+        const int row = mt * 16 + l15;
+        bf16x8v afrag = {};
+        if (row < M)
+          afrag = *reinterpret_cast<const bf16x8v*>(
+              Xl + row * SG_PROW + ks * 32 + l4 * 8);
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                          acc[mt], 0, 0, 0);
+      }
     }
   }
 
@@ -64,7 +83,7 @@ extern "C" void launch_skinny_gemm(
     int K, int64_t x_stride, int64_t c_stride, hipStream_t stream,
     bool* launched) {
   *launched = false;
-  if (M > 256 || N % 64 != 0 || K % 32 != 0) return;
+  if (M > 256 || N % 64 != 0 || K % SG_BK != 0) return;
   dim3 grid(N / 64, 1, 1);
  hipLaunchKernelGGL(( skinny_gemm_kernel), dim3(grid), dim3(SG_THREADS), 0, stream, 
       (uint16_t*)c, (const uint16_t*)x, (const uint16_t*)w,
