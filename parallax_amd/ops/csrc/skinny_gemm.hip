@@ -14,6 +14,7 @@
 #define SG_BK 128           // K elements staged per iteration
 #define SG_PROW (SG_BK + 8) // padded LDS row (elements)
 
+template <int MTILES>
 __global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
     uint16_t* __restrict__ c,        // [M, N] bf16 rows at c_stride
     const uint16_t* __restrict__ x,  // [M, K] rows at x_stride
@@ -29,20 +30,23 @@ __global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
 
   const int n = blockIdx.x * 64 + wid * 16 + l15;  // this lane's W row / C col
   const uint16_t* wrow = w + (size_t)n * K;
-  const int mtiles = (M + 15) / 16;
 
-  __shared__ uint16_t Xl[256 * SG_PROW];
+  __shared__ uint16_t Xl[MTILES * 16 * SG_PROW];
 
-  f32x4v acc[16] = {};
+  f32x4v acc[MTILES] = {};
   for (int k0 = 0; k0 < K; k0 += SG_BK) {
     __syncthreads();
     // stage X[:, k0:k0+128]: (row, 16B chunk) pairs over all threads so small
-    // M still engages the whole workgroup, chunks coalesce within a row
-    for (int idx = tid; idx < M * (SG_BK / 8); idx += SG_THREADS) {
+    // M still engages the whole workgroup, chunks coalesce within a row.
+    // Rows past M are zero-filled (read by padded MFMA rows).
+    for (int idx = tid; idx < MTILES * 16 * (SG_BK / 8); idx += SG_THREADS) {
       const int row = idx / (SG_BK / 8);
       const int cc = idx % (SG_BK / 8);
-      *reinterpret_cast<int4*>(Xl + row * SG_PROW + cc * 8) =
-          *reinterpret_cast<const int4*>(x + (size_t)row * x_stride + k0 + cc * 8);
+      int4 val = make_int4(0, 0, 0, 0);
+      if (row < M)
+        val = *reinterpret_cast<const int4*>(
+            x + (size_t)row * x_stride + k0 + cc * 8);
+      *reinterpret_cast<int4*>(Xl + row * SG_PROW + cc * 8) = val;
     }
     __syncthreads();
 
@@ -51,13 +55,12 @@ __global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
       const bf16x8v bfrag = *reinterpret_cast<const bf16x8v*>(
           wrow + k0 + ks * 32 + l4 * 8);
 #pragma unroll
-      for (int mt = 0; mt < 16; ++mt) {
-        if (mt >= mtiles) break;
+      for (int mt = 0; mt < MTILES; ++mt) {
         const int row = mt * 16 + l15;
-        bf16x8v afrag = {};
-        if (row < M)
-          afrag = *reinterpret_cast<const bf16x8v*>(
-              Xl + row * SG_PROW + ks * 32 + l4 * 8);
+        // stale LDS rows past M contribute to padded acc rows only, which are
+        // never stored — but keep them zeroed for NaN safety
+        const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(
+            Xl + row * SG_PROW + ks * 32 + l4 * 8);
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                           acc[mt], 0, 0, 0);
       }
@@ -66,8 +69,7 @@ __global__ __launch_bounds__(SG_THREADS) void skinny_gemm_kernel(
 
   const float b = bias != nullptr ? bf16_bits_to_f32(bias[n]) : 0.f;
 #pragma unroll
-  for (int mt = 0; mt < 16; ++mt) {
-    if (mt >= mtiles) break;
+  for (int mt = 0; mt < MTILES; ++mt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = mt * 16 + l4 * 4 + r;
@@ -84,8 +86,15 @@ extern "C" void launch_skinny_gemm(
   *launched = false;
   if (M > 256 || N % 64 != 0 || K % SG_BK != 0) return;
   dim3 grid(N / 64, 1, 1);
-  skinny_gemm_kernel<<<grid, SG_THREADS, 0, stream>>>(
-      (uint16_t*)c, (const uint16_t*)x, (const uint16_t*)w,
-      (const uint16_t*)bias, M, N, K, x_stride, c_stride);
+#define SG_LAUNCH(MT)                                                      \
+  skinny_gemm_kernel<MT><<<grid, SG_THREADS, 0, stream>>>(                 \
+      (uint16_t*)c, (const uint16_t*)x, (const uint16_t*)w,                \
+      (const uint16_t*)bias, M, N, K, x_stride, c_stride);                 \
   *launched = true;
+  if (M <= 16) { SG_LAUNCH(1) }
+  else if (M <= 32) { SG_LAUNCH(2) }
+  else if (M <= 64) { SG_LAUNCH(4) }
+  else if (M <= 128) { SG_LAUNCH(8) }
+  else { SG_LAUNCH(16) }
+#undef SG_LAUNCH
 }
