@@ -66,6 +66,13 @@ class ModelConfig:
     # -- attention sinks (gpt-oss) --------------------------------------------------
     attention_sinks: bool = False
 
+    # -- hybrid linear attention (qwen3-next gated deltanet) --------------------------
+    linear_num_key_heads: int = 0
+    linear_num_value_heads: int = 0
+    linear_key_head_dim: int = 0
+    linear_value_head_dim: int = 0
+    linear_conv_kernel_dim: int = 4
+
     raw: Dict[str, Any] = field(default_factory=dict, repr=False)
 
     @property
@@ -80,8 +87,12 @@ class ModelConfig:
     def is_mla(self) -> bool:
         return self.kv_lora_rank > 0
 
+    @property
+    def has_linear_layers(self) -> bool:
+        return bool(self.layer_types) and "linear_attention" in self.layer_types
+
     def layer_type(self, layer_idx: int) -> str:
-        """'attention' | 'sliding_attention' | 'mla_attention' | 'linear'."""
+        """'attention' | 'sliding_attention' | 'linear_attention' | 'mla_attention'."""
         if self.is_mla:
             return "mla_attention"
         if self.layer_types is not None:
@@ -132,7 +143,7 @@ class ModelConfig:
             and cfg.get("model_type") == "gpt_oss",
             attention_sinks=cfg.get("model_type") == "gpt_oss",
             qk_norm=cfg.get("use_qk_norm", False)
-            or cfg.get("model_type", "") in ("qwen3", "qwen3_moe"),
+            or cfg.get("model_type", "") in ("qwen3", "qwen3_moe", "qwen3_next"),
             sliding_window=sliding,
             layer_types=layer_types,
             eos_token_ids=[e for e in eos_ids if e is not None],
@@ -163,6 +174,12 @@ class ModelConfig:
             qk_nope_head_dim=cfg.get("qk_nope_head_dim") or 0,
             qk_rope_head_dim=cfg.get("qk_rope_head_dim") or 0,
             v_head_dim=cfg.get("v_head_dim") or 0,
+            # hybrid linear attention
+            linear_num_key_heads=cfg.get("linear_num_key_heads") or 0,
+            linear_num_value_heads=cfg.get("linear_num_value_heads") or 0,
+            linear_key_head_dim=cfg.get("linear_key_head_dim") or 0,
+            linear_value_head_dim=cfg.get("linear_value_head_dim") or 0,
+            linear_conv_kernel_dim=cfg.get("linear_conv_kernel_dim") or 4,
             raw=cfg,
         )
         return mc

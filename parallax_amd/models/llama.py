@@ -161,10 +161,7 @@ class LlamaShardModel(nn.Module):
         if self.is_first:
             self.embed_tokens = VocabEmbedding(cfg.vocab_size, cfg.hidden_size)
         self.layers = nn.ModuleList(
-            [
-                self.decoder_layer_cls(cfg, g, i)
-                for i, g in enumerate(range(start_layer, self.end_layer))
-            ]
+            self._build_layers(cfg, start_layer, self.end_layer)
         )
         if self.is_last:
             self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
@@ -175,6 +172,12 @@ class LlamaShardModel(nn.Module):
             build_rope_cache_for(cfg, rot_dim=cfg.rot_dim),
             persistent=False,
         )
+
+    def _build_layers(self, cfg: ModelConfig, start: int, end: int):
+        return [
+            self.decoder_layer_cls(cfg, g, i)
+            for i, g in enumerate(range(start, end))
+        ]
 
     @property
     def num_local_layers(self) -> int:

@@ -116,6 +116,32 @@ class Engine:
         self.model.rope_cache = self.model.rope_cache.float()
         self.model.eval()
 
+        # hybrid stacks: paged KV only for the full-attention layers
+        self.local_layer_types = [cfg.layer_type(g) for g in range(start, end)]
+        n_kv_layers = sum(
+            1 for t in self.local_layer_types if t != "linear_attention"
+        ) or 1
+        self.linear_cache = None
+        if cfg.has_linear_layers:
+            from .kv_cache import LinearStateCache
+
+            n_linear = sum(1 for t in self.local_layer_types if t == "linear_attention")
+            conv_dim = (
+                2 * cfg.linear_num_key_heads * cfg.linear_key_head_dim
+                + cfg.linear_num_value_heads * cfg.linear_value_head_dim
+            )
+            self.linear_cache = LinearStateCache(
+                max(1, n_linear),
+                conv_state_shape=(conv_dim, cfg.linear_conv_kernel_dim - 1),
+                recurrent_state_shape=(
+                    cfg.linear_num_value_heads, cfg.linear_key_head_dim,
+                    cfg.linear_value_head_dim,
+                ),
+                num_slots=args.max_batch_size + 8,
+                device=self.device,
+                dtype=args.dtype if args.dtype != torch.float32 else torch.float32,
+            )
+
         self.is_mla = cfg.is_mla
         if self.is_mla:
             mla_block_bytes = MLAKVCache.bytes_per_block(
@@ -142,7 +168,7 @@ class Engine:
                 assert self.device.type == "cuda", "fp8 KV needs the HIP kernels"
                 kv_dtype = torch.float8_e4m3fn
             spec = KVCacheSpec(
-                num_layers=end - start,
+                num_layers=n_kv_layers,
                 num_kv_heads=max(1, cfg.num_kv_heads // self.comm.tp_size),
                 head_dim=cfg.head_dim,
                 block_size=args.block_size,
@@ -153,7 +179,9 @@ class Engine:
             )
             self.kv_cache = PagedKVCache(spec, num_blocks, self.device)
         self.cache_manager = CacheManager(
-            args.block_size, num_blocks, enable_prefix_cache=args.enable_prefix_cache
+            args.block_size, num_blocks,
+            enable_prefix_cache=args.enable_prefix_cache and not cfg.has_linear_layers,
+            num_linear_slots=(args.max_batch_size + 8) if cfg.has_linear_layers else 0,
         )
         self.scheduler = Scheduler(
             self.cache_manager,
@@ -320,6 +348,11 @@ class Engine:
             query_lens=torch.tensor(query_lens, dtype=torch.int32, device=dev),
             kv_cache=None if self.is_mla else self.kv_cache,
             mla_cache=self.kv_cache if self.is_mla else None,
+            linear_cache=self.linear_cache,
+            linear_slots=torch.tensor(
+                [self.cache_manager.get(c.req.rid).linear_slot or 0 for c in chunks],
+                dtype=torch.int64, device=dev,
+            ) if self.linear_cache is not None else None,
             logits_indices=torch.tensor(logits_idx, dtype=torch.int64, device=dev),
             max_seq_len=max(seq_lens),
         )
@@ -347,6 +380,11 @@ class Engine:
             seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
             kv_cache=None if self.is_mla else self.kv_cache,
             mla_cache=self.kv_cache if self.is_mla else None,
+            linear_cache=self.linear_cache,
+            linear_slots=torch.tensor(
+                [self.cache_manager.get(r.rid).linear_slot or 0 for r in reqs],
+                dtype=torch.int64, device=dev,
+            ) if self.linear_cache is not None else None,
             logits_indices=torch.arange(len(reqs), dtype=torch.int64, device=dev),
             max_seq_len=max(seq_lens),
         )
