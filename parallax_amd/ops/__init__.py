@@ -306,7 +306,10 @@ def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
 
 # -- dense linear (skinny-M fast path) ------------------------------------------
 
-SKINNY_GEMM_MAX_M = int(os.environ.get("PARALLAX_SKINNY_GEMM_MAX_M", "256"))
+# Disabled by default: hipBLASLt reaches 3.4-5.8 TB/s of weight streaming at
+# decode shapes when measured in isolation (scripts/bench_gemm.py) and beats
+# this kernel's simple 2-barrier structure; opt in for experiments.
+SKINNY_GEMM_MAX_M = int(os.environ.get("PARALLAX_SKINNY_GEMM_MAX_M", "0"))
 
 
 def linear(
