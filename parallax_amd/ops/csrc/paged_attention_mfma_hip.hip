@@ -105,64 +105,78 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   const int kt_begin = tok_begin / DM_KTILE;
   const int kt_end = (tok_end + DM_KTILE - 1) / DM_KTILE;
 
+  // register double-buffered staging (async-STAGE split): loads for tile n+1
+  // are issued right after tile n's LDS write so HBM latency hides under the
+  // MFMA/softmax work (guide §6 G15)
+  const int stg_tok = tid & 63;
+  const int stg_dv = tid >> 6;
+  const int stg_d0 = stg_dv * (HEAD_DIM / 4);
+  int4 kreg[HEAD_DIM / 32], vreg[HEAD_DIM / 32];
+
+  auto load_tile = [&](int kt) {
+    const int gtok = kt * DM_KTILE + stg_tok;
+    const bool ok = gtok >= tok_begin && gtok < tok_end;
+    size_t row_off = 0;
+    if (ok) {
+      const int blk = btab[gtok / BLOCK_SIZE];
+      const int off = gtok % BLOCK_SIZE;
+      row_off = (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM;
+    }
+#pragma unroll
+    for (int c = 0; c < HEAD_DIM / 32; ++c) {
+      const int d = stg_d0 + c * 8;
+      int4 kval = make_int4(0, 0, 0, 0);
+      int4 vval = make_int4(0, 0, 0, 0);
+      if (ok) {
+        if (KV_FP8) {
+          // 8 fp8 bytes -> 8 bf16 (scale folded in)
+          const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
+              (const uint8_t*)k_cache_v + row_off + d);
+          const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
+              (const uint8_t*)v_cache_v + row_off + d);
+          uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
+          uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            ks[j] = f32_to_bf16_bits(
+                fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+            vsp[j] = f32_to_bf16_bits(
+                fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
+          }
+        } else {
+          kval = *reinterpret_cast<const int4*>(
+              (const uint16_t*)k_cache_v + row_off + d);
+          vval = *reinterpret_cast<const int4*>(
+              (const uint16_t*)v_cache_v + row_off + d);
+        }
+      }
+      kreg[c] = kval;
+      vreg[c] = vval;
+    }
+  };
+
+  load_tile(kt_begin);
+
   for (int kt = kt_begin; kt < kt_end; ++kt) {
     const int kbase = kt * DM_KTILE;
-    __syncthreads();
+    __syncthreads();  // previous tile's MFMA done reading LDS
 
-    // ---- stage K (row-major swz) + V (transposed swz) --------------------------
-    {
-      const int tok = tid & 63;
-      const int dv = tid >> 6;
-      const int d0 = dv * (HEAD_DIM / 4);
-      const int gtok = kbase + tok;
-      const bool ok = gtok >= tok_begin && gtok < tok_end;
-      size_t row_off = 0;
-      if (ok) {
-        const int blk = btab[gtok / BLOCK_SIZE];
-        const int off = gtok % BLOCK_SIZE;
-        row_off = (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM;
-      }
+    // ---- write the prefetched tile: K row-major swz, V transposed ---------------
 #pragma unroll
-      for (int c = 0; c < HEAD_DIM / 32; ++c) {
-        const int d = d0 + c * 8;
-        int4 kval = make_int4(0, 0, 0, 0);
-        int4 vval = make_int4(0, 0, 0, 0);
-        if (ok) {
-          if (KV_FP8) {
-            // 8 fp8 bytes -> 8 bf16 (scale folded in)
-            const uint8_t* k8 = (const uint8_t*)k_cache_v + row_off + d;
-            const uint8_t* v8 = (const uint8_t*)v_cache_v + row_off + d;
-            uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
-            uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
-            const uint64_t kraw = *reinterpret_cast<const uint64_t*>(k8);
-            const uint64_t vraw = *reinterpret_cast<const uint64_t*>(v8);
+    for (int c = 0; c < HEAD_DIM / 32; ++c) {
+      const int d = stg_d0 + c * 8;
+      const int kb = swz(stg_tok * HEAD_DIM * 2 + d * 2, stg_tok);
+      *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kreg[c];
+      const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vreg[c]);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              ks[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
-              vsp[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
-            }
-          } else {
-            kval = *reinterpret_cast<const int4*>(
-                (const uint16_t*)k_cache_v + row_off + d);
-            vval = *reinterpret_cast<const int4*>(
-                (const uint16_t*)v_cache_v + row_off + d);
-          }
-        }
-        const int kb = swz(tok * HEAD_DIM * 2 + d * 2, tok);
-        *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kval;
-
-        const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vval);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int dd = d + j;
-          const int vb = swz(dd * DM_KTILE * 2 + tok * 2, dd);
-          *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(VTl) + vb) = vs[j];
-        }
+      for (int j = 0; j < 8; ++j) {
+        const int dd = d + j;
+        const int vb = swz(dd * DM_KTILE * 2 + stg_tok * 2, dd);
+        *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(VTl) + vb) = vs[j];
       }
     }
     __syncthreads();
+    if (kt + 1 < kt_end) load_tile(kt + 1);  // in flight during the math below
 
     // ---- S^T = K . Q^T (wave w: k rows [16w, 16w+16)) ---------------------------
     f32x4v acc_s = {};
