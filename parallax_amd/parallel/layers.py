@@ -21,7 +21,11 @@ from .. import ops
 
 
 def _shard(dim_size: int, tp_size: int, tp_rank: int) -> tuple:
-    assert dim_size % tp_size == 0, f"{dim_size} not divisible by tp={tp_size}"
+    """(start, size) of this rank's shard; dims that do not divide evenly are
+    REPLICATED (e.g. odd vocab sizes in the lm_head) — the layer then behaves
+    as tp_size == 1 for that tensor."""
+    if dim_size % tp_size != 0:
+        return 0, dim_size
     per = dim_size // tp_size
     return per * tp_rank, per
 
@@ -33,9 +37,10 @@ class ColumnParallelLinear(nn.Module):
                  gather_output: bool = False):
         super().__init__()
         comm = get_comm()
-        self.tp_size = comm.tp_size
         self.full_out_features = out_features
         _, self.out_per_rank = _shard(out_features, comm.tp_size, comm.tp_rank)
+        self.replicated = self.out_per_rank == out_features and comm.tp_size > 1
+        self.tp_size = 1 if self.replicated else comm.tp_size
         self.weight = nn.Parameter(
             torch.empty(self.out_per_rank, in_features), requires_grad=False
         )
@@ -115,9 +120,10 @@ class RowParallelLinear(nn.Module):
     def __init__(self, in_features: int, out_features: int, bias: bool = False):
         super().__init__()
         comm = get_comm()
-        self.tp_size = comm.tp_size
         self.full_in_features = in_features
         _, self.in_per_rank = _shard(in_features, comm.tp_size, comm.tp_rank)
+        self.replicated = self.in_per_rank == in_features and comm.tp_size > 1
+        self.tp_size = 1 if self.replicated else comm.tp_size
         self.weight = nn.Parameter(
             torch.empty(out_features, self.in_per_rank), requires_grad=False
         )
@@ -139,7 +145,8 @@ class RowParallelLinear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         comm = get_comm()
         y = ops.linear(x, self.weight)
-        y = comm.tp_all_reduce(y)
+        if not self.replicated:
+            y = comm.tp_all_reduce(y)
         # after the all-reduce every rank holds the full sum; bias is added once
         if self.bias is not None:
             y = y + self.bias
