@@ -273,12 +273,73 @@ def mla_paged_attention_decode(
         ext.mla_paged_attention_decode(
             out, q_latent.contiguous(), q_rope.contiguous(), cache,
             block_tables.to(torch.int32), seq_lens.to(torch.int32), scale,
-            max_seq_len,
+            max_seq_len, q_latent.new_empty(0, dtype=torch.int32),
         )
         return out
     return ref.mla_paged_attention_decode(
         q_latent, q_rope, cache, block_tables, seq_lens, scale
     )
+
+
+def dsa_paged_attention_decode(
+    q_latent: torch.Tensor,
+    q_rope: torch.Tensor,
+    cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    topk_indices: torch.Tensor,
+    scale: float,
+    max_seq_len: Optional[int] = None,
+) -> torch.Tensor:
+    """DeepSeek-V3.2 sparse MLA decode: attention restricted to each row's
+    top-k token positions (-1 padded; a row starting -1 falls back dense)."""
+    if q_latent.is_cuda:
+        ext = _require_ext("dsa_paged_attention_decode")
+        out = torch.empty_like(q_latent)
+        if max_seq_len is None:
+            max_seq_len = int(seq_lens.max().item())
+        ext.mla_paged_attention_decode(
+            out, q_latent.contiguous(), q_rope.contiguous(), cache,
+            block_tables.to(torch.int32), seq_lens.to(torch.int32), scale,
+            max(max_seq_len, topk_indices.shape[1]),
+            topk_indices.to(torch.int32).contiguous(),
+        )
+        return out
+    return ref.dsa_paged_attention_decode(
+        q_latent, q_rope, cache, block_tables, seq_lens, topk_indices, scale
+    )
+
+
+def msa_paged_attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    token_positions: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    """MiniMax-style sparse attention over explicit token positions."""
+    if q.is_cuda:
+        ext = _require_ext("msa_paged_attention_decode")
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        ext.msa_paged_attention_decode(
+            out, q, k_cache, v_cache, block_tables.to(torch.int32),
+            seq_lens.to(torch.int32),
+            token_positions.to(torch.int32).contiguous(), scale,
+        )
+        return out
+    return ref.msa_paged_attention_decode(
+        q, k_cache, v_cache, block_tables, seq_lens, token_positions, scale
+    )
+
+
+# indexer / block-score helpers run as torch compositions on both CPU and GPU
+# (GEMV-scale work; the hot sparse gathers above are the HIP kernels)
+dsa_indexer_scores = ref.dsa_indexer_scores
+store_indexer_cache = ref.store_indexer_cache
+msa_block_scores = ref.msa_block_scores
+msa_topk_tokens = ref.msa_topk_tokens
 
 
 # -- activations ----------------------------------------------------------------------

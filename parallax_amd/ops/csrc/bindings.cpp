@@ -43,7 +43,12 @@ void launch_paged_attention_reduce(void*, const float*, const float*, int, int,
 void launch_mla_paged_attention_decode(void*, const void*, const void*,
                                        const void*, const int*, const int*,
                                        int, int, int, int, int, int, float,
-                                       int, int, float*, float*, hipStream_t,
+                                       int, int, float*, float*, const int*,
+                                       int, hipStream_t, bool*);
+void launch_msa_paged_attention_decode(void*, const void*, const void*,
+                                       const void*, const int*, const int*,
+                                       const int*, int, int, int, int, int,
+                                       int, int, int64_t, float, hipStream_t,
                                        bool*);
 void launch_build_moe_tiles(int*, int*, const int*, int, int, hipStream_t);
 void launch_moe_gate_up(void*, const void*, const void*, const int*,
@@ -318,7 +323,7 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
                                 torch::Tensor q_pe, torch::Tensor cache,
                                 torch::Tensor block_tables,
                                 torch::Tensor seq_lens, double scale,
-                                int64_t max_seq_len) {
+                                int64_t max_seq_len, torch::Tensor topk_indices) {
   CHECK_GPU(q_latent);
   CHECK_CONTIG(q_latent);
   CHECK_CONTIG(q_pe);
@@ -350,12 +355,20 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
     acc_ptr = tmp_acc.data_ptr<float>();
     ml_ptr = tmp_ml.data_ptr<float>();
   }
+  const int* idx_ptr = nullptr;
+  int max_topk = 0;
+  if (topk_indices.numel() > 0) {
+    TORCH_CHECK(topk_indices.scalar_type() == at::kInt &&
+                topk_indices.is_contiguous());
+    idx_ptr = topk_indices.data_ptr<int>();
+    max_topk = topk_indices.size(1);
+  }
   bool launched = false;
   launch_mla_paged_attention_decode(
       out.data_ptr(), q_latent.data_ptr(), q_pe.data_ptr(), cache.data_ptr(),
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, H, R, DR, BS,
       max_blocks, (float)scale, num_parts, part_tokens, acc_ptr, ml_ptr,
-      cur_stream(), &launched);
+      idx_ptr, max_topk, cur_stream(), &launched);
   TORCH_CHECK(launched, "no MLA kernel for R=", R, " DR=", DR, " BS=", BS);
   if (num_parts > 1)
     launch_paged_attention_reduce(out.data_ptr(), acc_ptr, ml_ptr, B, H, R,
@@ -419,7 +432,33 @@ bool skinny_gemm(torch::Tensor c, torch::Tensor x, torch::Tensor w,
   return launched;
 }
 
+void msa_paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                                torch::Tensor k_cache, torch::Tensor v_cache,
+                                torch::Tensor block_tables,
+                                torch::Tensor seq_lens,
+                                torch::Tensor token_positions, double scale) {
+  CHECK_GPU(q);
+  CHECK_BF16(q);
+  CHECK_BF16(k_cache);
+  TORCH_CHECK(token_positions.scalar_type() == at::kInt &&
+              token_positions.is_contiguous());
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hk = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  bool launched = false;
+  launch_msa_paged_attention_decode(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+      token_positions.data_ptr<int>(), token_positions.size(1), B, Hq, Hk, D,
+      BS, block_tables.size(1), row_stride(q, "q"), (float)scale, cur_stream(),
+      &launched);
+  TORCH_CHECK(launched, "no MSA kernel for D=", D, " BS=", BS, " G=", Hq / Hk);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("msa_paged_attention_decode", &msa_paged_attention_decode);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("moe_forward", &moe_forward);
   m.def("mla_paged_attention_decode", &mla_paged_attention_decode);

@@ -21,7 +21,7 @@
 #define MLA_KTILE 32
 #define MLA_HBLOCK 32
 
-template <int R, int DR, int BLOCK_SIZE, bool PARTITIONED>
+template <int R, int DR, int BLOCK_SIZE, bool PARTITIONED, bool SPARSE>
 __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
     uint16_t* __restrict__ out,       // [B, H, R]
     float* __restrict__ tmp_acc,      // [B, H, P, R]
@@ -32,15 +32,22 @@ __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     const int max_blocks, const int H, const float scale,
-    const int part_tokens) {
+    const int part_tokens,
+    const int* __restrict__ topk_indices,  // [B, max_topk] (SPARSE only)
+    const int max_topk) {
   constexpr int DK = R + DR;            // 576
   constexpr int KSTEPS = DK / 32;       // 18
   constexpr int PROW = 40;              // padded row length (bank spread)
   const int hb = blockIdx.x;            // head block (32 heads)
   const int seq = blockIdx.y;
   const int L = seq_lens[seq];
+  // DSA sparse mode: iterate the top-k index list; a row starting with -1
+  // falls back to dense (reference dsa_paged_attention semantics)
+  const int* idx_row = SPARSE ? topk_indices + (size_t)seq * max_topk : nullptr;
+  const bool sparse_row = SPARSE && idx_row[0] >= 0;
+  const int domain = sparse_row ? max_topk : L;
 
-  int tok_begin = 0, tok_end = L;
+  int tok_begin = 0, tok_end = domain;
   if (PARTITIONED) {
     const int p = blockIdx.z;
     tok_begin = max(tok_begin, p * part_tokens);
@@ -95,8 +102,13 @@ __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
     {
       const int tok = tid & 31;
       const int dv = tid >> 5;  // 8 slices of 72 dims (9 int4 each)
-      const int gtok = kbase + tok;
-      const bool ok = gtok >= tok_begin && gtok < tok_end;
+      const int j = kbase + tok;
+      int gtok = j;
+      bool ok = j >= tok_begin && j < tok_end;
+      if (sparse_row && ok) {
+        gtok = idx_row[j];
+        ok = gtok >= 0 && gtok < L;
+      }
       const uint16_t* crow = nullptr;
       if (ok) {
         const int blk = btab[gtok / BLOCK_SIZE];
@@ -151,9 +163,13 @@ __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
     float mx = -3.0e4f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int kglob = kbase + mw * 16 + l4 * 4 + r;
+      const int j = kbase + mw * 16 + l4 * 4 + r;
       float s = acc_s[r] * scale;
-      const bool vis = kglob >= tok_begin && kglob < tok_end && head_ok;
+      bool vis = j >= tok_begin && j < tok_end && head_ok;
+      if (sparse_row && vis) {
+        const int gt = idx_row[j];
+        vis = gt >= 0 && gt < L;
+      }
       s = vis ? s : -3.0e4f;
       acc_s[r] = s;
       mx = fmaxf(mx, s);
@@ -258,29 +274,37 @@ extern "C" void launch_mla_paged_attention_decode(
     void* out, const void* q_latent, const void* q_pe, const void* cache,
     const int* block_tables, const int* seq_lens, int B, int H, int R, int DR,
     int BS, int max_blocks, float scale, int num_parts, int part_tokens,
-    float* tmp_acc, float* tmp_ml, hipStream_t stream, bool* launched) {
+    float* tmp_acc, float* tmp_ml, const int* topk_indices, int max_topk,
+    hipStream_t stream, bool* launched) {
   *launched = false;
   if (R != 512 || DR != 64) return;  // DeepSeek V2/V3/R1/K2 geometry
   const int head_blocks = (H + MLA_HBLOCK - 1) / MLA_HBLOCK;
 
-#define MLA_LAUNCH(BSZ)                                                       \
+#define MLA_LAUNCH2(BSZ, SP)                                                  \
   if (num_parts <= 1) {                                                       \
     dim3 grid(head_blocks, B, 1);                                             \
-   hipLaunchKernelGGL(( mla_decode_kernel<512, 64, BSZ, false>), dim3(grid), dim3(MLA_THREADS), 0, stream,  \
+   hipLaunchKernelGGL(( mla_decode_kernel<512, 64, BSZ, false, SP>)                                \
+        , dim3(grid), dim3(MLA_THREADS), 0, stream,                                    \
         (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q_latent,          \
         (const uint16_t*)q_pe, (const uint16_t*)cache, block_tables,          \
-        seq_lens, max_blocks, H, scale, 0);                                   \
+        seq_lens, max_blocks, H, scale, 0, topk_indices, max_topk);           \
   } else {                                                                    \
     dim3 grid(head_blocks, B, num_parts);                                     \
-   hipLaunchKernelGGL(( mla_decode_kernel<512, 64, BSZ, true>), dim3(grid), dim3(MLA_THREADS), 0, stream,   \
+   hipLaunchKernelGGL(( mla_decode_kernel<512, 64, BSZ, true, SP>)                                 \
+        , dim3(grid), dim3(MLA_THREADS), 0, stream,                                    \
         nullptr, tmp_acc, tmp_ml, (const uint16_t*)q_latent,                  \
         (const uint16_t*)q_pe, (const uint16_t*)cache, block_tables,          \
-        seq_lens, max_blocks, H, scale, part_tokens);                         \
+        seq_lens, max_blocks, H, scale, part_tokens, topk_indices, max_topk); \
   }                                                                           \
   *launched = true;
+
+#define MLA_LAUNCH(BSZ)                                                       \
+  if (topk_indices != nullptr) { MLA_LAUNCH2(BSZ, true) }                     \
+  else { MLA_LAUNCH2(BSZ, false) }
 
   if (BS == 32) { MLA_LAUNCH(32) }
   else if (BS == 16) { MLA_LAUNCH(16) }
   else if (BS == 64) { MLA_LAUNCH(64) }
 #undef MLA_LAUNCH
+#undef MLA_LAUNCH2
 }

@@ -15,7 +15,8 @@
 #define DM_THREADS 256
 #define DM_KTILE 64
 
-template <int HEAD_DIM, int BLOCK_SIZE, bool PARTITIONED, bool KV_FP8>
+template <int HEAD_DIM, int BLOCK_SIZE, bool PARTITIONED, bool KV_FP8,
+          bool SPARSE = false>
 __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     uint16_t* __restrict__ out,        // [B, Hq, D] (final mode)
     float* __restrict__ tmp_acc,       // [B, Hq, P, D] (partitioned)
@@ -28,13 +29,19 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     const int max_blocks, const int Hk, const int G, const int64_t q_stride,
     const float scale, const int sliding_window, const float softcap,
     const float* __restrict__ sinks, const int part_tokens,
-    const float k_scale, const float v_scale) {
+    const float k_scale, const float v_scale,
+    const int* __restrict__ token_positions = nullptr,  // [B, max_pos] (SPARSE)
+    const int max_positions = 0) {
   const int seq = blockIdx.y;
   const int hk = blockIdx.x;
   const int L = seq_lens[seq];
+  // MSA sparse mode: iterate an explicit token-position list (-1 padded)
+  const int* pos_row =
+      SPARSE ? token_positions + (size_t)seq * max_positions : nullptr;
+  const int domain = SPARSE ? max_positions : L;
 
-  int tok_begin = 0, tok_end = L;
-  if (sliding_window > 0) tok_begin = max(0, L - sliding_window);
+  int tok_begin = 0, tok_end = domain;
+  if (!SPARSE && sliding_window > 0) tok_begin = max(0, L - sliding_window);
   if (PARTITIONED) {
     const int p = blockIdx.z;
     tok_begin = max(tok_begin, p * part_tokens);
@@ -113,8 +120,13 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   int4 kreg[HEAD_DIM / 32], vreg[HEAD_DIM / 32];
 
   auto load_tile = [&](int kt) {
-    const int gtok = kt * DM_KTILE + stg_tok;
-    const bool ok = gtok >= tok_begin && gtok < tok_end;
+    const int j = kt * DM_KTILE + stg_tok;
+    int gtok = j;
+    bool ok = j >= tok_begin && j < tok_end;
+    if (SPARSE && ok) {
+      gtok = pos_row[j];
+      ok = gtok >= 0 && gtok < L;
+    }
     size_t row_off = 0;
     if (ok) {
       const int blk = btab[gtok / BLOCK_SIZE];
@@ -195,10 +207,14 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     float mx = -3.0e4f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int kglob = kbase + 16 * wid + l4 * 4 + r;
+      const int j = kbase + 16 * wid + l4 * 4 + r;
       float s = acc_s[r] * scale;
       if (softcap > 0.f) s = softcap * tanhf(s / softcap);
-      const bool visible = kglob >= tok_begin && kglob < tok_end && l15 < G;
+      bool visible = j >= tok_begin && j < tok_end && l15 < G;
+      if (SPARSE && visible) {
+        const int gt = pos_row[j];
+        visible = gt >= 0 && gt < L;
+      }
       s = visible ? s : -3.0e4f;
       acc_s[r] = s;
       mx = fmaxf(mx, s);
@@ -301,6 +317,31 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
       }
     }
   }
+}
+
+extern "C" void launch_msa_paged_attention_decode(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    const int* block_tables, const int* seq_lens, const int* token_positions,
+    int max_positions, int B, int Hq, int Hk, int D, int BS, int max_blocks,
+    int64_t q_stride, float scale, hipStream_t stream, bool* launched) {
+  const int G = Hq / Hk;
+  *launched = false;
+  if (G > 16) return;
+  dim3 grid(Hk, B, 1);
+#define MSA_LAUNCH(HD, BSZ)                                                    \
+  paged_decode_mfma_kernel<HD, BSZ, false, false, true>                        \
+      <<<grid, DM_THREADS, 0, stream>>>(                                       \
+          (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q, k_cache,       \
+          v_cache, block_tables, seq_lens, max_blocks, Hk, G, q_stride, scale, \
+          -1, 0.f, nullptr, 0, 1.f, 1.f, token_positions, max_positions);      \
+  *launched = true;
+  if (D == 128 && BS == 32) { MSA_LAUNCH(128, 32) }
+  else if (D == 128 && BS == 16) { MSA_LAUNCH(128, 16) }
+  else if (D == 128 && BS == 64) { MSA_LAUNCH(128, 64) }
+  else if (D == 64 && BS == 32) { MSA_LAUNCH(64, 32) }
+  else if (D == 64 && BS == 16) { MSA_LAUNCH(64, 16) }
+  else if (D == 64 && BS == 64) { MSA_LAUNCH(64, 64) }
+#undef MSA_LAUNCH
 }
 
 extern "C" void launch_paged_decode_mfma(

@@ -362,3 +362,164 @@ def apply_penalties(
             counts = torch.bincount(ids, minlength=logits.shape[1]).to(logits.dtype)
             logits[i] -= pp * (counts > 0).to(logits.dtype) + fp * counts
     return logits
+
+
+# -- sparse attention (DSA: DeepSeek V3.2 top-k tokens; MSA: MiniMax block top-k) --
+
+
+def dsa_paged_attention_decode(
+    q_latent: torch.Tensor,      # [B, Hq, lora_rank]
+    q_rope: torch.Tensor,        # [B, Hq, rope_dim]
+    cache: torch.Tensor,         # [num_blocks, bs, lora_rank + rope_dim]
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    topk_indices: torch.Tensor,  # [B, index_topk] int; -1 padded; row starting
+                                 # with -1 => dense fallback (reference ops.py:182)
+    scale: float,
+) -> torch.Tensor:
+    B, Hq, R = q_latent.shape
+    bs = cache.shape[1]
+    out = torch.empty_like(q_latent)
+    for i in range(B):
+        L = int(seq_lens[i])
+        nb = (L + bs - 1) // bs
+        entries = cache[block_tables[i, :nb].long()].reshape(nb * bs, -1)[:L].float()
+        idx = topk_indices[i]
+        if int(idx[0]) >= 0:
+            keep = idx[(idx >= 0) & (idx < L)].long()
+            entries = entries[keep]
+        latent, rope = entries[:, :R], entries[:, R:]
+        logits = (
+            q_latent[i].float() @ latent.T + q_rope[i].float() @ rope.T
+        ) * scale
+        p = torch.softmax(logits, dim=-1)
+        out[i] = (p @ latent).to(q_latent.dtype)
+    return out
+
+
+def dsa_indexer_scores(
+    q_index: torch.Tensor,       # [B, index_heads, index_dim]
+    index_cache: torch.Tensor,   # [num_blocks, bs, index_heads, index_dim]
+    head_weights: torch.Tensor,  # [B, index_heads] per-head score weights
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+) -> torch.Tensor:
+    """Weighted q.k scores over the full context for top-k selection
+    (reference dsa_indexer, ops.py:325-367). Returns [B, max_ctx] with -inf
+    past each row's length; top-k itself is torch.topk on the result."""
+    B, Hi, Di = q_index.shape
+    bs = index_cache.shape[1]
+    max_ctx = int(seq_lens.max())
+    scores = torch.full((B, max_ctx), float("-inf"), dtype=torch.float32)
+    for i in range(B):
+        L = int(seq_lens[i])
+        nb = (L + bs - 1) // bs
+        keys = index_cache[block_tables[i, :nb].long()].reshape(nb * bs, Hi, Di)[:L]
+        s = torch.einsum("hd,lhd->hl", q_index[i].float(), keys.float())
+        s = torch.relu(s)  # per DeepSeek-V3.2 indexer: ReLU before head-weighting
+        scores[i, :L] = torch.einsum("h,hl->l", head_weights[i].float(), s)
+    return scores
+
+
+def store_indexer_cache(
+    index_keys: torch.Tensor,    # [T, index_heads, index_dim]
+    index_cache: torch.Tensor,   # [num_blocks, bs, index_heads, index_dim]
+    slot_mapping: torch.Tensor,
+) -> None:
+    bs = index_cache.shape[1]
+    valid = slot_mapping >= 0
+    slots = slot_mapping[valid].long()
+    index_cache[slots // bs, slots % bs] = index_keys[valid].to(index_cache.dtype)
+
+
+def msa_block_scores(
+    q: torch.Tensor,             # [B, Hq, D]
+    k_cache: torch.Tensor,       # [num_blocks, Hk, bs, D]
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    sparse_block: int,
+) -> torch.Tensor:
+    """Mean-pooled K per sparse block scored against the mean query
+    (reference msa_indexer phase 1). Returns [B, max_sparse_blocks] (-inf pad)."""
+    B, Hq, D = q.shape
+    Hk = k_cache.shape[1]
+    bs = k_cache.shape[2]
+    max_blocks = (int(seq_lens.max()) + sparse_block - 1) // sparse_block
+    out = torch.full((B, max_blocks), float("-inf"), dtype=torch.float32)
+    qm = q.float().mean(dim=1)  # [B, D]
+    for i in range(B):
+        L = int(seq_lens[i])
+        nb = (L + bs - 1) // bs
+        k = (
+            k_cache[block_tables[i, :nb].long()]
+            .transpose(1, 2).reshape(nb * bs, Hk, D)[:L]
+            .float().mean(dim=1)
+        )  # [L, D]
+        nsb = (L + sparse_block - 1) // sparse_block
+        for sb in range(nsb):
+            seg = k[sb * sparse_block : min((sb + 1) * sparse_block, L)]
+            out[i, sb] = seg.mean(dim=0) @ qm[i]
+    return out
+
+
+def msa_topk_tokens(
+    block_scores: torch.Tensor,  # [B, max_sparse_blocks]
+    seq_lens: torch.Tensor,
+    sparse_block: int,
+    topk_blocks: int,
+    init_blocks: int = 1,
+    local_blocks: int = 2,
+) -> torch.Tensor:
+    """Top-k sparse blocks per sequence expanded to token positions, always
+    keeping the first `init_blocks` and last `local_blocks` (reference
+    msa_block_topk_tokens). Returns [B, max_positions] int64, -1 padded."""
+    B = block_scores.shape[0]
+    rows = []
+    for i in range(B):
+        L = int(seq_lens[i])
+        nsb = (L + sparse_block - 1) // sparse_block
+        keep = set(range(min(init_blocks, nsb)))
+        keep |= set(range(max(0, nsb - local_blocks), nsb))
+        remaining = [b for b in range(nsb) if b not in keep]
+        if remaining and topk_blocks > 0:
+            sc = block_scores[i, remaining]
+            take = min(topk_blocks, len(remaining))
+            top = torch.topk(sc, take).indices.tolist()
+            keep |= {remaining[t] for t in top}
+        pos = []
+        for b in sorted(keep):
+            pos.extend(range(b * sparse_block, min((b + 1) * sparse_block, L)))
+        rows.append(torch.tensor(sorted(pos), dtype=torch.int64))
+    max_p = max(r.numel() for r in rows)
+    out = torch.full((B, max_p), -1, dtype=torch.int64)
+    for i, r in enumerate(rows):
+        out[i, : r.numel()] = r
+    return out
+
+
+def msa_paged_attention_decode(
+    q: torch.Tensor,             # [B, Hq, D]
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    token_positions: torch.Tensor,  # [B, max_positions] int, -1 padded
+    scale: float,
+) -> torch.Tensor:
+    """Exact attention over explicit token positions mapped through block
+    tables (reference msa_paged_attention, ops.py:594-662)."""
+    B, Hq, D = q.shape
+    Hk = k_cache.shape[1]
+    group = Hq // Hk
+    out = torch.empty_like(q)
+    for i in range(B):
+        L = int(seq_lens[i])
+        k, v = _gather_kv(k_cache, v_cache, block_tables[i], L)
+        pos = token_positions[i]
+        keep = pos[(pos >= 0) & (pos < L)].long()
+        kf = k.float()[keep].transpose(0, 1).repeat_interleave(group, dim=0)
+        vf = v.float()[keep].transpose(0, 1).repeat_interleave(group, dim=0)
+        logits = torch.einsum("hd,hld->hl", q[i].float(), kf) * scale
+        p = torch.softmax(logits, dim=-1)
+        out[i] = torch.einsum("hl,hld->hd", p, vf).to(q.dtype)
+    return out
