@@ -93,62 +93,75 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
   const int kv_limit = min(L, qpos0 + QTILE);
   const int kt_end = (kv_limit + KTILE - 1) / KTILE;
 
+  // register double-buffered staging: next tile's loads fly during this
+  // tile's QK/softmax/PV (guide §6 G15 async-STAGE split)
+  const int stg_tok = tid & 63;
+  const int stg_dv = tid >> 6;
+  const int stg_d = stg_dv * (HEAD_DIM / 4);
+  int4 kreg[HEAD_DIM / 32], vreg[HEAD_DIM / 32];
+
+  auto load_tile = [&](int kt) {
+    const int gtok = kt * KTILE + stg_tok;
+    const bool ok = gtok < L;
+    size_t row_off = 0;
+    if (ok) {
+      const int blk = btab[gtok / BS];
+      const int off = gtok % BS;
+      row_off = (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
+    }
+#pragma unroll
+    for (int c = 0; c < HEAD_DIM / 32; ++c) {  // 8 elems per step
+      int4 kval = make_int4(0, 0, 0, 0);
+      int4 vval = make_int4(0, 0, 0, 0);
+      if (ok) {
+        if (KV_FP8) {
+          const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
+              (const uint8_t*)k_cache_v + row_off + stg_d + c * 8);
+          const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
+              (const uint8_t*)v_cache_v + row_off + stg_d + c * 8);
+          uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
+          uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            ks[j] = f32_to_bf16_bits(
+                fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+            vsp[j] = f32_to_bf16_bits(
+                fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
+          }
+        } else {
+          kval = *reinterpret_cast<const int4*>(
+              (const uint16_t*)k_cache_v + row_off + stg_d + c * 8);
+          vval = *reinterpret_cast<const int4*>(
+              (const uint16_t*)v_cache_v + row_off + stg_d + c * 8);
+        }
+      }
+      kreg[c] = kval;
+      vreg[c] = vval;
+    }
+  };
+
+  if (kt_begin < kt_end) load_tile(kt_begin);
+
   for (int kt = kt_begin; kt < kt_end; ++kt) {
     const int kbase = kt * KTILE;
     __syncthreads();  // previous PV finished reading VTl/Pl
 
-    // ---- stage K row-major (swz) and V transposed (swz) -----------------------
-    {
-      const int tok = tid & 63;
-      const int dv = tid >> 6;           // 4 chunks of 32 dims
-      const int d = dv * (HEAD_DIM / 4);
-      const int gtok = kbase + tok;
-      const bool ok = gtok < L;
-      size_t row_off = 0;
-      if (ok) {
-        const int blk = btab[gtok / BS];
-        const int off = gtok % BS;
-        row_off = (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
-      }
+    // ---- write the prefetched tile: K row-major swz, V transposed --------------
 #pragma unroll
-      for (int c = 0; c < HEAD_DIM / 32; ++c) {  // 8 elems per step
-        int4 kval = make_int4(0, 0, 0, 0);
-        int4 vval = make_int4(0, 0, 0, 0);
-        if (ok) {
-          if (KV_FP8) {
-            const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
-                (const uint8_t*)k_cache_v + row_off + d + c * 8);
-            const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
-                (const uint8_t*)v_cache_v + row_off + d + c * 8);
-            uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
-            uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
+    for (int c = 0; c < HEAD_DIM / 32; ++c) {
+      const int d = stg_d + c * 8;
+      const int kb = swz(stg_tok * HEAD_DIM * 2 + d * 2, stg_tok);
+      *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kreg[c];
+      const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vreg[c]);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              ks[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
-              vsp[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
-            }
-          } else {
-            kval = *reinterpret_cast<const int4*>(
-                (const uint16_t*)k_cache_v + row_off + d + c * 8);
-            vval = *reinterpret_cast<const int4*>(
-                (const uint16_t*)v_cache_v + row_off + d + c * 8);
-          }
-        }
-        const int kb = swz(tok * HEAD_DIM * 2 + (d + c * 8) * 2, tok);
-        *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kval;
-
-        const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vval);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int dd = d + c * 8 + j;
-          const int vb = swz(dd * KTILE * 2 + tok * 2, dd);
-          *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(VTl) + vb) = vs[j];
-        }
+      for (int j = 0; j < 8; ++j) {
+        const int dd = d + j;
+        const int vb = swz(dd * KTILE * 2 + stg_tok * 2, dd);
+        *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(VTl) + vb) = vs[j];
       }
     }
     __syncthreads();
+    if (kt + 1 < kt_end) load_tile(kt + 1);
 
     // ---- S^T = K . Q^T ---------------------------------------------------------
     // wave w covers k rows [16w, 16w+16); acc col = q, row = k
