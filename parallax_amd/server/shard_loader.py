@@ -33,8 +33,10 @@ def list_safetensors_files(model_path: str) -> List[str]:
     return files, None
 
 
-def load_shard_weights(model, model_path: str) -> int:
-    """Load the weights this shard needs. Returns the number of tensors loaded."""
+def load_shard_weights(model, model_path: str, lora_path: str = None,
+                       lora_scale: float = 1.0) -> int:
+    """Load the weights this shard needs (optionally fusing a LoRA adapter).
+    Returns the number of tensors loaded."""
     from safetensors import safe_open
 
     files, weight_map = list_safetensors_files(model_path)
@@ -63,7 +65,93 @@ def load_shard_weights(model, model_path: str) -> int:
         "loaded %d tensors from %d file(s) for layers [%d,%d)",
         loaded, len(needed_files), model.start_layer, model.end_layer,
     )
+    if lora_path:
+        loaded += fuse_lora(model, lora_path, lora_scale)
     return loaded
+
+
+def fuse_lora(model, lora_path: str, scale: float = 1.0) -> int:
+    """Fuse a PEFT-format LoRA adapter into the base weights at load time
+    (reference shard_loader.py:114-228 behavior: W += scale * B @ A). The
+    adapter never exists as separate serving state — fused weights keep the
+    decode hot path unchanged."""
+    import json as _json
+
+    import torch
+    from safetensors import safe_open
+
+    cfg_path = os.path.join(lora_path, "adapter_config.json")
+    if os.path.exists(cfg_path):
+        with open(cfg_path) as f:
+            acfg = _json.load(f)
+        r = acfg.get("r", 8)
+        alpha = acfg.get("lora_alpha", r)
+        scale = scale * alpha / max(1, r)
+
+    adapter_file = None
+    for cand in ("adapter_model.safetensors", "adapter.safetensors"):
+        if os.path.exists(os.path.join(lora_path, cand)):
+            adapter_file = os.path.join(lora_path, cand)
+            break
+    if adapter_file is None:
+        raise FileNotFoundError(f"no adapter safetensors under {lora_path}")
+
+    pairs: Dict[str, Dict[str, "torch.Tensor"]] = {}
+    with safe_open(adapter_file, framework="pt", device="cpu") as f:
+        for name in f.keys():
+            # base_model.model.<hf name>.lora_A.weight / lora_B.weight
+            if ".lora_A." in name:
+                base, kind = name.split(".lora_A."), "A"
+            elif ".lora_B." in name:
+                base, kind = name.split(".lora_B."), "B"
+            else:
+                continue
+            key = base[0].replace("base_model.model.", "")
+            pairs.setdefault(key, {})[kind] = f.get_tensor(name)
+
+    fused = 0
+    for key, ab in pairs.items():
+        if "A" not in ab or "B" not in ab:
+            continue
+        delta = (ab["B"].float() @ ab["A"].float()) * scale
+        # route the DELTA through the model's normal weight router by fetching
+        # the current weight, adding, and re-loading
+        hf_name = key + ".weight"
+        cur = _read_current_weight(model, hf_name, delta.shape)
+        if cur is None:
+            continue
+        if model.load_hf_weight(hf_name, (cur + delta).to(torch.float32)):
+            fused += 1
+    logger.info("fused %d LoRA deltas from %s (scale %.3f)", fused, lora_path, scale)
+    return fused
+
+
+def _read_current_weight(model, hf_name: str, shape):
+    """Reconstruct the full (unsharded) current weight for a fusable target.
+    Works for the llama-family fused projections at tp_size 1 (LoRA fusion with
+    TP sharding re-loads from the base checkpoint first)."""
+    import torch
+
+    local = model.map_global_layer(hf_name)
+    if local is None:
+        return None
+    parts = local.split(".")
+    stem = parts[-2]
+    stacked = getattr(model, "_STACKED", {})
+    try:
+        if stem in stacked:
+            target, idx = stacked[stem]
+            module = model
+            for p in parts[:-2] + [target]:
+                module = module[int(p)] if p.isdigit() else getattr(module, p)
+            off = sum(module.shard_sizes[:idx])
+            return module.weight.data[off : off + shape[0]].float()
+        module = model
+        for p in parts[:-1]:
+            module = module[int(p)] if p.isdigit() else getattr(module, p)
+        return module.weight.data.float()
+    except (AttributeError, IndexError, KeyError):
+        return None
 
 
 def selective_file_list(model_path: str, start_layer: int, end_layer: int) -> List[str]:
