@@ -149,6 +149,35 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
             svc.node_endpoints.pop(body["node_id"], None)
         return {"status": "ok"}
 
+    @app.get("/")
+    async def dashboard():
+        from fastapi.responses import HTMLResponse
+
+        return HTMLResponse("""<!doctype html><html><head>
+<title>parallax_amd cluster</title>
+<style>body{font-family:monospace;margin:2em;background:#111;color:#eee}
+table{border-collapse:collapse}td,th{border:1px solid #444;padding:4px 10px}
+.ok{color:#6f6}.bad{color:#f66}</style></head><body>
+<h2>parallax_amd cluster</h2><div id=s>loading...</div>
+<script>
+async function load(){
+  const r = await fetch('/cluster/status'); const d = await r.json();
+  if(!d.initialized){document.getElementById('s').innerText='scheduler not initialized';return;}
+  let h = `<p>model: <b>${d.model}</b> | bootstrapped: ${d.bootstrapped} | `+
+          `nodes: ${d.num_nodes} | pipelines: ${d.num_pipelines}</p>`;
+  h += '<table><tr><th>node</th><th>hw</th><th>layers</th><th>inflight</th><th>ms/layer</th><th>alive</th></tr>';
+  for(const n of d.nodes){h += `<tr><td>${n.node_id}</td><td>${n.hardware}</td>`+
+    `<td>[${n.start_layer}, ${n.end_layer})</td><td>${n.current_requests}</td>`+
+    `<td>${(n.layer_latency_ms||0).toFixed(3)}</td>`+
+    `<td class="${n.active?'ok':'bad'}">${n.active?'yes':'NO'}</td></tr>`;}
+  h += '</table><p>pipelines:</p><ul>';
+  for(const p of d.pipelines){h += `<li>${p.join(' &rarr; ')}</li>`;}
+  h += '</ul>';
+  document.getElementById('s').innerHTML = h;
+}
+load(); setInterval(load, 3000);
+</script></body></html>""")
+
     @app.get("/cluster/status")
     async def cluster_status():
         if svc.scheduler is None:

@@ -104,14 +104,10 @@ def cmd_serve(args) -> None:
         app = create_app(server, tok, args.model_name or args.model or "model")
         uvicorn.run(app, host=args.host, port=args.port)
     else:
-        # non-head ranks just run the SPMD step loop
-        import time
-
+        # non-head ranks run the SPMD step loop; step() blocks in the ingress
+        # broadcast until rank 0 steps, so this does not spin while idle
         while True:
-            if engine.has_work:
-                engine.step()
-            else:
-                engine.step() if engine.scheduler.has_work else time.sleep(0.002)
+            engine.step()
 
 
 def cmd_chat(args) -> None:
