@@ -191,6 +191,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
 
     // ---- S^T = K . Q^T (wave w: k rows [16w, 16w+16)) ---------------------------
     f32x4v acc_s = {};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < HEAD_DIM / 32; ++s) {
       const int krow_i = 16 * wid + l15;
@@ -202,6 +203,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
           reinterpret_cast<const char*>(Ql) + qb);
       acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc_s, 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + per-column max --------------------------------------------------
     float mx = -3.0e4f;
@@ -262,6 +264,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
 #pragma unroll
         for (int rr = 0; rr < 4; ++rr) acc_o[mt][rr] *= r;
     }
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < DM_KTILE / 32; ++s) {
       const int pb = swz(l15 * DM_KTILE * 2 + s * 64 + l4 * 16, l15);
@@ -277,6 +280,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
                                                             acc_o[mt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- write out (O^T acc: col=q head, row=d) -----------------------------------
