@@ -104,3 +104,30 @@ def test_greedy_parity_chunked(tiny_dsv3):
         [prompt], [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)]
     )
     assert list(out.values())[0] == ref
+
+
+def test_mla_prefix_cache_reuse(tiny_dsv3):
+    """Kimi-K2-style prefix reuse: shared-prefix requests hit the block-radix
+    cache over the compressed MLA latent cache (BASELINE config 5 behavior)."""
+    hf, cfg = tiny_dsv3
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=256,
+                                 dtype=torch.float32))
+    m = get_model_class(cfg.architecture)(cfg).eval()
+    for name, t in hf.state_dict().items():
+        m.load_hf_weight(name, t)
+    m = m.float()
+    m.finalize_weights()
+    eng.model = m
+    shared = list(range(10, 42))  # 32 shared tokens = 4 full blocks
+    sp = SamplingParams(temperature=0.0, max_new_tokens=3, ignore_eos=True)
+    out1 = list(eng.generate([shared + [1, 2]], [sp]).values())[0]
+    hits0 = eng.cache_manager.radix.hit_tokens
+    out2 = list(eng.generate([shared + [3, 4, 5]], [sp]).values())[0]
+    assert eng.cache_manager.radix.hit_tokens > hits0  # prefix actually reused
+    # and reuse does not change the result
+    eng2 = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=256,
+                                  dtype=torch.float32,
+                                  enable_prefix_cache=False))
+    eng2.model = m
+    ref2 = list(eng2.generate([shared + [3, 4, 5]], [sp]).values())[0]
+    assert out2 == ref2
