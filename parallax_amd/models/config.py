@@ -63,6 +63,11 @@ class ModelConfig:
     qk_rope_head_dim: int = 0
     v_head_dim: int = 0
 
+    # -- DSA indexer (deepseek v3.2 sparse attention) ---------------------------
+    index_n_heads: int = 0
+    index_head_dim: int = 0
+    index_topk: int = 0
+
     # -- attention sinks (gpt-oss) --------------------------------------------------
     attention_sinks: bool = False
 
@@ -86,6 +91,10 @@ class ModelConfig:
     @property
     def is_mla(self) -> bool:
         return self.kv_lora_rank > 0
+
+    @property
+    def is_dsa(self) -> bool:
+        return self.index_topk > 0 and self.index_head_dim > 0
 
     @property
     def has_linear_layers(self) -> bool:
@@ -163,8 +172,9 @@ class ModelConfig:
             norm_topk_prob=cfg.get("norm_topk_prob", True),
             routed_scaling_factor=cfg.get("routed_scaling_factor", 1.0),
             scoring_func=cfg.get("scoring_func")
-            or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "kimi_k2",
-                                                       "glm4_moe", "glm4v_moe")
+            or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "deepseek_v32",
+                                                       "kimi_k2", "glm4_moe",
+                                                       "glm4v_moe")
                 else "softmax"),
             topk_group=cfg.get("topk_group", 0),
             n_group=cfg.get("n_group", 0),
@@ -174,6 +184,10 @@ class ModelConfig:
             qk_nope_head_dim=cfg.get("qk_nope_head_dim") or 0,
             qk_rope_head_dim=cfg.get("qk_rope_head_dim") or 0,
             v_head_dim=cfg.get("v_head_dim") or 0,
+            # DSA indexer
+            index_n_heads=cfg.get("index_n_heads") or 0,
+            index_head_dim=cfg.get("index_head_dim") or 0,
+            index_topk=cfg.get("index_topk") or 0,
             # hybrid linear attention
             linear_num_key_heads=cfg.get("linear_num_key_heads") or 0,
             linear_num_value_heads=cfg.get("linear_num_value_heads") or 0,

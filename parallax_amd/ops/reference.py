@@ -399,7 +399,10 @@ def dsa_paged_attention_decode(
 
 def dsa_indexer_scores(
     q_index: torch.Tensor,       # [B, index_heads, index_dim]
-    index_cache: torch.Tensor,   # [num_blocks, bs, index_heads, index_dim]
+    index_cache: torch.Tensor,   # [nb, bs, index_heads, index_dim] or
+                                 # [nb, bs, index_dim] (keys shared across
+                                 # heads, as in DeepSeek-V3.2 where only the
+                                 # indexer query is multi-headed)
     head_weights: torch.Tensor,  # [B, index_heads] per-head score weights
     block_tables: torch.Tensor,
     seq_lens: torch.Tensor,
@@ -409,13 +412,19 @@ def dsa_indexer_scores(
     past each row's length; top-k itself is torch.topk on the result."""
     B, Hi, Di = q_index.shape
     bs = index_cache.shape[1]
+    shared_k = index_cache.dim() == 3
     max_ctx = int(seq_lens.max())
-    scores = torch.full((B, max_ctx), float("-inf"), dtype=torch.float32)
+    scores = torch.full((B, max_ctx), float("-inf"), dtype=torch.float32,
+                        device=q_index.device)
     for i in range(B):
         L = int(seq_lens[i])
         nb = (L + bs - 1) // bs
-        keys = index_cache[block_tables[i, :nb].long()].reshape(nb * bs, Hi, Di)[:L]
-        s = torch.einsum("hd,lhd->hl", q_index[i].float(), keys.float())
+        keys = index_cache[block_tables[i, :nb].long()].reshape(nb * bs, -1, Di)[:L]
+        if shared_k:
+            s = torch.einsum("hd,ld->hl", q_index[i].float(),
+                             keys.squeeze(1).float())
+        else:
+            s = torch.einsum("hd,lhd->hl", q_index[i].float(), keys.float())
         s = torch.relu(s)  # per DeepSeek-V3.2 indexer: ReLU before head-weighting
         scores[i, :L] = torch.einsum("h,hl->l", head_weights[i].float(), s)
     return scores

@@ -147,6 +147,7 @@ class Engine:
             mla_block_bytes = MLAKVCache.bytes_per_block(
                 end - start, cfg.kv_lora_rank, cfg.qk_rope_head_dim,
                 args.block_size, args.dtype,
+                index_dim=cfg.index_head_dim if cfg.is_dsa else 0,
             )
             if args.num_kv_blocks:
                 num_blocks = args.num_kv_blocks
@@ -161,6 +162,7 @@ class Engine:
                 end - start, cfg.kv_lora_rank, cfg.qk_rope_head_dim,
                 args.block_size, num_blocks, self.device,
                 args.dtype if args.dtype != torch.float32 else torch.float32,
+                index_dim=cfg.index_head_dim if cfg.is_dsa else 0,
             )
         else:
             kv_dtype = args.dtype

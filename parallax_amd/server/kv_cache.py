@@ -78,6 +78,7 @@ class MLAKVCache:
         num_blocks: int,
         device: torch.device,
         dtype: torch.dtype = torch.bfloat16,
+        index_dim: int = 0,
     ):
         self.kv_lora_rank = kv_lora_rank
         self.rope_dim = rope_dim
@@ -88,16 +89,30 @@ class MLAKVCache:
             torch.zeros((num_blocks, block_size, self.entry_dim), dtype=dtype, device=device)
             for _ in range(num_layers)
         ]
+        # DSA (DeepSeek-V3.2): one indexer key per token per layer, shared
+        # across indexer query heads (reference dsa_cache.py)
+        self.index_dim = index_dim
+        self.index_caches: List[torch.Tensor] = (
+            [
+                torch.zeros((num_blocks, block_size, index_dim), dtype=dtype, device=device)
+                for _ in range(num_layers)
+            ]
+            if index_dim > 0
+            else []
+        )
 
     def layer(self, idx: int) -> torch.Tensor:
         return self.caches[idx]
 
+    def index_layer(self, idx: int) -> torch.Tensor:
+        return self.index_caches[idx]
+
     @staticmethod
     def bytes_per_block(
         num_layers: int, kv_lora_rank: int, rope_dim: int, block_size: int,
-        dtype: torch.dtype = torch.bfloat16,
+        dtype: torch.dtype = torch.bfloat16, index_dim: int = 0,
     ) -> int:
-        return num_layers * block_size * (kv_lora_rank + rope_dim) * dtype.itemsize
+        return num_layers * block_size * (kv_lora_rank + rope_dim + index_dim) * dtype.itemsize
 
 
 class LinearStateCache:
