@@ -7,6 +7,7 @@ from . import llama  # noqa: F401  (registers LlamaForCausalLM / Qwen2ForCausalL
 from . import qwen3  # noqa: F401
 from . import qwen3_moe  # noqa: F401
 from . import deepseek_v3  # noqa: F401
+from . import deepseek_v32  # noqa: F401
 from . import gpt_oss  # noqa: F401
 from . import glm4_moe  # noqa: F401
 from . import qwen3_next  # noqa: F401

@@ -57,6 +57,7 @@ class ModelConfig:
     moe_router_bias: bool = False     # deepseek v3 e_score_correction_bias
 
     # -- MLA (deepseek) ----------------------------------------------------------
+    rope_interleave: bool = False    # GPT-J pairwise rope (deepseek default)
     q_lora_rank: int = 0
     kv_lora_rank: int = 0
     qk_nope_head_dim: int = 0
@@ -179,6 +180,11 @@ class ModelConfig:
             topk_group=cfg.get("topk_group", 0),
             n_group=cfg.get("n_group", 0),
             # MLA
+            rope_interleave=cfg.get(
+                "rope_interleave",
+                cfg.get("model_type", "") in (
+                    "deepseek_v2", "deepseek_v3", "deepseek_v32", "kimi_k2"),
+            ),
             q_lora_rank=cfg.get("q_lora_rank") or 0,
             kv_lora_rank=cfg.get("kv_lora_rank") or 0,
             qk_nope_head_dim=cfg.get("qk_nope_head_dim") or 0,

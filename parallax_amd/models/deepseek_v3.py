@@ -40,6 +40,15 @@ class MLAAttention(nn.Module):
         self.r = cfg.kv_lora_rank
         self.q_lora_rank = cfg.q_lora_rank
         self.scale = (self.dn + self.dr) ** -0.5
+        # deepseek yarn: softmax scale gets mscale^2 (HF yarn_apply_mscale)
+        rs = cfg.rope_scaling or {}
+        if rs.get("rope_type", rs.get("type", "default")) != "default":
+            factor = rs.get("factor", 1.0)
+            mscale_all_dim = rs.get("mscale_all_dim", 0)
+            if mscale_all_dim and factor > 1:
+                m = 0.1 * mscale_all_dim * math.log(factor) + 1.0
+                self.scale *= m * m
+        self.rope_neox = not cfg.rope_interleave
         h = cfg.hidden_size
         H = cfg.num_heads  # full head count for weight shapes; TP shards q_b/kv_b/o
 
@@ -86,7 +95,8 @@ class MLAAttention(nn.Module):
 
         q_pe = q_pe.contiguous()
         k_pe = k_pe.contiguous()
-        ops.rope_inplace(q_pe, k_pe, meta.positions, rope_cache)
+        ops.rope_inplace(q_pe, k_pe, meta.positions, rope_cache,
+                         is_neox=self.rope_neox)
 
         cache = meta.mla_cache.layer(self.local_layer_idx)
         ops.mla_reshape_and_cache(latent, k_pe.squeeze(1), cache, meta.slot_mapping)
@@ -107,6 +117,11 @@ class MLAAttention(nn.Module):
                 "bhr,hvr->bhv", out_latent.float(), self.w_uv.float()
             ).to(x.dtype)
         return self.o_proj(attn.reshape(T, H * self.dv))
+
+    def _prefill_sparse_mask(self, i, t0, QL, L, meta):
+        """Hook for sparse-attention subclasses (DSA): additive [QL, L] mask
+        applied on top of the causal mask during prefill. None = dense."""
+        return None
 
     def _prefill_attention(self, q_nope, q_pe, meta: ForwardMeta, cache):
         """Non-absorbed prefill: gather latents for each request's context from
@@ -137,6 +152,9 @@ class MLAAttention(nn.Module):
             qpos = torch.arange(L - QL, L, device=logits.device).unsqueeze(-1)
             kpos = torch.arange(L, device=logits.device).unsqueeze(0)
             logits.masked_fill_((kpos > qpos).unsqueeze(0), float("-inf"))
+            extra = self._prefill_sparse_mask(i, t0, QL, L, meta)
+            if extra is not None:  # DSA (deepseek_v32) top-k restriction
+                logits = logits + extra.unsqueeze(0)
             p = logits.softmax(dim=-1)
             out[t0 : t0 + QL] = (
                 torch.einsum("hql,lhv->qhv", p, v_ctx).to(out.dtype)

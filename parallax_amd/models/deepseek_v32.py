@@ -75,7 +75,7 @@ class DSAMLAAttention(MLAAttention):
         super().__init__(cfg, layer_idx, local_layer_idx)
         assert self.q_lora_rank, "DeepSeek-V3.2 requires q_lora_rank"
         self.indexer = DSAIndexer(cfg)
-        self._idx_scores = None  # [T_chunk, max_ctx] fp32, set during prefill
+        self._idx_state = None  # (q_idx, w_idx, idx_cache) during prefill
 
     def forward(self, x: torch.Tensor, meta: ForwardMeta, rope_cache: torch.Tensor):
         T = x.shape[0]
@@ -90,7 +90,8 @@ class DSAMLAAttention(MLAAttention):
 
         q_pe = q_pe.contiguous()
         k_pe = k_pe.contiguous()
-        ops.rope_inplace(q_pe, k_pe, meta.positions, rope_cache)
+        ops.rope_inplace(q_pe, k_pe, meta.positions, rope_cache,
+                         is_neox=self.rope_neox)
 
         cache = meta.mla_cache.layer(self.local_layer_idx)
         ops.mla_reshape_and_cache(latent, k_pe.squeeze(1), cache, meta.slot_mapping)
@@ -103,12 +104,9 @@ class DSAMLAAttention(MLAAttention):
         ops.store_indexer_cache(k_idx, idx_cache, meta.slot_mapping)
 
         if meta.is_prefill:
-            self._idx_scores = ops.dsa_indexer_scores(
-                q_idx, idx_cache, w_idx, meta.block_tables, meta.seq_lens_indexer(meta)
-                if hasattr(meta, "seq_lens_indexer") else meta.seq_lens,
-            )
+            self._idx_state = (q_idx, w_idx, idx_cache)
             attn = self._prefill_attention(q_nope, q_pe, meta, cache)
-            self._idx_scores = None
+            self._idx_state = None
         else:
             scores = ops.dsa_indexer_scores(
                 q_idx, idx_cache, w_idx, meta.block_tables, meta.seq_lens
@@ -141,7 +139,15 @@ class DSAMLAAttention(MLAAttention):
         """Additive [QL, L] mask from the indexer's top-k per query row.
         Matches HF: non-selected causally-valid keys get fp32 min (not -inf),
         so rows degrade identically in edge cases."""
-        s = self._idx_scores[t0 : t0 + QL, :L].clone()  # [QL, L] fp32
+        q_idx, w_idx, idx_cache = self._idx_state
+        bs = idx_cache.shape[1]
+        nb = (L + bs - 1) // bs
+        keys = idx_cache[meta.block_tables[i, :nb].long()].reshape(
+            nb * bs, -1
+        )[:L].float()                                   # [L, Di]
+        qi = q_idx[t0 : t0 + QL].float()                # [QL, Hi, Di]
+        s = torch.relu(torch.einsum("qhd,ld->qhl", qi, keys))
+        s = torch.einsum("qh,qhl->ql", w_idx[t0 : t0 + QL], s)  # [QL, L] fp32
         qpos = torch.arange(L - QL, L, device=s.device).unsqueeze(-1)
         kpos = torch.arange(L, device=s.device).unsqueeze(0)
         causal = kpos > qpos
