@@ -11,3 +11,4 @@ from . import deepseek_v32  # noqa: F401
 from . import gpt_oss  # noqa: F401
 from . import glm4_moe  # noqa: F401
 from . import qwen3_next  # noqa: F401
+from . import minimax_m2  # noqa: F401

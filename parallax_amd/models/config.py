@@ -33,6 +33,7 @@ class ModelConfig:
     o_proj_bias: bool = False
     mlp_bias: bool = False
     qk_norm: bool = False            # qwen3-style per-head q/k RMSNorm
+    qk_norm_full: bool = False       # minimax-m2: RMSNorm over the FULL q/k width
     partial_rotary_factor: float = 1.0
     sliding_window: int = -1         # -1 = full attention
     # which layers use the sliding window ("full" layers interleave, gpt-oss)
@@ -153,7 +154,9 @@ class ModelConfig:
             and cfg.get("model_type") == "gpt_oss",
             attention_sinks=cfg.get("model_type") == "gpt_oss",
             qk_norm=cfg.get("use_qk_norm", False)
-            or cfg.get("model_type", "") in ("qwen3", "qwen3_moe", "qwen3_next"),
+            or cfg.get("model_type", "") in ("qwen3", "qwen3_moe", "qwen3_next",
+                                             "minimax_m2"),
+            qk_norm_full=cfg.get("model_type", "") == "minimax_m2",
             sliding_window=sliding,
             layer_types=layer_types,
             eos_token_ids=[e for e in eos_ids if e is not None],
@@ -175,7 +178,7 @@ class ModelConfig:
             scoring_func=cfg.get("scoring_func")
             or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "deepseek_v32",
                                                        "kimi_k2", "glm4_moe",
-                                                       "glm4v_moe")
+                                                       "glm4v_moe", "minimax_m2")
                 else "softmax"),
             topk_group=cfg.get("topk_group", 0),
             n_group=cfg.get("n_group", 0),

@@ -66,9 +66,17 @@ class LlamaAttention(nn.Module):
         self.o_proj = RowParallelLinear(
             cfg.num_heads * cfg.head_dim, h, bias=cfg.o_proj_bias
         )
+        self.qk_norm_full = cfg.qk_norm_full
         if self.qk_norm:
-            self.q_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
-            self.k_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
+            if self.qk_norm_full:
+                # minimax-m2: one RMSNorm over the whole concatenated q (and k)
+                # width — couples heads, so TP sharding would change the math
+                assert comm.tp_size == 1, "full-width qk_norm incompatible with TP"
+                self.q_norm = RMSNorm(cfg.num_heads * cfg.head_dim, cfg.rms_norm_eps)
+                self.k_norm = RMSNorm(cfg.num_kv_heads * cfg.head_dim, cfg.rms_norm_eps)
+            else:
+                self.q_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
+                self.k_norm = RMSNorm(cfg.head_dim, cfg.rms_norm_eps)
         self.sinks: Optional[nn.Parameter] = None
         if cfg.attention_sinks:
             self.sinks = nn.Parameter(
@@ -85,8 +93,12 @@ class LlamaAttention(nn.Module):
         k = k.view(T, self.num_kv_heads, self.head_dim)
         v = v.view(T, self.num_kv_heads, self.head_dim)
         if self.qk_norm:
-            q = self.q_norm(q.contiguous())
-            k = self.k_norm(k.contiguous())
+            if self.qk_norm_full:
+                q = self.q_norm(q.reshape(T, -1)).view(T, self.num_heads, self.head_dim)
+                k = self.k_norm(k.reshape(T, -1)).view(T, self.num_kv_heads, self.head_dim)
+            else:
+                q = self.q_norm(q.contiguous())
+                k = self.k_norm(k.contiguous())
             v = v.contiguous()
         k_cache, v_cache = meta.kv_cache.layer(self.local_layer_idx)
         ops.rope_and_cache(

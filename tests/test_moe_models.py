@@ -89,3 +89,27 @@ def test_qwen3_moe_parity():
     )
     assert cfg.qk_norm and cfg.is_moe
     _run_parity(hf, cfg)
+
+
+def test_minimax_m2_parity():
+    """MiniMax-M2: full-width q/k RMSNorm + sigmoid MoE with the correction
+    bias stored at mlp.e_score_correction_bias (block level, not in the gate)."""
+    torch.manual_seed(14)
+    hf_cfg = transformers.MiniMaxM2Config(
+        vocab_size=256, hidden_size=64, intermediate_size=48,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, num_local_experts=8, num_experts_per_tok=2,
+        max_position_embeddings=512, rope_theta=10000.0,
+        tie_word_embeddings=False,
+    )
+    hf = transformers.MiniMaxM2ForCausalLM(hf_cfg).eval()
+    # give the correction bias a non-trivial value so the test exercises it
+    with torch.no_grad():
+        for layer in hf.model.layers:
+            layer.mlp.e_score_correction_bias.uniform_(-0.5, 0.5)
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["MiniMaxM2ForCausalLM"]}
+    )
+    assert cfg.qk_norm and cfg.qk_norm_full and cfg.scoring_func == "sigmoid"
+    assert cfg.is_moe and cfg.num_experts == 8
+    _run_parity(hf, cfg)
