@@ -155,7 +155,8 @@ class ModelConfig:
             attention_sinks=cfg.get("model_type") == "gpt_oss",
             qk_norm=cfg.get("use_qk_norm", False)
             or cfg.get("model_type", "") in ("qwen3", "qwen3_moe", "qwen3_next",
-                                             "minimax_m2"),
+                                             "minimax_m2", "qwen3_5_text",
+                                             "qwen3_5_moe_text"),
             qk_norm_full=cfg.get("model_type", "") == "minimax_m2",
             sliding_window=sliding,
             layer_types=layer_types,

@@ -87,11 +87,14 @@ class GatedDeltaNet(nn.Module):
         a = a.reshape(T, self.hv)
         return q, k, v, z, b, a
 
+    def _project(self, x: torch.Tensor):
+        """-> (q [T,hk,dk], k [T,hk,dk], v [T,hv,dv], z [T,hv,dv], b, a [T,hv]).
+        Subclasses (qwen3_5: split in_proj_qkv/z/b/a) override this."""
+        return self._split_qkvz_ba(self.in_proj_qkvz(x), self.in_proj_ba(x))
+
     def forward(self, x: torch.Tensor, meta: ForwardMeta, rope_cache) -> torch.Tensor:
         T = x.shape[0]
-        q, k, v, z, b, a = self._split_qkvz_ba(
-            self.in_proj_qkvz(x), self.in_proj_ba(x)
-        )
+        q, k, v, z, b, a = self._project(x)
         mixed = torch.cat(
             [q.reshape(T, -1), k.reshape(T, -1), v.reshape(T, -1)], dim=-1
         )  # [T, conv_dim]
@@ -267,12 +270,14 @@ class Qwen3NextMoE(nn.Module):
 
 
 class Qwen3NextDecoderLayer(nn.Module):
+    deltanet_cls = GatedDeltaNet
+
     def __init__(self, cfg: ModelConfig, layer_idx: int, local_layer_idx: int,
                  kv_layer_idx: int, linear_layer_idx: int):
         super().__init__()
         self.is_linear = cfg.layer_type(layer_idx) == "linear_attention"
         if self.is_linear:
-            self.linear_attn = GatedDeltaNet(cfg, layer_idx, linear_layer_idx)
+            self.linear_attn = self.deltanet_cls(cfg, layer_idx, linear_layer_idx)
         else:
             self.self_attn = Qwen3NextAttention(cfg, layer_idx, kv_layer_idx)
         self.mlp = Qwen3NextMoE(cfg) if cfg.is_moe_layer(layer_idx) else LlamaMLP(cfg)
@@ -303,6 +308,7 @@ class Qwen3NextShardModel(LlamaShardModel):
         super().__init__(cfg, start_layer, end_layer)
 
     decoder_layer_cls = None  # constructed in _build_layers
+    hybrid_layer_cls = Qwen3NextDecoderLayer
 
     def _build_layers(self, cfg: ModelConfig, start: int, end: int):
         layers = []
@@ -310,10 +316,10 @@ class Qwen3NextShardModel(LlamaShardModel):
         lin_idx = 0
         for i, g in enumerate(range(start, end)):
             if cfg.layer_type(g) == "linear_attention":
-                layers.append(Qwen3NextDecoderLayer(cfg, g, i, 0, lin_idx))
+                layers.append(self.hybrid_layer_cls(cfg, g, i, 0, lin_idx))
                 lin_idx += 1
             else:
-                layers.append(Qwen3NextDecoderLayer(cfg, g, i, kv_idx, 0))
+                layers.append(self.hybrid_layer_cls(cfg, g, i, kv_idx, 0))
                 kv_idx += 1
         return layers
 
@@ -347,7 +353,9 @@ class Qwen3NextShardModel(LlamaShardModel):
                     getattr(la, sub).data.copy_(tensor.float().to(torch.bfloat16))
                 elif sub == "norm":
                     la.norm.weight.data.copy_(t)
-                elif sub in ("in_proj_qkvz", "in_proj_ba", "out_proj"):
+                elif sub in ("in_proj_qkvz", "in_proj_ba", "out_proj",
+                             "in_proj_qkv", "in_proj_z", "in_proj_b",
+                             "in_proj_a"):
                     getattr(la, sub).weight.data.copy_(t)
                 else:
                     return False
@@ -357,7 +365,7 @@ class Qwen3NextShardModel(LlamaShardModel):
                     0, t if parts[4] == "weight" else None,
                     t if parts[4] == "bias" else None)
                 return True
-            if parts[2] == "mlp":
+            if parts[2] == "mlp" and isinstance(layer.mlp, Qwen3NextMoE):
                 return self._load_moe_weight(layer, parts[3:], t)
         return super().load_hf_weight(name, tensor)
 

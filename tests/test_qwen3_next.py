@@ -100,3 +100,104 @@ def test_slot_recycling(tiny_next):
     out1 = list(eng.generate([p], sp).values())[0]
     out2 = list(eng.generate([p], sp).values())[0]
     assert out1 == out2
+
+
+# -- Qwen3.5 (split DeltaNet projections, dense MLP) ---------------------------
+
+
+@pytest.fixture(scope="module")
+def tiny_q35():
+    torch.manual_seed(23)
+    from transformers.models.qwen3_5.configuration_qwen3_5 import Qwen3_5TextConfig
+
+    hf_cfg = Qwen3_5TextConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, linear_num_key_heads=2, linear_num_value_heads=4,
+        linear_key_head_dim=16, linear_value_head_dim=16,
+        linear_conv_kernel_dim=3,
+        layer_types=["linear_attention", "full_attention"] * 2,
+        max_position_embeddings=512, tie_word_embeddings=False,
+        rope_theta=10000.0,
+    )
+    hf = transformers.Qwen3_5ForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["Qwen3_5ForCausalLM"]}
+    )
+    return hf, cfg
+
+
+def test_q35_weight_load_complete(tiny_q35):
+    hf, cfg = tiny_q35
+    assert cfg.qk_norm and cfg.has_linear_layers
+    m = get_model_class(cfg.architecture)(cfg)
+    missed = [n for n, t in hf.state_dict().items() if not m.load_hf_weight(n, t)]
+    assert missed == [], f"unrouted: {missed[:10]}"
+
+
+def test_q35_greedy_parity(tiny_q35):
+    hf, cfg = tiny_q35
+    prompt = [7, 42, 99, 5, 81, 23, 150, 3, 66, 12]
+    with torch.no_grad():
+        ref = hf.generate(
+            torch.tensor([prompt]), max_new_tokens=6, do_sample=False
+        )[0][len(prompt):].tolist()
+    eng = _engine_with_weights(hf, cfg)
+    out = eng.generate(
+        [prompt], [SamplingParams(temperature=0.0, max_new_tokens=6, ignore_eos=True)]
+    )
+    assert list(out.values())[0] == ref
+
+
+def test_q35_chunked_prefill_parity(tiny_q35):
+    hf, cfg = tiny_q35
+    prompt = list(range(3, 30))
+    with torch.no_grad():
+        ref = hf.generate(
+            torch.tensor([prompt]), max_new_tokens=4, do_sample=False
+        )[0][len(prompt):].tolist()
+    eng = _engine_with_weights(hf, cfg, prefill_chunk_size=8)
+    out = eng.generate(
+        [prompt], [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)]
+    )
+    assert list(out.values())[0] == ref
+
+
+def test_q35_moe_greedy_parity():
+    """Qwen3.5-MoE: split DeltaNet projections + qwen3_next-style MoE with a
+    sigmoid-gated shared expert."""
+    torch.manual_seed(29)
+    from transformers.models.qwen3_5_moe.configuration_qwen3_5_moe import (
+        Qwen3_5MoeTextConfig,
+    )
+
+    hf_cfg = Qwen3_5MoeTextConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=96,
+        moe_intermediate_size=32, shared_expert_intermediate_size=32,
+        num_experts=4, num_experts_per_tok=2, norm_topk_prob=True,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, linear_num_key_heads=2, linear_num_value_heads=4,
+        linear_key_head_dim=16, linear_value_head_dim=16,
+        linear_conv_kernel_dim=3,
+        layer_types=["linear_attention", "full_attention"] * 2,
+        max_position_embeddings=512, tie_word_embeddings=False,
+        rope_theta=10000.0,
+    )
+    hf = transformers.Qwen3_5MoeForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["Qwen3_5MoeForCausalLM"]}
+    )
+    assert cfg.is_moe and cfg.has_linear_layers
+    prompt = [7, 42, 99, 5, 81, 23, 150, 3, 66, 12]
+    with torch.no_grad():
+        ref = hf.generate(
+            torch.tensor([prompt]), max_new_tokens=5, do_sample=False
+        )[0][len(prompt):].tolist()
+    m = get_model_class(cfg.architecture)(cfg)
+    missed = [n for n, t in hf.state_dict().items() if not m.load_hf_weight(n, t)]
+    assert missed == [], f"unrouted: {missed[:10]}"
+    eng = _engine_with_weights(hf, cfg)
+    out = eng.generate(
+        [prompt], [SamplingParams(temperature=0.0, max_new_tokens=5, ignore_eos=True)]
+    )
+    assert list(out.values())[0] == ref
