@@ -13,3 +13,4 @@ from . import glm4_moe  # noqa: F401
 from . import qwen3_next  # noqa: F401
 from . import minimax_m2  # noqa: F401
 from . import qwen3_5  # noqa: F401
+from . import minimax  # noqa: F401

@@ -199,11 +199,20 @@ class ModelConfig:
             index_head_dim=cfg.get("index_head_dim") or 0,
             index_topk=cfg.get("index_topk") or 0,
             # hybrid linear attention
-            linear_num_key_heads=cfg.get("linear_num_key_heads") or 0,
-            linear_num_value_heads=cfg.get("linear_num_value_heads") or 0,
-            linear_key_head_dim=cfg.get("linear_key_head_dim") or 0,
-            linear_value_head_dim=cfg.get("linear_value_head_dim") or 0,
-            linear_conv_kernel_dim=cfg.get("linear_conv_kernel_dim") or 4,
+            linear_num_key_heads=cfg.get("linear_num_key_heads") or (
+                num_heads if cfg.get("model_type") in ("minimax", "minimax_m1")
+                else 0),
+            linear_num_value_heads=cfg.get("linear_num_value_heads") or (
+                num_heads if cfg.get("model_type") in ("minimax", "minimax_m1")
+                else 0),
+            linear_key_head_dim=cfg.get("linear_key_head_dim") or (
+                head_dim if cfg.get("model_type") in ("minimax", "minimax_m1")
+                else 0),
+            linear_value_head_dim=cfg.get("linear_value_head_dim") or (
+                head_dim if cfg.get("model_type") in ("minimax", "minimax_m1")
+                else 0),
+            linear_conv_kernel_dim=cfg.get("linear_conv_kernel_dim") or (
+                1 if cfg.get("model_type") in ("minimax", "minimax_m1") else 4),
             raw=cfg,
         )
         return mc

@@ -207,8 +207,11 @@ class LlamaShardModel(nn.Module):
         for layer in self.layers:
             hidden, residual = layer(hidden, residual, meta, self.rope_cache)
         if self.is_last:
-            hidden, _ = self.norm(hidden, residual)
-        else:
+            if residual is None:  # post-norm families (minimax) fold residuals
+                hidden = self.norm(hidden)
+            else:
+                hidden, _ = self.norm(hidden, residual)
+        elif residual is not None:
             hidden = hidden + residual
         return hidden
 
