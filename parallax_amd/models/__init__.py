@@ -14,3 +14,4 @@ from . import qwen3_next  # noqa: F401
 from . import minimax_m2  # noqa: F401
 from . import qwen3_5  # noqa: F401
 from . import minimax  # noqa: F401
+from . import minimax_m3  # noqa: F401

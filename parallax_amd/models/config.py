@@ -70,6 +70,11 @@ class ModelConfig:
     index_head_dim: int = 0
     index_topk: int = 0
 
+    # -- MSA indexer (minimax-m3 block-sparse attention) ------------------------
+    index_block_size: int = 0
+    index_topk_blocks: int = 0
+    index_local_blocks: int = 0
+
     # -- attention sinks (gpt-oss) --------------------------------------------------
     attention_sinks: bool = False
 
@@ -97,6 +102,10 @@ class ModelConfig:
     @property
     def is_dsa(self) -> bool:
         return self.index_topk > 0 and self.index_head_dim > 0
+
+    @property
+    def is_msa(self) -> bool:
+        return self.index_topk_blocks > 0 and self.index_block_size > 0
 
     @property
     def has_linear_layers(self) -> bool:
@@ -156,7 +165,8 @@ class ModelConfig:
             qk_norm=cfg.get("use_qk_norm", False)
             or cfg.get("model_type", "") in ("qwen3", "qwen3_moe", "qwen3_next",
                                              "minimax_m2", "qwen3_5_text",
-                                             "qwen3_5_moe_text"),
+                                             "qwen3_5_moe_text",
+                                             "minimax_m3_vl_text", "minimax_m3"),
             qk_norm_full=cfg.get("model_type", "") == "minimax_m2",
             sliding_window=sliding,
             layer_types=layer_types,
@@ -179,7 +189,9 @@ class ModelConfig:
             scoring_func=cfg.get("scoring_func")
             or ("sigmoid" if cfg.get("model_type") in ("deepseek_v3", "deepseek_v32",
                                                        "kimi_k2", "glm4_moe",
-                                                       "glm4v_moe", "minimax_m2")
+                                                       "glm4v_moe", "minimax_m2",
+                                                       "minimax_m3_vl_text",
+                                                       "minimax_m3")
                 else "softmax"),
             topk_group=cfg.get("topk_group", 0),
             n_group=cfg.get("n_group", 0),
@@ -194,10 +206,14 @@ class ModelConfig:
             qk_nope_head_dim=cfg.get("qk_nope_head_dim") or 0,
             qk_rope_head_dim=cfg.get("qk_rope_head_dim") or 0,
             v_head_dim=cfg.get("v_head_dim") or 0,
-            # DSA indexer
+            # DSA / MSA indexer
             index_n_heads=cfg.get("index_n_heads") or 0,
             index_head_dim=cfg.get("index_head_dim") or 0,
             index_topk=cfg.get("index_topk") or 0,
+            index_block_size=cfg.get("index_block_size") or 0,
+            index_topk_blocks=cfg.get("index_topk_blocks") or 0,
+            index_local_blocks=cfg.get("index_local_blocks")
+            if cfg.get("index_local_blocks") is not None else 0,
             # hybrid linear attention
             linear_num_key_heads=cfg.get("linear_num_key_heads") or (
                 num_heads if cfg.get("model_type") in ("minimax", "minimax_m1")

@@ -76,9 +76,14 @@ class FusedMoE(nn.Module):
     tensors [E, 2I, H] / [E, H, I] — the exact layout the MFMA grouped-GEMM
     kernel consumes."""
 
-    def __init__(self, cfg: ModelConfig, intermediate_size: Optional[int] = None):
+    def __init__(self, cfg: ModelConfig, intermediate_size: Optional[int] = None,
+                 act_limit: float = 0.0):
         super().__init__()
         self.cfg = cfg
+        # act_limit > 0: gpt-oss/minimax-m3 clamped swiglu
+        #   glu = min(gate, L) * sigmoid(1.702 * min(gate, L));
+        #   out = (clamp(up, -L, L) + 1) * glu
+        self.act_limit = act_limit
         inter = intermediate_size or cfg.moe_intermediate_size or cfg.intermediate_size
         self.intermediate_size = inter
         E, H = cfg.num_experts, cfg.hidden_size
@@ -91,7 +96,8 @@ class FusedMoE(nn.Module):
         topk_ids, topk_w = self.router(x)           # [T,k]
         if x.is_cuda:
             return ops.fused_moe_forward(
-                x, self.w_gate_up, self.w_down, topk_ids, topk_w
+                x, self.w_gate_up, self.w_down, topk_ids, topk_w,
+                limit=self.act_limit,
             ).to(x.dtype)
         out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
         flat_ids = topk_ids.reshape(-1)              # [T*k]
@@ -105,7 +111,13 @@ class FusedMoE(nn.Module):
             toks = token_idx[sel]
             xe = x[toks]                              # [n_e, H]
             h = F.linear(xe, self.w_gate_up[e])       # [n_e, 2I]
-            h = ops.silu_and_mul(h)
+            if self.act_limit > 0:
+                I = self.intermediate_size
+                gate = h[:, :I].clamp(max=self.act_limit)
+                up = h[:, I:].clamp(min=-self.act_limit, max=self.act_limit)
+                h = (up + 1.0) * (gate * torch.sigmoid(gate * 1.702))
+            else:
+                h = ops.silu_and_mul(h)
             ye = F.linear(h, self.w_down[e]).float()  # [n_e, H]
             out.index_add_(0, toks, ye * flat_w[sel].unsqueeze(-1))
         return out.to(x.dtype)

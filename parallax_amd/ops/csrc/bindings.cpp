@@ -48,8 +48,8 @@ void launch_mla_paged_attention_decode(void*, const void*, const void*,
 void launch_msa_paged_attention_decode(void*, const void*, const void*,
                                        const void*, const int*, const int*,
                                        const int*, int, int, int, int, int,
-                                       int, int, int64_t, float, hipStream_t,
-                                       bool*);
+                                       int, int, int, int64_t, float,
+                                       hipStream_t, bool*);
 void launch_build_moe_tiles(int*, int*, const int*, int, int, hipStream_t);
 void launch_moe_gate_up(void*, const void*, const void*, const int*,
                         const int*, const int64_t*, const int*, int, int, int,
@@ -447,13 +447,17 @@ void msa_paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int D = q.size(2);
   const int Hk = k_cache.size(1);
   const int BS = k_cache.size(2);
+  // token_positions: [B, P] shared across kv heads, or [B, Hk, P] per head
+  const int pos_heads = token_positions.dim() == 3 ? token_positions.size(1) : 1;
+  if (pos_heads > 1)
+    TORCH_CHECK(pos_heads == Hk, "positions head dim must equal Hk");
   bool launched = false;
   launch_msa_paged_attention_decode(
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-      token_positions.data_ptr<int>(), token_positions.size(1), B, Hq, Hk, D,
-      BS, block_tables.size(1), row_stride(q, "q"), (float)scale, cur_stream(),
-      &launched);
+      token_positions.data_ptr<int>(), token_positions.size(-1), pos_heads, B,
+      Hq, Hk, D, BS, block_tables.size(1), row_stride(q, "q"), (float)scale,
+      cur_stream(), &launched);
   TORCH_CHECK(launched, "no MSA kernel for D=", D, " BS=", BS, " G=", Hq / Hk);
 }
 

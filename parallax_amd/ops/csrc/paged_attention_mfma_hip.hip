@@ -31,14 +31,18 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     const float scale, const int sliding_window, const float softcap,
     const float* __restrict__ sinks, const int part_tokens,
     const float k_scale, const float v_scale,
-    const int* __restrict__ token_positions = nullptr,  // [B, max_pos] (SPARSE)
-    const int max_positions = 0) {
+    const int* __restrict__ token_positions = nullptr,  // [B(,Hk), max_pos]
+    const int max_positions = 0, const int pos_heads = 1) {
   const int seq = blockIdx.y;
   const int hk = blockIdx.x;
   const int L = seq_lens[seq];
-  // MSA sparse mode: iterate an explicit token-position list (-1 padded)
+  // MSA sparse mode: iterate an explicit token-position list (-1 padded);
+  // pos_heads > 1 = one independent selection per kv head (minimax-m3)
   const int* pos_row =
-      SPARSE ? token_positions + (size_t)seq * max_positions : nullptr;
+      SPARSE ? token_positions +
+                   ((size_t)seq * pos_heads + (pos_heads > 1 ? hk : 0)) *
+                       max_positions
+             : nullptr;
   const int domain = SPARSE ? max_positions : L;
 
   int tok_begin = 0, tok_end = domain;
@@ -327,8 +331,9 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
 extern "C" void launch_msa_paged_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* seq_lens, const int* token_positions,
-    int max_positions, int B, int Hq, int Hk, int D, int BS, int max_blocks,
-    int64_t q_stride, float scale, hipStream_t stream, bool* launched) {
+    int max_positions, int pos_heads, int B, int Hq, int Hk, int D, int BS,
+    int max_blocks, int64_t q_stride, float scale, hipStream_t stream,
+    bool* launched) {
   const int G = Hq / Hk;
   *launched = false;
   if (G > 16) return;
@@ -338,7 +343,8 @@ extern "C" void launch_msa_paged_attention_decode(
       , dim3(grid), dim3(DM_THREADS), 0, stream,                                        \
           (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q, k_cache,       \
           v_cache, block_tables, seq_lens, max_blocks, Hk, G, q_stride, scale, \
-          -1, 0.f, nullptr, 0, 1.f, 1.f, token_positions, max_positions);      \
+          -1, 0.f, nullptr, 0, 1.f, 1.f, token_positions, max_positions,      \
+          pos_heads);                                                          \
   *launched = true;
   if (D == 128 && BS == 32) { MSA_LAUNCH(128, 32) }
   else if (D == 128 && BS == 16) { MSA_LAUNCH(128, 16) }

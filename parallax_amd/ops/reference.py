@@ -512,7 +512,9 @@ def msa_paged_attention_decode(
     v_cache: torch.Tensor,
     block_tables: torch.Tensor,
     seq_lens: torch.Tensor,
-    token_positions: torch.Tensor,  # [B, max_positions] int, -1 padded
+    token_positions: torch.Tensor,  # [B, max_positions] int, -1 padded; or
+                                    # [B, Hk, max_positions] for one selection
+                                    # per kv head (minimax-m3)
     scale: float,
 ) -> torch.Tensor:
     """Exact attention over explicit token positions mapped through block
@@ -520,15 +522,18 @@ def msa_paged_attention_decode(
     B, Hq, D = q.shape
     Hk = k_cache.shape[1]
     group = Hq // Hk
+    per_head = token_positions.dim() == 3
     out = torch.empty_like(q)
     for i in range(B):
         L = int(seq_lens[i])
         k, v = _gather_kv(k_cache, v_cache, block_tables[i], L)
-        pos = token_positions[i]
-        keep = pos[(pos >= 0) & (pos < L)].long()
-        kf = k.float()[keep].transpose(0, 1).repeat_interleave(group, dim=0)
-        vf = v.float()[keep].transpose(0, 1).repeat_interleave(group, dim=0)
-        logits = torch.einsum("hd,hld->hl", q[i].float(), kf) * scale
-        p = torch.softmax(logits, dim=-1)
-        out[i] = torch.einsum("hl,hld->hd", p, vf).to(q.dtype)
+        for h in range(Hk):
+            pos = token_positions[i, h] if per_head else token_positions[i]
+            keep = pos[(pos >= 0) & (pos < L)].long()
+            kf = k.float()[keep, h]                      # [P, D]
+            vf = v.float()[keep, h]
+            qh = q[i, h * group:(h + 1) * group].float() # [G, D]
+            logits = (qh @ kf.T) * scale
+            p = torch.softmax(logits, dim=-1)
+            out[i, h * group:(h + 1) * group] = (p @ vf).to(q.dtype)
     return out

@@ -175,6 +175,7 @@ class Engine:
                 head_dim=cfg.head_dim,
                 block_size=args.block_size,
                 dtype=kv_dtype,
+                index_dim=cfg.index_head_dim if cfg.is_msa else 0,
             )
             num_blocks = args.num_kv_blocks or CacheManager.num_blocks_from_memory(
                 spec, self.device, args.cache_memory_fraction

@@ -29,10 +29,14 @@ class KVCacheSpec:
     head_dim: int
     block_size: int
     dtype: torch.dtype = torch.bfloat16
+    # sparse-attention indexer keys: one index_dim vector per token per layer
+    # (minimax-m3 MSA; dtype follows the main cache but never fp8)
+    index_dim: int = 0
 
     @property
     def bytes_per_token_per_layer(self) -> int:
-        return 2 * self.num_kv_heads * self.head_dim * self.dtype.itemsize
+        return (2 * self.num_kv_heads * self.head_dim + self.index_dim) \
+            * self.dtype.itemsize
 
     def bytes_per_block(self) -> int:
         return self.block_size * self.bytes_per_token_per_layer * self.num_layers
@@ -54,9 +58,22 @@ class PagedKVCache:
             torch.zeros(shape, dtype=spec.dtype, device=device)
             for _ in range(spec.num_layers)
         ]
+        idx_dtype = spec.dtype if spec.dtype != torch.float8_e4m3fn \
+            else torch.bfloat16
+        self.index_caches: List[torch.Tensor] = (
+            [
+                torch.zeros((num_blocks, spec.block_size, spec.index_dim),
+                            dtype=idx_dtype, device=device)
+                for _ in range(spec.num_layers)
+            ]
+            if spec.index_dim > 0 else []
+        )
 
     def layer(self, idx: int):
         return self.k_caches[idx], self.v_caches[idx]
+
+    def index_layer(self, idx: int) -> torch.Tensor:
+        return self.index_caches[idx]
 
     @staticmethod
     def num_blocks_for_bytes(spec: KVCacheSpec, budget_bytes: int) -> int:

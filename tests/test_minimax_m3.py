@@ -1,0 +1,103 @@
+"""MiniMax-M3 (MSA block-sparse attention with per-KV-head indexer, clamped
+MoE + shared experts, Gemma norms) greedy parity vs HF transformers, CPU fp32."""
+
+import pytest
+import torch
+
+transformers = pytest.importorskip("transformers")
+
+from parallax_amd.models import get_model_class
+from parallax_amd.models.config import ModelConfig
+from parallax_amd.server.engine import Engine, EngineArgs
+from parallax_amd.server.sampling_params import SamplingParams
+
+
+@pytest.fixture(scope="module")
+def tiny_m3():
+    torch.manual_seed(37)
+    from transformers.models.minimax_m3_vl.configuration_minimax_m3_vl import (
+        MiniMaxM3VLTextConfig,
+    )
+
+    hf_cfg = MiniMaxM3VLTextConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=32,
+        dense_intermediate_size=96, shared_intermediate_size=32,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, rotary_dim=8, num_local_experts=8, num_experts_per_tok=2,
+        routed_scaling_factor=1.5,
+        layer_types=["minimax_m3_sparse", "full_attention"] * 2,
+        mlp_layer_types=["sparse", "dense"] * 2,
+        index_n_heads=2, index_head_dim=16, index_block_size=4,
+        index_topk_blocks=2, index_local_blocks=1,
+        max_position_embeddings=512, tie_word_embeddings=False,
+        rope_parameters={"rope_type": "default", "rope_theta": 10000.0},
+        bos_token_id=1, eos_token_id=2,
+    )
+    hf = transformers.MiniMaxM3VLForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["MiniMaxM3ForCausalLM"]}
+    )
+    return hf, cfg
+
+
+def _engine_with(hf, cfg, **kw):
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=torch.float32, max_batch_size=8, **kw))
+    m = get_model_class(cfg.architecture)(cfg).eval()
+    for name, t in hf.state_dict().items():
+        m.load_hf_weight(name, t)
+    m = m.float()
+    eng.model = m
+    return eng
+
+
+def test_config(tiny_m3):
+    _, cfg = tiny_m3
+    assert cfg.is_msa and not cfg.is_dsa
+    assert cfg.index_block_size == 4 and cfg.index_topk_blocks == 2
+    assert cfg.qk_norm and cfg.scoring_func == "sigmoid"
+    # HF's rotary ignores config.rotary_dim — partial rotary comes only from
+    # rope_parameters.partial_rotary_factor (full-width here)
+    assert cfg.partial_rotary_factor == 1.0
+
+
+def test_weight_load_complete(tiny_m3):
+    hf, cfg = tiny_m3
+    m = get_model_class(cfg.architecture)(cfg)
+    missed = [n for n, t in hf.state_dict().items() if not m.load_hf_weight(n, t)]
+    assert missed == [], f"unrouted: {missed[:10]}"
+
+
+def test_greedy_parity_sparse(tiny_m3):
+    """Context (27+) spans ~7 key blocks of 4 with topk_blocks=2 + 1 local —
+    selection genuinely drops blocks in both prefill and decode."""
+    hf, cfg = tiny_m3
+    prompt = list(range(3, 30))
+    with torch.no_grad():
+        ref = hf.generate(
+            torch.tensor([prompt]), max_new_tokens=6, do_sample=False
+        )[0][len(prompt):].tolist()
+    eng = _engine_with(hf, cfg)
+    out = eng.generate(
+        [prompt], [SamplingParams(temperature=0.0, max_new_tokens=6, ignore_eos=True)]
+    )
+    assert list(out.values())[0] == ref
+
+
+def test_greedy_parity_chunked(tiny_m3):
+    """Chunked prefill must match HF for the first tokens AND be identical to
+    our own non-chunked run throughout. (Full 4-token HF parity on this prompt
+    hits a genuine fp near-tie in the block top-k — margin ~6e-3 after three
+    decode steps — so only the stable prefix is compared against HF.)"""
+    hf, cfg = tiny_m3
+    prompt = list(range(5, 38))
+    with torch.no_grad():
+        ref = hf.generate(
+            torch.tensor([prompt]), max_new_tokens=4, do_sample=False
+        )[0][len(prompt):].tolist()
+    sp = SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)
+    chunked = list(_engine_with(hf, cfg, prefill_chunk_size=10)
+                   .generate([prompt], [sp]).values())[0]
+    full = list(_engine_with(hf, cfg).generate([prompt], [sp]).values())[0]
+    assert chunked[:3] == ref[:3]
+    assert chunked == full
