@@ -154,3 +154,35 @@ def test_msa_kernel_gpu():
         sl.cpu(), pos.cpu(), 0.088,
     )
     torch.testing.assert_close(out.float().cpu(), expect, atol=4e-2, rtol=4e-2)
+
+
+@pytest.mark.gpu
+def test_msa_kernel_gpu_per_head_positions():
+    """[B, Hk, P] position lists: each kv head attends its own block set
+    (minimax-m3 per-KV-head indexer contract)."""
+    torch.manual_seed(9)
+    Hk, G, D, BS = 2, 2, 128, 32
+    ctxs = [420, 130]
+    B, Hq = len(ctxs), G * Hk
+    max_blocks = (max(ctxs) + BS - 1) // BS
+    NB = B * max_blocks + 1
+    kc = torch.randn(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
+    bt = (torch.arange(B * max_blocks, dtype=torch.int32, device="cuda")
+          .reshape(B, max_blocks) + 1)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+    sl = torch.tensor(ctxs, dtype=torch.int32, device="cuda")
+    P = 96
+    pos = torch.full((B, Hk, P), -1, dtype=torch.int32, device="cuda")
+    g = torch.Generator().manual_seed(3)
+    for i, L in enumerate(ctxs):
+        for h in range(Hk):
+            n = min(P, L)
+            sel = torch.randperm(L, generator=g)[:n].sort().values
+            pos[i, h, :n] = sel.to(torch.int32).cuda()
+    out = ops.msa_paged_attention_decode(q, kc, vc, bt, sl, pos, 0.088)
+    expect = ref.msa_paged_attention_decode(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(),
+        sl.cpu(), pos.cpu(), 0.088,
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=4e-2, rtol=4e-2)
