@@ -21,6 +21,7 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
+import torch.nn.functional as F
 
 from .. import ops
 from .config import ModelConfig
@@ -64,7 +65,14 @@ class DSAIndexer(nn.Module):
         cs = rope_cache[positions.long()]
         q = self.wq_b(q_resid).view(T, self.n_heads, self.head_dim)
         q = self._rope_neox(q, cs)
-        k = self.k_norm(self.wk(x).float()).to(x.dtype).unsqueeze(1)
+        # fp32 LayerNorm regardless of model dtype (HF computes it on the
+        # bf16 tensor; fp32 here is a superset in precision and keeps
+        # torch.layer_norm happy with mixed dtypes)
+        k = F.layer_norm(
+            self.wk(x).float(), (self.head_dim,),
+            self.k_norm.weight.float(), self.k_norm.bias.float(),
+            self.k_norm.eps,
+        ).to(x.dtype).unsqueeze(1)
         k = self._rope_neox(k, cs).squeeze(1)
         w = self.weights_proj(x).float() * self.weight_scale
         return q, k, w
@@ -108,9 +116,7 @@ class DSAMLAAttention(MLAAttention):
             attn = self._prefill_attention(q_nope, q_pe, meta, cache)
             self._idx_state = None
         else:
-            scores = ops.dsa_indexer_scores(
-                q_idx, idx_cache, w_idx, meta.block_tables, meta.seq_lens
-            )
+            scores = self._decode_scores(q_idx, w_idx, idx_cache, meta)
             topk_indices = self._decode_topk(scores, meta.seq_lens)
             q_latent = torch.einsum(
                 "bhd,hdr->bhr", q_nope.float(), self.w_uk.float()
@@ -118,11 +124,30 @@ class DSAMLAAttention(MLAAttention):
             out_latent = ops.dsa_paged_attention_decode(
                 q_latent.contiguous(), q_pe.contiguous(), cache,
                 meta.block_tables, meta.seq_lens, topk_indices, self.scale,
+                max_seq_len=meta.max_seq_len or None,
             )
             attn = torch.einsum(
                 "bhr,hvr->bhv", out_latent.float(), self.w_uv.float()
             ).to(x.dtype)
         return self.o_proj(attn.reshape(T, H * self.dv))
+
+    def _decode_scores(self, q_idx, w_idx, idx_cache, meta) -> torch.Tensor:
+        """[B, msl] fp32 weighted relu scores over the paged indexer cache.
+        Fully batched (no per-row host loops) so it is hipGraph-capturable;
+        shapes are fixed by meta.max_seq_len."""
+        B = q_idx.shape[0]
+        bs = idx_cache.shape[1]
+        msl = meta.max_seq_len or int(meta.seq_lens.max())
+        npages = min((msl + bs - 1) // bs, meta.block_tables.shape[1])
+        msl = min(msl, npages * bs)
+        keys = idx_cache[meta.block_tables[:, :npages].long()].reshape(
+            B, npages * bs, -1
+        )[:, :msl]                                       # [B, msl, Di]
+        s = torch.relu(torch.einsum("bhd,btd->bht", q_idx.float(), keys.float()))
+        scores = torch.einsum("bh,bht->bt", w_idx, s)    # [B, msl]
+        t = torch.arange(msl, device=scores.device)
+        valid = t.view(1, -1) < meta.seq_lens.view(B, 1)
+        return scores.masked_fill(~valid, float("-inf"))
 
     def _decode_topk(self, scores: torch.Tensor, seq_lens: torch.Tensor) -> torch.Tensor:
         """[B, index_topk] int32; rows with ctx <= topk use the dense fallback

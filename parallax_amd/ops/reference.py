@@ -431,14 +431,19 @@ def dsa_indexer_scores(
 
 
 def store_indexer_cache(
-    index_keys: torch.Tensor,    # [T, index_heads, index_dim]
-    index_cache: torch.Tensor,   # [num_blocks, bs, index_heads, index_dim]
+    index_keys: torch.Tensor,    # [T, index_heads, index_dim] or [T, index_dim]
+    index_cache: torch.Tensor,   # [num_blocks(+1 trash), bs, ...]
     slot_mapping: torch.Tensor,
 ) -> None:
+    """Scatter indexer keys by slot. hipGraph-capture safe: instead of
+    boolean compaction (data-dependent shapes) pad tokens (slot -1) are
+    redirected into the cache's LAST block, which the engine over-allocates
+    as a trash row (MLAKVCache/PagedKVCache index_caches)."""
     bs = index_cache.shape[1]
-    valid = slot_mapping >= 0
-    slots = slot_mapping[valid].long()
-    index_cache[slots // bs, slots % bs] = index_keys[valid].to(index_cache.dtype)
+    slots = slot_mapping.long()
+    trash = (index_cache.shape[0] - 1) * bs
+    slots = torch.where(slots >= 0, slots, torch.full_like(slots, trash))
+    index_cache[slots // bs, slots % bs] = index_keys.to(index_cache.dtype)
 
 
 def msa_block_scores(

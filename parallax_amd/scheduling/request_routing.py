@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import random
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List, Optional, Tuple
 
 from ..utils.logging_config import get_logger
 from .layer_allocation import Pipeline
@@ -116,3 +116,54 @@ class RequestRouter:
     @property
     def num_pipelines(self) -> int:
         return len(self.pipelines)
+
+
+def dijkstra_route(
+    layer_ranges: Dict[str, Tuple[int, int]],
+    num_layers: int,
+    latency: Optional[Dict[Tuple[str, str], float]] = None,
+) -> Optional[List[str]]:
+    """Schedulerless routing (reference p2p/server.py:592-626): given the
+    layer ranges each peer announced over the DHT, find the cheapest chain of
+    peers whose ranges concatenate to [0, num_layers). Edges connect node A
+    (covering [s, m)) to node B covering [m, e); edge cost = RTT estimate
+    (default 1 per hop, so min-hop). Returns node ids in pipeline order, or
+    None if the layer space cannot be covered."""
+    import heapq
+
+    starts: Dict[int, List[str]] = {}
+    for nid, (s, e) in layer_ranges.items():
+        if e > s:
+            starts.setdefault(s, []).append(nid)
+    # state = layer index reached; start = 0, goal = num_layers
+    best: Dict[int, float] = {0: 0.0}
+    prev: Dict[int, Tuple[int, str]] = {}
+    heap = [(0.0, 0, None)]
+    while heap:
+        cost, layer, from_nid = heapq.heappop(heap)
+        if layer >= num_layers:
+            break
+        if cost > best.get(layer, float("inf")):
+            continue
+        for nid in starts.get(layer, []):
+            _, e = layer_ranges[nid]
+            hop = 1.0
+            if latency is not None and from_nid is not None:
+                hop = latency.get((from_nid, nid), 1.0)
+            nxt = min(e, num_layers)
+            c = cost + hop
+            if c < best.get(nxt, float("inf")):
+                best[nxt] = c
+                prev[nxt] = (layer, nid)
+                heapq.heappush(heap, (c, nxt, nid))
+    if num_layers not in prev and num_layers not in best:
+        return None
+    if best.get(num_layers) is None:
+        return None
+    path = []
+    at = num_layers
+    while at != 0:
+        layer, nid = prev[at]
+        path.append(nid)
+        at = layer
+    return list(reversed(path))

@@ -137,7 +137,10 @@ class Engine:
                     cfg.linear_num_value_heads, cfg.linear_key_head_dim,
                     cfg.linear_value_head_dim,
                 ),
-                num_slots=args.max_batch_size + 8,
+                # +1: the last slot is a scratch slot for graph-capture pad
+                # rows (DecodeGraphRunner.linear_scratch_slot); the cache
+                # manager only hands out the first max_batch_size + 8
+                num_slots=args.max_batch_size + 9,
                 device=self.device,
                 dtype=args.dtype if args.dtype != torch.float32 else torch.float32,
             )
@@ -206,6 +209,8 @@ class Engine:
                 is_first_stage=self.comm.is_first_stage,
                 is_last_stage=self.comm.is_last_stage,
                 is_mla=self.is_mla,
+                linear_cache=self.linear_cache,
+                linear_scratch_slot=args.max_batch_size + 8,
             )
         self._pending_adds: List[InitialRequest] = []
         self._pending_aborts: List[str] = []
@@ -450,9 +455,15 @@ class Engine:
             hidden_in = comm.pp_recv(
                 (len(reqs), self.cfg.hidden_size), self.args.dtype, comm.pp_rank - 1
             )
+        lin_slots = None
+        if self.linear_cache is not None:
+            lin_slots = [
+                self.cache_manager.get(r.rid).linear_slot or 0 for r in reqs
+            ]
         out = self.graph_runner.run(input_ids, positions, slots, btabs, seq_lens,
                                     hidden_in=hidden_in,
-                                    rids=[r.rid for r in reqs])
+                                    rids=[r.rid for r in reqs],
+                                    linear_slots=lin_slots)
         if not comm.is_last_stage:
             comm.pp_send(out, comm.pp_rank + 1)
             return None

@@ -49,8 +49,15 @@ class DecodeGraphRunner:
         is_last_stage: bool,
         buckets: Optional[List[int]] = None,
         is_mla: bool = False,
+        linear_cache=None,
+        linear_scratch_slot: int = 0,
     ):
         self.is_mla = is_mla
+        # hybrid stacks (deltanet/lightning): per-request state slots; pad
+        # rows write into a dedicated scratch slot so capture/warmup replays
+        # never corrupt live state
+        self.linear_cache = linear_cache
+        self.linear_scratch_slot = linear_scratch_slot
         self.model = model
         self.kv_cache = kv_cache
         self.device = device
@@ -79,6 +86,11 @@ class DecodeGraphRunner:
         self.h_block_tables = torch.zeros(B, self.max_blocks, dtype=torch.int32,
                                           pin_memory=True)
         self.h_seq_lens = torch.ones(B, dtype=torch.int32, pin_memory=True)
+        if linear_cache is not None:
+            self.linear_slots = torch.full((B,), linear_scratch_slot,
+                                           dtype=torch.int64, device=dev)
+            self.h_linear_slots = torch.full((B,), linear_scratch_slot,
+                                             dtype=torch.int64, pin_memory=True)
 
         self.ctx_buckets: List[int] = []
         c = 1024
@@ -106,6 +118,9 @@ class DecodeGraphRunner:
             mla_cache=self.kv_cache if self.is_mla else None,
             logits_indices=None,
             max_seq_len=ctx_bucket or self.max_model_len,
+            linear_cache=self.linear_cache,
+            linear_slots=self.linear_slots[:bucket]
+            if self.linear_cache is not None else None,
         )
 
     def _forward(self, bucket: int, ctx_bucket: Optional[int] = None) -> torch.Tensor:
@@ -129,6 +144,8 @@ class DecodeGraphRunner:
         self.slot_mapping.fill_(-1)
         self.block_tables.zero_()
         self.seq_lens.fill_(1)
+        if self.linear_cache is not None:
+            self.linear_slots.fill_(self.linear_scratch_slot)
         torch.cuda.synchronize()
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
@@ -166,6 +183,7 @@ class DecodeGraphRunner:
         seq_lens: List[int],
         hidden_in: Optional[torch.Tensor] = None,
         rids: Optional[List[str]] = None,
+        linear_slots: Optional[List[int]] = None,
     ) -> torch.Tensor:
         """Returns logits[:B] (last stage) or hidden[:B] (other stages)."""
         B = len(seq_lens)
@@ -209,6 +227,13 @@ class DecodeGraphRunner:
         self.positions[:bucket].copy_(self.h_positions[:bucket], non_blocking=True)
         self.slot_mapping[:bucket].copy_(self.h_slot_mapping[:bucket], non_blocking=True)
         self.seq_lens[:bucket].copy_(self.h_seq_lens[:bucket], non_blocking=True)
+        if self.linear_cache is not None:
+            self.h_linear_slots[:B] = torch.tensor(
+                linear_slots or [self.linear_scratch_slot] * B, dtype=torch.int64
+            )
+            self.h_linear_slots[B:bucket] = self.linear_scratch_slot
+            self.linear_slots[:bucket].copy_(self.h_linear_slots[:bucket],
+                                             non_blocking=True)
         if hidden_in is not None:
             self.hidden_in[:B].copy_(hidden_in)
 

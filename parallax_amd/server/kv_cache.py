@@ -60,9 +60,10 @@ class PagedKVCache:
         ]
         idx_dtype = spec.dtype if spec.dtype != torch.float8_e4m3fn \
             else torch.bfloat16
+        # +1 trash block for graph-safe pad-token stores (see MLAKVCache)
         self.index_caches: List[torch.Tensor] = (
             [
-                torch.zeros((num_blocks, spec.block_size, spec.index_dim),
+                torch.zeros((num_blocks + 1, spec.block_size, spec.index_dim),
                             dtype=idx_dtype, device=device)
                 for _ in range(spec.num_layers)
             ]
@@ -109,9 +110,13 @@ class MLAKVCache:
         # DSA (DeepSeek-V3.2): one indexer key per token per layer, shared
         # across indexer query heads (reference dsa_cache.py)
         self.index_dim = index_dim
+        # +1 trash block at the end: graph-safe indexer-cache stores write
+        # pad tokens (slot -1) there instead of branching (see
+        # ops.reference.store_indexer_cache)
         self.index_caches: List[torch.Tensor] = (
             [
-                torch.zeros((num_blocks, block_size, index_dim), dtype=dtype, device=device)
+                torch.zeros((num_blocks + 1, block_size, index_dim),
+                            dtype=dtype, device=device)
                 for _ in range(num_layers)
             ]
             if index_dim > 0

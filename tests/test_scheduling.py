@@ -209,3 +209,28 @@ def test_roofline_latency_sane():
     lat = n.node_latency_ms(model)
     # 70B decode on one MI355X: ~>1 ms, < 1 s
     assert 0.5 < lat < 1000.0
+
+
+def test_dijkstra_route_min_hop():
+    from parallax_amd.scheduling.request_routing import dijkstra_route
+
+    ranges = {
+        "a": (0, 16), "b": (16, 32), "c": (0, 32),
+        "d": (0, 8), "e": (8, 32),
+    }
+    # single node covering everything wins (1 hop)
+    assert dijkstra_route(ranges, 32) == ["c"]
+    # without c: 2-hop chains; either (a,b) or (d,e)
+    del ranges["c"]
+    path = dijkstra_route(ranges, 32)
+    assert path in (["a", "b"], ["d", "e"])
+    # unreachable coverage
+    assert dijkstra_route({"a": (0, 16), "b": (20, 32)}, 32) is None
+
+
+def test_dijkstra_route_latency_weighted():
+    from parallax_amd.scheduling.request_routing import dijkstra_route
+
+    ranges = {"a": (0, 16), "b": (16, 32), "d": (0, 16), "e": (16, 32)}
+    lat = {("a", "b"): 10.0, ("a", "e"): 0.1, ("d", "e"): 10.0, ("d", "b"): 10.0}
+    assert dijkstra_route(ranges, 32, latency=lat) == ["a", "e"]
