@@ -39,7 +39,8 @@ def test_deepseek_v32_engine_gpu():
         "moe_intermediate_size": 128, "n_routed_experts": 8,
         "num_experts_per_tok": 2, "n_shared_experts": 1, "n_group": 2,
         "topk_group": 1, "first_k_dense_replace": 1, "q_lora_rank": 96,
-        "kv_lora_rank": 64, "qk_nope_head_dim": 64, "qk_rope_head_dim": 32,
+        # the MLA HIP kernel is compiled for the production dims (R=512 DR=64)
+        "kv_lora_rank": 512, "qk_nope_head_dim": 64, "qk_rope_head_dim": 64,
         "v_head_dim": 64, "index_n_heads": 4, "index_head_dim": 64,
         "index_topk": 8, "max_position_embeddings": 2048,
         "rope_theta": 10000.0, "eos_token_id": None,
