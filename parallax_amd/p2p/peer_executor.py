@@ -90,16 +90,45 @@ class PeerExecutor:
                 end_layer - start_layer, cfg.kv_lora_rank, cfg.qk_rope_head_dim,
                 block_size, num_kv_blocks, self.device,
                 dtype if dtype != torch.float32 else torch.float32,
+                index_dim=cfg.index_head_dim if cfg.is_dsa else 0,
             )
         else:
             spec = KVCacheSpec(
                 num_layers=end_layer - start_layer, num_kv_heads=cfg.num_kv_heads,
                 head_dim=cfg.head_dim, block_size=block_size, dtype=dtype,
+                index_dim=cfg.index_head_dim if cfg.is_msa else 0,
             )
             self.kv_cache = PagedKVCache(spec, num_kv_blocks, self.device)
+        # hybrid stacks: per-request conv/recurrent state slots
+        self.linear_cache = None
+        num_linear_slots = 0
+        if cfg.has_linear_layers:
+            from ..server.kv_cache import LinearStateCache
+
+            local_types = [cfg.layer_type(g) for g in range(start_layer, end_layer)]
+            n_linear = sum(1 for t in local_types if t == "linear_attention")
+            conv_dim = (
+                2 * cfg.linear_num_key_heads * cfg.linear_key_head_dim
+                + cfg.linear_num_value_heads * cfg.linear_value_head_dim
+            )
+            num_linear_slots = 64
+            self.linear_cache = LinearStateCache(
+                max(1, n_linear),
+                conv_state_shape=(conv_dim, cfg.linear_conv_kernel_dim - 1),
+                recurrent_state_shape=(
+                    cfg.linear_num_value_heads, cfg.linear_key_head_dim,
+                    cfg.linear_value_head_dim,
+                ),
+                num_slots=num_linear_slots + 1,
+                device=self.device,
+                dtype=dtype if dtype != torch.float32 else torch.float32,
+            )
         # non-head peers never see token content: radix prefix match disabled
+        # (and linear-state slots do not compose with prefix reuse)
         self.cache_manager = CacheManager(
-            block_size, num_kv_blocks, enable_prefix_cache=self.is_head
+            block_size, num_kv_blocks,
+            enable_prefix_cache=self.is_head and not cfg.has_linear_layers,
+            num_linear_slots=num_linear_slots,
         )
         self.scheduler = (
             Scheduler(self.cache_manager, eos_token_ids=cfg.eos_token_ids)
@@ -228,7 +257,8 @@ class PeerExecutor:
             btabs.append(state.block_table)
             seq_lens.append(c.start + c.num_tokens)
             qlens.append(c.num_tokens)
-        meta = self._meta(True, positions, slots, btabs, seq_lens, qlens)
+        meta = self._meta(True, positions, slots, btabs, seq_lens, qlens,
+                          rids=[c.req.rid for c in chunks])
         return meta, torch.tensor(input_ids, dtype=torch.long, device=self.device), None
 
     def _head_decode_meta(self, reqs):
@@ -242,7 +272,8 @@ class PeerExecutor:
                 state.block_table, pos, 1, self.block_size))
             btabs.append(state.block_table)
             seq_lens.append(r.total_len)
-        meta = self._meta(False, positions, slots, btabs, seq_lens, None)
+        meta = self._meta(False, positions, slots, btabs, seq_lens, None,
+                          rids=[r.rid for r in reqs])
         return meta, torch.tensor(input_ids, dtype=torch.long, device=self.device)
 
     # -- non-head -------------------------------------------------------------------------
@@ -297,7 +328,8 @@ class PeerExecutor:
                 if is_prefill and p.input_ids:
                     ctx[1].extend(p.input_ids)
         meta = self._meta(is_prefill, positions, slots, btabs, seq_lens,
-                          qlens if is_prefill else None)
+                          qlens if is_prefill else None,
+                          rids=[p.rid for p in pkts])
         hidden = torch.cat(hiddens, dim=0)
         with torch.inference_mode():
             hidden = self.model(hidden, meta)
@@ -353,8 +385,15 @@ class PeerExecutor:
 
     # -- shared ---------------------------------------------------------------------------
 
-    def _meta(self, is_prefill, positions, slots, btabs, seq_lens, qlens) -> ForwardMeta:
+    def _meta(self, is_prefill, positions, slots, btabs, seq_lens, qlens,
+              rids=None) -> ForwardMeta:
         dev = self.device
+        linear_slots = None
+        if self.linear_cache is not None and rids is not None:
+            linear_slots = torch.tensor(
+                [self.cache_manager.get(r).linear_slot or 0 for r in rids],
+                dtype=torch.int64, device=dev,
+            )
         return ForwardMeta(
             is_prefill=is_prefill,
             positions=torch.tensor(positions, dtype=torch.int32, device=dev),
@@ -365,6 +404,8 @@ class PeerExecutor:
             if qlens else None,
             kv_cache=None if self.is_mla else self.kv_cache,
             mla_cache=self.kv_cache if self.is_mla else None,
+            linear_cache=self.linear_cache,
+            linear_slots=linear_slots,
             max_seq_len=max(seq_lens),
         )
 

@@ -98,3 +98,82 @@ def test_tcp_transport_roundtrip():
     assert a.recv(timeout=5) == b"pong"
     a.close()
     b.close()
+
+
+def _run_p2p(cfg, sd, n_stages, prompts, n_new=5):
+    registry = {}
+    peer_ids = [f"peer{i}" for i in range(n_stages)]
+    base, rem = divmod(cfg.num_layers, n_stages)
+    spans, pos = [], 0
+    for i in range(n_stages):
+        n = base + (1 if i < rem else 0)
+        spans.append((pos, pos + n))
+        pos += n
+    peers = []
+    for pid, (s, e) in zip(peer_ids, spans):
+        t = LoopbackTransport(pid, registry)
+        px = PeerExecutor(cfg, s, e, pid, t, dtype=torch.float32,
+                          num_kv_blocks=128, block_size=8)
+        for name, w in sd.items():
+            px.model.load_hf_weight(name, w)
+        if hasattr(px.model, "finalize_weights"):
+            px.model.finalize_weights()
+        peers.append(px)
+    head = peers[0]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=n_new, ignore_eos=True)
+    rids = [head.submit(p, sp, peer_ids) for p in prompts]
+    outputs = {rid: [] for rid in rids}
+    done = set()
+    for _ in range(3000):
+        for px in peers:
+            px.step(recv_timeout=0.001)
+        for out in head.drain_outputs():
+            outputs[out.rid].append(out.token_id)
+            if out.finished:
+                done.add(out.rid)
+        if len(done) == len(rids):
+            break
+    return [outputs[r] for r in rids]
+
+
+def test_p2p_hybrid_linear_stack():
+    """Qwen3-Next-style hybrid (deltanet + attention) over 2 peers: linear
+    state slots live per peer and must reproduce the single-engine output."""
+    import transformers
+
+    from parallax_amd.models import get_model_class
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.server.engine import Engine, EngineArgs
+
+    torch.manual_seed(41)
+    hf_cfg = transformers.Qwen3NextConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        moe_intermediate_size=32, shared_expert_intermediate_size=32,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=32, num_experts=4, num_experts_per_tok=2, norm_topk_prob=True,
+        decoder_sparse_step=1, mlp_only_layers=[],
+        linear_num_key_heads=2, linear_num_value_heads=4,
+        linear_key_head_dim=16, linear_value_head_dim=16,
+        linear_conv_kernel_dim=4,
+        layer_types=["linear_attention", "full_attention"] * 2,
+        max_position_embeddings=512, tie_word_embeddings=False,
+        rope_parameters={"rope_type": "default", "rope_theta": 10000.0,
+                         "partial_rotary_factor": 0.25},
+    )
+    hf = transformers.Qwen3NextForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["Qwen3NextForCausalLM"]}
+    )
+    prompts = [[7, 42, 99, 5, 81, 23], [3, 9, 27]]
+    sd = dict(hf.state_dict())
+    # single-engine baseline
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=torch.float32, max_batch_size=8))
+    m = get_model_class(cfg.architecture)(cfg).eval()
+    for name, w in sd.items():
+        m.load_hf_weight(name, w)
+    eng.model = m.float()
+    sp = SamplingParams(temperature=0.0, max_new_tokens=5, ignore_eos=True)
+    expect = list(eng.generate(prompts, [sp] * 2).values())
+    got = _run_p2p(cfg, sd, 2, prompts)
+    assert got == expect
