@@ -371,6 +371,7 @@ def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
 # decode shapes when measured in isolation (scripts/bench_gemm.py) and beats
 # this kernel's simple 2-barrier structure; opt in for experiments.
 SKINNY_GEMM_MAX_M = int(os.environ.get("PARALLAX_SKINNY_GEMM_MAX_M", "0"))
+LT_GEMM = os.environ.get("PARALLAX_LT_GEMM", "1") != "0"
 
 
 def linear(
@@ -394,6 +395,22 @@ def linear(
         if ext.skinny_gemm(out, x, weight,
                            bias if bias is not None else x.new_empty(0)):
             return out
+    if (
+        LT_GEMM
+        and x.is_cuda
+        and x.dim() == 2
+        and x.dtype == torch.bfloat16
+        and weight.dtype == torch.bfloat16
+        and x.is_contiguous()
+        and weight.is_contiguous()
+    ):
+        # tuned hipBLASLt: the default heuristic's decode-shape tile picks
+        # are ~2.5x off the weight-streaming roofline (profiles/README.md);
+        # first call per shape times the heuristic candidates (engine warmup,
+        # outside graph capture) and later calls replay the winner
+        ext = _require_ext("lt_linear")
+        y = ext.lt_linear(x, weight)
+        return y if bias is None else y + bias
     return torch.nn.functional.linear(x, weight, bias)
 
 

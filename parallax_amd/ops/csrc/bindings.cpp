@@ -461,7 +461,11 @@ void msa_paged_attention_decode(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(launched, "no MSA kernel for D=", D, " BS=", BS, " G=", Hq / Hk);
 }
 
+torch::Tensor lt_linear(torch::Tensor x, torch::Tensor w);  // lt_gemm.cpp
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("lt_linear", &lt_linear,
+        "tuned hipBLASLt y = x @ w^T (bf16, fp32 accum)");
   m.def("msa_paged_attention_decode", &msa_paged_attention_decode);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("moe_forward", &moe_forward);

@@ -18,6 +18,7 @@ CSRC = os.path.join(ROOT, "parallax_amd", "ops", "csrc")
 
 sources = [
     os.path.join(CSRC, "bindings.cpp"),
+    os.path.join(CSRC, "lt_gemm.cpp"),
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "paged_attention.hip"),
     os.path.join(CSRC, "paged_attention_mfma.hip"),
@@ -34,6 +35,7 @@ setup(
         CUDAExtension(
             name="parallax_amd.ops._C",
             sources=sources,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": [

@@ -343,3 +343,21 @@ def test_skinny_gemm_strided_rows():
     out = ops.linear(x, w)
     expect = torch.nn.functional.linear(x.float(), w.float())
     torch.testing.assert_close(out.float(), expect, atol=8e-2, rtol=8e-2)
+
+
+@pytest.mark.parametrize("M,N,K", [
+    (1, 6144, 4096), (64, 4096, 4096), (128, 57344, 4096),
+    (128, 4096, 14336), (200, 512, 256), (16384, 6144, 4096),
+])
+def test_lt_linear_tuned(M, N, K):
+    """Tuned hipBLASLt path: numerics vs fp32 F.linear, and the tuned algo
+    must keep giving identical results on repeat calls (graph determinism)."""
+    torch.manual_seed(0)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") / 8
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / 8
+    ext = ops._require_ext("lt_linear")
+    y = ext.lt_linear(x, w)
+    ref = torch.nn.functional.linear(x.float(), w.float())
+    torch.testing.assert_close(y.float(), ref, atol=2e-1, rtol=2e-2)
+    y2 = ext.lt_linear(x, w)
+    assert torch.equal(y, y2)
