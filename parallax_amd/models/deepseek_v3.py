@@ -83,13 +83,14 @@ class MLAAttention(nn.Module):
         T = x.shape[0]
         H = self.num_heads
         if self.q_lora_rank:
-            q = self.q_b_proj(self.q_a_layernorm(self.q_a_proj(x)))
+            q = self.q_b_proj(self.q_a_layernorm(
+                ops.linear(x, self.q_a_proj.weight)))
         else:
             q = self.q_proj(x)
         q = q.view(T, H, self.dn + self.dr)
         q_nope, q_pe = q[..., : self.dn], q[..., self.dn :]
 
-        kv_a = self.kv_a_proj_with_mqa(x)
+        kv_a = ops.linear(x, self.kv_a_proj_with_mqa.weight)
         latent = self.kv_a_layernorm(kv_a[:, : self.r])
         k_pe = kv_a[:, self.r :].unsqueeze(1)  # [T, 1, dr]
 

@@ -217,7 +217,9 @@ class LlamaShardModel(nn.Module):
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         assert self.is_last, "only the last shard computes logits"
-        return self.lm_head(hidden.to(self.lm_head.weight.dtype)).float()
+        w = self.lm_head.weight
+        return ops.linear(hidden.to(w.dtype).contiguous(), w,
+                          self.lm_head.bias).float()
 
     # -- weight loading ---------------------------------------------------------
 
