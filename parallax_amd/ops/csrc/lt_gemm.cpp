@@ -22,6 +22,7 @@
 
 #include <torch/extension.h>
 
+#include <algorithm>
 #include <map>
 #include <mutex>
 #include <tuple>
@@ -125,19 +126,29 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
   hipEvent_t ev0, ev1;
   C10_HIP_CHECK(hipEventCreate(&ev0));
   C10_HIP_CHECK(hipEventCreate(&ev1));
-  int best = -1;
-  float best_ms = 1e30f;
-  for (int i = 0; i < n_results; ++i) {
-    if (run(results[i].algo) != HIPBLAS_STATUS_SUCCESS) continue;  // warm/verify
+  auto time_algo = [&](int i, int iters) -> float {
+    if (run(results[i].algo) != HIPBLAS_STATUS_SUCCESS) return 1e30f;  // warm
     C10_HIP_CHECK(hipEventRecord(ev0, stream));
-    for (int it = 0; it < 8; ++it) (void)run(results[i].algo);
+    for (int it = 0; it < iters; ++it) (void)run(results[i].algo);
     C10_HIP_CHECK(hipEventRecord(ev1, stream));
     C10_HIP_CHECK(hipEventSynchronize(ev1));
-    float ms = 0.f;
+    float ms = 1e30f;
     C10_HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+    return ms;
+  };
+  // screen every candidate with one timed iteration, then re-time the best
+  // few — keeps one-off tuning of huge prefill shapes to ~tens of ms
+  std::vector<std::pair<float, int>> screened;
+  for (int i = 0; i < n_results; ++i) screened.emplace_back(time_algo(i, 1), i);
+  std::sort(screened.begin(), screened.end());
+  int best = -1;
+  float best_ms = 1e30f;
+  for (int r = 0; r < std::min<int>(4, (int)screened.size()); ++r) {
+    if (screened[r].first >= 1e30f) continue;
+    float ms = time_algo(screened[r].second, 6);
     if (ms < best_ms) {
       best_ms = ms;
-      best = i;
+      best = screened[r].second;
     }
   }
   C10_HIP_CHECK(hipEventDestroy(ev0));
