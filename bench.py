@@ -138,20 +138,41 @@ def main():
     ap.add_argument("--micro-batches", type=int, default=0,
                     help="0 = one per pipeline stage")
     ap.add_argument("--kv-dtype", default="auto", choices=["auto", "fp8"])
+    ap.add_argument("--parallelism", default="auto", choices=["auto", "dp", "pp"],
+                    help="auto: DP replicas when the model fits one GPU "
+                         "(288 GB HBM3E), PP layer split otherwise")
     ap.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
     args = ap.parse_args()
 
     from parallax_amd.models.config import ModelConfig
-    from parallax_amd.parallel.comm import init_distributed
+    from parallax_amd.parallel.comm import CommContext, init_distributed
     from parallax_amd.server.engine import Engine, EngineArgs
     from parallax_amd.server.sampling_params import SamplingParams
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     n_gpus = args.gpus if world == 1 else world
-    comm = init_distributed(pp_size=world, tp_size=1)
 
     cfg_fn, model_name = MODELS[args.model]
     cfg = ModelConfig.from_hf_config(cfg_fn())
+
+    # parallelism: the scheduler serves replicas (DP) whenever a whole model
+    # fits one 288 GB GPU — that is the reference's multi-pipeline deployment
+    # for the PP=1 configs — and splits layers (PP) only when it must
+    mode = args.parallelism
+    if mode == "auto":
+        params_gb = (
+            cfg.num_layers * cfg.per_layer_param_bytes()
+            + cfg.embedding_bytes() + cfg.lm_head_bytes()
+        ) / 1e9
+        mode = "dp" if (world == 1 or params_gb < 190.0) else "pp"
+    comm_world = init_distributed(pp_size=world, tp_size=1)
+    if mode == "dp" and world > 1:
+        # independent engine per rank; torch.distributed only for barriers
+        comm = CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                           pp_rank=0, tp_rank=0, device=comm_world.device)
+    else:
+        mode = "pp" if world > 1 else mode
+        comm = comm_world
     if args.cpu:
         cfg.num_layers = 2
         cfg.vocab_size = 1024
@@ -161,12 +182,13 @@ def main():
 
     use_gpu = torch.cuda.is_available() and not args.cpu
     global_batch = args.batch_per_gpu * world
+    engine_batch = args.batch_per_gpu if (mode == "dp") else global_batch
     total_new_tokens = args.warmup + args.steps + 8
     eargs = EngineArgs(
         block_size=args.block_size,
-        max_batch_size=max(128, global_batch),
-        max_num_tokens_per_batch=max(16384, global_batch * 2),
-        micro_batches=args.micro_batches or world,
+        max_batch_size=max(128, engine_batch),
+        max_num_tokens_per_batch=max(16384, engine_batch * 2),
+        micro_batches=args.micro_batches or (1 if mode == "dp" else world),
         dtype=torch.bfloat16 if use_gpu else torch.float32,
         num_kv_blocks=None if use_gpu else 4096,
         enable_prefix_cache=False,  # unique synthetic prompts; skip radix overhead
@@ -175,21 +197,22 @@ def main():
     )
     engine = Engine(cfg, eargs, comm=comm, random_weights=True)
 
-    # synthetic prompts, unique tokens so nothing prefix-shares
-    g = torch.Generator().manual_seed(1234 + 7)
+    # synthetic prompts, unique tokens so nothing prefix-shares (per-replica
+    # seeds in DP mode so replicas do not share content)
+    g = torch.Generator().manual_seed(1234 + 7 + comm_world.rank * 1000)
     sp = SamplingParams(
         temperature=1.0, top_p=1.0, top_k=-1,
         max_new_tokens=total_new_tokens, ignore_eos=True,
     )
     if comm.rank == 0:
-        for i in range(global_batch):
+        for i in range(engine_batch):
             prompt = torch.randint(0, cfg.vocab_size, (args.prompt_len,), generator=g).tolist()
             engine.submit(prompt, sp)
 
     def sync():
         if use_gpu:
             torch.cuda.synchronize()
-        comm.barrier()
+        comm_world.barrier()
 
     # prefill everything (untimed; ends when all requests are decoding)
     t_prefill0 = time.perf_counter()
@@ -210,23 +233,23 @@ def main():
     t0 = time.perf_counter()
     for _ in range(args.steps):
         outs = engine.step()
-        assert len(outs) == global_batch, f"batch shrank: {len(outs)}"
+        assert len(outs) == engine_batch, f"batch shrank: {len(outs)}"
     sync()
     elapsed = time.perf_counter() - t0
 
     # max over ranks
-    if comm.world_size > 1:
+    if comm_world.world_size > 1:
         import torch.distributed as dist
 
         t = torch.tensor([elapsed], dtype=torch.float64)
-        if comm.device.type == "cuda":
-            t = t.to(comm.device)
+        if comm_world.device.type == "cuda":
+            t = t.to(comm_world.device)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
     tokens = global_batch * args.steps
     value = tokens / elapsed
-    if comm.rank == 0:
+    if comm_world.rank == 0:
         result = {
             "metric": "output_tokens_per_sec",
             "value": round(value, 2),
@@ -244,7 +267,8 @@ def main():
                 "model": model_name,
                 "global_batch": global_batch,
                 "seq_len": args.prompt_len,
-                "parallelism": f"pp{world}",
+                "parallelism": f"{mode if world > 1 else 'pp'}{world}"
+                if world > 1 else "pp1",
                 "kv_dtype": args.kv_dtype,
                 "micro_batches": eargs.micro_batches,
                 "prefill_s": round(prefill_s, 3),
