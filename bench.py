@@ -132,9 +132,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="deepseek-r1-distill-llama-8b", choices=MODELS)
     ap.add_argument("--batch-per-gpu", type=int, default=512,
-                    help="decode batch per GPU (global batch = N * this); 512
-                    is the measured throughput knee (21 ms/step = 47 tok/s per
-                    stream on the flagship, see profiles/README.md)")
+                    help="decode batch per GPU (global batch = N * this); "
+                         "512 is the measured throughput knee (21 ms/step = "
+                         "47 tok/s per stream, see profiles/README.md)")
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--block-size", type=int, default=32)
     ap.add_argument("--micro-batches", type=int, default=0,
