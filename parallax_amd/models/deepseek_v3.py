@@ -248,18 +248,18 @@ class DeepseekV3ShardModel(LlamaShardModel):
             return True
         if parts[0] == "experts":
             if parts[1] == "gate_up_proj":  # fused [E, 2I, H] (transformers >= 5)
-                moe.experts.w_gate_up.data.copy_(t)
+                moe.experts.load_fused_gate_up(t)
                 return True
             if parts[1] == "down_proj":     # fused [E, H, I]
-                moe.experts.w_down.data.copy_(t)
+                moe.experts.load_fused_down(t)
                 return True
             e, proj = int(parts[1]), parts[2]
             if proj == "gate_proj":
-                moe.experts.w_gate_up.data[e, :inter].copy_(t)
+                moe.experts.load_expert_gate(e, t)
             elif proj == "up_proj":
-                moe.experts.w_gate_up.data[e, inter:].copy_(t)
+                moe.experts.load_expert_up(e, t)
             elif proj == "down_proj":
-                moe.experts.w_down.data[e].copy_(t)
+                moe.experts.load_expert_down(e, t)
             return True
         if parts[0] == "shared_experts":
             proj = parts[1]

@@ -217,9 +217,11 @@ class LlamaShardModel(nn.Module):
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         assert self.is_last, "only the last shard computes logits"
-        w = self.lm_head.weight
-        return ops.linear(hidden.to(w.dtype).contiguous(), w,
-                          self.lm_head.bias).float()
+        # lm_head is a ColumnParallelLinear (gathers under TP) and already
+        # routes through ops.linear -> the tuned hipBLASLt path
+        return self.lm_head(
+            hidden.to(self.lm_head.weight.dtype).contiguous()
+        ).float()
 
     # -- weight loading ---------------------------------------------------------
 

@@ -215,19 +215,18 @@ class MiniMaxShardModel(LlamaShardModel):
                 return True
             if parts[3] == "experts":
                 if parts[4] == "gate_up_proj":   # fused [E, 2I, H]
-                    layer.mlp.experts.w_gate_up.data.copy_(t)
+                    layer.mlp.experts.load_fused_gate_up(t)
                     return True
                 if parts[4] == "down_proj":      # fused [E, H, I]
-                    layer.mlp.experts.w_down.data.copy_(t)
+                    layer.mlp.experts.load_fused_down(t)
                     return True
-                inter = layer.mlp.experts.intermediate_size
                 e, proj = int(parts[4]), parts[5]
                 if proj in ("gate_proj", "w1"):
-                    layer.mlp.experts.w_gate_up.data[e, :inter].copy_(t)
+                    layer.mlp.experts.load_expert_gate(e, t)
                 elif proj in ("up_proj", "w3"):
-                    layer.mlp.experts.w_gate_up.data[e, inter:].copy_(t)
+                    layer.mlp.experts.load_expert_up(e, t)
                 elif proj in ("down_proj", "w2"):
-                    layer.mlp.experts.w_down.data[e].copy_(t)
+                    layer.mlp.experts.load_expert_down(e, t)
                 else:
                     return False
                 return True

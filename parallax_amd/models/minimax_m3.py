@@ -297,9 +297,9 @@ class MiniMaxM3ShardModel(LlamaShardModel):
                     return True
                 if parts[3] == "experts":
                     if parts[4] == "gate_up_proj":
-                        mlp.experts.w_gate_up.data.copy_(t)
+                        mlp.experts.load_fused_gate_up(t)
                     elif parts[4] == "down_proj":
-                        mlp.experts.w_down.data.copy_(t)
+                        mlp.experts.load_fused_down(t)
                     else:
                         return False
                     return True

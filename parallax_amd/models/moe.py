@@ -16,6 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import ops
+from ..parallel.comm import get_comm
 from .config import ModelConfig
 
 
@@ -74,7 +75,15 @@ class MoERouter(nn.Module):
 class FusedMoE(nn.Module):
     """Routed experts with fused gate_up/down weights stored as stacked 3-D
     tensors [E, 2I, H] / [E, H, I] — the exact layout the MFMA grouped-GEMM
-    kernel consumes."""
+    kernel consumes.
+
+    Expert parallelism: when the engine runs TP > 1 and the expert count
+    divides, experts are SHARDED across the TP group (each rank holds
+    E/tp_size experts, computes its local experts' contributions for the
+    whole batch, and the partial outputs all-reduce) — a real EP
+    implementation where the reference only plumbs moe_ep_size=1
+    (sglang/model_runner.py:65-66). Non-divisible counts fall back to
+    replication."""
 
     def __init__(self, cfg: ModelConfig, intermediate_size: Optional[int] = None,
                  act_limit: float = 0.0):
@@ -87,18 +96,70 @@ class FusedMoE(nn.Module):
         inter = intermediate_size or cfg.moe_intermediate_size or cfg.intermediate_size
         self.intermediate_size = inter
         E, H = cfg.num_experts, cfg.hidden_size
+        comm = get_comm()
+        self._comm = comm
+        self.ep_size = comm.tp_size if (comm.tp_size > 1 and E % comm.tp_size == 0) else 1
+        self.ep_rank = comm.tp_rank if self.ep_size > 1 else 0
+        self.num_local_experts = E // self.ep_size
+        self.expert_offset = self.ep_rank * self.num_local_experts
         self.router = MoERouter(cfg)
-        self.w_gate_up = nn.Parameter(torch.empty(E, 2 * inter, H), requires_grad=False)
-        self.w_down = nn.Parameter(torch.empty(E, H, inter), requires_grad=False)
+        self.w_gate_up = nn.Parameter(
+            torch.empty(self.num_local_experts, 2 * inter, H), requires_grad=False
+        )
+        self.w_down = nn.Parameter(
+            torch.empty(self.num_local_experts, H, inter), requires_grad=False
+        )
+
+    # -- EP-aware weight loading (used by every family's loader) ----------------
+
+    def local_expert(self, e: int) -> Optional[int]:
+        l = e - self.expert_offset
+        return l if 0 <= l < self.num_local_experts else None
+
+    def load_expert_gate(self, e: int, t: torch.Tensor) -> None:
+        l = self.local_expert(e)
+        if l is not None:
+            self.w_gate_up.data[l, : self.intermediate_size].copy_(t)
+
+    def load_expert_up(self, e: int, t: torch.Tensor) -> None:
+        l = self.local_expert(e)
+        if l is not None:
+            self.w_gate_up.data[l, self.intermediate_size :].copy_(t)
+
+    def load_expert_down(self, e: int, t: torch.Tensor) -> None:
+        l = self.local_expert(e)
+        if l is not None:
+            self.w_down.data[l].copy_(t)
+
+    def load_fused_gate_up(self, t: torch.Tensor) -> None:
+        o = self.expert_offset
+        self.w_gate_up.data.copy_(t[o : o + self.num_local_experts])
+
+    def load_fused_down(self, t: torch.Tensor) -> None:
+        o = self.expert_offset
+        self.w_down.data.copy_(t[o : o + self.num_local_experts])
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
-        topk_ids, topk_w = self.router(x)           # [T,k]
+        topk_ids, topk_w = self.router(x)           # [T,k] global expert ids
+        if self.ep_size > 1:
+            # remap to local ids; foreign experts get the sentinel id
+            # num_local_experts, which the grouped-GEMM segment table and the
+            # CPU loop both skip — no wasted compute, partials all-reduce
+            local = topk_ids - self.expert_offset
+            valid = (local >= 0) & (local < self.num_local_experts)
+            topk_ids = torch.where(
+                valid, local, torch.full_like(local, self.num_local_experts)
+            )
+            topk_w = topk_w * valid
         if x.is_cuda:
-            return ops.fused_moe_forward(
+            out = ops.fused_moe_forward(
                 x, self.w_gate_up, self.w_down, topk_ids, topk_w,
                 limit=self.act_limit,
-            ).to(x.dtype)
+            )
+            if self.ep_size > 1:
+                out = self._comm.tp_all_reduce(out)
+            return out.to(x.dtype)
         out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
         flat_ids = topk_ids.reshape(-1)              # [T*k]
         flat_w = topk_w.reshape(-1)
@@ -107,6 +168,8 @@ class FusedMoE(nn.Module):
         )
         used_experts = torch.unique(flat_ids)
         for e in used_experts.tolist():
+            if e >= self.num_local_experts:
+                continue  # EP sentinel
             sel = (flat_ids == e).nonzero(as_tuple=True)[0]
             toks = token_idx[sel]
             xe = x[toks]                              # [n_e, H]
@@ -120,6 +183,8 @@ class FusedMoE(nn.Module):
                 h = ops.silu_and_mul(h)
             ye = F.linear(h, self.w_down[e]).float()  # [n_e, H]
             out.index_add_(0, toks, ye * flat_w[sel].unsqueeze(-1))
+        if self.ep_size > 1:
+            out = self._comm.tp_all_reduce(out)
         return out.to(x.dtype)
 
 

@@ -52,19 +52,19 @@ class Qwen3MoEShardModel(LlamaShardModel):
             return True
         if parts[3] == "experts":
             if parts[4] == "gate_up_proj":  # fused [E, 2I, H] (transformers >= 5)
-                layer.mlp.experts.w_gate_up.data.copy_(t)
+                layer.mlp.experts.load_fused_gate_up(t)
                 return True
             if parts[4] == "down_proj":
-                layer.mlp.experts.w_down.data.copy_(t)
+                layer.mlp.experts.load_fused_down(t)
                 return True
             e = int(parts[4])
             proj = parts[5]
             if proj == "gate_proj":
-                layer.mlp.experts.w_gate_up.data[e, :inter].copy_(t)
+                layer.mlp.experts.load_expert_gate(e, t)
             elif proj == "up_proj":
-                layer.mlp.experts.w_gate_up.data[e, inter:].copy_(t)
+                layer.mlp.experts.load_expert_up(e, t)
             elif proj == "down_proj":
-                layer.mlp.experts.w_down.data[e].copy_(t)
+                layer.mlp.experts.load_expert_down(e, t)
             return True
         if parts[3].startswith("shared_expert"):
             if parts[3] == "shared_expert_gate":

@@ -378,9 +378,9 @@ class Qwen3NextShardModel(LlamaShardModel):
             return True
         if parts[0] == "experts":
             if parts[1] == "gate_up_proj":
-                mlp.experts.w_gate_up.data.copy_(t)
+                mlp.experts.load_fused_gate_up(t)
             elif parts[1] == "down_proj":
-                mlp.experts.w_down.data.copy_(t)
+                mlp.experts.load_fused_down(t)
             else:
                 return False
             return True
