@@ -49,6 +49,7 @@ class StepOutput:
     token_id: int
     finished: bool
     finish_reason: Optional[str] = None
+    logprob: Optional[float] = None
 
 
 @dataclass
@@ -280,8 +281,8 @@ class Engine:
             self.scheduler.complete_prefill_chunk(chunk)
 
         if sample_reqs:
-            token_ids = self._sample_and_broadcast(logits_parts, sample_reqs)
-            for req, tok in zip(sample_reqs, token_ids):
+            sampled = self._sample_and_broadcast(logits_parts, sample_reqs)
+            for req, (tok, lp) in zip(sample_reqs, sampled):
                 finished = self.scheduler.commit_token(req.rid, tok)
                 outputs.append(
                     StepOutput(
@@ -289,6 +290,7 @@ class Engine:
                         token_id=tok,
                         finished=finished is not None,
                         finish_reason=req.status.finish_reason,
+                        logprob=lp,
                     )
                 )
         return outputs
@@ -489,23 +491,24 @@ class Engine:
 
     def _sample_and_broadcast(
         self, logits_parts: List[torch.Tensor], sample_reqs: List[InitialRequest]
-    ) -> List[int]:
+    ) -> List[tuple]:
+        """Returns [(token_id, logprob-or-None)] for each sampled request."""
         comm = self.comm
         if comm.pp_size == 1:
             logits = torch.cat(logits_parts, dim=0)
-            return self.sampler.sample(logits, sample_reqs)
-        token_ids: List[Optional[int]]
+            return self.sampler.sample_with_logprobs(logits, sample_reqs)
+        sampled: List[Optional[tuple]]
         if comm.is_last_stage and comm.tp_rank == 0:
             logits = torch.cat(logits_parts, dim=0)
-            token_ids = self.sampler.sample(logits, sample_reqs)
+            sampled = self.sampler.sample_with_logprobs(logits, sample_reqs)
         else:
-            token_ids = [None] * len(sample_reqs)
+            sampled = [None] * len(sample_reqs)
         if comm.world_size > 1:
-            payload = [token_ids]
+            payload = [sampled]
             src = comm.stage_rank(comm.pp_size - 1) - comm.tp_rank  # tp_rank 0 of last stage
             dist.broadcast_object_list(payload, src=src)
-            token_ids = payload[0]
-        return token_ids  # type: ignore[return-value]
+            sampled = payload[0]
+        return sampled  # type: ignore[return-value]
 
     # -- weight refit (runtime weight update, reference §3.5) ---------------------------------
 

@@ -41,15 +41,27 @@ def create_app(
         """Drain a request stream fully (non-streaming path)."""
         loop = asyncio.get_event_loop()
         token_ids: List[int] = []
+        logprobs: List[float] = []
         finish_reason = "stop"
         while True:
             out = await loop.run_in_executor(None, stream.out_queue.get)
             if out is None:
                 break
             token_ids.append(out.token_id)
+            if out.logprob is not None:
+                logprobs.append(out.logprob)
             if out.finished:
                 finish_reason = out.finish_reason or "stop"
-        return token_ids, finish_reason
+        return token_ids, finish_reason, logprobs
+
+    def _chat_logprobs(token_ids, logprobs):
+        """OpenAI chat logprobs block: one entry per sampled token."""
+        return {
+            "content": [
+                {"token": tokenizer.decode([t]), "logprob": lp}
+                for t, lp in zip(token_ids, logprobs)
+            ]
+        }
 
     @app.get("/health")
     async def health():
@@ -128,13 +140,15 @@ def create_app(
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        token_ids, finish_reason = await _collect(stream, sp)
+        token_ids, finish_reason, logprobs = await _collect(stream, sp)
         return JSONResponse({
             "id": rid, "object": "chat.completion", "created": created,
             "model": model_name,
             "choices": [{
                 "index": 0,
                 "message": {"role": "assistant", "content": tokenizer.decode(token_ids)},
+                "logprobs": _chat_logprobs(token_ids, logprobs)
+                if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,
             }],
             "usage": {
@@ -157,12 +171,14 @@ def create_app(
         sp = _params(body)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         stream = server.submit(prompt_ids, sp, rid=rid)
-        token_ids, finish_reason = await _collect(stream, sp)
+        token_ids, finish_reason, logprobs = await _collect(stream, sp)
         return JSONResponse({
             "id": rid, "object": "text_completion", "created": int(time.time()),
             "model": model_name,
             "choices": [{
                 "index": 0, "text": tokenizer.decode(token_ids),
+                "logprobs": {"token_logprobs": logprobs}
+                if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,
             }],
             "usage": {

@@ -24,6 +24,15 @@ class Sampler:
 
     def sample(self, logits: torch.Tensor, reqs: List[InitialRequest]) -> List[int]:
         """logits: [B, vocab] fp32, row i belongs to reqs[i]. Returns token ids."""
+        return [t for t, _ in self.sample_with_logprobs(logits, reqs)]
+
+    def sample_with_logprobs(
+        self, logits: torch.Tensor, reqs: List[InitialRequest]
+    ) -> List[tuple]:
+        """[(token_id, logprob-or-None)] — the logprob is log_softmax of the
+        post-penalty, pre-temperature logits at the sampled token, computed
+        only for requests with sampling_params.logprobs (reference wire fields
+        token_prob / return_probs)."""
         B = logits.shape[0]
         assert B == len(reqs)
         sp = [r.sampling_params for r in reqs]
@@ -49,4 +58,13 @@ class Sampler:
             [s.min_p for s in sp],
             generator=self.generator,
         )
-        return tokens.tolist()
+        tok_list = tokens.tolist()
+        if not any(s.logprobs for s in sp):
+            return [(t, None) for t in tok_list]
+        lp = torch.log_softmax(logits.float(), dim=-1).gather(
+            1, tokens.view(-1, 1).to(logits.device)
+        ).squeeze(1).tolist()
+        return [
+            (t, lp[i] if sp[i].logprobs else None)
+            for i, t in enumerate(tok_list)
+        ]

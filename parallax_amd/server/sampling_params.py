@@ -27,6 +27,9 @@ class SamplingParams:
     seed: Optional[int] = None
     # structured output hook (reference carries json_schema through the wire)
     json_schema: Optional[str] = None
+    # return per-token logprob of the sampled token (reference wire field
+    # token_prob / return_probs, proto forward.proto)
+    logprobs: bool = False
 
     def __post_init__(self):
         if self.temperature < 0.0:
@@ -60,6 +63,7 @@ class SamplingParams:
             "ignore_eos": self.ignore_eos,
             "seed": self.seed,
             "json_schema": self.json_schema,
+            "logprobs": self.logprobs,
         }
 
     @classmethod
@@ -90,4 +94,5 @@ class SamplingParams:
             stop=stop,
             ignore_eos=bool(body.get("ignore_eos", False)),
             seed=body.get("seed"),
+            logprobs=bool(body.get("logprobs", False)),
         )

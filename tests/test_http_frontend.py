@@ -98,3 +98,16 @@ def test_concurrent_requests(client):
 def test_stats(client):
     s = client.get("/stats").json()
     assert s["total_requests"] >= 1 and s["total_output_tokens"] >= 1
+
+
+def test_logprobs_roundtrip(client):
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 4, "temperature": 0.0, "ignore_eos": True,
+        "logprobs": True,
+    })
+    assert r.status_code == 200
+    lp = r.json()["choices"][0]["logprobs"]
+    assert lp is not None and len(lp["content"]) == 4
+    for e in lp["content"]:
+        assert e["logprob"] <= 0.0
