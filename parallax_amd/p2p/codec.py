@@ -79,8 +79,9 @@ def encode_control(kind: str, rids: List[str]) -> bytes:
     return msgpack.packb({"kind": kind, "rids": rids}, use_bin_type=True)
 
 
-def encode_tokens(tokens: List[Tuple[str, int]]) -> bytes:
-    """Sampled (rid, token_id) pairs from the last stage back to the head."""
+def encode_tokens(tokens) -> bytes:
+    """Sampled (rid, token_id) or (rid, token_id, logprob) tuples from the
+    last stage back to the head (logprob = reference proto token_prob)."""
     return msgpack.packb({"kind": "token", "tokens": tokens}, use_bin_type=True)
 
 

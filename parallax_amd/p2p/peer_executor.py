@@ -45,6 +45,7 @@ class PeerOutput:
     token_id: int
     finished: bool
     finish_reason: Optional[str] = None
+    logprob: Optional[float] = None
 
 
 class PeerExecutor:
@@ -190,13 +191,15 @@ class PeerExecutor:
                 break
             msg = codec.decode(data)
             if msg["kind"] == "token":
-                for rid, tok in msg["tokens"]:
+                for entry in msg["tokens"]:
+                    rid, tok = entry[0], entry[1]
+                    lp = entry[2] if len(entry) > 2 else None
                     finished = self.scheduler.commit_token(rid, tok)
                     req = finished or self.scheduler.running.get(rid)
                     if req is not None:
                         self.finished_outputs.append(
                             PeerOutput(rid, tok, finished is not None,
-                                       req.status.finish_reason)
+                                       req.status.finish_reason, logprob=lp)
                         )
                     if finished is not None:
                         self._broadcast_control("release", [rid],
@@ -363,15 +366,16 @@ class PeerExecutor:
                                         sampling_params=sp)
                     fr.output_token_ids = out_ids
                     fake_reqs.append(fr)
-                tokens = self.sampler.sample(logits, fake_reqs)
-                for p, tok in zip(sample_pkts, tokens):
+                sampled = self.sampler.sample_with_logprobs(logits, fake_reqs)
+                for p, (tok, _) in zip(sample_pkts, sampled):
                     if p.rid in self._sampling_ctx:
                         self._sampling_ctx[p.rid][2].append(tok)
                 # token goes to the head = first entry of the routing table
                 head = sample_pkts[0].routing_table[0]
                 self.transport.send(
                     head, codec.encode_tokens(
-                        [(p.rid, int(t)) for p, t in zip(sample_pkts, tokens)]
+                        [(p.rid, int(t), lp)
+                         for p, (t, lp) in zip(sample_pkts, sampled)]
                     ),
                 )
         else:
