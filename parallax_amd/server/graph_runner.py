@@ -184,6 +184,7 @@ class DecodeGraphRunner:
         hidden_in: Optional[torch.Tensor] = None,
         rids: Optional[List[str]] = None,
         linear_slots: Optional[List[int]] = None,
+        input_ids_dev: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         """Returns logits[:B] (last stage) or hidden[:B] (other stages)."""
         B = len(seq_lens)
@@ -223,7 +224,17 @@ class DecodeGraphRunner:
         self._cached_rids = list(rids) if rids is not None else []
         self._cached_btlens = btlens
 
-        self.input_ids[:bucket].copy_(self.h_input_ids[:bucket], non_blocking=True)
+        if input_ids_dev is None:
+            self.input_ids[:bucket].copy_(self.h_input_ids[:bucket],
+                                          non_blocking=True)
+        else:
+            # async decode: previous step's sampled tokens feed this step's
+            # input ids without a host round-trip (device-to-device, ordered
+            # after the sampler kernels on the same stream)
+            self.input_ids[:B].copy_(input_ids_dev.view(-1)[:B])
+            if bucket > B:
+                self.input_ids[B:bucket].copy_(self.h_input_ids[B:bucket],
+                                               non_blocking=True)
         self.positions[:bucket].copy_(self.h_positions[:bucket], non_blocking=True)
         self.slot_mapping[:bucket].copy_(self.h_slot_mapping[:bucket], non_blocking=True)
         self.seq_lens[:bucket].copy_(self.h_seq_lens[:bucket], non_blocking=True)

@@ -1,0 +1,78 @@
+"""Async decode pipelining: the one-step-late commit path must produce
+exactly the same tokens as the synchronous engine, fall back safely for
+value-dependent finishes (EOS/stop), and drain cleanly at boundaries."""
+
+import pytest
+import torch
+
+from parallax_amd.models.config import ModelConfig
+from parallax_amd.server import engine as engine_mod
+from parallax_amd.server.engine import Engine, EngineArgs
+from parallax_amd.server.sampling_params import SamplingParams
+
+
+def _cfg():
+    return ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=211, hidden_size=64,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+        intermediate_size=128, max_position_embeddings=256,
+        eos_token_ids=[5], rope_theta=10000.0,
+    )
+
+
+def _run(async_on, sps, prompts, monkeypatch):
+    monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+    eng = Engine(_cfg(), EngineArgs(block_size=8, num_kv_blocks=128,
+                                    dtype=torch.float32, seed=7),
+                 random_weights=True)
+    return eng.generate(prompts, sps)
+
+
+PROMPTS = [[1, 2, 3, 4, 5, 6], [9, 8, 7]]
+
+
+def test_async_matches_sync_ignore_eos(monkeypatch):
+    sps = [SamplingParams(temperature=0.0, max_new_tokens=8, ignore_eos=True)] * 2
+    a = _run(True, sps, PROMPTS, monkeypatch)
+    b = _run(False, sps, PROMPTS, monkeypatch)
+    assert list(a.values()) == list(b.values())
+    assert all(len(v) == 8 for v in a.values())
+
+
+def test_async_matches_sync_sampled(monkeypatch):
+    """Temperature sampling: the RNG call sequence must be identical, so
+    sampled tokens match bit-for-bit too."""
+    sps = [SamplingParams(temperature=0.8, top_p=0.9, max_new_tokens=10,
+                          ignore_eos=True)] * 2
+    a = _run(True, sps, PROMPTS, monkeypatch)
+    b = _run(False, sps, PROMPTS, monkeypatch)
+    assert list(a.values()) == list(b.values())
+
+
+def test_eos_sensitive_falls_back(monkeypatch):
+    """ignore_eos=False keeps requests on the sync path (finish depends on
+    token values) — results identical either way."""
+    sps = [SamplingParams(temperature=0.0, max_new_tokens=12)] * 2
+    a = _run(True, sps, PROMPTS, monkeypatch)
+    b = _run(False, sps, PROMPTS, monkeypatch)
+    assert list(a.values()) == list(b.values())
+
+
+def test_mixed_lengths_transition(monkeypatch):
+    """Different max_new_tokens: the shorter request exits the async envelope
+    2 tokens early, forcing drain -> sync -> (smaller) async transitions."""
+    sps = [
+        SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True),
+        SamplingParams(temperature=0.0, max_new_tokens=12, ignore_eos=True),
+    ]
+    a = _run(True, sps, PROMPTS, monkeypatch)
+    b = _run(False, sps, PROMPTS, monkeypatch)
+    assert list(a.values()) == list(b.values())
+    assert [len(v) for v in a.values()] == [4, 12]
+
+
+def test_no_placeholders_leak(monkeypatch):
+    sps = [SamplingParams(temperature=0.0, max_new_tokens=9, ignore_eos=True)] * 2
+    a = _run(True, sps, PROMPTS, monkeypatch)
+    for toks in a.values():
+        assert all(t >= 0 for t in toks)
