@@ -220,7 +220,10 @@ class Engine:
             )
         self._pending_adds: List[InitialRequest] = []
         self._inflight = None       # async decode pipelining state
-        self._async_pinned = None
+        # double-buffered pinned staging: step N's D2H copy must not overwrite
+        # the buffer step N-1's finalize has yet to read
+        self._async_pinned = [None, None]
+        self._async_pin_idx = 0
         self._pending_aborts: List[str] = []
         self.step_count = 0
         logger.info(
@@ -407,13 +410,14 @@ class Engine:
                 generator=self.sampler.generator,
             )
         if self.device.type == "cuda":
-            pinned = self._async_pinned
+            self._async_pin_idx ^= 1
+            pinned = self._async_pinned[self._async_pin_idx]
             if pinned is None or pinned.numel() < len(reqs):
                 pinned = torch.empty(
                     max(len(reqs), self.args.max_batch_size),
                     dtype=tokens_dev.dtype, pin_memory=True,
                 )
-                self._async_pinned = pinned
+                self._async_pinned[self._async_pin_idx] = pinned
             pinned[: len(reqs)].copy_(tokens_dev, non_blocking=True)
             event = torch.cuda.Event()
             event.record()
