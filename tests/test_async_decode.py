@@ -76,3 +76,25 @@ def test_no_placeholders_leak(monkeypatch):
     a = _run(True, sps, PROMPTS, monkeypatch)
     for toks in a.values():
         assert all(t >= 0 for t in toks)
+
+
+@pytest.mark.gpu
+def test_async_matches_sync_gpu(monkeypatch):
+    """On the GPU (graphs + device-fed input ids + pinned D2H staging) the
+    async pipeline must produce the same greedy tokens as the sync path."""
+    cfg = _cfg()
+    cfg.vocab_size = 512
+    cfg.hidden_size = 256
+    cfg.head_dim = 64
+
+    def run(async_on):
+        monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+        eng = Engine(cfg, EngineArgs(num_kv_blocks=128, seed=7),
+                     random_weights=True)
+        sps = [SamplingParams(temperature=0.0, max_new_tokens=12,
+                              ignore_eos=True)] * 2
+        return list(eng.generate([[1, 2, 3, 4, 5], [9, 8, 7]], sps).values())
+
+    a, b = run(True), run(False)
+    assert a == b
+    assert all(len(v) == 12 for v in a)
