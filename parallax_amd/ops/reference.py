@@ -306,13 +306,26 @@ def sample_tokens(
         (not g) and (0 < k < V or p < 1.0 or m > 0.0)
         for g, k, p, m in zip(greedy, tks, tps, mps)
     )
-    out = logits.argmax(dim=-1)  # greedy default for every row
     if all(greedy):
-        return out
+        return logits.argmax(dim=-1)
 
     t_gpu = torch.tensor(
         [max(t, 1e-6) for t in temps], dtype=torch.float32, device=dev
     ).unsqueeze(-1)
+    if not need_filter:
+        # gumbel-max: argmax(logits/T + G), G ~ Gumbel(0,1). Identical
+        # distribution to softmax+multinomial but pure elementwise+argmax —
+        # torch.multinomial over a [B, 128k] prob matrix is the single most
+        # expensive sampler kernel on ROCm
+        u = torch.rand(B, V, device=dev, generator=generator)
+        gumbel = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))
+        sampled = (logits / t_gpu + gumbel).argmax(dim=-1)
+        if not any(greedy):
+            return sampled
+        greedy_mask = torch.tensor(greedy, dtype=torch.bool, device=dev)
+        return torch.where(greedy_mask, logits.argmax(dim=-1), sampled)
+
+    out = logits.argmax(dim=-1)  # greedy default for every row
     probs = torch.softmax(logits / t_gpu, dim=-1)
     if need_filter:
         sorted_p, idx = torch.sort(probs, dim=-1, descending=True)
