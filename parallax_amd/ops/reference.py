@@ -318,7 +318,9 @@ def sample_tokens(
         # torch.multinomial over a [B, 128k] prob matrix is the single most
         # expensive sampler kernel on ROCm
         u = torch.rand(B, V, device=dev, generator=generator)
-        gumbel = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))
+        gumbel = -torch.log(
+            (-torch.log(u.clamp_min(1e-20))).clamp_min(1e-20)
+        )
         sampled = (logits / t_gpu + gumbel).argmax(dim=-1)
         if not any(greedy):
             return sampled
