@@ -117,11 +117,30 @@ def deepseek_v3() -> dict:
     }
 
 
+def kimi_k2() -> dict:
+    """Kimi-K2 (1T MoE): DeepSeek-V3 architecture class with 384 routed
+    experts — the BASELINE prefix-reuse config (Kimi-K2 PP=8 @ 256k ctx)."""
+    cfg = deepseek_v3()
+    cfg.update({
+        "architectures": ["KimiK2ForCausalLM"],
+        "model_type": "kimi_k2",
+        "vocab_size": 163840,
+        "n_routed_experts": 384,
+        "num_experts_per_tok": 8,
+        "n_group": 1,
+        "topk_group": 1,
+        "first_k_dense_replace": 1,
+        "max_position_embeddings": 262144,
+    })
+    return cfg
+
+
 MODELS = {
     "deepseek-r1-distill-llama-8b": (deepseek_r1_distill_llama_8b, "DeepSeek-R1-Distill-Llama-8B"),
     "qwen2-0.5b": (qwen2_05b, "Qwen2-0.5B"),
     "qwen2-72b": (qwen2_72b, "Qwen2-72B"),
     "deepseek-v3": (deepseek_v3, "DeepSeek-V3-671B"),
+    "kimi-k2": (kimi_k2, "Kimi-K2"),
 }
 
 
@@ -143,6 +162,10 @@ def main():
     ap.add_argument("--parallelism", default="auto", choices=["auto", "dp", "pp"],
                     help="auto: DP replicas when the model fits one GPU "
                          "(288 GB HBM3E), PP layer split otherwise")
+    ap.add_argument("--shared-prefix", type=int, default=0,
+                    help="tokens of prompt shared across all requests "
+                         "(exercises block-radix prefix reuse; BASELINE "
+                         "Kimi-K2 config)")
     ap.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
     args = ap.parse_args()
 
@@ -193,7 +216,7 @@ def main():
         micro_batches=args.micro_batches or (1 if mode == "dp" else world),
         dtype=torch.bfloat16 if use_gpu else torch.float32,
         num_kv_blocks=None if use_gpu else 4096,
-        enable_prefix_cache=False,  # unique synthetic prompts; skip radix overhead
+        enable_prefix_cache=args.shared_prefix > 0,
         kv_cache_dtype=args.kv_dtype,
         seed=0,
     )
@@ -207,8 +230,14 @@ def main():
         max_new_tokens=total_new_tokens, ignore_eos=True,
     )
     if comm.rank == 0:
+        shared = torch.randint(
+            0, cfg.vocab_size, (args.shared_prefix,), generator=g
+        ).tolist() if args.shared_prefix else []
         for i in range(engine_batch):
-            prompt = torch.randint(0, cfg.vocab_size, (args.prompt_len,), generator=g).tolist()
+            tail = max(1, args.prompt_len - len(shared))
+            prompt = shared + torch.randint(
+                0, cfg.vocab_size, (tail,), generator=g
+            ).tolist()
             engine.submit(prompt, sp)
 
     def sync():
@@ -272,6 +301,7 @@ def main():
                 "parallelism": f"{mode if world > 1 else 'pp'}{world}"
                 if world > 1 else "pp1",
                 "kv_dtype": args.kv_dtype,
+                "shared_prefix": args.shared_prefix,
                 "micro_batches": eargs.micro_batches,
                 "prefill_s": round(prefill_s, 3),
             },
