@@ -25,8 +25,8 @@ void launch_rope_and_cache(void*, const void*, const void*, void*, void*,
 void launch_reshape_and_cache(const void*, const void*, void*, void*,
                               const int64_t*, int, int, int, int, hipStream_t);
 void launch_mla_reshape_and_cache(const void*, const void*, void*,
-                                  const int64_t*, int, int, int, int,
-                                  hipStream_t);
+                                  const int64_t*, int, int, int, int, bool,
+                                  float, hipStream_t);
 void launch_act_and_mul(void*, const void*, int64_t, int, bool, hipStream_t);
 void launch_paged_attention_decode(void*, const void*, const void*, const void*,
                                    const int*, const int*, int, int, int, int,
@@ -44,7 +44,7 @@ void launch_mla_paged_attention_decode(void*, const void*, const void*,
                                        const void*, const int*, const int*,
                                        int, int, int, int, int, int, float,
                                        int, int, float*, float*, const int*,
-                                       int, hipStream_t, bool*);
+                                       int, bool, float, hipStream_t, bool*);
 void launch_msa_paged_attention_decode(void*, const void*, const void*,
                                        const void*, const int*, const int*,
                                        const int*, int, int, int, int, int,
@@ -180,7 +180,7 @@ void mla_reshape_and_cache(torch::Tensor latent, torch::Tensor k_rope,
   launch_mla_reshape_and_cache(latent.data_ptr(), k_rope.data_ptr(),
                                cache.data_ptr(),
                                slot_mapping.data_ptr<int64_t>(), T, R, DR, BS,
-                               cur_stream());
+                               is_fp8(cache), 1.0f, cur_stream());
 }
 
 void silu_and_mul(torch::Tensor out, torch::Tensor x) {
@@ -328,7 +328,8 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
   CHECK_CONTIG(q_latent);
   CHECK_CONTIG(q_pe);
   CHECK_BF16(q_latent);
-  CHECK_BF16(cache);
+  TORCH_CHECK(cache.scalar_type() == at::kBFloat16 || is_fp8(cache),
+              "MLA cache must be bf16 or fp8_e4m3");
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
   const int B = q_latent.size(0);
@@ -368,7 +369,7 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
       out.data_ptr(), q_latent.data_ptr(), q_pe.data_ptr(), cache.data_ptr(),
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, H, R, DR, BS,
       max_blocks, (float)scale, num_parts, part_tokens, acc_ptr, ml_ptr,
-      idx_ptr, max_topk, cur_stream(), &launched);
+      idx_ptr, max_topk, is_fp8(cache), 1.0f, cur_stream(), &launched);
   TORCH_CHECK(launched, "no MLA kernel for R=", R, " DR=", DR, " BS=", BS);
   if (num_parts > 1)
     launch_paged_attention_reduce(out.data_ptr(), acc_ptr, ml_ptr, B, H, R,

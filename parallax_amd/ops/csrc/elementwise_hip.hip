@@ -268,30 +268,61 @@ extern "C" void launch_reshape_and_cache(
 }
 
 // MLA variant: latent [T, R] + rope [T, dr] -> cache [NB, BS, R+dr]
+template <bool FP8>
 __global__ void mla_reshape_and_cache_kernel(
     const uint16_t* __restrict__ latent, const uint16_t* __restrict__ k_rope,
-    uint16_t* __restrict__ cache, const int64_t* __restrict__ slot_mapping,
-    const int R, const int DR, const int BS) {
+    void* __restrict__ cache, const int64_t* __restrict__ slot_mapping,
+    const int R, const int DR, const int BS, const float inv_scale) {
   const int t = blockIdx.x;
   const int64_t slot = slot_mapping[t];
   if (slot < 0) return;
   const int64_t blk = slot / BS, off = slot % BS;
-  uint16_t* dst = cache + ((size_t)blk * BS + off) * (R + DR);
-  for (int i = threadIdx.x * 8; i < R; i += blockDim.x * 8)
-    *reinterpret_cast<int4*>(dst + i) =
-        *reinterpret_cast<const int4*>(latent + (size_t)t * R + i);
-  for (int i = threadIdx.x * 8; i < DR; i += blockDim.x * 8)
-    *reinterpret_cast<int4*>(dst + R + i) =
-        *reinterpret_cast<const int4*>(k_rope + (size_t)t * DR + i);
+  const size_t row = ((size_t)blk * BS + off) * (R + DR);
+  if (FP8) {
+    uint8_t* dst = (uint8_t*)cache + row;
+    for (int i = threadIdx.x * 8; i < R; i += blockDim.x * 8) {
+      uint64_t packed = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        packed |= (uint64_t)f32_to_fp8_e4m3(
+                      bf16_bits_to_f32(latent[(size_t)t * R + i + j]) *
+                      inv_scale)
+                  << (8 * j);
+      *reinterpret_cast<uint64_t*>(dst + i) = packed;
+    }
+    for (int i = threadIdx.x * 8; i < DR; i += blockDim.x * 8) {
+      uint64_t packed = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        packed |= (uint64_t)f32_to_fp8_e4m3(
+                      bf16_bits_to_f32(k_rope[(size_t)t * DR + i + j]) *
+                      inv_scale)
+                  << (8 * j);
+      *reinterpret_cast<uint64_t*>(dst + R + i) = packed;
+    }
+  } else {
+    uint16_t* dst = (uint16_t*)cache + row;
+    for (int i = threadIdx.x * 8; i < R; i += blockDim.x * 8)
+      *reinterpret_cast<int4*>(dst + i) =
+          *reinterpret_cast<const int4*>(latent + (size_t)t * R + i);
+    for (int i = threadIdx.x * 8; i < DR; i += blockDim.x * 8)
+      *reinterpret_cast<int4*>(dst + R + i) =
+          *reinterpret_cast<const int4*>(k_rope + (size_t)t * DR + i);
+  }
 }
 
 extern "C" void launch_mla_reshape_and_cache(
     const void* latent, const void* k_rope, void* cache,
-    const int64_t* slot_mapping, int T, int R, int DR, int BS,
-    hipStream_t stream) {
- hipLaunchKernelGGL(( mla_reshape_and_cache_kernel), dim3(T), dim3(128), 0, stream, 
-      (const uint16_t*)latent, (const uint16_t*)k_rope, (uint16_t*)cache,
-      slot_mapping, R, DR, BS);
+    const int64_t* slot_mapping, int T, int R, int DR, int BS, bool fp8,
+    float inv_scale, hipStream_t stream) {
+  if (fp8)
+   hipLaunchKernelGGL(( mla_reshape_and_cache_kernel<true>), dim3(T), dim3(128), 0, stream, 
+        (const uint16_t*)latent, (const uint16_t*)k_rope, cache, slot_mapping,
+        R, DR, BS, inv_scale);
+  else
+   hipLaunchKernelGGL(( mla_reshape_and_cache_kernel<false>), dim3(T), dim3(128), 0, stream, 
+        (const uint16_t*)latent, (const uint16_t*)k_rope, cache, slot_mapping,
+        R, DR, BS, inv_scale);
 }
 
 // ---- activations ----------------------------------------------------------------------

@@ -20,20 +20,21 @@
 #define MLA_KTILE 32
 #define MLA_HBLOCK 32
 
-template <int R, int DR, int BLOCK_SIZE, bool PARTITIONED, bool SPARSE>
+template <int R, int DR, int BLOCK_SIZE, bool PARTITIONED, bool SPARSE,
+          bool KV_FP8 = false>
 __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
     uint16_t* __restrict__ out,       // [B, H, R]
     float* __restrict__ tmp_acc,      // [B, H, P, R]
-    float* __restrict__ tmp_ml,       // [B, H, P, 2]
+    float* __restrict__ tmp_ml,      // [B, H, P, 2]
     const uint16_t* __restrict__ q_latent,  // [B, H, R]
     const uint16_t* __restrict__ q_pe,      // [B, H, DR]
-    const uint16_t* __restrict__ cache,     // [NB, BS, R+DR]
+    const void* __restrict__ cache_v,       // [NB, BS, R+DR] bf16 | fp8(e4m3)
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     const int max_blocks, const int H, const float scale,
     const int part_tokens,
     const int* __restrict__ topk_indices,  // [B, max_topk] (SPARSE only)
-    const int max_topk) {
+    const int max_topk, const float c_scale = 1.f) {
   constexpr int DK = R + DR;            // 576
   constexpr int KSTEPS = DK / 32;       // 18
   constexpr int PROW = 40;              // padded row length (bank spread)
@@ -108,17 +109,31 @@ __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
         gtok = idx_row[j];
         ok = gtok >= 0 && gtok < L;
       }
-      const uint16_t* crow = nullptr;
+      size_t crow_tok = 0;
       if (ok) {
         const int blk = btab[gtok / BLOCK_SIZE];
         const int off = gtok % BLOCK_SIZE;
-        crow = cache + ((size_t)blk * BLOCK_SIZE + off) * DK;
+        crow_tok = (size_t)blk * BLOCK_SIZE + off;
       }
 #pragma unroll
       for (int c = 0; c < 9; ++c) {
         const int d = (dv * 9 + c) * 8;
         int4 val = make_int4(0, 0, 0, 0);
-        if (ok) val = *reinterpret_cast<const int4*>(crow + d);
+        if (ok) {
+          if (KV_FP8) {
+            // 8 fp8 bytes -> 8 bf16 (one cache-wide scale)
+            const uint64_t raw = *reinterpret_cast<const uint64_t*>(
+                (const uint8_t*)cache_v + crow_tok * DK + d);
+            uint16_t* vs16 = reinterpret_cast<uint16_t*>(&val);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              vs16[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * c_scale);
+          } else {
+            val = *reinterpret_cast<const int4*>(
+                (const uint16_t*)cache_v + crow_tok * DK + d);
+          }
+        }
         const int byte = swz(tok * DK * 2 + d * 2, tok);
         *reinterpret_cast<int4*>(reinterpret_cast<char*>(Cl) + byte) = val;
         const uint16_t* vs = reinterpret_cast<const uint16_t*>(&val);
@@ -274,28 +289,33 @@ extern "C" void launch_mla_paged_attention_decode(
     const int* block_tables, const int* seq_lens, int B, int H, int R, int DR,
     int BS, int max_blocks, float scale, int num_parts, int part_tokens,
     float* tmp_acc, float* tmp_ml, const int* topk_indices, int max_topk,
-    hipStream_t stream, bool* launched) {
+    bool kv_fp8, float c_scale, hipStream_t stream, bool* launched) {
   *launched = false;
   if (R != 512 || DR != 64) return;  // DeepSeek V2/V3/R1/K2 geometry
   const int head_blocks = (H + MLA_HBLOCK - 1) / MLA_HBLOCK;
 
-#define MLA_LAUNCH2(BSZ, SP)                                                  \
+#define MLA_LAUNCH3(BSZ, SP, FP8)                                             \
   if (num_parts <= 1) {                                                       \
     dim3 grid(head_blocks, B, 1);                                             \
-    mla_decode_kernel<512, 64, BSZ, false, SP>                                \
+    mla_decode_kernel<512, 64, BSZ, false, SP, FP8>                           \
         <<<grid, MLA_THREADS, 0, stream>>>(                                   \
         (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q_latent,          \
-        (const uint16_t*)q_pe, (const uint16_t*)cache, block_tables,          \
-        seq_lens, max_blocks, H, scale, 0, topk_indices, max_topk);           \
+        (const uint16_t*)q_pe, cache, block_tables,                           \
+        seq_lens, max_blocks, H, scale, 0, topk_indices, max_topk, c_scale);  \
   } else {                                                                    \
     dim3 grid(head_blocks, B, num_parts);                                     \
-    mla_decode_kernel<512, 64, BSZ, true, SP>                                 \
+    mla_decode_kernel<512, 64, BSZ, true, SP, FP8>                            \
         <<<grid, MLA_THREADS, 0, stream>>>(                                   \
         nullptr, tmp_acc, tmp_ml, (const uint16_t*)q_latent,                  \
-        (const uint16_t*)q_pe, (const uint16_t*)cache, block_tables,          \
-        seq_lens, max_blocks, H, scale, part_tokens, topk_indices, max_topk); \
+        (const uint16_t*)q_pe, cache, block_tables,                           \
+        seq_lens, max_blocks, H, scale, part_tokens, topk_indices, max_topk,  \
+        c_scale);                                                             \
   }                                                                           \
   *launched = true;
+
+#define MLA_LAUNCH2(BSZ, SP)                                                  \
+  if (kv_fp8) { MLA_LAUNCH3(BSZ, SP, true) }                                  \
+  else { MLA_LAUNCH3(BSZ, SP, false) }
 
 #define MLA_LAUNCH(BSZ)                                                       \
   if (topk_indices != nullptr) { MLA_LAUNCH2(BSZ, true) }                     \
@@ -306,4 +326,5 @@ extern "C" void launch_mla_paged_attention_decode(
   else if (BS == 64) { MLA_LAUNCH(64) }
 #undef MLA_LAUNCH
 #undef MLA_LAUNCH2
+#undef MLA_LAUNCH3
 }

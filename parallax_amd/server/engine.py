@@ -167,10 +167,13 @@ class Engine:
                 else:
                     budget = 1 << 30
                 num_blocks = max(16, budget // mla_block_bytes)
+            mla_dtype = args.dtype if args.dtype != torch.float32 else torch.float32
+            if args.kv_cache_dtype == "fp8":
+                assert self.device.type == "cuda", "fp8 KV needs the HIP kernels"
+                mla_dtype = torch.float8_e4m3fn
             self.kv_cache = MLAKVCache(
                 end - start, cfg.kv_lora_rank, cfg.qk_rope_head_dim,
-                args.block_size, num_blocks, self.device,
-                args.dtype if args.dtype != torch.float32 else torch.float32,
+                args.block_size, num_blocks, self.device, mla_dtype,
                 index_dim=cfg.index_head_dim if cfg.is_dsa else 0,
             )
         else:

@@ -110,13 +110,14 @@ class MLAKVCache:
         # DSA (DeepSeek-V3.2): one indexer key per token per layer, shared
         # across indexer query heads (reference dsa_cache.py)
         self.index_dim = index_dim
+        idx_dtype = dtype if dtype != torch.float8_e4m3fn else torch.bfloat16
         # +1 trash block at the end: graph-safe indexer-cache stores write
         # pad tokens (slot -1) there instead of branching (see
         # ops.reference.store_indexer_cache)
         self.index_caches: List[torch.Tensor] = (
             [
                 torch.zeros((num_blocks + 1, block_size, index_dim),
-                            dtype=dtype, device=device)
+                            dtype=idx_dtype, device=device)
                 for _ in range(num_layers)
             ]
             if index_dim > 0

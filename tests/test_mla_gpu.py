@@ -70,3 +70,41 @@ def test_deepseek_engine_gpu_smoke():
         [SamplingParams(temperature=0.0, max_new_tokens=5, ignore_eos=True)] * 2,
     )
     assert all(len(v) == 5 for v in out.values())
+
+
+@pytest.mark.gpu
+def test_mla_decode_fp8_cache():
+    """FP8-E4M3 compressed cache: quantize-on-store + dequantize-in-staging
+    must track the bf16 kernel within fp8 rounding error."""
+    torch.manual_seed(3)
+    B, H, R, DR, BS = 4, 32, 512, 64, 32
+    ctxs = [300, 77, 512, 40]
+    max_blocks = (max(ctxs) + BS - 1) // BS
+    NB = B * max_blocks + 1
+    latent = torch.randn(sum(ctxs), R, dtype=torch.bfloat16, device="cuda") / 4
+    rope = torch.randn(sum(ctxs), DR, dtype=torch.bfloat16, device="cuda") / 4
+    cache8 = torch.zeros(NB, BS, R + DR, dtype=torch.float8_e4m3fn, device="cuda")
+    cache16 = torch.zeros(NB, BS, R + DR, dtype=torch.bfloat16, device="cuda")
+    bt = (torch.arange(B * max_blocks, dtype=torch.int32, device="cuda")
+          .reshape(B, max_blocks) + 1)
+    slots = []
+    t0 = 0
+    for i, L in enumerate(ctxs):
+        for p in range(L):
+            blk = int(bt[i, p // BS])
+            slots.append(blk * BS + p % BS)
+        t0 += L
+    sm = torch.tensor(slots, dtype=torch.int64, device="cuda")
+    ops.mla_reshape_and_cache(latent, rope, cache8, sm)
+    ops.mla_reshape_and_cache(latent, rope, cache16, sm)
+    sl = torch.tensor(ctxs, dtype=torch.int32, device="cuda")
+    ql = torch.randn(B, H, R, dtype=torch.bfloat16, device="cuda") / 8
+    qp = torch.randn(B, H, DR, dtype=torch.bfloat16, device="cuda") / 8
+    scale = 1.0 / math.sqrt(R + DR)
+    out8 = ops.mla_paged_attention_decode(ql, qp, cache8, bt, sl, scale)
+    out16 = ops.mla_paged_attention_decode(ql, qp, cache16, bt, sl, scale)
+    # fp8 rounding on both K and V sides: loose tolerance, but the softmax
+    # keeps outputs in the same range
+    torch.testing.assert_close(out8.float(), out16.float(), atol=0.12, rtol=0.12)
+    # and the fp8 store actually quantized (bytes differ from bf16 view)
+    assert cache8.dtype == torch.float8_e4m3fn
