@@ -25,7 +25,7 @@ from typing import Dict, Optional
 
 import httpx
 from fastapi import FastAPI, HTTPException, Request
-from fastapi.responses import JSONResponse
+from fastapi.responses import JSONResponse, StreamingResponse
 
 from ..models.config import ModelConfig
 from ..scheduling.model_info import ModelInfo
@@ -219,12 +219,55 @@ load(); setInterval(load, 3000);
             raise HTTPException(502, f"head peer {head} has no HTTP endpoint")
         url = f"http://{ep['host']}:{ep['port']}/v1/chat/completions"
         body.setdefault("routing_table", decision.routing_table)
+        t0 = time.monotonic()
+
+        def _log_usage(usage: dict) -> None:
+            # reference gateway parses the final SSE usage chunk and logs
+            # per-request TPS/TTFT (request_handler.py:190-202)
+            if not usage:
+                return
+            logger.info(
+                "request via %s: in=%s out=%s ttft_ms=%s tps=%s e2e_ms=%.0f",
+                head, usage.get("prompt_tokens"), usage.get("completion_tokens"),
+                usage.get("ttft_ms"), usage.get("tps"),
+                (time.monotonic() - t0) * 1e3,
+            )
+
+        if body.get("stream"):
+            # SSE relay: stream chunks through while holding the routing slot
+            async def relay():
+                last_usage = {}
+                try:
+                    async with httpx.AsyncClient(timeout=600.0) as client:
+                        async with client.stream("POST", url, json=body) as resp:
+                            async for chunk in resp.aiter_bytes():
+                                if b'"usage"' in chunk:
+                                    try:
+                                        import json as _json
+
+                                        for line in chunk.decode().splitlines():
+                                            if line.startswith("data: {"):
+                                                d = _json.loads(line[6:])
+                                                if d.get("usage"):
+                                                    last_usage = d["usage"]
+                                    except Exception:
+                                        pass
+                                yield chunk
+                finally:
+                    _log_usage(last_usage)
+                    with svc._lock:
+                        svc.scheduler.complete_request(decision.routing_table)
+
+            return StreamingResponse(relay(), media_type="text/event-stream")
+
         try:
             async with httpx.AsyncClient(timeout=600.0) as client:
                 for attempt in range(FORWARD_RETRIES):
                     try:
                         resp = await client.post(url, json=body)
-                        return JSONResponse(resp.json(), status_code=resp.status_code)
+                        payload = resp.json()
+                        _log_usage(payload.get("usage") or {})
+                        return JSONResponse(payload, status_code=resp.status_code)
                     except httpx.HTTPError:
                         if attempt == FORWARD_RETRIES - 1:
                             raise
