@@ -13,7 +13,7 @@ import uuid
 from typing import List
 
 from fastapi import FastAPI, HTTPException, Request
-from fastapi.responses import JSONResponse
+from fastapi.responses import JSONResponse, StreamingResponse
 
 from ..server.sampling_params import SamplingParams
 from ..server.tokenizer_util import TokenizerWrapper
@@ -43,6 +43,54 @@ def create_head_app(
         sp = SamplingParams.from_openai(body)
         rid, q = agent.submit(prompt_ids, sp, routing_table)
         loop = asyncio.get_event_loop()
+        created = int(time.time())
+
+        if body.get("stream"):
+            async def sse():
+                import json as _json
+
+                token_ids: List[int] = []
+                sent_len = 0
+                t0 = time.monotonic()
+                first_t = None
+                while True:
+                    out = await loop.run_in_executor(None, q.get)
+                    if out is None:
+                        break
+                    if first_t is None:
+                        first_t = time.monotonic()
+                    token_ids.append(out.token_id)
+                    text = tokenizer.decode(token_ids)
+                    delta, nl = text[sent_len:], len(text)
+                    sent_len = nl
+                    chunk = {
+                        "id": rid, "object": "chat.completion.chunk",
+                        "created": created, "model": model_name,
+                        "choices": [{
+                            "index": 0, "delta": {"content": delta},
+                            "finish_reason": out.finish_reason
+                            if out.finished else None,
+                        }],
+                    }
+                    yield f"data: {_json.dumps(chunk)}\n\n"
+                elapsed = time.monotonic() - t0
+                usage = {
+                    "prompt_tokens": len(prompt_ids),
+                    "completion_tokens": len(token_ids),
+                    "total_tokens": len(prompt_ids) + len(token_ids),
+                    "ttft_ms": round(
+                        ((first_t or time.monotonic()) - t0) * 1e3, 2),
+                    "tps": round(len(token_ids) / max(elapsed, 1e-6), 2),
+                }
+                yield "data: " + _json.dumps({
+                    "id": rid, "object": "chat.completion.chunk",
+                    "created": created, "model": model_name,
+                    "choices": [], "usage": usage,
+                }) + "\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
         token_ids, finish_reason = [], "stop"
         t0 = time.monotonic()
         first_t = None

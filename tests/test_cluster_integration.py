@@ -148,6 +148,23 @@ def test_cluster_end_to_end():
         assert body["usage"]["completion_tokens"] == 5
         assert body["choices"][0]["message"]["content"]
 
+        # SSE streaming relayed through the gateway
+        chunks = []
+        with client.stream("POST", f"{base}/v1/chat/completions", json={
+            "model": "tiny-llama",
+            "messages": [{"role": "user", "content": "stream me"}],
+            "max_tokens": 4, "temperature": 0.0, "stream": True,
+        }) as resp:
+            assert resp.status_code == 200
+            for line in resp.iter_lines():
+                if line.startswith("data: "):
+                    chunks.append(line[6:])
+        assert chunks[-1] == "[DONE]"
+        import json as _json
+        deltas = [_json.loads(c) for c in chunks[:-1]]
+        usage = [d for d in deltas if d.get("usage")]
+        assert usage and usage[-1]["usage"]["completion_tokens"] == 4
+
     for a in agents:
         a.stop()
     backend_srv.should_exit = True
