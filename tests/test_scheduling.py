@@ -234,3 +234,50 @@ def test_dijkstra_route_latency_weighted():
     ranges = {"a": (0, 16), "b": (16, 32), "d": (0, 16), "e": (16, 32)}
     lat = {("a", "b"): 10.0, ("a", "e"): 0.1, ("d", "e"): 10.0, ("d", "b"): 10.0}
     assert dijkstra_route(ranges, 32, latency=lat) == ["a", "e"]
+
+
+def test_dp_allocator_prefers_parallel_pipelines():
+    """Z(k) = k^2/(T + stages/k * RTT): with 4 over-half-capacity nodes the
+    DP allocator should build 2 two-stage pipelines (throughput 4) rather
+    than one four-stage chain (reference DP objective,
+    layer_allocation.py:758+)."""
+    model = llama70b_info()
+    nodes = [build_node(f"n{i}", small_gpu(0.6, f"half{i}")) for i in range(4)]
+    set_full_rtt(nodes, rtt=2.0)
+    pipelines = DynamicProgrammingLayerAllocator(model).allocate_from_standby(nodes)
+    assert pipelines and len(pipelines) == 2
+    for p in pipelines:
+        assert p.covers(80)
+        spans = sorted((n.start_layer, n.end_layer) for n in p.nodes)
+        for (s0, e0), (s1, e1) in zip(spans, spans[1:]):
+            assert e0 == s1  # contiguous coverage
+
+
+def test_greedy_lookahead_closes_with_smallest_node():
+    """Greedy look-ahead: when a pipeline needs a closer, pick the SMALLEST
+    node able to close it so big nodes seed the next pipeline
+    (layer_allocation.py:582-755 behavior)."""
+    model = llama70b_info()
+    nodes = [
+        build_node("big0", small_gpu(0.55, "b0")),   # ~64-layer capacity
+        build_node("big1", small_gpu(0.55, "b1")),
+        build_node("small0", small_gpu(0.25, "s0")), # ~29-layer capacity
+        build_node("small1", small_gpu(0.25, "s1")),
+    ]
+    set_full_rtt(nodes)
+    pipelines = GreedyLayerAllocator(model).allocate_from_standby(nodes)
+    assert len(pipelines) == 2
+    for p in pipelines:
+        ids = {n.node_id for n in p.nodes}
+        # each pipeline pairs one big with one small — big nodes are not
+        # burned as closers for each other
+        assert len(ids & {"big0", "big1"}) == 1
+        assert len(ids & {"small0", "small1"}) == 1
+
+
+def test_allocation_fails_cleanly_when_undercapacity():
+    model = llama70b_info()
+    nodes = [build_node("tiny", small_gpu(0.05, "t"))]
+    set_full_rtt(nodes)
+    assert GreedyLayerAllocator(model).allocate_from_standby(nodes) == []
+    assert water_fill_layers(nodes, model, 80) is None
