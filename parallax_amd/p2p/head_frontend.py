@@ -32,6 +32,48 @@ def create_head_app(
     async def health():
         return {"status": "ok", "node_id": agent.node_id}
 
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "owned_by": "parallax_amd"}]}
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        prompt = body.get("prompt")
+        if prompt is None:
+            raise HTTPException(400, "prompt required")
+        if isinstance(prompt, list) and prompt and isinstance(prompt[0], int):
+            prompt_ids = prompt
+        else:
+            prompt_ids = tokenizer.encode(prompt)
+        routing_table: List[str] = body.get("routing_table") or [agent.node_id]
+        sp = SamplingParams.from_openai(body)
+        rid, q = agent.submit(prompt_ids, sp, routing_table)
+        loop = asyncio.get_event_loop()
+        token_ids, finish_reason = [], "stop"
+        while True:
+            out = await loop.run_in_executor(None, q.get)
+            if out is None:
+                break
+            token_ids.append(out.token_id)
+            if out.finished:
+                finish_reason = out.finish_reason or "stop"
+        return JSONResponse({
+            "id": rid, "object": "text_completion",
+            "created": int(time.time()), "model": model_name,
+            "choices": [{
+                "index": 0, "text": tokenizer.decode(token_ids),
+                "finish_reason": finish_reason,
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": len(token_ids),
+                "total_tokens": len(prompt_ids) + len(token_ids),
+            },
+        })
+
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
         body = await request.json()
