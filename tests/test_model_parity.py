@@ -92,3 +92,55 @@ def test_prefix_cache_reuse_same_output(tiny_llama):
     eng_fresh = _build_engine(hf, cfg, enable_prefix_cache=False)
     ref2 = list(eng_fresh.generate([p2], [sp]).values())[0]
     assert out2 == ref2
+
+
+def test_staggered_admission_parity():
+    """Continuous batching: a request admitted while another is mid-decode
+    must not perturb either one's greedy output (mixed prefill+decode
+    batches, the core serving path)."""
+    import transformers
+
+    torch.manual_seed(61)
+    hf_cfg = transformers.LlamaConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=256, rope_theta=10000.0,
+        tie_word_embeddings=False,
+    )
+    hf = transformers.LlamaForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["LlamaForCausalLM"]}
+    )
+    p1, p2 = [7, 42, 99, 5, 81], [9, 8, 7, 6, 5, 4, 3]
+    refs = []
+    with torch.no_grad():
+        for p in (p1, p2):
+            refs.append(hf.generate(torch.tensor([p]), max_new_tokens=8,
+                                    do_sample=False)[0][len(p):].tolist())
+
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=torch.float32))
+    m = get_model_class(cfg.architecture)(cfg).eval()
+    for name, t in hf.state_dict().items():
+        m.load_hf_weight(name, t)
+    eng.model = m.float()
+    sp = SamplingParams(temperature=0.0, max_new_tokens=8, ignore_eos=True)
+    outs = {}
+    r1 = eng.submit(p1, sp)
+    outs[r1] = []
+    # r1 prefills and decodes 3 tokens before r2 arrives
+    for _ in range(4):
+        for o in eng.step():
+            outs[o.rid].append(o.token_id)
+    r2 = eng.submit(p2, sp)
+    outs[r2] = []
+    for _ in range(40):
+        for o in eng.step():
+            outs[o.rid].append(o.token_id)
+        if not eng.has_work:
+            break
+    assert outs[r1] == refs[0]
+    assert outs[r2] == refs[1]

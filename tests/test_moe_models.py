@@ -113,3 +113,37 @@ def test_minimax_m2_parity():
     assert cfg.qk_norm and cfg.qk_norm_full and cfg.scoring_func == "sigmoid"
     assert cfg.is_moe and cfg.num_experts == 8
     _run_parity(hf, cfg)
+
+
+def test_gpt_oss_chunked_prefill_parity():
+    """Sliding-window + sinks with chunked prefill: window masks must span
+    chunk boundaries exactly."""
+    torch.manual_seed(15)
+    hf_cfg = transformers.GptOssConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, num_local_experts=4, num_experts_per_tok=2,
+        sliding_window=8, max_position_embeddings=512, rope_theta=10000.0,
+        layer_types=["sliding_attention", "full_attention"] * 2,
+        tie_word_embeddings=False, attention_bias=True,
+        rope_parameters={"rope_type": "default", "rope_theta": 10000.0},
+    )
+    hf = transformers.GptOssForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["GptOssForCausalLM"]}
+    )
+    prompt = list(range(3, 33))
+    with torch.no_grad():
+        ref = hf.generate(torch.tensor([prompt]), max_new_tokens=4,
+                          do_sample=False)[0][len(prompt):].tolist()
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=torch.float32, prefill_chunk_size=16))
+    m = get_model_class(cfg.architecture)(cfg).eval()
+    for name, t in hf.state_dict().items():
+        m.load_hf_weight(name, t)
+    eng.model = m.float()
+    out = eng.generate(
+        [prompt],
+        [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)],
+    )
+    assert list(out.values())[0] == ref
