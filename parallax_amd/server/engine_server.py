@@ -43,9 +43,12 @@ class EngineServer:
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
-        # serving metrics (parity with reference request_metrics)
+        # serving metrics (parity with reference request_metrics /
+        # SharedState.update_metrics EWMA)
         self.total_requests = 0
         self.total_output_tokens = 0
+        self.step_ms_ewma: Optional[float] = None
+        self.last_batch_tokens = 0
 
     # -- lifecycle ------------------------------------------------------------
 
@@ -91,8 +94,13 @@ class EngineServer:
             if not has_work:
                 time.sleep(self.idle_sleep_s)
                 continue
+            t0 = time.monotonic()
             with self._lock:
                 outputs = self.engine.step()
+            dt_ms = (time.monotonic() - t0) * 1e3
+            self.step_ms_ewma = dt_ms if self.step_ms_ewma is None \
+                else 0.1 * dt_ms + 0.9 * self.step_ms_ewma
+            self.last_batch_tokens = len(outputs)
             for out in outputs:
                 self.total_output_tokens += 1
                 stream = self._streams.get(out.rid)
@@ -112,4 +120,8 @@ class EngineServer:
             "waiting": len(self.engine.scheduler.wait_queue),
             "free_kv_blocks": self.engine.cache_manager.num_free_blocks,
             "prefix_cache_hit_rate": self.engine.cache_manager.radix.hit_rate,
+            "step_ms_ewma": round(self.step_ms_ewma, 3)
+            if self.step_ms_ewma is not None else None,
+            "last_batch_tokens": self.last_batch_tokens,
+            "engine_steps": self.engine.step_count,
         }

@@ -90,3 +90,39 @@ def test_ep2_matches_single(tmp_path):
              nprocs=2, join=True)
     got = torch.load(out_file)
     assert got == expected
+
+
+def test_ep_loader_helpers_route_local_slice():
+    """FusedMoE EP loader helpers: per-expert and fused loads only touch the
+    local expert slice (unit-level, fake 2-way comm)."""
+    from parallax_amd.parallel import comm as comm_mod
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.models.moe import FusedMoE
+
+    cfg = ModelConfig(
+        architecture="Qwen3MoeForCausalLM", model_type="qwen3_moe",
+        vocab_size=64, hidden_size=16, num_layers=1, num_heads=2,
+        num_kv_heads=2, head_dim=8, intermediate_size=32,
+        moe_intermediate_size=8, num_experts=4, num_experts_per_tok=2,
+    )
+    saved = comm_mod._CTX
+    try:
+        comm_mod._CTX = comm_mod.CommContext(
+            world_size=2, rank=1, pp_size=1, tp_size=2, pp_rank=0, tp_rank=1,
+            device=torch.device("cpu"))
+        moe = FusedMoE(cfg)
+        assert moe.ep_size == 2 and moe.ep_rank == 1
+        assert moe.num_local_experts == 2 and moe.expert_offset == 2
+        assert moe.w_gate_up.shape[0] == 2
+        # expert 0/1 are foreign: loads are no-ops; 2/3 land at local 0/1
+        g = torch.randn(8, 16)
+        moe.w_gate_up.data.zero_()
+        moe.load_expert_gate(0, g)
+        assert moe.w_gate_up.abs().sum() == 0
+        moe.load_expert_gate(3, g)
+        assert torch.equal(moe.w_gate_up.data[1, :8], g)
+        full = torch.randn(4, 16, 8)
+        moe.load_fused_down(full)
+        assert torch.equal(moe.w_down.data, full[2:4])
+    finally:
+        comm_mod._CTX = saved
