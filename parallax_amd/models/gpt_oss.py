@@ -14,7 +14,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .config import ModelConfig
-from .llama import LlamaDecoderLayer, LlamaShardModel, RMSNorm
+from .llama import LlamaDecoderLayer, LlamaShardModel
 from .registry import register_model
 
 

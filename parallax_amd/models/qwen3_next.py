@@ -20,7 +20,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import ops
-from ..parallel.layers import MergedColumnParallelLinear, RowParallelLinear
+from ..parallel.layers import MergedColumnParallelLinear
 from .config import ModelConfig
 from .forward_meta import ForwardMeta
 from .llama import LlamaAttention, LlamaMLP, LlamaShardModel, RMSNorm

@@ -12,7 +12,6 @@ for the [num_blocks, num_kv_heads, block_size, head_dim] MI355X layout.
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional, Tuple
 
 import torch

@@ -16,7 +16,7 @@ from typing import Any, Dict, List, Optional, Tuple
 import msgpack
 import torch
 
-from ..server.request import IntermediateRequest, RequestStatus
+from ..server.request import IntermediateRequest
 from ..server.sampling_params import SamplingParams
 
 _DTYPES = {

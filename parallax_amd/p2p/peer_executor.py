@@ -28,7 +28,7 @@ from ..server.kv_cache import (
     build_block_table_tensor,
     slot_mapping_for_positions,
 )
-from ..server.request import InitialRequest, IntermediateRequest, RequestStatus
+from ..server.request import InitialRequest, IntermediateRequest
 from ..server.sampler import Sampler
 from ..server.sampling_params import SamplingParams
 from ..server.scheduler import Scheduler

@@ -37,10 +37,10 @@ from .kv_cache import (
     build_block_table_tensor,
     slot_mapping_for_positions,
 )
-from .request import InitialRequest, RequestStatus, new_request_id
+from .request import InitialRequest, new_request_id
 from .sampler import Sampler
 from .sampling_params import SamplingParams
-from .scheduler import PrefillChunk, ScheduledBatch, Scheduler
+from .scheduler import PrefillChunk, Scheduler
 
 logger = get_logger("server.engine")
 
