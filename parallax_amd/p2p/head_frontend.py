@@ -57,7 +57,8 @@ def create_head_app(
             out = await loop.run_in_executor(None, q.get)
             if out is None:
                 break
-            token_ids.append(out.token_id)
+            if out.token_id >= 0:
+                token_ids.append(out.token_id)
             if out.finished:
                 finish_reason = out.finish_reason or "stop"
         return JSONResponse({
@@ -101,6 +102,8 @@ def create_head_app(
                         break
                     if first_t is None:
                         first_t = time.monotonic()
+                    if out.token_id < 0:
+                        continue
                     token_ids.append(out.token_id)
                     text = tokenizer.decode(token_ids)
                     delta, nl = text[sent_len:], len(text)
@@ -142,7 +145,8 @@ def create_head_app(
                 break
             if first_t is None:
                 first_t = time.monotonic()
-            token_ids.append(out.token_id)
+            if out.token_id >= 0:
+                token_ids.append(out.token_id)
             if out.finished:
                 finish_reason = out.finish_reason or "stop"
         elapsed = time.monotonic() - t0

@@ -280,9 +280,14 @@ class Engine:
             self._pending_aborts or timeout_sweep
         ):
             outputs.extend(self._finalize_inflight())
-        self.scheduler.sweep_aborted()
+        for req in self.scheduler.sweep_aborted():
+            outputs.append(StepOutput(rid=req.rid, token_id=-1, finished=True,
+                                      finish_reason=req.status.finish_reason))
         if timeout_sweep:
-            self.scheduler.sweep_timeouts()
+            for req in self.scheduler.sweep_timeouts():
+                outputs.append(StepOutput(rid=req.rid, token_id=-1,
+                                          finished=True,
+                                          finish_reason=req.status.finish_reason))
         self.scheduler.admit_requests()
         batch = self.scheduler.form_batch()
         self.step_count += 1
@@ -727,7 +732,8 @@ class Engine:
         for _ in range(max_steps):
             for out in self.step():
                 if out.rid in outputs:
-                    outputs[out.rid].append(out.token_id)
+                    if out.token_id >= 0:
+                        outputs[out.rid].append(out.token_id)
                     if out.finished:
                         done[out.rid] = outputs[out.rid]
             # check AFTER the step: step() syncs ingress across ranks, so the

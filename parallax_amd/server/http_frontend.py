@@ -37,22 +37,43 @@ def create_app(
             sp.stop_token_ids = []
         return sp
 
+    def _find_stop(text: str, stops: List[str]):
+        """Earliest stop-string hit in the decoded text, or -1."""
+        best = -1
+        for st in stops:
+            i = text.find(st)
+            if i >= 0 and (best < 0 or i < best):
+                best = i
+        return best
+
     async def _collect(stream, sp: SamplingParams):
-        """Drain a request stream fully (non-streaming path)."""
+        """Drain a request stream fully (non-streaming path). Stop STRINGS
+        are matched here on the detokenized text (token-id stops happen in
+        the engine) — on a hit the request is aborted and the text truncated
+        (reference: frontend-side stop matching)."""
         loop = asyncio.get_event_loop()
         token_ids: List[int] = []
         logprobs: List[float] = []
         finish_reason = "stop"
+        stop_at: int = -1
         while True:
             out = await loop.run_in_executor(None, stream.out_queue.get)
             if out is None:
                 break
+            if out.token_id < 0:  # abort/timeout terminator
+                finish_reason = out.finish_reason or "abort"
+                continue
             token_ids.append(out.token_id)
             if out.logprob is not None:
                 logprobs.append(out.logprob)
+            if sp.stop and stop_at < 0:
+                hit = _find_stop(tokenizer.decode(token_ids), sp.stop)
+                if hit >= 0:
+                    stop_at = hit
+                    server.abort(out.rid)
             if out.finished:
                 finish_reason = out.finish_reason or "stop"
-        return token_ids, finish_reason, logprobs
+        return token_ids, finish_reason, logprobs, stop_at
 
     def _chat_logprobs(token_ids, logprobs):
         """OpenAI chat logprobs block: one entry per sampled token."""
@@ -113,6 +134,8 @@ def create_app(
                         break
                     if first_token_t is None:
                         first_token_t = time.monotonic()
+                    if out.token_id < 0:
+                        continue
                     token_ids.append(out.token_id)
                     text = tokenizer.decode(token_ids)
                     delta, sent_len = text[sent_len:], len(text)
@@ -140,13 +163,17 @@ def create_app(
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        token_ids, finish_reason, logprobs = await _collect(stream, sp)
+        token_ids, finish_reason, logprobs, stop_at = await _collect(stream, sp)
+        content = tokenizer.decode(token_ids)
+        if stop_at >= 0:
+            content = content[:stop_at]
+            finish_reason = "stop"
         return JSONResponse({
             "id": rid, "object": "chat.completion", "created": created,
             "model": model_name,
             "choices": [{
                 "index": 0,
-                "message": {"role": "assistant", "content": tokenizer.decode(token_ids)},
+                "message": {"role": "assistant", "content": content},
                 "logprobs": _chat_logprobs(token_ids, logprobs)
                 if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,
@@ -171,12 +198,16 @@ def create_app(
         sp = _params(body)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         stream = server.submit(prompt_ids, sp, rid=rid)
-        token_ids, finish_reason, logprobs = await _collect(stream, sp)
+        token_ids, finish_reason, logprobs, stop_at = await _collect(stream, sp)
+        text = tokenizer.decode(token_ids)
+        if stop_at >= 0:
+            text = text[:stop_at]
+            finish_reason = "stop"
         return JSONResponse({
             "id": rid, "object": "text_completion", "created": int(time.time()),
             "model": model_name,
             "choices": [{
-                "index": 0, "text": tokenizer.decode(token_ids),
+                "index": 0, "text": text,
                 "logprobs": {"token_logprobs": logprobs}
                 if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,

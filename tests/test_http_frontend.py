@@ -111,3 +111,26 @@ def test_logprobs_roundtrip(client):
     assert lp is not None and len(lp["content"]) == 4
     for e in lp["content"]:
         assert e["logprob"] <= 0.0
+
+
+def test_stop_string_truncates_and_aborts(client):
+    """Stop STRINGS are matched on detokenized text at the frontend: first
+    run greedy unconstrained, then re-run with a stop string taken from the
+    middle of that output — the response must truncate before it."""
+    base = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 8, "temperature": 0.0, "ignore_eos": True,
+    }).json()["choices"][0]["message"]["content"]
+    # tokens decode as "<id> <id> ..."; pick the 4th token's text as the stop
+    parts = base.split(" ")
+    assert len(parts) == 8
+    stop = parts[3]
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 8, "temperature": 0.0, "ignore_eos": True,
+        "stop": stop,
+    }).json()
+    content = r["choices"][0]["message"]["content"]
+    assert stop not in content
+    assert content == base[: base.find(stop)]
+    assert r["choices"][0]["finish_reason"] == "stop"
