@@ -1,5 +1,9 @@
 """hipGraph-captured decode forward.
 
+No reference analogue: the reference delegates decode batching to SGLang/vLLM
+(their CUDA-graph runners, sglang/model_runner.py); this is the MI355X-native
+equivalent designed for this engine's SPMD stage loop.
+
 The decode step of a 32-layer shard launches ~350 kernels; at ~10-20 us of
 launch+gap each that dominates the ~3 ms of real work. This runner captures the
 whole per-stage decode forward (embedding -> layers -> final norm -> lm_head)
