@@ -44,6 +44,14 @@ class NodeAgent:
         self.node_id = node_id or f"node-{new_request_id()[:8]}"
         self.host = host
         self.http_port = http_port
+        if hardware is None and device is not None and device.type == "cuda":
+            from ..scheduling.node import detect_hardware
+
+            hw = detect_hardware()
+            hardware = {"name": hw.name, "num_gpus": hw.num_gpus,
+                        "memory_gb": hw.memory_gb,
+                        "tflops_bf16": hw.tflops_bf16,
+                        "memory_bandwidth_gbps": hw.memory_bandwidth_gbps}
         self.hardware = hardware or {"name": "MI355X", "num_gpus": 1,
                                      "memory_gb": 288.0}
         self.model_path = model_path

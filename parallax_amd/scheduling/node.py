@@ -35,6 +35,27 @@ MI355X = NodeHardware()
 MI355X_x8 = NodeHardware(name="MI355X x8", num_gpus=8)
 
 
+def detect_hardware() -> NodeHardware:
+    """Probe the local machine (reference server_info.HardwareInfo.detect):
+    on a ROCm GPU read name/VRAM from torch and assume MI355X-class compute
+    ratios scaled by memory; CPU fallback is a tiny test profile."""
+    import torch
+
+    if torch.cuda.is_available():
+        props = torch.cuda.get_device_properties(0)
+        mem_gb = props.total_memory / (1 << 30)
+        scale = mem_gb / 288.0
+        return NodeHardware(
+            name=props.name or "ROCm GPU",
+            num_gpus=torch.cuda.device_count(),
+            memory_gb=mem_gb,
+            tflops_bf16=2500.0 * max(scale, 0.05),
+            memory_bandwidth_gbps=8000.0 * max(scale, 0.05),
+        )
+    return NodeHardware(name="cpu-test", num_gpus=1, memory_gb=16.0,
+                        tflops_bf16=1.0, memory_bandwidth_gbps=50.0)
+
+
 @dataclass
 class Node:
     node_id: str

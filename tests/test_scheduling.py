@@ -281,3 +281,12 @@ def test_allocation_fails_cleanly_when_undercapacity():
     set_full_rtt(nodes)
     assert GreedyLayerAllocator(model).allocate_from_standby(nodes) == []
     assert water_fill_layers(nodes, model, 80) is None
+
+
+def test_detect_hardware_cpu_profile():
+    from parallax_amd.scheduling.node import detect_hardware
+
+    hw = detect_hardware()
+    assert hw.memory_gb > 0 and hw.tflops_bf16 > 0
+    n = Node(node_id="d", hardware=hw)
+    assert n.decoder_layer_capacity(llama70b_info()) >= 0
