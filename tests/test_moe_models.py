@@ -147,3 +147,37 @@ def test_gpt_oss_chunked_prefill_parity():
         [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)],
     )
     assert list(out.values())[0] == ref
+
+
+def test_glm4_moe_chunked_prefill_parity():
+    """Partial rotary (0.5) + sigmoid MoE with page-aligned chunked prefill."""
+    torch.manual_seed(12)
+    hf_cfg = transformers.Glm4MoeConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        moe_intermediate_size=32, num_hidden_layers=3,
+        num_attention_heads=4, num_key_value_heads=2, head_dim=16,
+        n_routed_experts=4, num_experts_per_tok=2, n_shared_experts=1,
+        n_group=1, topk_group=1, first_k_dense_replace=1,
+        norm_topk_prob=True, routed_scaling_factor=1.0,
+        max_position_embeddings=512, use_qk_norm=True,
+        tie_word_embeddings=False, attention_bias=False,
+    )
+    hf = transformers.Glm4MoeForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["Glm4MoeForCausalLM"]}
+    )
+    prompt = list(range(3, 31))
+    with torch.no_grad():
+        ref = hf.generate(torch.tensor([prompt]), max_new_tokens=4,
+                          do_sample=False)[0][len(prompt):].tolist()
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=torch.float32, prefill_chunk_size=16))
+    m = get_model_class(cfg.architecture)(cfg).eval()
+    for name, t in hf.state_dict().items():
+        m.load_hf_weight(name, t)
+    eng.model = m.float()
+    out = eng.generate(
+        [prompt],
+        [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)],
+    )
+    assert list(out.values())[0] == ref
