@@ -49,3 +49,25 @@ def test_bench_shared_prefix_contract():
     assert r.returncode == 0, r.stderr[-2000:]
     j = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][0])
     assert j["config"]["shared_prefix"] == 64
+
+
+@pytest.mark.parametrize("mode,expect", [("auto", "dp2"), ("pp", "pp2")])
+def test_bench_two_rank_contract(mode, expect):
+    """The driver's SCALE runs launch bench.py under torch.distributed.run:
+    both the DP-replica mode (auto for models that fit one GPU) and forced PP
+    must produce the contract line with the right parallelism tag."""
+    port = 29750 + (0 if mode == "auto" else 1)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "bench.py", "--cpu", "--steps", "2",
+         "--warmup", "1", "--batch-per-gpu", "4", "--parallelism", mode],
+        capture_output=True, text=True, timeout=600, cwd=".",
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    j = json.loads(lines[0])
+    assert j["config"]["parallelism"] == expect
+    assert j["config"]["global_batch"] == 8
+    assert j["n_gpus"] == 0  # cpu run
