@@ -5,6 +5,7 @@ bookkeeping, no tensors needed.
 """
 
 import pytest
+import torch
 
 from parallax_amd.server.allocator import BlockAllocator, OutOfBlocksError
 from parallax_amd.server.cache_manager import CacheManager
@@ -87,3 +88,43 @@ def test_cache_manager_decode_growth():
     assert len(cm.get("r1").block_table) == 2
     cm.append_tokens("r1", 5)  # idempotent
     assert len(cm.get("r1").block_table) == 2
+
+
+def test_linear_state_cache_snapshot_restore():
+    from parallax_amd.server.kv_cache import LinearStateCache
+
+    c = LinearStateCache(num_layers=2, conv_state_shape=(6, 3),
+                         recurrent_state_shape=(2, 4, 4), num_slots=4,
+                         device=torch.device("cpu"), dtype=torch.float32)
+    for l in range(2):
+        c.conv_states[l][1].fill_(l + 1.0)
+        c.recurrent_states[l][1].fill_(l + 10.0)
+    snap = c.snapshot(1)
+    c.reset_slot(1)
+    assert c.conv_states[0][1].abs().sum() == 0
+    c.restore(1, snap)
+    assert torch.all(c.conv_states[1][1] == 2.0)
+    assert torch.all(c.recurrent_states[0][1] == 10.0)
+    # snapshot is a deep copy: later mutation doesn't corrupt it
+    c.conv_states[0][1].fill_(99.0)
+    c.restore(1, snap)
+    assert torch.all(c.conv_states[0][1] == 1.0)
+
+
+def test_mla_cache_index_trash_block():
+    """The +1 trash block convention: store with slot -1 lands in the last
+    block and never corrupts addressable blocks."""
+    from parallax_amd.server.kv_cache import MLAKVCache
+    from parallax_amd.ops import reference as ref
+
+    c = MLAKVCache(num_layers=1, kv_lora_rank=8, rope_dim=4, block_size=4,
+                   num_blocks=3, device=torch.device("cpu"),
+                   dtype=torch.float32, index_dim=8)
+    idx = c.index_layer(0)
+    assert idx.shape[0] == 4  # 3 + trash
+    keys = torch.randn(2, 8)
+    ref.store_indexer_cache(keys, idx, torch.tensor([5, -1]))
+    assert torch.equal(idx[1, 1], keys[0])
+    # the pad token went to the trash block, not block 0..2
+    assert idx[:3].abs().sum() == keys[0].abs().sum()
+    assert idx[3].abs().sum() > 0
