@@ -23,6 +23,15 @@ from .node_agent import NodeAgent
 logger = get_logger("p2p.head_frontend")
 
 
+def _find_stop(text: str, stops) -> int:
+    best = -1
+    for st in stops or []:
+        i = text.find(st)
+        if i >= 0 and (best < 0 or i < best):
+            best = i
+    return best
+
+
 def create_head_app(
     agent: NodeAgent, tokenizer: TokenizerWrapper, model_name: str = "model"
 ) -> FastAPI:
@@ -139,6 +148,7 @@ def create_head_app(
         token_ids, finish_reason = [], "stop"
         t0 = time.monotonic()
         first_t = None
+        stop_at = -1
         while True:
             out = await loop.run_in_executor(None, q.get)
             if out is None:
@@ -147,6 +157,11 @@ def create_head_app(
                 first_t = time.monotonic()
             if out.token_id >= 0:
                 token_ids.append(out.token_id)
+                if sp.stop and stop_at < 0:
+                    hit = _find_stop(tokenizer.decode(token_ids), sp.stop)
+                    if hit >= 0:
+                        stop_at = hit
+                        agent.abort(out.rid) if hasattr(agent, "abort") else None
             if out.finished:
                 finish_reason = out.finish_reason or "stop"
         elapsed = time.monotonic() - t0
@@ -158,8 +173,9 @@ def create_head_app(
             "choices": [{
                 "index": 0,
                 "message": {"role": "assistant",
-                            "content": tokenizer.decode(token_ids)},
-                "finish_reason": finish_reason,
+                            "content": tokenizer.decode(token_ids)[:stop_at]
+                            if stop_at >= 0 else tokenizer.decode(token_ids)},
+                "finish_reason": "stop" if stop_at >= 0 else finish_reason,
             }],
             "usage": {
                 "prompt_tokens": len(prompt_ids),

@@ -133,6 +133,10 @@ class NodeAgent:
         t1.start()
         t2.start()
 
+    def abort(self, rid: str) -> None:
+        if self.executor is not None and self.executor.is_head:
+            self.executor.abort(rid)
+
     def stop(self) -> None:
         self._stop.set()
         for t in self._threads:

@@ -164,6 +164,12 @@ class PeerExecutor:
         return req.rid
 
     @property
+    def abort(self, rid: str) -> None:
+        """Client-requested abort (head only): swept on the next step."""
+        req = self.scheduler.running.get(rid) if self.scheduler else None
+        if req is not None:
+            req.abort_requested = True
+
     def has_work(self) -> bool:
         if self.is_head:
             return self.scheduler.has_work
@@ -206,7 +212,14 @@ class PeerExecutor:
                                                 finished.routing_table)
             recv_timeout = 0.0  # only block on the first recv
 
-        # 2. schedule local work
+        # 2. abort sweep: release downstream state and terminate streams
+        for req in self.scheduler.sweep_aborted():
+            self.finished_outputs.append(
+                PeerOutput(req.rid, -1, True, req.status.finish_reason)
+            )
+            self._broadcast_control("release", [req.rid], req.routing_table)
+
+        # 3. schedule local work
         self.scheduler.admit_requests()
         batch = self.scheduler.form_batch()
         if batch.is_empty:
