@@ -163,13 +163,13 @@ class PeerExecutor:
         self.scheduler.add_request(req)
         return req.rid
 
-    @property
     def abort(self, rid: str) -> None:
         """Client-requested abort (head only): swept on the next step."""
         req = self.scheduler.running.get(rid) if self.scheduler else None
         if req is not None:
             req.abort_requested = True
 
+    @property
     def has_work(self) -> bool:
         if self.is_head:
             return self.scheduler.has_work

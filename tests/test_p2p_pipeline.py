@@ -177,3 +177,44 @@ def test_p2p_hybrid_linear_stack():
     expect = list(eng.generate(prompts, [sp] * 2).values())
     got = _run_p2p(cfg, sd, 2, prompts)
     assert got == expect
+
+
+def test_p2p_abort_releases_downstream():
+    """Head-side abort mid-generation: the client stream gets a terminator
+    PeerOutput (token_id -1) and the downstream peer's cache state is
+    released via the broadcast."""
+    cfg = tiny_cfg()
+    registry = {}
+    sd = full_state_dict(cfg)
+    peers = []
+    for pid, (s, e) in zip(["h", "t"], [(0, 2), (2, 4)]):
+        t = LoopbackTransport(pid, registry)
+        px = PeerExecutor(cfg, s, e, pid, t, dtype=torch.float32,
+                          num_kv_blocks=128, block_size=8)
+        for name, w in sd.items():
+            px.model.load_hf_weight(name, w)
+        peers.append(px)
+    head, tail = peers
+    sp = SamplingParams(temperature=0.0, max_new_tokens=64, ignore_eos=True)
+    rid = head.submit([1, 2, 3, 4], sp, ["h", "t"])
+    toks = []
+    finished = None
+    for i in range(400):
+        for px in peers:
+            px.step(recv_timeout=0.001)
+        for out in head.drain_outputs():
+            if out.token_id >= 0:
+                toks.append(out.token_id)
+            if out.finished:
+                finished = out
+        if len(toks) >= 5 and finished is None:
+            head.abort(rid)
+        if finished is not None:
+            # let release broadcasts drain
+            for _ in range(20):
+                for px in peers:
+                    px.step(recv_timeout=0.001)
+            break
+    assert finished is not None and finished.finish_reason == "abort"
+    assert 5 <= len(toks) < 64
+    assert rid not in tail._peer_positions  # downstream state released
