@@ -31,6 +31,9 @@ sources = [s for s in sources if os.path.exists(s)]
 
 setup(
     name="parallax_amd_ext",
+    entry_points={
+        "console_scripts": ["parallax_amd = parallax_amd.cli:main"],
+    },
     ext_modules=[
         CUDAExtension(
             name="parallax_amd.ops._C",
