@@ -178,8 +178,7 @@ async function load(){
 load(); setInterval(load, 3000);
 </script></body></html>""")
 
-    @app.get("/cluster/status")
-    async def cluster_status():
+    def _status() -> dict:
         if svc.scheduler is None:
             return {"initialized": False}
         with svc._lock:
@@ -187,6 +186,30 @@ load(); setInterval(load, 3000);
         status["initialized"] = True
         status["endpoints"] = svc.node_endpoints
         return status
+
+    @app.get("/cluster/status")
+    async def cluster_status():
+        return _status()
+
+    @app.get("/cluster/status_stream")
+    async def cluster_status_stream(interval_s: float = 2.0, count: int = 0):
+        """NDJSON status stream (reference /cluster/status NDJSON,
+        backend/main.py:172-193): one JSON line per tick; count=0 streams
+        until the client disconnects."""
+        import json as _json
+
+        from fastapi.responses import StreamingResponse
+
+        async def ndjson():
+            n = 0
+            while True:
+                yield _json.dumps(_status()) + "\n"
+                n += 1
+                if count and n >= count:
+                    break
+                await asyncio.sleep(interval_s)
+
+        return StreamingResponse(ndjson(), media_type="application/x-ndjson")
 
     @app.post("/weight/refit")
     async def weight_refit():

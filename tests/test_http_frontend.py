@@ -134,3 +134,17 @@ def test_stop_string_truncates_and_aborts(client):
     assert stop not in content
     assert content == base[: base.find(stop)]
     assert r["choices"][0]["finish_reason"] == "stop"
+
+
+def test_cluster_status_stream_ndjson():
+    import json as _json
+
+    from parallax_amd.backend.service import SchedulerService, create_backend_app
+
+    app = create_backend_app(SchedulerService())
+    with TestClient(app) as c:
+        with c.stream("GET", "/cluster/status_stream?count=2&interval_s=0.01") as r:
+            lines = [l for l in r.iter_lines() if l.strip()]
+    assert len(lines) == 2
+    for l in lines:
+        assert _json.loads(l)["initialized"] is False
