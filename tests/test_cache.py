@@ -128,3 +128,21 @@ def test_mla_cache_index_trash_block():
     # the pad token went to the trash block, not block 0..2
     assert idx[:3].abs().sum() == keys[0].abs().sum()
     assert idx[3].abs().sum() > 0
+
+
+def test_radix_does_not_evict_referenced_blocks():
+    """Blocks still referenced by a live request must survive LRU pressure:
+    eviction only drops the TREE's refcount; the block returns to the free
+    list only when the owning request also releases it."""
+    alloc = BlockAllocator(8, block_size=4)
+    radix = BlockRadixCache(block_size=4, allocator=alloc)
+    a_blocks = alloc.allocate(2)          # live request's hold (rc=1)
+    radix.insert(list(range(8)), a_blocks)  # tree's hold (rc=2)
+    # blocks still referenced by the request are NOT evictable at all
+    assert radix.evict(10) == 0
+    assert alloc.num_free_blocks == 6
+    # request releases its hold -> the leaf becomes evictable, LRU order
+    for b in a_blocks:
+        alloc.decref(b)
+    assert radix.evict(10) == 2
+    assert alloc.num_free_blocks == 8
