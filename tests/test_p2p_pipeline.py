@@ -218,3 +218,34 @@ def test_p2p_abort_releases_downstream():
     assert finished is not None and finished.finish_reason == "abort"
     assert 5 <= len(toks) < 64
     assert rid not in tail._peer_positions  # downstream state released
+
+
+def test_codec_large_tensor_and_empty_batch():
+    """Wire codec edge cases: multi-MB bf16 hidden states round-trip intact;
+    empty forward batches survive."""
+    h = torch.randn(2048, 1024, dtype=torch.bfloat16)
+    req = IntermediateRequest(
+        rid="big", routing_table=["a"], current_position=0,
+        num_new_tokens=2048, is_prefill=True, hidden_states=h,
+    )
+    msg = codec.decode(codec.encode_forward([req]))
+    assert torch.equal(msg["reqs"][0].hidden_states, h)
+    empty = codec.decode(codec.encode_forward([]))
+    assert empty["kind"] == "forward" and empty["reqs"] == []
+
+
+def test_tcp_transport_many_frames_in_order():
+    from parallax_amd.p2p.transport import TcpTransport
+
+    a = TcpTransport("a", "127.0.0.1", 0)
+    b = TcpTransport("b", "127.0.0.1", 0)
+    a.set_peer_addr("b", "127.0.0.1", b.port)
+    payloads = [bytes([i]) * (1000 * (i + 1)) for i in range(20)]
+    for p in payloads:
+        a.send("b", p)
+    got = []
+    for _ in range(20):
+        got.append(b.recv(timeout=5))
+    assert got == payloads
+    a.close()
+    b.close()
