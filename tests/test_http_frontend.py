@@ -148,3 +148,15 @@ def test_cluster_status_stream_ndjson():
     assert len(lines) == 2
     for l in lines:
         assert _json.loads(l)["initialized"] is False
+
+
+def test_tokenizer_wrapper_synthetic_roundtrip():
+    from parallax_amd.server.tokenizer_util import TokenizerWrapper
+
+    tok = TokenizerWrapper(vocab_size=128)
+    ids = tok.encode("hello")
+    assert all(0 <= i < 128 for i in ids)
+    text = tok.decode(ids)
+    assert text  # decodable
+    chat_ids = tok.chat_prompt_ids([{"role": "user", "content": "hi"}])
+    assert chat_ids and all(0 <= i < 128 for i in chat_ids)
