@@ -77,3 +77,41 @@ def test_shard_loader_selective_files(tmp_path):
         json.dump(idx, f)
     files = selective_file_list(str(tmp_path / "m"), 1, 2)
     assert "b.safetensors" in files and "a.safetensors" in files  # endpoints travel
+
+
+def test_http_update_weights_endpoint(tmp_path):
+    """POST /update_weights reloads the checkpoint through the serving stack
+    (reference /weight/refit -> node reload path, single-host variant)."""
+    from fastapi.testclient import TestClient
+
+    from parallax_amd.server.engine_server import EngineServer
+    from parallax_amd.server.http_frontend import create_app
+    from parallax_amd.server.tokenizer_util import TokenizerWrapper
+
+    torch.manual_seed(5)
+    hf_cfg = transformers.LlamaConfig(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, rope_theta=10000.0,
+        tie_word_embeddings=False,
+    )
+    hf = transformers.LlamaForCausalLM(hf_cfg).eval()
+    ckpt = str(tmp_path / "ckpt")
+    save_checkpoint(ckpt, hf, hf_cfg)
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["LlamaForCausalLM"]}
+    )
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=64,
+                                 dtype=torch.float32),
+                 model_path=ckpt)
+    server = EngineServer(eng)
+    server.start()
+    tok = TokenizerWrapper(vocab_size=cfg.vocab_size)
+    app = create_app(server, tok, "tiny")
+    try:
+        with TestClient(app) as c:
+            r = c.post("/update_weights", json={"model_path": ckpt})
+            assert r.status_code == 200
+            assert r.json()["updated_tensors"] > 0
+    finally:
+        server.stop()
