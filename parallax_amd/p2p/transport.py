@@ -60,8 +60,14 @@ class TcpTransport(Transport):
     (reconnect on failure). peer addresses are set via set_peer_addr (from the
     scheduler's cluster view)."""
 
-    def __init__(self, peer_id: str, host: str = "0.0.0.0", port: int = 0):
+    #: refuse frames claiming more than this many bytes (a corrupt or malicious
+    #: peer must not be able to drive unbounded allocation on the receiver)
+    DEFAULT_MAX_FRAME_BYTES = 512 * 1024 * 1024
+
+    def __init__(self, peer_id: str, host: str = "0.0.0.0", port: int = 0,
+                 max_frame_bytes: int = DEFAULT_MAX_FRAME_BYTES):
         self.peer_id = peer_id
+        self.max_frame_bytes = max_frame_bytes
         self.inbox: "queue.Queue[bytes]" = queue.Queue()
         self._peers: Dict[str, tuple] = {}
         self._conns: Dict[str, socket.socket] = {}
@@ -122,6 +128,12 @@ class TcpTransport(Transport):
                 if header is None:
                     break
                 (n,) = struct.unpack("<Q", header)
+                if n > self.max_frame_bytes:
+                    logger.warning(
+                        "dropping connection: frame of %d bytes exceeds max %d",
+                        n, self.max_frame_bytes,
+                    )
+                    break
                 payload = self._read_exact(conn, n)
                 if payload is None:
                     break

@@ -34,7 +34,9 @@ def create_app(
     def _params(body: dict) -> SamplingParams:
         sp = SamplingParams.from_openai(body)
         if tokenizer.eos_token_id is not None and not sp.stop_token_ids:
-            sp.stop_token_ids = []
+            # default stop tokens to the tokenizer's EOS (it may differ from the
+            # model-config eos_token_ids the engine applies)
+            sp.stop_token_ids = [tokenizer.eos_token_id]
         return sp
 
     def _find_stop(text: str, stops: List[str]):

@@ -206,8 +206,15 @@ class Scheduler:
 
     def _release(self, req: InitialRequest) -> None:
         self.running.pop(req.rid, None)
-        # publish the full sequence for future prefix hits, then drop our hold
-        self.cache.free_request(req.rid, req.all_token_ids)
+        # Publish only tokens whose KV was actually computed (the reference
+        # bounds insertion by context_len, cache_manager.insert_full_blocks_to_cache):
+        # the final sampled token is never forwarded so its KV slot is unwritten,
+        # and a request aborted mid-prefill has no complete KV at all.
+        if req.prefill_done:
+            computed = req.num_prefilled_tokens + max(0, req.num_output_tokens - 1)
+            self.cache.free_request(req.rid, req.all_token_ids[:computed])
+        else:
+            self.cache.free_request(req.rid, None)
         self.finished_reqs.append(req)
 
     # -- introspection ----------------------------------------------------------------
