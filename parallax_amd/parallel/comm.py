@@ -4,8 +4,9 @@ CPU tests.
 MI355X-first design: one process per GPU; the world is laid out
 as pp_size x tp_size with rank = pp_rank * tp_size + tp_rank. Pipeline
 hidden-state transport is torch.distributed send/recv (RCCL point-to-point over
-xGMI; ~153 GB/s per link) on the default stream for now, with a dedicated HIP
-comm stream planned for overlap. TP all-reduce uses the per-stage group.
+xGMI; ~153 GB/s per link); the async variants (pp_irecv/pp_send_async) ride the
+process group's internal comm stream so micro-batch sends overlap the next
+group's compute. TP all-reduce uses the per-stage group.
 Replaces the reference's NCCL-inside-SGLang init (sglang/model_runner.py:97-218)
 and its Lattica hidden-state RPC for the in-host path.
 """
@@ -82,6 +83,53 @@ class CommContext:
         t = torch.empty(shape, dtype=dtype, device=self.device)
         dist.recv(t, self.stage_rank(src_pp_rank))
         return t
+
+    # -- PP async point-to-point (micro-batch pipelining) ---------------------
+    #
+    # With the RCCL backend, isend/irecv enqueue on the process group's
+    # internal comm stream (ordered after the caller's current stream), and
+    # Work.wait() inserts a stream dependency rather than blocking the host —
+    # so a stage can compute micro-batch i while micro-batch i-1 is in flight
+    # over xGMI and micro-batch i+1's recv is already posted.
+
+    def pp_irecv(self, shape, dtype: torch.dtype, src_pp_rank: int):
+        """Post a receive; returns (buffer, work). Call work.wait() before
+        reading the buffer (stream-ordered on RCCL, host-blocking on gloo)."""
+        t = torch.empty(shape, dtype=dtype, device=self.device)
+        work = dist.irecv(t, self.stage_rank(src_pp_rank))
+        return t, work
+
+    def pp_send_async(self, t: torch.Tensor, dst_pp_rank: int, slot: int = 0) -> None:
+        """Copy `t` into a per-slot staging buffer and isend it. The caller's
+        stream only pays for the staging copy; the send itself overlaps the
+        next micro-batch's compute. `slot` keys the staging buffer (use the
+        micro-batch index) so a graph-output buffer can be reused immediately
+        after this returns."""
+        if not hasattr(self, "_send_slots"):
+            self._send_slots = {}
+        t = t.contiguous()
+        key = (dst_pp_rank, slot)
+        prev = self._send_slots.get(key)
+        buf = None
+        if prev is not None:
+            pwork, pbuf = prev
+            if pwork is not None:
+                pwork.wait()  # slot reuse: previous send must drain first
+            if pbuf.shape == t.shape and pbuf.dtype == t.dtype:
+                buf = pbuf
+        if buf is None:
+            buf = torch.empty_like(t)
+        buf.copy_(t)
+        work = dist.isend(buf, self.stage_rank(dst_pp_rank))
+        self._send_slots[key] = (work, buf)
+
+    def pp_flush_sends(self) -> None:
+        """Wait (stream-ordered) for all outstanding async sends."""
+        for key, (work, _buf) in list(getattr(self, "_send_slots", {}).items()):
+            if work is not None:
+                work.wait()
+                w, b = self._send_slots[key]
+                self._send_slots[key] = (None, b)
 
     def barrier(self) -> None:
         if dist.is_initialized():

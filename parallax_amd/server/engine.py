@@ -229,6 +229,11 @@ class Engine:
         self._async_pin_idx = 0
         self._pending_aborts: List[str] = []
         self.step_count = 0
+        # device-tensor PP sync state (no host-object broadcasts in the step path)
+        self._flag_buf: Optional[torch.Tensor] = None
+        self._tok_bcast_buf: Optional[torch.Tensor] = None
+        self._lp_bcast_buf: Optional[torch.Tensor] = None
+        self.object_sync_count = 0  # payload broadcasts (adds/aborts only)
         logger.info(
             "engine up: layers [%d,%d) of %d, %d KV blocks x %d tokens, mla=%s, device %s",
             start, end, cfg.num_layers, num_blocks, args.block_size,
@@ -467,29 +472,58 @@ class Engine:
     # -- ingress replication ----------------------------------------------------------
 
     def _sync_ingress(self) -> None:
+        """Replicate pending adds/aborts to every rank.
+
+        Steady-state decode pays exactly ONE tiny device-tensor broadcast (the
+        payload-length flag, usually 0) — never an object broadcast. Only when
+        requests actually arrive or abort does a second uint8-tensor broadcast
+        carry the serialized payload (both are RCCL device collectives on GPU;
+        the reference's per-hop pickle+RPC cost, p2p/server.py:628-755, has no
+        analogue here). ``object_sync_count`` counts payload broadcasts so
+        tests can assert the decode hot path stays object-free."""
         if self.comm.world_size > 1:
+            import pickle
+
+            if self._flag_buf is None:
+                self._flag_buf = torch.zeros(
+                    1, dtype=torch.int64, device=self.device
+                )
+            payload_bytes = b""
             if self.comm.rank == 0:
-                payload = [
-                    [
-                        (r.rid, r.prompt_token_ids, r.sampling_params.to_dict())
-                        for r in self._pending_adds
-                    ],
-                    list(self._pending_aborts),
-                ]
-            else:
-                payload = [None, None]
-            dist.broadcast_object_list(payload, src=0)
-            if self.comm.rank != 0:
-                adds, aborts = payload
-                self._pending_adds = [
-                    InitialRequest(
-                        rid=rid,
-                        prompt_token_ids=toks,
-                        sampling_params=SamplingParams.from_dict(sp),
+                if self._pending_adds or self._pending_aborts:
+                    payload_bytes = pickle.dumps(
+                        (
+                            [
+                                (r.rid, r.prompt_token_ids,
+                                 r.sampling_params.to_dict())
+                                for r in self._pending_adds
+                            ],
+                            list(self._pending_aborts),
+                        )
                     )
-                    for rid, toks, sp in adds
-                ]
-                self._pending_aborts = list(aborts)
+                self._flag_buf.fill_(len(payload_bytes))
+            dist.broadcast(self._flag_buf, src=0)
+            n = int(self._flag_buf.item())
+            if n > 0:
+                self.object_sync_count += 1
+                if self.comm.rank == 0:
+                    buf = torch.frombuffer(
+                        bytearray(payload_bytes), dtype=torch.uint8
+                    ).to(self.device)
+                else:
+                    buf = torch.empty(n, dtype=torch.uint8, device=self.device)
+                dist.broadcast(buf, src=0)
+                if self.comm.rank != 0:
+                    adds, aborts = pickle.loads(bytes(buf.cpu().numpy()))
+                    self._pending_adds = [
+                        InitialRequest(
+                            rid=rid,
+                            prompt_token_ids=toks,
+                            sampling_params=SamplingParams.from_dict(sp),
+                        )
+                        for rid, toks, sp in adds
+                    ]
+                    self._pending_aborts = list(aborts)
         for req in self._pending_adds:
             self.scheduler.add_request(req)
         for rid in self._pending_aborts:
@@ -603,12 +637,22 @@ class Engine:
             logits = self._pipeline_forward(meta, input_ids, need)
         return logits, samp_reqs
 
-    def _decode_one(self, reqs: List[InitialRequest]) -> Optional[torch.Tensor]:
-        """One decode forward for one (micro-)batch, graph-captured on GPU."""
+    def _decode_fwd(
+        self, reqs: List[InitialRequest], hidden_in: Optional[torch.Tensor]
+    ) -> torch.Tensor:
+        """One decode forward for one (micro-)batch — compute only, no PP
+        comm. Returns hidden states (intermediate stages) or logits (last
+        stage); graph-captured on GPU."""
         if self.graph_runner is None:
             meta, ids = self._build_decode_meta(reqs)
-            return self._pipeline_forward(meta, ids, meta.logits_indices)
-        comm = self.comm
+            if self.comm.is_first_stage:
+                hidden = self.model.embed(ids).to(self.args.dtype)
+            else:
+                hidden = hidden_in
+            hidden = self.model(hidden, meta)
+            if self.comm.is_last_stage:
+                return self.model.compute_logits(hidden[meta.logits_indices])
+            return hidden
         input_ids, positions, slots, btabs, seq_lens = [], [], [], [], []
         for r in reqs:
             state = self.cache_manager.get(r.rid)
@@ -621,20 +665,25 @@ class Engine:
             )
             btabs.append(state.block_table)
             seq_lens.append(r.total_len)
-        hidden_in = None
-        if not comm.is_first_stage:
-            hidden_in = comm.pp_recv(
-                (len(reqs), self.cfg.hidden_size), self.args.dtype, comm.pp_rank - 1
-            )
         lin_slots = None
         if self.linear_cache is not None:
             lin_slots = [
                 self.cache_manager.get(r.rid).linear_slot or 0 for r in reqs
             ]
-        out = self.graph_runner.run(input_ids, positions, slots, btabs, seq_lens,
-                                    hidden_in=hidden_in,
-                                    rids=[r.rid for r in reqs],
-                                    linear_slots=lin_slots)
+        return self.graph_runner.run(input_ids, positions, slots, btabs, seq_lens,
+                                     hidden_in=hidden_in,
+                                     rids=[r.rid for r in reqs],
+                                     linear_slots=lin_slots)
+
+    def _decode_one(self, reqs: List[InitialRequest]) -> Optional[torch.Tensor]:
+        """Single-micro-batch decode with blocking PP transport."""
+        comm = self.comm
+        hidden_in = None
+        if not comm.is_first_stage:
+            hidden_in = comm.pp_recv(
+                (len(reqs), self.cfg.hidden_size), self.args.dtype, comm.pp_rank - 1
+            )
+        out = self._decode_fwd(reqs, hidden_in)
         if not comm.is_last_stage:
             comm.pp_send(out, comm.pp_rank + 1)
             return None
@@ -645,14 +694,33 @@ class Engine:
         with torch.inference_mode():
             if mb == 1 or len(reqs) < mb:
                 return self._decode_one(reqs)
-            # micro-batch pipelining: split the decode batch, pipeline the chunks
+            # Micro-batch pipelining with comm/compute overlap: all receives
+            # are pre-posted (they land in issue order per RCCL peer pair),
+            # and each group's output is staged + isent on the comm stream so
+            # this stage starts group i+1's forward while group i's hidden
+            # states are still in flight over xGMI.
             groups: List[List[InitialRequest]] = [list(x) for x in _split(reqs, mb)]
+            comm = self.comm
+            h = self.cfg.hidden_size
+            recvs: List[tuple] = []
+            if not comm.is_first_stage:
+                for g in groups:
+                    recvs.append(
+                        comm.pp_irecv((len(g), h), self.args.dtype, comm.pp_rank - 1)
+                    )
             logits_parts = []
-            for g in groups:
-                out = self._decode_one(g)
-                if out is not None:
+            for i, g in enumerate(groups):
+                hidden_in = None
+                if not comm.is_first_stage:
+                    buf, work = recvs[i]
+                    work.wait()  # stream-ordered on RCCL
+                    hidden_in = buf
+                out = self._decode_fwd(g, hidden_in)
+                if not comm.is_last_stage:
+                    comm.pp_send_async(out, comm.pp_rank + 1, slot=i)
+                else:
                     logits_parts.append(out.clone())  # graph output buffer is reused
-            if self.comm.is_last_stage:
+            if comm.is_last_stage:
                 return torch.cat(logits_parts, dim=0)
             return None
 
@@ -661,23 +729,47 @@ class Engine:
     def _sample_and_broadcast(
         self, logits_parts: List[torch.Tensor], sample_reqs: List[InitialRequest]
     ) -> List[tuple]:
-        """Returns [(token_id, logprob-or-None)] for each sampled request."""
+        """Returns [(token_id, logprob-or-None)] for each sampled request.
+
+        PP>1: the last stage samples on-device and the token ids travel as ONE
+        int64 device-tensor RCCL broadcast (plus a float tensor only when a
+        request asked for logprobs) — no pickled host objects on the decode
+        critical path. Every rank's replicated scheduler state agrees on the
+        request order and count, so shapes need no negotiation."""
         comm = self.comm
         if comm.pp_size == 1:
             logits = torch.cat(logits_parts, dim=0)
             return self.sampler.sample_with_logprobs(logits, sample_reqs)
-        sampled: List[Optional[tuple]]
+        B = len(sample_reqs)
+        need_lp = any(r.sampling_params.logprobs for r in sample_reqs)
+        if self._tok_bcast_buf is None or self._tok_bcast_buf.numel() < B:
+            cap = max(B, self.args.max_batch_size)
+            self._tok_bcast_buf = torch.zeros(
+                cap, dtype=torch.int64, device=self.device
+            )
+            self._lp_bcast_buf = torch.zeros(
+                cap, dtype=torch.float32, device=self.device
+            )
         if comm.is_last_stage and comm.tp_rank == 0:
             logits = torch.cat(logits_parts, dim=0)
-            sampled = self.sampler.sample_with_logprobs(logits, sample_reqs)
-        else:
-            sampled = [None] * len(sample_reqs)
-        if comm.world_size > 1:
-            payload = [sampled]
-            src = comm.stage_rank(comm.pp_size - 1) - comm.tp_rank  # tp_rank 0 of last stage
-            dist.broadcast_object_list(payload, src=src)
-            sampled = payload[0]
-        return sampled  # type: ignore[return-value]
+            tokens_dev, lp_dev = self.sampler.sample_device(
+                logits, sample_reqs, want_logprobs=need_lp
+            )
+            self._tok_bcast_buf[:B].copy_(tokens_dev)
+            if need_lp:
+                self._lp_bcast_buf[:B].copy_(lp_dev)
+        src = comm.stage_rank(comm.pp_size - 1) - comm.tp_rank  # tp_rank 0 of last stage
+        dist.broadcast(self._tok_bcast_buf[:B], src=src)
+        if need_lp:
+            dist.broadcast(self._lp_bcast_buf[:B], src=src)
+        toks = self._tok_bcast_buf[:B].tolist()
+        if not need_lp:
+            return [(t, None) for t in toks]
+        lps = self._lp_bcast_buf[:B].tolist()
+        return [
+            (t, lps[i] if sample_reqs[i].sampling_params.logprobs else None)
+            for i, t in enumerate(toks)
+        ]
 
     # -- weight refit (runtime weight update, reference §3.5) ---------------------------------
 

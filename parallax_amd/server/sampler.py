@@ -26,13 +26,15 @@ class Sampler:
         """logits: [B, vocab] fp32, row i belongs to reqs[i]. Returns token ids."""
         return [t for t, _ in self.sample_with_logprobs(logits, reqs)]
 
-    def sample_with_logprobs(
-        self, logits: torch.Tensor, reqs: List[InitialRequest]
-    ) -> List[tuple]:
-        """[(token_id, logprob-or-None)] — the logprob is log_softmax of the
-        post-penalty, pre-temperature logits at the sampled token, computed
-        only for requests with sampling_params.logprobs (reference wire fields
-        token_prob / return_probs)."""
+    def sample_device(
+        self,
+        logits: torch.Tensor,
+        reqs: List[InitialRequest],
+        want_logprobs: Optional[bool] = None,
+    ):
+        """Device-resident sampling: returns (tokens int64 [B] on device,
+        logprobs fp32 [B] on device or None). No host synchronization — the
+        PP token broadcast sends these tensors directly over RCCL."""
         B = logits.shape[0]
         assert B == len(reqs)
         sp = [r.sampling_params for r in reqs]
@@ -58,13 +60,29 @@ class Sampler:
             [s.min_p for s in sp],
             generator=self.generator,
         )
-        tok_list = tokens.tolist()
-        if not any(s.logprobs for s in sp):
-            return [(t, None) for t in tok_list]
+        if want_logprobs is None:
+            want_logprobs = any(s.logprobs for s in sp)
+        if not want_logprobs:
+            return tokens, None
         lp = torch.log_softmax(logits.float(), dim=-1).gather(
             1, tokens.view(-1, 1).to(logits.device)
-        ).squeeze(1).tolist()
+        ).squeeze(1)
+        return tokens, lp
+
+    def sample_with_logprobs(
+        self, logits: torch.Tensor, reqs: List[InitialRequest]
+    ) -> List[tuple]:
+        """[(token_id, logprob-or-None)] — the logprob is log_softmax of the
+        post-penalty, pre-temperature logits at the sampled token, computed
+        only for requests with sampling_params.logprobs (reference wire fields
+        token_prob / return_probs)."""
+        sp = [r.sampling_params for r in reqs]
+        tokens, lp = self.sample_device(logits, reqs)
+        tok_list = tokens.tolist()
+        if lp is None:
+            return [(t, None) for t in tok_list]
+        lp_list = lp.tolist()
         return [
-            (t, lp[i] if sp[i].logprobs else None)
+            (t, lp_list[i] if sp[i].logprobs else None)
             for i, t in enumerate(tok_list)
         ]

@@ -109,3 +109,70 @@ def test_pp_matches_single_process(tmp_path, world):
     mp.spawn(_pp_worker, args=(world, port, out_file), nprocs=world, join=True)
     got = torch.load(out_file)
     assert got == expected
+
+
+MANY_PROMPTS = [[5, 9, 13, 2, 7], [3, 3, 3, 99], [1] * 9, [7, 8], [41, 2, 2, 2, 6, 6]]
+
+
+def _pp4_counter_worker(rank, world, port, out_file):
+    os.environ.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+        RANK=str(rank), WORLD_SIZE=str(world),
+    )
+    import torch as _t
+
+    from parallax_amd.parallel.comm import init_distributed
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    comm = init_distributed(pp_size=world, tp_size=1, backend="gloo",
+                            device=_t.device("cpu"))
+    cfg = tiny_cfg()
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=_t.float32, micro_batches=4), comm=comm)
+    for name, t in full_state_dict(cfg).items():
+        eng.model.load_hf_weight(name, t)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6, ignore_eos=True)
+    if rank == 0:
+        for p in MANY_PROMPTS:
+            eng.submit(p, sp)
+    outs = {}
+    # fixed step count on every rank: all ranks run the same iterations
+    for _ in range(12):
+        for out in eng.step():
+            if out.token_id >= 0:
+                outs.setdefault(out.rid, []).append(out.token_id)
+    if rank == 0:
+        _t.save({"object_sync_count": eng.object_sync_count,
+                 "outs": list(outs.values())}, out_file)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_pp4_decode_path_is_object_free(tmp_path):
+    """steady-state PP decode must not pickle host objects —
+    the only payload broadcast is the one carrying the initial adds."""
+    from parallax_amd.parallel import comm as comm_mod
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    cfg = tiny_cfg()
+    comm_mod._CTX = None
+    ctx = comm_mod.CommContext(
+        world_size=1, rank=0, pp_size=1, tp_size=1, pp_rank=0, tp_rank=0,
+        device=torch.device("cpu"),
+    )
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=torch.float32), comm=ctx)
+    for name, t in full_state_dict(cfg).items():
+        eng.model.load_hf_weight(name, t)
+    sp = [SamplingParams(temperature=0.0, max_new_tokens=6, ignore_eos=True)] * len(MANY_PROMPTS)
+    expected = sorted(eng.generate(MANY_PROMPTS, sp).values())
+
+    out_file = str(tmp_path / "pp4.pt")
+    mp.spawn(_pp4_counter_worker, args=(4, 29640, out_file), nprocs=4, join=True)
+    got = torch.load(out_file)
+    assert got["object_sync_count"] == 1  # the initial adds payload, nothing else
+    assert sorted(got["outs"]) == expected
