@@ -147,8 +147,8 @@ MODELS = {
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=32)
-    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--steps", type=int, default=128)
+    ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="deepseek-r1-distill-llama-8b", choices=MODELS)
     ap.add_argument("--batch-per-gpu", type=int, default=512,
                     help="decode batch per GPU (global batch = N * this); "
@@ -245,10 +245,15 @@ def main():
             torch.cuda.synchronize()
         comm_world.barrier()
 
-    # prefill everything (untimed; ends when all requests are decoding)
+    # prefill everything (untimed; ends when all requests are decoding).
+    # Per-request TTFT = submit -> first sampled token (all requests are
+    # submitted at t0, so this is the saturated-arrival TTFT distribution).
     t_prefill0 = time.perf_counter()
+    first_token_s = {}
     while True:
-        engine.step()
+        for out in engine.step():
+            if out.rid not in first_token_s and out.token_id >= 0:
+                first_token_s[out.rid] = time.perf_counter() - t_prefill0
         running = engine.scheduler.running
         if running and all(r.prefill_done and r.num_output_tokens >= 1 for r in running.values()):
             break
@@ -256,6 +261,8 @@ def main():
             raise RuntimeError("all requests finished during prefill phase?")
     sync()
     prefill_s = time.perf_counter() - t_prefill0
+    ttfts = sorted(first_token_s.values())
+    p50_ttft_ms = round(ttfts[len(ttfts) // 2] * 1000, 1) if ttfts else None
 
     for _ in range(args.warmup):
         engine.step()
@@ -294,6 +301,7 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
+            "p50_ttft_ms": p50_ttft_ms,
             "config": {
                 "model": model_name,
                 "global_batch": global_batch,
@@ -304,6 +312,7 @@ def main():
                 "shared_prefix": args.shared_prefix,
                 "micro_batches": eargs.micro_batches,
                 "prefill_s": round(prefill_s, 3),
+                "p50_ttft_ms": p50_ttft_ms,
             },
         }
         print(json.dumps(result))
