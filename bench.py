@@ -159,6 +159,11 @@ def main():
     ap.add_argument("--micro-batches", type=int, default=0,
                     help="0 = one per pipeline stage")
     ap.add_argument("--kv-dtype", default="auto", choices=["auto", "fp8"])
+    ap.add_argument("--weight-dtype", default="auto",
+                    choices=["auto", "bf16", "fp8"],
+                    help="auto: fp8 W8A8 expert weights for the MoE configs "
+                         "whose BASELINE entry names fp8 (deepseek-v3, "
+                         "kimi-k2), bf16 otherwise")
     ap.add_argument("--parallelism", default="auto", choices=["auto", "dp", "pp"],
                     help="auto: DP replicas when the model fits one GPU "
                          "(288 GB HBM3E), PP layer split otherwise")
@@ -220,6 +225,11 @@ def main():
         kv_cache_dtype=args.kv_dtype,
         seed=0,
     )
+    wdtype = args.weight_dtype
+    if wdtype == "auto":
+        wdtype = "fp8" if args.model in ("deepseek-v3", "kimi-k2") and use_gpu else "bf16"
+    if wdtype == "fp8":
+        eargs.moe_weight_dtype = "fp8"
     engine = Engine(cfg, eargs, comm=comm, random_weights=True)
 
     # synthetic prompts, unique tokens so nothing prefix-shares (per-replica
@@ -309,6 +319,7 @@ def main():
                 "parallelism": f"{mode if world > 1 else 'pp'}{world}"
                 if world > 1 else "pp1",
                 "kv_dtype": args.kv_dtype,
+                "weight_dtype": wdtype,
                 "shared_prefix": args.shared_prefix,
                 "micro_batches": eargs.micro_batches,
                 "prefill_s": round(prefill_s, 3),

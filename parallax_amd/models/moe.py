@@ -103,12 +103,53 @@ class FusedMoE(nn.Module):
         self.num_local_experts = E // self.ep_size
         self.expert_offset = self.ep_rank * self.num_local_experts
         self.router = MoERouter(cfg)
+        self.fp8 = False
         self.w_gate_up = nn.Parameter(
             torch.empty(self.num_local_experts, 2 * inter, H), requires_grad=False
         )
         self.w_down = nn.Parameter(
             torch.empty(self.num_local_experts, H, inter), requires_grad=False
         )
+
+    # -- fp8 (W8A8) path ---------------------------------------------------------
+
+    def quantize_fp8(self) -> None:
+        """Convert expert weights to fp8-E4M3 with per-output-channel dequant
+        scales — the BASELINE DeepSeek-V3 "fp8 MFMA" configuration. The GPU
+        path then runs the fp8 grouped-GEMM kernels (ops.fused_moe_forward_fp8)
+        with per-token activation quantization; the CPU fallback dequantizes.
+        Quantization is chunked over experts to bound the fp32 temporary."""
+        if self.fp8:
+            return
+        E = self.num_local_experts
+        gu, dn = self.w_gate_up.data, self.w_down.data
+        q_gu = torch.empty(gu.shape, dtype=torch.float8_e4m3fn, device=gu.device)
+        s_gu = torch.empty(gu.shape[:2], dtype=torch.float32, device=gu.device)
+        q_dn = torch.empty(dn.shape, dtype=torch.float8_e4m3fn, device=dn.device)
+        s_dn = torch.empty(dn.shape[:2], dtype=torch.float32, device=dn.device)
+        step = max(1, min(8, E))
+        for e0 in range(0, E, step):
+            sl = slice(e0, min(e0 + step, E))
+            q_gu[sl], s_gu[sl] = ops.quantize_fp8_weight(gu[sl])
+            q_dn[sl], s_dn[sl] = ops.quantize_fp8_weight(dn[sl])
+        del self._parameters["w_gate_up"], self._parameters["w_down"]
+        self.register_buffer("w_gate_up_fp8", q_gu)
+        self.register_buffer("w_gu_scale", s_gu)
+        self.register_buffer("w_down_fp8", q_dn)
+        self.register_buffer("w_down_scale", s_dn)
+        self.fp8 = True
+
+    def _expert_gu(self, e: int, dtype: torch.dtype) -> torch.Tensor:
+        if self.fp8:
+            return (self.w_gate_up_fp8[e].float()
+                    * self.w_gu_scale[e].unsqueeze(-1)).to(dtype)
+        return self.w_gate_up[e]
+
+    def _expert_down(self, e: int, dtype: torch.dtype) -> torch.Tensor:
+        if self.fp8:
+            return (self.w_down_fp8[e].float()
+                    * self.w_down_scale[e].unsqueeze(-1)).to(dtype)
+        return self.w_down[e]
 
     # -- EP-aware weight loading (used by every family's loader) ----------------
 
@@ -153,10 +194,17 @@ class FusedMoE(nn.Module):
             )
             topk_w = topk_w * valid
         if x.is_cuda:
-            out = ops.fused_moe_forward(
-                x, self.w_gate_up, self.w_down, topk_ids, topk_w,
-                limit=self.act_limit,
-            )
+            if self.fp8:
+                out = ops.fused_moe_forward_fp8(
+                    x, self.w_gate_up_fp8, self.w_gu_scale,
+                    self.w_down_fp8, self.w_down_scale, topk_ids, topk_w,
+                    limit=self.act_limit,
+                )
+            else:
+                out = ops.fused_moe_forward(
+                    x, self.w_gate_up, self.w_down, topk_ids, topk_w,
+                    limit=self.act_limit,
+                )
             if self.ep_size > 1:
                 out = self._comm.tp_all_reduce(out)
             return out.to(x.dtype)
@@ -173,7 +221,7 @@ class FusedMoE(nn.Module):
             sel = (flat_ids == e).nonzero(as_tuple=True)[0]
             toks = token_idx[sel]
             xe = x[toks]                              # [n_e, H]
-            h = F.linear(xe, self.w_gate_up[e])       # [n_e, 2I]
+            h = F.linear(xe, self._expert_gu(e, xe.dtype))  # [n_e, 2I]
             if self.act_limit > 0:
                 I = self.intermediate_size
                 gate = h[:, :I].clamp(max=self.act_limit)
@@ -181,7 +229,7 @@ class FusedMoE(nn.Module):
                 h = (up + 1.0) * (gate * torch.sigmoid(gate * 1.702))
             else:
                 h = ops.silu_and_mul(h)
-            ye = F.linear(h, self.w_down[e]).float()  # [n_e, H]
+            ye = F.linear(h, self._expert_down(e, h.dtype)).float()  # [n_e, H]
             out.index_add_(0, toks, ye * flat_w[sel].unsqueeze(-1))
         if self.ep_size > 1:
             out = self._comm.tp_all_reduce(out)

@@ -445,6 +445,47 @@ def fused_moe_forward(
     return out
 
 
+def quantize_fp8_weight(w: torch.Tensor):
+    """Per-output-channel fp8-E4M3 quantization of a weight tensor whose rows
+    are output channels along dim -2 ([..., N, K] -> fp8 [..., N, K] +
+    fp32 dequant scale [..., N]). OCP e4m3fn — the gfx950-native encoding."""
+    amax = w.float().abs().amax(dim=-1).clamp_min(1e-8)
+    scale = amax / 448.0
+    q = (w.float() / scale.unsqueeze(-1)).clamp(-448.0, 448.0)
+    return q.to(torch.float8_e4m3fn), scale.contiguous()
+
+
+def fused_moe_forward_fp8(
+    x: torch.Tensor,             # [T, H] bf16
+    w_gate_up: torch.Tensor,     # [E, 2I, H] fp8 e4m3
+    w_gu_scale: torch.Tensor,    # [E, 2I] fp32
+    w_down: torch.Tensor,        # [E, H, I] fp8 e4m3
+    w_down_scale: torch.Tensor,  # [E, H] fp32
+    topk_ids: torch.Tensor,      # [T, k] long
+    topk_weights: torch.Tensor,  # [T, k] float
+    activation: str = "silu",
+    limit: float = 0.0,
+) -> torch.Tensor:
+    """W8A8 grouped-GEMM MoE on the fp8 MFMA path (activations quantized
+    per-token on device). Returns fp32 [T, H]."""
+    assert x.is_cuda
+    ext = _require_ext("moe_forward_fp8")
+    T, H = x.shape
+    E = w_gate_up.shape[0]
+    k = topk_ids.shape[1]
+    flat = topk_ids.reshape(-1)
+    sorted_ids, perm = torch.sort(flat)
+    seg = torch.searchsorted(
+        sorted_ids, torch.arange(E + 1, device=x.device, dtype=sorted_ids.dtype)
+    ).to(torch.int32)
+    route_w = topk_weights.reshape(-1).float()[perm].contiguous()
+    out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+    ext.moe_forward_fp8(out, x.contiguous(), w_gate_up, w_gu_scale, w_down,
+                        w_down_scale, perm.contiguous(), seg.contiguous(),
+                        route_w, k, activation == "gelu", limit)
+    return out
+
+
 # -- sampling (torch ops; GPU path uses torch's ROCm kernels — not a hot spot
 #    relative to the model forward, custom kernel is a later optimization) -------
 

@@ -57,6 +57,16 @@ void launch_moe_gate_up(void*, const void*, const void*, const int*,
 void launch_moe_down(void*, const void*, const void*, const float*, const int*,
                      const int*, const int64_t*, const int*, int, int, int,
                      int, int, hipStream_t);
+void launch_quantize_fp8_rows(void*, float*, const void*, int, int,
+                              hipStream_t);
+void launch_moe_gate_up_fp8(void*, const void*, const float*, const void*,
+                            const float*, const int*, const int*,
+                            const int64_t*, const int*, int, int, int, int,
+                            int, bool, float, hipStream_t);
+void launch_moe_down_fp8(void*, const void*, const float*, const void*,
+                         const float*, const float*, const int*, const int*,
+                         const int64_t*, const int*, int, int, int, int, int,
+                         hipStream_t);
 void launch_skinny_gemm(void*, const void*, const void*, const void*, int,
                         int, int, int64_t, int64_t, hipStream_t, bool*);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
@@ -415,6 +425,64 @@ void moe_forward(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
                   stream);
 }
 
+void moe_forward_fp8(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
+                     torch::Tensor w_gu_scale, torch::Tensor w_down,
+                     torch::Tensor w_down_scale, torch::Tensor perm,
+                     torch::Tensor seg_offsets, torch::Tensor route_w,
+                     int64_t topk, bool gelu, double limit) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_BF16(x);
+  TORCH_CHECK(w_gu.scalar_type() == at::kFloat8_e4m3fn && w_gu.is_contiguous());
+  TORCH_CHECK(w_down.scalar_type() == at::kFloat8_e4m3fn &&
+              w_down.is_contiguous());
+  TORCH_CHECK(w_gu_scale.scalar_type() == at::kFloat &&
+              w_down_scale.scalar_type() == at::kFloat);
+  TORCH_CHECK(out.scalar_type() == at::kFloat && out.is_contiguous());
+  TORCH_CHECK(perm.scalar_type() == at::kLong);
+  TORCH_CHECK(seg_offsets.scalar_type() == at::kInt);
+  TORCH_CHECK(route_w.scalar_type() == at::kFloat);
+  const int T = x.size(0);
+  const int E = w_gu.size(0);
+  const int I = w_gu.size(1) / 2;
+  const int H = w_gu.size(2);
+  const int n_assign = perm.size(0);
+  TORCH_CHECK(H % 32 == 0 && I % 64 == 0,
+              "MoE dims must be multiples of 32/64 (H=", H, " I=", I, ")");
+  const int max_tiles = (n_assign + 15) / 16 + E;
+  auto iopts = x.options().dtype(at::kInt);
+  auto fopts = x.options().dtype(at::kFloat);
+  auto bopts = x.options().dtype(at::kByte);
+  torch::Tensor tile_expert = torch::empty({max_tiles}, iopts);
+  torch::Tensor tile_row0 = torch::empty({max_tiles}, iopts);
+  torch::Tensor x_fp8 = torch::empty({T, H}, bopts);
+  torch::Tensor x_scale = torch::empty({T}, fopts);
+  torch::Tensor h_buf =
+      torch::empty({n_assign, I}, x.options().dtype(at::kBFloat16));
+  torch::Tensor h_fp8 = torch::empty({n_assign, I}, bopts);
+  torch::Tensor h_scale = torch::empty({std::max(n_assign, 1)}, fopts);
+  auto stream = cur_stream();
+  launch_quantize_fp8_rows(x_fp8.data_ptr(), x_scale.data_ptr<float>(),
+                           x.data_ptr(), T, H, stream);
+  launch_build_moe_tiles(tile_expert.data_ptr<int>(), tile_row0.data_ptr<int>(),
+                         seg_offsets.data_ptr<int>(), E, max_tiles, stream);
+  launch_moe_gate_up_fp8(
+      h_buf.data_ptr(), x_fp8.data_ptr(), x_scale.data_ptr<float>(),
+      w_gu.data_ptr(), w_gu_scale.data_ptr<float>(),
+      tile_expert.data_ptr<int>(), tile_row0.data_ptr<int>(),
+      perm.data_ptr<int64_t>(), seg_offsets.data_ptr<int>(), E, (int)topk, H,
+      I, max_tiles, gelu, (float)limit, stream);
+  if (n_assign > 0)
+    launch_quantize_fp8_rows(h_fp8.data_ptr(), h_scale.data_ptr<float>(),
+                             h_buf.data_ptr(), n_assign, I, stream);
+  launch_moe_down_fp8(
+      out.data_ptr(), h_fp8.data_ptr(), h_scale.data_ptr<float>(),
+      w_down.data_ptr(), w_down_scale.data_ptr<float>(),
+      route_w.data_ptr<float>(), tile_expert.data_ptr<int>(),
+      tile_row0.data_ptr<int>(), perm.data_ptr<int64_t>(),
+      seg_offsets.data_ptr<int>(), E, (int)topk, H, I, max_tiles, stream);
+}
+
 bool skinny_gemm(torch::Tensor c, torch::Tensor x, torch::Tensor w,
                  torch::Tensor bias) {
   CHECK_GPU(x);
@@ -470,6 +538,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("msa_paged_attention_decode", &msa_paged_attention_decode);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("moe_forward", &moe_forward);
+  m.def("moe_forward_fp8", &moe_forward_fp8);
   m.def("mla_paged_attention_decode", &mla_paged_attention_decode);
   m.def("prefill_attention", &prefill_attention);
   m.def("rmsnorm", &rmsnorm);

@@ -75,6 +75,7 @@ class EngineArgs:
     max_model_len: int = 8192           # context ceiling (sizes graph buffers)
     enable_graphs: bool = True          # hipGraph-captured decode forward
     kv_cache_dtype: str = "auto"        # "auto" (= engine dtype) | "fp8"(e4m3)
+    moe_weight_dtype: str = "auto"      # "auto" (= engine dtype) | "fp8"(W8A8)
 
 
 def partition_layers(num_layers: int, pp_size: int, pp_rank: int) -> Tuple[int, int]:
@@ -121,6 +122,15 @@ class Engine:
         # rope cache stays fp32
         self.model.rope_cache = self.model.rope_cache.float()
         self.model.eval()
+        if args.moe_weight_dtype == "fp8":
+            from ..models.moe import FusedMoE
+
+            n_q = 0
+            for m in self.model.modules():
+                if isinstance(m, FusedMoE):
+                    m.quantize_fp8()
+                    n_q += 1
+            logger.info("fp8 MoE: quantized %d expert blocks (W8A8)", n_q)
 
         # hybrid stacks: paged KV only for the full-attention layers
         self.local_layer_types = [cfg.layer_type(g) for g in range(start, end)]

@@ -181,3 +181,32 @@ def test_glm4_moe_chunked_prefill_parity():
         [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)],
     )
     assert list(out.values())[0] == ref
+
+
+def test_fused_moe_fp8_cpu_fallback():
+    """quantize_fp8 keeps the CPU expert-loop output close to the bf16 module
+    (per-channel W8 quantization error only)."""
+    import torch
+
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.models.moe import FusedMoE
+
+    cfg = ModelConfig(
+        architecture="Qwen3MoeForCausalLM", model_type="qwen3_moe",
+        vocab_size=64, hidden_size=64, num_layers=1, num_heads=2,
+        num_kv_heads=2, head_dim=32, intermediate_size=128,
+        moe_intermediate_size=64, num_experts=8, num_experts_per_tok=2,
+        max_position_embeddings=128, eos_token_ids=[],
+    )
+    torch.manual_seed(0)
+    moe = FusedMoE(cfg)
+    moe.router.weight.data.normal_(0, 0.2)
+    moe.w_gate_up.data.normal_(0, 0.05)
+    moe.w_down.data.normal_(0, 0.05)
+    x = torch.randn(12, cfg.hidden_size) * 0.5
+    ref = moe(x)
+    moe.quantize_fp8()
+    assert moe.fp8 and not hasattr(moe, "w_gate_up")
+    got = moe(x)
+    rel = (got - ref).norm() / ref.norm().clamp_min(1e-6)
+    assert rel < 0.08, f"fp8 CPU fallback rel error {rel:.3f}"
