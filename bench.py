@@ -171,6 +171,9 @@ def main():
                     help="tokens of prompt shared across all requests "
                          "(exercises block-radix prefix reuse; BASELINE "
                          "Kimi-K2 config)")
+    ap.add_argument("--layers", type=int, default=0,
+                    help="override num_hidden_layers (reduced-layer single-GPU "
+                         "runs of the PP=8 configs; headline runs use 0 = full)")
     ap.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
     args = ap.parse_args()
 
@@ -184,6 +187,9 @@ def main():
 
     cfg_fn, model_name = MODELS[args.model]
     cfg = ModelConfig.from_hf_config(cfg_fn())
+    if args.layers:
+        cfg.num_layers = args.layers
+        model_name += f"-{args.layers}L"
 
     # parallelism: the scheduler serves replicas (DP) whenever a whole model
     # fits one 288 GB GPU — that is the reference's multi-pipeline deployment
