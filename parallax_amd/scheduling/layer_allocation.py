@@ -244,11 +244,21 @@ class GreedyLayerAllocator:
 
 
 class DynamicProgrammingLayerAllocator:
-    """Chooses the number of pipelines k maximizing
-    Z(k) = k^alpha / (T_comp + (stages/k) * RTT), alpha = 2 (reference :758-965:
-    throughput grows with replicas, per-request latency grows with stage count),
-    using a DP over capacity-sorted nodes to find the minimum-stage partition
-    into k complete pipelines."""
+    """Memoized state-space DP over capacity-sorted nodes (behavior parity
+    with the reference's dp(i, open_residuals, finished) search,
+    layer_allocation.py:758-965): for each target pipeline count k it finds
+    the MINIMUM-STAGE partition of the pool into k complete pipelines —
+    interleaving pipeline construction, so heterogeneous pools like
+    capacities (40,30,20 | 35,30,25) over 90 layers partition into two
+    3-stage pipelines where a round-robin grouping fails — then scores each
+    feasible k with Z(k) = k^alpha / (T_comp + (stages/k) * RTT) and
+    backtracks the winner's assignment.
+
+    State: (i = next node index, open_residuals = sorted tuple of layers
+    still needed by open pipelines, finished). Transitions: skip node /
+    assign node to an open pipeline (closing it when the node can also
+    afford the lm_head) / start a new pipeline (node hosts the embedding).
+    Suffix-capacity pruning bounds the search."""
 
     alpha = 2.0
     default_rtt_ms = 5.0
@@ -257,38 +267,60 @@ class DynamicProgrammingLayerAllocator:
         self.model = model
 
     def allocate_from_standby(self, standby: List[Node]) -> List[Pipeline]:
-        best: Tuple[float, List[Pipeline]] = (-1.0, [])
+        L = self.model.num_layers
+        nodes = sorted(
+            standby, key=lambda n: n.decoder_layer_capacity(self.model),
+            reverse=True,
+        )
+        N = len(nodes)
+        cap_mid = [n.decoder_layer_capacity(self.model) for n in nodes]
+        cap_first = [
+            n.decoder_layer_capacity(self.model, is_first=True) for n in nodes
+        ]
+        cap_last = [
+            n.decoder_layer_capacity(self.model, is_last=True) for n in nodes
+        ]
+        cap_solo = [
+            n.decoder_layer_capacity(self.model, is_first=True, is_last=True)
+            for n in nodes
+        ]
+        total_cap = sum(cap_mid)
         greedy = GreedyLayerAllocator(self.model)
-        max_k = max(1, len(standby))
+        if N == 0 or L <= 0 or total_cap < L:
+            return greedy.allocate_from_standby(standby)
+        suffix = [0] * (N + 1)
+        for i in range(N - 1, -1, -1):
+            suffix[i] = suffix[i + 1] + cap_mid[i]
+        max_k = min(N, total_cap // L)
+
+        best_score, best_groups = -1.0, None
         for k in range(1, max_k + 1):
-            # try to build exactly k pipelines from a fresh copy of assignments
-            for n in standby:
-                n.clear_assignment()
-            pipes = self._build_k(list(standby), k)
+            groups = self._solve_k(
+                nodes, cap_mid, cap_first, cap_last, cap_solo, suffix, L, k
+            )
+            if groups is None:
+                continue
+            pipes = self._apply_groups(groups, L)
             if pipes is None:
                 continue
             score = self._score(pipes, k)
-            if score > best[0]:
-                best = (score, pipes)
-        if best[1]:
-            # re-apply the winning assignment (nodes were mutated per k-trial)
-            for n in standby:
-                n.clear_assignment()
-            winning = self._build_k(list(standby), len(best[1]))
-            return winning or []
-        for n in standby:
+            if score > best_score:
+                best_score, best_groups = score, groups
+        for n in nodes:
             n.clear_assignment()
-        return greedy.allocate_from_standby(standby)
+        if best_groups is None:
+            return greedy.allocate_from_standby(standby)
+        pipes = self._apply_groups(best_groups, L)
+        return pipes if pipes is not None else greedy.allocate_from_standby(standby)
 
-    def _build_k(self, pool: List[Node], k: int) -> Optional[List[Pipeline]]:
-        """Split the capacity-sorted pool round-robin into k groups, then
-        water-fill each group; DP-style fallback shrinks groups that fail."""
-        L = self.model.num_layers
-        pool = sorted(pool, key=lambda n: n.decoder_layer_capacity(self.model), reverse=True)
-        groups: List[List[Node]] = [[] for _ in range(k)]
-        for i, n in enumerate(pool):
-            groups[i % k].append(n)
-        pipes = []
+    def _apply_groups(
+        self, groups: List[List[Node]], L: int
+    ) -> Optional[List[Pipeline]]:
+        """Water-fill spans within each DP-chosen node group."""
+        for g in groups:
+            for n in g:
+                n.clear_assignment()
+        pipes: List[Pipeline] = []
         for g in groups:
             spans = water_fill_layers(g, self.model, L)
             if spans is None:
@@ -298,6 +330,114 @@ class DynamicProgrammingLayerAllocator:
                 n.model = self.model
             pipes.append(Pipeline([n for n in g if n.num_layers_hosted > 0]))
         return pipes
+
+    def _solve_k(
+        self,
+        nodes: List[Node],
+        cap_mid: List[int],
+        cap_first: List[int],
+        cap_last: List[int],
+        cap_solo: List[int],
+        suffix: List[int],
+        L: int,
+        k: int,
+    ) -> Optional[List[List[Node]]]:
+        """Min-stage partition into exactly k pipelines, or None."""
+        N = len(nodes)
+        INF = float("inf")
+        memo: Dict[Tuple[int, Tuple[int, ...], int], float] = {}
+        action: Dict[Tuple[int, Tuple[int, ...], int], Tuple] = {}
+
+        def dp(i: int, open_res: Tuple[int, ...], finished: int) -> float:
+            if finished == k and not open_res:
+                return 0.0
+            if i == N:
+                return INF
+            key = (i, open_res, finished)
+            if key in memo:
+                return memo[key]
+            new_needed = k - finished - len(open_res)
+            # pruning: overshot target / not enough capacity or nodes left
+            if (
+                new_needed < 0
+                or suffix[i] < sum(open_res) + max(0, new_needed) * L
+                or finished + len(open_res) + (N - i) < k
+            ):
+                memo[key] = INF
+                return INF
+
+            best_cost = dp(i + 1, open_res, finished)
+            best_act: Tuple = ("skip",)
+
+            # assign node i to open pipeline j
+            for j, rj in enumerate(open_res):
+                if rj <= cap_last[i]:
+                    # node covers the remaining layers AND the lm_head: close
+                    lst = list(open_res)
+                    lst.pop(j)
+                    c = 1 + dp(i + 1, tuple(lst), finished + 1)
+                    if c < best_cost:
+                        best_cost, best_act = c, ("close", j)
+                else:
+                    # keep open; if the node could cover the layers but not
+                    # the lm_head, it must leave >= 1 layer for a closer
+                    r_after = max(1, rj - cap_mid[i])
+                    lst = list(open_res)
+                    lst[j] = r_after
+                    lst.sort()
+                    c = 1 + dp(i + 1, tuple(lst), finished)
+                    if c < best_cost:
+                        best_cost, best_act = c, ("assign", j)
+
+            # start a new pipeline with node i as the embedding host
+            if new_needed > 0:
+                if L <= cap_solo[i]:
+                    c = 1 + dp(i + 1, open_res, finished + 1)
+                    if c < best_cost:
+                        best_cost, best_act = c, ("solo",)
+                else:
+                    r_new = max(1, L - cap_first[i])
+                    lst = sorted(list(open_res) + [r_new])
+                    c = 1 + dp(i + 1, tuple(lst), finished)
+                    if c < best_cost:
+                        best_cost, best_act = c, ("start", r_new)
+
+            memo[key] = best_cost
+            action[key] = best_act
+            return best_cost
+
+        if dp(0, (), 0) == INF:
+            return None
+
+        # backtrack: replay the recorded decisions
+        groups: List[List[Node]] = []
+        open_list: List[Tuple[int, List[Node]]] = []  # (residual, group) sorted
+        i, finished = 0, 0
+        while not (finished == k and not open_list):
+            if i >= N:  # inconsistent path (should not happen)
+                return None
+            key = (i, tuple(r for r, _ in open_list), finished)
+            act = action.get(key, ("skip",))
+            if act[0] == "skip":
+                pass
+            elif act[0] == "close":
+                _, g = open_list.pop(act[1])
+                g.append(nodes[i])
+                groups.append(g)
+                finished += 1
+            elif act[0] == "assign":
+                rj, g = open_list.pop(act[1])
+                g.append(nodes[i])
+                open_list.append((max(1, rj - cap_mid[i]), g))
+                open_list.sort(key=lambda t: t[0])
+            elif act[0] == "solo":
+                groups.append([nodes[i]])
+                finished += 1
+            elif act[0] == "start":
+                open_list.append((act[1], [nodes[i]]))
+                open_list.sort(key=lambda t: t[0])
+            i += 1
+        return groups
 
     def _score(self, pipes: List[Pipeline], k: int) -> float:
         stages = sum(len(p.nodes) for p in pipes)

@@ -95,6 +95,63 @@ def test_dp_allocator_covers():
     assert pipes and all(p.covers(80) for p in pipes)
 
 
+class FixedCapNode(Node):
+    """Fake node with a pinned layer capacity (capacity formula bypassed so
+    partition tests can state exact heterogeneous pools)."""
+
+    def __init__(self, nid, cap, hw=None):
+        super().__init__(node_id=nid, hardware=hw or MI355X)
+        self._cap = cap
+
+    def decoder_layer_capacity(self, model=None, is_first=False, is_last=False):
+        return self._cap
+
+
+def test_dp_allocator_finds_interleaved_partition():
+    """a heterogeneous pool where greedy/round-robin grouping
+    cannot form two pipelines but the memoized dp(i, open_residuals, finished)
+    search can. Capacities (40,35,30,30,25,20) over 90 layers: the only
+    2-pipeline partition is {40,30,20} + {35,30,25}; round-robin over the
+    capacity-sorted pool yields {40,30,25}=95 / {35,30,20}=85 < 90 and fails."""
+    model = llama70b_info()
+    model.cfg.num_layers = 90
+    caps = {"a40": 40, "b35": 35, "c30": 30, "d30": 30, "e25": 25, "f20": 20}
+    nodes = [FixedCapNode(nid, c) for nid, c in caps.items()]
+    set_full_rtt(nodes)
+
+    # the round-robin grouping (the old heuristic) cannot build k=2
+    ordered = sorted(nodes, key=lambda n: n.decoder_layer_capacity(model),
+                     reverse=True)
+    rr = [[], []]
+    for i, n in enumerate(ordered):
+        rr[i % 2].append(n)
+    assert min(sum(n.decoder_layer_capacity(model) for n in g) for g in rr) < 90
+
+    pipes = DynamicProgrammingLayerAllocator(model).allocate_from_standby(nodes)
+    assert len(pipes) == 2, f"DP should find 2 pipelines, got {len(pipes)}"
+    for p in pipes:
+        assert p.covers(90)
+    groups = [sorted(caps[nid] for nid in p.node_ids) for p in pipes]
+    assert sorted(groups) == [[20, 30, 40], [25, 30, 35]]
+
+
+def test_dp_allocator_min_stages_prefers_fewer_nodes():
+    """With one huge node and several small ones, k=1 via the huge node alone
+    is a 1-stage pipeline; the DP must not smear layers across extra stages
+    when a node pool supports more replicas."""
+    model = llama70b_info()
+    big = FixedCapNode("big", 100)
+    smalls = [FixedCapNode(f"s{i}", 45) for i in range(2)]
+    nodes = [big] + smalls
+    set_full_rtt(nodes)
+    pipes = DynamicProgrammingLayerAllocator(model).allocate_from_standby(nodes)
+    # 80 layers: {big} alone and {s0,s1} together -> k=2 beats k=1
+    assert len(pipes) == 2
+    sizes = sorted(len(p.nodes) for p in pipes)
+    assert sizes == [1, 2]
+    assert all(p.covers(80) for p in pipes)
+
+
 def make_scheduler(n_nodes=2, **kw):
     model = llama70b_info()
     sched = ClusterScheduler(model, min_nodes_bootstrapping=n_nodes, **kw)
