@@ -1,0 +1,272 @@
+"""json_schema constrained decoding: a character-level pushdown automaton
+compiled from a JSON-schema subset, driving token-level logit masks in the
+sampler.
+
+Reference parity: the reference carries `json_schema` on SamplingParams
+(src/parallax/server/sampling/sampling_params.py:25) and delegates enforcement
+to its backends; here enforcement is native. Schema subset: object (properties
+generated in schema order, required == all listed), string, integer, number,
+boolean, null, enum (distinct literals), array (items + minItems/maxItems),
+and arbitrary nesting of those.
+
+Design: states are immutable tuples (a stack of frames, top last), so the
+per-state token mask is cacheable. `advance` consumes one character;
+delimiters that terminate numbers are re-processed against the remaining
+stack, which is what makes `{"a":12}`-style adjacency work. Token masking
+tests each vocab string against the automaton; masks are cached per
+(schema, state) so steady-state decoding costs one dict lookup.
+"""
+
+from __future__ import annotations
+
+import json
+from typing import Dict, List, Optional, Tuple
+
+State = Tuple  # tuple of frames; () == complete
+
+_DIGITS = set("0123456789")
+
+
+class JsonSchemaFSM:
+    def __init__(self, schema: dict):
+        # flatten schema nodes into a list so frames can reference them by id
+        self.nodes: List[dict] = []
+        self.root = self._compile(schema)
+
+    def _compile(self, schema: dict) -> int:
+        nid = len(self.nodes)
+        self.nodes.append({})
+        node: Dict = {}
+        if "enum" in schema:
+            node = {"kind": "enum",
+                    "options": tuple(json.dumps(v, separators=(",", ":"))
+                                     for v in schema["enum"])}
+        else:
+            t = schema.get("type", "object")
+            if t == "object":
+                props = schema.get("properties", {})
+                order = [k for k in props]
+                node = {
+                    "kind": "object",
+                    "keys": tuple(order),
+                    "children": tuple(self._compile(props[k]) for k in order),
+                }
+            elif t == "array":
+                node = {
+                    "kind": "array",
+                    "item": self._compile(schema.get("items", {"type": "string"})),
+                    "min": int(schema.get("minItems", 0)),
+                    "max": int(schema.get("maxItems", 1 << 30)),
+                }
+            elif t in ("string", "integer", "number", "boolean", "null"):
+                node = {"kind": t}
+            else:
+                raise ValueError(f"unsupported schema type: {t!r}")
+        self.nodes[nid] = node
+        return nid
+
+    # -- state construction ------------------------------------------------------
+
+    def initial(self) -> State:
+        return (("val", self.root),)
+
+    def is_complete(self, state: State) -> bool:
+        return len(state) == 0
+
+    # -- the automaton ----------------------------------------------------------
+
+    def _expand(self, state: State) -> State:
+        """Expand a top-of-stack val frame into concrete frames (no input)."""
+        while state and state[-1][0] == "val":
+            nid = state[-1][1]
+            node = self.nodes[nid]
+            rest = state[:-1]
+            kind = node["kind"]
+            if kind == "enum":
+                state = rest + (("alt", node["options"], ""),)
+            elif kind == "string":
+                state = rest + (("str",), ("lit", '"', 0))
+            elif kind == "integer":
+                state = rest + (("int", "start"),)
+            elif kind == "number":
+                state = rest + (("num", "start"),)
+            elif kind == "boolean":
+                state = rest + (("alt", ("true", "false"), ""),)
+            elif kind == "null":
+                state = rest + (("lit", "null", 0),)
+            elif kind == "object":
+                frames: List[Tuple] = [("lit", "}", 0)]
+                for i in range(len(node["keys"]) - 1, -1, -1):
+                    frames.append(("val", node["children"][i]))
+                    sep = "{" if i == 0 else ","
+                    frames.append(("lit", f'{sep}"{node["keys"][i]}":', 0))
+                if not node["keys"]:
+                    frames = [("lit", "{}", 0)]
+                # stack top is state[-1]: frames were built closing-brace
+                # first, so appending in built order puts '{"key":' on top
+                state = rest + tuple(frames)
+            elif kind == "array":
+                state = rest + (("arr", nid, 0, "item_or_close"),
+                                ("lit", "[", 0))
+            else:  # pragma: no cover
+                raise AssertionError(kind)
+        return state
+
+    def advance(self, state: State, ch: str) -> Optional[State]:
+        state = self._expand(state)
+        if not state:
+            return None  # complete: no more characters accepted
+        top = state[-1]
+        rest = state[:-1]
+        tag = top[0]
+
+        if tag == "lit":
+            _, text, i = top
+            if ch != text[i]:
+                return None
+            if i + 1 == len(text):
+                return rest
+            return rest + (("lit", text, i + 1),)
+
+        if tag == "str":
+            if ch == '"':
+                return rest
+            if ch == "\\":
+                return rest + (top, ("esc",))
+            if ord(ch) < 0x20:
+                return None
+            return state
+
+        if tag == "esc":
+            if ch in '"\\/bfnrt':
+                return rest
+            return None
+
+        if tag == "alt":
+            _, options, prefix = top
+            p = prefix + ch
+            live = [o for o in options if o.startswith(p)]
+            if not live:
+                return None
+            if p in live and len(live) == 1:
+                return rest
+            return rest + (("alt", options, p),)
+
+        if tag in ("int", "num"):
+            _, phase = top
+            is_num = tag == "num"
+            if phase == "start":
+                if ch == "-":
+                    return rest + ((tag, "int0"),)
+                if ch == "0":
+                    return rest + ((tag, "z"),)
+                if ch in _DIGITS:
+                    return rest + ((tag, "int"),)
+                return None
+            if phase == "int0":
+                if ch == "0":
+                    return rest + ((tag, "z"),)
+                if ch in _DIGITS:
+                    return rest + ((tag, "int"),)
+                return None
+            if phase in ("z", "int", "frac", "exp"):
+                if ch in _DIGITS and phase != "z":
+                    return rest + ((tag, phase),)
+                if is_num and ch == "." and phase in ("z", "int"):
+                    return rest + ((tag, "frac0"),)
+                if is_num and ch in "eE" and phase in ("z", "int", "frac"):
+                    return rest + ((tag, "exp0"),)
+                if ch in _DIGITS and phase == "z":
+                    return None  # no leading zeros
+                # number complete: delimiter re-processed by the rest
+                return self.advance(rest, ch)
+            if phase == "frac0":
+                return rest + ((tag, "frac"),) if ch in _DIGITS else None
+            if phase == "exp0":
+                if ch in "+-":
+                    return rest + ((tag, "exp1"),)
+                return rest + ((tag, "exp"),) if ch in _DIGITS else None
+            if phase == "exp1":
+                return rest + ((tag, "exp"),) if ch in _DIGITS else None
+            return None
+
+        if tag == "arr":
+            _, nid, n, expect = top
+            node = self.nodes[nid]
+            if expect == "item_or_close":
+                if ch == "]" and n >= node["min"]:
+                    return rest
+                if n >= node["max"]:
+                    return None
+                nxt = rest + (("arr", nid, n + 1, "sep_or_close"),
+                              ("val", node["item"]))
+                return self.advance(nxt, ch)
+            if expect == "sep_or_close":
+                if ch == "]" and n >= node["min"]:
+                    return rest
+                if ch == "," and n < node["max"]:
+                    return rest + (("arr", nid, n + 1, "sep_or_close"),
+                                   ("val", node["item"]))
+                return None
+            return None
+
+        return None  # pragma: no cover
+
+    def advance_str(self, state: State, s: str) -> Optional[State]:
+        for ch in s:
+            state = self.advance(state, ch)
+            if state is None:
+                return None
+        return self._expand(state)
+
+
+class GrammarMatcher:
+    """Per-request masking state: tracks consumed output tokens and yields the
+    set of allowed next-token ids (EOS once the schema is satisfied). Masks
+    are cached per (schema, state) across all requests."""
+
+    _fsm_cache: Dict[str, JsonSchemaFSM] = {}
+    _mask_cache: Dict[Tuple[str, State], List[int]] = {}
+
+    def __init__(self, schema_json: str, vocab: List[str], eos_ids: List[int]):
+        self.schema_json = schema_json
+        fsm = self._fsm_cache.get(schema_json)
+        if fsm is None:
+            fsm = JsonSchemaFSM(json.loads(schema_json))
+            self._fsm_cache[schema_json] = fsm
+        self.fsm = fsm
+        self.vocab = vocab
+        self.eos_ids = [e for e in eos_ids if e is not None]
+        self.state: Optional[State] = fsm.initial()
+        self._consumed = 0
+
+    def catch_up(self, output_token_ids: List[int]) -> None:
+        for tid in output_token_ids[self._consumed:]:
+            if self.state is None:
+                break
+            if tid in self.eos_ids:
+                continue
+            s = self.vocab[tid] if 0 <= tid < len(self.vocab) else ""
+            nxt = self.fsm.advance_str(self.state, s)
+            if nxt is not None:
+                self.state = nxt
+            # else: token was sampled outside the grammar (mask raced an
+            # abort or vocab mismatch) — freeze rather than crash
+        self._consumed = len(output_token_ids)
+
+    def allowed_ids(self) -> List[int]:
+        if self.state is None:
+            return list(self.eos_ids)
+        if self.fsm.is_complete(self.state):
+            return list(self.eos_ids)
+        key = (self.schema_json, self.state)
+        ids = self._mask_cache.get(key)
+        if ids is None:
+            adv = self.fsm.advance_str
+            st = self.state
+            ids = [
+                i for i, s in enumerate(self.vocab)
+                if s and adv(st, s) is not None
+            ]
+            self._mask_cache[key] = ids
+        return ids

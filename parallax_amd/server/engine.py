@@ -269,6 +269,11 @@ class Engine:
     def abort(self, rid: str) -> None:
         self._pending_aborts.append(rid)
 
+    def set_grammar_vocab(self, vocab: List[str]) -> None:
+        """Enable json_schema constrained decoding: vocab[i] is token i's
+        text. Must be called on every rank that samples (in practice: all)."""
+        self.sampler.grammar_vocab = vocab
+
     @property
     def has_work(self) -> bool:
         return self.scheduler.has_work or bool(self._pending_adds)
@@ -377,7 +382,7 @@ class Engine:
             sp = r.sampling_params
             if not (sp.ignore_eos or not r.eos_token_ids):
                 return False
-            if sp.stop_token_ids or sp.logprobs:
+            if sp.stop_token_ids or sp.logprobs or sp.json_schema:
                 return False
             if (sp.repetition_penalty != 1.0 or sp.presence_penalty != 0.0
                     or sp.frequency_penalty != 0.0):

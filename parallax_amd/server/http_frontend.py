@@ -37,6 +37,9 @@ def create_app(
             # default stop tokens to the tokenizer's EOS (it may differ from the
             # model-config eos_token_ids the engine applies)
             sp.stop_token_ids = [tokenizer.eos_token_id]
+        if sp.json_schema and server.engine.sampler.grammar_vocab is None:
+            # one-time id->text table for constrained decoding
+            server.engine.set_grammar_vocab(tokenizer.vocab_strings())
         return sp
 
     def _find_stop(text: str, stops: List[str]):

@@ -94,6 +94,9 @@ class InitialRequest(Request):
     finish_time: Optional[float] = None
     # set when the client disconnected / abort was requested
     abort_requested: bool = False
+    # lazily-built constrained-decoding matcher (sampling_params.json_schema);
+    # lives on the rank that samples, never serialized
+    grammar: Optional[object] = field(default=None, repr=False, compare=False)
 
     @property
     def prompt_len(self) -> int:

@@ -17,10 +17,47 @@ class Sampler:
     def __init__(self, device: torch.device, seed: Optional[int] = None):
         self.device = device
         self.generator = None
+        # id -> token string table; enables json_schema constrained decoding
+        # (set via Engine.set_grammar_vocab)
+        self.grammar_vocab: Optional[List[str]] = None
         if seed is not None:
             self.generator = torch.Generator(
                 device=device if device.type == "cuda" else "cpu"
             ).manual_seed(seed)
+
+    def _apply_grammar_masks(
+        self, logits: torch.Tensor, reqs: List[InitialRequest]
+    ) -> torch.Tensor:
+        """Mask logits rows of requests with a json_schema to the token set
+        the schema FSM allows next (EOS once the schema is satisfied)."""
+        if self.grammar_vocab is None:
+            return logits
+        rows = [
+            i for i, r in enumerate(reqs) if r.sampling_params.json_schema
+        ]
+        if not rows:
+            return logits
+        from .constrained import GrammarMatcher
+
+        logits = logits.clone()
+        for i in rows:
+            r = reqs[i]
+            if r.grammar is None:
+                eos = list(r.eos_token_ids) + list(
+                    r.sampling_params.stop_token_ids
+                )
+                r.grammar = GrammarMatcher(
+                    r.sampling_params.json_schema, self.grammar_vocab, eos
+                )
+            r.grammar.catch_up(r.output_token_ids)
+            ids = r.grammar.allowed_ids()
+            if not ids:
+                continue  # no EOS registered: leave unconstrained
+            idx = torch.tensor(ids, dtype=torch.long, device=logits.device)
+            row = torch.full_like(logits[i], float("-inf"))
+            row[idx] = logits[i, idx]
+            logits[i] = row
+        return logits
 
     def sample(self, logits: torch.Tensor, reqs: List[InitialRequest]) -> List[int]:
         """logits: [B, vocab] fp32, row i belongs to reqs[i]. Returns token ids."""
@@ -38,6 +75,7 @@ class Sampler:
         B = logits.shape[0]
         assert B == len(reqs)
         sp = [r.sampling_params for r in reqs]
+        logits = self._apply_grammar_masks(logits, reqs)
         need_penalties = any(
             s.repetition_penalty != 1.0 or s.presence_penalty != 0.0
             or s.frequency_penalty != 0.0

@@ -95,4 +95,21 @@ class SamplingParams:
             ignore_eos=bool(body.get("ignore_eos", False)),
             seed=body.get("seed"),
             logprobs=bool(body.get("logprobs", False)),
+            json_schema=_extract_json_schema(body),
         )
+
+
+def _extract_json_schema(body: dict):
+    """OpenAI structured-output surfaces: `response_format={"type":
+    "json_schema", "json_schema": {"schema": {...}}}` or a raw `json_schema`
+    field (the reference's wire name). Returns the schema as a JSON string."""
+    import json as _json
+
+    raw = body.get("json_schema")
+    rf = body.get("response_format")
+    if raw is None and isinstance(rf, dict) and rf.get("type") == "json_schema":
+        inner = rf.get("json_schema") or {}
+        raw = inner.get("schema", inner)
+    if raw is None:
+        return None
+    return raw if isinstance(raw, str) else _json.dumps(raw)

@@ -64,6 +64,20 @@ class TokenizerWrapper:
             return self.hf.decode(ids, skip_special_tokens=True)
         return self.synthetic.decode(ids)
 
+    def vocab_strings(self) -> List[str]:
+        """id -> token text table for constrained decoding. HF: per-id decode
+        (one-time cost); synthetic: byte tokens map back to their character."""
+        if self.hf is not None:
+            size = len(self.hf)
+            return [self.hf.decode([i]) for i in range(size)]
+        v = self.synthetic
+        out = [""] * v.vocab_size
+        for b in range(256):
+            tid = 3 + (b % (v.vocab_size - 3))
+            if 0 <= tid < v.vocab_size and not out[tid]:
+                out[tid] = chr(b)
+        return out
+
     def chat_prompt_ids(self, messages: List[dict]) -> List[int]:
         if self.hf is not None and getattr(self.hf, "chat_template", None):
             return self.hf.apply_chat_template(

@@ -116,8 +116,8 @@ def test_moe_fp8_engine_decode():
         "n_routed_experts": 16, "num_experts_per_tok": 4, "n_shared_experts": 1,
         "n_group": 4, "topk_group": 2, "routed_scaling_factor": 2.5,
         "norm_topk_prob": True, "first_k_dense_replace": 1,
-        "q_lora_rank": 64, "kv_lora_rank": 64, "qk_nope_head_dim": 32,
-        "qk_rope_head_dim": 16, "v_head_dim": 32, "rms_norm_eps": 1e-6,
+        "q_lora_rank": 128, "kv_lora_rank": 512, "qk_nope_head_dim": 64,
+        "qk_rope_head_dim": 64, "v_head_dim": 64, "rms_norm_eps": 1e-6,
         "rope_theta": 10000.0, "max_position_embeddings": 2048,
     })
     eng = Engine(cfg, EngineArgs(num_kv_blocks=128, moe_weight_dtype="fp8"),
