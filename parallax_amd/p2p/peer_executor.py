@@ -141,6 +141,14 @@ class PeerExecutor:
         # non-head cache position bookkeeping: rid -> tokens cached so far
         self._peer_positions: Dict[str, int] = {}
         self.finished_outputs: List[PeerOutput] = []
+        # packets that raced a blocking refit fetch; drained before transport
+        self._deferred_packets: List[bytes] = []
+        self._pending_refit_manifest: Optional[tuple] = None
+
+    def _recv(self, timeout: float) -> Optional[bytes]:
+        if self._deferred_packets:
+            return self._deferred_packets.pop(0)
+        return self.transport.recv(timeout=timeout)
 
     # -- head API ------------------------------------------------------------------
 
@@ -192,11 +200,13 @@ class PeerExecutor:
     def _head_step(self, recv_timeout: float) -> None:
         # 1. drain token packets from the last stage
         while True:
-            data = self.transport.recv(timeout=recv_timeout)
+            data = self._recv(recv_timeout)
             if data is None:
                 break
             msg = codec.decode(data)
-            if msg["kind"] == "token":
+            if msg["kind"].startswith("refit_"):
+                self._handle_refit(msg)
+            elif msg["kind"] == "token":
                 for entry in msg["tokens"]:
                     rid, tok = entry[0], entry[1]
                     lp = entry[2] if len(entry) > 2 else None
@@ -296,19 +306,21 @@ class PeerExecutor:
 
     def _peer_step(self, recv_timeout: float) -> None:
         packets: List[IntermediateRequest] = []
-        data = self.transport.recv(timeout=recv_timeout)
+        data = self._recv(recv_timeout)
         if data is None:
             return
         while data is not None:
             msg = codec.decode(data)
             if msg["kind"] == "forward":
                 packets.extend(msg["reqs"])
+            elif msg["kind"].startswith("refit_"):
+                self._handle_refit(msg)
             elif msg["kind"] in ("release", "abort"):
                 for rid in msg["rids"]:
                     self.cache_manager.free_request(rid)
                     self._peer_positions.pop(rid, None)
                     self._sampling_ctx.pop(rid, None)
-            data = self.transport.recv(timeout=0.0)
+            data = self._recv(0.0)
         if not packets:
             return
         prefills = [p for p in packets if p.is_prefill]
@@ -437,6 +449,84 @@ class PeerExecutor:
             by_peer[nxt].append(p)
         for peer, group in by_peer.items():
             self.transport.send(peer, codec.encode_forward(group))
+
+    # -- weight refit over the transport (reference p2p/server.py:224-338) --------
+
+    def set_refit_publisher(self, publisher) -> None:
+        """Serve weight chunks to peers (the trainer/origin side)."""
+        self._refit_publisher = publisher
+
+    def _handle_refit(self, msg: dict) -> None:
+        from . import refit as refit_mod
+
+        if msg["kind"] == "refit_get":
+            pub = getattr(self, "_refit_publisher", None)
+            if pub is None:
+                logger.warning("refit_get but no publisher attached")
+                return
+            refit_mod.answer_refit_get(pub, self.transport, msg)
+        elif msg["kind"] == "refit_chunk":
+            logger.debug("stray refit_chunk (no fetch in progress)")
+        elif msg["kind"] == "refit_manifest":
+            # a pushed manifest: fetch our shard's files, then hot-reload
+            self._pending_refit_manifest = (
+                msg["manifest"], msg.get("publisher", "")
+            )
+
+    def refit_from_peer(
+        self, publisher_peer: str, manifest: dict, dest_root: str,
+        reload_weights: bool = True, timeout: float = 60.0,
+    ) -> str:
+        """Pull the manifest's chunks covering this peer's layer range over
+        the transport, verify CIDs, reassemble, GC old versions and hot-reload
+        the shard weights. Chunk replies are consumed by the step loop (or
+        drained here when no loop is running)."""
+        import msgpack
+
+        from . import refit as refit_mod
+
+        fetcher = refit_mod.RefitFetcher(dest_root)
+
+        def get(name: str, idx: int) -> bytes:
+            self.transport.send(
+                publisher_peer,
+                msgpack.packb(
+                    {"kind": "refit_get", "name": name, "idx": idx,
+                     "reply_to": self.peer_id},
+                    use_bin_type=True,
+                ),
+            )
+            deadline = time.monotonic() + timeout
+            while time.monotonic() < deadline:
+                data = self.transport.recv(timeout=0.05)
+                if data is None:
+                    continue
+                msg = codec.decode(data)
+                if msg["kind"] == "refit_chunk":
+                    if msg.get("name") == name and msg.get("idx") == idx:
+                        return msg["data"]
+                    continue  # stale reply from a retried request
+                if msg["kind"] == "refit_get":
+                    self._handle_refit(msg)
+                else:
+                    # pipeline packet raced the refit: step() drains these first
+                    self._deferred_packets.append(data)
+            raise refit_mod.RefitError(
+                f"timeout fetching {name}[{idx}] from {publisher_peer}"
+            )
+
+        vdir = fetcher.fetch(
+            manifest, get,
+            layer_range=(self.model.start_layer, self.model.end_layer),
+        )
+        if reload_weights:
+            from ..server.shard_loader import load_shard_weights
+
+            n = load_shard_weights(self.model, vdir)
+            if hasattr(self.model, "finalize_weights"):
+                self.model.finalize_weights()
+            logger.info("refit: reloaded %d tensors from %s", n, vdir)
+        return vdir
 
     def _broadcast_control(self, kind: str, rids: List[str], routing_table: List[str]) -> None:
         for peer in routing_table:

@@ -115,3 +115,128 @@ def test_http_update_weights_endpoint(tmp_path):
             assert r.json()["updated_tensors"] > 0
     finally:
         server.stop()
+
+
+# -- chunked refit distribution over the P2P transport (reference
+#    p2p/server.py:224-338 CID-checksummed block store + 3-version GC) -----------
+
+
+def _tiny_cfg():
+    return ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=128, hidden_size=32,
+        num_layers=2, num_heads=2, num_kv_heads=2, head_dim=16,
+        intermediate_size=64, max_position_embeddings=128, eos_token_ids=[],
+    )
+
+
+def _save_tiny_ckpt(tmpdir, seed):
+    """Synthetic sharded checkpoint: per-layer safetensors + index map."""
+    from safetensors.torch import save_file
+
+    os.makedirs(tmpdir, exist_ok=True)
+    g = torch.Generator().manual_seed(seed)
+    cfg = _tiny_cfg()
+    weight_map = {}
+    sd0, sd1, shared = {}, {}, {}
+    def rand(*s):
+        return torch.randn(*s, generator=g) * 0.05
+    for i, sd in enumerate((sd0, sd1)):
+        p = f"model.layers.{i}."
+        hd = cfg.num_heads * cfg.head_dim
+        kvd = cfg.num_kv_heads * cfg.head_dim
+        sd[p + "self_attn.q_proj.weight"] = rand(hd, cfg.hidden_size)
+        sd[p + "self_attn.k_proj.weight"] = rand(kvd, cfg.hidden_size)
+        sd[p + "self_attn.v_proj.weight"] = rand(kvd, cfg.hidden_size)
+        sd[p + "self_attn.o_proj.weight"] = rand(cfg.hidden_size, hd)
+        sd[p + "mlp.gate_proj.weight"] = rand(cfg.intermediate_size, cfg.hidden_size)
+        sd[p + "mlp.up_proj.weight"] = rand(cfg.intermediate_size, cfg.hidden_size)
+        sd[p + "mlp.down_proj.weight"] = rand(cfg.hidden_size, cfg.intermediate_size)
+        sd[p + "input_layernorm.weight"] = torch.ones(cfg.hidden_size)
+        sd[p + "post_attention_layernorm.weight"] = torch.ones(cfg.hidden_size)
+    shared["model.embed_tokens.weight"] = rand(cfg.vocab_size, cfg.hidden_size)
+    shared["model.norm.weight"] = torch.ones(cfg.hidden_size)
+    shared["lm_head.weight"] = rand(cfg.vocab_size, cfg.hidden_size)
+    for fname, sd in (("layer0.safetensors", sd0), ("layer1.safetensors", sd1),
+                      ("shared.safetensors", shared)):
+        save_file(sd, os.path.join(tmpdir, fname))
+        for k in sd:
+            weight_map[k] = fname
+    with open(os.path.join(tmpdir, "model.safetensors.index.json"), "w") as f:
+        json.dump({"weight_map": weight_map}, f)
+    return cfg
+
+
+def test_refit_chunks_over_tcp_transport(tmp_path):
+    """Two peers over real TCP: the fetcher pulls only its layer range's
+    files, verifies every chunk CID, reassembles and hot-reloads; a corrupted
+    chunk stream is rejected with RefitError."""
+    import threading
+
+    from parallax_amd.p2p.peer_executor import PeerExecutor
+    from parallax_amd.p2p.refit import RefitError, RefitFetcher, RefitPublisher
+    from parallax_amd.p2p.transport import TcpTransport
+
+    ckpt = str(tmp_path / "ckpt_v1")
+    cfg = _save_tiny_ckpt(ckpt, seed=7)
+
+    t_pub = TcpTransport("pub", host="127.0.0.1")
+    t_sub = TcpTransport("sub", host="127.0.0.1")
+    t_pub.set_peer_addr("sub", "127.0.0.1", t_sub.port)
+    t_sub.set_peer_addr("pub", "127.0.0.1", t_pub.port)
+    try:
+        publisher = RefitPublisher(ckpt, version=1, chunk_size=1024)
+        pub_peer = PeerExecutor(cfg, 0, 1, "pub", t_pub, random_weights=True)
+        pub_peer.set_refit_publisher(publisher)
+        sub_peer = PeerExecutor(cfg, 1, 2, "sub", t_sub, random_weights=True)
+
+        stop = threading.Event()
+
+        def pump():
+            while not stop.is_set():
+                pub_peer.step(recv_timeout=0.05)
+
+        th = threading.Thread(target=pump, daemon=True)
+        th.start()
+        before = {k: v.clone() for k, v in sub_peer.model.state_dict().items()}
+        vdir = sub_peer.refit_from_peer(
+            "pub", publisher.manifest, str(tmp_path / "versions"), timeout=20
+        )
+        stop.set()
+        th.join(timeout=5)
+
+        # layer-range filter: only layer1 + shared files fetched
+        got = sorted(os.listdir(vdir))
+        assert "layer1.safetensors" in got and "shared.safetensors" in got
+        assert "layer0.safetensors" not in got
+        # weights actually changed to the checkpoint's values
+        after = sub_peer.model.state_dict()
+        changed = any(
+            not torch.equal(before[k], after[k]) for k in before
+        )
+        assert changed
+
+        # corrupted stream -> RefitError (checksum rejection after retries)
+        bad = RefitFetcher(str(tmp_path / "bad"))
+        def corrupt_get(name, idx):
+            data = bytearray(publisher.get_chunk(name, idx))
+            data[0] ^= 0xFF
+            return bytes(data)
+        with pytest.raises(RefitError):
+            bad.fetch(publisher.manifest, corrupt_get, layer_range=(1, 2))
+    finally:
+        t_pub.close()
+        t_sub.close()
+
+
+def test_refit_version_gc(tmp_path):
+    """The fetcher keeps only the newest 3 version directories."""
+    from parallax_amd.p2p.refit import RefitFetcher, RefitPublisher
+
+    ckpt = str(tmp_path / "ck")
+    _save_tiny_ckpt(ckpt, seed=1)
+    fetcher = RefitFetcher(str(tmp_path / "vers"))
+    for v in range(1, 6):
+        pub = RefitPublisher(ckpt, version=v, chunk_size=2048)
+        fetcher.fetch(pub.manifest, pub.get_chunk)
+    kept = sorted(os.listdir(str(tmp_path / "vers")))
+    assert kept == ["v3", "v4", "v5"]
