@@ -336,10 +336,75 @@ def msa_paged_attention_decode(
 
 # indexer / block-score helpers run as torch compositions on both CPU and GPU
 # (GEMV-scale work; the hot sparse gathers above are the HIP kernels)
-dsa_indexer_scores = ref.dsa_indexer_scores
-store_indexer_cache = ref.store_indexer_cache
-msa_block_scores = ref.msa_block_scores
-msa_topk_tokens = ref.msa_topk_tokens
+def dsa_indexer_scores(q_index, index_cache, head_weights, block_tables,
+                       seq_lens):
+    """Weighted relu(q.k) scores over the paged index cache (DeepSeek-V3.2
+    indexer). HIP MFMA kernel on GPU for the shared-key layout; torch
+    reference elsewhere (reference Metal analogue: dsa_indexer.metal)."""
+    if (
+        q_index.is_cuda and index_cache.dim() == 3
+        and q_index.dtype == torch.bfloat16
+        and index_cache.dtype == torch.bfloat16
+        and q_index.shape[1] <= 64 and q_index.shape[2] in (64, 128)
+    ):
+        ext = _require_ext("dsa_indexer_scores")
+        max_ctx = int(seq_lens.max())
+        return ext.dsa_indexer_scores(
+            q_index, index_cache, head_weights.float(),
+            block_tables.int(), seq_lens.int(), max_ctx,
+        )
+    return ref.dsa_indexer_scores(
+        q_index, index_cache, head_weights, block_tables, seq_lens
+    )
+
+
+def store_indexer_cache(index_keys, index_cache, slot_mapping):
+    """Scatter indexer keys by slot (pad slots -1 land in the trash block)."""
+    if index_keys.is_cuda and index_keys.dtype == torch.bfloat16 \
+            and index_cache.dtype == torch.bfloat16:
+        ext = _require_ext("store_indexer_cache")
+        bs = index_cache.shape[1]
+        trash_slot = (index_cache.shape[0] - 1) * bs
+        ext.store_indexer_cache(
+            index_keys.reshape(index_keys.shape[0], -1),
+            index_cache, slot_mapping.long(), trash_slot,
+        )
+        return
+    ref.store_indexer_cache(index_keys, index_cache, slot_mapping)
+
+
+def msa_block_scores(q, k_cache, block_tables, seq_lens, sparse_block):
+    """Mean-pooled per-sparse-block K scores vs the head-mean query
+    (MiniMax-M3 indexer phase 1; reference msa_indexer.metal)."""
+    if q.is_cuda and q.dtype == torch.bfloat16 \
+            and k_cache.dtype == torch.bfloat16 and q.shape[2] in (64, 128):
+        ext = _require_ext("msa_block_scores")
+        max_sb = (int(seq_lens.max()) + sparse_block - 1) // sparse_block
+        return ext.msa_block_scores(
+            q, k_cache, block_tables.int(), seq_lens.int(), sparse_block,
+            max(1, max_sb),
+        )
+    return ref.msa_block_scores(q, k_cache, block_tables, seq_lens, sparse_block)
+
+
+def msa_topk_tokens(block_scores, seq_lens, sparse_block, topk_blocks,
+                    init_blocks=1, local_blocks=2):
+    """Top-k sparse blocks expanded to sorted token positions (-1 padded),
+    always keeping the init/local blocks (phase 2)."""
+    if block_scores.is_cuda:
+        ext = _require_ext("msa_topk_tokens")
+        keep_blocks = topk_blocks + init_blocks + local_blocks
+        max_sb = block_scores.shape[1]
+        max_positions = min(int(seq_lens.max()),
+                            min(keep_blocks, max_sb) * sparse_block)
+        return ext.msa_topk_tokens(
+            block_scores.float(), seq_lens.int(), sparse_block, topk_blocks,
+            init_blocks, local_blocks, max(1, max_positions),
+        )
+    return ref.msa_topk_tokens(
+        block_scores, seq_lens, sparse_block, topk_blocks, init_blocks,
+        local_blocks,
+    )
 
 
 # -- activations ----------------------------------------------------------------------

@@ -57,6 +57,16 @@ void launch_moe_gate_up(void*, const void*, const void*, const int*,
 void launch_moe_down(void*, const void*, const void*, const float*, const int*,
                      const int*, const int64_t*, const int*, int, int, int,
                      int, int, hipStream_t);
+void launch_dsa_indexer_scores(float*, const void*, const void*, const float*,
+                               const int*, const int*, int, int, int, int,
+                               int, int, hipStream_t, bool*);
+void launch_store_indexer_cache(void*, const void*, const int64_t*, int,
+                                int64_t, int, hipStream_t);
+void launch_msa_block_scores(float*, const void*, const void*, const int*,
+                             const int*, int, int, int, int, int, int, int,
+                             int, hipStream_t, bool*);
+void launch_msa_topk_tokens(int64_t*, const float*, const int*, int, int, int,
+                            int, int, int, int, hipStream_t);
 void launch_quantize_fp8_rows(void*, float*, const void*, int, int,
                               hipStream_t);
 void launch_moe_gate_up_fp8(void*, const void*, const float*, const void*,
@@ -425,6 +435,89 @@ void moe_forward(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
                   stream);
 }
 
+torch::Tensor dsa_indexer_scores(torch::Tensor q_index,
+                                 torch::Tensor index_cache,
+                                 torch::Tensor head_weights,
+                                 torch::Tensor block_tables,
+                                 torch::Tensor seq_lens, int64_t max_ctx) {
+  CHECK_GPU(q_index);
+  CHECK_BF16(q_index);
+  CHECK_BF16(index_cache);
+  TORCH_CHECK(index_cache.dim() == 3, "index cache must be [NB, BS, Di]");
+  TORCH_CHECK(head_weights.scalar_type() == at::kFloat);
+  const int B = q_index.size(0);
+  const int Hi = q_index.size(1);
+  const int Di = q_index.size(2);
+  const int BS = index_cache.size(1);
+  auto scores = torch::empty(
+      {B, max_ctx}, q_index.options().dtype(at::kFloat));
+  bool launched = false;
+  launch_dsa_indexer_scores(
+      scores.data_ptr<float>(), q_index.contiguous().data_ptr(),
+      index_cache.data_ptr(), head_weights.contiguous().data_ptr<float>(),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), B, Hi, Di, BS,
+      block_tables.size(1), (int)max_ctx, cur_stream(), &launched);
+  TORCH_CHECK(launched, "no dsa_indexer kernel for Hi=", Hi, " Di=", Di);
+  return scores;
+}
+
+void store_indexer_cache(torch::Tensor index_keys, torch::Tensor index_cache,
+                         torch::Tensor slot_mapping, int64_t trash_slot) {
+  CHECK_GPU(index_keys);
+  CHECK_BF16(index_keys);
+  CHECK_BF16(index_cache);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  const int T = index_keys.size(0);
+  const int DI = index_keys.size(-1);
+  TORCH_CHECK(DI % 8 == 0);
+  launch_store_indexer_cache(index_cache.data_ptr(),
+                             index_keys.contiguous().data_ptr(),
+                             slot_mapping.data_ptr<int64_t>(), T, trash_slot,
+                             DI, cur_stream());
+}
+
+torch::Tensor msa_block_scores(torch::Tensor q, torch::Tensor k_cache,
+                               torch::Tensor block_tables,
+                               torch::Tensor seq_lens, int64_t sparse_block,
+                               int64_t max_sparse_blocks) {
+  CHECK_GPU(q);
+  CHECK_BF16(q);
+  CHECK_BF16(k_cache);
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hk = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  auto out = torch::empty(
+      {B, max_sparse_blocks}, q.options().dtype(at::kFloat));
+  bool launched = false;
+  launch_msa_block_scores(out.data_ptr<float>(), q.contiguous().data_ptr(),
+                          k_cache.data_ptr(), block_tables.data_ptr<int>(),
+                          seq_lens.data_ptr<int>(), B, Hq, Hk, D, BS,
+                          block_tables.size(1), (int)sparse_block,
+                          (int)max_sparse_blocks, cur_stream(), &launched);
+  TORCH_CHECK(launched, "no msa_block_scores kernel for D=", D);
+  return out;
+}
+
+torch::Tensor msa_topk_tokens(torch::Tensor block_scores,
+                              torch::Tensor seq_lens, int64_t sparse_block,
+                              int64_t topk_blocks, int64_t init_blocks,
+                              int64_t local_blocks, int64_t max_positions) {
+  CHECK_GPU(block_scores);
+  TORCH_CHECK(block_scores.scalar_type() == at::kFloat);
+  const int B = block_scores.size(0);
+  auto out = torch::empty(
+      {B, max_positions}, block_scores.options().dtype(at::kLong));
+  launch_msa_topk_tokens(out.data_ptr<int64_t>(),
+                         block_scores.contiguous().data_ptr<float>(),
+                         seq_lens.data_ptr<int>(), B, (int)sparse_block,
+                         block_scores.size(1), (int)topk_blocks,
+                         (int)init_blocks, (int)local_blocks,
+                         (int)max_positions, cur_stream());
+  return out;
+}
+
 void moe_forward_fp8(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
                      torch::Tensor w_gu_scale, torch::Tensor w_down,
                      torch::Tensor w_down_scale, torch::Tensor perm,
@@ -539,6 +632,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("moe_forward", &moe_forward);
   m.def("moe_forward_fp8", &moe_forward_fp8);
+  m.def("dsa_indexer_scores", &dsa_indexer_scores);
+  m.def("store_indexer_cache", &store_indexer_cache);
+  m.def("msa_block_scores", &msa_block_scores);
+  m.def("msa_topk_tokens", &msa_topk_tokens);
   m.def("mla_paged_attention_decode", &mla_paged_attention_decode);
   m.def("prefill_attention", &prefill_attention);
   m.def("rmsnorm", &rmsnorm);

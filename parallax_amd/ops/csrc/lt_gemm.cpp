@@ -133,7 +133,8 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
             s.handle, hipblaslt_ext::GemmType::HIPBLASLT_GEMM, HIPBLAS_OP_T,
             HIPBLAS_OP_N, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
             HIPBLAS_COMPUTE_32F, all) == HIPBLAS_STATUS_SUCCESS) {
-      // big-M shapes take ~ms per timed probe: cap the screened set
+      // the screening loop below is additionally time-budgeted, but cap the
+      // supported-check pass too (it walks thousands of entries)
       const size_t cap = (M <= 1024) ? 2048 : 384;
       size_t kept = 0;
       const float alpha1 = 1.f, beta1 = 0.f;
@@ -174,8 +175,22 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
   };
   // screen every candidate with one timed iteration, then re-time the best
   // few — keeps one-off tuning of huge prefill shapes to ~tens of ms
+  // Per-shape screening budget: the heuristic's own candidates (first in the
+  // list) are always timed; the widened set is screened until the budget runs
+  // out. Bounds the one-time tuning stall a previously-unseen shape causes
+  // mid-serving (PARALLAX_LT_TUNE_MS, default 150 ms/shape).
+  static const float kBudgetMs = [] {
+    const char* e = getenv("PARALLAX_LT_TUNE_MS");
+    return e ? (float)atof(e) : 150.f;
+  }();
   std::vector<std::pair<float, int>> screened;
-  for (int i = 0; i < n_results; ++i) screened.emplace_back(time_algo(i, 1), i);
+  float spent_ms = 0.f;
+  for (int i = 0; i < n_results; ++i) {
+    if (i >= n_heur && spent_ms > kBudgetMs) break;
+    const float ms = time_algo(i, 1);
+    if (ms < 1e29f) spent_ms += 2.f * ms;  // warm + timed run
+    screened.emplace_back(ms, i);
+  }
   std::sort(screened.begin(), screened.end());
   int best = -1;
   float best_ms = 1e30f;

@@ -26,6 +26,7 @@ sources = [
     os.path.join(CSRC, "prefill_attention.hip"),
     os.path.join(CSRC, "moe.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
+    os.path.join(CSRC, "indexer.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 
