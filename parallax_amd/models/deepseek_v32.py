@@ -140,6 +140,15 @@ class DSAMLAAttention(MLAAttention):
         msl = meta.max_seq_len or int(meta.seq_lens.max())
         npages = min((msl + bs - 1) // bs, meta.block_tables.shape[1])
         msl = min(msl, npages * bs)
+        if q_idx.is_cuda and q_idx.dtype == torch.bfloat16 \
+                and idx_cache.dim() == 3:
+            # HIP MFMA score kernel: streams the paged index cache directly
+            # (no [B, msl, Di] gather materialization — at 256k ctx that
+            # gather alone is 64 MB/req/layer)
+            return ops.dsa_indexer_scores(
+                q_idx, idx_cache, w_idx.float(), meta.block_tables,
+                meta.seq_lens, max_ctx=msl,
+            )
         keys = idx_cache[meta.block_tables[:, :npages].long()].reshape(
             B, npages * bs, -1
         )[:, :msl]                                       # [B, msl, Di]

@@ -337,10 +337,12 @@ def msa_paged_attention_decode(
 # indexer / block-score helpers run as torch compositions on both CPU and GPU
 # (GEMV-scale work; the hot sparse gathers above are the HIP kernels)
 def dsa_indexer_scores(q_index, index_cache, head_weights, block_tables,
-                       seq_lens):
+                       seq_lens, max_ctx=None):
     """Weighted relu(q.k) scores over the paged index cache (DeepSeek-V3.2
     indexer). HIP MFMA kernel on GPU for the shared-key layout; torch
-    reference elsewhere (reference Metal analogue: dsa_indexer.metal)."""
+    reference elsewhere (reference Metal analogue: dsa_indexer.metal).
+    Pass max_ctx (e.g. the graph ctx bucket) to stay hipGraph-capturable —
+    int(seq_lens.max()) syncs the host otherwise."""
     if (
         q_index.is_cuda and index_cache.dim() == 3
         and q_index.dtype == torch.bfloat16
@@ -348,7 +350,8 @@ def dsa_indexer_scores(q_index, index_cache, head_weights, block_tables,
         and q_index.shape[1] <= 64 and q_index.shape[2] in (64, 128)
     ):
         ext = _require_ext("dsa_indexer_scores")
-        max_ctx = int(seq_lens.max())
+        if max_ctx is None:
+            max_ctx = int(seq_lens.max())
         return ext.dsa_indexer_scores(
             q_index, index_cache, head_weights.float(),
             block_tables.int(), seq_lens.int(), max_ctx,

@@ -186,3 +186,78 @@ def test_msa_kernel_gpu_per_head_positions():
         sl.cpu(), pos.cpu(), 0.088,
     )
     torch.testing.assert_close(out.float().cpu(), expect, atol=4e-2, rtol=4e-2)
+
+
+@pytest.mark.gpu
+def test_dsa_indexer_scores_kernel_gpu():
+    """HIP MFMA indexer score pass vs the fp32 torch reference (VERDICT item
+    5; reference Metal analogue dsa_indexer.metal)."""
+    torch.manual_seed(0)
+    B, Hi, Di, bs = 3, 64, 128, 32
+    seq_lens = torch.tensor([70, 128, 15], dtype=torch.int32, device="cuda")
+    max_blocks = 4
+    nb = B * max_blocks + 1
+    cache = (torch.randn(nb, bs, Di, device="cuda") * 0.3).bfloat16()
+    bt = torch.arange(B * max_blocks, dtype=torch.int32, device="cuda") \
+        .reshape(B, max_blocks).contiguous()
+    q = (torch.randn(B, Hi, Di, device="cuda") * 0.3).bfloat16()
+    w = torch.rand(B, Hi, device="cuda")
+
+    got = ops.dsa_indexer_scores(q, cache, w, bt, seq_lens)
+    exp = ref.dsa_indexer_scores(q.float().cpu(), cache.float().cpu(),
+                                 w.cpu(), bt.cpu(), seq_lens.cpu())
+    for i in range(B):
+        L = int(seq_lens[i])
+        torch.testing.assert_close(got[i, :L].cpu(), exp[i, :L],
+                                   atol=5e-2, rtol=5e-2)
+        assert (got[i, L:].cpu() < -1e29).all()
+
+
+@pytest.mark.gpu
+def test_store_indexer_cache_kernel_gpu():
+    torch.manual_seed(1)
+    T, Di, bs, nb = 10, 128, 16, 4  # last block = trash
+    cache = torch.zeros(nb, bs, Di, dtype=torch.bfloat16, device="cuda")
+    keys = (torch.randn(T, Di, device="cuda") * 0.5).bfloat16()
+    slots = torch.tensor([0, 5, 17, 31, -1, 40, 2, -1, 33, 47],
+                         dtype=torch.int64, device="cuda")
+    ops.store_indexer_cache(keys, cache, slots)
+    flat = cache.reshape(-1, Di)
+    for t, s in enumerate(slots.tolist()):
+        if s >= 0:
+            torch.testing.assert_close(flat[s], keys[t])
+    # pad rows landed in the trash block, not in live slots
+    assert (flat[: (nb - 1) * bs][~torch.isin(
+        torch.arange((nb - 1) * bs, device="cuda"),
+        slots[slots >= 0])].float().abs().sum(dim=-1) == 0).all()
+
+
+@pytest.mark.gpu
+def test_msa_indexer_kernels_gpu():
+    """HIP block-score + top-k expansion vs the torch reference (per-row
+    position SETS compared; widths may differ by padding)."""
+    torch.manual_seed(2)
+    B, Hq, Hk, D, bs = 2, 8, 2, 128, 16
+    seq_lens = torch.tensor([150, 83], dtype=torch.int32, device="cuda")
+    max_blocks = 10
+    kc = (torch.randn(B * max_blocks + 1, Hk, bs, D, device="cuda") * 0.3).bfloat16()
+    bt = torch.arange(B * max_blocks, dtype=torch.int32, device="cuda") \
+        .reshape(B, max_blocks).contiguous()
+    q = (torch.randn(B, Hq, D, device="cuda") * 0.3).bfloat16()
+
+    sparse_block = 32
+    got = ops.msa_block_scores(q, kc, bt, seq_lens, sparse_block)
+    exp = ref.msa_block_scores(q.float().cpu(), kc.float().cpu(), bt.cpu(),
+                               seq_lens.cpu(), sparse_block)
+    for i in range(B):
+        nsb = (int(seq_lens[i]) + sparse_block - 1) // sparse_block
+        torch.testing.assert_close(got[i, :nsb].cpu(), exp[i, :nsb],
+                                   atol=5e-2, rtol=5e-2)
+
+    pos_g = ops.msa_topk_tokens(got, seq_lens, sparse_block, topk_blocks=2)
+    pos_e = ref.msa_topk_tokens(exp, seq_lens.cpu(), sparse_block,
+                                topk_blocks=2)
+    for i in range(B):
+        sg = set(pos_g[i][pos_g[i] >= 0].tolist())
+        se = set(pos_e[i][pos_e[i] >= 0].tolist())
+        assert sg == se, f"row {i}: {sorted(sg)[:8]} vs {sorted(se)[:8]}"
