@@ -12,6 +12,7 @@ from . import gpt_oss  # noqa: F401
 from . import glm4_moe  # noqa: F401
 from . import qwen3_next  # noqa: F401
 from . import minimax_m2  # noqa: F401
+from . import step3p5  # noqa: F401
 from . import qwen3_5  # noqa: F401
 from . import minimax  # noqa: F401
 from . import minimax_m3  # noqa: F401

@@ -77,6 +77,7 @@ class ModelConfig:
 
     # -- attention sinks (gpt-oss) --------------------------------------------------
     attention_sinks: bool = False
+    use_attn_gate: bool = False    # step3p5 head-wise sigmoid output gate
 
     # -- hybrid linear attention (qwen3-next gated deltanet) --------------------------
     linear_num_key_heads: int = 0
@@ -162,11 +163,14 @@ class ModelConfig:
             o_proj_bias=cfg.get("attention_bias", False)
             and cfg.get("model_type") == "gpt_oss",
             attention_sinks=cfg.get("model_type") == "gpt_oss",
+            use_attn_gate=cfg.get("use_head_wise_attn_gate",
+                                  cfg.get("model_type") == "step3p5"),
             qk_norm=cfg.get("use_qk_norm", False)
             or cfg.get("model_type", "") in ("qwen3", "qwen3_moe", "qwen3_next",
                                              "minimax_m2", "qwen3_5_text",
                                              "qwen3_5_moe_text",
-                                             "minimax_m3_vl_text", "minimax_m3"),
+                                             "minimax_m3_vl_text", "minimax_m3",
+                                             "step3p5"),
             qk_norm_full=cfg.get("model_type", "") == "minimax_m2",
             sliding_window=sliding,
             layer_types=layer_types,

@@ -118,6 +118,10 @@ class LlamaAttention(nn.Module):
                 self.scale, self.sliding_window, sinks=sinks,
                 max_seq_len=meta.max_seq_len or None,
             )
+        return self._project_out(attn, x, T)
+
+    def _project_out(self, attn: torch.Tensor, x: torch.Tensor, T: int):
+        """Output projection; subclasses may gate `attn` first (step3p5)."""
         return self.o_proj(attn.reshape(T, self.num_heads * self.head_dim))
 
 

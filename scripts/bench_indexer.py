@@ -11,6 +11,10 @@ import time
 
 import torch
 
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 from parallax_amd import ops
 
 
