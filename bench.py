@@ -237,6 +237,13 @@ def main():
     if wdtype == "fp8":
         eargs.moe_weight_dtype = "fp8"
     engine = Engine(cfg, eargs, comm=comm, random_weights=True)
+    if use_gpu:
+        # pre-tune hipBLASLt algo picks for the Ms this run will see:
+        # prefill chunks, decode batch (graph bucket), per-chunk sampling rows
+        chunk_m = min(eargs.max_num_tokens_per_batch,
+                      engine_batch * args.prompt_len)
+        sample_m = max(1, eargs.max_num_tokens_per_batch // max(1, args.prompt_len))
+        engine.warmup_gemms([chunk_m, engine_batch, sample_m])
 
     # synthetic prompts, unique tokens so nothing prefix-shares (per-replica
     # seeds in DP mode so replicas do not share content)

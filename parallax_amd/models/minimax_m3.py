@@ -206,8 +206,8 @@ class M3SparseAttention(LlamaAttention):
             kf = (
                 k_cache[tabs].permute(0, 2, 1, 3).reshape(npages * bs_page, Hk, D)[:L]
             ).float()                                     # [L, Hk, D]
-            vf = (
-                v_cache[tabs].permute(0, 2, 1, 3).reshape(npages * bs_page, Hk, D)[:L]
+            vf = (  # V cache is transposed [.., Hk, D, bs]
+                v_cache[tabs].permute(0, 3, 1, 2).reshape(npages * bs_page, Hk, D)[:L]
             ).float()
             keys_i = idx_cache[tabs].reshape(npages * bs_page, -1)[:L].float()
             qi = q_idx[t0 : t0 + QL].float()              # [QL, Hi, Di]

@@ -148,10 +148,13 @@ __global__ void rope_and_cache_kernel(
   const int64_t blk = slot >= 0 ? slot / BS : 0;
   const int64_t off = slot >= 0 ? slot % BS : 0;
   const size_t base_off = (((size_t)blk * Hk) * BS + off) * D;
+  // v cache is transposed [NB, Hk, D, BS]: per-(h,d) element lands at column
+  // `off` of the d row
+  const size_t vbase_off = ((size_t)blk * Hk) * (size_t)D * BS + off;
   uint16_t* kdst16 = FP8 ? nullptr : (uint16_t*)k_cache_v + base_off;
-  uint16_t* vdst16 = FP8 ? nullptr : (uint16_t*)v_cache_v + base_off;
+  uint16_t* vdst16 = FP8 ? nullptr : (uint16_t*)v_cache_v + vbase_off;
   uint8_t* kdst8 = FP8 ? (uint8_t*)k_cache_v + base_off : nullptr;
-  uint8_t* vdst8 = FP8 ? (uint8_t*)v_cache_v + base_off : nullptr;
+  uint8_t* vdst8 = FP8 ? (uint8_t*)v_cache_v + vbase_off : nullptr;
 
   // q: rope in place
   for (int i = threadIdx.x; i < Hq * half; i += blockDim.x) {
@@ -196,15 +199,18 @@ __global__ void rope_and_cache_kernel(
         kdst16[(size_t)h * BS * D + d] = f32_to_bf16_bits(x);
     }
   }
-  // v: copy (quantized when FP8)
+  // v: transpose-scatter into the [.., D, BS] layout (quantized when FP8) —
+  // one 2 B store per element at decode (1 token); the attention kernels'
+  // fragment loads read V^T contiguously in exchange
   for (int i = threadIdx.x; i < Hk * D; i += blockDim.x) {
     const int h = i / D;
     const int d = i % D;
     const float x = bf16_bits_to_f32(v[(size_t)t * v_stride + h * D + d]);
+    const size_t dst = ((size_t)h * D + d) * BS;
     if (FP8)
-      vdst8[(size_t)h * BS * D + d] = f32_to_fp8_e4m3(x * inv_v_scale);
+      vdst8[dst] = f32_to_fp8_e4m3(x * inv_v_scale);
     else
-      vdst16[(size_t)h * BS * D + d] = f32_to_bf16_bits(x);
+      vdst16[dst] = f32_to_bf16_bits(x);
   }
 }
 
@@ -247,12 +253,16 @@ __global__ void reshape_and_cache_kernel(
   for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
     const int h = (i * 8) / D;
     const int d = (i * 8) % D;
-    // dst row: cache[blk][h][off][d]
+    // k row: cache[blk][h][off][d]
     const size_t dst = (((size_t)blk * Hk + h) * BS + off) * D + d;
     *reinterpret_cast<int4*>(k_cache + dst) =
         *reinterpret_cast<const int4*>(ksrc + i * 8);
-    *reinterpret_cast<int4*>(v_cache + dst) =
-        *reinterpret_cast<const int4*>(vsrc + i * 8);
+    // v transposed: cache[blk][h][d + j][off]
+    const int4 vv = *reinterpret_cast<const int4*>(vsrc + i * 8);
+    const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v_cache[(((size_t)blk * Hk + h) * D + d + j) * BS + off] = ve[j];
   }
 }
 

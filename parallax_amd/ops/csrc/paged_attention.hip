@@ -190,11 +190,17 @@ __global__ __launch_bounds__(ATTN_THREADS) void paged_attention_kernel(
     for (int it = 0; it < VIT; ++it) {
       const int tok2 = base_tok + tp + it * NT_PAR;
       if (tok2 >= tok_begin && tok2 < tok_end) {
+        // v cache is transposed [NB, Hk, D, BS]: this token's 8 d-elements
+        // are strided by BLOCK_SIZE
         const int blk = btab[tok2 / BLOCK_SIZE];
         const int off = tok2 % BLOCK_SIZE;
-        vv[it] = load_bf16x8(
-            v_cache + (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM +
-            dc * 8);
+        const size_t vbase =
+            (((size_t)blk * Hk + hk) * HEAD_DIM + dc * 8) * BLOCK_SIZE + off;
+        alignas(16) uint16_t e[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          e[j] = v_cache[vbase + (size_t)j * BLOCK_SIZE];
+        vv[it].raw = *reinterpret_cast<const int4*>(e);
       } else {
         vv[it].raw = make_int4(0, 0, 0, 0);
       }

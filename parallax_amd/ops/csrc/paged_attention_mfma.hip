@@ -1,14 +1,24 @@
 // MFMA decode attention: the GQA query group rides the matrix cores.
 //
-// The VALU decode kernel (paged_attention.hip) computes G x D dot products per
-// token with unpack+fma chains — fine at G<=4, VALU-bound at G>=8. Here the
-// whole KV group is one MFMA N-tile (G <= 16 query heads = 16 columns):
-//   S^T[64tok x 16q] = K_tile · Q^T      (A = K rows, B = Q rows — natural)
-//   O^T[D x 16q]    += V^T · P           (V transposed during staging)
-// using the same LDS layout machinery as the prefill kernel (XOR-swizzled rows,
-// ds_read_b128 fragments). One workgroup per (sequence, kv_head, partition);
-// partitions are fixed-size (graph-capture stable) and combine through the
-// same reduce kernel as the VALU path.
+// Round-2 restructure: decode attention has ZERO K/V reuse inside a workgroup
+// (each K row feeds one QK^T fragment, each V^T row feeds one PV fragment),
+// so the round-1 LDS staging (and its 32-scalar-store V transpose) was pure
+// overhead. The V cache is now stored TRANSPOSED ([NB, Hk, D, BS], see
+// kv_cache.py), which makes both MFMA operand fragments DIRECT 16-byte HBM
+// loads:
+//   S^T[64tok x 16q] = K_tile . Q^T   A-frag: 8 consecutive d of one token
+//   O^T[D x 16q]    += V^T . P        A-frag: 8 consecutive tokens of one d
+// Q lives in registers; only P (2 KB) and the softmax reductions use LDS, so
+// occupancy is bounded by registers, not LDS, and the kernel runs at the KV
+// stream rate. K/V fragments are register double-buffered across the 64-token
+// tiles (issue tile n+1's loads before tile n's math; the compiler's
+// per-register waitcnts do the rest — guide §6 G15).
+//
+// One workgroup per (sequence, kv_head, partition); partitions are fixed-size
+// (graph-capture stable) and combine through the shared reduce kernel.
+// Sliding window, softcap, sinks, fp8 KV (dequant at load), and the MSA
+// SPARSE mode (explicit token-position lists; per-element V gather since
+// sparse tokens are not contiguous) ride the same template.
 
 #include "common.h"
 
@@ -23,7 +33,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     float* __restrict__ tmp_ml,        // [B, Hq, P, 2]
     const uint16_t* __restrict__ q,    // [B, Hq, D] rows at q_stride
     const void* __restrict__ k_cache_v,  // [NB, Hk, BS, D] bf16 | fp8
-    const void* __restrict__ v_cache_v,
+    const void* __restrict__ v_cache_v,  // [NB, Hk, D, BS] (transposed)
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     const int max_blocks, const int Hk, const int G, const int64_t q_stride,
@@ -35,8 +45,6 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   const int seq = blockIdx.y;
   const int hk = blockIdx.x;
   const int L = seq_lens[seq];
-  // MSA sparse mode: iterate an explicit token-position list (-1 padded);
-  // pos_heads > 1 = one independent selection per kv head (minimax-m3)
   const int* pos_row =
       SPARSE ? token_positions +
                    ((size_t)seq * pos_heads + (pos_heads > 1 ? hk : 0)) *
@@ -71,36 +79,25 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     return;
   }
 
-  __shared__ uint16_t Kl[DM_KTILE * HEAD_DIM];   // row = token, swz
-  __shared__ uint16_t VTl[HEAD_DIM * DM_KTILE];  // row = dim,   swz
-  __shared__ uint16_t Ql[16 * HEAD_DIM];         // row = q head, swz
-  __shared__ uint16_t Pl[16 * DM_KTILE];         // row = q head, swz
+  __shared__ uint16_t Pl[16 * DM_KTILE];  // row = q head, swz
   __shared__ float m_s[16], l_s[16], resc[16];
   __shared__ float wred[4][16];
 
-  // ---- stage the G query heads (zeros pad to 16 rows) ---------------------------
+  constexpr int KC = HEAD_DIM / 32;      // 16-B K/Q chunks per row
+  constexpr int MT = HEAD_DIM / 16 / 4;  // PV M-tiles per wave
+
+  // ---- Q fragments in registers (head l15; zero-pad heads >= G) ---------------
+  bf16x8v qfrag[KC];
   {
-    const int qrow = tid & 15;
-    const int dv = tid >> 4;  // 16 chunks
-    const bool ok = qrow < G;
+    const bool ok = l15 < G;
 #pragma unroll
-    for (int c = 0; c < HEAD_DIM / 128; ++c) {
-      const int d = (dv + c * 16) * 8;
+    for (int s = 0; s < KC; ++s) {
       int4 val = make_int4(0, 0, 0, 0);
       if (ok)
         val = *reinterpret_cast<const int4*>(
-            q + (size_t)seq * q_stride + (hk * G + qrow) * HEAD_DIM + d);
-      const int byte = swz(qrow * HEAD_DIM * 2 + d * 2, qrow);
-      *reinterpret_cast<int4*>(reinterpret_cast<char*>(Ql) + byte) = val;
-    }
-    if (HEAD_DIM == 64 && dv < 8) {  // D=64: only 8 chunks
-      const int d = dv * 8;
-      int4 val = make_int4(0, 0, 0, 0);
-      if (ok)
-        val = *reinterpret_cast<const int4*>(
-            q + (size_t)seq * q_stride + (hk * G + qrow) * HEAD_DIM + d);
-      const int byte = swz(qrow * HEAD_DIM * 2 + d * 2, qrow);
-      *reinterpret_cast<int4*>(reinterpret_cast<char*>(Ql) + byte) = val;
+            q + (size_t)seq * q_stride + (hk * G + l15) * HEAD_DIM + s * 32 +
+            l4 * 8);
+      qfrag[s] = *reinterpret_cast<const bf16x8v*>(&val);
     }
   }
   if (tid < 16) {
@@ -108,104 +105,120 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     l_s[tid] = 0.f;
   }
 
-  constexpr int MT = HEAD_DIM / 16 / 4;  // M-tiles of O^T per wave
   f32x4v acc_o[MT] = {};
-
   const int* btab = block_tables + (size_t)seq * max_blocks;
   const int kt_begin = tok_begin / DM_KTILE;
   const int kt_end = (tok_end + DM_KTILE - 1) / DM_KTILE;
 
-  // register double-buffered staging (async-STAGE split): loads for tile n+1
-  // are issued right after tile n's LDS write so HBM latency hides under the
-  // MFMA/softmax work (guide §6 G15)
-  const int stg_tok = tid & 63;
-  const int stg_dv = tid >> 6;
-  const int stg_d0 = stg_dv * (HEAD_DIM / 4);
-  int4 kreg[HEAD_DIM / 32], vreg[HEAD_DIM / 32];
+  // logical token -> global token (sparse indirection; clamped for address
+  // safety — masked scores / zero p neutralize the values)
+  auto map_tok = [&](int j, bool& valid) -> int {
+    valid = j >= tok_begin && j < tok_end;
+    int gt = j;
+    if (SPARSE) {
+      gt = valid ? pos_row[j] : 0;
+      valid = valid && gt >= 0 && gt < L;
+    }
+    return min(max(gt, 0), L - 1);
+  };
 
-  auto load_tile = [&](int kt) {
-    const int j = kt * DM_KTILE + stg_tok;
-    int gtok = j;
-    bool ok = j >= tok_begin && j < tok_end;
-    if (SPARSE && ok) {
-      gtok = pos_row[j];
-      ok = gtok >= 0 && gtok < L;
-    }
-    size_t row_off = 0;
-    if (ok) {
-      const int blk = btab[gtok / BLOCK_SIZE];
-      const int off = gtok % BLOCK_SIZE;
-      row_off = (((size_t)blk * Hk + hk) * BLOCK_SIZE + off) * HEAD_DIM;
-    }
+  // ---- register double-buffered tile loads ------------------------------------
+  // K: lane covers token (16*wid + l15), all KC d-chunks.
+  // V^T: lane covers d rows (wid*MT+mt)*16 + l15, token chunk s*32 + l4*8.
+  int4 kreg[2][KC];
+  int4 vreg[2][2][MT];
+  bool kvalid[2];
+
+  auto load_k = [&](int kt, int buf) {
+    bool valid;
+    const int gtok = map_tok(kt * DM_KTILE + 16 * wid + l15, valid);
+    kvalid[buf] = valid;
+    const int blk = btab[gtok / BLOCK_SIZE];
+    const size_t row =
+        (((size_t)blk * Hk + hk) * BLOCK_SIZE + gtok % BLOCK_SIZE) * HEAD_DIM;
 #pragma unroll
-    for (int c = 0; c < HEAD_DIM / 32; ++c) {
-      const int d = stg_d0 + c * 8;
-      int4 kval = make_int4(0, 0, 0, 0);
-      int4 vval = make_int4(0, 0, 0, 0);
-      if (ok) {
-        if (KV_FP8) {
-          // 8 fp8 bytes -> 8 bf16 (scale folded in)
-          const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
-              (const uint8_t*)k_cache_v + row_off + d);
-          const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
-              (const uint8_t*)v_cache_v + row_off + d);
-          uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
-          uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
+    for (int s = 0; s < KC; ++s) {
+      if (KV_FP8) {
+        const uint64_t raw = *reinterpret_cast<const uint64_t*>(
+            (const uint8_t*)k_cache_v + row + s * 32 + l4 * 8);
+        uint16_t* e = reinterpret_cast<uint16_t*>(&kreg[buf][s]);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            ks[j] = f32_to_bf16_bits(
-                fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
-            vsp[j] = f32_to_bf16_bits(
-                fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
-          }
-        } else {
-          kval = *reinterpret_cast<const int4*>(
-              (const uint16_t*)k_cache_v + row_off + d);
-          vval = *reinterpret_cast<const int4*>(
-              (const uint16_t*)v_cache_v + row_off + d);
-        }
+        for (int j = 0; j < 8; ++j)
+          e[j] = f32_to_bf16_bits(
+              fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * k_scale);
+      } else {
+        kreg[buf][s] = *reinterpret_cast<const int4*>(
+            (const uint16_t*)k_cache_v + row + s * 32 + l4 * 8);
       }
-      kreg[c] = kval;
-      vreg[c] = vval;
     }
   };
 
-  load_tile(kt_begin);
+  auto load_v = [&](int kt, int buf) {
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      const int j0 = kt * DM_KTILE + s * 32 + l4 * 8;
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        const int drow = (wid * MT + mt) * 16 + l15;
+        if (SPARSE) {
+          // sparse tokens are scattered: per-element gather
+          alignas(16) uint16_t e[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            bool valid;
+            const int gt = map_tok(j0 + j, valid);
+            const int blk = btab[gt / BLOCK_SIZE];
+            const size_t addr = (((size_t)blk * Hk + hk) * HEAD_DIM + drow) *
+                                    BLOCK_SIZE + gt % BLOCK_SIZE;
+            if (KV_FP8)
+              e[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32(((const uint8_t*)v_cache_v)[addr]) * v_scale);
+            else
+              e[j] = ((const uint16_t*)v_cache_v)[addr];
+          }
+          vreg[buf][s][mt] = *reinterpret_cast<const int4*>(e);
+        } else {
+          bool valid;
+          const int gt = map_tok(j0, valid);  // 8 tokens stay in one block
+          const int blk = btab[gt / BLOCK_SIZE];
+          const size_t row = (((size_t)blk * Hk + hk) * HEAD_DIM + drow) *
+                                 BLOCK_SIZE + gt % BLOCK_SIZE;
+          if (KV_FP8) {
+            const uint64_t raw = *reinterpret_cast<const uint64_t*>(
+                (const uint8_t*)v_cache_v + row);
+            uint16_t* e = reinterpret_cast<uint16_t*>(&vreg[buf][s][mt]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              e[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * v_scale);
+          } else {
+            vreg[buf][s][mt] = *reinterpret_cast<const int4*>(
+                (const uint16_t*)v_cache_v + row);
+          }
+        }
+      }
+    }
+  };
+
+  load_k(kt_begin, 0);
+  load_v(kt_begin, 0);
 
   for (int kt = kt_begin; kt < kt_end; ++kt) {
     const int kbase = kt * DM_KTILE;
-    __syncthreads();  // previous tile's MFMA done reading LDS
-
-    // ---- write the prefetched tile: K row-major swz, V transposed ---------------
-#pragma unroll
-    for (int c = 0; c < HEAD_DIM / 32; ++c) {
-      const int d = stg_d0 + c * 8;
-      const int kb = swz(stg_tok * HEAD_DIM * 2 + d * 2, stg_tok);
-      *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kreg[c];
-      const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vreg[c]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int dd = d + j;
-        const int vb = swz(dd * DM_KTILE * 2 + stg_tok * 2, dd);
-        *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(VTl) + vb) = vs[j];
-      }
+    const int cur = (kt - kt_begin) & 1;
+    if (kt + 1 < kt_end) {
+      load_k(kt + 1, cur ^ 1);
+      load_v(kt + 1, cur ^ 1);
     }
-    __syncthreads();
-    if (kt + 1 < kt_end) load_tile(kt + 1);  // in flight during the math below
 
     // ---- S^T = K . Q^T (wave w: k rows [16w, 16w+16)) ---------------------------
     f32x4v acc_s = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < HEAD_DIM / 32; ++s) {
-      const int krow_i = 16 * wid + l15;
-      const int ka = swz(krow_i * HEAD_DIM * 2 + s * 64 + l4 * 16, krow_i);
-      const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<const char*>(Kl) + ka);
-      const int qb = swz(l15 * HEAD_DIM * 2 + s * 64 + l4 * 16, l15);
-      const bf16x8v bfrag = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<const char*>(Ql) + qb);
-      acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc_s, 0, 0, 0);
+    for (int s = 0; s < KC; ++s) {
+      const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(&kreg[cur][s]);
+      acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, qfrag[s], acc_s,
+                                                      0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
@@ -268,27 +281,29 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
 #pragma unroll
         for (int rr = 0; rr < 4; ++rr) acc_o[mt][rr] *= r;
     }
+    // barrier: l_s's wred read + everyone's resc read must complete before
+    // the next tile's wred/resc writes (threads race ahead through PV)
+    __syncthreads();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < DM_KTILE / 32; ++s) {
+    for (int s = 0; s < 2; ++s) {
       const int pb = swz(l15 * DM_KTILE * 2 + s * 64 + l4 * 16, l15);
       const bf16x8v bfrag = *reinterpret_cast<const bf16x8v*>(
           reinterpret_cast<const char*>(Pl) + pb);
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
-        const int drow = (wid * MT + mt) * 16 + l15;
-        const int va = swz(drow * DM_KTILE * 2 + s * 64 + l4 * 16, drow);
-        const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(
-            reinterpret_cast<const char*>(VTl) + va);
+        const bf16x8v afrag =
+            *reinterpret_cast<const bf16x8v*>(&vreg[cur][s][mt]);
         acc_o[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                             acc_o[mt], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
+    // no trailing barrier: the next tile's two reduction barriers precede any
+    // Pl/resc rewrite, so this tile's PV reads are already protected
   }
 
   // ---- write out (O^T acc: col=q head, row=d) -----------------------------------
-  __syncthreads();
   const int g = l15;
   if (g < G) {
     const size_t hq = (size_t)hk * G + g;

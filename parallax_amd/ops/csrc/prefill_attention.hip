@@ -93,48 +93,72 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
   const int kt_end = (kv_limit + KTILE - 1) / KTILE;
 
   // register double-buffered staging: next tile's loads fly during this
-  // tile's QK/softmax/PV (guide §6 G15 async-STAGE split)
+  // tile's QK/softmax/PV (guide §6 G15 async-STAGE split).
+  // K: per-token rows (natural layout). V: the cache is TRANSPOSED
+  // ([NB, Hk, D, BS]), so V^T rows stage as vectorized row-major copies —
+  // no per-element transpose scatter.
   const int stg_tok = tid & 63;
   const int stg_dv = tid >> 6;
   const int stg_d = stg_dv * (HEAD_DIM / 4);
-  int4 kreg[HEAD_DIM / 32], vreg[HEAD_DIM / 32];
+  int4 kreg[HEAD_DIM / 32];
+  // V^T: threads_per_row threads cover each d row's KTILE tokens
+  constexpr int VT_TPR = PF_THREADS / HEAD_DIM;        // 2 @128, 4 @64
+  constexpr int VT_TOK = KTILE / VT_TPR;               // tokens per thread
+  const int vt_d = tid / VT_TPR;                       // d row
+  const int vt_t0 = (tid % VT_TPR) * VT_TOK;           // first token
+  int4 vreg[VT_TOK / 8];
 
   auto load_tile = [&](int kt) {
-    const int gtok = kt * KTILE + stg_tok;
-    const bool ok = gtok < L;
-    size_t row_off = 0;
-    if (ok) {
-      const int blk = btab[gtok / BS];
-      const int off = gtok % BS;
-      row_off = (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
+    {
+      const int gtok = kt * KTILE + stg_tok;
+      const bool ok = gtok < L;
+      size_t row_off = 0;
+      if (ok) {
+        const int blk = btab[gtok / BS];
+        const int off = gtok % BS;
+        row_off = (((size_t)blk * Hk + hk) * BS + off) * HEAD_DIM;
+      }
+#pragma unroll
+      for (int c = 0; c < HEAD_DIM / 32; ++c) {  // 8 elems per step
+        int4 kval = make_int4(0, 0, 0, 0);
+        if (ok) {
+          if (KV_FP8) {
+            const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
+                (const uint8_t*)k_cache_v + row_off + stg_d + c * 8);
+            uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              ks[j] = f32_to_bf16_bits(
+                  fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+          } else {
+            kval = *reinterpret_cast<const int4*>(
+                (const uint16_t*)k_cache_v + row_off + stg_d + c * 8);
+          }
+        }
+        kreg[c] = kval;
+      }
     }
 #pragma unroll
-    for (int c = 0; c < HEAD_DIM / 32; ++c) {  // 8 elems per step
-      int4 kval = make_int4(0, 0, 0, 0);
+    for (int c = 0; c < VT_TOK / 8; ++c) {  // 8 tokens per step (one block)
+      const int gtok = kt * KTILE + vt_t0 + c * 8;
       int4 vval = make_int4(0, 0, 0, 0);
-      if (ok) {
+      if (gtok < L) {
+        const int blk = btab[gtok / BS];
+        const size_t row = (((size_t)blk * Hk + hk) * HEAD_DIM + vt_d) * BS +
+                           gtok % BS;
         if (KV_FP8) {
-          const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
-              (const uint8_t*)k_cache_v + row_off + stg_d + c * 8);
           const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
-              (const uint8_t*)v_cache_v + row_off + stg_d + c * 8);
-          uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
+              (const uint8_t*)v_cache_v + row);
           uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            ks[j] = f32_to_bf16_bits(
-                fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+          for (int j = 0; j < 8; ++j)
             vsp[j] = f32_to_bf16_bits(
                 fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
-          }
         } else {
-          kval = *reinterpret_cast<const int4*>(
-              (const uint16_t*)k_cache_v + row_off + stg_d + c * 8);
           vval = *reinterpret_cast<const int4*>(
-              (const uint16_t*)v_cache_v + row_off + stg_d + c * 8);
+              (const uint16_t*)v_cache_v + row);
         }
       }
-      kreg[c] = kval;
       vreg[c] = vval;
     }
   };
@@ -145,19 +169,17 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
     const int kbase = kt * KTILE;
     __syncthreads();  // previous PV finished reading VTl/Pl
 
-    // ---- write the prefetched tile: K row-major swz, V transposed --------------
+    // ---- write the prefetched tile: K token rows, V^T d rows (both swz) --------
 #pragma unroll
     for (int c = 0; c < HEAD_DIM / 32; ++c) {
       const int d = stg_d + c * 8;
       const int kb = swz(stg_tok * HEAD_DIM * 2 + d * 2, stg_tok);
       *reinterpret_cast<int4*>(reinterpret_cast<char*>(Kl) + kb) = kreg[c];
-      const uint16_t* vs = reinterpret_cast<const uint16_t*>(&vreg[c]);
+    }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int dd = d + j;
-        const int vb = swz(dd * KTILE * 2 + stg_tok * 2, dd);
-        *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(VTl) + vb) = vs[j];
-      }
+    for (int c = 0; c < VT_TOK / 8; ++c) {
+      const int vb = swz(vt_d * KTILE * 2 + (vt_t0 + c * 8) * 2, vt_d);
+      *reinterpret_cast<int4*>(reinterpret_cast<char*>(VTl) + vb) = vreg[c];
     }
     __syncthreads();
     if (kt + 1 < kt_end) load_tile(kt + 1);

@@ -99,7 +99,7 @@ def reshape_and_cache(
     k: torch.Tensor,             # [T, Hk, D]
     v: torch.Tensor,             # [T, Hk, D]
     k_cache: torch.Tensor,       # [num_blocks, Hk, block_size, D]
-    v_cache: torch.Tensor,
+    v_cache: torch.Tensor,       # [num_blocks, Hk, D, block_size] (transposed)
     slot_mapping: torch.Tensor,  # [T] int (block*bs + off); -1 = skip (padding)
 ) -> None:
     block_size = k_cache.shape[2]
@@ -108,7 +108,7 @@ def reshape_and_cache(
     blk = slots // block_size
     off = slots % block_size
     k_cache[blk, :, off] = k[valid].to(k_cache.dtype)
-    v_cache[blk, :, off] = v[valid].to(v_cache.dtype)
+    v_cache[blk, :, :, off] = v[valid].to(v_cache.dtype)
 
 
 def mla_reshape_and_cache(
@@ -135,12 +135,12 @@ def _gather_kv(
     block_table: torch.Tensor,  # [max_blocks] int
     seq_len: int,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Gather one request's KV as [seq_len, Hk, D]."""
+    """Gather one request's KV as [seq_len, Hk, D] (V cache is transposed)."""
     block_size = k_cache.shape[2]
     nb = (seq_len + block_size - 1) // block_size
     blocks = block_table[:nb].long()
     k = k_cache[blocks].transpose(1, 2).reshape(nb * block_size, *k_cache.shape[1:2], k_cache.shape[3])
-    v = v_cache[blocks].transpose(1, 2).reshape(nb * block_size, *v_cache.shape[1:2], v_cache.shape[3])
+    v = v_cache[blocks].permute(0, 3, 1, 2).reshape(nb * block_size, *v_cache.shape[1:2], v_cache.shape[2])
     return k[:seq_len], v[:seq_len]
 
 

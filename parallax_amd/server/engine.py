@@ -269,6 +269,30 @@ class Engine:
     def abort(self, rid: str) -> None:
         self._pending_aborts.append(rid)
 
+    def warmup_gemms(self, ms: List[int]) -> None:
+        """Pre-tune the hipBLASLt algo picks for every 2-D weight shape at
+        the given M values. Per-shape tuning is one-time but otherwise lands
+        inside the first prefill and pollutes TTFT; a long-lived server calls
+        this at startup (the bench does too, before the timed phases)."""
+        if self.device.type != "cuda":
+            return
+        seen = set()
+        with torch.inference_mode():
+            for m in sorted({int(x) for x in ms if x and x > 0}):
+                for name, p in self.model.named_parameters():
+                    if p.dim() != 2 or "embed_tokens" in name \
+                            or p.dtype != self.args.dtype:
+                        continue
+                    key = (m, p.shape[0], p.shape[1])
+                    if key in seen:
+                        continue
+                    seen.add(key)
+                    x = torch.zeros(m, p.shape[1], dtype=p.dtype,
+                                    device=self.device)
+                    ops.linear(x, p)
+        torch.cuda.synchronize()
+        logger.info("gemm warmup: tuned %d shapes", len(seen))
+
     def set_grammar_vocab(self, vocab: List[str]) -> None:
         """Enable json_schema constrained decoding: vocab[i] is token i's
         text. Must be called on every rank that samples (in practice: all)."""

@@ -49,13 +49,18 @@ class PagedKVCache:
         self.spec = spec
         self.num_blocks = num_blocks
         self.device = device
-        shape = (num_blocks, spec.num_kv_heads, spec.block_size, spec.head_dim)
+        # K rows are tokens (QK^T fragments read 8 consecutive d per token);
+        # V is stored TRANSPOSED [.., D, block] so PV fragments read 8
+        # consecutive tokens per d straight from HBM — the decode/prefill
+        # kernels consume V^T without any LDS transpose staging
+        k_shape = (num_blocks, spec.num_kv_heads, spec.block_size, spec.head_dim)
+        v_shape = (num_blocks, spec.num_kv_heads, spec.head_dim, spec.block_size)
         self.k_caches: List[torch.Tensor] = [
-            torch.zeros(shape, dtype=spec.dtype, device=device)
+            torch.zeros(k_shape, dtype=spec.dtype, device=device)
             for _ in range(spec.num_layers)
         ]
         self.v_caches: List[torch.Tensor] = [
-            torch.zeros(shape, dtype=spec.dtype, device=device)
+            torch.zeros(v_shape, dtype=spec.dtype, device=device)
             for _ in range(spec.num_layers)
         ]
         idx_dtype = spec.dtype if spec.dtype != torch.float8_e4m3fn \

@@ -21,7 +21,7 @@ def run_case(B, Hk, G, D, ctx, max_seq_len, BS=32, iters=50):
     max_blocks = (max(max_seq_len, ctx) + BS - 1) // BS
     nb = B * ((ctx + BS - 1) // BS) + 1
     kc = torch.randn(nb, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
-    vc = torch.randn(nb, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(nb, Hk, D, BS, dtype=torch.bfloat16, device="cuda")
     bt = torch.zeros(B, max_blocks, dtype=torch.int32, device="cuda")
     nblk = (ctx + BS - 1) // BS
     bt[:, :nblk] = (

@@ -71,12 +71,12 @@ def test_reshape_and_cache():
     k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
     v = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
     kc = torch.zeros(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
-    vc = torch.zeros_like(kc)
+    vc = torch.zeros(NB, Hk, D, BS, dtype=torch.bfloat16, device="cuda")
     slots = torch.randperm(NB * BS, device="cuda")[:T].to(torch.int64)
     slots[5] = -1  # padding skip
     ops.reshape_and_cache(k, v, kc, vc, slots)
     kc_ref = torch.zeros(NB, Hk, BS, D).float()
-    vc_ref = torch.zeros_like(kc_ref)
+    vc_ref = torch.zeros(NB, Hk, D, BS).float()
     ref.reshape_and_cache(k.float().cpu(), v.float().cpu(), kc_ref, vc_ref, slots.cpu())
     _assert_close(kc, kc_ref, atol=1e-2)
     _assert_close(vc, vc_ref, atol=1e-2)
@@ -96,7 +96,7 @@ def _make_paged_kv(B, Hk, D, BS, max_ctx, seed=0):
     max_blocks = (max_ctx + BS - 1) // BS
     NB = B * max_blocks + 1
     kc = torch.randn(NB, Hk, BS, D, generator=g, dtype=torch.bfloat16, device="cuda")
-    vc = torch.randn(NB, Hk, BS, D, generator=g, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(NB, Hk, D, BS, generator=g, dtype=torch.bfloat16, device="cuda")
     bt = torch.arange(B * max_blocks, dtype=torch.int32, device="cuda").reshape(B, max_blocks) + 1
     return kc, vc, bt
 
@@ -212,7 +212,7 @@ def test_rope_and_cache_fused():
     k = qkv[:, Hq * D : (Hq + Hk) * D].view(T, Hk, D)
     v = qkv[:, (Hq + Hk) * D :].view(T, Hk, D)
     kc = torch.zeros(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
-    vc = torch.zeros_like(kc)
+    vc = torch.zeros(NB, Hk, D, BS, dtype=torch.bfloat16, device="cuda")
     pos = torch.randint(0, 500, (T,), dtype=torch.int32, device="cuda")
     cs = ops.build_rope_cache(512, D, 10000.0).cuda()
     slots = torch.randperm(NB * BS, device="cuda")[:T].to(torch.int64)
@@ -221,7 +221,7 @@ def test_rope_and_cache_fused():
     q_ref = q.float().cpu().clone()
     k_ref = k.float().cpu().clone()
     kc_ref = torch.zeros(NB, Hk, BS, D).float()
-    vc_ref = torch.zeros_like(kc_ref)
+    vc_ref = torch.zeros(NB, Hk, D, BS).float()
     ref.rope_inplace(q_ref, k_ref, pos.cpu(), cs.cpu(), True)
     ref.reshape_and_cache(k_ref, v.float().cpu(), kc_ref, vc_ref, slots.cpu())
 
@@ -261,7 +261,7 @@ def test_rope_and_cache_fp8():
     k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
     v = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
     kc = torch.zeros(NB, Hk, BS, D, dtype=torch.float8_e4m3fn, device="cuda")
-    vc = torch.zeros_like(kc)
+    vc = torch.zeros(NB, Hk, D, BS, dtype=torch.float8_e4m3fn, device="cuda")
     pos = torch.randint(0, 400, (T,), dtype=torch.int32, device="cuda")
     cs = ops.build_rope_cache(512, D, 10000.0).cuda()
     slots = torch.randperm(NB * BS, device="cuda")[:T].to(torch.int64)
@@ -273,7 +273,7 @@ def test_rope_and_cache_fp8():
     # cache holds fp8-quantized roped K / raw V
     blk, off = slots.cpu() // BS, slots.cpu() % BS
     got_k = kc.cpu().float()[blk, :, off]
-    got_v = vc.cpu().float()[blk, :, off]
+    got_v = vc.cpu().float()[blk, :, :, off]
     torch.testing.assert_close(got_k, _fp8_roundtrip_cpu(k_ref), atol=8e-2, rtol=8e-2)
     torch.testing.assert_close(got_v, _fp8_roundtrip_cpu(v.float().cpu()),
                                atol=8e-2, rtol=8e-2)

@@ -18,7 +18,7 @@ def _setup(query_lens, seq_lens, Hq, Hk, D, BS=32, seed=0):
     max_blocks = (max(seq_lens) + BS - 1) // BS
     NB = B * max_blocks + 1
     kc = torch.randn(NB, Hk, BS, D, generator=g, dtype=torch.bfloat16, device="cuda")
-    vc = torch.randn(NB, Hk, BS, D, generator=g, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(NB, Hk, D, BS, generator=g, dtype=torch.bfloat16, device="cuda")
     bt = (
         torch.arange(B * max_blocks, dtype=torch.int32, device="cuda").reshape(B, max_blocks)
         + 1

@@ -67,7 +67,7 @@ def test_msa_pipeline_cpu():
     max_blocks = (max(ctxs) + BS - 1) // BS
     NB = B * max_blocks + 1
     kc = torch.randn(NB, Hk, BS, D)
-    vc = torch.randn(NB, Hk, BS, D)
+    vc = torch.randn(NB, Hk, D, BS)
     bt = (torch.arange(B * max_blocks, dtype=torch.int32).reshape(B, max_blocks) + 1)
     q = torch.randn(B, Hq, D) * 0.2
     sl = torch.tensor(ctxs, dtype=torch.int32)
@@ -141,7 +141,7 @@ def test_msa_kernel_gpu():
     max_blocks = (max(ctxs) + BS - 1) // BS
     NB = B * max_blocks + 1
     kc = torch.randn(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
-    vc = torch.randn(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(NB, Hk, D, BS, dtype=torch.bfloat16, device="cuda")
     bt = (torch.arange(B * max_blocks, dtype=torch.int32, device="cuda")
           .reshape(B, max_blocks) + 1)
     q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
@@ -167,7 +167,7 @@ def test_msa_kernel_gpu_per_head_positions():
     max_blocks = (max(ctxs) + BS - 1) // BS
     NB = B * max_blocks + 1
     kc = torch.randn(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
-    vc = torch.randn(NB, Hk, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(NB, Hk, D, BS, dtype=torch.bfloat16, device="cuda")
     bt = (torch.arange(B * max_blocks, dtype=torch.int32, device="cuda")
           .reshape(B, max_blocks) + 1)
     q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
