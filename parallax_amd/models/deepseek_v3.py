@@ -125,10 +125,12 @@ class MLAAttention(nn.Module):
         return None
 
     def _prefill_attention(self, q_nope, q_pe, meta: ForwardMeta, cache):
-        """Non-absorbed prefill: gather latents for each request's context from
-        the paged cache, expand through kv_b, dense causal attention.
-        (torch/rocBLAS compute; a fused MLA prefill kernel is a later
-        optimization — prefill is compute-bound and batched GEMMs run on MFMA.)"""
+        """Absorbed, L-chunked online-softmax prefill over the COMPRESSED
+        cache: q_nope folds through w_uk so QK^T and PV both contract against
+        the latent directly (same math as the absorbed decode kernel) —
+        no [L, H, dn+dv] expansion and no [QL, L] full-logits buffer, which
+        is what makes 256k-context prefill (BASELINE Kimi-K2 config) fit.
+        Sparse subclasses (DSA) still take the dense masked path below."""
         H, dn, dv, r = self.num_heads, self.dn, self.dv, self.r
         bs = cache.shape[1]
         out = torch.empty(
@@ -141,25 +143,56 @@ class MLAAttention(nn.Module):
             nb = (L + bs - 1) // bs
             entries = cache[meta.block_tables[i, :nb].long()].reshape(nb * bs, -1)[:L]
             lat, kpe_ctx = entries[:, :r], entries[:, r:]        # [L, r], [L, dr]
-            kv = torch.einsum("lr,hdr->lhd", lat.float(),
-                              torch.cat([self.w_uk, self.w_uv], dim=1).float())
-            k_nope_ctx, v_ctx = kv[:, :, :dn], kv[:, :, dn:]
             qi_n = q_nope[t0 : t0 + QL].float()                  # [QL, H, dn]
             qi_p = q_pe[t0 : t0 + QL].float()                    # [QL, H, dr]
-            logits = (
-                torch.einsum("qhd,lhd->hql", qi_n, k_nope_ctx)
-                + torch.einsum("qhd,ld->hql", qi_p, kpe_ctx.float())
-            ) * self.scale
-            qpos = torch.arange(L - QL, L, device=logits.device).unsqueeze(-1)
-            kpos = torch.arange(L, device=logits.device).unsqueeze(0)
-            logits.masked_fill_((kpos > qpos).unsqueeze(0), float("-inf"))
             extra = self._prefill_sparse_mask(i, t0, QL, L, meta)
             if extra is not None:  # DSA (deepseek_v32) top-k restriction
+                kv = torch.einsum("lr,hdr->lhd", lat.float(),
+                                  torch.cat([self.w_uk, self.w_uv], dim=1).float())
+                k_nope_ctx, v_ctx = kv[:, :, :dn], kv[:, :, dn:]
+                logits = (
+                    torch.einsum("qhd,lhd->hql", qi_n, k_nope_ctx)
+                    + torch.einsum("qhd,ld->hql", qi_p, kpe_ctx.float())
+                ) * self.scale
+                qpos = torch.arange(L - QL, L, device=logits.device).unsqueeze(-1)
+                kpos = torch.arange(L, device=logits.device).unsqueeze(0)
+                logits.masked_fill_((kpos > qpos).unsqueeze(0), float("-inf"))
                 logits = logits + extra.unsqueeze(0)
-            p = logits.softmax(dim=-1)
-            out[t0 : t0 + QL] = (
-                torch.einsum("hql,lhv->qhv", p, v_ctx).to(out.dtype)
-            )
+                p = logits.softmax(dim=-1)
+                out[t0 : t0 + QL] = (
+                    torch.einsum("hql,lhv->qhv", p, v_ctx).to(out.dtype)
+                )
+                t0 += QL
+                continue
+
+            q_lat = torch.einsum("qhd,hdr->qhr", qi_n, self.w_uk.float())
+            dev = q_lat.device
+            qpos = torch.arange(L - QL, L, device=dev).view(1, -1, 1)
+            # chunk so the [H, QL, Lc] score buffer stays ~<=2 GB
+            lc_size = max(1024, (1 << 29) // max(1, H * QL))
+            m = torch.full((H, QL, 1), float("-inf"), device=dev)
+            lsum = torch.zeros(H, QL, 1, device=dev)
+            acc = torch.zeros(QL, H, r, device=dev)
+            for c0 in range(0, L, lc_size):
+                c1 = min(c0 + lc_size, L)
+                latc = lat[c0:c1].float()
+                s = (
+                    torch.einsum("qhr,lr->hql", q_lat, latc)
+                    + torch.einsum("qhd,ld->hql", qi_p, kpe_ctx[c0:c1].float())
+                ) * self.scale
+                kpos = torch.arange(c0, c1, device=dev).view(1, 1, -1)
+                s.masked_fill_(kpos > qpos, float("-inf"))
+                m_new = torch.maximum(m, s.amax(dim=-1, keepdim=True))
+                p = torch.exp(s - m_new)
+                resc = torch.exp(m - m_new)
+                lsum = lsum * resc + p.sum(dim=-1, keepdim=True)
+                acc = acc * resc.squeeze(-1).permute(1, 0).unsqueeze(-1) \
+                    + torch.einsum("hql,lr->qhr", p, latc)
+                m = m_new
+            out_lat = acc / lsum.squeeze(-1).permute(1, 0).unsqueeze(-1)
+            out[t0 : t0 + QL] = torch.einsum(
+                "qhr,hvr->qhv", out_lat, self.w_uv.float()
+            ).to(out.dtype)
             t0 += QL
         return out
 

@@ -125,14 +125,15 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   // ---- register double-buffered tile loads ------------------------------------
   // K: lane covers token (16*wid + l15), all KC d-chunks.
   // V^T: lane covers d rows (wid*MT+mt)*16 + l15, token chunk s*32 + l4*8.
-  int4 kreg[2][KC];
-  int4 vreg[2][2][MT];
-  bool kvalid[2];
+  // Two NAMED buffer sets with a compile-time parity branch: runtime-indexed
+  // register arrays land in scratch memory (guide common-mistake #20 — the
+  // first cut of this kernel measured 272 B/lane of scratch and ran 2x slow).
+  int4 kregA[KC], kregB[KC];
+  int4 vregA[2][MT], vregB[2][MT];
 
-  auto load_k = [&](int kt, int buf) {
+  auto load_k = [&](int kt, int4 (&kreg)[KC]) {
     bool valid;
     const int gtok = map_tok(kt * DM_KTILE + 16 * wid + l15, valid);
-    kvalid[buf] = valid;
     const int blk = btab[gtok / BLOCK_SIZE];
     const size_t row =
         (((size_t)blk * Hk + hk) * BLOCK_SIZE + gtok % BLOCK_SIZE) * HEAD_DIM;
@@ -141,19 +142,19 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
       if (KV_FP8) {
         const uint64_t raw = *reinterpret_cast<const uint64_t*>(
             (const uint8_t*)k_cache_v + row + s * 32 + l4 * 8);
-        uint16_t* e = reinterpret_cast<uint16_t*>(&kreg[buf][s]);
+        uint16_t* e = reinterpret_cast<uint16_t*>(&kreg[s]);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           e[j] = f32_to_bf16_bits(
               fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * k_scale);
       } else {
-        kreg[buf][s] = *reinterpret_cast<const int4*>(
+        kreg[s] = *reinterpret_cast<const int4*>(
             (const uint16_t*)k_cache_v + row + s * 32 + l4 * 8);
       }
     }
   };
 
-  auto load_v = [&](int kt, int buf) {
+  auto load_v = [&](int kt, int4 (&vreg)[2][MT]) {
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
       const int j0 = kt * DM_KTILE + s * 32 + l4 * 8;
@@ -176,7 +177,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
             else
               e[j] = ((const uint16_t*)v_cache_v)[addr];
           }
-          vreg[buf][s][mt] = *reinterpret_cast<const int4*>(e);
+          vreg[s][mt] = *reinterpret_cast<const int4*>(e);
         } else {
           bool valid;
           const int gt = map_tok(j0, valid);  // 8 tokens stay in one block
@@ -186,13 +187,13 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
           if (KV_FP8) {
             const uint64_t raw = *reinterpret_cast<const uint64_t*>(
                 (const uint8_t*)v_cache_v + row);
-            uint16_t* e = reinterpret_cast<uint16_t*>(&vreg[buf][s][mt]);
+            uint16_t* e = reinterpret_cast<uint16_t*>(&vreg[s][mt]);
 #pragma unroll
             for (int j = 0; j < 8; ++j)
               e[j] = f32_to_bf16_bits(
                   fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * v_scale);
           } else {
-            vreg[buf][s][mt] = *reinterpret_cast<const int4*>(
+            vreg[s][mt] = *reinterpret_cast<const int4*>(
                 (const uint16_t*)v_cache_v + row);
           }
         }
@@ -200,23 +201,16 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     }
   };
 
-  load_k(kt_begin, 0);
-  load_v(kt_begin, 0);
-
-  for (int kt = kt_begin; kt < kt_end; ++kt) {
+  auto tile_math = [&](int kt, const int4 (&kreg)[KC],
+                       const int4 (&vreg)[2][MT]) {
     const int kbase = kt * DM_KTILE;
-    const int cur = (kt - kt_begin) & 1;
-    if (kt + 1 < kt_end) {
-      load_k(kt + 1, cur ^ 1);
-      load_v(kt + 1, cur ^ 1);
-    }
 
     // ---- S^T = K . Q^T (wave w: k rows [16w, 16w+16)) ---------------------------
     f32x4v acc_s = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < KC; ++s) {
-      const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(&kreg[cur][s]);
+      const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(&kreg[s]);
       acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, qfrag[s], acc_s,
                                                       0, 0, 0);
     }
@@ -293,7 +287,7 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
         const bf16x8v afrag =
-            *reinterpret_cast<const bf16x8v*>(&vreg[cur][s][mt]);
+            *reinterpret_cast<const bf16x8v*>(&vreg[s][mt]);
         acc_o[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                             acc_o[mt], 0, 0, 0);
       }
@@ -301,6 +295,24 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     __builtin_amdgcn_s_setprio(0);
     // no trailing barrier: the next tile's two reduction barriers precede any
     // Pl/resc rewrite, so this tile's PV reads are already protected
+  };
+
+  load_k(kt_begin, kregA);
+  load_v(kt_begin, vregA);
+  for (int kt = kt_begin; kt < kt_end; ++kt) {
+    if (((kt - kt_begin) & 1) == 0) {
+      if (kt + 1 < kt_end) {
+        load_k(kt + 1, kregB);
+        load_v(kt + 1, vregB);
+      }
+      tile_math(kt, kregA, vregA);
+    } else {
+      if (kt + 1 < kt_end) {
+        load_k(kt + 1, kregA);
+        load_v(kt + 1, vregA);
+      }
+      tile_math(kt, kregB, vregB);
+    }
   }
 
   // ---- write out (O^T acc: col=q head, row=d) -----------------------------------
