@@ -42,21 +42,70 @@ def build_random_prompts(num: int, input_len: int, vocab: int = 30000,
     ]
 
 
+def build_dataset(name: str, num: int, input_len: int, output_len: int,
+                  dataset_path: Optional[str], seed: int = 0):
+    """Returns (prompts, output_lens). Dataset modes (reference
+    benchmark_serving.py:122-144 dataset breadth):
+    - random: fixed-length random-token prompts
+    - sharegpt: a LOCAL ShareGPT-format JSON (offline env: no downloader);
+      human turns become prompts, the following reply's length caps output
+    - synthetic-sharegpt: offline lognormal length mix approximating the
+      ShareGPT distribution (no file needed)"""
+    rng = random.Random(seed)
+    if name == "random":
+        return build_random_prompts(num, input_len, seed=seed), [output_len] * num
+    if name == "sharegpt":
+        if not dataset_path:
+            raise SystemExit("--dataset-path required for --dataset-name sharegpt")
+        with open(dataset_path) as f:
+            data = json.load(f)
+        pairs = []
+        for conv in data:
+            turns = conv.get("conversations") or conv.get("items") or []
+            for i in range(len(turns) - 1):
+                if turns[i].get("from") in ("human", "user"):
+                    prompt = turns[i].get("value", "")
+                    reply = turns[i + 1].get("value", "")
+                    if prompt and reply:
+                        pairs.append((prompt, max(1, len(reply.split()))))
+        if not pairs:
+            raise SystemExit(f"no usable turns in {dataset_path}")
+        rng.shuffle(pairs)
+        pairs = [pairs[i % len(pairs)] for i in range(num)]
+        return [p for p, _ in pairs], [o for _, o in pairs]
+    if name == "synthetic-sharegpt":
+        prompts, outs = [], []
+        for _ in range(num):
+            ilen = max(4, min(4096, int(rng.lognormvariate(4.9, 1.0))))
+            olen = max(4, min(1024, int(rng.lognormvariate(4.8, 0.9))))
+            prompts.append(" ".join(str(rng.randrange(30000))
+                                    for _ in range(ilen)))
+            outs.append(olen)
+        return prompts, outs
+    raise SystemExit(f"unknown dataset {name}")
+
+
 async def one_request(client: httpx.AsyncClient, base_url: str, prompt: str,
-                      output_len: int) -> RequestResult:
+                      output_len: int, backend: str = "chat") -> RequestResult:
+    """backend: "chat" (/v1/chat/completions) or "completions"
+    (/v1/completions) — the reference ships per-backend request functions
+    (backend_request_func.py); both endpoints stream SSE here."""
     res = RequestResult()
     t0 = time.perf_counter()
     last_t = t0
-    try:
-        async with client.stream(
-            "POST", f"{base_url}/v1/chat/completions",
-            json={
-                "model": "bench", "stream": True,
+    if backend == "completions":
+        url = f"{base_url}/v1/completions"
+        body = {"model": "bench", "stream": True, "prompt": prompt,
+                "max_tokens": output_len, "temperature": 1.0,
+                "ignore_eos": True}
+    else:
+        url = f"{base_url}/v1/chat/completions"
+        body = {"model": "bench", "stream": True,
                 "messages": [{"role": "user", "content": prompt}],
                 "max_tokens": output_len, "temperature": 1.0,
-                "ignore_eos": True,
-            },
-        ) as r:
+                "ignore_eos": True}
+    try:
+        async with client.stream("POST", url, json=body) as r:
             if r.status_code != 200:
                 res.error = f"http {r.status_code}"
                 return res
@@ -85,16 +134,21 @@ async def one_request(client: httpx.AsyncClient, base_url: str, prompt: str,
 
 
 async def run_benchmark(base_url: str, num_prompts: int, request_rate: float,
-                        input_len: int, output_len: int, seed: int = 0) -> dict:
-    prompts = build_random_prompts(num_prompts, input_len, seed=seed)
+                        input_len: int, output_len: int, seed: int = 0,
+                        dataset: str = "random",
+                        dataset_path: Optional[str] = None,
+                        backend: str = "chat") -> dict:
+    prompts, out_lens = build_dataset(
+        dataset, num_prompts, input_len, output_len, dataset_path, seed
+    )
     rng = random.Random(seed)
     results: List[RequestResult] = []
     t_start = time.perf_counter()
     async with httpx.AsyncClient(timeout=600.0) as client:
         tasks = []
-        for prompt in prompts:
+        for prompt, olen in zip(prompts, out_lens):
             tasks.append(asyncio.create_task(
-                one_request(client, base_url, prompt, output_len)))
+                one_request(client, base_url, prompt, olen, backend)))
             if request_rate != float("inf"):
                 # Poisson arrivals at the requested rate (reference behavior)
                 await asyncio.sleep(rng.expovariate(request_rate))
@@ -145,10 +199,18 @@ def main():
     ap.add_argument("--input-len", type=int, default=512)
     ap.add_argument("--output-len", type=int, default=128)
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--dataset-name", default="random",
+                    choices=["random", "sharegpt", "synthetic-sharegpt"])
+    ap.add_argument("--dataset-path", default=None,
+                    help="local ShareGPT-format JSON (offline env)")
+    ap.add_argument("--backend", default="chat",
+                    choices=["chat", "completions"])
     args = ap.parse_args()
     result = asyncio.run(run_benchmark(
         args.base_url, args.num_prompts, args.request_rate,
         args.input_len, args.output_len, args.seed,
+        dataset=args.dataset_name, dataset_path=args.dataset_path,
+        backend=args.backend,
     ))
     print(json.dumps(result, indent=2))
 

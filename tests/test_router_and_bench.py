@@ -77,3 +77,25 @@ def test_benchmark_harness_end_to_end():
     assert result["itl_ms"] is not None
     srv.should_exit = True
     server.stop()
+
+
+def test_benchmark_dataset_modes(tmp_path):
+    """Dataset breadth (reference benchmark_serving.py dataset modes):
+    random / local-sharegpt / synthetic-sharegpt builders."""
+    import json as _json
+
+    from parallax_amd.benchmark.benchmark_serving import build_dataset
+
+    p, o = build_dataset("random", 4, 8, 16, None)
+    assert len(p) == 4 and o == [16] * 4
+    p, o = build_dataset("synthetic-sharegpt", 16, 8, 16, None, seed=2)
+    assert len(p) == 16 and min(o) >= 4
+    path = tmp_path / "sg.json"
+    path.write_text(_json.dumps([{"conversations": [
+        {"from": "human", "value": "q1 words"},
+        {"from": "gpt", "value": "a1 has four words"},
+        {"from": "human", "value": "q2"},
+        {"from": "gpt", "value": "a2 reply"},
+    ]}]))
+    p, o = build_dataset("sharegpt", 4, 8, 16, str(path))
+    assert len(p) == 4 and set(p) <= {"q1 words", "q2"}
