@@ -85,17 +85,13 @@ def cmd_serve(args) -> None:
         from bench import MODELS  # synthetic configs
 
         cfg = ModelConfig.from_hf_config(MODELS[args.model][0]())
-    eargs = EngineArgs(
-        block_size=args.block_size,
-        max_batch_size=args.max_batch_size,
-        max_num_tokens_per_batch=args.max_num_tokens_per_batch,
-        micro_batches=world,
-        dtype=torch.bfloat16 if torch.cuda.is_available() else torch.float32,
-        num_kv_blocks=None if torch.cuda.is_available() else 4096,
-        max_model_len=args.max_model_len,
-    )
+    eargs = engine_args_from_cli(args, world=world)
     engine = Engine(cfg, eargs, comm=comm, model_path=args.model_path,
-                    random_weights=args.model_path is None)
+                    random_weights=args.model_path is None,
+                    lora_path=getattr(args, "lora_path", None))
+    if torch.cuda.is_available():
+        engine.warmup_gemms([args.max_num_tokens_per_batch,
+                             args.max_batch_size])
     if comm.rank == 0:
         server = EngineServer(engine)
         server.start()
@@ -128,6 +124,80 @@ def cmd_chat(args) -> None:
         print(f"assistant> {msg['content']}")
 
 
+def add_engine_args(p: argparse.ArgumentParser) -> None:
+    """Worker/engine flag surface (reference server_args.py breadth): every
+    EngineArgs knob is reachable from the CLI."""
+    p.add_argument("--block-size", type=int, default=32,
+                   help="KV page size in tokens")
+    p.add_argument("--max-batch-size", type=int, default=128)
+    p.add_argument("--max-num-tokens-per-batch", type=int, default=16384)
+    p.add_argument("--prefill-chunk-size", type=int, default=8192)
+    p.add_argument("--max-model-len", type=int, default=8192,
+                   help="context ceiling (sizes graph buffers)")
+    p.add_argument("--cache-memory-fraction", type=float, default=0.80,
+                   help="fraction of free HBM for the KV cache")
+    p.add_argument("--num-kv-blocks", type=int, default=None,
+                   help="explicit KV block count (overrides the fraction)")
+    p.add_argument("--kv-cache-dtype", default="auto",
+                   choices=["auto", "fp8"], help="fp8 = e4m3 KV storage")
+    p.add_argument("--moe-weight-dtype", default="auto",
+                   choices=["auto", "fp8"], help="fp8 = W8A8 expert weights")
+    p.add_argument("--dtype", default="auto",
+                   choices=["auto", "bfloat16", "float16", "float32"])
+    p.add_argument("--micro-batches", type=int, default=0,
+                   help="decode micro-batches in flight (0 = one per stage)")
+    p.add_argument("--disable-prefix-cache", action="store_true",
+                   help="turn off the block-radix prefix cache")
+    p.add_argument("--disable-graphs", action="store_true",
+                   help="turn off hipGraph-captured decode")
+    p.add_argument("--request-timeout", type=float, default=600.0,
+                   help="per-request abort timeout (s)")
+    p.add_argument("--seed", type=int, default=0, help="sampling seed")
+    p.add_argument("--start-layer", type=int, default=None,
+                   help="explicit layer range start (decentralized mode)")
+    p.add_argument("--end-layer", type=int, default=None)
+    p.add_argument("--lora-path", default=None,
+                   help="LoRA adapter directory fused at load time")
+    p.add_argument("--log-level", default=None,
+                   help="logging level (DEBUG/INFO/WARNING)")
+
+
+def engine_args_from_cli(args, world: int = 1):
+    import torch
+
+    from .server.engine import EngineArgs
+
+    if args.log_level:
+        import logging
+
+        logging.getLogger("parallax_amd").setLevel(args.log_level.upper())
+    use_gpu = torch.cuda.is_available()
+    if args.dtype == "auto":
+        dtype = torch.bfloat16 if use_gpu else torch.float32
+    else:
+        dtype = getattr(torch, args.dtype)
+    return EngineArgs(
+        block_size=args.block_size,
+        max_batch_size=args.max_batch_size,
+        max_num_tokens_per_batch=args.max_num_tokens_per_batch,
+        prefill_chunk_size=args.prefill_chunk_size,
+        cache_memory_fraction=args.cache_memory_fraction,
+        num_kv_blocks=args.num_kv_blocks if args.num_kv_blocks
+        else (None if use_gpu else 4096),
+        micro_batches=args.micro_batches or world,
+        enable_prefix_cache=not args.disable_prefix_cache,
+        dtype=dtype,
+        seed=args.seed,
+        request_timeout_s=args.request_timeout,
+        start_layer=args.start_layer,
+        end_layer=args.end_layer,
+        max_model_len=args.max_model_len,
+        enable_graphs=not args.disable_graphs,
+        kv_cache_dtype=args.kv_cache_dtype,
+        moe_weight_dtype=args.moe_weight_dtype,
+    )
+
+
 def main(argv=None) -> None:
     p = argparse.ArgumentParser(prog="parallax_amd")
     sub = p.add_subparsers(dest="cmd", required=True)
@@ -158,10 +228,7 @@ def main(argv=None) -> None:
     ps.add_argument("--model-path", default=None)
     ps.add_argument("--model", default="deepseek-r1-distill-llama-8b")
     ps.add_argument("--model-name", default=None)
-    ps.add_argument("--block-size", type=int, default=32)
-    ps.add_argument("--max-batch-size", type=int, default=128)
-    ps.add_argument("--max-num-tokens-per-batch", type=int, default=16384)
-    ps.add_argument("--max-model-len", type=int, default=8192)
+    add_engine_args(ps)
     ps.set_defaults(fn=cmd_serve)
 
     pc = sub.add_parser("chat", help="terminal chat client")

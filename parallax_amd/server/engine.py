@@ -94,6 +94,7 @@ class Engine:
         comm: Optional[CommContext] = None,
         model_path: Optional[str] = None,
         random_weights: bool = False,
+        lora_path: Optional[str] = None,
     ):
         self.cfg = cfg
         self.args = args
@@ -115,7 +116,7 @@ class Engine:
         elif model_path is not None:
             from .shard_loader import load_shard_weights
 
-            load_shard_weights(self.model, model_path)
+            load_shard_weights(self.model, model_path, lora_path=lora_path)
         if hasattr(self.model, "finalize_weights"):
             self.model.finalize_weights()
         self.model = self.model.to(device=self.device, dtype=args.dtype)
