@@ -153,6 +153,14 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
     async def dashboard():
         from fastapi.responses import HTMLResponse
 
+        # full single-file app (setup wizard + node monitor + chat UI — the
+        # dependency-free equivalent of the reference's React dashboard);
+        # the inline page below is the fallback if the static file is missing
+        static = os.path.join(os.path.dirname(__file__), "static",
+                              "dashboard.html")
+        if os.path.exists(static):
+            with open(static) as f:
+                return HTMLResponse(f.read())
         return HTMLResponse("""<!doctype html><html><head>
 <title>parallax_amd cluster</title>
 <style>body{font-family:monospace;margin:2em;background:#111;color:#eee}

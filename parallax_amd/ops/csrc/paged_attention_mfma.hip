@@ -3,16 +3,20 @@
 // Round-2 restructure: decode attention has ZERO K/V reuse inside a workgroup
 // (each K row feeds one QK^T fragment, each V^T row feeds one PV fragment),
 // so the round-1 LDS staging (and its 32-scalar-store V transpose) was pure
-// overhead. The V cache is now stored TRANSPOSED ([NB, Hk, D, BS], see
-// kv_cache.py), which makes both MFMA operand fragments DIRECT 16-byte HBM
-// loads:
-//   S^T[64tok x 16q] = K_tile . Q^T   A-frag: 8 consecutive d of one token
-//   O^T[D x 16q]    += V^T . P        A-frag: 8 consecutive tokens of one d
-// Q lives in registers; only P (2 KB) and the softmax reductions use LDS, so
-// occupancy is bounded by registers, not LDS, and the kernel runs at the KV
-// stream rate. K/V fragments are register double-buffered across the 64-token
-// tiles (issue tile n+1's loads before tile n's math; the compiler's
-// per-register waitcnts do the rest — guide §6 G15).
+// overhead. The V cache is stored TRANSPOSED ([NB, Hk, D, BS], kv_cache.py),
+// which makes both MFMA operand fragments DIRECT 16-byte HBM loads:
+//   S^T[KT tok x 16q] = K_tile . Q^T   A-frag: 8 consecutive d of one token
+//   O^T[D x 16q]     += V^T . P        A-frag: 8 consecutive tokens of one d
+// Q lives in registers; only P and the softmax reductions use LDS, so
+// occupancy is bounded by registers, not LDS. K fragments are register
+// double-buffered across tiles with NAMED buffers (a runtime-indexed buffer
+// array goes to scratch — guide common-mistake #20; the first cut measured
+// 272 B/lane of scratch and ran 2x slow). V single-buffers at tile start so
+// its latency hides under QK+softmax.
+//
+// KT (tokens per tile) is a template parameter: 128 halves the per-byte
+// barrier/softmax overhead vs 64 at higher register pressure; the launcher
+// picks via PARALLAX_DM_KTILE (A/B'd on hardware).
 //
 // One workgroup per (sequence, kv_head, partition); partitions are fixed-size
 // (graph-capture stable) and combine through the shared reduce kernel.
@@ -22,11 +26,12 @@
 
 #include "common.h"
 
+#include <stdlib.h>
+
 #define DM_THREADS 256
-#define DM_KTILE 64
 
 template <int HEAD_DIM, int BLOCK_SIZE, bool PARTITIONED, bool KV_FP8,
-          bool SPARSE = false>
+          bool SPARSE = false, int KT = 64>
 __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     uint16_t* __restrict__ out,        // [B, Hq, D] (final mode)
     float* __restrict__ tmp_acc,       // [B, Hq, P, D] (partitioned)
@@ -79,12 +84,14 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     return;
   }
 
-  __shared__ uint16_t Pl[16 * DM_KTILE];  // row = q head, swz
+  __shared__ uint16_t Pl[16 * KT];  // row = q head, swz
   __shared__ float m_s[16], l_s[16], resc[16];
   __shared__ float wred[4][16];
 
   constexpr int KC = HEAD_DIM / 32;      // 16-B K/Q chunks per row
   constexpr int MT = HEAD_DIM / 16 / 4;  // PV M-tiles per wave
+  constexpr int MS = KT / 64;            // token m-tiles per wave (QK)
+  constexpr int SC = KT / 32;            // PV K-steps
 
   // ---- Q fragments in registers (head l15; zero-pad heads >= G) ---------------
   bf16x8v qfrag[KC];
@@ -107,8 +114,8 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
 
   f32x4v acc_o[MT] = {};
   const int* btab = block_tables + (size_t)seq * max_blocks;
-  const int kt_begin = tok_begin / DM_KTILE;
-  const int kt_end = (tok_end + DM_KTILE - 1) / DM_KTILE;
+  const int kt_begin = tok_begin / KT;
+  const int kt_end = (tok_end + KT - 1) / KT;
 
   // logical token -> global token (sparse indirection; clamped for address
   // safety — masked scores / zero p neutralize the values)
@@ -122,42 +129,43 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     return min(max(gt, 0), L - 1);
   };
 
-  // ---- register double-buffered tile loads ------------------------------------
-  // K: lane covers token (16*wid + l15), all KC d-chunks.
+  // K: wave wid owns token rows [wid*16*MS, +16*MS); lane l15 picks the row
+  // within each 16-row m-tile, all KC d-chunks per row.
+  int4 kregA[MS][KC], kregB[MS][KC];
   // V^T: lane covers d rows (wid*MT+mt)*16 + l15, token chunk s*32 + l4*8.
-  // Two NAMED buffer sets with a compile-time parity branch: runtime-indexed
-  // register arrays land in scratch memory (guide common-mistake #20 — the
-  // first cut of this kernel measured 272 B/lane of scratch and ran 2x slow).
-  int4 kregA[KC], kregB[KC];
-  int4 vregA[2][MT], vregB[2][MT];
+  int4 vreg[SC][MT];
 
-  auto load_k = [&](int kt, int4 (&kreg)[KC]) {
-    bool valid;
-    const int gtok = map_tok(kt * DM_KTILE + 16 * wid + l15, valid);
-    const int blk = btab[gtok / BLOCK_SIZE];
-    const size_t row =
-        (((size_t)blk * Hk + hk) * BLOCK_SIZE + gtok % BLOCK_SIZE) * HEAD_DIM;
+  auto load_k = [&](int kt, int4 (&kreg)[MS][KC]) {
 #pragma unroll
-    for (int s = 0; s < KC; ++s) {
-      if (KV_FP8) {
-        const uint64_t raw = *reinterpret_cast<const uint64_t*>(
-            (const uint8_t*)k_cache_v + row + s * 32 + l4 * 8);
-        uint16_t* e = reinterpret_cast<uint16_t*>(&kreg[s]);
+    for (int ms = 0; ms < MS; ++ms) {
+      bool valid;
+      const int gtok =
+          map_tok(kt * KT + (wid * MS + ms) * 16 + l15, valid);
+      const int blk = btab[gtok / BLOCK_SIZE];
+      const size_t row =
+          (((size_t)blk * Hk + hk) * BLOCK_SIZE + gtok % BLOCK_SIZE) * HEAD_DIM;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          e[j] = f32_to_bf16_bits(
-              fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * k_scale);
-      } else {
-        kreg[s] = *reinterpret_cast<const int4*>(
-            (const uint16_t*)k_cache_v + row + s * 32 + l4 * 8);
+      for (int s = 0; s < KC; ++s) {
+        if (KV_FP8) {
+          const uint64_t raw = *reinterpret_cast<const uint64_t*>(
+              (const uint8_t*)k_cache_v + row + s * 32 + l4 * 8);
+          uint16_t* e = reinterpret_cast<uint16_t*>(&kreg[ms][s]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            e[j] = f32_to_bf16_bits(
+                fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * k_scale);
+        } else {
+          kreg[ms][s] = *reinterpret_cast<const int4*>(
+              (const uint16_t*)k_cache_v + row + s * 32 + l4 * 8);
+        }
       }
     }
   };
 
-  auto load_v = [&](int kt, int4 (&vreg)[2][MT]) {
+  auto load_v = [&](int kt) {
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const int j0 = kt * DM_KTILE + s * 32 + l4 * 8;
+    for (int s = 0; s < SC; ++s) {
+      const int j0 = kt * KT + s * 32 + l4 * 8;
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
         const int drow = (wid * MT + mt) * 16 + l15;
@@ -201,37 +209,40 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     }
   };
 
-  auto tile_math = [&](int kt, const int4 (&kreg)[KC],
-                       const int4 (&vreg)[2][MT]) {
-    const int kbase = kt * DM_KTILE;
+  auto tile_math = [&](int kt, const int4 (&kreg)[MS][KC]) {
+    const int kbase = kt * KT;
 
-    // ---- S^T = K . Q^T (wave w: k rows [16w, 16w+16)) ---------------------------
-    f32x4v acc_s = {};
+    // ---- S^T = K . Q^T (wave w: token rows [w*16*MS, +16*MS)) -------------------
+    f32x4v acc_s[MS] = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < KC; ++s) {
-      const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(&kreg[s]);
-      acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, qfrag[s], acc_s,
-                                                      0, 0, 0);
-    }
+    for (int ms = 0; ms < MS; ++ms)
+#pragma unroll
+      for (int s = 0; s < KC; ++s) {
+        const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(&kreg[ms][s]);
+        acc_s[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, qfrag[s], acc_s[ms], 0, 0, 0);
+      }
     __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + per-column max --------------------------------------------------
     float mx = -3.0e4f;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int j = kbase + 16 * wid + l4 * 4 + r;
-      float s = acc_s[r] * scale;
-      if (softcap > 0.f) s = softcap * tanhf(s / softcap);
-      bool visible = j >= tok_begin && j < tok_end && l15 < G;
-      if (SPARSE && visible) {
-        const int gt = pos_row[j];
-        visible = gt >= 0 && gt < L;
+    for (int ms = 0; ms < MS; ++ms)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int j = kbase + (wid * MS + ms) * 16 + l4 * 4 + r;
+        float s = acc_s[ms][r] * scale;
+        if (softcap > 0.f) s = softcap * tanhf(s / softcap);
+        bool visible = j >= tok_begin && j < tok_end && l15 < G;
+        if (SPARSE && visible) {
+          const int gt = pos_row[j];
+          visible = gt >= 0 && gt < L;
+        }
+        s = visible ? s : -3.0e4f;
+        acc_s[ms][r] = s;
+        mx = fmaxf(mx, s);
       }
-      s = visible ? s : -3.0e4f;
-      acc_s[r] = s;
-      mx = fmaxf(mx, s);
-    }
     mx = fmaxf(mx, __shfl_xor(mx, 16, WAVE_SIZE));
     mx = fmaxf(mx, __shfl_xor(mx, 32, WAVE_SIZE));
     if (lane < 16) wred[wid][lane] = mx;
@@ -250,19 +261,21 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     {
       const float m = m_s[l15];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const float p = __expf(acc_s[r] - m);
-        sm += p;
-        const int kk = 16 * wid + l4 * 4 + r;
-        const int pb = swz(l15 * DM_KTILE * 2 + kk * 2, l15);
-        *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(Pl) + pb) =
-            f32_to_bf16_bits(p);
-      }
+      for (int ms = 0; ms < MS; ++ms)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float p = __expf(acc_s[ms][r] - m);
+          sm += p;
+          const int kk = (wid * MS + ms) * 16 + l4 * 4 + r;
+          const int pb = swz(l15 * KT * 2 + kk * 2, l15);
+          *reinterpret_cast<uint16_t*>(reinterpret_cast<char*>(Pl) + pb) =
+              f32_to_bf16_bits(p);
+        }
       sm += __shfl_xor(sm, 16, WAVE_SIZE);
       sm += __shfl_xor(sm, 32, WAVE_SIZE);
     }
     if (lane < 16) wred[wid][lane] = sm;
-    __syncthreads();
+    __syncthreads();  // Pl + wred writes from every wave before l_s / PV
     if (tid < 16)
       l_s[tid] = l_s[tid] * resc[tid] + wred[0][tid] + wred[1][tid] +
                  wred[2][tid] + wred[3][tid];
@@ -280,14 +293,13 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
     __syncthreads();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const int pb = swz(l15 * DM_KTILE * 2 + s * 64 + l4 * 16, l15);
+    for (int s = 0; s < SC; ++s) {
+      const int pb = swz(l15 * KT * 2 + s * 64 + l4 * 16, l15);
       const bf16x8v bfrag = *reinterpret_cast<const bf16x8v*>(
           reinterpret_cast<const char*>(Pl) + pb);
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
-        const bf16x8v afrag =
-            *reinterpret_cast<const bf16x8v*>(&vreg[s][mt]);
+        const bf16x8v afrag = *reinterpret_cast<const bf16x8v*>(&vreg[s][mt]);
         acc_o[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                             acc_o[mt], 0, 0, 0);
       }
@@ -298,20 +310,16 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   };
 
   load_k(kt_begin, kregA);
-  load_v(kt_begin, vregA);
   for (int kt = kt_begin; kt < kt_end; ++kt) {
+    // V for THIS tile issues first (its latency hides under QK+softmax);
+    // next tile's K prefetches into the other named buffer.
+    load_v(kt);
     if (((kt - kt_begin) & 1) == 0) {
-      if (kt + 1 < kt_end) {
-        load_k(kt + 1, kregB);
-        load_v(kt + 1, vregB);
-      }
-      tile_math(kt, kregA, vregA);
+      if (kt + 1 < kt_end) load_k(kt + 1, kregB);
+      tile_math(kt, kregA);
     } else {
-      if (kt + 1 < kt_end) {
-        load_k(kt + 1, kregA);
-        load_v(kt + 1, vregA);
-      }
-      tile_math(kt, kregB, vregB);
+      if (kt + 1 < kt_end) load_k(kt + 1, kregA);
+      tile_math(kt, kregB);
     }
   }
 
@@ -354,6 +362,14 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   }
 }
 
+static int dm_ktile() {
+  static const int kt = [] {
+    const char* e = getenv("PARALLAX_DM_KTILE");
+    return (e && atoi(e) == 64) ? 64 : 128;
+  }();
+  return kt;
+}
+
 extern "C" void launch_msa_paged_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* seq_lens, const int* token_positions,
@@ -364,8 +380,9 @@ extern "C" void launch_msa_paged_attention_decode(
   *launched = false;
   if (G > 16) return;
   dim3 grid(Hk, B, 1);
+  // sparse tiles gather per-element anyway: KT=64 keeps the tail waste low
 #define MSA_LAUNCH(HD, BSZ)                                                    \
-  paged_decode_mfma_kernel<HD, BSZ, false, false, true>                        \
+  paged_decode_mfma_kernel<HD, BSZ, false, false, true, 64>                    \
       <<<grid, DM_THREADS, 0, stream>>>(                                       \
           (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q, k_cache,       \
           v_cache, block_tables, seq_lens, max_blocks, Hk, G, q_stride, scale, \
@@ -392,17 +409,17 @@ extern "C" void launch_paged_decode_mfma(
   *launched = false;
   if (G > 16) return;
 
-#define DM_LAUNCH2(HD, BSZ, FP8)                                               \
+#define DM_LAUNCH3(HD, BSZ, FP8, KTV)                                          \
   if (num_parts <= 1) {                                                        \
     dim3 grid(Hk, B, 1);                                                       \
-    paged_decode_mfma_kernel<HD, BSZ, false, FP8>                              \
+    paged_decode_mfma_kernel<HD, BSZ, false, FP8, false, KTV>                  \
         <<<grid, DM_THREADS, 0, stream>>>(                                     \
             (uint16_t*)out, nullptr, nullptr, (const uint16_t*)q, k_cache,     \
             v_cache, block_tables, seq_lens, max_blocks, Hk, G, q_stride,      \
             scale, sliding_window, softcap, sinks, 0, k_scale, v_scale);       \
   } else {                                                                     \
     dim3 grid(Hk, B, num_parts);                                               \
-    paged_decode_mfma_kernel<HD, BSZ, true, FP8>                               \
+    paged_decode_mfma_kernel<HD, BSZ, true, FP8, false, KTV>                   \
         <<<grid, DM_THREADS, 0, stream>>>(                                     \
             nullptr, tmp_acc, tmp_ml, (const uint16_t*)q, k_cache, v_cache,    \
             block_tables, seq_lens, max_blocks, Hk, G, q_stride, scale,        \
@@ -411,7 +428,13 @@ extern "C" void launch_paged_decode_mfma(
   *launched = true;
 
 #define DM_LAUNCH(HD, BSZ)                                                     \
-  if (kv_fp8) { DM_LAUNCH2(HD, BSZ, true) } else { DM_LAUNCH2(HD, BSZ, false) }
+  if (kv_fp8) {                                                                \
+    if (dm_ktile() == 128) { DM_LAUNCH3(HD, BSZ, true, 128) }                  \
+    else { DM_LAUNCH3(HD, BSZ, true, 64) }                                     \
+  } else {                                                                     \
+    if (dm_ktile() == 128) { DM_LAUNCH3(HD, BSZ, false, 128) }                 \
+    else { DM_LAUNCH3(HD, BSZ, false, 64) }                                    \
+  }
 
   if (D == 128 && BS == 32) { DM_LAUNCH(128, 32) }
   else if (D == 128 && BS == 16) { DM_LAUNCH(128, 16) }
@@ -420,5 +443,5 @@ extern "C" void launch_paged_decode_mfma(
   else if (D == 64 && BS == 16) { DM_LAUNCH(64, 16) }
   else if (D == 64 && BS == 64) { DM_LAUNCH(64, 64) }
 #undef DM_LAUNCH
-#undef DM_LAUNCH2
+#undef DM_LAUNCH3
 }

@@ -50,6 +50,9 @@ if __name__ == "__main__":
     # flagship decode shape (8B: Hk=8, G=4) at various batch/ctx
     for B, ctx in [(64, 520), (128, 520), (256, 520), (64, 2048), (64, 8192), (8, 8192), (1, 131072)]:
         run_case(B, 8, 4, 128, ctx, max_seq_len=8192 if ctx <= 8192 else ctx)
+    # VERDICT target shape: batch 512, ctx 4096 (>=5.5 TB/s effective)
+    run_case(512, 8, 4, 128, 4096, 4096, iters=20)
+    run_case(512, 8, 4, 128, 640, 1024)
     # effect of the graph-mode fixed partition count (msl >> ctx)
     for msl in [512, 1024, 2048, 8192, 32768]:
         run_case(64, 8, 4, 128, 520, max_seq_len=msl)
