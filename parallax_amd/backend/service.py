@@ -19,6 +19,7 @@ Endpoints:
 from __future__ import annotations
 
 import asyncio
+import os
 import threading
 import time
 from typing import Dict, Optional

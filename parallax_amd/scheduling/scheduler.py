@@ -211,6 +211,8 @@ class ClusterScheduler:
             "bootstrapped": self.bootstrapped,
             "num_nodes": len(self.nodes),
             "num_pipelines": len(self.pipelines),
+            "num_layers": self.model.num_layers,
+            "last_refit_time": self.last_refit_time,
             "nodes": [
                 {
                     "node_id": n.node_id,
