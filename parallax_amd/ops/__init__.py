@@ -557,5 +557,41 @@ def fused_moe_forward_fp8(
 # -- sampling (torch ops; GPU path uses torch's ROCm kernels — not a hot spot
 #    relative to the model forward, custom kernel is a later optimization) -------
 
-sample_tokens = ref.sample_tokens
+_SAMPLE_CALLS = 0
+
+
+def sample_tokens(logits, temperatures, top_ps, top_ks, min_ps,
+                  generator=None):
+    """Batched sampling. GPU rows with no top-k/p/min-p filtering take the
+    fused one-pass gumbel-max HIP kernel (the torch composition makes ~6
+    passes over the [B, vocab] buffer); filtered rows fall back to the torch
+    path. Seed = generator seed + a per-call counter so repeated calls draw
+    fresh noise deterministically."""
+    global _SAMPLE_CALLS
+    if logits.is_cuda and has_extension():
+        B, V = logits.shape
+        temps = [float(t) for t in temperatures]
+        greedy = [t <= 0.0 for t in temps]
+        need_filter = any(
+            (not g) and (0 < int(k) < V or float(p) < 1.0 or float(m) > 0.0)
+            for g, k, p, m in zip(greedy, top_ks, top_ps, min_ps)
+        )
+        if not need_filter:
+            ext = _require_ext("sample_gumbel")
+            _SAMPLE_CALLS += 1
+            base = generator.initial_seed() if generator is not None else 0x5eed
+            inv_t = torch.tensor(
+                [1.0 / max(t, 1e-6) for t in temps], dtype=torch.float32,
+                device=logits.device,
+            )
+            gmask = torch.tensor(greedy, dtype=torch.uint8,
+                                 device=logits.device)
+            return ext.sample_gumbel(
+                logits.float().contiguous(), inv_t, gmask,
+                (base + _SAMPLE_CALLS) & 0x7FFFFFFFFFFF,
+            )
+    return ref.sample_tokens(logits, temperatures, top_ps, top_ks, min_ps,
+                             generator)
+
+
 apply_penalties = ref.apply_penalties

@@ -77,6 +77,8 @@ void launch_moe_down_fp8(void*, const void*, const float*, const void*,
                          const float*, const float*, const int*, const int*,
                          const int64_t*, const int*, int, int, int, int, int,
                          hipStream_t);
+void launch_sample_gumbel(int64_t*, const float*, const float*,
+                          const uint8_t*, uint64_t, int, int, hipStream_t);
 void launch_skinny_gemm(void*, const void*, const void*, const void*, int,
                         int, int, int64_t, int64_t, hipStream_t, bool*);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
@@ -255,10 +257,17 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     sinks_ptr = sinks_f.data_ptr<float>();
   }
 
-  // flash-decoding split: fixed 256-token partitions (graph-capture friendly —
-  // num_parts depends only on max_seq_len, and partitions past a sequence's
-  // actual length exit immediately / are skipped by the reduce kernel)
+  // Flash-decoding split, graph-capture friendly (num_parts depends only on
+  // max_seq_len and the batch bucket; partitions past a sequence's actual
+  // length exit immediately / are skipped by the reduce kernel). Partition
+  // size grows with B*Hk: when the batch alone fills the chip, bigger
+  // partitions cut empty-partition sweep + reduce cost and pick the KT=128
+  // tile path (A/B'd: profiles/README.md round 2).
+  const int wg_base = B * Hk;
   int part_tokens = 256;
+  if (wg_base >= 4096) part_tokens = 2048;
+  else if (wg_base >= 1024) part_tokens = 1024;
+  else if (wg_base >= 512) part_tokens = 512;
   int num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
   if (num_parts > 128) {
     part_tokens = (int)((max_seq_len + 127) / 128);
@@ -360,7 +369,12 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
   const int max_blocks = block_tables.size(1);
   TORCH_CHECK(cache.size(2) == R + DR, "cache entry dim mismatch");
 
+  // same grid-fill-aware partition sizing as the GQA decode path
+  const int mla_wg_base = B * ((H + 15) / 16);
   int part_tokens = 256;
+  if (mla_wg_base >= 4096) part_tokens = 2048;
+  else if (mla_wg_base >= 1024) part_tokens = 1024;
+  else if (mla_wg_base >= 512) part_tokens = 512;
   int num_parts = (int)((max_seq_len + part_tokens - 1) / part_tokens);
   if (num_parts > 128) {
     part_tokens = (int)((max_seq_len + 127) / 128);
@@ -576,6 +590,23 @@ void moe_forward_fp8(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
       seg_offsets.data_ptr<int>(), E, (int)topk, H, I, max_tiles, stream);
 }
 
+torch::Tensor sample_gumbel(torch::Tensor logits, torch::Tensor inv_temp,
+                            torch::Tensor greedy, int64_t seed) {
+  CHECK_GPU(logits);
+  CHECK_CONTIG(logits);
+  TORCH_CHECK(logits.scalar_type() == at::kFloat, "logits must be fp32");
+  TORCH_CHECK(inv_temp.scalar_type() == at::kFloat);
+  TORCH_CHECK(greedy.scalar_type() == at::kByte);
+  const int B = logits.size(0);
+  const int V = logits.size(1);
+  auto out = torch::empty({B}, logits.options().dtype(at::kLong));
+  launch_sample_gumbel(out.data_ptr<int64_t>(), logits.data_ptr<float>(),
+                       inv_temp.contiguous().data_ptr<float>(),
+                       greedy.contiguous().data_ptr<uint8_t>(),
+                       (uint64_t)seed, B, V, cur_stream());
+  return out;
+}
+
 bool skinny_gemm(torch::Tensor c, torch::Tensor x, torch::Tensor w,
                  torch::Tensor bias) {
   CHECK_GPU(x);
@@ -630,6 +661,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "tuned hipBLASLt y = x @ w^T (bf16, fp32 accum)");
   m.def("msa_paged_attention_decode", &msa_paged_attention_decode);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("sample_gumbel", &sample_gumbel);
   m.def("moe_forward", &moe_forward);
   m.def("moe_forward_fp8", &moe_forward_fp8);
   m.def("dsa_indexer_scores", &dsa_indexer_scores);

@@ -362,12 +362,16 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
   }
 }
 
-static int dm_ktile() {
-  static const int kt = [] {
+static int dm_ktile(int span_tokens) {
+  // A/B on MI355X (profiles/README.md round 2): KT=128 wins when each
+  // workgroup walks a long token span (ctx 8k: 5.63 vs 5.08 TB/s), KT=64
+  // wins on short spans (less boundary-tile waste). Env overrides for A/B.
+  static const int forced = [] {
     const char* e = getenv("PARALLAX_DM_KTILE");
-    return (e && atoi(e) == 64) ? 64 : 128;
+    return e ? atoi(e) : 0;
   }();
-  return kt;
+  if (forced == 64 || forced == 128) return forced;
+  return span_tokens >= 2048 ? 128 : 64;
 }
 
 extern "C" void launch_msa_paged_attention_decode(
@@ -427,12 +431,15 @@ extern "C" void launch_paged_decode_mfma(
   }                                                                            \
   *launched = true;
 
+  const int span = num_parts > 1 ? part_tokens : 1 << 30;
+  const int kt_pick = dm_ktile(span);
+
 #define DM_LAUNCH(HD, BSZ)                                                     \
   if (kv_fp8) {                                                                \
-    if (dm_ktile() == 128) { DM_LAUNCH3(HD, BSZ, true, 128) }                  \
+    if (kt_pick == 128) { DM_LAUNCH3(HD, BSZ, true, 128) }                     \
     else { DM_LAUNCH3(HD, BSZ, true, 64) }                                     \
   } else {                                                                     \
-    if (dm_ktile() == 128) { DM_LAUNCH3(HD, BSZ, false, 128) }                 \
+    if (kt_pick == 128) { DM_LAUNCH3(HD, BSZ, false, 128) }                    \
     else { DM_LAUNCH3(HD, BSZ, false, 64) }                                    \
   }
 

@@ -27,6 +27,7 @@ sources = [
     os.path.join(CSRC, "moe.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "indexer.hip"),
+    os.path.join(CSRC, "sampler.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 

@@ -361,3 +361,33 @@ def test_lt_linear_tuned(M, N, K):
     torch.testing.assert_close(y.float(), ref, atol=2e-1, rtol=2e-2)
     y2 = ext.lt_linear(x, w)
     assert torch.equal(y, y2)
+
+
+def test_sample_gumbel_kernel():
+    """Fused one-pass gumbel-max sampler: greedy rows equal argmax; sampled
+    rows concentrate on the dominant logit; fixed generator seed reproduces."""
+    torch.manual_seed(14)
+    B, V = 8, 1000
+    logits = torch.randn(B, V, device="cuda")
+    logits[:, 7] += 12.0  # dominant token
+    gen = torch.Generator(device="cuda").manual_seed(123)
+
+    toks = ops.sample_tokens(logits, [0.0] * B, [1.0] * B, [-1] * B,
+                             [0.0] * B, generator=gen)
+    assert (toks == logits.argmax(-1)).all()
+
+    hits = 0
+    for _ in range(20):
+        t = ops.sample_tokens(logits, [1.0] * B, [1.0] * B, [-1] * B,
+                              [0.0] * B, generator=gen)
+        hits += int((t == 7).sum())
+    assert hits > 0.9 * 20 * B, f"dominant token sampled {hits}/160"
+
+    # draws vary across calls (the counter advances)
+    a = ops.sample_tokens(torch.zeros(1, V, device="cuda"), [1.0], [1.0],
+                          [-1], [0.0], generator=gen)
+    b = ops.sample_tokens(torch.zeros(1, V, device="cuda"), [1.0], [1.0],
+                          [-1], [0.0], generator=gen)
+    c = ops.sample_tokens(torch.zeros(1, V, device="cuda"), [1.0], [1.0],
+                          [-1], [0.0], generator=gen)
+    assert len({int(a), int(b), int(c)}) > 1
