@@ -370,7 +370,7 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
   TORCH_CHECK(cache.size(2) == R + DR, "cache entry dim mismatch");
 
   // same grid-fill-aware partition sizing as the GQA decode path
-  const int mla_wg_base = B * ((H + 15) / 16);
+  const int mla_wg_base = B * ((H + 31) / 32);  // MLA_HBLOCK = 32
   int part_tokens = 256;
   if (mla_wg_base >= 4096) part_tokens = 2048;
   else if (mla_wg_base >= 1024) part_tokens = 1024;

@@ -665,17 +665,70 @@ class Engine:
         return self.model.compute_logits(hidden[need_logits_at])
 
     def _run_prefill(self, chunks: List[PrefillChunk]):
-        meta, input_ids = self._build_prefill_meta(chunks)
         samp_reqs = [c.req for c in chunks if c.is_last_chunk]
-        # logits only at last tokens of final chunks
+        mb = max(1, self.args.micro_batches) if self.comm.pp_size > 1 else 1
+        groups: List[List[PrefillChunk]] = (
+            [list(g) for g in _split(chunks, mb)] if mb > 1 and len(chunks) >= mb
+            else [chunks]
+        )
+        logits_parts: List[torch.Tensor] = []
+        with torch.inference_mode():
+            if len(groups) == 1:
+                logits = self._prefill_group(chunks)
+                return logits, samp_reqs
+            # micro-batch pipelining: stage i computes group g while its
+            # hidden states for group g-1 are in flight to stage i+1 (same
+            # pre-posted-irecv + staged-isend pattern as _run_decode)
+            comm = self.comm
+            h = self.cfg.hidden_size
+            recvs: List[tuple] = []
+            if not comm.is_first_stage:
+                for g in groups:
+                    T = sum(c.num_tokens for c in g)
+                    recvs.append(
+                        comm.pp_irecv((T, h), self.args.dtype, comm.pp_rank - 1)
+                    )
+            for i, g in enumerate(groups):
+                hidden_in = None
+                if not comm.is_first_stage:
+                    buf, work = recvs[i]
+                    work.wait()
+                    hidden_in = buf
+                out = self._prefill_group(g, hidden_in=hidden_in,
+                                          send_slot=i if not comm.is_last_stage
+                                          else None)
+                if comm.is_last_stage and out is not None:
+                    logits_parts.append(out)
+        if self.comm.is_last_stage and logits_parts:
+            return torch.cat(logits_parts, dim=0), samp_reqs
+        return None, samp_reqs
+
+    def _prefill_group(self, chunks: List[PrefillChunk],
+                       hidden_in: Optional[torch.Tensor] = None,
+                       send_slot: Optional[int] = None):
+        """Forward one prefill (micro-)batch through the local shard. When
+        hidden_in/send_slot are given the PP transport is async (pipelined);
+        otherwise the blocking single-group path runs."""
+        meta, input_ids = self._build_prefill_meta(chunks)
         keep = torch.tensor(
             [i for i, c in enumerate(chunks) if c.is_last_chunk], dtype=torch.int64
         )
         need = meta.logits_indices.cpu()[keep].to(self.device) if keep.numel() else \
             torch.empty(0, dtype=torch.int64, device=self.device)
-        with torch.inference_mode():
-            logits = self._pipeline_forward(meta, input_ids, need)
-        return logits, samp_reqs
+        comm = self.comm
+        if hidden_in is None and send_slot is None:
+            return self._pipeline_forward(meta, input_ids, need)
+        if comm.is_first_stage:
+            hidden = self.model.embed(input_ids).to(self.args.dtype)
+        else:
+            hidden = hidden_in
+        hidden = self.model(hidden, meta)
+        if not comm.is_last_stage:
+            comm.pp_send_async(hidden, comm.pp_rank + 1, slot=1000 + (send_slot or 0))
+            return None
+        if need.numel() == 0:
+            return None
+        return self.model.compute_logits(hidden[need])
 
     def _decode_fwd(
         self, reqs: List[InitialRequest], hidden_in: Optional[torch.Tensor]
