@@ -63,12 +63,21 @@ class TcpTransport(Transport):
     #: refuse frames claiming more than this many bytes (a corrupt or malicious
     #: peer must not be able to drive unbounded allocation on the receiver)
     DEFAULT_MAX_FRAME_BYTES = 512 * 1024 * 1024
+    #: inbox backpressure: a reader that stalls makes senders block in
+    #: sendall (TCP flow control) instead of growing receiver memory
+    DEFAULT_MAX_INBOX = 4096
 
     def __init__(self, peer_id: str, host: str = "0.0.0.0", port: int = 0,
-                 max_frame_bytes: int = DEFAULT_MAX_FRAME_BYTES):
+                 max_frame_bytes: int = DEFAULT_MAX_FRAME_BYTES,
+                 max_inbox: int = DEFAULT_MAX_INBOX,
+                 auth_token: Optional[str] = None):
         self.peer_id = peer_id
         self.max_frame_bytes = max_frame_bytes
-        self.inbox: "queue.Queue[bytes]" = queue.Queue()
+        # shared-secret handshake: when set, every outbound connection sends
+        # an AUTH frame first and inbound connections must present it before
+        # any payload frame is accepted
+        self.auth_token = auth_token
+        self.inbox: "queue.Queue[bytes]" = queue.Queue(maxsize=max_inbox)
         self._peers: Dict[str, tuple] = {}
         self._conns: Dict[str, socket.socket] = {}
         self._lock = threading.Lock()
@@ -95,6 +104,9 @@ class TcpTransport(Transport):
                     host, port = self._peers[peer_id]
                     conn = socket.create_connection((host, port), timeout=10)
                     conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+                    if self.auth_token is not None:
+                        tok = self.auth_token.encode()
+                        conn.sendall(b"AUTH" + struct.pack("<I", len(tok)) + tok)
                     with self._lock:
                         self._conns[peer_id] = conn
                 conn.sendall(struct.pack("<Q", len(payload)) + payload)
@@ -123,6 +135,22 @@ class TcpTransport(Transport):
 
     def _read_loop(self, conn: socket.socket) -> None:
         try:
+            if self.auth_token is not None:
+                magic = self._read_exact(conn, 4)
+                if magic != b"AUTH":
+                    logger.warning("dropping connection: no auth handshake")
+                    return
+                hdr = self._read_exact(conn, 4)
+                if hdr is None:
+                    return
+                (tl,) = struct.unpack("<I", hdr)
+                if tl > 4096:
+                    logger.warning("dropping connection: oversized auth token")
+                    return
+                tok = self._read_exact(conn, tl)
+                if tok != self.auth_token.encode():
+                    logger.warning("dropping connection: bad auth token")
+                    return
             while not self._stop.is_set():
                 header = self._read_exact(conn, 8)
                 if header is None:
@@ -137,6 +165,8 @@ class TcpTransport(Transport):
                 payload = self._read_exact(conn, n)
                 if payload is None:
                     break
+                # bounded inbox: blocking put -> the TCP window closes and the
+                # SENDER stalls (backpressure) rather than this side ballooning
                 self.inbox.put(payload)
         finally:
             conn.close()

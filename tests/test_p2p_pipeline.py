@@ -249,3 +249,28 @@ def test_tcp_transport_many_frames_in_order():
     assert got == payloads
     a.close()
     b.close()
+
+
+def test_tcp_auth_handshake():
+    """Auth-token transports accept matching peers and drop mismatched or
+    unauthenticated connections (VERDICT weak #8)."""
+    import time as _time
+
+    from parallax_amd.p2p.transport import TcpTransport
+
+    a = TcpTransport("a", host="127.0.0.1", auth_token="s3cret")
+    b = TcpTransport("b", host="127.0.0.1", auth_token="s3cret")
+    bad = TcpTransport("x", host="127.0.0.1", auth_token="wrong")
+    naked = TcpTransport("n", host="127.0.0.1")  # no token at all
+    try:
+        a.set_peer_addr("b", "127.0.0.1", b.port)
+        bad.set_peer_addr("b", "127.0.0.1", b.port)
+        naked.set_peer_addr("b", "127.0.0.1", b.port)
+        a.send("b", b"hello")
+        assert b.recv(timeout=5.0) == b"hello"
+        bad.send("b", b"evil")
+        naked.send("b", b"sneaky")
+        assert b.recv(timeout=0.8) is None  # both rejected at handshake
+    finally:
+        for t in (a, b, bad, naked):
+            t.close()
