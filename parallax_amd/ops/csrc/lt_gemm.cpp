@@ -135,7 +135,11 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
             HIPBLAS_COMPUTE_32F, all) == HIPBLAS_STATUS_SUCCESS) {
       // the screening loop below is additionally time-budgeted, but cap the
       // supported-check pass too (it walks thousands of entries)
-      const size_t cap = (M <= 1024) ? 2048 : 384;
+      static const size_t cap_env = [] {
+        const char* e = getenv("PARALLAX_LT_TUNE_CAP");
+        return e ? (size_t)atoll(e) : (size_t)0;
+      }();
+      const size_t cap = cap_env ? cap_env : ((M <= 1024) ? 2048 : 384);
       size_t kept = 0;
       const float alpha1 = 1.f, beta1 = 0.f;
       for (auto& r : all) {

@@ -164,9 +164,9 @@ def test_kimi_k2_256k_prefix_reuse_smoke():
     requests. Asserts the second request's prefill is served from cache."""
     cfg = ModelConfig.from_hf_config({
         "architectures": ["KimiK2ForCausalLM"], "model_type": "kimi_k2",
-        "vocab_size": 2048, "hidden_size": 1024, "num_hidden_layers": 1,
-        "num_attention_heads": 16, "num_key_value_heads": 16,
-        "intermediate_size": 2048, "moe_intermediate_size": 512,
+        "vocab_size": 2048, "hidden_size": 512, "num_hidden_layers": 1,
+        "num_attention_heads": 8, "num_key_value_heads": 8,
+        "intermediate_size": 1024, "moe_intermediate_size": 256,
         "n_routed_experts": 384, "num_experts_per_tok": 8,
         "n_shared_experts": 1, "n_group": 1, "topk_group": 1,
         "routed_scaling_factor": 2.5, "norm_topk_prob": True,
