@@ -513,6 +513,32 @@ def fused_moe_forward(
     return out
 
 
+_FP8_LIN_BUFS = {}
+
+
+def linear_fp8(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
+               bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """W8A8 dense GEMM through hipBLASLt fp8: x quantized per-tensor on the
+    fly into persistent per-shape buffers (graph-capture safe: the scale and
+    xq pointers baked into a captured graph never move)."""
+    ext = _require_ext("lt_linear_fp8")
+    M, K = x.shape
+    key = (M, K, x.device.index)
+    bufs = _FP8_LIN_BUFS.get(key)
+    if bufs is None:
+        bufs = (
+            torch.empty(M, K, dtype=torch.float8_e4m3fn, device=x.device),
+            torch.ones(1, dtype=torch.float32, device=x.device),
+        )
+        _FP8_LIN_BUFS[key] = bufs
+    xq, xs = bufs
+    xf = x.float()
+    xs.copy_((xf.abs().amax() / 448.0).clamp_min(1e-8).reshape(1))
+    xq.copy_((xf / xs).clamp(-448.0, 448.0))
+    y = ext.lt_linear_fp8(xq, w_q, xs, w_scale)
+    return y if bias is None else y + bias
+
+
 def quantize_fp8_weight(w: torch.Tensor):
     """Per-output-channel fp8-E4M3 quantization of a weight tensor whose rows
     are output channels along dim -2 ([..., N, K] -> fp8 [..., N, K] +

@@ -655,8 +655,12 @@ void msa_paged_attention_decode(torch::Tensor out, torch::Tensor q,
 }
 
 torch::Tensor lt_linear(torch::Tensor x, torch::Tensor w);  // lt_gemm.cpp
+torch::Tensor lt_linear_fp8(torch::Tensor x_q, torch::Tensor w_q,
+                            torch::Tensor x_scale, torch::Tensor w_scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("lt_linear_fp8", &lt_linear_fp8,
+        "fp8-E4M3 W8A8 y = (xq*sx) @ (wq*sw)^T (bf16 out)");
   m.def("lt_linear", &lt_linear,
         "tuned hipBLASLt y = x @ w^T (bf16, fp32 accum)");
   m.def("msa_paged_attention_decode", &msa_paged_attention_decode);

@@ -49,7 +49,7 @@ struct CachedShape;
 struct LtState {
   hipblasLtHandle_t handle = nullptr;
   void* workspace = nullptr;
-  std::map<std::tuple<int64_t, int64_t, int64_t>,
+  std::map<std::tuple<int64_t, int64_t, int64_t, int>,
            std::unique_ptr<CachedShape>> shapes;
   std::mutex mu;
 };
@@ -74,7 +74,8 @@ struct Descs {
   }
 };
 
-void make_descs(Descs& d, int64_t M, int64_t N, int64_t K) {
+void make_descs(Descs& d, int64_t M, int64_t N, int64_t K,
+                hipDataType in_type = HIP_R_16BF) {
   LT_CHECK(hipblasLtMatmulDescCreate(&d.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
   hipblasOperation_t ta = HIPBLAS_OP_T, tb = HIPBLAS_OP_N;
   LT_CHECK(hipblasLtMatmulDescSetAttribute(d.op, HIPBLASLT_MATMUL_DESC_TRANSA,
@@ -82,9 +83,9 @@ void make_descs(Descs& d, int64_t M, int64_t N, int64_t K) {
   LT_CHECK(hipblasLtMatmulDescSetAttribute(d.op, HIPBLASLT_MATMUL_DESC_TRANSB,
                                            &tb, sizeof(tb)));
   // A = w: [K, N] column-major view of the row-major [N, K] buffer, op T
-  LT_CHECK(hipblasLtMatrixLayoutCreate(&d.a, HIP_R_16BF, K, N, K));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&d.a, in_type, K, N, K));
   // B = x: [K, M] column-major view of the row-major [M, K] buffer
-  LT_CHECK(hipblasLtMatrixLayoutCreate(&d.b, HIP_R_16BF, K, M, K));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&d.b, in_type, K, M, K));
   // C = y^T: [N, M] column-major = row-major y [M, N]
   LT_CHECK(hipblasLtMatrixLayoutCreate(&d.c, HIP_R_16BF, N, M, N));
 }
@@ -96,10 +97,11 @@ struct CachedShape {
 
 hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
                                  const void* wp, const void* xp, void* yp,
-                                 hipStream_t stream) {
+                                 hipStream_t stream,
+                                 hipDataType in_type = HIP_R_16BF) {
   LtState& s = state();
   Descs d;
-  make_descs(d, M, N, K);
+  make_descs(d, M, N, K, in_type);
 
   hipblasLtMatmulPreference_t pref;
   LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
@@ -131,7 +133,7 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
     std::vector<hipblasLtMatmulHeuristicResult_t> all;
     if (hipblaslt_ext::getAllAlgos(
             s.handle, hipblaslt_ext::GemmType::HIPBLASLT_GEMM, HIPBLAS_OP_T,
-            HIPBLAS_OP_N, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
+            HIPBLAS_OP_N, in_type, in_type, HIP_R_16BF, HIP_R_16BF,
             HIPBLAS_COMPUTE_32F, all) == HIPBLAS_STATUS_SUCCESS) {
       // the screening loop below is additionally time-budgeted, but cap the
       // supported-check pass too (it walks thousands of entries)
@@ -235,7 +237,7 @@ torch::Tensor lt_linear(torch::Tensor x, torch::Tensor w) {
   CachedShape* cs = nullptr;
   {
     std::lock_guard<std::mutex> g(s.mu);
-    auto key = std::make_tuple(M, N, K);
+    auto key = std::make_tuple(M, N, K, 0);
     auto it = s.shapes.find(key);
     if (it == s.shapes.end()) {
       hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
@@ -255,6 +257,61 @@ torch::Tensor lt_linear(torch::Tensor x, torch::Tensor w) {
   const float alpha = 1.f, beta = 0.f;
   LT_CHECK(hipblasLtMatmul(s.handle, cs->descs.op, &alpha, w.data_ptr(),
                            cs->descs.a, x.data_ptr(), cs->descs.b, &beta,
+                           y.data_ptr(), cs->descs.c, y.data_ptr(),
+                           cs->descs.c, &cs->algo, s.workspace,
+                           kWorkspaceBytes, stream));
+  return y;
+}
+
+// fp8-E4M3 W8A8 GEMM (BASELINE DeepSeek-V3 "fp8 MFMA" dense route):
+// y[M,N] bf16 = (x_q[M,K] * sx) . (w_q[N,K] * sw)^T with per-tensor DEVICE
+// scale pointers (hipblaslt reads them at kernel time, so graph capture
+// bakes only the persistent buffer addresses the caller guarantees).
+torch::Tensor lt_linear_fp8(torch::Tensor x_q, torch::Tensor w_q,
+                            torch::Tensor x_scale, torch::Tensor w_scale) {
+  TORCH_CHECK(x_q.is_cuda() && w_q.is_cuda());
+  TORCH_CHECK(x_q.scalar_type() == at::kFloat8_e4m3fn &&
+              w_q.scalar_type() == at::kFloat8_e4m3fn,
+              "lt_linear_fp8: fp8_e4m3fn inputs required");
+  TORCH_CHECK(x_q.is_contiguous() && w_q.is_contiguous());
+  TORCH_CHECK(x_scale.scalar_type() == at::kFloat && x_scale.is_cuda());
+  TORCH_CHECK(w_scale.scalar_type() == at::kFloat && w_scale.is_cuda());
+  const int64_t M = x_q.size(0), K = x_q.size(1), N = w_q.size(0);
+  TORCH_CHECK(w_q.size(1) == K, "lt_linear_fp8: K mismatch");
+
+  auto y = at::empty({M, N}, x_q.options().dtype(at::kBFloat16));
+  hipStream_t stream = at::hip::getCurrentHIPStream().stream();
+  LtState& s = state();
+
+  CachedShape* cs = nullptr;
+  {
+    std::lock_guard<std::mutex> g(s.mu);
+    auto key = std::make_tuple(M, N, K, 1);
+    auto it = s.shapes.find(key);
+    if (it == s.shapes.end()) {
+      hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+      (void)hipStreamIsCapturing(stream, &cap);
+      TORCH_CHECK(cap == hipStreamCaptureStatusNone,
+                  "lt_linear_fp8: first call for shape ", M, "x", N, "x", K,
+                  " happened during graph capture; warm this shape up first");
+      auto entry = std::make_unique<CachedShape>();
+      make_descs(entry->descs, M, N, K, HIP_R_8F_E4M3);
+      entry->algo = tune_shape(M, N, K, w_q.data_ptr(), x_q.data_ptr(),
+                               y.data_ptr(), stream, HIP_R_8F_E4M3);
+      it = s.shapes.emplace(key, std::move(entry)).first;
+    }
+    cs = it->second.get();
+  }
+  // per-call device scale pointers (A = w, B = x in our TN formulation)
+  const void* sa = w_scale.data_ptr();
+  const void* sb = x_scale.data_ptr();
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      cs->descs.op, HIPBLASLT_MATMUL_DESC_A_SCALE_POINTER, &sa, sizeof(sa)));
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      cs->descs.op, HIPBLASLT_MATMUL_DESC_B_SCALE_POINTER, &sb, sizeof(sb)));
+  const float alpha = 1.f, beta = 0.f;
+  LT_CHECK(hipblasLtMatmul(s.handle, cs->descs.op, &alpha, w_q.data_ptr(),
+                           cs->descs.a, x_q.data_ptr(), cs->descs.b, &beta,
                            y.data_ptr(), cs->descs.c, y.data_ptr(),
                            cs->descs.c, &cs->algo, s.workspace,
                            kWorkspaceBytes, stream));

@@ -20,6 +20,33 @@ from .comm import get_comm
 from .. import ops
 
 
+class _Fp8WeightMixin:
+    """Per-tensor fp8-E4M3 weight storage for the hipBLASLt W8A8 route
+    (BASELINE DeepSeek-V3 fp8 config). quantize_fp8() swaps the bf16 weight
+    parameter for an fp8 buffer + device scale; forward then dispatches
+    ops.linear_fp8 (activations quantize per-call on device)."""
+
+    fp8 = False
+
+    def quantize_fp8(self) -> None:
+        if self.fp8:
+            return
+        w = self.weight.data
+        amax = w.float().abs().amax().clamp_min(1e-8)
+        scale = (amax / 448.0).reshape(1).float()
+        q = (w.float() / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+        del self._parameters["weight"]
+        self.register_buffer("weight_fp8", q)
+        self.register_buffer("weight_scale", scale.to(w.device))
+        self.fp8 = True
+
+    def _linear(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fp8:
+            return ops.linear_fp8(x, self.weight_fp8, self.weight_scale,
+                                  getattr(self, "bias", None))
+        return ops.linear(x, self.weight, getattr(self, "bias", None))
+
+
 def _shard(dim_size: int, tp_size: int, tp_rank: int) -> tuple:
     """(start, size) of this rank's shard; dims that do not divide evenly are
     REPLICATED (e.g. odd vocab sizes in the lm_head) — the layer then behaves
