@@ -236,6 +236,7 @@ def main():
         wdtype = "fp8" if args.model in ("deepseek-v3", "kimi-k2") and use_gpu else "bf16"
     if wdtype == "fp8":
         eargs.moe_weight_dtype = "fp8"
+        eargs.linear_weight_dtype = "fp8"   # dense GEMM route too (BASELINE)
     engine = Engine(cfg, eargs, comm=comm, random_weights=True)
     if use_gpu:
         # pre-tune hipBLASLt algo picks for the Ms this run will see:

@@ -520,7 +520,12 @@ def linear_fp8(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
                bias: Optional[torch.Tensor] = None) -> torch.Tensor:
     """W8A8 dense GEMM through hipBLASLt fp8: x quantized per-tensor on the
     fly into persistent per-shape buffers (graph-capture safe: the scale and
-    xq pointers baked into a captured graph never move)."""
+    xq pointers baked into a captured graph never move). CPU fallback
+    dequantizes (plumbing tests)."""
+    if not x.is_cuda:
+        w = (w_q.float() * w_scale.float()).to(torch.float32)
+        y = torch.nn.functional.linear(x.float(), w).to(x.dtype)
+        return y if bias is None else y + bias
     ext = _require_ext("lt_linear_fp8")
     M, K = x.shape
     key = (M, K, x.device.index)

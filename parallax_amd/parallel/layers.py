@@ -57,7 +57,7 @@ def _shard(dim_size: int, tp_size: int, tp_rank: int) -> tuple:
     return per * tp_rank, per
 
 
-class ColumnParallelLinear(nn.Module):
+class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
     """Y = X W^T with W sharded along output features."""
 
     def __init__(self, in_features: int, out_features: int, bias: bool = False,
@@ -88,13 +88,13 @@ class ColumnParallelLinear(nn.Module):
             self.bias.data.copy_(b[start : start + per])
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = ops.linear(x, self.weight, self.bias)
+        y = self._linear(x)
         if self.gather_output and self.tp_size > 1:
             y = get_comm().tp_all_gather(y, dim=-1)
         return y
 
 
-class MergedColumnParallelLinear(nn.Module):
+class MergedColumnParallelLinear(nn.Module, _Fp8WeightMixin):
     """Several column-parallel projections fused into one GEMM (e.g. QKV or
     gate+up). Each sub-projection is sharded independently so per-rank layout is
     [q_shard | k_shard | v_shard]."""
@@ -135,13 +135,13 @@ class MergedColumnParallelLinear(nn.Module):
             off += per
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return ops.linear(x, self.weight, self.bias)
+        return self._linear(x)
 
     def split_output(self, y: torch.Tensor) -> List[torch.Tensor]:
         return list(torch.split(y, self.shard_sizes, dim=-1))
 
 
-class RowParallelLinear(nn.Module):
+class RowParallelLinear(nn.Module, _Fp8WeightMixin):
     """Y = X W^T with W sharded along input features; all-reduce on forward."""
 
     def __init__(self, in_features: int, out_features: int, bias: bool = False):
@@ -171,7 +171,10 @@ class RowParallelLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         comm = get_comm()
-        y = ops.linear(x, self.weight)
+        if self.fp8:
+            y = ops.linear_fp8(x, self.weight_fp8, self.weight_scale)
+        else:
+            y = ops.linear(x, self.weight)
         if not self.replicated:
             y = comm.tp_all_reduce(y)
         # after the all-reduce every rank holds the full sum; bias is added once

@@ -76,6 +76,7 @@ class EngineArgs:
     enable_graphs: bool = True          # hipGraph-captured decode forward
     kv_cache_dtype: str = "auto"        # "auto" (= engine dtype) | "fp8"(e4m3)
     moe_weight_dtype: str = "auto"      # "auto" (= engine dtype) | "fp8"(W8A8)
+    linear_weight_dtype: str = "auto"   # dense GEMMs: "auto" | "fp8"(W8A8)
 
 
 def partition_layers(num_layers: int, pp_size: int, pp_rank: int) -> Tuple[int, int]:
@@ -132,6 +133,21 @@ class Engine:
                     m.quantize_fp8()
                     n_q += 1
             logger.info("fp8 MoE: quantized %d expert blocks (W8A8)", n_q)
+        if args.linear_weight_dtype == "fp8":
+            from ..parallel.layers import (
+                ColumnParallelLinear, MergedColumnParallelLinear,
+                RowParallelLinear,
+            )
+
+            n_q = 0
+            for name, m in self.model.named_modules():
+                if isinstance(m, (ColumnParallelLinear,
+                                  MergedColumnParallelLinear,
+                                  RowParallelLinear)) \
+                        and "lm_head" not in name:
+                    m.quantize_fp8()
+                    n_q += 1
+            logger.info("fp8 dense: quantized %d linears (W8A8)", n_q)
 
         # hybrid stacks: paged KV only for the full-attention layers
         self.local_layer_types = [cfg.layer_type(g) for g in range(start, end)]
@@ -291,6 +307,18 @@ class Engine:
                     x = torch.zeros(m, p.shape[1], dtype=p.dtype,
                                     device=self.device)
                     ops.linear(x, p)
+                # fp8-quantized linears hold buffers, not parameters
+                for name, mod in self.model.named_modules():
+                    w = getattr(mod, "weight_fp8", None)
+                    if w is None or not getattr(mod, "fp8", False):
+                        continue
+                    key = (m, w.shape[0], w.shape[1], "fp8")
+                    if key in seen:
+                        continue
+                    seen.add(key)
+                    x = torch.zeros(m, w.shape[1], dtype=self.args.dtype,
+                                    device=self.device)
+                    ops.linear_fp8(x, w, mod.weight_scale)
         torch.cuda.synchronize()
         logger.info("gemm warmup: tuned %d shapes", len(seen))
 

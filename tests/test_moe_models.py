@@ -210,3 +210,38 @@ def test_fused_moe_fp8_cpu_fallback():
     got = moe(x)
     rel = (got - ref).norm() / ref.norm().clamp_min(1e-6)
     assert rel < 0.08, f"fp8 CPU fallback rel error {rel:.3f}"
+
+
+def test_fp8_dense_linear_cpu_fallback():
+    """linear_weight_dtype=fp8: dense linears quantize to per-tensor fp8;
+    CPU fallback dequantizes and stays close to the bf16 engine."""
+    import torch
+
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=128, hidden_size=64,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+        intermediate_size=128, max_position_embeddings=128, eos_token_ids=[],
+    )
+    ctx = CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                      pp_rank=0, tp_rank=0, device=torch.device("cpu"))
+    torch.manual_seed(0)
+    a = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=64,
+                               dtype=torch.float32), comm=ctx,
+               random_weights=True)
+    torch.manual_seed(0)
+    b = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=64,
+                               dtype=torch.float32,
+                               linear_weight_dtype="fp8"), comm=ctx,
+               random_weights=True)
+    # lm_head stays bf16/fp32; everything else quantized
+    assert not b.model.lm_head.fp8
+    assert b.model.layers[0].self_attn.qkv_proj.fp8
+    sp = [SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)]
+    out_a = list(a.generate([[5, 9, 13, 2]], sp).values())[0]
+    out_b = list(b.generate([[5, 9, 13, 2]], sp).values())[0]
+    assert len(out_a) == 4 and len(out_b) == 4  # runs end-to-end
