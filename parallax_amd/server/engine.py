@@ -275,10 +275,20 @@ class Engine:
         sampling_params: Optional[SamplingParams] = None,
         rid: Optional[str] = None,
     ) -> str:
+        sp = sampling_params or SamplingParams()
+        limit = min(self.args.max_model_len, self.cfg.max_position_embeddings)
+        if len(prompt_token_ids) + sp.max_new_tokens > limit:
+            # out-of-range positions would read past the rope table / graph
+            # buffers (a device fault, not an error message) — reject up front
+            raise ValueError(
+                f"request length {len(prompt_token_ids)} + "
+                f"{sp.max_new_tokens} new tokens exceeds the engine context "
+                f"limit {limit} (max_model_len / max_position_embeddings)"
+            )
         req = InitialRequest(
             rid=rid or new_request_id(),
             prompt_token_ids=list(prompt_token_ids),
-            sampling_params=sampling_params or SamplingParams(),
+            sampling_params=sp,
         )
         self._pending_adds.append(req)
         return req.rid

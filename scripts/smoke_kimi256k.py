@@ -27,7 +27,7 @@ def main():
         "routed_scaling_factor": 2.5, "norm_topk_prob": True,
         "first_k_dense_replace": 0, "q_lora_rank": 512, "kv_lora_rank": 512,
         "qk_nope_head_dim": 128, "qk_rope_head_dim": 64, "v_head_dim": 128,
-        "max_position_embeddings": 262144, "eos_token_id": None,
+        "max_position_embeddings": 262144 + 512, "eos_token_id": None,
     })
     t0 = time.time()
     eng = Engine(

@@ -172,7 +172,7 @@ def test_kimi_k2_256k_prefix_reuse_smoke():
         "routed_scaling_factor": 2.5, "norm_topk_prob": True,
         "first_k_dense_replace": 0, "q_lora_rank": 512, "kv_lora_rank": 512,
         "qk_nope_head_dim": 128, "qk_rope_head_dim": 64, "v_head_dim": 128,
-        "max_position_embeddings": 262144, "eos_token_id": None,
+        "max_position_embeddings": 262144 + 512, "eos_token_id": None,
     })
     CTX = 256 * 1024
     eng = Engine(
