@@ -141,7 +141,7 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
         const char* e = getenv("PARALLAX_LT_TUNE_CAP");
         return e ? (size_t)atoll(e) : (size_t)0;
       }();
-      const size_t cap = cap_env ? cap_env : ((M <= 1024) ? 2048 : 384);
+      const size_t cap = cap_env ? cap_env : ((M <= 1024) ? 100000 : 384);
       size_t kept = 0;
       const float alpha1 = 1.f, beta1 = 0.f;
       for (auto& r : all) {
@@ -187,12 +187,15 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
   // mid-serving (PARALLAX_LT_TUNE_MS, default 150 ms/shape).
   static const float kBudgetMs = [] {
     const char* e = getenv("PARALLAX_LT_TUNE_MS");
-    return e ? (float)atof(e) : 150.f;
+    return e ? (float)atof(e) : 1500.f;
   }();
+  // tiny-M shapes (prefill lm_head row counts) appear mid-serving with
+  // varying M: keep their one-time tuning stall small
+  const float budget_ms = (M <= 64) ? std::min(kBudgetMs, 150.f) : kBudgetMs;
   std::vector<std::pair<float, int>> screened;
   float spent_ms = 0.f;
   for (int i = 0; i < n_results; ++i) {
-    if (i >= n_heur && spent_ms > kBudgetMs) break;
+    if (i >= n_heur && spent_ms > budget_ms) break;
     const float ms = time_algo(i, 1);
     if (ms < 1e29f) spent_ms += 2.f * ms;  // warm + timed run
     screened.emplace_back(ms, i);
