@@ -141,7 +141,7 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
         const char* e = getenv("PARALLAX_LT_TUNE_CAP");
         return e ? (size_t)atoll(e) : (size_t)0;
       }();
-      const size_t cap = cap_env ? cap_env : ((M <= 1024) ? 100000 : 384);
+      const size_t cap = cap_env ? cap_env : (size_t)100000;  // budget-bound
       size_t kept = 0;
       const float alpha1 = 1.f, beta1 = 0.f;
       for (auto& r : all) {

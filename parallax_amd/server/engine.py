@@ -612,18 +612,20 @@ class Engine:
     # -- forward passes ------------------------------------------------------------------
 
     def _build_prefill_meta(self, chunks: List[PrefillChunk]) -> Tuple[ForwardMeta, torch.Tensor]:
-        positions, input_ids, slot_mapping = [], [], []
+        # vectorized host build: a 16k-token chunk used to cost ~10 ms of
+        # per-token Python loops per step — at 16 chunked-prefill steps that
+        # was a visible TTFT slice
+        bs = self.args.block_size
+        pos_parts, slot_parts, input_ids = [], [], []
         block_tables, seq_lens, query_lens, logits_idx = [], [], [], []
         t = 0
         for c in chunks:
             state = self.cache_manager.get(c.req.rid)
-            positions.extend(range(c.start, c.start + c.num_tokens))
+            pos = torch.arange(c.start, c.start + c.num_tokens, dtype=torch.int64)
+            bt = torch.tensor(state.block_table, dtype=torch.int64)
+            slot_parts.append(bt[pos // bs] * bs + pos % bs)
+            pos_parts.append(pos)
             input_ids.extend(c.req.prompt_token_ids[c.start : c.start + c.num_tokens])
-            slot_mapping.extend(
-                slot_mapping_for_positions(
-                    state.block_table, c.start, c.num_tokens, self.args.block_size
-                )
-            )
             block_tables.append(state.block_table)
             seq_lens.append(c.start + c.num_tokens)
             query_lens.append(c.num_tokens)
@@ -632,8 +634,8 @@ class Engine:
         dev = self.device
         meta = ForwardMeta(
             is_prefill=True,
-            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
-            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
+            positions=torch.cat(pos_parts).to(device=dev, dtype=torch.int32),
+            slot_mapping=torch.cat(slot_parts).to(dev),
             block_tables=build_block_table_tensor(block_tables, dev),
             seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
             query_lens=torch.tensor(query_lens, dtype=torch.int32, device=dev),
