@@ -150,10 +150,12 @@ def main():
     ap.add_argument("--steps", type=int, default=128)
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="deepseek-r1-distill-llama-8b", choices=MODELS)
-    ap.add_argument("--batch-per-gpu", type=int, default=512,
-                    help="decode batch per GPU (global batch = N * this); "
-                         "512 is the measured throughput knee (21 ms/step = "
-                         "47 tok/s per stream, see profiles/README.md)")
+    ap.add_argument("--batch-per-gpu", type=int, default=1024,
+                    help="decode batch per GPU (global batch = N * this). "
+                         "Measured sweep (profiles/README.md round 2): 128 -> "
+                         "15.6k tok/s @ 574 ms TTFT, 512 -> 27.1k @ 1.7 s, "
+                         "1024 -> 31.1k @ 3.2 s, 2048 -> 33.5k @ 6.3 s; 1024 "
+                         "is the default throughput/latency point")
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--block-size", type=int, default=32)
     ap.add_argument("--micro-batches", type=int, default=0,
