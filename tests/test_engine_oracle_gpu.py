@@ -75,6 +75,32 @@ FAMILIES = {
         "n_routed_experts": 8, "num_experts_per_tok": 2,
         "routed_scaling_factor": 1.0,
     },
+    "deepseek_v32": {
+        # tiny ctx stays under index_topk -> the dense fallback path, so the
+        # composition check is stable (the sparse kernels have their own
+        # numerics tests)
+        "architectures": ["DeepseekV32ForCausalLM"], "model_type": "deepseek_v32",
+        "vocab_size": 512, "hidden_size": 128, "num_hidden_layers": 2,
+        "num_attention_heads": 4, "num_key_value_heads": 4,
+        "intermediate_size": 256, "moe_intermediate_size": 64,
+        "n_routed_experts": 8, "num_experts_per_tok": 2, "n_shared_experts": 1,
+        "n_group": 2, "topk_group": 1, "routed_scaling_factor": 1.0,
+        "first_k_dense_replace": 1, "q_lora_rank": 64, "kv_lora_rank": 512,
+        "qk_nope_head_dim": 64, "qk_rope_head_dim": 64, "v_head_dim": 64,
+        "index_n_heads": 4, "index_head_dim": 64, "index_topk": 64,
+    },
+    "qwen3_next": {
+        "architectures": ["Qwen3NextForCausalLM"], "model_type": "qwen3_next",
+        "vocab_size": 512, "hidden_size": 128, "num_hidden_layers": 2,
+        "num_attention_heads": 4, "num_key_value_heads": 2, "head_dim": 64,
+        "intermediate_size": 256, "moe_intermediate_size": 64,
+        "num_experts": 8, "num_experts_per_tok": 2,
+        "shared_expert_intermediate_size": 64,
+        "linear_num_key_heads": 2, "linear_num_value_heads": 4,
+        "linear_key_head_dim": 32, "linear_value_head_dim": 32,
+        "linear_conv_kernel_dim": 4, "partial_rotary_factor": 0.25,
+        "layer_types": ["linear_attention", "full_attention"],
+    },
     "step3p5": {
         "architectures": ["Step3p5ForCausalLM"], "model_type": "step3p5",
         "vocab_size": 512, "hidden_size": 128, "num_hidden_layers": 2,
