@@ -493,6 +493,7 @@ def fused_moe_forward(
     topk_weights: torch.Tensor,  # [T, k] float
     activation: str = "silu",
     limit: float = 0.0,
+    bias_gate_up: Optional[torch.Tensor] = None,  # [E, 2I] bf16 (gpt-oss)
 ) -> torch.Tensor:
     """Grouped-GEMM MoE (device-side routing; graph-capture safe). Returns
     fp32 [T, H]."""
@@ -509,7 +510,9 @@ def fused_moe_forward(
     route_w = topk_weights.reshape(-1).float()[perm].contiguous()
     out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
     ext.moe_forward(out, x.contiguous(), w_gate_up, w_down, perm.contiguous(),
-                    seg.contiguous(), route_w, k, activation == "gelu", limit)
+                    seg.contiguous(), route_w, k, activation == "gelu", limit,
+                    bias_gate_up if bias_gate_up is not None
+                    else x.new_empty(0))
     return out
 
 

@@ -53,7 +53,7 @@ void launch_msa_paged_attention_decode(void*, const void*, const void*,
 void launch_build_moe_tiles(int*, int*, const int*, int, int, hipStream_t);
 void launch_moe_gate_up(void*, const void*, const void*, const int*,
                         const int*, const int64_t*, const int*, int, int, int,
-                        int, int, bool, float, hipStream_t);
+                        int, int, bool, float, const void*, hipStream_t);
 void launch_moe_down(void*, const void*, const void*, const float*, const int*,
                      const int*, const int64_t*, const int*, int, int, int,
                      int, int, hipStream_t);
@@ -413,7 +413,8 @@ void mla_paged_attention_decode(torch::Tensor out, torch::Tensor q_latent,
 void moe_forward(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
                  torch::Tensor w_down, torch::Tensor perm,
                  torch::Tensor seg_offsets, torch::Tensor route_w,
-                 int64_t topk, bool gelu, double limit) {
+                 int64_t topk, bool gelu, double limit,
+                 torch::Tensor bias_gu) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_BF16(x);
@@ -438,10 +439,14 @@ void moe_forward(torch::Tensor out, torch::Tensor x, torch::Tensor w_gu,
   auto stream = cur_stream();
   launch_build_moe_tiles(tile_expert.data_ptr<int>(), tile_row0.data_ptr<int>(),
                          seg_offsets.data_ptr<int>(), E, max_tiles, stream);
+  TORCH_CHECK(bias_gu.numel() == 0 ||
+              (bias_gu.scalar_type() == at::kBFloat16 &&
+               bias_gu.is_contiguous() && bias_gu.numel() == (int64_t)E * 2 * I));
   launch_moe_gate_up(h_buf.data_ptr(), x.data_ptr(), w_gu.data_ptr(),
                      tile_expert.data_ptr<int>(), tile_row0.data_ptr<int>(),
                      perm.data_ptr<int64_t>(), seg_offsets.data_ptr<int>(), E,
-                     (int)topk, H, I, max_tiles, gelu, (float)limit, stream);
+                     (int)topk, H, I, max_tiles, gelu, (float)limit,
+                     bias_gu.numel() ? bias_gu.data_ptr() : nullptr, stream);
   launch_moe_down(out.data_ptr(), h_buf.data_ptr(), w_down.data_ptr(),
                   route_w.data_ptr<float>(), tile_expert.data_ptr<int>(),
                   tile_row0.data_ptr<int>(), perm.data_ptr<int64_t>(),

@@ -52,7 +52,8 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_kernel(
     const int64_t* __restrict__ perm,      // [T*k] sorted-assignment -> flat idx
     const int* __restrict__ seg_offsets,   // [E+1]
     const int E, const int topk, const int H, const int I,
-    const float limit) {                   // >0: gpt-oss clamped act
+    const float limit,                     // >0: gpt-oss clamped act
+    const uint16_t* __restrict__ bias_gu = nullptr) {  // [E, 2I] or null
   const int tile = blockIdx.x;
   const int e = tile_expert[tile];
   if (e < 0) return;
@@ -106,6 +107,10 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_kernel(
       if (grow >= rows_end) continue;
       const int n = n0 + l15;              // column within I
       float g = acc_g[mt][r], u = acc_u[mt][r];
+      if (bias_gu != nullptr) {
+        g += bf16_bits_to_f32(bias_gu[(size_t)e * 2 * I + n]);
+        u += bf16_bits_to_f32(bias_gu[(size_t)e * 2 * I + I + n]);
+      }
       float a;
       if (limit > 0.f) {                   // gpt-oss clamped act
         g = fminf(g, limit);
@@ -448,16 +453,18 @@ extern "C" void launch_moe_gate_up(
     void* h_buf, const void* x, const void* w_gu, const int* tile_expert,
     const int* tile_row0, const int64_t* perm, const int* seg_offsets, int E,
     int topk, int H, int I, int max_tiles, bool gelu, float limit,
-    hipStream_t stream) {
+    const void* bias_gu, hipStream_t stream) {
   dim3 grid(max_tiles, ceil_div(I, 64), 1);
   if (gelu)
     moe_gate_up_kernel<true><<<grid, MOE_THREADS, 0, stream>>>(
         (uint16_t*)h_buf, (const uint16_t*)x, (const uint16_t*)w_gu,
-        tile_expert, tile_row0, perm, seg_offsets, E, topk, H, I, limit);
+        tile_expert, tile_row0, perm, seg_offsets, E, topk, H, I, limit,
+        (const uint16_t*)bias_gu);
   else
     moe_gate_up_kernel<false><<<grid, MOE_THREADS, 0, stream>>>(
         (uint16_t*)h_buf, (const uint16_t*)x, (const uint16_t*)w_gu,
-        tile_expert, tile_row0, perm, seg_offsets, E, topk, H, I, limit);
+        tile_expert, tile_row0, perm, seg_offsets, E, topk, H, I, limit,
+        (const uint16_t*)bias_gu);
 }
 
 extern "C" void launch_moe_down(
