@@ -15,11 +15,15 @@ import torch
 
 from parallax_amd import ops
 
-SHAPES = [  # (M, N, K, label) — DeepSeek-R1-Distill-Llama-8B decode @ batch 512
-    (512, 6144, 4096, "qkv"),
-    (512, 4096, 4096, "o_proj"),
-    (512, 28672, 4096, "gate_up"),
-    (512, 4096, 14336, "down"),
+SHAPES = [  # (M, N, K, label) — DeepSeek-R1-Distill-Llama-8B decode shapes
+    (512, 6144, 4096, "qkv@512"),
+    (512, 4096, 4096, "o_proj@512"),
+    (512, 28672, 4096, "gate_up@512"),
+    (512, 4096, 14336, "down@512"),
+    (1024, 6144, 4096, "qkv@1024"),
+    (1024, 4096, 4096, "o_proj@1024"),
+    (1024, 28672, 4096, "gate_up@1024"),
+    (1024, 4096, 14336, "down@1024"),
 ]
 
 
