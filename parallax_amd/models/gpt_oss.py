@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from .config import ModelConfig
 from .llama import LlamaDecoderLayer, LlamaShardModel
 from .registry import register_model
