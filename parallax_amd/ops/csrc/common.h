@@ -114,6 +114,27 @@ DEVINL float fp8_e4m3_to_f32(uint8_t b) {
   return static_cast<float>(v);
 }
 
+typedef float f32x2v __attribute__((ext_vector_type(2)));
+
+// 8 fp8-E4M3 bytes -> 8 bf16 (scaled): packed HW converts
+// (v_cvt_pk_f32_fp8 on dword halves) instead of 8 byte-extract + scalar
+// convert chains — the fp8-KV fragment-load path was VALU-bound on the
+// scalar version (profiles/README.md round 2 fp8-KV note).
+DEVINL void fp8x8_to_bf16x8(uint64_t raw, float scale, uint16_t* out8) {
+  const uint32_t lo = (uint32_t)raw;
+  const uint32_t hi = (uint32_t)(raw >> 32);
+  f32x2v f[4];
+  f[0] = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
+  f[1] = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
+  f[2] = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
+  f[3] = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    out8[2 * i] = f32_to_bf16_bits(f[i][0] * scale);
+    out8[2 * i + 1] = f32_to_bf16_bits(f[i][1] * scale);
+  }
+}
+
 DEVINL uint8_t f32_to_fp8_e4m3(float f) {
   __hip_fp8_e4m3 v(f);
   return v.__x;

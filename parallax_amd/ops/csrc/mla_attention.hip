@@ -121,14 +121,10 @@ __global__ __launch_bounds__(MLA_THREADS) void mla_decode_kernel(
         int4 val = make_int4(0, 0, 0, 0);
         if (ok) {
           if (KV_FP8) {
-            // 8 fp8 bytes -> 8 bf16 (one cache-wide scale)
+            // 8 fp8 bytes -> 8 bf16 (one cache-wide scale), packed converts
             const uint64_t raw = *reinterpret_cast<const uint64_t*>(
                 (const uint8_t*)cache_v + crow_tok * DK + d);
-            uint16_t* vs16 = reinterpret_cast<uint16_t*>(&val);
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              vs16[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * c_scale);
+            fp8x8_to_bf16x8(raw, c_scale, reinterpret_cast<uint16_t*>(&val));
           } else {
             val = *reinterpret_cast<const int4*>(
                 (const uint16_t*)cache_v + crow_tok * DK + d);

@@ -149,11 +149,8 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
         if (KV_FP8) {
           const uint64_t raw = *reinterpret_cast<const uint64_t*>(
               (const uint8_t*)k_cache_v + row + s * 32 + l4 * 8);
-          uint16_t* e = reinterpret_cast<uint16_t*>(&kreg[ms][s]);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            e[j] = f32_to_bf16_bits(
-                fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * k_scale);
+          fp8x8_to_bf16x8(raw, k_scale,
+                          reinterpret_cast<uint16_t*>(&kreg[ms][s]));
         } else {
           kreg[ms][s] = *reinterpret_cast<const int4*>(
               (const uint16_t*)k_cache_v + row + s * 32 + l4 * 8);
@@ -195,11 +192,8 @@ __global__ __launch_bounds__(DM_THREADS) void paged_decode_mfma_kernel(
           if (KV_FP8) {
             const uint64_t raw = *reinterpret_cast<const uint64_t*>(
                 (const uint8_t*)v_cache_v + row);
-            uint16_t* e = reinterpret_cast<uint16_t*>(&vreg[s][mt]);
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              e[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((raw >> (8 * j)) & 0xff) * v_scale);
+            fp8x8_to_bf16x8(raw, v_scale,
+                            reinterpret_cast<uint16_t*>(&vreg[s][mt]));
           } else {
             vreg[s][mt] = *reinterpret_cast<const int4*>(
                 (const uint16_t*)v_cache_v + row);

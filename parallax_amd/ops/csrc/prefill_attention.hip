@@ -125,11 +125,8 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
           if (KV_FP8) {
             const uint64_t kraw = *reinterpret_cast<const uint64_t*>(
                 (const uint8_t*)k_cache_v + row_off + stg_d + c * 8);
-            uint16_t* ks = reinterpret_cast<uint16_t*>(&kval);
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              ks[j] = f32_to_bf16_bits(
-                  fp8_e4m3_to_f32((kraw >> (8 * j)) & 0xff) * k_scale);
+            fp8x8_to_bf16x8(kraw, k_scale,
+                            reinterpret_cast<uint16_t*>(&kval));
           } else {
             kval = *reinterpret_cast<const int4*>(
                 (const uint16_t*)k_cache_v + row_off + stg_d + c * 8);
@@ -149,11 +146,8 @@ __global__ __launch_bounds__(PF_THREADS) void prefill_attention_kernel(
         if (KV_FP8) {
           const uint64_t vraw = *reinterpret_cast<const uint64_t*>(
               (const uint8_t*)v_cache_v + row);
-          uint16_t* vsp = reinterpret_cast<uint16_t*>(&vval);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vsp[j] = f32_to_bf16_bits(
-                fp8_e4m3_to_f32((vraw >> (8 * j)) & 0xff) * v_scale);
+          fp8x8_to_bf16x8(vraw, v_scale,
+                          reinterpret_cast<uint16_t*>(&vval));
         } else {
           vval = *reinterpret_cast<const int4*>(
               (const uint16_t*)v_cache_v + row);
