@@ -63,10 +63,13 @@ class PeerExecutor:
         random_weights: bool = False,
         model_path: Optional[str] = None,
         seed: int = 0,
+        refit_dir: Optional[str] = None,
     ):
         self.cfg = cfg
         self.peer_id = peer_id
         self.transport = transport
+        # version dirs for pushed weight refits (None = manifest pushes ignored)
+        self.refit_dir = refit_dir
         self.device = device or torch.device("cpu")
         self.dtype = dtype
         self.block_size = block_size
@@ -190,6 +193,13 @@ class PeerExecutor:
     # -- the step ----------------------------------------------------------------------
 
     def step(self, recv_timeout: float = 0.01) -> None:
+        if self._pending_refit_manifest is not None and self.refit_dir:
+            manifest, publisher = self._pending_refit_manifest
+            self._pending_refit_manifest = None
+            try:
+                self.refit_from_peer(publisher, manifest, self.refit_dir)
+            except Exception as e:  # keep serving on a failed refit
+                logger.error("pushed refit failed: %s", e)
         if self.is_head:
             self._head_step(recv_timeout)
         else:

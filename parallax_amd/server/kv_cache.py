@@ -1,13 +1,16 @@
 """Paged KV cache tensors, MI355X-first layout.
 
 Layout choice (differs from the reference's vLLM-packed Metal layout,
-src/parallax/server/cache/kv_cache.py:84-193): K and V both use
-``[num_blocks, num_kv_heads, block_size, head_dim]``. For one (block, head) the
-``block_size x head_dim`` tile is contiguous, so the decode kernel streams KV
-with 16 B/lane coalesced loads straight into LDS/registers, and prefill append
-(reshape_and_cache) writes each token's head vector contiguously. 288 GB HBM3E
-means block count is sized generously from a memory fraction rather than packed
-tightly.
+src/parallax/server/cache/kv_cache.py:84-193): K uses
+``[num_blocks, num_kv_heads, block_size, head_dim]`` (token-major rows: a
+QK^T A-fragment is 8 consecutive d of one token) and V is stored TRANSPOSED,
+``[num_blocks, num_kv_heads, head_dim, block_size]`` (d-major rows: a PV
+A-fragment is 8 consecutive tokens of one d) — so BOTH decode-attention MFMA
+operands are direct 16 B HBM loads with no LDS transpose staging, and the
+prefill kernel stages V^T as plain vectorized row copies. The cache write
+pays a per-element transpose scatter once per token instead. 288 GB HBM3E
+means block count is sized generously from a memory fraction rather than
+packed tightly.
 
 Also here: ``MLAKVCache`` — DeepSeek-style compressed latent cache (kv_lora_rank
 latent + rope dims per token, no per-head expansion), and ``LinearStateCache`` —

@@ -240,3 +240,48 @@ def test_refit_version_gc(tmp_path):
         fetcher.fetch(pub.manifest, pub.get_chunk)
     kept = sorted(os.listdir(str(tmp_path / "vers")))
     assert kept == ["v3", "v4", "v5"]
+
+
+def test_refit_manifest_push_triggers_reload(tmp_path):
+    """A pushed refit_manifest packet makes the peer fetch its shard's chunks
+    and hot-reload on its next step (reference heartbeat-triggered refit)."""
+    import threading
+
+    import msgpack
+
+    from parallax_amd.p2p.peer_executor import PeerExecutor
+    from parallax_amd.p2p.refit import RefitPublisher
+    from parallax_amd.p2p.transport import LoopbackTransport
+
+    ckpt = str(tmp_path / "ckpt")
+    cfg = _save_tiny_ckpt(ckpt, seed=11)
+    registry = {}
+    t_pub = LoopbackTransport("pub", registry)
+    t_sub = LoopbackTransport("sub", registry)
+    publisher = RefitPublisher(ckpt, version=3, chunk_size=512)
+    pub_peer = PeerExecutor(cfg, 0, 1, "pub", t_pub, random_weights=True)
+    pub_peer.set_refit_publisher(publisher)
+    sub_peer = PeerExecutor(cfg, 1, 2, "sub", t_sub, random_weights=True,
+                            refit_dir=str(tmp_path / "vers"))
+    before = {k: v.clone() for k, v in sub_peer.model.state_dict().items()}
+
+    stop = threading.Event()
+
+    def pump():
+        while not stop.is_set():
+            pub_peer.step(recv_timeout=0.02)
+
+    th = threading.Thread(target=pump, daemon=True)
+    th.start()
+    t_pub.send("sub", msgpack.packb(
+        {"kind": "refit_manifest", "manifest": publisher.manifest,
+         "publisher": "pub"}, use_bin_type=True))
+    sub_peer.step(recv_timeout=0.2)   # receive the manifest
+    sub_peer.step(recv_timeout=0.05)  # consume: fetch + reload
+    stop.set()
+    th.join(timeout=5)
+    after = sub_peer.model.state_dict()
+    assert any(not torch.equal(before[k], after[k]) for k in before)
+    import os as _os
+
+    assert _os.path.isdir(str(tmp_path / "vers" / "v3"))
