@@ -21,6 +21,10 @@ logger = get_logger("cli")
 def cmd_run(args) -> None:
     import uvicorn
 
+    from .utils.banner import print_banner
+
+    print_banner("scheduler")
+
     from .backend.service import SchedulerService, create_backend_app
 
     svc = SchedulerService()
@@ -69,6 +73,11 @@ def cmd_join(args) -> None:
 def cmd_serve(args) -> None:
     import torch
     import uvicorn
+
+    from .utils.banner import print_banner
+
+    if int(os.environ.get("RANK", "0")) == 0:
+        print_banner("serve")
 
     from .models.config import ModelConfig
     from .parallel.comm import init_distributed
