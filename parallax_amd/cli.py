@@ -151,6 +151,9 @@ def add_engine_args(p: argparse.ArgumentParser) -> None:
                    choices=["auto", "fp8"], help="fp8 = e4m3 KV storage")
     p.add_argument("--moe-weight-dtype", default="auto",
                    choices=["auto", "fp8"], help="fp8 = W8A8 expert weights")
+    p.add_argument("--linear-weight-dtype", default="auto",
+                   choices=["auto", "fp8"],
+                   help="fp8 = W8A8 dense GEMMs (lm_head stays full precision)")
     p.add_argument("--dtype", default="auto",
                    choices=["auto", "bfloat16", "float16", "float32"])
     p.add_argument("--micro-batches", type=int, default=0,
@@ -204,6 +207,7 @@ def engine_args_from_cli(args, world: int = 1):
         enable_graphs=not args.disable_graphs,
         kv_cache_dtype=args.kv_cache_dtype,
         moe_weight_dtype=args.moe_weight_dtype,
+        linear_weight_dtype=args.linear_weight_dtype,
     )
 
 
