@@ -35,7 +35,8 @@ from ..utils.logging_config import get_logger
 
 logger = get_logger("server.graph_runner")
 
-DEFAULT_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256]
+DEFAULT_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256,
+                   384, 512, 768]
 
 
 class DecodeGraphRunner:
