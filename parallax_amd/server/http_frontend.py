@@ -123,7 +123,10 @@ def create_app(
         prompt_ids = tokenizer.chat_prompt_ids(messages)
         sp = _params(body)
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
-        stream = server.submit(prompt_ids, sp, rid=rid)
+        try:
+            stream = server.submit(prompt_ids, sp, rid=rid)
+        except ValueError as e:  # over context limit etc -> clean client error
+            raise HTTPException(400, str(e))
         created = int(time.time())
 
         if body.get("stream"):
@@ -202,7 +205,10 @@ def create_app(
             prompt_ids = tokenizer.encode(prompt)
         sp = _params(body)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
-        stream = server.submit(prompt_ids, sp, rid=rid)
+        try:
+            stream = server.submit(prompt_ids, sp, rid=rid)
+        except ValueError as e:  # over context limit etc -> clean client error
+            raise HTTPException(400, str(e))
         token_ids, finish_reason, logprobs, stop_at = await _collect(stream, sp)
         text = tokenizer.decode(token_ids)
         if stop_at >= 0:
