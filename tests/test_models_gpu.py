@@ -113,3 +113,36 @@ def test_minimax_m2_engine_gpu():
     })
     assert cfg.qk_norm_full
     _run_twice(cfg)
+
+
+def test_constrained_decoding_gpu():
+    """json_schema masks compose with the fused gumbel sampler on GPU."""
+    import json
+
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+    from parallax_amd.server.tokenizer_util import TokenizerWrapper
+
+    tok = TokenizerWrapper(vocab_size=512)
+    vocab = tok.vocab_strings()
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=128,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=32,
+        intermediate_size=256, max_position_embeddings=512,
+        eos_token_ids=[tok.eos_token_id],
+    )
+    eng = Engine(cfg, EngineArgs(block_size=16, num_kv_blocks=64),
+                 random_weights=True)
+    eng.set_grammar_vocab(vocab)
+    schema = json.dumps({"type": "object", "properties": {
+        "a": {"type": "integer"}, "b": {"type": "boolean"}}})
+    sp = SamplingParams(temperature=1.0, max_new_tokens=150,
+                        json_schema=schema)
+    out = eng.generate([[5, 9, 13]], [sp])
+    toks = list(out.values())[0]
+    text = "".join(
+        vocab[t] for t in toks if t != tok.eos_token_id and t < len(vocab)
+    )
+    obj = json.loads(text)
+    assert isinstance(obj.get("a"), int) and isinstance(obj.get("b"), bool)
