@@ -56,13 +56,12 @@ def create_app(
         are matched here on the detokenized text (token-id stops happen in
         the engine) — on a hit the request is aborted and the text truncated
         (reference: frontend-side stop matching)."""
-        loop = asyncio.get_event_loop()
         token_ids: List[int] = []
         logprobs: List[float] = []
         finish_reason = "stop"
         stop_at: int = -1
         while True:
-            out = await loop.run_in_executor(None, stream.out_queue.get)
+            out = await stream.aget()
             if out is None:
                 break
             if out.token_id < 0:  # abort/timeout terminator
@@ -124,20 +123,22 @@ def create_app(
         sp = _params(body)
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         try:
-            stream = server.submit(prompt_ids, sp, rid=rid)
+            stream = server.submit(
+                prompt_ids, sp, rid=rid,
+                aio_loop=asyncio.get_running_loop(), aio_queue=asyncio.Queue(),
+            )
         except ValueError as e:  # over context limit etc -> clean client error
             raise HTTPException(400, str(e))
         created = int(time.time())
 
         if body.get("stream"):
             async def sse():
-                loop = asyncio.get_event_loop()
                 token_ids: List[int] = []
                 sent_len = 0
                 t_start = time.monotonic()
                 first_token_t = None
                 while True:
-                    out = await loop.run_in_executor(None, stream.out_queue.get)
+                    out = await stream.aget()
                     if out is None:
                         break
                     if first_token_t is None:
@@ -206,16 +207,61 @@ def create_app(
         sp = _params(body)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         try:
-            stream = server.submit(prompt_ids, sp, rid=rid)
+            stream = server.submit(
+                prompt_ids, sp, rid=rid,
+                aio_loop=asyncio.get_running_loop(), aio_queue=asyncio.Queue(),
+            )
         except ValueError as e:  # over context limit etc -> clean client error
             raise HTTPException(400, str(e))
+        created = int(time.time())
+
+        if body.get("stream"):
+            async def sse():
+                token_ids: List[int] = []
+                sent_len = 0
+                t_start = time.monotonic()
+                first_t = None
+                while True:
+                    out = await stream.aget()
+                    if out is None:
+                        break
+                    if first_t is None:
+                        first_t = time.monotonic()
+                    if out.token_id < 0:
+                        continue
+                    token_ids.append(out.token_id)
+                    text = tokenizer.decode(token_ids)
+                    delta, sent_len = text[sent_len:], len(text)
+                    chunk = {
+                        "id": rid, "object": "text_completion",
+                        "created": created, "model": model_name,
+                        "choices": [{
+                            "index": 0, "text": delta,
+                            "finish_reason": out.finish_reason
+                            if out.finished else None,
+                        }],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                elapsed = time.monotonic() - t_start
+                usage = {
+                    "prompt_tokens": len(prompt_ids),
+                    "completion_tokens": len(token_ids),
+                    "total_tokens": len(prompt_ids) + len(token_ids),
+                    "ttft_ms": round(((first_t or time.monotonic()) - t_start) * 1e3, 2),
+                    "tps": round(len(token_ids) / max(elapsed, 1e-6), 2),
+                }
+                yield f"data: {json.dumps({'id': rid, 'object': 'text_completion', 'created': created, 'model': model_name, 'choices': [], 'usage': usage})}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
         token_ids, finish_reason, logprobs, stop_at = await _collect(stream, sp)
         text = tokenizer.decode(token_ids)
         if stop_at >= 0:
             text = text[:stop_at]
             finish_reason = "stop"
         return JSONResponse({
-            "id": rid, "object": "text_completion", "created": int(time.time()),
+            "id": rid, "object": "text_completion", "created": created,
             "model": model_name,
             "choices": [{
                 "index": 0, "text": text,
