@@ -59,11 +59,12 @@ def create_head_app(
             prompt_ids = tokenizer.encode(prompt)
         routing_table: List[str] = body.get("routing_table") or [agent.node_id]
         sp = SamplingParams.from_openai(body)
-        rid, q = agent.submit(prompt_ids, sp, routing_table)
-        loop = asyncio.get_event_loop()
+        aq = asyncio.Queue()
+        rid, _ = agent.submit(prompt_ids, sp, routing_table,
+                              aio_loop=asyncio.get_running_loop(), aio_queue=aq)
         token_ids, finish_reason = [], "stop"
         while True:
-            out = await loop.run_in_executor(None, q.get)
+            out = await aq.get()
             if out is None:
                 break
             if out.token_id >= 0:
@@ -93,8 +94,9 @@ def create_head_app(
         routing_table: List[str] = body.get("routing_table") or [agent.node_id]
         prompt_ids = tokenizer.chat_prompt_ids(messages)
         sp = SamplingParams.from_openai(body)
-        rid, q = agent.submit(prompt_ids, sp, routing_table)
-        loop = asyncio.get_event_loop()
+        aq = asyncio.Queue()
+        rid, _ = agent.submit(prompt_ids, sp, routing_table,
+                              aio_loop=asyncio.get_running_loop(), aio_queue=aq)
         created = int(time.time())
 
         if body.get("stream"):
@@ -106,7 +108,7 @@ def create_head_app(
                 t0 = time.monotonic()
                 first_t = None
                 while True:
-                    out = await loop.run_in_executor(None, q.get)
+                    out = await aq.get()
                     if out is None:
                         break
                     if first_t is None:
@@ -150,7 +152,7 @@ def create_head_app(
         first_t = None
         stop_at = -1
         while True:
-            out = await loop.run_in_executor(None, q.get)
+            out = await aq.get()
             if out is None:
                 break
             if first_t is None:

@@ -24,6 +24,17 @@ from .transport import TcpTransport
 logger = get_logger("p2p.node_agent")
 
 
+class _AioBridgeQueue:
+    """Queue facade whose put() hops into an asyncio loop."""
+
+    def __init__(self, loop, aio_queue):
+        self.loop = loop
+        self.aio_queue = aio_queue
+
+    def put(self, item) -> None:
+        self.loop.call_soon_threadsafe(self.aio_queue.put_nowait, item)
+
+
 class NodeAgent:
     def __init__(
         self,
@@ -214,12 +225,18 @@ class NodeAgent:
     # -- head request API (used by the head HTTP frontend) ----------------------------
 
     def submit(self, prompt_ids: List[int], sp: SamplingParams,
-               routing_table: List[str]):
+               routing_table: List[str], aio_loop=None, aio_queue=None):
+        """aio_loop/aio_queue: asyncio delivery bridge — outputs are pushed
+        with call_soon_threadsafe so async consumers never block an executor
+        thread (see engine_server.RequestStream.deliver)."""
         import queue as queue_mod
 
         assert self.executor is not None and self.executor.is_head
         rid = new_request_id()
-        q: "queue_mod.Queue[Optional[PeerOutput]]" = queue_mod.Queue()
+        if aio_queue is not None:
+            q = _AioBridgeQueue(aio_loop, aio_queue)
+        else:
+            q = queue_mod.Queue()
         with self._lock:
             self._streams[rid] = q
             self.executor.submit(prompt_ids, sp, routing_table, rid=rid)
