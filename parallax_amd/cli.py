@@ -101,10 +101,14 @@ def cmd_serve(args) -> None:
     if torch.cuda.is_available():
         engine.warmup_gemms([args.max_num_tokens_per_batch,
                              args.max_batch_size])
+    tok = TokenizerWrapper(args.model_path, vocab_size=cfg.vocab_size)
+    if world > 1:
+        # constrained decoding masks logits on the SAMPLING rank (the last
+        # pipeline stage), so every rank needs the id->text table up front
+        engine.set_grammar_vocab(tok.vocab_strings())
     if comm.rank == 0:
         server = EngineServer(engine)
         server.start()
-        tok = TokenizerWrapper(args.model_path, vocab_size=cfg.vocab_size)
         app = create_app(server, tok, args.model_name or args.model or "model")
         uvicorn.run(app, host=args.host, port=args.port)
     else:
