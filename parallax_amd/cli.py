@@ -99,8 +99,7 @@ def cmd_serve(args) -> None:
                     random_weights=args.model_path is None,
                     lora_path=getattr(args, "lora_path", None))
     if torch.cuda.is_available():
-        engine.warmup_gemms([args.max_num_tokens_per_batch,
-                             args.max_batch_size])
+        engine.warmup_serving()
     tok = TokenizerWrapper(args.model_path, vocab_size=cfg.vocab_size)
     if world > 1:
         # constrained decoding masks logits on the SAMPLING rank (the last

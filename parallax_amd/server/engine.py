@@ -332,6 +332,26 @@ class Engine:
         torch.cuda.synchronize()
         logger.info("gemm warmup: tuned %d shapes", len(seen))
 
+    def warmup_serving(self) -> None:
+        """Pre-tune the GEMM algo picks for EVERY decode graph bucket plus the
+        prefill chunk shapes. Without this, serving pays a multi-second
+        hipBLASLt tuning stall the first time each batch-size bucket appears
+        (measured: 44 s p50 TTFT at request-rate 16 from exactly this)."""
+        hot = [self.args.max_num_tokens_per_batch,
+               self.args.prefill_chunk_size, self.args.max_batch_size]
+        minor = (list(self.graph_runner.buckets)
+                 if self.graph_runner is not None else [])
+        prev = os.environ.get("PARALLAX_LT_TUNE_MS")
+        self.warmup_gemms(hot)  # deep search for the steady-state shapes
+        try:
+            os.environ["PARALLAX_LT_TUNE_MS"] = "100"
+            self.warmup_gemms(minor)  # quick picks for transient buckets
+        finally:
+            if prev is None:
+                os.environ.pop("PARALLAX_LT_TUNE_MS", None)
+            else:
+                os.environ["PARALLAX_LT_TUNE_MS"] = prev
+
     def set_grammar_vocab(self, vocab: List[str]) -> None:
         """Enable json_schema constrained decoding: vocab[i] is token i's
         text. Must be called on every rank that samples (in practice: all)."""
