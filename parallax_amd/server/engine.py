@@ -335,8 +335,10 @@ class Engine:
     def warmup_serving(self) -> None:
         """Pre-tune the GEMM algo picks for EVERY decode graph bucket plus the
         prefill chunk shapes. Without this, serving pays a multi-second
-        hipBLASLt tuning stall the first time each batch-size bucket appears
-        (measured: 44 s p50 TTFT at request-rate 16 from exactly this)."""
+        hipBLASLt tuning stall the first time each batch-size bucket appears,
+        and a multi-second graph capture the first time each (batch, ctx)
+        bucket pair appears (measured: 44 s p50 TTFT at request-rate 16 from
+        exactly these two stalls compounding during the arrival burst)."""
         hot = [self.args.max_num_tokens_per_batch,
                self.args.prefill_chunk_size, self.args.max_batch_size]
         minor = (list(self.graph_runner.buckets)
@@ -351,6 +353,8 @@ class Engine:
                 os.environ.pop("PARALLAX_LT_TUNE_MS", None)
             else:
                 os.environ["PARALLAX_LT_TUNE_MS"] = prev
+        if self.graph_runner is not None:
+            self.graph_runner.capture_all()
 
     def set_grammar_vocab(self, vocab: List[str]) -> None:
         """Enable json_schema constrained decoding: vocab[i] is token i's

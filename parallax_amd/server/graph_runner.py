@@ -26,6 +26,7 @@ Capture-shape invariants:
 
 from __future__ import annotations
 
+import time
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -107,6 +108,7 @@ class DecodeGraphRunner:
         self._graphs: Dict[Tuple[int, int], torch.cuda.CUDAGraph] = {}
         self._outputs: Dict[Tuple[int, int], torch.Tensor] = {}
         self._pool = None
+        self._warmed: set = set()
         # block-table row cache: rows only change when a request crosses a
         # block boundary or the batch composition changes
         self._cached_rids: List[str] = []
@@ -155,8 +157,12 @@ class DecodeGraphRunner:
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
-            for _ in range(2):  # warmup (caching allocator, rocBLAS heuristics)
+            # warmup (caching allocator, rocBLAS heuristics) — per batch
+            # bucket only: GEMM shapes don't depend on the ctx bucket
+            n_warm = 0 if bucket in self._warmed else 2
+            for _ in range(n_warm):
                 self._forward(bucket, ctx_bucket)
+        self._warmed.add(bucket)
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
@@ -166,6 +172,20 @@ class DecodeGraphRunner:
             self._pool = graph.pool()
         self._graphs[(bucket, ctx_bucket)] = graph
         self._outputs[(bucket, ctx_bucket)] = out
+
+    def capture_all(self) -> None:
+        """Pre-capture every (batch bucket, ctx bucket) graph so serving never
+        stalls on an on-demand capture (a capture costs seconds; during a
+        request burst those stalls compound into tens of seconds of TTFT).
+        Large batches first: their allocations establish the memory pool's
+        high-water mark so later captures reuse it."""
+        t0 = time.monotonic()
+        for b in sorted(self.buckets, reverse=True):
+            for c in self.ctx_buckets:
+                if (b, c) not in self._graphs:
+                    self._capture(b, c)
+        logger.info("pre-captured %d decode graphs in %.1fs",
+                    len(self._graphs), time.monotonic() - t0)
 
     def bucket_for(self, batch: int) -> int:
         for b in self.buckets:
