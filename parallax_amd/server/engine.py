@@ -592,15 +592,19 @@ class Engine:
                 )
             payload_bytes = b""
             if self.comm.rank == 0:
-                if self._pending_adds or self._pending_aborts:
+                # snapshot: the HTTP thread may append concurrently; ranks
+                # must drain exactly what was serialized (SPMD determinism)
+                n_adds = len(self._pending_adds)
+                n_abrt = len(self._pending_aborts)
+                if n_adds or n_abrt:
                     payload_bytes = pickle.dumps(
                         (
                             [
                                 (r.rid, r.prompt_token_ids,
                                  r.sampling_params.to_dict())
-                                for r in self._pending_adds
+                                for r in self._pending_adds[:n_adds]
                             ],
-                            list(self._pending_aborts),
+                            self._pending_aborts[:n_abrt],
                         )
                     )
                 self._flag_buf.fill_(len(payload_bytes))
@@ -626,12 +630,23 @@ class Engine:
                         for rid, toks, sp in adds
                     ]
                     self._pending_aborts = list(aborts)
-        for req in self._pending_adds:
+        if self.comm.world_size > 1 and self.comm.rank == 0:
+            adds = self._pending_adds[:n_adds]
+            aborts = self._pending_aborts[:n_abrt]
+            del self._pending_adds[:n_adds]
+            del self._pending_aborts[:n_abrt]
+        else:
+            # lock-free handoff from the submit thread: snapshot the count,
+            # consume exactly that many (appends racing in stay for next step)
+            n_a, n_b = len(self._pending_adds), len(self._pending_aborts)
+            adds = self._pending_adds[:n_a]
+            aborts = self._pending_aborts[:n_b]
+            del self._pending_adds[:n_a]
+            del self._pending_aborts[:n_b]
+        for req in adds:
             self.scheduler.add_request(req)
-        for rid in self._pending_aborts:
+        for rid in aborts:
             self.scheduler.abort_request(rid)
-        self._pending_adds = []
-        self._pending_aborts = []
 
     # -- forward passes ------------------------------------------------------------------
 

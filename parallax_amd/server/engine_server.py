@@ -93,24 +93,24 @@ class EngineServer:
     ) -> RequestStream:
         rid = rid or new_request_id()
         stream = RequestStream(rid=rid, aio_loop=aio_loop, aio_queue=aio_queue)
-        with self._lock:
-            self._streams[rid] = stream
-            self.engine.submit(prompt_token_ids, sampling_params, rid=rid)
-            self.total_requests += 1
+        # lock-free: engine.submit is validation + a list append the step
+        # loop drains with snapshot semantics. Taking the step lock here
+        # would block the HTTP event loop for a full engine step (up to
+        # ~1 s during a prefill chunk) on EVERY arriving request.
+        self._streams[rid] = stream
+        self.engine.submit(prompt_token_ids, sampling_params, rid=rid)
+        self.total_requests += 1
         return stream
 
     def abort(self, rid: str) -> None:
-        with self._lock:
-            self.engine.abort(rid)
+        self.engine.abort(rid)
 
     # -- the loop -----------------------------------------------------------------
 
     def _run_loop(self) -> None:
         logger.info("engine step loop running")
         while not self._stop.is_set():
-            with self._lock:
-                has_work = self.engine.has_work
-            if not has_work:
+            if not self.engine.has_work:
                 time.sleep(self.idle_sleep_s)
                 continue
             t0 = time.monotonic()
@@ -127,8 +127,7 @@ class EngineServer:
                     stream.deliver(out)
                     if out.finished:
                         stream.deliver(None)
-                        with self._lock:
-                            self._streams.pop(out.rid, None)
+                        self._streams.pop(out.rid, None)
         logger.info("engine step loop stopped")
 
     def stats(self) -> dict:

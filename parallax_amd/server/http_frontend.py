@@ -33,9 +33,12 @@ def create_app(
 
     def _params(body: dict) -> SamplingParams:
         sp = SamplingParams.from_openai(body)
-        if tokenizer.eos_token_id is not None and not sp.stop_token_ids:
+        if (tokenizer.eos_token_id is not None and not sp.stop_token_ids
+                and not sp.ignore_eos):
             # default stop tokens to the tokenizer's EOS (it may differ from the
-            # model-config eos_token_ids the engine applies)
+            # model-config eos_token_ids the engine applies); ignore_eos
+            # requests must stay stop-free or they lose async decode
+            # pipelining (Engine._async_eligible)
             sp.stop_token_ids = [tokenizer.eos_token_id]
         if sp.json_schema and server.engine.sampler.grammar_vocab is None:
             # one-time id->text table for constrained decoding
