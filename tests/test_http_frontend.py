@@ -160,3 +160,49 @@ def test_tokenizer_wrapper_synthetic_roundtrip():
     assert text  # decodable
     chat_ids = tok.chat_prompt_ids([{"role": "user", "content": "hi"}])
     assert chat_ids and all(0 <= i < 128 for i in chat_ids)
+
+
+def test_concurrent_submit_storm():
+    """Lock-free submit: many threads submitting while the step loop runs must
+    neither lose requests (snapshot ingress drain) nor deadlock."""
+    import threading
+
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=64,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+        intermediate_size=128, max_position_embeddings=512,
+        eos_token_ids=[2],
+    )
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=256,
+                                 dtype=torch.float32), random_weights=True)
+    server = EngineServer(eng)
+    server.start()
+    try:
+        results = {}
+
+        def worker(wid):
+            sp = SamplingParams(temperature=0.0, max_new_tokens=4,
+                                ignore_eos=True)
+            streams = [server.submit(list(range(3, 11)), sp,
+                                     rid=f"w{wid}-{i}") for i in range(8)]
+            got = 0
+            for st in streams:
+                while True:
+                    out = st.out_queue.get(timeout=30)
+                    if out is None:
+                        break
+                    got += 1
+            results[wid] = got
+
+        ths = [threading.Thread(target=worker, args=(w,)) for w in range(6)]
+        for t in ths:
+            t.start()
+        for t in ths:
+            t.join(timeout=60)
+        assert all(not t.is_alive() for t in ths)
+        # every one of the 6*8 requests produced exactly max_new_tokens
+        assert results == {w: 8 * 4 for w in range(6)}
+    finally:
+        server.stop()
