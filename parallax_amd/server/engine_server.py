@@ -23,6 +23,14 @@ from .sampling_params import SamplingParams
 logger = get_logger("server.engine_server")
 
 
+def _deliver_batch(items) -> None:
+    """Runs inside the consumer event loop: fan a step's outputs out to the
+    per-request asyncio queues (queue waiter wakeups are loop-internal and
+    ~10x cheaper than cross-thread signalling)."""
+    for q, item in items:
+        q.put_nowait(item)
+
+
 @dataclass
 class RequestStream:
     rid: str
@@ -120,14 +128,29 @@ class EngineServer:
             self.step_ms_ewma = dt_ms if self.step_ms_ewma is None \
                 else 0.1 * dt_ms + 0.9 * self.step_ms_ewma
             self.last_batch_tokens = len(outputs)
+            # batch delivery: ONE call_soon_threadsafe per (step, loop), not
+            # one per token — per-token signalling (lock + self-pipe write
+            # ~15 us each) capped the whole server at ~3k tok/s when the
+            # engine produces 30k+
+            aio_batches: Dict[object, list] = {}
             for out in outputs:
                 self.total_output_tokens += 1
                 stream = self._streams.get(out.rid)
-                if stream is not None:
-                    stream.deliver(out)
+                if stream is None:
+                    continue
+                if stream.aio_queue is not None and stream.aio_loop is not None:
+                    items = aio_batches.setdefault(stream.aio_loop, [])
+                    items.append((stream.aio_queue, out))
                     if out.finished:
-                        stream.deliver(None)
+                        items.append((stream.aio_queue, None))
                         self._streams.pop(out.rid, None)
+                else:
+                    stream.out_queue.put(out)
+                    if out.finished:
+                        stream.out_queue.put(None)
+                        self._streams.pop(out.rid, None)
+            for loop, items in aio_batches.items():
+                loop.call_soon_threadsafe(_deliver_batch, items)
         logger.info("engine step loop stopped")
 
     def stats(self) -> dict:

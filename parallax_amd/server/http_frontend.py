@@ -140,24 +140,44 @@ def create_app(
                 sent_len = 0
                 t_start = time.monotonic()
                 first_token_t = None
-                while True:
+                done = False
+                while not done:
                     out = await stream.aget()
                     if out is None:
                         break
+                    # greedy drain: under load several tokens are already
+                    # queued — emit them as ONE chunk so the event loop does
+                    # O(wakeups) work, not O(tokens) (the per-token SSE path
+                    # caps a single asyncio loop at ~3k tok/s)
+                    batch = [out]
+                    while True:
+                        try:
+                            nxt = stream.aio_queue.get_nowait()
+                        except asyncio.QueueEmpty:
+                            break
+                        if nxt is None:
+                            done = True
+                            break
+                        batch.append(nxt)
                     if first_token_t is None:
                         first_token_t = time.monotonic()
-                    if out.token_id < 0:
-                        continue
-                    token_ids.append(out.token_id)
+                    finish = None
+                    for o in batch:
+                        if o.token_id >= 0:
+                            token_ids.append(o.token_id)
+                        if o.finished:
+                            finish = o.finish_reason or "stop"
                     text = tokenizer.decode(token_ids)
                     delta, sent_len = text[sent_len:], len(text)
+                    if not delta and finish is None:
+                        continue
                     chunk = {
                         "id": rid, "object": "chat.completion.chunk",
                         "created": created, "model": model_name,
                         "choices": [{
                             "index": 0,
                             "delta": {"content": delta},
-                            "finish_reason": out.finish_reason if out.finished else None,
+                            "finish_reason": finish,
                         }],
                     }
                     yield f"data: {json.dumps(chunk)}\n\n"
@@ -224,24 +244,39 @@ def create_app(
                 sent_len = 0
                 t_start = time.monotonic()
                 first_t = None
-                while True:
+                done = False
+                while not done:
                     out = await stream.aget()
                     if out is None:
                         break
+                    batch = [out]  # greedy drain (see chat_completions)
+                    while True:
+                        try:
+                            nxt = stream.aio_queue.get_nowait()
+                        except asyncio.QueueEmpty:
+                            break
+                        if nxt is None:
+                            done = True
+                            break
+                        batch.append(nxt)
                     if first_t is None:
                         first_t = time.monotonic()
-                    if out.token_id < 0:
-                        continue
-                    token_ids.append(out.token_id)
+                    finish = None
+                    for o in batch:
+                        if o.token_id >= 0:
+                            token_ids.append(o.token_id)
+                        if o.finished:
+                            finish = o.finish_reason or "stop"
                     text = tokenizer.decode(token_ids)
                     delta, sent_len = text[sent_len:], len(text)
+                    if not delta and finish is None:
+                        continue
                     chunk = {
                         "id": rid, "object": "text_completion",
                         "created": created, "model": model_name,
                         "choices": [{
                             "index": 0, "text": delta,
-                            "finish_reason": out.finish_reason
-                            if out.finished else None,
+                            "finish_reason": finish,
                         }],
                     }
                     yield f"data: {json.dumps(chunk)}\n\n"
