@@ -136,6 +136,7 @@ def create_app(
 
         if body.get("stream"):
             async def sse():
+              try:
                 token_ids: List[int] = []
                 sent_len = 0
                 t_start = time.monotonic()
@@ -192,6 +193,10 @@ def create_app(
                 }
                 yield f"data: {json.dumps({'id': rid, 'object': 'chat.completion.chunk', 'created': created, 'model': model_name, 'choices': [], 'usage': usage})}\n\n"
                 yield "data: [DONE]\n\n"
+              except asyncio.CancelledError:
+                # client went away mid-stream: stop generating for it
+                server.abort(rid)
+                raise
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
@@ -240,6 +245,7 @@ def create_app(
 
         if body.get("stream"):
             async def sse():
+              try:
                 token_ids: List[int] = []
                 sent_len = 0
                 t_start = time.monotonic()
@@ -290,6 +296,9 @@ def create_app(
                 }
                 yield f"data: {json.dumps({'id': rid, 'object': 'text_completion', 'created': created, 'model': model_name, 'choices': [], 'usage': usage})}\n\n"
                 yield "data: [DONE]\n\n"
+              except asyncio.CancelledError:
+                server.abort(rid)
+                raise
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
