@@ -148,7 +148,11 @@ async def run_benchmark(base_url: str, num_prompts: int, request_rate: float,
     rng = random.Random(seed)
     results: List[RequestResult] = []
     t_start = time.perf_counter()
-    async with httpx.AsyncClient(timeout=600.0) as client:
+    # default pool caps at 100 connections: with streaming responses held
+    # open for the whole generation, that silently throttles the benchmark to
+    # 100 concurrent requests and inflates TTFT by the queueing delay
+    limits = httpx.Limits(max_connections=None, max_keepalive_connections=None)
+    async with httpx.AsyncClient(timeout=600.0, limits=limits) as client:
         tasks = []
         for prompt, olen in zip(prompts, out_lens):
             tasks.append(asyncio.create_task(
