@@ -189,6 +189,11 @@ hipblasLtMatmulAlgo_t tune_shape(int64_t M, int64_t N, int64_t K,
   // the hot shapes a deep search and minor graph buckets a quick one
   const char* be = getenv("PARALLAX_LT_TUNE_MS");
   const float kBudgetMs = be ? (float)atof(be) : 1500.f;
+  // budget 0 = steady-state serving: a previously-unseen shape (arbitrary
+  // prefill token counts) takes the heuristic's top pick with NO timing
+  // loop — a mid-serving tuning stall is worse than a slightly sub-optimal
+  // algo on a shape that appears once
+  if (kBudgetMs <= 0.f) return results[0].algo;
   // tiny-M shapes (prefill lm_head row counts) appear mid-serving with
   // varying M: keep their one-time tuning stall small
   const float budget_ms = (M <= 64) ? std::min(kBudgetMs, 150.f) : kBudgetMs;
