@@ -127,8 +127,9 @@ def test_constrained_decoding_gpu():
     tok = TokenizerWrapper(vocab_size=512)
     vocab = tok.vocab_strings()
     cfg = ModelConfig(
-        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=128,
-        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=32,
+        # head_dim 64: the HIP prefill kernel supports head_dim 64/128 only
+        architecture="LlamaForCausalLM", vocab_size=512, hidden_size=256,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=64,
         intermediate_size=256, max_position_embeddings=512,
         eos_token_ids=[tok.eos_token_id],
     )
