@@ -7,6 +7,7 @@ restarts the executors, :757-874). Fresh design over HTTP + TcpTransport.
 
 from __future__ import annotations
 
+import os
 import threading
 import time
 from typing import Dict, List, Optional
@@ -137,6 +138,12 @@ class NodeAgent:
 
     def start(self) -> None:
         assert self.executor is not None, "join() first"
+        if self.device is not None and self.device.type == "cuda":
+            # WAN nodes run the eager forward path (no graph runner): cap the
+            # one-time per-shape hipBLASLt search so a previously-unseen
+            # prefill shape stalls the pipeline stage by ~0.1 s, not ~1.5 s
+            # (RPC latency dominates end-to-end here anyway)
+            os.environ.setdefault("PARALLAX_LT_TUNE_MS", "100")
         t1 = threading.Thread(target=self._step_loop, daemon=True, name="peer-step")
         t2 = threading.Thread(target=self._heartbeat_loop, daemon=True,
                               name="peer-heartbeat")
