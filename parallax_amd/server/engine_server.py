@@ -106,7 +106,11 @@ class EngineServer:
         # would block the HTTP event loop for a full engine step (up to
         # ~1 s during a prefill chunk) on EVERY arriving request.
         self._streams[rid] = stream
-        self.engine.submit(prompt_token_ids, sampling_params, rid=rid)
+        try:
+            self.engine.submit(prompt_token_ids, sampling_params, rid=rid)
+        except Exception:
+            self._streams.pop(rid, None)  # rejected: don't leak the stream
+            raise
         self.total_requests += 1
         return stream
 
