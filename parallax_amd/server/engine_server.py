@@ -121,10 +121,22 @@ class EngineServer:
 
     def _run_loop(self) -> None:
         logger.info("engine step loop running")
+        # multi-rank engines: non-head ranks block inside step()'s ingress
+        # broadcast while idle. Step periodically even with no work so the
+        # RCCL collective completes well inside any watchdog timeout (an
+        # empty step costs one 8-byte flag broadcast).
+        heartbeat_s = 5.0 if self.engine.comm.world_size > 1 else None
+        last_step = time.monotonic()
         while not self._stop.is_set():
             if not self.engine.has_work:
+                if (heartbeat_s is not None
+                        and time.monotonic() - last_step > heartbeat_s):
+                    with self._lock:
+                        self.engine.step()
+                    last_step = time.monotonic()
                 time.sleep(self.idle_sleep_s)
                 continue
+            last_step = time.monotonic()
             t0 = time.monotonic()
             with self._lock:
                 outputs = self.engine.step()
