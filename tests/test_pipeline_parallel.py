@@ -176,3 +176,65 @@ def test_pp4_decode_path_is_object_free(tmp_path):
     got = torch.load(out_file)
     assert got["object_sync_count"] == 1  # the initial adds payload, nothing else
     assert sorted(got["outs"]) == expected
+
+
+def _pp_mid_stream_worker(rank, world, port, out_file):
+    """Rank 0 submits requests from a SEPARATE THREAD while the step loop
+    runs (the serving ingress pattern): the snapshot drain must replicate
+    exactly the serialized adds to the other ranks. Every rank runs the SAME
+    fixed number of steps (idle steps are just the flag broadcast), so the
+    ranks stay in lockstep regardless of when the submits land."""
+    os.environ.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+        RANK=str(rank), WORLD_SIZE=str(world),
+    )
+    import threading
+    import time as _time
+
+    import torch as _t
+
+    from parallax_amd.parallel.comm import init_distributed
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    comm = init_distributed(pp_size=world, tp_size=1, backend="gloo",
+                            device=_t.device("cpu"))
+    cfg = tiny_cfg()
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                 dtype=_t.float32), comm=comm)
+    for name, t in full_state_dict(cfg).items():
+        eng.model.load_hf_weight(name, t)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)
+    N, STEPS = 12, 300
+
+    th = None
+    if rank == 0:
+        def submitter():
+            for i in range(N):
+                eng.submit(PROMPTS[i % len(PROMPTS)], sp, rid=f"m{i}")
+                _time.sleep(0.002)  # land mid-step on purpose
+
+        th = threading.Thread(target=submitter)
+        th.start()
+    counts = {}
+    for _ in range(STEPS):
+        for out in eng.step():
+            if out.token_id >= 0:
+                counts[out.rid] = counts.get(out.rid, 0) + 1
+    if th is not None:
+        th.join()
+    if rank == 0:
+        _t.save(counts, out_file)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_pp_mid_stream_submission(tmp_path):
+    """Serving ingress under PP: threaded submits racing the step loop."""
+    out_file = str(tmp_path / "mid_stream.pt")
+    mp.spawn(_pp_mid_stream_worker, args=(2, 29655, out_file), nprocs=2,
+             join=True)
+    counts = torch.load(out_file)
+    assert len(counts) == 12 and all(v == 4 for v in counts.values())
