@@ -168,6 +168,8 @@ def add_engine_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--request-timeout", type=float, default=600.0,
                    help="per-request abort timeout (s)")
     p.add_argument("--seed", type=int, default=0, help="sampling seed")
+    p.add_argument("--enable-routing-stats", action="store_true",
+                   help="per-expert MoE routing counters (GET /stats)")
     p.add_argument("--start-layer", type=int, default=None,
                    help="explicit layer range start (decentralized mode)")
     p.add_argument("--end-layer", type=int, default=None)
@@ -209,6 +211,7 @@ def engine_args_from_cli(args, world: int = 1):
         max_model_len=args.max_model_len,
         enable_graphs=not args.disable_graphs,
         kv_cache_dtype=args.kv_cache_dtype,
+        enable_routing_stats=args.enable_routing_stats,
         moe_weight_dtype=args.moe_weight_dtype,
         linear_weight_dtype=args.linear_weight_dtype,
     )

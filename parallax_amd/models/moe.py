@@ -110,6 +110,17 @@ class FusedMoE(nn.Module):
         self.w_down = nn.Parameter(
             torch.empty(self.num_local_experts, H, inter), requires_grad=False
         )
+        # expert-routing observability (reference parity:
+        # vllm_executor enable_return_routed_experts): persistent GLOBAL
+        # per-expert token counters, graph-safe (fixed-shape scatter_add
+        # on a buffer that exists before capture). None = disabled, free.
+        self.routing_counts: Optional[torch.Tensor] = None
+
+    def enable_routing_stats(self) -> None:
+        self.routing_counts = torch.zeros(
+            self.cfg.num_experts + 1, dtype=torch.int64,
+            device=self.w_gate_up.device,
+        )
 
     # -- fp8 (W8A8) path ---------------------------------------------------------
 
@@ -183,6 +194,11 @@ class FusedMoE(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
         topk_ids, topk_w = self.router(x)           # [T,k] global expert ids
+        if self.routing_counts is not None:
+            self.routing_counts.scatter_add_(
+                0, topk_ids.reshape(-1),
+                torch.ones_like(topk_ids.reshape(-1)),
+            )
         if self.ep_size > 1:
             # remap to local ids; foreign experts get the sentinel id
             # num_local_experts, which the grouped-GEMM segment table and the

@@ -76,6 +76,7 @@ class EngineArgs:
     enable_graphs: bool = True          # hipGraph-captured decode forward
     kv_cache_dtype: str = "auto"        # "auto" (= engine dtype) | "fp8"(e4m3)
     moe_weight_dtype: str = "auto"      # "auto" (= engine dtype) | "fp8"(W8A8)
+    enable_routing_stats: bool = False  # per-expert MoE routing counters
     linear_weight_dtype: str = "auto"   # dense GEMMs: "auto" | "fp8"(W8A8)
 
 
@@ -133,6 +134,12 @@ class Engine:
                     m.quantize_fp8()
                     n_q += 1
             logger.info("fp8 MoE: quantized %d expert blocks (W8A8)", n_q)
+        if args.enable_routing_stats:
+            from ..models.moe import FusedMoE
+
+            for m in self.model.modules():
+                if isinstance(m, FusedMoE):
+                    m.enable_routing_stats()  # before any graph capture
         if args.linear_weight_dtype == "fp8":
             from ..parallel.layers import (
                 ColumnParallelLinear, MergedColumnParallelLinear,
@@ -295,6 +302,19 @@ class Engine:
 
     def abort(self, rid: str) -> None:
         self._pending_aborts.append(rid)
+
+    def routing_stats(self) -> Optional[Dict[str, List[int]]]:
+        """Per-expert routed-token counts per MoE layer (None if disabled).
+        Reference parity: enable_return_routed_experts observability."""
+        if not self.args.enable_routing_stats:
+            return None
+        from ..models.moe import FusedMoE
+
+        out: Dict[str, List[int]] = {}
+        for name, m in self.model.named_modules():
+            if isinstance(m, FusedMoE) and m.routing_counts is not None:
+                out[name] = m.routing_counts[:-1].tolist()
+        return out
 
     def warmup_gemms(self, ms: List[int]) -> None:
         """Pre-tune the hipBLASLt algo picks for every 2-D weight shape at

@@ -181,4 +181,5 @@ class EngineServer:
             if self.step_ms_ewma is not None else None,
             "last_batch_tokens": self.last_batch_tokens,
             "engine_steps": self.engine.step_count,
+            "moe_routing": self.engine.routing_stats(),
         }
