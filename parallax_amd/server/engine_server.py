@@ -74,6 +74,12 @@ class EngineServer:
         self.total_output_tokens = 0
         self.step_ms_ewma: Optional[float] = None
         self.last_batch_tokens = 0
+        # on-demand tracing (the reference has no tracer; here a
+        # chrome trace of the live step loop is one POST /profile away)
+        self._profile_req: Optional[tuple] = None  # (steps, path)
+        self._profiler = None
+        self._profile_steps_left = 0
+        self.last_trace_path: Optional[str] = None
 
     # -- lifecycle ------------------------------------------------------------
 
@@ -117,6 +123,11 @@ class EngineServer:
     def abort(self, rid: str) -> None:
         self.engine.abort(rid)
 
+    def profile_next(self, steps: int, path: str) -> None:
+        """Trace the next `steps` engine steps into a chrome trace at
+        `path` (picked up by the step loop; safe to call while serving)."""
+        self._profile_req = (max(1, int(steps)), path)
+
     # -- the loop -----------------------------------------------------------------
 
     def _run_loop(self) -> None:
@@ -137,9 +148,33 @@ class EngineServer:
                 time.sleep(self.idle_sleep_s)
                 continue
             last_step = time.monotonic()
+            if self._profile_req is not None and self._profiler is None:
+                steps, path = self._profile_req
+                self._profile_req = None
+                import torch
+                from torch.profiler import ProfilerActivity, profile
+
+                acts = [ProfilerActivity.CPU]
+                if torch.cuda.is_available():
+                    acts.append(ProfilerActivity.CUDA)
+                self._profiler = profile(activities=acts)
+                self._profiler.__enter__()
+                self._profile_steps_left = steps
+                self._trace_path = path
             t0 = time.monotonic()
             with self._lock:
                 outputs = self.engine.step()
+            if self._profiler is not None:
+                self._profile_steps_left -= 1
+                if self._profile_steps_left <= 0:
+                    self._profiler.__exit__(None, None, None)
+                    try:
+                        self._profiler.export_chrome_trace(self._trace_path)
+                        self.last_trace_path = self._trace_path
+                        logger.info("chrome trace written to %s",
+                                    self._trace_path)
+                    finally:
+                        self._profiler = None
             dt_ms = (time.monotonic() - t0) * 1e3
             self.step_ms_ewma = dt_ms if self.step_ms_ewma is None \
                 else 0.1 * dt_ms + 0.9 * self.step_ms_ewma

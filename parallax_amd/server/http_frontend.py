@@ -106,6 +106,14 @@ def create_app(
     async def stats():
         return server.stats()
 
+    @app.post("/profile")
+    async def profile(request: Request):
+        """Trace the next N engine steps to a chrome trace file."""
+        body = await request.json()
+        path = body.get("path", "/tmp/parallax_trace.json")
+        server.profile_next(int(body.get("steps", 10)), path)
+        return {"tracing_next_steps": int(body.get("steps", 10)), "path": path}
+
     @app.post("/update_weights")
     async def update_weights(request: Request):
         body = await request.json()

@@ -206,3 +206,24 @@ def test_concurrent_submit_storm():
         assert results == {w: 8 * 4 for w in range(6)}
     finally:
         server.stop()
+
+
+def test_profile_endpoint(client, tmp_path):
+    """POST /profile traces the next engine steps into a chrome trace."""
+    import time as _time
+
+    path = str(tmp_path / "trace.json")
+    r = client.post("/profile", json={"steps": 2, "path": path})
+    assert r.status_code == 200
+    # drive some steps through the engine
+    r = client.post("/v1/completions", json={
+        "prompt": [5, 9, 13], "max_tokens": 4, "temperature": 0.0,
+        "ignore_eos": True,
+    })
+    assert r.status_code == 200
+    import os as _os
+    for _ in range(100):
+        if _os.path.exists(path):
+            break
+        _time.sleep(0.05)
+    assert _os.path.exists(path) and _os.path.getsize(path) > 100
