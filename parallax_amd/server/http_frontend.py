@@ -19,7 +19,7 @@ from fastapi.responses import JSONResponse, StreamingResponse
 from ..utils.logging_config import get_logger
 from .engine_server import EngineServer
 from .sampling_params import SamplingParams
-from .tokenizer_util import TokenizerWrapper
+from .tokenizer_util import IncrementalDetokenizer, TokenizerWrapper
 
 logger = get_logger("server.http_frontend")
 
@@ -145,8 +145,8 @@ def create_app(
         if body.get("stream"):
             async def sse():
               try:
-                token_ids: List[int] = []
-                sent_len = 0
+                detok = IncrementalDetokenizer(tokenizer)
+                token_ids = detok.ids
                 t_start = time.monotonic()
                 first_token_t = None
                 done = False
@@ -171,13 +171,13 @@ def create_app(
                     if first_token_t is None:
                         first_token_t = time.monotonic()
                     finish = None
+                    new_ids = []
                     for o in batch:
                         if o.token_id >= 0:
-                            token_ids.append(o.token_id)
+                            new_ids.append(o.token_id)
                         if o.finished:
                             finish = o.finish_reason or "stop"
-                    text = tokenizer.decode(token_ids)
-                    delta, sent_len = text[sent_len:], len(text)
+                    delta = detok.push(new_ids) if new_ids else ""
                     if not delta and finish is None:
                         continue
                     chunk = {
@@ -254,8 +254,8 @@ def create_app(
         if body.get("stream"):
             async def sse():
               try:
-                token_ids: List[int] = []
-                sent_len = 0
+                detok = IncrementalDetokenizer(tokenizer)
+                token_ids = detok.ids
                 t_start = time.monotonic()
                 first_t = None
                 done = False
@@ -276,13 +276,13 @@ def create_app(
                     if first_t is None:
                         first_t = time.monotonic()
                     finish = None
+                    new_ids = []
                     for o in batch:
                         if o.token_id >= 0:
-                            token_ids.append(o.token_id)
+                            new_ids.append(o.token_id)
                         if o.finished:
                             finish = o.finish_reason or "stop"
-                    text = tokenizer.decode(token_ids)
-                    delta, sent_len = text[sent_len:], len(text)
+                    delta = detok.push(new_ids) if new_ids else ""
                     if not delta and finish is None:
                         continue
                     chunk = {

@@ -85,3 +85,37 @@ class TokenizerWrapper:
             )
         text = self.synthetic.apply_chat_template(messages)
         return self.encode(text)
+
+
+class IncrementalDetokenizer:
+    """Streaming detokenization with a sliding window (the standard
+    incremental scheme: byte-level BPE merges across token boundaries, so
+    emit only text that can no longer change; a trailing U+FFFD marks an
+    incomplete multibyte sequence that must wait for more tokens).
+
+    Replaces full-sequence re-decoding in the SSE path — that was O(n^2)
+    characters per stream over a generation."""
+
+    def __init__(self, tok: "TokenizerWrapper"):
+        self.tok = tok
+        self.ids: List[int] = []
+        self.prefix_offset = 0
+        self.read_offset = 0
+
+    def push(self, new_ids: List[int]) -> str:
+        """Append tokens; return the newly-stable text delta ('' if the tail
+        is still ambiguous)."""
+        self.ids.extend(new_ids)
+        prefix_text = self.tok.decode(self.ids[self.prefix_offset:self.read_offset])
+        new_text = self.tok.decode(self.ids[self.prefix_offset:])
+        if new_text.endswith("�"):
+            return ""  # incomplete utf-8 sequence: wait for the next token
+        delta = new_text[len(prefix_text):]
+        self.prefix_offset = self.read_offset
+        self.read_offset = len(self.ids)
+        return delta
+
+    @property
+    def text_so_far(self) -> str:
+        """Full decode (used for final bookkeeping, not per token)."""
+        return self.tok.decode(self.ids)

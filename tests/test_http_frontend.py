@@ -227,3 +227,19 @@ def test_profile_endpoint(client, tmp_path):
             break
         _time.sleep(0.05)
     assert _os.path.exists(path) and _os.path.getsize(path) > 100
+
+
+def test_incremental_detokenizer_matches_full_decode():
+    from parallax_amd.server.tokenizer_util import (
+        IncrementalDetokenizer, TokenizerWrapper,
+    )
+
+    tok = TokenizerWrapper(vocab_size=512)
+    ids = [5, 9, 13, 2, 7, 300, 301, 55]
+    detok = IncrementalDetokenizer(tok)
+    pieces = [detok.push([i]) for i in ids]
+    assert "".join(pieces) == tok.decode(ids)
+    # batched pushes too (the SSE coalescing path)
+    detok2 = IncrementalDetokenizer(tok)
+    got = detok2.push(ids[:3]) + detok2.push(ids[3:])
+    assert got == tok.decode(ids)
