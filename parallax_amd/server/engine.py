@@ -268,6 +268,7 @@ class Engine:
         self._tok_bcast_buf: Optional[torch.Tensor] = None
         self._lp_bcast_buf: Optional[torch.Tensor] = None
         self.object_sync_count = 0  # payload broadcasts (adds/aborts only)
+        self._deferred_free: List[InitialRequest] = []
         logger.info(
             "engine up: layers [%d,%d) of %d, %d KV blocks x %d tokens, mla=%s, device %s",
             start, end, cfg.num_layers, num_blocks, args.block_size,
@@ -488,9 +489,13 @@ class Engine:
             return False
         for r in batch.decode_reqs:
             sp = r.sampling_params
-            if not (sp.ignore_eos or not r.eos_token_ids):
-                return False
-            if sp.stop_token_ids or sp.logprobs or sp.json_schema:
+            # eos/stop-token finishes are handled ONE STEP LATE by
+            # _finalize_inflight (rollback: truncate the zombie placeholder,
+            # defer the cache release until the in-flight step's event is
+            # synced) — so the common serving case keeps the pipelining.
+            # logprobs/json_schema/penalties stay sync: they need token
+            # VALUES on the host before the next forward.
+            if sp.logprobs or sp.json_schema:
                 return False
             if (sp.repetition_penalty != 1.0 or sp.presence_penalty != 0.0
                     or sp.frequency_penalty != 0.0):
@@ -577,18 +582,44 @@ class Engine:
         f = flight if flight is not None else self._inflight
         if f is None:
             return []
-        if flight is None:
+        drained = flight is None  # no newer flight exists after this one
+        if drained:
             self._inflight = None
         if f["event"] is not None:
             f["event"].synchronize()
+        # the event sync above covers the zombie KV writes that deferred
+        # these frees (see release_keep_cache)
+        for req in self._deferred_free:
+            self.scheduler.free_cache(req)
+        self._deferred_free.clear()
         toks = f["pinned"][: len(f["reqs"])].tolist()
         outputs = []
         for req, tok, idx in zip(f["reqs"], toks, f["out_idx"]):
-            if idx < len(req.output_token_ids) and \
-                    req.output_token_ids[idx] == self.PLACEHOLDER_TOKEN:
+            if idx >= len(req.output_token_ids):
+                continue  # zombie: request finished before this step's token
+            if req.output_token_ids[idx] == self.PLACEHOLDER_TOKEN:
                 req.output_token_ids[idx] = tok
+            finished = False
+            if req.rid in self.scheduler.running:
+                # one-step-late eos/stop/length detection: drop any newer
+                # placeholders, test, restore them if the request continues
+                tail = len(req.output_token_ids) - (idx + 1)
+                del req.output_token_ids[idx + 1:]
+                if req.check_finished():
+                    finished = True
+                    self.scheduler.release_keep_cache(req)
+                    if drained:
+                        self.scheduler.free_cache(req)
+                    else:
+                        self._deferred_free.append(req)
+                else:
+                    req.output_token_ids.extend(
+                        [self.PLACEHOLDER_TOKEN] * tail
+                    )
             outputs.append(
-                StepOutput(rid=req.rid, token_id=tok, finished=False)
+                StepOutput(rid=req.rid, token_id=tok, finished=finished,
+                           finish_reason=req.status.finish_reason
+                           if finished else None)
             )
         return outputs
 

@@ -206,6 +206,19 @@ class Scheduler:
 
     def _release(self, req: InitialRequest) -> None:
         self.running.pop(req.rid, None)
+        self.free_cache(req)
+        self.finished_reqs.append(req)
+
+    def release_keep_cache(self, req: InitialRequest) -> None:
+        """Remove from the running set (so the next batch excludes it) but
+        keep its KV blocks allocated: the async-pipelined step already in
+        flight still WRITES this request's last KV slot — freeing now would
+        let a concurrent allocation adopt the block mid-write. The engine
+        calls free_cache() after that step's completion event is synced."""
+        self.running.pop(req.rid, None)
+        self.finished_reqs.append(req)
+
+    def free_cache(self, req: InitialRequest) -> None:
         # Publish only tokens whose KV was actually computed (the reference
         # bounds insertion by context_len, cache_manager.insert_full_blocks_to_cache):
         # the final sampled token is never forwarded so its KV slot is unwritten,
@@ -215,7 +228,6 @@ class Scheduler:
             self.cache.free_request(req.rid, req.all_token_ids[:computed])
         else:
             self.cache.free_request(req.rid, None)
-        self.finished_reqs.append(req)
 
     # -- introspection ----------------------------------------------------------------
 

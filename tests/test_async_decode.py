@@ -50,8 +50,8 @@ def test_async_matches_sync_sampled(monkeypatch):
 
 
 def test_eos_sensitive_falls_back(monkeypatch):
-    """ignore_eos=False keeps requests on the sync path (finish depends on
-    token values) — results identical either way."""
+    """ignore_eos=False requests now ride the async path too (one-step-late
+    finish detection) — results identical either way."""
     sps = [SamplingParams(temperature=0.0, max_new_tokens=12)] * 2
     a = _run(True, sps, PROMPTS, monkeypatch)
     b = _run(False, sps, PROMPTS, monkeypatch)
@@ -98,3 +98,53 @@ def test_async_matches_sync_gpu(monkeypatch):
     a, b = run(True), run(False)
     assert a == b
     assert all(len(v) == 12 for v in a)
+
+
+
+def _run_counted(async_on, sps, prompts, monkeypatch, engine_args=None):
+    """Like _run, but also reports how many async enqueues happened."""
+    monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+    eng = Engine(_cfg(), engine_args or EngineArgs(
+        block_size=8, num_kv_blocks=128, dtype=torch.float32, seed=7),
+        random_weights=True)
+    n_async = 0
+    orig = eng._enqueue_async
+
+    def counted(*a, **kw):
+        nonlocal n_async
+        n_async += 1
+        return orig(*a, **kw)
+
+    eng.__dict__["_enqueue_async"] = counted
+    out = eng.generate(prompts, sps)
+    return out, n_async, eng
+
+
+def test_stop_token_finishes_on_async_path(monkeypatch):
+    """A mid-stream stop token must (a) actually run through the async
+    pipeline, (b) finish at the same position as the sync engine, and
+    (c) return every KV block (deferred frees flushed)."""
+    probe = [SamplingParams(temperature=0.0, max_new_tokens=12,
+                            ignore_eos=True)] * 2
+    base, _, _ = _run_counted(False, probe, PROMPTS, monkeypatch)
+    toks0 = list(base.values())[0]
+    stop = toks0[4]  # finish request 0 after its 5th token
+
+    args = EngineArgs(block_size=8, num_kv_blocks=128, dtype=torch.float32,
+                      seed=7, enable_prefix_cache=False)
+    sps = [SamplingParams(temperature=0.0, max_new_tokens=12, ignore_eos=True,
+                          stop_token_ids=[stop])] * 2
+    a, n_async, eng = _run_counted(True, sps, PROMPTS, monkeypatch,
+                                   engine_args=args)
+    b, _, _ = _run_counted(False, sps, PROMPTS, monkeypatch,
+                           engine_args=args)
+    assert list(a.values()) == list(b.values())
+    assert n_async > 0, "stop-token requests should use the async path now"
+    a0 = list(a.values())[0]
+    assert a0[-1] == stop and len(a0) == 5
+    # all placeholders resolved, no zombies leaked
+    for toks in a.values():
+        assert all(t >= 0 for t in toks)
+    # deferred cache frees flushed: every block back in the pool
+    assert eng.cache_manager.num_free_blocks == 128
+    assert not eng._deferred_free and eng._inflight is None
