@@ -20,11 +20,20 @@ def _cfg():
     )
 
 
+def _cpu_comm():
+    # pin to CPU: these fp32 tests must behave the same on a GPU box
+    # (the bf16 HIP kernels are exercised by the @pytest.mark.gpu variant)
+    from parallax_amd.parallel.comm import CommContext
+
+    return CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                       pp_rank=0, tp_rank=0, device=torch.device("cpu"))
+
+
 def _run(async_on, sps, prompts, monkeypatch):
     monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
     eng = Engine(_cfg(), EngineArgs(block_size=8, num_kv_blocks=128,
                                     dtype=torch.float32, seed=7),
-                 random_weights=True)
+                 comm=_cpu_comm(), random_weights=True)
     return eng.generate(prompts, sps)
 
 
@@ -106,7 +115,7 @@ def _run_counted(async_on, sps, prompts, monkeypatch, engine_args=None):
     monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
     eng = Engine(_cfg(), engine_args or EngineArgs(
         block_size=8, num_kv_blocks=128, dtype=torch.float32, seed=7),
-        random_weights=True)
+        comm=_cpu_comm(), random_weights=True)
     n_async = 0
     orig = eng._enqueue_async
 
