@@ -412,6 +412,12 @@ class Engine:
         for req in self.scheduler.sweep_aborted():
             outputs.append(StepOutput(rid=req.rid, token_id=-1, finished=True,
                                       finish_reason=req.status.finish_reason))
+        for req in self.scheduler.drain_aborted_waiting():
+            outputs.append(StepOutput(rid=req.rid, token_id=-1, finished=True,
+                                      finish_reason=req.status.finish_reason))
+        # the single-host engine reports finishes via StepOutput; drain the
+        # scheduler's bookkeeping list so a long-lived server doesn't grow it
+        self.scheduler.drain_finished()
         if timeout_sweep:
             for req in self.scheduler.sweep_timeouts():
                 outputs.append(StepOutput(rid=req.rid, token_id=-1,

@@ -71,6 +71,9 @@ class Scheduler:
         self.wait_queue: Deque[InitialRequest] = deque()
         self.running: Dict[str, InitialRequest] = {}
         self.finished_reqs: List[InitialRequest] = []
+        # aborts that hit the wait queue never pass through the running-set
+        # sweeps, so the engine reports them to consumers from this list
+        self.aborted_waiting: List[InitialRequest] = []
 
     # -- ingress ---------------------------------------------------------------
 
@@ -90,6 +93,7 @@ class Scheduler:
                 r.status = RequestStatus.FINISHED_ABORT
                 del self.wait_queue[i]
                 self.finished_reqs.append(r)
+                self.aborted_waiting.append(r)
                 return True
         return False
 
@@ -238,6 +242,10 @@ class Scheduler:
     @property
     def num_running(self) -> int:
         return len(self.running)
+
+    def drain_aborted_waiting(self) -> List[InitialRequest]:
+        out, self.aborted_waiting = self.aborted_waiting, []
+        return out
 
     def drain_finished(self) -> List[InitialRequest]:
         out, self.finished_reqs = self.finished_reqs, []
