@@ -385,7 +385,11 @@ class Engine:
 
     @property
     def has_work(self) -> bool:
-        return self.scheduler.has_work or bool(self._pending_adds)
+        # an in-flight async step / deferred KV frees still need one more
+        # step() to drain even when the scheduler itself is empty (the last
+        # running request can finish one step late)
+        return (self.scheduler.has_work or bool(self._pending_adds)
+                or self._inflight is not None or bool(self._deferred_free))
 
     # -- the step ---------------------------------------------------------------------
 
@@ -439,13 +443,28 @@ class Engine:
                 return outputs
             if prev is not None:  # membership changed: drain, then refill
                 outputs.extend(self._finalize_inflight())
+                # the drain can detect one-step-late finishes: requests the
+                # formed batch still lists may have just been released
+                batch.decode_reqs = [
+                    r for r in batch.decode_reqs
+                    if r.rid in self.scheduler.running
+                ]
+                if not batch.decode_reqs:
+                    return outputs
             self._enqueue_async(batch.decode_reqs, input_dev=None)
             return outputs
 
         if self._inflight is not None:
-            # finalize only patches token values (no finishes possible in the
-            # async envelope), so the batch formed above remains valid
+            # the drain patches token values AND may finish requests
+            # one step late — drop freed requests from the formed batch
+            # before any forward touches their (released) cache state
             outputs.extend(self._finalize_inflight())
+            batch.decode_reqs = [
+                r for r in batch.decode_reqs
+                if r.rid in self.scheduler.running
+            ]
+            if batch.is_empty:
+                return outputs
         sample_reqs: List[InitialRequest] = []
         logits_parts: List[torch.Tensor] = []
 
