@@ -165,3 +165,49 @@ def test_async_pipelining_equivalence(reqs, seed):
     tokens_b, finish_b = run(False)
     assert tokens_a == tokens_b
     assert finish_a == finish_b
+
+
+@settings(max_examples=15, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(reqs=st.lists(req_strategy, min_size=2, max_size=6),
+       share_prefix=st.booleans(),
+       seed=st.integers(min_value=0, max_value=2))
+def test_prefix_cache_accounting(reqs, share_prefix, seed):
+    """With the radix prefix cache ENABLED: after everything drains, every
+    block is either free or held by the radix (free + cached = pool), and
+    same-prefix requests must not corrupt accounting across async finishes."""
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=97, hidden_size=32,
+        num_layers=2, num_heads=2, num_kv_heads=1, head_dim=16,
+        intermediate_size=64, max_position_embeddings=256,
+        eos_token_ids=[5],
+    )
+    comm = CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                       pp_rank=0, tp_rank=0, device=torch.device("cpu"))
+    eng = Engine(
+        cfg,
+        EngineArgs(block_size=8, num_kv_blocks=NUM_BLOCKS,
+                   dtype=torch.float32, seed=seed, max_batch_size=8),
+        comm=comm, random_weights=True,
+    )
+    finish = {}
+    for i, r in enumerate(reqs):
+        sp = SamplingParams(
+            temperature=r["temperature"], max_new_tokens=r["max_new"],
+            ignore_eos=r["ignore_eos"],
+            stop_token_ids=[r["stop_tok"]] if r["stop_tok"] else [],
+        )
+        base = list(range(3, 27)) if share_prefix \
+            else list(range(3 + i, 3 + i + r["prompt_len"]))
+        eng.submit(base + [50 + i], sp, rid=f"p{i}")
+    for _ in range(500):
+        if not eng.has_work:
+            break
+        for out in eng.step():
+            if out.finished:
+                finish[out.rid] = out.finish_reason
+    assert not eng.has_work
+    assert len(finish) == len(reqs)
+    cm = eng.cache_manager
+    assert cm.allocator.num_free_blocks + cm.radix.num_cached_blocks \
+        == NUM_BLOCKS
