@@ -347,3 +347,44 @@ def test_detect_hardware_cpu_profile():
     assert hw.memory_gb > 0 and hw.tflops_bf16 > 0
     n = Node(node_id="d", hardware=hw)
     assert n.decoder_layer_capacity(llama70b_info()) >= 0
+
+
+try:
+    from hypothesis import HealthCheck, given, settings
+    from hypothesis import strategies as st
+    HAVE_HYP = True
+except ImportError:  # pragma: no cover
+    HAVE_HYP = False
+
+
+@pytest.mark.skipif(not HAVE_HYP, reason="hypothesis not installed")
+@settings(max_examples=40, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(fracs=st.lists(st.sampled_from([0.15, 0.25, 0.5, 1.0]),
+                      min_size=1, max_size=6),
+       algo=st.sampled_from(["greedy", "dp"]))
+def test_allocators_structural_invariants(fracs, algo):
+    """For ANY node mix, every produced pipeline must cover [0, L)
+    contiguously with disjoint spans, respect each node's layer capacity,
+    and never reuse a node across pipelines."""
+    model = llama70b_info()
+    nodes = [build_node(f"n{i}", small_gpu(f, name=f"hw{i}"))
+             for i, f in enumerate(fracs)]
+    set_full_rtt(nodes)
+    cls = (GreedyLayerAllocator if algo == "greedy"
+           else DynamicProgrammingLayerAllocator)
+    pipelines = cls(model).allocate_from_standby(list(nodes))
+    L = model.num_layers
+    used = set()
+    for pipe in pipelines:
+        spans = [(n.start_layer, n.end_layer) for n in pipe.nodes]
+        # contiguous cover of [0, L)
+        assert spans[0][0] == 0 and spans[-1][1] == L
+        for (s0, e0), (s1, e1) in zip(spans, spans[1:]):
+            assert e0 == s1 and e0 > s0
+        assert all(e > s for s, e in spans)
+        for n in pipe.nodes:
+            assert n.node_id not in used, "node reused across pipelines"
+            used.add(n.node_id)
+            cap = n.decoder_layer_capacity(model)
+            assert n.end_layer - n.start_layer <= cap
