@@ -4,10 +4,11 @@ sampler.
 
 Reference parity: the reference carries `json_schema` on SamplingParams
 (src/parallax/server/sampling/sampling_params.py:25) and delegates enforcement
-to its backends; here enforcement is native. Schema subset: object (properties
-generated in schema order, required == all listed), string, integer, number,
-boolean, null, enum (distinct literals), array (items + minItems/maxItems),
-and arbitrary nesting of those.
+to its backends; here enforcement is native. Schema subset: object (properties generated in
+schema order; a "required" list makes the others optional — the automaton
+lets the model skip any run of optional keys but never a required one),
+string, integer, number, boolean, null, enum (distinct literals), array
+(items + minItems/maxItems), and arbitrary nesting of those.
 
 Design: states are immutable tuples (a stack of frames, top last), so the
 per-state token mask is cacheable. `advance` consumes one character;
@@ -46,10 +47,15 @@ class JsonSchemaFSM:
             if t == "object":
                 props = schema.get("properties", {})
                 order = [k for k in props]
+                # no "required" list = every property required (strict-mode
+                # default; also the pre-existing behavior)
+                req = frozenset(schema["required"]) if "required" in schema \
+                    else frozenset(order)
                 node = {
                     "kind": "object",
                     "keys": tuple(order),
                     "children": tuple(self._compile(props[k]) for k in order),
+                    "required": req,
                 }
             elif t == "array":
                 node = {
@@ -95,16 +101,7 @@ class JsonSchemaFSM:
             elif kind == "null":
                 state = rest + (("lit", "null", 0),)
             elif kind == "object":
-                frames: List[Tuple] = [("lit", "}", 0)]
-                for i in range(len(node["keys"]) - 1, -1, -1):
-                    frames.append(("val", node["children"][i]))
-                    sep = "{" if i == 0 else ","
-                    frames.append(("lit", f'{sep}"{node["keys"][i]}":', 0))
-                if not node["keys"]:
-                    frames = [("lit", "{}", 0)]
-                # stack top is state[-1]: frames were built closing-brace
-                # first, so appending in built order puts '{"key":' on top
-                state = rest + tuple(frames)
+                state = rest + (("obj", nid, 0, "open"),)
             elif kind == "array":
                 state = rest + (("arr", nid, 0, "item_or_close"),
                                 ("lit", "[", 0))
@@ -189,6 +186,57 @@ class JsonSchemaFSM:
             if phase == "exp1":
                 return rest + ((tag, "exp"),) if ch in _DIGITS else None
             return None
+
+        if tag == "obj":
+            _, nid, i, mode = top
+            node = self.nodes[nid]
+            keys, req = node["keys"], node["required"]
+            may_close = not any(k in req for k in keys[i:])
+            if mode == "open":
+                if ch != "{":
+                    return None
+                return rest + (("obj", nid, 0, "key_or_close"),)
+            if mode == "key_or_close":
+                if ch == "}" and may_close:
+                    return rest
+                if i >= len(keys):
+                    return None
+                return self.advance(rest + (("objkey", nid, i, ""),), ch)
+            if mode == "key":  # after a comma: a key MUST follow
+                if i >= len(keys):
+                    return None
+                return self.advance(rest + (("objkey", nid, i, ""),), ch)
+            if mode == "sep_or_close":
+                if ch == "}" and may_close:
+                    return rest
+                if ch == "," and i < len(keys):
+                    return rest + (("obj", nid, i, "key"),)
+                return None
+            return None
+
+        if tag == "objkey":
+            # lazily choose WHICH key comes next by matching characters
+            # against the candidate quoted keys (schema order; a run of
+            # optional keys may be skipped, a required key may not)
+            _, nid, i, prefix = top
+            node = self.nodes[nid]
+            keys, req = node["keys"], node["required"]
+            opts = []
+            for j in range(i, len(keys)):
+                opts.append((f'"{keys[j]}":', j))
+                if keys[j] in req:
+                    break
+            p = prefix + ch
+            live = [(o, j) for o, j in opts if o.startswith(p)]
+            if not live:
+                return None
+            for o, j in live:
+                # quoted forms are never prefixes of each other, so an exact
+                # match is unique and final
+                if o == p:
+                    return rest + (("obj", nid, j + 1, "sep_or_close"),
+                                   ("val", node["children"][j]))
+            return rest + (("objkey", nid, i, p),)
 
         if tag == "arr":
             _, nid, n, expect = top

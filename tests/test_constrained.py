@@ -103,3 +103,50 @@ def test_engine_schema_conformance_synthetic_tokenizer():
     )
     obj = json.loads(text)
     assert isinstance(obj.get("a"), int) and isinstance(obj.get("b"), bool)
+
+
+def test_optional_properties():
+    """A "required" subset lets the model skip optional keys (in order) but
+    never a required one, and close only once all required keys are done."""
+    import json
+
+    from parallax_amd.server.constrained import JsonSchemaFSM
+
+    fsm = JsonSchemaFSM(json.loads(json.dumps({
+        "type": "object",
+        "properties": {"a": {"type": "integer"}, "b": {"type": "boolean"},
+                       "c": {"type": "string"}},
+        "required": ["a", "c"],
+    })))
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    assert accepts('{"a":1,"b":true,"c":"x"}')   # all keys
+    assert accepts('{"a":1,"c":"x"}')            # optional b skipped
+    assert not accepts('{"a":1}')                # required c missing
+    assert not accepts('{"b":true,"c":"x"}')     # required a skipped
+    assert not accepts('{"c":"x","a":1}')        # order violated
+    assert not accepts('{"a":1,"b":true}')       # closed before required c
+    assert not accepts('{"a":1,,"c":"x"}')
+
+
+def test_all_optional_object():
+    import json
+
+    from parallax_amd.server.constrained import JsonSchemaFSM
+
+    fsm = JsonSchemaFSM({"type": "object", "required": [],
+                         "properties": {"x": {"type": "integer"}}})
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    assert accepts("{}")
+    assert accepts('{"x":7}')
+    assert not accepts('{"x":}')
+    # round-trip with python's parser for everything the FSM accepts
+    for t in ("{}", '{"x":7}'):
+        json.loads(t)
