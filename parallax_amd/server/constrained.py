@@ -8,7 +8,8 @@ to its backends; here enforcement is native. Schema subset: object (properties g
 schema order; a "required" list makes the others optional — the automaton
 lets the model skip any run of optional keys but never a required one),
 string, integer, number, boolean, null, enum (distinct literals), array
-(items + minItems/maxItems), and arbitrary nesting of those.
+(items + minItems/maxItems), anyOf/oneOf unions (nullable fields), and
+arbitrary nesting of those.
 
 Design: states are immutable tuples (a stack of frames, top last), so the
 per-state token mask is cacheable. `advance` consumes one character;
@@ -38,7 +39,11 @@ class JsonSchemaFSM:
         nid = len(self.nodes)
         self.nodes.append({})
         node: Dict = {}
-        if "enum" in schema:
+        if "anyOf" in schema or "oneOf" in schema:
+            branches = schema.get("anyOf") or schema.get("oneOf")
+            node = {"kind": "union",
+                    "children": tuple(self._compile(b) for b in branches)}
+        elif "enum" in schema:
             node = {"kind": "enum",
                     "options": tuple(json.dumps(v, separators=(",", ":"))
                                      for v in schema["enum"])}
@@ -77,7 +82,17 @@ class JsonSchemaFSM:
         return (("val", self.root),)
 
     def is_complete(self, state: State) -> bool:
-        return len(state) == 0
+        state = self._expand(state)
+        if len(state) == 0:
+            return True
+        top, rest = state[-1], state[:-1]
+        if top[0] == "union":
+            return any(self.is_complete(sub) for sub in top[1]) \
+                and self.is_complete(rest)
+        if top[0] in ("int", "num") and top[1] in ("z", "int", "frac", "exp"):
+            # a number is complete at any digit boundary
+            return self.is_complete(rest)
+        return False
 
     # -- the automaton ----------------------------------------------------------
 
@@ -102,6 +117,9 @@ class JsonSchemaFSM:
                 state = rest + (("lit", "null", 0),)
             elif kind == "object":
                 state = rest + (("obj", nid, 0, "open"),)
+            elif kind == "union":
+                subs = tuple((("val", c),) for c in node["children"])
+                state = rest + (("union", subs),)
             elif kind == "array":
                 state = rest + (("arr", nid, 0, "item_or_close"),
                                 ("lit", "[", 0))
@@ -185,6 +203,23 @@ class JsonSchemaFSM:
                 return rest + ((tag, "exp"),) if ch in _DIGITS else None
             if phase == "exp1":
                 return rest + ((tag, "exp"),) if ch in _DIGITS else None
+            return None
+
+        if tag == "union":
+            # nondeterministic branch embedded in one frame: advance every
+            # live branch; a branch completing hands the character to the
+            # enclosing context only when no branch can consume it
+            # (maximal munch, same as the number frames)
+            subs = top[1]
+            new_subs = []
+            for sub in subs:
+                adv = self.advance(sub, ch)
+                if adv is not None:
+                    new_subs.append(adv)
+            if new_subs:
+                return rest + (("union", tuple(new_subs)),)
+            if any(self.is_complete(sub) for sub in subs):
+                return self.advance(rest, ch)
             return None
 
         if tag == "obj":

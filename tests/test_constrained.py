@@ -150,3 +150,46 @@ def test_all_optional_object():
     # round-trip with python's parser for everything the FSM accepts
     for t in ("{}", '{"x":7}'):
         json.loads(t)
+
+
+def test_anyof_union():
+    """anyOf/oneOf: the automaton accepts any branch (nullable fields are
+    the common case in tool schemas)."""
+    from parallax_amd.server.constrained import JsonSchemaFSM
+
+    fsm = JsonSchemaFSM({
+        "type": "object",
+        "properties": {
+            "v": {"anyOf": [{"type": "integer"}, {"type": "null"}]},
+            "w": {"oneOf": [{"type": "string"},
+                            {"type": "array", "items": {"type": "integer"}}]},
+        },
+    })
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    assert accepts('{"v":3,"w":"hi"}')
+    assert accepts('{"v":null,"w":[1,2]}')
+    assert accepts('{"v":-7,"w":[]}')
+    assert not accepts('{"v":true,"w":"hi"}')   # neither branch
+    assert not accepts('{"v":3,"w":7}')
+
+
+def test_union_number_boundary():
+    """A union branch completing at a delimiter hands the char onward."""
+    from parallax_amd.server.constrained import JsonSchemaFSM
+
+    fsm = JsonSchemaFSM({"type": "array",
+                         "items": {"anyOf": [{"type": "number"},
+                                             {"type": "boolean"}]}})
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    assert accepts("[1.5,true,2]")
+    assert accepts("[]")
+    assert not accepts("[1.5,]")
+    assert not accepts("[null]")
