@@ -8,8 +8,8 @@ to its backends; here enforcement is native. Schema subset: object (properties g
 schema order; a "required" list makes the others optional — the automaton
 lets the model skip any run of optional keys but never a required one),
 string, integer, number, boolean, null, enum (distinct literals), array
-(items + minItems/maxItems), anyOf/oneOf unions (nullable fields), and
-arbitrary nesting of those.
+(items + minItems/maxItems), anyOf/oneOf unions (nullable fields),
+$ref/$defs including recursive references, and arbitrary nesting of those.
 
 Design: states are immutable tuples (a stack of frames, top last), so the
 per-state token mask is cacheable. `advance` consumes one character;
@@ -33,11 +33,27 @@ class JsonSchemaFSM:
     def __init__(self, schema: dict):
         # flatten schema nodes into a list so frames can reference them by id
         self.nodes: List[dict] = []
+        defs = schema.get("$defs") or schema.get("definitions") or {}
+        # reserve ids for every named def first so $ref cycles (recursive
+        # schemas: trees, linked structures) resolve to stable ids
+        self._defs: Dict[str, int] = {}
+        for name in defs:
+            self._defs[name] = len(self.nodes)
+            self.nodes.append({})
+        for name, sub in defs.items():
+            self._compile(sub, nid=self._defs[name])
         self.root = self._compile(schema)
 
-    def _compile(self, schema: dict) -> int:
-        nid = len(self.nodes)
-        self.nodes.append({})
+    def _compile(self, schema: dict, nid: Optional[int] = None) -> int:
+        if "$ref" in schema:
+            ref = schema["$ref"]
+            name = ref.rsplit("/", 1)[-1]
+            if name not in self._defs:
+                raise ValueError(f"unresolvable $ref: {ref!r}")
+            return self._defs[name]
+        if nid is None:
+            nid = len(self.nodes)
+            self.nodes.append({})
         node: Dict = {}
         if "anyOf" in schema or "oneOf" in schema:
             branches = schema.get("anyOf") or schema.get("oneOf")

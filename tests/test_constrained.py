@@ -193,3 +193,33 @@ def test_union_number_boundary():
     assert accepts("[]")
     assert not accepts("[1.5,]")
     assert not accepts("[null]")
+
+
+def test_recursive_ref_schema():
+    """$defs + $ref, including self-reference (a binary-tree schema)."""
+    from parallax_amd.server.constrained import JsonSchemaFSM
+
+    fsm = JsonSchemaFSM({
+        "$defs": {
+            "tree": {
+                "type": "object",
+                "properties": {
+                    "v": {"type": "integer"},
+                    "kids": {"type": "array",
+                             "items": {"$ref": "#/$defs/tree"},
+                             "maxItems": 2},
+                },
+                "required": ["v"],
+            }
+        },
+        "$ref": "#/$defs/tree",
+    })
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    assert accepts('{"v":1}')
+    assert accepts('{"v":1,"kids":[{"v":2},{"v":3,"kids":[{"v":4}]}]}')
+    assert not accepts('{"kids":[]}')          # required v missing
+    assert not accepts('{"v":1,"kids":[5]}')   # item not a tree
