@@ -68,6 +68,16 @@ class JsonSchemaFSM:
             if t == "object":
                 props = schema.get("properties", {})
                 order = [k for k in props]
+                if not order and isinstance(
+                        schema.get("additionalProperties"), dict):
+                    # free-form object: arbitrary string keys, value schema
+                    # from additionalProperties (this is what makes
+                    # response_format json_object expressible)
+                    node = {"kind": "freeobject",
+                            "value": self._compile(
+                                schema["additionalProperties"])}
+                    self.nodes[nid] = node
+                    return nid
                 # no "required" list = every property required (strict-mode
                 # default; also the pre-existing behavior)
                 req = frozenset(schema["required"]) if "required" in schema \
@@ -133,6 +143,8 @@ class JsonSchemaFSM:
                 state = rest + (("lit", "null", 0),)
             elif kind == "object":
                 state = rest + (("obj", nid, 0, "open"),)
+            elif kind == "freeobject":
+                state = rest + (("fobj", nid, "open"),)
             elif kind == "union":
                 subs = tuple((("val", c),) for c in node["children"])
                 state = rest + (("union", subs),)
@@ -262,6 +274,28 @@ class JsonSchemaFSM:
                     return rest
                 if ch == "," and i < len(keys):
                     return rest + (("obj", nid, i, "key"),)
+                return None
+            return None
+
+        if tag == "fobj":
+            _, nid, mode = top
+            node = self.nodes[nid]
+            if mode == "open":
+                return rest + (("fobj", nid, "key_or_close"),) \
+                    if ch == "{" else None
+            key_frames = (("fobj", nid, "sep_or_close"),
+                          ("val", node["value"]), ("lit", ":", 0), ("str",))
+            if mode == "key_or_close":
+                if ch == "}":
+                    return rest
+                return rest + key_frames if ch == '"' else None
+            if mode == "key":  # after a comma a key MUST follow
+                return rest + key_frames if ch == '"' else None
+            if mode == "sep_or_close":
+                if ch == "}":
+                    return rest
+                if ch == ",":
+                    return rest + (("fobj", nid, "key"),)
                 return None
             return None
 

@@ -110,6 +110,22 @@ def _extract_json_schema(body: dict):
     if raw is None and isinstance(rf, dict) and rf.get("type") == "json_schema":
         inner = rf.get("json_schema") or {}
         raw = inner.get("schema", inner)
+    if raw is None and isinstance(rf, dict) and rf.get("type") == "json_object":
+        # schema-less JSON mode: any object with any keys and any JSON values
+        raw = ANY_JSON_OBJECT_SCHEMA
     if raw is None:
         return None
     return raw if isinstance(raw, str) else _json.dumps(raw)
+
+
+#: "any JSON value" expressed in the FSM's schema subset (recursive union)
+ANY_JSON_OBJECT_SCHEMA = {
+    "$defs": {"any": {"anyOf": [
+        {"type": "string"}, {"type": "number"}, {"type": "boolean"},
+        {"type": "null"},
+        {"type": "array", "items": {"$ref": "#/$defs/any"}},
+        {"type": "object", "additionalProperties": {"$ref": "#/$defs/any"}},
+    ]}},
+    "type": "object",
+    "additionalProperties": {"$ref": "#/$defs/any"},
+}

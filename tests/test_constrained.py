@@ -223,3 +223,33 @@ def test_recursive_ref_schema():
     assert accepts('{"v":1,"kids":[{"v":2},{"v":3,"kids":[{"v":4}]}]}')
     assert not accepts('{"kids":[]}')          # required v missing
     assert not accepts('{"v":1,"kids":[5]}')   # item not a tree
+
+
+def test_free_form_object_and_json_object_mode():
+    """additionalProperties (arbitrary keys) + the json_object schema used by
+    response_format={"type": "json_object"}."""
+    import json
+
+    from parallax_amd.server.constrained import JsonSchemaFSM
+    from parallax_amd.server.sampling_params import ANY_JSON_OBJECT_SCHEMA
+
+    fsm = JsonSchemaFSM(ANY_JSON_OBJECT_SCHEMA)
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    for t in ('{}', '{"a":1}', '{"x":{"y":[1,"two",null,true]},"z":-3.5}',
+              '{"k":"v","k2":[{"deep":{}}]}'):
+        assert accepts(t), t
+        json.loads(t)  # everything accepted is valid JSON
+    for t in ('[]', '"str"', '{"a":}', '{:1}', '{"a" 1}'):
+        assert not accepts(t), t
+
+
+def test_json_object_response_format_plumbing():
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    sp = SamplingParams.from_openai(
+        {"response_format": {"type": "json_object"}})
+    assert sp.json_schema and "additionalProperties" in sp.json_schema
