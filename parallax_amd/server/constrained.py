@@ -360,6 +360,7 @@ class GrammarMatcher:
 
     _fsm_cache: Dict[str, JsonSchemaFSM] = {}
     _mask_cache: Dict[Tuple[str, State], List[int]] = {}
+    _trie_cache: Dict[int, tuple] = {}  # id(vocab) -> (vocab, trie root)
 
     def __init__(self, schema_json: str, vocab: List[str], eos_ids: List[int]):
         self.schema_json = schema_json
@@ -387,6 +388,27 @@ class GrammarMatcher:
             # abort or vocab mismatch) — freeze rather than crash
         self._consumed = len(output_token_ids)
 
+    def _trie(self):
+        """Prefix trie over the vocab: computing a mask walks the trie,
+        advancing the automaton per EDGE — a rejected prefix prunes its whole
+        subtree, so a mask costs O(grammar-reachable prefixes), not
+        O(vocab x token length). At 128k vocab the linear scan was a
+        multi-hundred-ms stall per previously-unseen automaton state; the
+        richer schemas (recursion, unions) make unseen states common."""
+        entry = self._trie_cache.get(id(self.vocab))
+        if entry is not None and entry[0] is self.vocab:
+            return entry[1]
+        root: Dict = {"ids": [], "kids": {}}
+        for i, tok in enumerate(self.vocab):
+            if not tok:
+                continue
+            node = root
+            for ch in tok:
+                node = node["kids"].setdefault(ch, {"ids": [], "kids": {}})
+            node["ids"].append(i)
+        self._trie_cache[id(self.vocab)] = (self.vocab, root)
+        return root
+
     def allowed_ids(self) -> List[int]:
         if self.state is None:
             return list(self.eos_ids)
@@ -395,11 +417,16 @@ class GrammarMatcher:
         key = (self.schema_json, self.state)
         ids = self._mask_cache.get(key)
         if ids is None:
-            adv = self.fsm.advance_str
-            st = self.state
-            ids = [
-                i for i, s in enumerate(self.vocab)
-                if s and adv(st, s) is not None
-            ]
+            ids = []
+            adv = self.fsm.advance
+            stack = [(self.state, self._trie())]
+            while stack:
+                st, node = stack.pop()
+                ids.extend(node["ids"])
+                for ch, child in node["kids"].items():
+                    nxt = adv(st, ch)
+                    if nxt is not None:
+                        stack.append((nxt, child))
+            ids.sort()
             self._mask_cache[key] = ids
         return ids
