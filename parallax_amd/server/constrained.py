@@ -95,7 +95,11 @@ class JsonSchemaFSM:
                     "min": int(schema.get("minItems", 0)),
                     "max": int(schema.get("maxItems", 1 << 30)),
                 }
-            elif t in ("string", "integer", "number", "boolean", "null"):
+            elif t == "string":
+                node = {"kind": "string",
+                        "minlen": int(schema.get("minLength", 0)),
+                        "maxlen": int(schema.get("maxLength", 1 << 30))}
+            elif t in ("integer", "number", "boolean", "null"):
                 node = {"kind": t}
             else:
                 raise ValueError(f"unsupported schema type: {t!r}")
@@ -132,7 +136,14 @@ class JsonSchemaFSM:
             if kind == "enum":
                 state = rest + (("alt", node["options"], ""),)
             elif kind == "string":
-                state = rest + (("str",), ("lit", '"', 0))
+                # bounded strings carry a length counter; unbounded ones use
+                # the counterless frame so every string state is shared (one
+                # cached mask covers all inside-string positions)
+                if node["minlen"] or node["maxlen"] < (1 << 30):
+                    state = rest + (("strn", node["minlen"], node["maxlen"],
+                                     0), ("lit", '"', 0))
+                else:
+                    state = rest + (("str",), ("lit", '"', 0))
             elif kind == "integer":
                 state = rest + (("int", "start"),)
             elif kind == "number":
@@ -179,6 +190,17 @@ class JsonSchemaFSM:
             if ord(ch) < 0x20:
                 return None
             return state
+
+        if tag == "strn":
+            _, mn, mx, n = top
+            if ch == '"':
+                return rest if n >= mn else None
+            if ch == "\\":
+                return rest + (("strn", mn, mx, n + 1), ("esc",)) \
+                    if n < mx else None
+            if ord(ch) < 0x20 or n >= mx:
+                return None
+            return rest + (("strn", mn, mx, n + 1),)
 
         if tag == "esc":
             if ch in '"\\/bfnrt':

@@ -253,3 +253,20 @@ def test_json_object_response_format_plumbing():
     sp = SamplingParams.from_openai(
         {"response_format": {"type": "json_object"}})
     assert sp.json_schema and "additionalProperties" in sp.json_schema
+
+
+def test_string_length_bounds():
+    from parallax_amd.server.constrained import JsonSchemaFSM
+
+    fsm = JsonSchemaFSM({"type": "object", "properties": {
+        "s": {"type": "string", "minLength": 2, "maxLength": 4}}})
+
+    def accepts(text):
+        st = fsm.advance_str(fsm.initial(), text)
+        return st is not None and fsm.is_complete(st)
+
+    assert accepts('{"s":"ab"}')
+    assert accepts('{"s":"abcd"}')
+    assert accepts('{"s":"a\\n"}')     # escape counts as one char
+    assert not accepts('{"s":"a"}')    # too short
+    assert not accepts('{"s":"abcde"}')  # too long
