@@ -179,7 +179,14 @@ class NodeAgent:
             if ex is None:
                 time.sleep(0.05)
                 continue
-            ex.step(recv_timeout=0.005)
+            try:
+                ex.step(recv_timeout=0.005)
+            except Exception:
+                # a step must never kill the node (reference keeps serving on
+                # batch errors); affected requests abort via the sweeps
+                logger.exception("executor step failed; node keeps serving")
+                time.sleep(0.1)
+                continue
             if ex.is_head:
                 for out in ex.drain_outputs():
                     q = self._streams.get(out.rid)

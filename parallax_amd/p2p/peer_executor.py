@@ -232,7 +232,17 @@ class PeerExecutor:
                                                 finished.routing_table)
             recv_timeout = 0.0  # only block on the first recv
 
-        # 2. abort sweep: release downstream state and terminate streams
+        # 2. timeout sweep: requests stranded by a dead downstream peer
+        # terminate with a timeout finish instead of hanging forever
+        self._sweep_counter = getattr(self, "_sweep_counter", 0) + 1
+        if self._sweep_counter % 64 == 0:
+            for req in self.scheduler.sweep_timeouts():
+                self.finished_outputs.append(
+                    PeerOutput(req.rid, -1, True, req.status.finish_reason)
+                )
+                self._broadcast_control("release", [req.rid],
+                                        req.routing_table)
+        # abort sweep: release downstream state and terminate streams
         for req in self.scheduler.sweep_aborted():
             self.finished_outputs.append(
                 PeerOutput(req.rid, -1, True, req.status.finish_reason)
@@ -458,7 +468,18 @@ class PeerExecutor:
                 continue
             by_peer[nxt].append(p)
         for peer, group in by_peer.items():
-            self.transport.send(peer, codec.encode_forward(group))
+            try:
+                self.transport.send(peer, codec.encode_forward(group))
+            except (ConnectionError, OSError, KeyError) as e:
+                # dead/unknown next hop: abort the affected requests so the
+                # head terminates their streams (reference: batch error ->
+                # abort + ERROR to client, sglang_executor.py:505-546) and
+                # keep serving everything else
+                logger.error("send to %s failed (%s): aborting %d reqs",
+                             peer, e, len(group))
+                if self.is_head:
+                    for pkt in group:
+                        self.scheduler.abort_request(pkt.rid)
 
     # -- weight refit over the transport (reference p2p/server.py:224-338) --------
 
@@ -541,4 +562,10 @@ class PeerExecutor:
     def _broadcast_control(self, kind: str, rids: List[str], routing_table: List[str]) -> None:
         for peer in routing_table:
             if peer != self.peer_id:
-                self.transport.send(peer, codec.encode_control(kind, rids))
+                try:
+                    self.transport.send(peer, codec.encode_control(kind, rids))
+                except (ConnectionError, OSError, KeyError) as e:
+                    # a dead peer cannot receive the release; its state is
+                    # reclaimed when it rejoins (fresh executor) or expires
+                    logger.warning("control %s to %s failed: %s",
+                                   kind, peer, e)

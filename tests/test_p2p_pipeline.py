@@ -274,3 +274,27 @@ def test_tcp_auth_handshake():
     finally:
         for t in (a, b, bad, naked):
             t.close()
+
+
+def test_dead_peer_aborts_requests_not_node():
+    """A routing table naming a nonexistent peer must terminate the affected
+    streams (abort) instead of crashing the head or hanging the client."""
+    cfg = tiny_cfg()
+    registry = {}
+    t = LoopbackTransport("head", registry)
+    head = PeerExecutor(cfg, 0, cfg.num_layers // 2, "head", t,
+                        dtype=torch.float32, num_kv_blocks=128, block_size=8)
+    for name, w in full_state_dict(cfg).items():
+        head.model.load_hf_weight(name, w)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=4, ignore_eos=True)
+    rid = head.submit([5, 9, 13], sp, ["head", "ghost-peer"])
+    finished = []
+    for _ in range(200):
+        head.step(recv_timeout=0.001)  # must not raise
+        finished += [o for o in head.drain_outputs() if o.finished]
+        if finished:
+            break
+    assert finished and finished[0].rid == rid
+    assert finished[0].finish_reason == "abort"
+    # head keeps serving: nothing left running
+    assert not head.scheduler.running
