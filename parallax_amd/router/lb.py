@@ -15,7 +15,8 @@ from typing import Dict, List, Optional
 
 import httpx
 from fastapi import FastAPI, HTTPException, Request
-from fastapi.responses import JSONResponse
+from fastapi.responses import (HTMLResponse, JSONResponse,
+                                StreamingResponse)
 
 from ..utils.logging_config import get_logger
 
@@ -136,8 +137,10 @@ def create_router_app(lb: LoadBalancer) -> FastAPI:
             lb.set_strategy(body["strategy"])
         return {"status": "ok"}
 
-    @app.post("/v1/chat/completions")
-    async def chat(request: Request):
+    async def _relay(request: Request, path: str):
+        """Proxy one OpenAI request to the picked cluster. Streaming requests
+        relay the SSE bytes as they arrive (the reference router relays SSE,
+        request_handler.py:190-245); metrics update from the usage chunk."""
         body = await request.json()
         await lb.check_health()
         ep = lb.pick()
@@ -145,20 +148,82 @@ def create_router_app(lb: LoadBalancer) -> FastAPI:
             raise HTTPException(503, "no healthy cluster")
         ep.inflight += 1
         t0 = time.monotonic()
-        try:
-            async with httpx.AsyncClient(timeout=600.0) as client:
-                r = await client.post(f"{ep.url}/v1/chat/completions", json=body)
-            data = r.json()
-            usage = data.get("usage", {})
-            ep.update_metrics(
-                usage.get("ttft_ms", (time.monotonic() - t0) * 1e3),
-                usage.get("tps"),
-            )
-            return JSONResponse(data, status_code=r.status_code)
-        except httpx.HTTPError as e:
-            ep.healthy = False
-            raise HTTPException(502, f"cluster {ep.url} failed: {e}")
-        finally:
-            ep.inflight -= 1
+        if not body.get("stream"):
+            try:
+                async with httpx.AsyncClient(timeout=600.0) as client:
+                    r = await client.post(f"{ep.url}{path}", json=body)
+                data = r.json()
+                usage = data.get("usage", {})
+                ep.update_metrics(
+                    usage.get("ttft_ms", (time.monotonic() - t0) * 1e3),
+                    usage.get("tps"),
+                )
+                return JSONResponse(data, status_code=r.status_code)
+            except httpx.HTTPError as e:
+                ep.healthy = False
+                raise HTTPException(502, f"cluster {ep.url} failed: {e}")
+            finally:
+                ep.inflight -= 1
+
+        async def sse():
+            import json as _json
+
+            first_t = None
+            try:
+                async with httpx.AsyncClient(timeout=600.0) as client:
+                    async with client.stream(
+                        "POST", f"{ep.url}{path}", json=body
+                    ) as r:
+                        async for line in r.aiter_lines():
+                            if first_t is None and line.startswith("data:"):
+                                first_t = time.monotonic()
+                            if line.startswith("data: ") and '"usage"' in line:
+                                try:
+                                    msg = _json.loads(line[6:])
+                                    u = msg.get("usage") or {}
+                                    ep.update_metrics(
+                                        u.get("ttft_ms",
+                                              ((first_t or time.monotonic())
+                                               - t0) * 1e3),
+                                        u.get("tps"),
+                                    )
+                                except ValueError:
+                                    pass
+                            yield line + "\n"
+            except httpx.HTTPError:
+                ep.healthy = False
+                yield 'data: {"error": "upstream failed"}\n\n'
+            finally:
+                ep.inflight -= 1
+
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
+    @app.post("/v1/chat/completions")
+    async def chat(request: Request):
+        return await _relay(request, "/v1/chat/completions")
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        return await _relay(request, "/v1/completions")
+
+    @app.get("/")
+    async def dashboard():
+        rows = "".join(
+            f"<tr><td>{u}</td><td>{'up' if e.healthy else 'DOWN'}</td>"
+            f"<td>{e.ewma_ttft_ms and round(e.ewma_ttft_ms, 1)}</td>"
+            f"<td>{e.ewma_tps and round(e.ewma_tps, 1)}</td>"
+            f"<td>{e.inflight}</td></tr>"
+            for u, e in lb.endpoints.items()
+        )
+        html = (
+            "<html><head><title>parallax-amd router</title>"
+            "<meta http-equiv=refresh content=3>"
+            "<style>body{font-family:monospace;background:#111;color:#ddd}"
+            "td,th{padding:4px 12px;border-bottom:1px solid #333}</style>"
+            f"</head><body><h2>router · strategy: {lb.strategy}</h2>"
+            "<table><tr><th>endpoint</th><th>health</th><th>ttft ms</th>"
+            f"<th>tps</th><th>inflight</th></tr>{rows}</table></body></html>"
+        )
+        return HTMLResponse(html)
 
     return app
