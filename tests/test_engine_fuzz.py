@@ -52,7 +52,24 @@ req_strategy = st.fixed_dictionaries({
     "stop_tok": st.sampled_from([None, 5, 17, 42]),
     "temperature": st.sampled_from([0.0, 0.8]),
     "abort_after": st.sampled_from([None, None, None, 0, 2, 5]),
+    # a json_schema request rides the sync path; mixing it with async-
+    # eligible requests churns batch membership through both paths
+    "grammar": st.sampled_from([False, False, False, True]),
 })
+
+TINY_SCHEMA = '{"type": "object", "properties": {"k": {"type": "boolean"}}}'
+
+
+def _sp(r):
+    import json  # noqa: F401
+
+    return SamplingParams(
+        temperature=r["temperature"],
+        max_new_tokens=r["max_new"],
+        ignore_eos=r["ignore_eos"],
+        stop_token_ids=[r["stop_tok"]] if r["stop_tok"] else [],
+        json_schema=TINY_SCHEMA if r.get("grammar") else None,
+    )
 
 
 @settings(max_examples=25, deadline=None,
@@ -63,6 +80,11 @@ req_strategy = st.fixed_dictionaries({
        seed=st.integers(min_value=0, max_value=3))
 def test_lifecycle_invariants(reqs, submit_gaps, seed):
     eng = _engine(seed)
+    if any(r.get("grammar") for r in reqs):
+        eng.set_grammar_vocab(
+            [""] + ['{"k":', "true", "false", "}", '{"k"', ":t", "rue}"]
+            + [f"<{i}>" for i in range(90)]
+        )
     free0 = eng.cache_manager.num_free_blocks
     assert free0 == NUM_BLOCKS
 
@@ -76,13 +98,7 @@ def test_lifecycle_invariants(reqs, submit_gaps, seed):
         if pending and (steps % (1 + submit_gaps[pending[0][0] % 6]) == 0):
             i, r = pending.pop(0)
             rid = f"f{i}"
-            sp = SamplingParams(
-                temperature=r["temperature"],
-                max_new_tokens=r["max_new"],
-                ignore_eos=r["ignore_eos"],
-                stop_token_ids=[r["stop_tok"]] if r["stop_tok"] else [],
-            )
-            eng.submit(list(range(3, 3 + r["prompt_len"])), sp, rid=rid)
+            eng.submit(list(range(3, 3 + r["prompt_len"])), _sp(r), rid=rid)
             submitted.add(rid)
             if r["abort_after"] is not None:
                 abort_at[rid] = steps + r["abort_after"]
