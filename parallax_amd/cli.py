@@ -87,7 +87,11 @@ def cmd_serve(args) -> None:
     from .server.tokenizer_util import TokenizerWrapper
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
-    comm = init_distributed(pp_size=world, tp_size=1)
+    tp = max(1, getattr(args, "tp_size", 1))
+    if world % tp != 0:
+        raise SystemExit(f"--tp-size {tp} must divide the launched world "
+                         f"size {world} (torchrun --nproc-per-node)")
+    comm = init_distributed(pp_size=world // tp, tp_size=tp)
     if args.model_path:
         cfg = ModelConfig.from_pretrained(args.model_path)
     else:
@@ -170,6 +174,9 @@ def add_engine_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--seed", type=int, default=0, help="sampling seed")
     p.add_argument("--enable-routing-stats", action="store_true",
                    help="per-expert MoE routing counters (GET /stats)")
+    p.add_argument("--tp-size", type=int, default=1,
+                   help="tensor-parallel degree; ranks = tp x pp "
+                        "(torchrun --nproc-per-node must equal tp*pp)")
     p.add_argument("--start-layer", type=int, default=None,
                    help="explicit layer range start (decentralized mode)")
     p.add_argument("--end-layer", type=int, default=None)

@@ -243,3 +243,13 @@ def test_incremental_detokenizer_matches_full_decode():
     detok2 = IncrementalDetokenizer(tok)
     got = detok2.push(ids[:3]) + detok2.push(ids[3:])
     assert got == tok.decode(ids)
+
+
+def test_serve_tp_size_validation(monkeypatch):
+    """--tp-size must divide the launched world size."""
+    import parallax_amd.cli as cli
+
+    monkeypatch.setenv("WORLD_SIZE", "3")
+    with pytest.raises(SystemExit):
+        cli.cmd_serve(type("A", (), {"tp_size": 2, "model_path": None,
+                                     "model": "x", "model_name": None})())
