@@ -208,25 +208,38 @@ def create_app(
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        token_ids, finish_reason, logprobs, stop_at = await _collect(stream, sp)
-        content = tokenizer.decode(token_ids)
-        if stop_at >= 0:
-            content = content[:stop_at]
-            finish_reason = "stop"
-        return JSONResponse({
-            "id": rid, "object": "chat.completion", "created": created,
-            "model": model_name,
-            "choices": [{
-                "index": 0,
+        streams = [stream]
+        n = max(1, int(body.get("n", 1)))
+        for k in range(1, n):  # extra choices share the prompt (prefix cache)
+            streams.append(server.submit(
+                prompt_ids, sp, rid=f"{rid}-{k}",
+                aio_loop=asyncio.get_running_loop(),
+                aio_queue=asyncio.Queue(),
+            ))
+        results = await asyncio.gather(*(_collect(st, sp) for st in streams))
+        choices, total_completion = [], 0
+        for idx, (token_ids, finish_reason, logprobs, stop_at) in \
+                enumerate(results):
+            content = tokenizer.decode(token_ids)
+            if stop_at >= 0:
+                content = content[:stop_at]
+                finish_reason = "stop"
+            total_completion += len(token_ids)
+            choices.append({
+                "index": idx,
                 "message": {"role": "assistant", "content": content},
                 "logprobs": _chat_logprobs(token_ids, logprobs)
                 if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,
-            }],
+            })
+        return JSONResponse({
+            "id": rid, "object": "chat.completion", "created": created,
+            "model": model_name,
+            "choices": choices,
             "usage": {
                 "prompt_tokens": len(prompt_ids),
-                "completion_tokens": len(token_ids),
-                "total_tokens": len(prompt_ids) + len(token_ids),
+                "completion_tokens": total_completion,
+                "total_tokens": len(prompt_ids) + total_completion,
             },
         })
 
@@ -310,24 +323,37 @@ def create_app(
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        token_ids, finish_reason, logprobs, stop_at = await _collect(stream, sp)
-        text = tokenizer.decode(token_ids)
-        if stop_at >= 0:
-            text = text[:stop_at]
-            finish_reason = "stop"
-        return JSONResponse({
-            "id": rid, "object": "text_completion", "created": created,
-            "model": model_name,
-            "choices": [{
-                "index": 0, "text": text,
+        streams = [stream]
+        n = max(1, int(body.get("n", 1)))
+        for k in range(1, n):
+            streams.append(server.submit(
+                prompt_ids, sp, rid=f"{rid}-{k}",
+                aio_loop=asyncio.get_running_loop(),
+                aio_queue=asyncio.Queue(),
+            ))
+        results = await asyncio.gather(*(_collect(st, sp) for st in streams))
+        choices, total_completion = [], 0
+        for idx, (token_ids, finish_reason, logprobs, stop_at) in \
+                enumerate(results):
+            text = tokenizer.decode(token_ids)
+            if stop_at >= 0:
+                text = text[:stop_at]
+                finish_reason = "stop"
+            total_completion += len(token_ids)
+            choices.append({
+                "index": idx, "text": text,
                 "logprobs": {"token_logprobs": logprobs}
                 if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,
-            }],
+            })
+        return JSONResponse({
+            "id": rid, "object": "text_completion", "created": created,
+            "model": model_name,
+            "choices": choices,
             "usage": {
                 "prompt_tokens": len(prompt_ids),
-                "completion_tokens": len(token_ids),
-                "total_tokens": len(prompt_ids) + len(token_ids),
+                "completion_tokens": total_completion,
+                "total_tokens": len(prompt_ids) + total_completion,
             },
         })
 

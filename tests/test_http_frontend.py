@@ -253,3 +253,22 @@ def test_serve_tp_size_validation(monkeypatch):
     with pytest.raises(SystemExit):
         cli.cmd_serve(type("A", (), {"tp_size": 2, "model_path": None,
                                      "model": "x", "model_name": None})())
+
+
+def test_n_choices(client):
+    """OpenAI n>1: multiple independent completions per request."""
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-test-model",
+        "messages": [{"role": "user", "content": "hello"}],
+        "max_tokens": 4, "temperature": 1.0, "n": 3, "ignore_eos": True,
+    })
+    assert r.status_code == 200
+    body = r.json()
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+    assert body["usage"]["completion_tokens"] == 12
+    r2 = client.post("/v1/completions", json={
+        "prompt": [5, 9, 13], "max_tokens": 3, "n": 2, "temperature": 0.0,
+        "ignore_eos": True,
+    })
+    ch = r2.json()["choices"]
+    assert len(ch) == 2 and ch[0]["text"] == ch[1]["text"]  # greedy: identical
