@@ -76,6 +76,14 @@ class Sampler:
         assert B == len(reqs)
         sp = [r.sampling_params for r in reqs]
         logits = self._apply_grammar_masks(logits, reqs)
+        if any(s.logit_bias for s in sp):
+            logits = logits.clone()
+            V = logits.shape[-1]
+            for i, s_i in enumerate(sp):
+                if s_i.logit_bias:
+                    for tid, b in s_i.logit_bias.items():
+                        if 0 <= tid < V:
+                            logits[i, tid] += b
         need_penalties = any(
             s.repetition_penalty != 1.0 or s.presence_penalty != 0.0
             or s.frequency_penalty != 0.0

@@ -7,7 +7,7 @@ Capability parity with the reference's src/parallax/server/sampling/sampling_par
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional, Union
+from typing import Dict, List, Optional, Union
 
 
 @dataclass
@@ -27,6 +27,8 @@ class SamplingParams:
     seed: Optional[int] = None
     # structured output hook (reference carries json_schema through the wire)
     json_schema: Optional[str] = None
+    # OpenAI logit_bias: token id -> additive bias (applied pre-sampling)
+    logit_bias: Optional[Dict[int, float]] = None
     # return per-token logprob of the sampled token (reference wire field
     # token_prob / return_probs, proto forward.proto)
     logprobs: bool = False
@@ -64,6 +66,7 @@ class SamplingParams:
             "seed": self.seed,
             "json_schema": self.json_schema,
             "logprobs": self.logprobs,
+            "logit_bias": self.logit_bias,
         }
 
     @classmethod
@@ -96,6 +99,9 @@ class SamplingParams:
             seed=body.get("seed"),
             logprobs=bool(body.get("logprobs", False)),
             json_schema=_extract_json_schema(body),
+            logit_bias={int(k): float(v)
+                        for k, v in body["logit_bias"].items()}
+            if isinstance(body.get("logit_bias"), dict) else None,
         )
 
 

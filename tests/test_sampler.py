@@ -86,3 +86,34 @@ def test_sampler_logprobs_only_for_requesting_rows():
     assert out[0][0] == 1 and out[1][0] == 1
     assert out[0][1] is not None and out[0][1] <= 0.0
     assert out[1][1] is None
+
+
+def test_logit_bias_forces_token():
+    """A large positive bias makes the token win at temperature 0."""
+    import torch
+
+    from parallax_amd.server.request import InitialRequest
+    from parallax_amd.server.sampler import Sampler
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    s = Sampler(torch.device("cpu"), seed=0)
+    logits = torch.randn(2, 32)
+    reqs = [
+        InitialRequest(rid="a", prompt_token_ids=[1],
+                       sampling_params=SamplingParams(
+                           temperature=0.0, logit_bias={7: 100.0})),
+        InitialRequest(rid="b", prompt_token_ids=[1],
+                       sampling_params=SamplingParams(temperature=0.0)),
+    ]
+    toks, _ = s.sample_device(logits, reqs)
+    assert toks[0].item() == 7
+    assert toks[1].item() == logits[1].argmax().item()
+
+
+def test_logit_bias_openai_parse():
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    sp = SamplingParams.from_openai({"logit_bias": {"42": -5, "7": 3.5}})
+    assert sp.logit_bias == {42: -5.0, 7: 3.5}
+    rt = SamplingParams.from_dict(sp.to_dict())
+    assert rt.logit_bias == sp.logit_bias
