@@ -333,6 +333,7 @@ def create_app(
             ))
         results = await asyncio.gather(*(_collect(st, sp) for st in streams))
         choices, total_completion = [], 0
+        echo_text = tokenizer.decode(prompt_ids) if body.get("echo") else ""
         for idx, (token_ids, finish_reason, logprobs, stop_at) in \
                 enumerate(results):
             text = tokenizer.decode(token_ids)
@@ -341,7 +342,7 @@ def create_app(
                 finish_reason = "stop"
             total_completion += len(token_ids)
             choices.append({
-                "index": idx, "text": text,
+                "index": idx, "text": echo_text + text,
                 "logprobs": {"token_logprobs": logprobs}
                 if sp.logprobs and logprobs else None,
                 "finish_reason": finish_reason,
