@@ -227,3 +227,80 @@ def test_prefix_cache_accounting(reqs, share_prefix, seed):
     cm = eng.cache_manager
     assert cm.allocator.num_free_blocks + cm.radix.num_cached_blocks \
         == NUM_BLOCKS
+
+
+@settings(max_examples=10, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(reqs=st.lists(req_strategy, min_size=1, max_size=4),
+       n_stages=st.sampled_from([2, 3]))
+def test_p2p_pipeline_fuzz(reqs, n_stages):
+    """Random request mixes over a loopback PeerExecutor chain: every request
+    terminates, tokens match the single-host engine (greedy), and every
+    peer's cache state drains."""
+    from parallax_amd.p2p.peer_executor import PeerExecutor
+    from parallax_amd.p2p.transport import LoopbackTransport
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=97, hidden_size=32,
+        num_layers=4, num_heads=2, num_kv_heads=1, head_dim=16,
+        intermediate_size=64, max_position_embeddings=256,
+        eos_token_ids=[5],
+    )
+    from tests.test_pipeline_parallel import full_state_dict
+
+    # one HF-style weight set shared by the single-host engine and the chain
+    sd = full_state_dict(cfg)
+    comm = CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                       pp_rank=0, tp_rank=0, device=torch.device("cpu"))
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=96,
+                                 dtype=torch.float32, seed=0), comm=comm)
+    for name, w in sd.items():
+        eng.model.load_hf_weight(name, w)
+
+    sps = []
+    for r in reqs:
+        sps.append(SamplingParams(
+            temperature=0.0, max_new_tokens=r["max_new"],
+            ignore_eos=r["ignore_eos"],
+            stop_token_ids=[r["stop_tok"]] if r["stop_tok"] else [],
+        ))
+    prompts = [list(range(3, 3 + r["prompt_len"])) for r in reqs]
+    expected = eng.generate(prompts, sps)
+
+    registry = {}
+    peer_ids = [f"fz{i}" for i in range(n_stages)]
+    base, rem = divmod(cfg.num_layers, n_stages)
+    spans, pos = [], 0
+    for i in range(n_stages):
+        n = base + (1 if i < rem else 0)
+        spans.append((pos, pos + n))
+        pos += n
+    peers = []
+    for pid, (s0, e0) in zip(peer_ids, spans):
+        t = LoopbackTransport(pid, registry)
+        px = PeerExecutor(cfg, s0, e0, pid, t, dtype=torch.float32,
+                          num_kv_blocks=96, block_size=8, seed=0)
+        for name, w in sd.items():
+            px.model.load_hf_weight(name, w)
+        peers.append(px)
+    head = peers[0]
+    rids = [head.submit(p, sp, peer_ids) for p, sp in zip(prompts, sps)]
+    got = {rid: [] for rid in rids}
+    done = set()
+    for _ in range(3000):
+        for px in peers:
+            px.step(recv_timeout=0.0005)
+        for out in head.drain_outputs():
+            if out.token_id >= 0:
+                got[out.rid].append(out.token_id)
+            if out.finished:
+                done.add(out.rid)
+        if len(done) == len(rids):
+            break
+    assert len(done) == len(rids), "pipeline failed to drain"
+    assert [got[r] for r in rids] == list(expected.values())
+    for _ in range(30):  # let release packets land
+        for px in peers:
+            px.step(recv_timeout=0.0005)
+    for px in peers[1:]:
+        assert not px._peer_positions
