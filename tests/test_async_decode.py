@@ -157,3 +157,54 @@ def test_stop_token_finishes_on_async_path(monkeypatch):
     # deferred cache frees flushed: every block back in the pool
     assert eng.cache_manager.num_free_blocks == 128
     assert not eng._deferred_free and eng._inflight is None
+
+
+def test_min_new_tokens_defers_stop_on_async_path(monkeypatch):
+    """A stop token sampled before min_new_tokens must not finish the
+    request — on the async path the finish check runs one step late but
+    with identical semantics."""
+    probe = [SamplingParams(temperature=0.0, max_new_tokens=10,
+                            ignore_eos=True)] * 2
+    base, _, _ = _run_counted(False, probe, PROMPTS, monkeypatch)
+    toks0 = list(base.values())[0]
+    stop = toks0[1]  # would stop after token 2
+    if stop in toks0[4:]:  # need a token that doesn't recur later
+        stop = None
+    sps = [SamplingParams(temperature=0.0, max_new_tokens=10, ignore_eos=True,
+                          stop_token_ids=[stop] if stop is not None else [],
+                          min_new_tokens=6)] * 2
+    a, n_async, _ = _run_counted(True, sps, PROMPTS, monkeypatch)
+    b, _, _ = _run_counted(False, sps, PROMPTS, monkeypatch)
+    assert list(a.values()) == list(b.values())
+    if stop is not None:
+        # the early stop token was ignored (min_new_tokens), so request 0
+        # ran past position 2
+        assert len(list(a.values())[0]) >= 6
+
+
+def test_abort_races_async_finish(monkeypatch):
+    """Abort submitted the same step a request finishes one-step-late must
+    not double-release or hang."""
+    from parallax_amd.server.engine import Engine, EngineArgs
+
+    monkeypatch.setattr(engine_mod, "ASYNC_DECODE", True)
+    eng = Engine(_cfg(), EngineArgs(block_size=8, num_kv_blocks=128,
+                                    dtype=torch.float32, seed=7,
+                                    enable_prefix_cache=False),
+                 comm=_cpu_comm(), random_weights=True)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=8, ignore_eos=True)
+    eng.submit(PROMPTS[0], sp, rid="r0")
+    eng.submit(PROMPTS[1], sp, rid="r1")
+    finished = {}
+    for i in range(200):
+        if i == 3:
+            eng.abort("r0")
+            eng.abort("r0")  # double abort is idempotent
+        for out in eng.step():
+            if out.finished:
+                finished[out.rid] = out.finish_reason
+        if not eng.has_work:
+            break
+    assert set(finished) == {"r0", "r1"}
+    assert finished["r0"] == "abort" and finished["r1"] == "length"
+    assert eng.cache_manager.num_free_blocks == 128
