@@ -96,7 +96,7 @@ DEVINL float block_reduce_sum(float v, float* lds_scratch) {
   if (lane == 0) lds_scratch[wid] = v;
   __syncthreads();
   int nw = (blockDim.x + WAVE_SIZE - 1) / WAVE_SIZE;
-  v = (threadIdx.x < nw) ? lds_scratch[threadIdx.x] : 0.f;
+  v = (threadIdx.x < (unsigned)nw) ? lds_scratch[threadIdx.x] : 0.f;
   if (wid == 0) {
 #pragma unroll
     for (int off = MAX_WAVES / 2; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE_SIZE);
