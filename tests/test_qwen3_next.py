@@ -201,3 +201,33 @@ def test_q35_moe_greedy_parity():
         [prompt], [SamplingParams(temperature=0.0, max_new_tokens=5, ignore_eos=True)]
     )
     assert list(out.values())[0] == ref
+
+
+def test_async_late_finish_matches_sync_hybrid(tiny_next, monkeypatch):
+    """Hybrid (linear-state) stacks under the async pipeline with
+    value-dependent finishes: a zombie step writes a finished request's
+    conv/recurrent state slot, which must stay allocated until the in-flight
+    event drains — results must equal the sync engine exactly."""
+    from parallax_amd.server import engine as engine_mod
+
+    hf, cfg = tiny_next
+    prompts = [[7, 42, 99, 5, 81], [3, 9, 13, 2]]
+
+    def run(async_on, sps):
+        monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+        eng = _engine_with_weights(hf, cfg)
+        return eng.generate(prompts, sps), eng
+
+    probe = [SamplingParams(temperature=0.0, max_new_tokens=10,
+                            ignore_eos=True)] * 2
+    base, _ = run(False, probe)
+    stop = list(base.values())[0][3]  # stop request 0 mid-stream
+
+    sps = [SamplingParams(temperature=0.0, max_new_tokens=10, ignore_eos=True,
+                          stop_token_ids=[stop])] * 2
+    a, eng_a = run(True, sps)
+    b, _ = run(False, sps)
+    assert list(a.values()) == list(b.values())
+    # all slots drained: linear-state slots and KV blocks back in the pools
+    assert not eng_a.scheduler.running
+    assert eng_a._inflight is None and not eng_a._deferred_free
