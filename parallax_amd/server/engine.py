@@ -520,7 +520,8 @@ class Engine:
             # synced) — so the common serving case keeps the pipelining.
             # logprobs/json_schema/penalties stay sync: they need token
             # VALUES on the host before the next forward.
-            if sp.logprobs or sp.json_schema or sp.logit_bias:
+            if sp.logprobs or sp.json_schema or sp.logit_bias \
+                    or sp.seed is not None:
                 return False
             if (sp.repetition_penalty != 1.0 or sp.presence_penalty != 0.0
                     or sp.frequency_penalty != 0.0):

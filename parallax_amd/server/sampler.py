@@ -106,6 +106,21 @@ class Sampler:
             [s.min_p for s in sp],
             generator=self.generator,
         )
+        # per-request seed (OpenAI `seed`): rows re-sample with their own
+        # generator keyed by (seed, position), so the draw is reproducible
+        # regardless of batch composition
+        for i, s_i in enumerate(sp):
+            if s_i.seed is not None and s_i.temperature > 0.0:
+                from ..ops import reference as ref_ops
+
+                g = torch.Generator(device=logits.device)
+                g.manual_seed((int(s_i.seed) * 1000003
+                               + len(reqs[i].output_token_ids))
+                              & 0x7FFFFFFFFFFFFFFF)
+                tokens[i] = ref_ops.sample_tokens(
+                    logits[i:i + 1], [s_i.temperature], [s_i.top_p],
+                    [s_i.top_k], [s_i.min_p], g,
+                )[0]
         if want_logprobs is None:
             want_logprobs = any(s.logprobs for s in sp)
         if not want_logprobs:

@@ -117,3 +117,42 @@ def test_logit_bias_openai_parse():
     assert sp.logit_bias == {42: -5.0, 7: 3.5}
     rt = SamplingParams.from_dict(sp.to_dict())
     assert rt.logit_bias == sp.logit_bias
+
+
+def test_per_request_seed_reproducible():
+    """Same seed -> same sampled stream regardless of batch composition."""
+    import torch
+
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=211, hidden_size=64,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+        intermediate_size=128, max_position_embeddings=256, eos_token_ids=[],
+    )
+
+    def make():
+        comm = CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                           pp_rank=0, tp_rank=0,
+                           device=torch.device("cpu"))
+        return Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                      dtype=torch.float32, seed=0),
+                      comm=comm, random_weights=True)
+
+    sp_seeded = SamplingParams(temperature=0.9, max_new_tokens=6,
+                               ignore_eos=True, seed=1234)
+    sp_other = SamplingParams(temperature=0.9, max_new_tokens=6,
+                              ignore_eos=True)
+    # run 1: alone; run 2: alongside another request (different batch shape)
+    a = make().generate([[5, 9, 13]], [sp_seeded])
+    eng = make()
+    b = eng.generate([[5, 9, 13], [7, 8]], [sp_seeded, sp_other])
+    assert list(a.values())[0] == list(b.values())[0]
+    # a different seed diverges
+    sp2 = SamplingParams(temperature=0.9, max_new_tokens=6,
+                         ignore_eos=True, seed=99)
+    c = make().generate([[5, 9, 13]], [sp2])
+    assert list(c.values())[0] != list(a.values())[0]
