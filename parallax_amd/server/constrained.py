@@ -22,6 +22,7 @@ tests each vocab string against the automaton; masks are cached per
 from __future__ import annotations
 
 import json
+from collections import OrderedDict
 from typing import Dict, List, Optional, Tuple
 
 State = Tuple  # tuple of frames; () == complete
@@ -381,7 +382,10 @@ class GrammarMatcher:
     are cached per (schema, state) across all requests."""
 
     _fsm_cache: Dict[str, JsonSchemaFSM] = {}
-    _mask_cache: Dict[Tuple[str, State], List[int]] = {}
+    # mask cache is LRU-bounded: recursive schemas produce unboundedly many
+    # distinct automaton states over a long-lived server
+    _mask_cache: "OrderedDict[Tuple[str, State], List[int]]" = OrderedDict()
+    _mask_cache_max = 20000
     _trie_cache: Dict[int, tuple] = {}  # id(vocab) -> (vocab, trie root)
 
     def __init__(self, schema_json: str, vocab: List[str], eos_ids: List[int]):
@@ -451,4 +455,8 @@ class GrammarMatcher:
                         stack.append((nxt, child))
             ids.sort()
             self._mask_cache[key] = ids
+            if len(self._mask_cache) > self._mask_cache_max:
+                self._mask_cache.popitem(last=False)
+        else:
+            self._mask_cache.move_to_end(key)
         return ids
