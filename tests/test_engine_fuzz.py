@@ -27,7 +27,7 @@ from parallax_amd.server.sampling_params import SamplingParams
 NUM_BLOCKS = 96
 
 
-def _engine(seed: int) -> Engine:
+def _engine(seed: int, prefix_cache: bool = False) -> Engine:
     cfg = ModelConfig(
         architecture="LlamaForCausalLM", vocab_size=97, hidden_size=32,
         num_layers=2, num_heads=2, num_kv_heads=1, head_dim=16,
@@ -40,7 +40,7 @@ def _engine(seed: int) -> Engine:
         cfg,
         EngineArgs(block_size=8, num_kv_blocks=NUM_BLOCKS,
                    dtype=torch.float32, seed=seed,
-                   enable_prefix_cache=False, max_batch_size=8),
+                   enable_prefix_cache=prefix_cache, max_batch_size=8),
         comm=comm, random_weights=True,
     )
 
@@ -136,8 +136,9 @@ def test_lifecycle_invariants(reqs, submit_gaps, seed):
 @settings(max_examples=20, deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(reqs=st.lists(req_strategy, min_size=1, max_size=5),
-       seed=st.integers(min_value=0, max_value=2))
-def test_async_pipelining_equivalence(reqs, seed):
+       seed=st.integers(min_value=0, max_value=2),
+       share_prefix=st.booleans())
+def test_async_pipelining_equivalence(reqs, seed, share_prefix):
     """For any mix of eos-sensitive / stop-token GREEDY requests, the
     async-pipelined engine must emit exactly the sync engine's tokens and
     finish reasons (the one-step-late rollback is unobservable). Sampled
@@ -151,7 +152,7 @@ def test_async_pipelining_equivalence(reqs, seed):
         old = engine_mod.ASYNC_DECODE
         engine_mod.ASYNC_DECODE = async_on
         try:
-            eng = _engine(seed)
+            eng = _engine(seed, prefix_cache=share_prefix)
             sps, prompts, rids = [], [], []
             for i, r in enumerate(reqs):
                 sps.append(SamplingParams(
@@ -160,7 +161,13 @@ def test_async_pipelining_equivalence(reqs, seed):
                     ignore_eos=r["ignore_eos"],
                     stop_token_ids=[r["stop_tok"]] if r["stop_tok"] else [],
                 ))
-                prompts.append(list(range(3, 3 + r["prompt_len"])))
+                if share_prefix:
+                    # same long base + unique tail: later requests admit
+                    # with most tokens prefix-cache hit (needs an engine
+                    # with the radix enabled — see below)
+                    prompts.append(list(range(3, 27)) + [60 + len(prompts)])
+                else:
+                    prompts.append(list(range(3, 3 + r["prompt_len"])))
             for i, (p, sp) in enumerate(zip(prompts, sps)):
                 eng.submit(p, sp, rid=f"e{i}")
             tokens, finish = {}, {}
