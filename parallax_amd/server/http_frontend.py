@@ -124,6 +124,89 @@ def create_app(
             n = server.engine.update_weights_from_disk(path)
         return {"updated_tensors": n}
 
+    async def _sse_multi(streams, all_rids, make_chunk, make_usage):
+        """Merge n request streams into one SSE stream (OpenAI streams n>1
+        as interleaved chunks labeled by choice index). Each pump drains its
+        queue greedily and detokenizes incrementally; client disconnect
+        aborts every underlying request."""
+        q: asyncio.Queue = asyncio.Queue()
+
+        async def pump(idx, st):
+            detok = IncrementalDetokenizer(tokenizer)
+            first_t = None
+            done = False
+            while not done:
+                out = await st.aget()
+                if out is None:
+                    break
+                # greedy drain: under load several tokens are already queued
+                # — one chunk per wakeup keeps the loop at O(steps), not
+                # O(tokens) (per-token SSE capped the loop at ~3k tok/s)
+                batch = [out]
+                while True:
+                    try:
+                        nxt = st.aio_queue.get_nowait()
+                    except asyncio.QueueEmpty:
+                        break
+                    if nxt is None:
+                        done = True
+                        break
+                    batch.append(nxt)
+                if first_t is None:
+                    first_t = time.monotonic()
+                finish = None
+                new_ids = []
+                for o in batch:
+                    if o.token_id >= 0:
+                        new_ids.append(o.token_id)
+                    if o.finished:
+                        finish = o.finish_reason or "stop"
+                delta = detok.push(new_ids) if new_ids else ""
+                if delta or finish is not None:
+                    await q.put(("chunk", idx, delta, finish))
+            await q.put(("end", idx, len(detok.ids), first_t))
+
+        tasks = [asyncio.create_task(pump(i, st))
+                 for i, st in enumerate(streams)]
+        t_start = time.monotonic()
+        first_any = None
+        total_tokens = 0
+        active = len(streams)
+        try:
+            while active:
+                kind, idx, a, b = await q.get()
+                if kind == "chunk":
+                    if first_any is None:
+                        first_any = time.monotonic()
+                    yield make_chunk(idx, a, b)
+                else:
+                    active -= 1
+                    total_tokens += a
+                    if first_any is None and b is not None:
+                        first_any = b
+            yield make_usage(total_tokens,
+                             first_any or time.monotonic(), t_start)
+            yield "data: [DONE]\n\n"
+        except asyncio.CancelledError:
+            # client went away mid-stream: stop generating everywhere
+            for t in tasks:
+                t.cancel()
+            for r in all_rids:
+                server.abort(r)
+            raise
+
+    def _submit_n(prompt_ids, sp, rid, n):
+        """Submit the primary + n-1 extra same-prompt requests (the prefix
+        cache makes the extra prefills nearly free)."""
+        loop = asyncio.get_running_loop()
+        streams = [server.submit(prompt_ids, sp, rid=rid,
+                                 aio_loop=loop, aio_queue=asyncio.Queue())]
+        for k in range(1, n):
+            streams.append(server.submit(
+                prompt_ids, sp, rid=f"{rid}-{k}",
+                aio_loop=loop, aio_queue=asyncio.Queue()))
+        return streams
+
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
         body = await request.json()
@@ -133,89 +216,47 @@ def create_app(
         prompt_ids = tokenizer.chat_prompt_ids(messages)
         sp = _params(body)
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        n = max(1, int(body.get("n", 1)))
         try:
-            stream = server.submit(
-                prompt_ids, sp, rid=rid,
-                aio_loop=asyncio.get_running_loop(), aio_queue=asyncio.Queue(),
-            )
+            streams = _submit_n(prompt_ids, sp, rid, n)
         except ValueError as e:  # over context limit etc -> clean client error
             raise HTTPException(400, str(e))
         created = int(time.time())
 
         if body.get("stream"):
-            async def sse():
-              try:
-                detok = IncrementalDetokenizer(tokenizer)
-                token_ids = detok.ids
-                t_start = time.monotonic()
-                first_token_t = None
-                done = False
-                while not done:
-                    out = await stream.aget()
-                    if out is None:
-                        break
-                    # greedy drain: under load several tokens are already
-                    # queued — emit them as ONE chunk so the event loop does
-                    # O(wakeups) work, not O(tokens) (the per-token SSE path
-                    # caps a single asyncio loop at ~3k tok/s)
-                    batch = [out]
-                    while True:
-                        try:
-                            nxt = stream.aio_queue.get_nowait()
-                        except asyncio.QueueEmpty:
-                            break
-                        if nxt is None:
-                            done = True
-                            break
-                        batch.append(nxt)
-                    if first_token_t is None:
-                        first_token_t = time.monotonic()
-                    finish = None
-                    new_ids = []
-                    for o in batch:
-                        if o.token_id >= 0:
-                            new_ids.append(o.token_id)
-                        if o.finished:
-                            finish = o.finish_reason or "stop"
-                    delta = detok.push(new_ids) if new_ids else ""
-                    if not delta and finish is None:
-                        continue
-                    chunk = {
-                        "id": rid, "object": "chat.completion.chunk",
-                        "created": created, "model": model_name,
-                        "choices": [{
-                            "index": 0,
-                            "delta": {"content": delta},
-                            "finish_reason": finish,
-                        }],
-                    }
-                    yield f"data: {json.dumps(chunk)}\n\n"
+            def make_chunk(idx, delta, finish):
+                chunk = {
+                    "id": rid, "object": "chat.completion.chunk",
+                    "created": created, "model": model_name,
+                    "choices": [{
+                        "index": idx,
+                        "delta": {"content": delta},
+                        "finish_reason": finish,
+                    }],
+                }
+                return f"data: {json.dumps(chunk)}\n\n"
+
+            def make_usage(total, first_t, t_start):
                 # final usage chunk (reference logs TPS/TTFT from this)
                 elapsed = time.monotonic() - t_start
                 usage = {
                     "prompt_tokens": len(prompt_ids),
-                    "completion_tokens": len(token_ids),
-                    "total_tokens": len(prompt_ids) + len(token_ids),
-                    "ttft_ms": round(((first_token_t or time.monotonic()) - t_start) * 1e3, 2),
-                    "tps": round(len(token_ids) / max(elapsed, 1e-6), 2),
+                    "completion_tokens": total,
+                    "total_tokens": len(prompt_ids) + total,
+                    "ttft_ms": round((first_t - t_start) * 1e3, 2),
+                    "tps": round(total / max(elapsed, 1e-6), 2),
                 }
-                yield f"data: {json.dumps({'id': rid, 'object': 'chat.completion.chunk', 'created': created, 'model': model_name, 'choices': [], 'usage': usage})}\n\n"
-                yield "data: [DONE]\n\n"
-              except asyncio.CancelledError:
-                # client went away mid-stream: stop generating for it
-                server.abort(rid)
-                raise
+                return "data: " + json.dumps({
+                    "id": rid, "object": "chat.completion.chunk",
+                    "created": created, "model": model_name,
+                    "choices": [], "usage": usage,
+                }) + "\n\n"
 
-            return StreamingResponse(sse(), media_type="text/event-stream")
+            return StreamingResponse(
+                _sse_multi(streams, [st.rid for st in streams],
+                           make_chunk, make_usage),
+                media_type="text/event-stream")
 
-        streams = [stream]
-        n = max(1, int(body.get("n", 1)))
-        for k in range(1, n):  # extra choices share the prompt (prefix cache)
-            streams.append(server.submit(
-                prompt_ids, sp, rid=f"{rid}-{k}",
-                aio_loop=asyncio.get_running_loop(),
-                aio_queue=asyncio.Queue(),
-            ))
         results = await asyncio.gather(*(_collect(st, sp) for st in streams))
         choices, total_completion = [], 0
         for idx, (token_ids, finish_reason, logprobs, stop_at) in \
@@ -255,82 +296,45 @@ def create_app(
             prompt_ids = tokenizer.encode(prompt)
         sp = _params(body)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
+        n = max(1, int(body.get("n", 1)))
         try:
-            stream = server.submit(
-                prompt_ids, sp, rid=rid,
-                aio_loop=asyncio.get_running_loop(), aio_queue=asyncio.Queue(),
-            )
+            streams = _submit_n(prompt_ids, sp, rid, n)
         except ValueError as e:  # over context limit etc -> clean client error
             raise HTTPException(400, str(e))
         created = int(time.time())
 
         if body.get("stream"):
-            async def sse():
-              try:
-                detok = IncrementalDetokenizer(tokenizer)
-                token_ids = detok.ids
-                t_start = time.monotonic()
-                first_t = None
-                done = False
-                while not done:
-                    out = await stream.aget()
-                    if out is None:
-                        break
-                    batch = [out]  # greedy drain (see chat_completions)
-                    while True:
-                        try:
-                            nxt = stream.aio_queue.get_nowait()
-                        except asyncio.QueueEmpty:
-                            break
-                        if nxt is None:
-                            done = True
-                            break
-                        batch.append(nxt)
-                    if first_t is None:
-                        first_t = time.monotonic()
-                    finish = None
-                    new_ids = []
-                    for o in batch:
-                        if o.token_id >= 0:
-                            new_ids.append(o.token_id)
-                        if o.finished:
-                            finish = o.finish_reason or "stop"
-                    delta = detok.push(new_ids) if new_ids else ""
-                    if not delta and finish is None:
-                        continue
-                    chunk = {
-                        "id": rid, "object": "text_completion",
-                        "created": created, "model": model_name,
-                        "choices": [{
-                            "index": 0, "text": delta,
-                            "finish_reason": finish,
-                        }],
-                    }
-                    yield f"data: {json.dumps(chunk)}\n\n"
+            def make_chunk(idx, delta, finish):
+                chunk = {
+                    "id": rid, "object": "text_completion",
+                    "created": created, "model": model_name,
+                    "choices": [{
+                        "index": idx, "text": delta,
+                        "finish_reason": finish,
+                    }],
+                }
+                return f"data: {json.dumps(chunk)}\n\n"
+
+            def make_usage(total, first_t, t_start):
                 elapsed = time.monotonic() - t_start
                 usage = {
                     "prompt_tokens": len(prompt_ids),
-                    "completion_tokens": len(token_ids),
-                    "total_tokens": len(prompt_ids) + len(token_ids),
-                    "ttft_ms": round(((first_t or time.monotonic()) - t_start) * 1e3, 2),
-                    "tps": round(len(token_ids) / max(elapsed, 1e-6), 2),
+                    "completion_tokens": total,
+                    "total_tokens": len(prompt_ids) + total,
+                    "ttft_ms": round((first_t - t_start) * 1e3, 2),
+                    "tps": round(total / max(elapsed, 1e-6), 2),
                 }
-                yield f"data: {json.dumps({'id': rid, 'object': 'text_completion', 'created': created, 'model': model_name, 'choices': [], 'usage': usage})}\n\n"
-                yield "data: [DONE]\n\n"
-              except asyncio.CancelledError:
-                server.abort(rid)
-                raise
+                return "data: " + json.dumps({
+                    "id": rid, "object": "text_completion",
+                    "created": created, "model": model_name,
+                    "choices": [], "usage": usage,
+                }) + "\n\n"
 
-            return StreamingResponse(sse(), media_type="text/event-stream")
+            return StreamingResponse(
+                _sse_multi(streams, [st.rid for st in streams],
+                           make_chunk, make_usage),
+                media_type="text/event-stream")
 
-        streams = [stream]
-        n = max(1, int(body.get("n", 1)))
-        for k in range(1, n):
-            streams.append(server.submit(
-                prompt_ids, sp, rid=f"{rid}-{k}",
-                aio_loop=asyncio.get_running_loop(),
-                aio_queue=asyncio.Queue(),
-            ))
         results = await asyncio.gather(*(_collect(st, sp) for st in streams))
         choices, total_completion = [], 0
         echo_text = tokenizer.decode(prompt_ids) if body.get("echo") else ""

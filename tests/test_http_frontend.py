@@ -272,3 +272,27 @@ def test_n_choices(client):
     })
     ch = r2.json()["choices"]
     assert len(ch) == 2 and ch[0]["text"] == ch[1]["text"]  # greedy: identical
+
+
+def test_streaming_n_choices(client):
+    """stream + n>1: interleaved chunks labeled by choice index (previously
+    silently returned only choice 0)."""
+    import json as _json
+
+    seen_idx = set()
+    usage = None
+    with client.stream("POST", "/v1/completions", json={
+        "prompt": [5, 9, 13], "max_tokens": 4, "n": 2,
+        "temperature": 1.0, "ignore_eos": True, "stream": True,
+    }) as r:
+        assert r.status_code == 200
+        for line in r.iter_lines():
+            if not line.startswith("data: ") or line == "data: [DONE]":
+                continue
+            msg = _json.loads(line[6:])
+            for c in msg.get("choices", []):
+                seen_idx.add(c["index"])
+            if msg.get("usage"):
+                usage = msg["usage"]
+    assert seen_idx == {0, 1}
+    assert usage and usage["completion_tokens"] == 8
