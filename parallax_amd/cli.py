@@ -174,6 +174,9 @@ def add_engine_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--seed", type=int, default=0, help="sampling seed")
     p.add_argument("--enable-routing-stats", action="store_true",
                    help="per-expert MoE routing counters (GET /stats)")
+    p.add_argument("--decode-priority", action="store_true",
+                   help="form decode batches before prefills (lower TPOT "
+                        "under arrival bursts at the cost of TTFT)")
     p.add_argument("--tp-size", type=int, default=1,
                    help="tensor-parallel degree; ranks = tp x pp "
                         "(torchrun --nproc-per-node must equal tp*pp)")
@@ -219,6 +222,7 @@ def engine_args_from_cli(args, world: int = 1):
         enable_graphs=not args.disable_graphs,
         kv_cache_dtype=args.kv_cache_dtype,
         enable_routing_stats=args.enable_routing_stats,
+        prefill_priority=not args.decode_priority,
         moe_weight_dtype=args.moe_weight_dtype,
         linear_weight_dtype=args.linear_weight_dtype,
     )

@@ -77,6 +77,7 @@ class EngineArgs:
     kv_cache_dtype: str = "auto"        # "auto" (= engine dtype) | "fp8"(e4m3)
     moe_weight_dtype: str = "auto"      # "auto" (= engine dtype) | "fp8"(W8A8)
     enable_routing_stats: bool = False  # per-expert MoE routing counters
+    prefill_priority: bool = True       # prefills before decodes in a step
     linear_weight_dtype: str = "auto"   # dense GEMMs: "auto" | "fp8"(W8A8)
 
 
@@ -239,6 +240,7 @@ class Engine:
             prefill_chunk_size=args.prefill_chunk_size,
             request_timeout_s=args.request_timeout_s,
             eos_token_ids=cfg.eos_token_ids,
+            prefill_priority=args.prefill_priority,
         )
         self.sampler = Sampler(self.device, args.seed)
         self.graph_runner = None

@@ -74,6 +74,7 @@ class EngineServer:
         self.total_output_tokens = 0
         self.step_ms_ewma: Optional[float] = None
         self.last_batch_tokens = 0
+        self.tps_ewma: Optional[float] = None  # output tokens/s over steps
         # on-demand tracing (the reference has no tracer; here a
         # chrome trace of the live step loop is one POST /profile away)
         self._profile_req: Optional[tuple] = None  # (steps, path)
@@ -179,6 +180,9 @@ class EngineServer:
             self.step_ms_ewma = dt_ms if self.step_ms_ewma is None \
                 else 0.1 * dt_ms + 0.9 * self.step_ms_ewma
             self.last_batch_tokens = len(outputs)
+            inst_tps = len(outputs) / max(dt_ms * 1e-3, 1e-9)
+            self.tps_ewma = inst_tps if self.tps_ewma is None \
+                else 0.1 * inst_tps + 0.9 * self.tps_ewma
             # batch delivery: ONE call_soon_threadsafe per (step, loop), not
             # one per token — per-token signalling (lock + self-pipe write
             # ~15 us each) capped the whole server at ~3k tok/s when the
@@ -216,5 +220,7 @@ class EngineServer:
             if self.step_ms_ewma is not None else None,
             "last_batch_tokens": self.last_batch_tokens,
             "engine_steps": self.engine.step_count,
+            "output_tps_ewma": round(self.tps_ewma, 1)
+            if self.tps_ewma is not None else None,
             "moe_routing": self.engine.routing_stats(),
         }

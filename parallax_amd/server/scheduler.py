@@ -58,6 +58,7 @@ class Scheduler:
         prefill_chunk_size: int = 8192,
         request_timeout_s: float = 600.0,
         eos_token_ids: Optional[List[int]] = None,
+        prefill_priority: bool = True,
     ):
         self.cache = cache_manager
         self.max_batch_size = max_batch_size
@@ -67,6 +68,9 @@ class Scheduler:
         self.prefill_chunk_size = max(bs, (prefill_chunk_size // bs) * bs)
         self.request_timeout_s = request_timeout_s
         self.eos_token_ids = eos_token_ids or []
+        # reference --prefill-priority: prefills before decodes (default) or
+        # decodes first (lower TPOT under arrival bursts at the cost of TTFT)
+        self.prefill_priority = prefill_priority
 
         self.wait_queue: Deque[InitialRequest] = deque()
         self.running: Dict[str, InitialRequest] = {}
@@ -127,8 +131,10 @@ class Scheduler:
     def form_batch(self) -> ScheduledBatch:
         batch = ScheduledBatch()
         token_budget = self.max_num_tokens_per_batch
+        if not self.prefill_priority:
+            token_budget -= self._form_decodes(batch, token_budget)
 
-        # prefills first (possibly chunked)
+        # prefills (possibly chunked)
         for req in self.running.values():
             if token_budget <= 0:
                 break
@@ -147,9 +153,15 @@ class Scheduler:
                 )
                 token_budget -= chunk
 
-        # then decodes
+        if self.prefill_priority:
+            self._form_decodes(batch, token_budget)
+        return batch
+
+    def _form_decodes(self, batch: ScheduledBatch, token_budget: int) -> int:
+        used = 0
         for req in self.running.values():
-            if token_budget <= 0 or len(batch.decode_reqs) >= self.max_batch_size:
+            if used >= token_budget \
+                    or len(batch.decode_reqs) >= self.max_batch_size:
                 break
             if req.status is RequestStatus.DECODING and not req.is_finished:
                 try:
@@ -161,8 +173,8 @@ class Scheduler:
                     req.abort_requested = True
                     continue
                 batch.decode_reqs.append(req)
-                token_budget -= 1
-        return batch
+                used += 1
+        return used
 
     # -- step completion -----------------------------------------------------------
 

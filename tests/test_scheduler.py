@@ -109,3 +109,35 @@ def test_timeout_sweep():
     s.admit_requests()
     timed = s.sweep_timeouts()
     assert len(timed) == 1 and s.num_running == 0
+
+
+def test_decode_priority_ordering():
+    """prefill_priority=False forms decodes before prefills: with a tight
+    token budget the decode slots win."""
+    import torch
+
+    from parallax_amd.server.cache_manager import CacheManager
+    from parallax_amd.server.request import InitialRequest, RequestStatus
+    from parallax_amd.server.sampling_params import SamplingParams
+    from parallax_amd.server.scheduler import Scheduler
+
+    cache = CacheManager(8, 64, enable_prefix_cache=False)
+    sched = Scheduler(cache, max_num_tokens_per_batch=4,
+                      prefill_priority=False)
+    # one decoding request + one big pending prefill
+    r1 = InitialRequest(rid="d", prompt_token_ids=[1, 2, 3],
+                        sampling_params=SamplingParams(max_new_tokens=8,
+                                                       ignore_eos=True))
+    r2 = InitialRequest(rid="p", prompt_token_ids=list(range(3, 30)),
+                        sampling_params=SamplingParams(max_new_tokens=8,
+                                                       ignore_eos=True))
+    sched.add_request(r1)
+    sched.add_request(r2)
+    sched.admit_requests()
+    r1.num_prefilled_tokens = r1.prompt_len
+    r1.status = RequestStatus.DECODING
+    r1.commit_new_token(9)
+    batch = sched.form_batch()
+    assert [r.rid for r in batch.decode_reqs] == ["d"]
+    # decode took 1 of the 4 budget; prefill gets a page-aligned remainder
+    assert sum(c.num_tokens for c in batch.prefill_chunks) <= 3
