@@ -14,6 +14,9 @@ from typing import Dict, List, Optional, Tuple
 from .model_info import ModelInfo
 
 # memory split defaults (reference p2p/server.py:368-369)
+# reference defaults (p2p/server.py:368-369); overridable per node so a
+# deployment can trade parameter share against KV share (--param-mem-ratio /
+# --kvcache-mem-ratio in the reference's server_args)
 PARAM_MEM_RATIO = 0.65
 KVCACHE_MEM_RATIO = 0.25
 
@@ -68,6 +71,10 @@ class Node:
 
     # dynamic state from heartbeats
     current_requests: int = 0
+    # memory split (overridable per node; reference --param-mem-ratio /
+    # --kvcache-mem-ratio)
+    param_mem_ratio: float = PARAM_MEM_RATIO
+    kvcache_mem_ratio: float = KVCACHE_MEM_RATIO
     measured_layer_latency_ms: Optional[float] = None  # EWMA from node_update
     rtt_ms: Dict[str, float] = field(default_factory=dict)  # peer -> RTT
     last_heartbeat: float = field(default_factory=time.monotonic)
@@ -102,7 +109,7 @@ class Node:
         self, model: ModelInfo, is_first: bool = False, is_last: bool = False
     ) -> int:
         """Max decoder layers this node can host (reference node.py:274-307)."""
-        budget = self.hardware.total_memory_bytes * PARAM_MEM_RATIO
+        budget = self.hardware.total_memory_bytes * self.param_mem_ratio
         if is_first:
             budget -= model.embedding_io_bytes()
         if is_last and not model.cfg.tie_word_embeddings:
@@ -111,7 +118,7 @@ class Node:
         return max(0, int(budget // max(1, per_layer)))
 
     def kv_budget_bytes(self) -> int:
-        return int(self.hardware.total_memory_bytes * KVCACHE_MEM_RATIO)
+        return int(self.hardware.total_memory_bytes * self.kvcache_mem_ratio)
 
     def max_requests(self, model: ModelInfo, avg_context: int = 2048) -> int:
         """KV-budget-bounded concurrent batch size (reference node.py:212-246)."""
