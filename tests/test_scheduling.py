@@ -388,3 +388,88 @@ def test_allocators_structural_invariants(fracs, algo):
             used.add(n.node_id)
             cap = n.decoder_layer_capacity(model)
             assert n.end_layer - n.start_layer <= cap
+
+
+@pytest.mark.skipif(not HAVE_HYP, reason="hypothesis not installed")
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(fracs=st.lists(st.sampled_from([0.12, 0.2, 0.3, 0.5]),
+                      min_size=2, max_size=5))
+def test_dp_minimum_stages_vs_bruteforce(fracs):
+    """For every feasible pipeline count k, the DP's chosen partition must
+    use no more total stages than ANY brute-force partition of the pool
+    into k feasible pipelines (exhaustive set-partition check, N<=5)."""
+    from itertools import combinations
+
+    model = llama70b_info()
+    nodes = [build_node(f"b{i}", small_gpu(f, name=f"b{i}"))
+             for i, f in enumerate(fracs)]
+    set_full_rtt(nodes)
+    L = model.num_layers
+    caps = {n.node_id: n.decoder_layer_capacity(model) for n in nodes}
+
+    def feasible(group):
+        # position-aware: the first node also hosts the embedding and the
+        # last the lm_head, shrinking their layer capacity — pick the
+        # best (first, last) assignment for the group
+        if len(group) == 1:
+            n = group[0]
+            return n.decoder_layer_capacity(
+                model, is_first=True, is_last=True) >= L
+        best = -1
+        for f in group:
+            for last in group:
+                if f is last:
+                    continue
+                tot = sum(
+                    n.decoder_layer_capacity(
+                        model, is_first=(n is f), is_last=(n is last))
+                    for n in group
+                )
+                best = max(best, tot)
+        return best >= L
+
+    def best_bruteforce(k):
+        """Min total nodes used across partitions into k feasible groups."""
+        ids = list(range(len(nodes)))
+
+        def rec(remaining, k_left):
+            if k_left == 0:
+                return 0
+            if len(remaining) < k_left:
+                return None
+            best = None
+            # first group: any subset containing remaining[0]... also allow
+            # leaving nodes unused: choose subsets of remaining
+            head = remaining[0]
+            rest = remaining[1:]
+            # option: head unused
+            r = rec(rest, k_left)
+            if r is not None:
+                best = r
+            for sz in range(1, len(rest) + 2):
+                for combo in combinations(rest, sz - 1):
+                    group = [nodes[head]] + [nodes[c] for c in combo]
+                    if not feasible(group):
+                        continue
+                    rem2 = [x for x in rest if x not in combo]
+                    r = rec(rem2, k_left - 1)
+                    if r is not None:
+                        tot = sz + r
+                        if best is None or tot < best:
+                            best = tot
+            return best
+
+        return rec(ids, k)
+
+    alloc = DynamicProgrammingLayerAllocator(model)
+    pipes = alloc.allocate_from_standby(list(nodes))
+    if not pipes:
+        # DP says infeasible for every k: brute force must agree for k=1
+        assert best_bruteforce(1) is None
+        return
+    k = len(pipes)
+    dp_stages = sum(len(p.nodes) for p in pipes)
+    bf = best_bruteforce(k)
+    assert bf is not None
+    assert dp_stages <= bf, (dp_stages, bf, fracs, k)
