@@ -270,3 +270,49 @@ def test_string_length_bounds():
     assert accepts('{"s":"a\\n"}')     # escape counts as one char
     assert not accepts('{"s":"a"}')    # too short
     assert not accepts('{"s":"abcde"}')  # too long
+
+
+@pytest.mark.parametrize("schema,check", [
+    ({"type": "object",
+      "properties": {"a": {"type": "integer"}, "b": {"type": "boolean"},
+                     "c": {"type": "string", "maxLength": 6}},
+      "required": ["a"]},
+     lambda o: isinstance(o.get("a"), int)
+     and set(o) <= {"a", "b", "c"}),
+    ({"type": "object",
+      "properties": {"v": {"anyOf": [{"type": "null"},
+                                     {"type": "number"}]}}},
+     lambda o: o["v"] is None or isinstance(o["v"], (int, float))),
+    ({"$defs": {"t": {"type": "object",
+                      "properties": {"n": {"type": "integer"},
+                                     "k": {"type": "array",
+                                           "items": {"$ref": "#/$defs/t"},
+                                           "maxItems": 2}},
+                      "required": ["n"]}},
+      "$ref": "#/$defs/t"},
+     lambda o: isinstance(o["n"], int)),
+    ({"type": "object",
+      "additionalProperties": {"type": "integer"}},
+     lambda o: all(isinstance(v, int) for v in o.values())),
+])
+def test_masked_walk_conforms_across_features(schema, check):
+    """Random masked walks terminate in valid JSON for every schema feature
+    family (optional keys, unions, recursion, free-form objects)."""
+    import random
+
+    vocab = [""] * 3 + [chr(c) for c in range(32, 127)]
+    rng = random.Random(11)
+    for trial in range(4):
+        m = GrammarMatcher(json.dumps(schema), vocab, eos_ids=[2])
+        out = []
+        for _ in range(800):
+            m.catch_up(out)
+            ids = m.allowed_ids()
+            assert ids, "dead end in the automaton"
+            if ids == [2]:
+                break
+            out.append(rng.choice(ids))
+        else:
+            pytest.fail("walk did not terminate")
+        obj = json.loads("".join(vocab[t] for t in out))
+        assert check(obj)
