@@ -122,6 +122,8 @@ def cmd_serve(args) -> None:
 
 
 def cmd_chat(args) -> None:
+    import json as _json
+
     import httpx
 
     url = args.url.rstrip("/") + "/v1/chat/completions"
@@ -133,11 +135,23 @@ def cmd_chat(args) -> None:
         except EOFError:
             break
         history.append({"role": "user", "content": user})
-        r = httpx.post(url, json={"model": "default", "messages": history,
-                                  "max_tokens": args.max_tokens}, timeout=600)
-        msg = r.json()["choices"][0]["message"]
-        history.append(msg)
-        print(f"assistant> {msg['content']}")
+        print("assistant> ", end="", flush=True)
+        text = ""
+        with httpx.stream("POST", url, json={
+            "model": "default", "messages": history,
+            "max_tokens": args.max_tokens, "stream": True,
+        }, timeout=600) as r:
+            for line in r.iter_lines():
+                if not line.startswith("data: ") or line == "data: [DONE]":
+                    continue
+                msg = _json.loads(line[6:])
+                for c in msg.get("choices", []):
+                    delta = (c.get("delta") or {}).get("content") or ""
+                    if delta:
+                        text += delta
+                        print(delta, end="", flush=True)
+        print()
+        history.append({"role": "assistant", "content": text})
 
 
 def add_engine_args(p: argparse.ArgumentParser) -> None:
