@@ -131,3 +131,38 @@ def test_short_prompt_dense_fallback(tiny_dsv32):
         [prompt], [SamplingParams(temperature=0.0, max_new_tokens=3, ignore_eos=True)]
     )
     assert list(out.values())[0] == ref
+
+
+def test_dsa_engine_mixed_lifecycle(tiny_dsv32, monkeypatch):
+    """DSA (MLA + sparse indexer) engine under mixed finishes and aborts:
+    stop tokens, an abort mid-stream, and async-on/off equivalence — the
+    sparse index cache and latent cache must drain cleanly."""
+    from parallax_amd.server import engine as engine_mod
+
+    hf, cfg = tiny_dsv32
+
+    def run(async_on):
+        monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+        eng = _engine_with(hf, cfg)
+        sp_a = SamplingParams(temperature=0.0, max_new_tokens=8,
+                              ignore_eos=True)
+        eng.submit([7, 42, 99, 5, 81, 23, 150, 3], sp_a, rid="a")
+        eng.submit([3, 9, 13, 2, 7, 7, 7, 1], sp_a, rid="b")
+        tokens, finish = {}, {}
+        for i in range(200):
+            if i == 4:
+                eng.abort("b")
+            for out in eng.step():
+                if out.token_id >= 0:
+                    tokens.setdefault(out.rid, []).append(out.token_id)
+                if out.finished:
+                    finish[out.rid] = out.finish_reason
+            if not eng.has_work:
+                break
+        assert not eng.has_work
+        assert set(finish) == {"a", "b"}
+        assert finish["b"] == "abort" and finish["a"] == "length"
+        assert not eng.scheduler.running and eng._inflight is None
+        return tokens["a"]
+
+    assert run(True) == run(False)
