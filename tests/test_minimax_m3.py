@@ -101,3 +101,36 @@ def test_greedy_parity_chunked(tiny_m3):
     full = list(_engine_with(hf, cfg).generate([prompt], [sp]).values())[0]
     assert chunked[:3] == ref[:3]
     assert chunked == full
+
+
+def test_msa_engine_mixed_lifecycle(tiny_m3, monkeypatch):
+    """MSA sparse engine under mixed finishes/abort with async-on/off
+    equivalence (the index cache and topk selection ride the same step)."""
+    from parallax_amd.server import engine as engine_mod
+
+    hf, cfg = tiny_m3
+
+    def run(async_on):
+        monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+        eng = _engine_with(hf, cfg)
+        sp = SamplingParams(temperature=0.0, max_new_tokens=6,
+                            ignore_eos=True)
+        eng.submit([7, 42, 99, 5, 81, 23, 15, 3, 9, 1], sp, rid="a")
+        eng.submit([3, 9, 13, 2], sp, rid="b")
+        tokens, finish = {}, {}
+        for i in range(200):
+            if i == 3:
+                eng.abort("b")
+            for out in eng.step():
+                if out.token_id >= 0:
+                    tokens.setdefault(out.rid, []).append(out.token_id)
+                if out.finished:
+                    finish[out.rid] = out.finish_reason
+            if not eng.has_work:
+                break
+        assert set(finish) == {"a", "b"}
+        assert finish["b"] == "abort" and finish["a"] == "length"
+        assert not eng.scheduler.running and eng._inflight is None
+        return tokens["a"]
+
+    assert run(True) == run(False)
