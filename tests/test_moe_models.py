@@ -281,3 +281,56 @@ def test_routing_stats_counts():
     t = per_layer.pop()
     assert t % cfg.num_experts_per_tok == 0
     assert t >= 4 * cfg.num_experts_per_tok
+
+
+def test_gpt_oss_async_lifecycle(monkeypatch):
+    """gpt-oss (sliding window + sinks + biased MoE) under the async
+    pipeline with an abort and a stop finish: async == sync."""
+    from parallax_amd.models import get_model_class
+    from parallax_amd.server import engine as engine_mod
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    torch.manual_seed(11)
+    hf_cfg = transformers.GptOssConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, num_local_experts=4, num_experts_per_tok=2,
+        sliding_window=8, max_position_embeddings=512, rope_theta=10000.0,
+        layer_types=["sliding_attention", "full_attention"] * 2,
+        tie_word_embeddings=False, attention_bias=True,
+        rope_parameters={"rope_type": "default", "rope_theta": 10000.0},
+    )
+    hf = transformers.GptOssForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig.from_hf_config(
+        hf_cfg.to_dict() | {"architectures": ["GptOssForCausalLM"]}
+    )
+
+    def run(async_on):
+        monkeypatch.setattr(engine_mod, "ASYNC_DECODE", async_on)
+        eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=128,
+                                     dtype=torch.float32, max_batch_size=8))
+        m = get_model_class(cfg.architecture)(cfg).eval()
+        for name, t in hf.state_dict().items():
+            m.load_hf_weight(name, t)
+        eng.model = m.float()
+        sp = SamplingParams(temperature=0.0, max_new_tokens=7,
+                            ignore_eos=True)
+        eng.submit([7, 42, 99, 5, 81, 23, 15, 3, 9, 1, 2, 8], sp, rid="a")
+        eng.submit([3, 9, 13, 2, 7], sp, rid="b")
+        tokens, finish = {}, {}
+        for i in range(200):
+            if i == 3:
+                eng.abort("b")
+            for out in eng.step():
+                if out.token_id >= 0:
+                    tokens.setdefault(out.rid, []).append(out.token_id)
+                if out.finished:
+                    finish[out.rid] = out.finish_reason
+            if not eng.has_work:
+                break
+        assert set(finish) == {"a", "b"}
+        assert finish["b"] == "abort" and finish["a"] == "length"
+        return tokens["a"]
+
+    assert run(True) == run(False)
