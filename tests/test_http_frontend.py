@@ -332,3 +332,43 @@ def test_serving_soak_leak_free(client):
     st = client.get("/stats").json()
     assert st["running"] == 0 and st["waiting"] == 0
     assert st["total_requests"] >= 120
+
+
+def test_incremental_detok_bytelevel_bpe():
+    """The sliding-window algorithm must stay exact on REAL byte-level BPE,
+    including multi-byte UTF-8 sequences split across tokens (the U+FFFD
+    wait case)."""
+    tokenizers = pytest.importorskip("tokenizers")
+    from tokenizers import Tokenizer, decoders, models, pre_tokenizers, trainers
+
+    tok = Tokenizer(models.BPE(unk_token=None))
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(
+        vocab_size=300, special_tokens=["<eos>"],
+        initial_alphabet=pre_tokenizers.ByteLevel.alphabet(),
+    )
+    tok.train_from_iterator(
+        ["hello world", "straße café 東京", "the quick brown fox"], trainer)
+
+    class _HF:  # minimal adapter over the raw tokenizers object
+        eos_token_id = 0
+
+        def decode(self, ids, **kw):
+            return tok.decode(list(ids))
+
+        def encode(self, text):
+            return tok.encode(text).ids
+
+    from parallax_amd.server.tokenizer_util import (
+        IncrementalDetokenizer, TokenizerWrapper,
+    )
+
+    tw = TokenizerWrapper()
+    tw.hf = _HF()
+    for text in ["hello straße 東京 fox", "café café 東京東京",
+                 "the quick brown fox jumps"]:
+        ids = tok.encode(text).ids
+        detok = IncrementalDetokenizer(tw)
+        pieces = [detok.push([i]) for i in ids]
+        assert "".join(pieces) == tw.decode(ids)
