@@ -332,6 +332,8 @@ def test_serving_soak_leak_free(client):
     st = client.get("/stats").json()
     assert st["running"] == 0 and st["waiting"] == 0
     assert st["total_requests"] >= 120
+    assert st["output_tps_ewma"] is not None and st["output_tps_ewma"] > 0
+    assert st["step_ms_ewma"] is not None
 
 
 def test_incremental_detok_bytelevel_bpe():
