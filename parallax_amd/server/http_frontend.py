@@ -124,15 +124,20 @@ def create_app(
             n = server.engine.update_weights_from_disk(path)
         return {"updated_tensors": n}
 
-    async def _sse_multi(streams, all_rids, make_chunk, make_usage):
+    async def _sse_multi(streams, all_rids, make_chunk, make_usage,
+                         stop_strings=None):
         """Merge n request streams into one SSE stream (OpenAI streams n>1
         as interleaved chunks labeled by choice index). Each pump drains its
-        queue greedily and detokenizes incrementally; client disconnect
-        aborts every underlying request."""
+        queue greedily and detokenizes incrementally; stop STRINGS truncate
+        the stream at the match (token-id stops happen in the engine);
+        client disconnect aborts every underlying request."""
         q: asyncio.Queue = asyncio.Queue()
+        stops = stop_strings or []
+        tail_keep = max((len(x) for x in stops), default=0)
 
         async def pump(idx, st):
             detok = IncrementalDetokenizer(tokenizer)
+            emitted = ""  # tail window for cross-chunk stop-string hits
             first_t = None
             done = False
             while not done:
@@ -162,6 +167,17 @@ def create_app(
                     if o.finished:
                         finish = o.finish_reason or "stop"
                 delta = detok.push(new_ids) if new_ids else ""
+                if delta and stops:
+                    tail = emitted[-tail_keep:]
+                    hit = _find_stop(tail + delta, stops)
+                    if hit >= 0:
+                        # truncate at the match (a hit inside the already-
+                        # emitted tail keeps nothing more) and stop
+                        delta = delta[: max(0, hit - len(tail))]
+                        server.abort(st.rid)
+                        await q.put(("chunk", idx, delta, "stop"))
+                        break
+                    emitted += delta
                 if delta or finish is not None:
                     await q.put(("chunk", idx, delta, finish))
             await q.put(("end", idx, len(detok.ids), first_t))
@@ -254,7 +270,7 @@ def create_app(
 
             return StreamingResponse(
                 _sse_multi(streams, [st.rid for st in streams],
-                           make_chunk, make_usage),
+                           make_chunk, make_usage, stop_strings=sp.stop),
                 media_type="text/event-stream")
 
         results = await asyncio.gather(*(_collect(st, sp) for st in streams))
@@ -332,7 +348,7 @@ def create_app(
 
             return StreamingResponse(
                 _sse_multi(streams, [st.rid for st in streams],
-                           make_chunk, make_usage),
+                           make_chunk, make_usage, stop_strings=sp.stop),
                 media_type="text/event-stream")
 
         results = await asyncio.gather(*(_collect(st, sp) for st in streams))

@@ -374,3 +374,38 @@ def test_incremental_detok_bytelevel_bpe():
         detok = IncrementalDetokenizer(tw)
         pieces = [detok.push([i]) for i in ids]
         assert "".join(pieces) == tw.decode(ids)
+
+
+def test_streaming_stop_string(client):
+    """Stop strings truncate the SSE stream at the match (previously only
+    the non-streaming path matched them)."""
+    import json as _json
+
+    # discover the greedy text first
+    r = client.post("/v1/completions", json={
+        "prompt": [5, 9, 13], "max_tokens": 6, "temperature": 0.0,
+        "ignore_eos": True,
+    })
+    full = r.json()["choices"][0]["text"]
+    # stop at the 3rd token's text
+    probe = client.post("/v1/completions", json={
+        "prompt": [5, 9, 13], "max_tokens": 3, "temperature": 0.0,
+        "ignore_eos": True,
+    }).json()["choices"][0]["text"]
+    stop = probe[len(probe) // 2:]
+    text, finish = "", None
+    with client.stream("POST", "/v1/completions", json={
+        "prompt": [5, 9, 13], "max_tokens": 6, "temperature": 0.0,
+        "ignore_eos": True, "stream": True, "stop": [stop],
+    }) as r:
+        for line in r.iter_lines():
+            if not line.startswith("data: ") or line == "data: [DONE]":
+                continue
+            msg = _json.loads(line[6:])
+            for c in msg.get("choices", []):
+                text += c.get("text", "")
+                if c.get("finish_reason"):
+                    finish = c["finish_reason"]
+    assert finish == "stop"
+    assert stop not in text
+    assert full.startswith(text)
