@@ -177,7 +177,8 @@ def create_app(
                         server.abort(st.rid)
                         await q.put(("chunk", idx, delta, "stop"))
                         break
-                    emitted += delta
+                    emitted = (emitted + delta)[-tail_keep:] \
+                        if tail_keep else ""
                 if delta or finish is not None:
                     await q.put(("chunk", idx, delta, finish))
             await q.put(("end", idx, len(detok.ids), first_t))
