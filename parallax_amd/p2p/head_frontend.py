@@ -119,6 +119,24 @@ def create_head_app(
                     text = tokenizer.decode(token_ids)
                     delta, nl = text[sent_len:], len(text)
                     sent_len = nl
+                    if sp.stop:
+                        hit = _find_stop(text, sp.stop)
+                        if hit >= 0:
+                            # truncate at the stop string and terminate
+                            delta = text[min(sent_len - len(delta), hit):hit]
+                            agent.executor.abort(rid)
+                            chunk = {
+                                "id": rid,
+                                "object": "chat.completion.chunk",
+                                "created": created, "model": model_name,
+                                "choices": [{
+                                    "index": 0,
+                                    "delta": {"content": delta},
+                                    "finish_reason": "stop",
+                                }],
+                            }
+                            yield f"data: {_json.dumps(chunk)}\n\n"
+                            break
                     chunk = {
                         "id": rid, "object": "chat.completion.chunk",
                         "created": created, "model": model_name,
