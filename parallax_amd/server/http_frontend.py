@@ -231,7 +231,10 @@ def create_app(
         if not messages:
             raise HTTPException(400, "messages required")
         prompt_ids = tokenizer.chat_prompt_ids(messages)
-        sp = _params(body)
+        try:
+            sp = _params(body)
+        except (ValueError, TypeError) as e:  # bad sampling params -> 400
+            raise HTTPException(400, str(e))
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         n = max(1, int(body.get("n", 1)))
         try:
@@ -311,7 +314,10 @@ def create_app(
             prompt_ids = prompt
         else:
             prompt_ids = tokenizer.encode(prompt)
-        sp = _params(body)
+        try:
+            sp = _params(body)
+        except (ValueError, TypeError) as e:  # bad sampling params -> 400
+            raise HTTPException(400, str(e))
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         n = max(1, int(body.get("n", 1)))
         try:

@@ -409,3 +409,13 @@ def test_streaming_stop_string(client):
     assert finish == "stop"
     assert stop not in text
     assert full.startswith(text)
+
+
+def test_bad_params_are_400(client):
+    for body in (
+        {"prompt": [5], "max_tokens": 0},
+        {"prompt": [5], "max_tokens": -3},
+        {"prompt": [5], "temperature": "hot"},
+    ):
+        r = client.post("/v1/completions", json=body)
+        assert r.status_code == 400, body
