@@ -120,8 +120,11 @@ def create_app(
         path = body.get("model_path")
         if not path:
             raise HTTPException(400, "model_path required")
-        with server._lock:
-            n = server.engine.update_weights_from_disk(path)
+        try:
+            with server._lock:
+                n = server.engine.update_weights_from_disk(path)
+        except (FileNotFoundError, OSError, ValueError) as e:
+            raise HTTPException(400, f"refit failed: {e}")
         return {"updated_tensors": n}
 
     async def _sse_multi(streams, all_rids, make_chunk, make_usage,

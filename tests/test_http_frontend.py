@@ -419,3 +419,9 @@ def test_bad_params_are_400(client):
     ):
         r = client.post("/v1/completions", json=body)
         assert r.status_code == 400, body
+
+
+def test_update_weights_bad_path_is_400(client):
+    r = client.post("/update_weights",
+                    json={"model_path": "/nonexistent/ckpt"})
+    assert r.status_code == 400
