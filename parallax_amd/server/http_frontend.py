@@ -233,7 +233,10 @@ def create_app(
         messages = body.get("messages")
         if not messages:
             raise HTTPException(400, "messages required")
-        prompt_ids = tokenizer.chat_prompt_ids(messages)
+        try:
+            prompt_ids = tokenizer.chat_prompt_ids(messages)
+        except (TypeError, ValueError, KeyError, AttributeError) as e:
+            raise HTTPException(400, f"bad messages: {e}")
         try:
             sp = _params(body)
         except (ValueError, TypeError) as e:  # bad sampling params -> 400
@@ -313,10 +316,14 @@ def create_app(
         prompt = body.get("prompt")
         if prompt is None:
             raise HTTPException(400, "prompt required")
-        if isinstance(prompt, list) and prompt and isinstance(prompt[0], int):
-            prompt_ids = prompt
-        else:
-            prompt_ids = tokenizer.encode(prompt)
+        try:
+            if isinstance(prompt, list) and prompt \
+                    and isinstance(prompt[0], int):
+                prompt_ids = [int(t) for t in prompt]
+            else:
+                prompt_ids = tokenizer.encode(prompt)
+        except (TypeError, ValueError, AttributeError) as e:
+            raise HTTPException(400, f"bad prompt: {e}")
         try:
             sp = _params(body)
         except (ValueError, TypeError) as e:  # bad sampling params -> 400

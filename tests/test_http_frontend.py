@@ -425,3 +425,12 @@ def test_update_weights_bad_path_is_400(client):
     r = client.post("/update_weights",
                     json={"model_path": "/nonexistent/ckpt"})
     assert r.status_code == 400
+
+
+def test_garbage_prompts_are_400(client):
+    assert client.post("/v1/completions",
+                       json={"prompt": [1, "x", None]}).status_code == 400
+    assert client.post("/v1/chat/completions",
+                       json={"messages": ["not-a-dict"]}).status_code == 400
+    assert client.post("/v1/completions",
+                       json={"prompt": {"weird": 1}}).status_code == 400
