@@ -80,9 +80,15 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
     app = FastAPI(title="parallax_amd-scheduler", version="0.1.0")
     app.state.service = svc
 
+    def _require(body: dict, *keys):
+        missing = [k for k in keys if k not in body]
+        if missing:
+            raise HTTPException(400, f"missing field(s): {missing}")
+
     @app.post("/scheduler/init")
     async def scheduler_init(request: Request):
         body = await request.json()
+        _require(body, "hf_config")
         svc.init_model(
             body.get("model_name", "model"),
             body["hf_config"],
@@ -97,7 +103,8 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
         if svc.scheduler is None:
             raise HTTPException(503, "scheduler not initialized")
         body = await request.json()
-        hw = body.get("hardware", {})
+        _require(body, "node_id")
+        hw = body.get("hardware", {}) or {}
         node = Node(
             node_id=body["node_id"],
             hardware=NodeHardware(
@@ -127,6 +134,7 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
         if svc.scheduler is None:
             raise HTTPException(503, "scheduler not initialized")
         body = await request.json()
+        _require(body, "node_id")
         with svc._lock:
             assignment = svc.scheduler.node_update(
                 body["node_id"],
@@ -144,6 +152,7 @@ def create_backend_app(service: Optional[SchedulerService] = None) -> FastAPI:
     @app.post("/node/leave")
     async def node_leave(request: Request):
         body = await request.json()
+        _require(body, "node_id")
         with svc._lock:
             if svc.scheduler is not None:
                 svc.scheduler.node_leave(body["node_id"])
