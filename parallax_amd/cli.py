@@ -34,7 +34,8 @@ def cmd_run(args) -> None:
         with open(os.path.join(args.model_path, "config.json")) as f:
             hf_cfg = json.load(f)
         svc.init_model(args.model_name or args.model_path, hf_cfg,
-                       min_nodes=args.min_nodes)
+                       min_nodes=args.min_nodes, allocator=args.allocator,
+                       routing=args.routing)
     app = create_backend_app(svc)
     uvicorn.run(app, host=args.host, port=args.port)
 
@@ -252,6 +253,12 @@ def main(argv=None) -> None:
     pr.add_argument("--model-path", default=None)
     pr.add_argument("--model-name", default=None)
     pr.add_argument("--min-nodes", type=int, default=1)
+    pr.add_argument("--allocator", default="greedy",
+                    choices=["greedy", "dp"],
+                    help="layer allocator (greedy look-ahead or memoized DP)")
+    pr.add_argument("--routing", default="round_robin",
+                    choices=["round_robin", "random", "latency"],
+                    help="request routing across pipelines")
     pr.set_defaults(fn=cmd_run)
 
     pj = sub.add_parser("join", help="join a scheduler as a worker node")
