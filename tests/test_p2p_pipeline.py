@@ -298,3 +298,60 @@ def test_dead_peer_aborts_requests_not_node():
     assert finished[0].finish_reason == "abort"
     # head keeps serving: nothing left running
     assert not head.scheduler.running
+
+
+def test_head_frontend_stream_stop_string():
+    """The WAN head's SSE path truncates at a stop string and aborts the
+    executor request (fake agent, no network)."""
+    import asyncio
+
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from parallax_amd.p2p.head_frontend import create_head_app
+    from parallax_amd.p2p.peer_executor import PeerOutput
+    from parallax_amd.server.tokenizer_util import TokenizerWrapper
+
+    aborted = []
+
+    class FakeExec:
+        def abort(self, rid):
+            aborted.append(rid)
+
+    class FakeAgent:
+        node_id = "head"
+        executor = FakeExec()
+
+        def submit(self, prompt_ids, sp, routing, aio_loop=None,
+                   aio_queue=None):
+            # feed 5 tokens then a terminator into the asyncio queue
+            for i, t in enumerate([10, 11, 12, 13, 14]):
+                aio_loop.call_soon_threadsafe(
+                    aio_queue.put_nowait,
+                    PeerOutput("r1", t, i == 4,
+                               "length" if i == 4 else None))
+            aio_loop.call_soon_threadsafe(aio_queue.put_nowait, None)
+            return "r1", aio_queue
+
+    tok = TokenizerWrapper(vocab_size=64)
+    app = create_head_app(FakeAgent(), tok, "fake")
+    with TestClient(app) as c:
+        stop = tok.decode([12])  # third token's text
+        text, finish = "", None
+        with c.stream("POST", "/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "x"}],
+            "max_tokens": 5, "stream": True, "stop": [stop],
+        }) as r:
+            import json as _json
+
+            for line in r.iter_lines():
+                if not line.startswith("data: ") or line == "data: [DONE]":
+                    continue
+                msg = _json.loads(line[6:])
+                for ch in msg.get("choices", []):
+                    text += (ch.get("delta") or {}).get("content") or ""
+                    if ch.get("finish_reason"):
+                        finish = ch["finish_reason"]
+    assert finish == "stop"
+    assert stop not in text
+    assert aborted == ["r1"]
