@@ -169,3 +169,20 @@ def test_cluster_end_to_end():
         a.stop()
     backend_srv.should_exit = True
     head_srv.should_exit = True
+
+
+def test_dashboard_served():
+    """The scheduler serves the single-file dashboard at /."""
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from parallax_amd.backend.service import SchedulerService, create_backend_app
+
+    app = create_backend_app(SchedulerService())
+    with TestClient(app) as c:
+        r = c.get("/")
+        assert r.status_code == 200
+        for marker in ("cluster", "chat", "setup", "fetch(\"/cluster/status\")"):
+            assert marker in r.text
+        # uninitialized scheduler: status endpoint responds cleanly
+        assert c.get("/cluster/status").json()["initialized"] is False
