@@ -355,3 +355,48 @@ def test_head_frontend_stream_stop_string():
     assert finish == "stop"
     assert stop not in text
     assert aborted == ["r1"]
+
+
+try:
+    from hypothesis import HealthCheck, given, settings
+    from hypothesis import strategies as st
+    _HYP = True
+except ImportError:  # pragma: no cover
+    _HYP = False
+
+
+@pytest.mark.skipif(not _HYP, reason="hypothesis not installed")
+@settings(max_examples=40, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(rows=st.integers(min_value=0, max_value=9),
+       cols=st.integers(min_value=1, max_value=33),
+       pos=st.integers(min_value=0, max_value=100000),
+       tok=st.integers(min_value=-1, max_value=200000),
+       temp=st.sampled_from([0.0, 0.7, 1.0]),
+       prefill=st.booleans())
+def test_codec_roundtrip_property(rows, cols, pos, tok, temp, prefill):
+    """Wire codec: arbitrary packet shapes survive encode/decode exactly
+    (hidden-state bytes, positions, token ids, sampling params)."""
+    from parallax_amd.p2p import codec
+    from parallax_amd.server.request import IntermediateRequest
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    h = torch.randn(rows, cols) if rows else None
+    pkt = IntermediateRequest(
+        rid="x" * 8, current_position=pos, hidden_states=h,
+        next_token_id=tok if tok >= 0 else None,
+        routing_table=["a", "b"], is_prefill=prefill,
+        num_new_tokens=max(1, rows),
+        sampling_params=SamplingParams(temperature=temp, max_new_tokens=5),
+    )
+    out = codec.decode(codec.encode_forward([pkt]))
+    assert out["kind"] == "forward"
+    got = out["reqs"][0]
+    assert got.rid == pkt.rid and got.current_position == pos
+    assert got.next_token_id == pkt.next_token_id
+    assert got.is_prefill == prefill
+    assert got.sampling_params.temperature == temp
+    if h is None:
+        assert got.hidden_states is None or got.hidden_states.numel() == 0
+    else:
+        assert torch.equal(got.hidden_states, h)
