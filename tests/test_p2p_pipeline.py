@@ -400,3 +400,37 @@ def test_codec_roundtrip_property(rows, cols, pos, tok, temp, prefill):
         assert got.hidden_states is None or got.hidden_states.numel() == 0
     else:
         assert torch.equal(got.hidden_states, h)
+
+
+def test_tcp_transport_survives_garbage():
+    """Random bytes, oversized frame claims and wrong auth must not kill the
+    accept loop or leak into the inbox."""
+    import socket as sk
+    import time as _time
+
+    from parallax_amd.p2p.transport import TcpTransport
+
+    t = TcpTransport("victim", host="127.0.0.1", port=0,
+                     max_frame_bytes=1 << 20, auth_token="sekret")
+    try:
+        for payload in (b"\x00" * 3,                      # truncated header
+                        b"\xff" * 12,                     # absurd frame length
+                        b"AUTH" + (99999).to_bytes(4, "little") + b"x",
+                        b"GARBAGEGARBAGE"):
+            with sk.create_connection(("127.0.0.1", t.port), timeout=5) as c:
+                c.sendall(payload)
+                _time.sleep(0.05)
+        # a well-formed authed frame still arrives after all that
+        import struct
+
+        with sk.create_connection(("127.0.0.1", t.port), timeout=5) as c:
+            tok = b"sekret"
+            c.sendall(b"AUTH" + struct.pack("<I", len(tok)) + tok)
+            msg = b"hello"
+            c.sendall(struct.pack("<Q", len(msg)) + msg)
+            _time.sleep(0.2)
+        got = t.recv(timeout=5)
+        assert got == b"hello"
+        assert t.recv(timeout=0.1) is None  # nothing from the garbage
+    finally:
+        t.close() if hasattr(t, "close") else None
