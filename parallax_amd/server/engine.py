@@ -295,8 +295,15 @@ class Engine:
                 f"{sp.max_new_tokens} new tokens exceeds the engine context "
                 f"limit {limit} (max_model_len / max_position_embeddings)"
             )
+        rid = rid or new_request_id()
+        if rid in self.scheduler.running \
+                or any(r.rid == rid for r in self._pending_adds) \
+                or any(r.rid == rid for r in self.scheduler.wait_queue):
+            # a duplicate rid would collide in the cache-state and running
+            # maps and silently corrupt block accounting
+            raise ValueError(f"duplicate request id {rid!r}")
         req = InitialRequest(
-            rid=rid or new_request_id(),
+            rid=rid,
             prompt_token_ids=list(prompt_token_ids),
             sampling_params=sp,
         )

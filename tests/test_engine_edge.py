@@ -48,3 +48,36 @@ def test_request_timeout_sweep_aborts():
         if aborted or not eng.has_work:
             break
     assert aborted or not eng.has_work
+
+
+def test_duplicate_rid_rejected():
+    import torch
+
+    from parallax_amd.models.config import ModelConfig
+    from parallax_amd.parallel.comm import CommContext
+    from parallax_amd.server.engine import Engine, EngineArgs
+    from parallax_amd.server.sampling_params import SamplingParams
+
+    cfg = ModelConfig(
+        architecture="LlamaForCausalLM", vocab_size=128, hidden_size=32,
+        num_layers=1, num_heads=2, num_kv_heads=1, head_dim=16,
+        intermediate_size=64, max_position_embeddings=128, eos_token_ids=[],
+    )
+    comm = CommContext(world_size=1, rank=0, pp_size=1, tp_size=1,
+                       pp_rank=0, tp_rank=0, device=torch.device("cpu"))
+    eng = Engine(cfg, EngineArgs(block_size=8, num_kv_blocks=32,
+                                 dtype=torch.float32), comm=comm,
+                 random_weights=True)
+    sp = SamplingParams(temperature=0.0, max_new_tokens=2, ignore_eos=True)
+    eng.submit([3, 4], sp, rid="dup")
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        eng.submit([5, 6], sp, rid="dup")          # still pending
+    eng.step()
+    with _pytest.raises(ValueError):
+        eng.submit([5, 6], sp, rid="dup")          # now running
+    while eng.has_work:
+        eng.step()
+    eng.submit([5, 6], sp, rid="dup")              # finished: rid reusable
+    while eng.has_work:
+        eng.step()
