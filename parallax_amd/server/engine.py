@@ -296,9 +296,11 @@ class Engine:
                 f"limit {limit} (max_model_len / max_position_embeddings)"
             )
         rid = rid or new_request_id()
+        # list() snapshots are GIL-atomic; the step thread mutates these
+        # structures concurrently (lock-free submit)
         if rid in self.scheduler.running \
-                or any(r.rid == rid for r in self._pending_adds) \
-                or any(r.rid == rid for r in self.scheduler.wait_queue):
+                or any(r.rid == rid for r in list(self._pending_adds)) \
+                or any(r.rid == rid for r in list(self.scheduler.wait_queue)):
             # a duplicate rid would collide in the cache-state and running
             # maps and silently corrupt block accounting
             raise ValueError(f"duplicate request id {rid!r}")
