@@ -165,8 +165,12 @@ class PeerExecutor:
         assert self.is_head
         from ..server.request import new_request_id
 
+        rid = rid or new_request_id()
+        if rid in self.scheduler.running \
+                or any(r.rid == rid for r in list(self.scheduler.wait_queue)):
+            raise ValueError(f"duplicate request id {rid!r}")
         req = InitialRequest(
-            rid=rid or new_request_id(),
+            rid=rid,
             prompt_token_ids=list(prompt_token_ids),
             sampling_params=sampling_params,
             routing_table=list(routing_table),
